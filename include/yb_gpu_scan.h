@@ -1,0 +1,326 @@
+/*
+ * include/yb_gpu_scan.h — the drop-in C ABI for the MI355X-native DocDB
+ * SST-block scan-and-filter path.
+ *
+ * This ABI is the boundary a YugabyteDB tablet server would call through in
+ * place of the CPU DocRowwiseIterator row loop. Each entry point names the
+ * reference interface it replaces (paths relative to yugabyte/yugabyte-db):
+ *
+ *   yb_gpu_scan_open        ~ YQLStorageIf::GetIterator + DocRowwiseIterator
+ *                             ctor/Init  (src/yb/docdb/ql_storage_interface.h:
+ *                             37-120, doc_rowwise_iterator.h:52-81,
+ *                             doc_rowwise_iterator.cc:165-231)
+ *   yb_gpu_scan_feed_blocks ~ the BoundedRocksDbIterator data source — the
+ *                             already-flushed, decompressed data blocks a
+ *                             tablet scan walks (rocksdb/table/block.cc)
+ *   yb_gpu_scan_execute     ~ PgsqlReadOperation::ExecuteScalar row loop
+ *                             (src/yb/docdb/pgsql_operation.cc:2808-2931)
+ *   yb_gpu_scan_aggregate   ~ EvalAggregate/PopulateAggregate
+ *                             (src/yb/docdb/pgsql_operation.cc:3171-3186)
+ *   yb_gpu_scan_next_batch  ~ YQLRowwiseIteratorIf::PgFetchNext(PgTableRow*)
+ *                             batched (src/yb/docdb/
+ *                             ql_rowwise_iterator_interface.h:32-97)
+ *   yb_gpu_scan_paging_state~ SetPagingState (pgsql_operation.cc:2908-2922)
+ *
+ * Error convention mirrors yb::Status (src/yb/util/status_fwd.h): int code
+ * (0 = OK) + message via yb_gpu_last_error. No exceptions cross the ABI.
+ * Threading: one handle == one HIP stream, single caller per handle
+ * (CheckInitOnce, doc_rowwise_iterator.cc:128-135); different handles are
+ * independent (one per tablet/GPU).
+ *
+ * The same library also exports the synthetic tablet generator (the write
+ * path needed to build benchmark/parity datasets — BlockBuilder
+ * rocksdb/table/block_builder.cc, RowPacker dockv/packed_row.cc).
+ */
+#ifndef YB_GPU_SCAN_H
+#define YB_GPU_SCAN_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- shared descriptors -------------------------------------------------- */
+
+/* Value-column data types (DataType subset — src/yb/common/value.messages.h).
+ * Values match oracle/orcl.h orcl_dtype_t for test convenience. */
+typedef enum {
+  YBG_T_BOOL = 0,
+  YBG_T_INT8 = 1,
+  YBG_T_INT16 = 2,
+  YBG_T_INT32 = 3,
+  YBG_T_INT64 = 4,
+  YBG_T_UINT32 = 5,
+  YBG_T_UINT64 = 6,
+  YBG_T_FLOAT = 7,
+  YBG_T_DOUBLE = 8,
+  YBG_T_STRING = 9,
+} ybg_dtype_t;
+
+typedef enum {
+  YBG_KT_INT64 = 0,
+  YBG_KT_INT32 = 1,
+  YBG_KT_STRING = 2,
+} ybg_keytype_t;
+
+/* src/yb/rocksdb/types.h:50-56 KeyValueEncodingFormat */
+typedef enum {
+  YBG_ENC_SHARED_PREFIX = 0,
+  YBG_ENC_THREE_SHARED_PARTS = 1,
+} ybg_kv_format_t;
+
+#define YBG_MAX_COLS 32
+#define YBG_MAX_KEYCOLS 8
+#define YBG_MAX_PREDS 8
+#define YBG_MAX_AGGS 8
+#define YBG_MAX_HT 16
+
+typedef struct {
+  int32_t column_id;
+  int32_t dtype;    /* ybg_dtype_t */
+  int32_t nullable; /* affects V1 packing varlen-ness (schema_packing.cc:45-49) */
+} ybg_value_col_t;
+
+typedef struct {
+  int32_t has_hash;
+  int32_t num_hash_cols;
+  int32_t num_range_cols;
+  int32_t key_types[YBG_MAX_KEYCOLS]; /* ybg_keytype_t */
+  int32_t num_value_cols;
+  ybg_value_col_t value_cols[YBG_MAX_COLS];
+} ybg_schema_t;
+
+/* Encoded read-time limits (EncodedReadHybridTime —
+ * src/yb/docdb/intent_aware_iterator.h:61-77). */
+typedef struct {
+  uint8_t read[YBG_MAX_HT];         int32_t read_len;
+  uint8_t local_limit[YBG_MAX_HT];  int32_t local_limit_len;
+  uint8_t global_limit[YBG_MAX_HT]; int32_t global_limit_len;
+} ybg_read_time_t;
+
+/* Helper: build the three encoded limits from plain hybrid times
+ * (micros<<12|logical) with write_id = kMaxWriteId
+ * (intent_aware_iterator.cc:1446-1455). */
+void ybg_read_time_init(ybg_read_time_t *rt, uint64_t read_ht,
+                        uint64_t local_limit_ht, uint64_t global_limit_ht);
+
+typedef enum {
+  YBG_PRED_GT = 0, YBG_PRED_GE, YBG_PRED_LT, YBG_PRED_LE,
+  YBG_PRED_EQ, YBG_PRED_NE,
+} ybg_pred_op_t;
+
+typedef struct {
+  int32_t is_key_col;
+  int32_t col;
+  int32_t op;       /* ybg_pred_op_t */
+  uint64_t datum;   /* numeric rhs bit pattern (dtype of the column) */
+  const uint8_t *bytes; /* string rhs */
+  uint64_t bytes_len;
+} ybg_pred_t;
+
+typedef enum {
+  YBG_AGG_COUNT = 0,
+  YBG_AGG_COUNT_STAR = 1,
+  YBG_AGG_SUM_INT64 = 2,
+  YBG_AGG_SUM_DOUBLE = 3,
+  YBG_AGG_MIN_INT64 = 4,
+  YBG_AGG_MAX_INT64 = 5,
+  YBG_AGG_MIN_DOUBLE = 6,
+  YBG_AGG_MAX_DOUBLE = 7,
+} ybg_agg_op_t;
+
+typedef struct {
+  int32_t op;  /* ybg_agg_op_t */
+  int32_t col; /* value column index */
+} ybg_agg_t;
+
+/* Scan spec — the YQLScanSpec/DocPgsqlScanSpec surface reduced to what the
+ * hot path consumes (qlexpr/ql_scanspec.h:200-267 bounds;
+ * pgsql_operation.cc:602-668 predicates; :3171-3186 aggregates). */
+typedef struct {
+  ybg_schema_t schema;
+  int32_t kv_format; /* ybg_kv_format_t */
+  ybg_read_time_t read_time;
+  int32_t num_preds;
+  ybg_pred_t preds[YBG_MAX_PREDS];
+  int32_t num_aggs;
+  ybg_agg_t aggs[YBG_MAX_AGGS];
+  const uint8_t *lower_bound; uint64_t lower_bound_len; /* incl., encoded DocKey */
+  const uint8_t *upper_bound; uint64_t upper_bound_len; /* excl. */
+  int32_t emit_rows;  /* 1: materialize matching rows (next_batch) */
+  uint64_t row_limit; /* 0 = unlimited; else paging after this many rows */
+} ybg_scan_spec_t;
+
+/* ---- scan handle --------------------------------------------------------- */
+
+typedef struct ybg_scan ybg_scan_t;
+
+/* Last error message for the calling thread. */
+const char *yb_gpu_last_error(void);
+
+/* Open a scan. Returns 0 and sets *out on success; nonzero error code
+ * otherwise. Requires a visible MI355X (gfx950) device: this is the GPU
+ * product path — there is NO CPU fallback. */
+int yb_gpu_scan_open(const ybg_scan_spec_t *spec, ybg_scan_t **out);
+
+/* Feed the tablet's data blocks. blocks: concatenated raw block contents
+ * (uncompressed, no 5-byte file trailer); offsets[i]..offsets[i+1] delimit
+ * block i (offsets has n+1 entries). The memory is borrowed for the handle's
+ * lifetime. `device` nonzero means `blocks` is already a device pointer
+ * (resident in HBM); otherwise it is copied host->device once here. */
+int yb_gpu_scan_feed_blocks(ybg_scan_t *s, const uint8_t *blocks,
+                            const uint64_t *offsets, uint64_t n_blocks,
+                            int device);
+
+/* Run the scan asynchronously on the handle's stream. */
+int yb_gpu_scan_execute(ybg_scan_t *s);
+
+/* Block until the scan completes; returns status. */
+int yb_gpu_scan_wait(ybg_scan_t *s);
+
+typedef struct {
+  int64_t value_i64;
+  double value_f64;
+  int32_t is_null;
+  int32_t pad_;
+} ybg_agg_result_t;
+
+typedef struct {
+  uint64_t rows_scanned;  /* visible rows visited */
+  uint64_t rows_matched;  /* rows passing predicates */
+  uint64_t entries_seen;  /* KV entries decoded */
+  ybg_agg_result_t aggs[YBG_MAX_AGGS];
+} ybg_scan_result_t;
+
+/* Fetch aggregate results (implies wait). */
+int yb_gpu_scan_aggregate(ybg_scan_t *s, ybg_scan_result_t *out);
+
+/* Materialized row batch (PgTableRow analog — dockv/pg_row.h:91-179).
+ * Row order within the batch is by (sort_key) = the row's position in the
+ * tablet (block index << 32 | entry ordinal); rows are emitted unsorted —
+ * callers needing key order sort by sort_key. */
+typedef struct {
+  uint64_t n_rows;
+  uint64_t n_key_cols;
+  uint64_t n_value_cols;
+  const uint64_t *sort_key;   /* [n_rows] */
+  const uint64_t *key_datums; /* [n_rows * n_key_cols] */
+  const uint64_t *datums;     /* [n_rows * n_value_cols]; string cols: offset
+                                 into varlen heap (hi32 = len) */
+  const uint32_t *null_masks; /* [n_rows] bit i = value col i NULL */
+  const uint8_t *varlen;      /* varlen heap */
+  uint64_t varlen_size;
+} ybg_row_batch_t;
+
+/* Fetch the materialized matching rows (host-visible; implies wait).
+ * Valid until the next execute/close on this handle. */
+int yb_gpu_scan_next_batch(ybg_scan_t *s, ybg_row_batch_t *out);
+
+/* Resumable position after row_limit was hit: encoded DocKey of the next
+ * undelivered row (pgsql_operation.cc:2796-2806, 2908-2922). Returns length,
+ * 0 if the scan is complete. */
+int yb_gpu_scan_paging_state(ybg_scan_t *s, uint8_t *key_out, size_t cap,
+                             size_t *len_out);
+
+/* Kernel-time accounting for the last execute (HIP events on the handle's
+ * stream): total kernel ms and the dominant (decode) kernel's ms. */
+int yb_gpu_scan_kernel_ms(ybg_scan_t *s, double *total_ms, double *decode_ms);
+
+int yb_gpu_scan_close(ybg_scan_t *s);
+
+/* Returns 1 if a gfx950-class HIP device is visible. */
+int yb_gpu_available(void);
+
+/* ---- synthetic tablet generator (write path) ----------------------------- */
+
+typedef struct ybg_builder ybg_builder_t;
+
+/* Create a tablet builder. block_size_target: flush threshold in bytes
+ * (--db_block_size_bytes, dockv/packed_row.cc:39; configs use 4096).
+ * restart_interval: 16 (rocksdb/table.h:150). */
+ybg_builder_t *ybg_builder_create(const ybg_schema_t *schema, int kv_format,
+                                  size_t block_size_target,
+                                  int restart_interval);
+
+/* Row key for the builder: datums for int key cols, bytes for string cols. */
+typedef struct {
+  uint16_t hash;       /* kUInt16Hash prefix value (when schema.has_hash) */
+  uint64_t datums[YBG_MAX_KEYCOLS];
+  const uint8_t *strs[YBG_MAX_KEYCOLS];
+  uint64_t str_lens[YBG_MAX_KEYCOLS];
+} ybg_key_t;
+
+/* Column values for a packed row: numeric bit patterns / string slices;
+ * null[i] nonzero => NULL. */
+typedef struct {
+  uint64_t datums[YBG_MAX_COLS];
+  const uint8_t *strs[YBG_MAX_COLS];
+  uint64_t str_lens[YBG_MAX_COLS];
+  uint8_t null[YBG_MAX_COLS];
+} ybg_rowvals_t;
+
+/* Append entries IN KEY ORDER (caller guarantees DocDB ordering:
+ * user_key asc; for equal user keys impossible here since HT differs).
+ * ht = micros<<12|logical. seq: rocksdb sequence number for the internal
+ * key suffix (dbformat.h:84-110; initial seqno 1<<50 per
+ * docdb_rocksdb_util.cc:170). packed_version: 1 or 2. */
+int ybg_builder_add_packed_row(ybg_builder_t *b, const ybg_key_t *key,
+                               uint64_t ht, uint32_t write_id, uint64_t seq,
+                               int packed_version, const ybg_rowvals_t *vals);
+
+/* Column update entry ('K' + svarint(column_id) subkey; value = single V1
+ * value). null nonzero writes a column tombstone. */
+int ybg_builder_add_column_update(ybg_builder_t *b, const ybg_key_t *key,
+                                  int value_col_idx, uint64_t ht,
+                                  uint32_t write_id, uint64_t seq,
+                                  uint64_t datum, const uint8_t *str,
+                                  uint64_t str_len, int null);
+
+/* Row tombstone (bare key, value = kTombstone 'X'). */
+int ybg_builder_add_row_tombstone(ybg_builder_t *b, const ybg_key_t *key,
+                                  uint64_t ht, uint32_t write_id, uint64_t seq);
+
+/* Append a fully custom entry (tests): raw user key + raw value. The
+ * internal key suffix fixed64(seq<<8|kTypeValue) is appended here. */
+int ybg_builder_add_raw(ybg_builder_t *b, const uint8_t *user_key,
+                        size_t user_key_len, uint64_t seq,
+                        const uint8_t *value, size_t value_len);
+
+/* Finish: flush last block. Returns pointers valid until destroy. */
+int ybg_builder_finish(ybg_builder_t *b, const uint8_t **data,
+                       const uint64_t **offsets, uint64_t *n_blocks,
+                       uint64_t *total_bytes, uint64_t *n_entries);
+
+void ybg_builder_destroy(ybg_builder_t *b);
+
+/* One-call multi-threaded benchmark dataset generator.
+ * Produces `rows` rows of the given schema in one tablet: hash prefix walks
+ * 0..65535 monotonically, one int64 range key column ascending, packed-row
+ * (version as given) values seeded deterministically (seed). versions>1
+ * writes that many MVCC versions per row at increasing HTs (newest at
+ * ht_base + (versions-1)*ht_step). Caller frees *data with ybg_free. */
+typedef struct {
+  uint64_t rows;
+  uint64_t seed;
+  int packed_version;      /* 1 or 2 */
+  int kv_format;           /* ybg_kv_format_t */
+  uint32_t block_size;     /* e.g. 4096 */
+  int restart_interval;    /* 16 */
+  int versions;            /* MVCC versions per row (config 4: 5) */
+  uint64_t ht_base_micros; /* first version's HT physical micros */
+  uint64_t ht_step_micros; /* HT increment between versions */
+  int nthreads;            /* 0 = hw concurrency */
+} ybg_gen_params_t;
+
+int ybg_generate(const ybg_schema_t *schema, const ybg_gen_params_t *p,
+                 uint8_t **data, uint64_t **offsets, uint64_t *n_blocks,
+                 uint64_t *total_bytes, uint64_t *n_entries);
+
+void ybg_free(void *p);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* YB_GPU_SCAN_H */

@@ -1,0 +1,677 @@
+// yugabyte-db_amd/csrc/sstgen.cc — synthetic tablet (SST data block) writer:
+// the write-path subset needed to build benchmark and parity datasets.
+// Produces blocks in the reference's on-disk format:
+//   BlockBuilder           src/yb/rocksdb/table/block_builder.cc
+//   ThreeSharedParts sizes src/yb/rocksdb/table/block_builder_internal.h:100-239
+//   DocKey                 src/yb/dockv/doc_key.h:40-63
+//   DocHybridTime          src/yb/common/doc_hybrid_time.cc:39-76
+//   Packed rows V1/V2      src/yb/dockv/packed_row.cc:453-461, 523-544
+//   Single V1 values       src/yb/dockv/primitive_value.cc:1066-1125
+#include "../../include/yb_gpu_scan.h"
+#include "codec.h"
+
+#include <algorithm>
+#include <atomic>
+#include <cstdlib>
+#include <memory>
+#include <thread>
+
+namespace ybg {
+namespace {
+
+// ---------------------------------------------------------------------------
+// Block builder (restates rocksdb::BlockBuilder, block_builder.cc:63-412)
+// ---------------------------------------------------------------------------
+
+struct ComponentSizes {
+  size_t prev_ns1 = 0, ns1 = 0, middle = 0, prev_ns2 = 0, ns2 = 0;
+};
+
+// block_builder.cc:118-141 (FindMaxSharedSubstringAtTheSamePos). Note the
+// reference only records a run when it is TERMINATED by a mismatch — a run
+// reaching the end of the slice is not counted; we replicate that.
+static std::pair<size_t, size_t> MaxSharedRun(const uint8_t* a,
+                                              const uint8_t* b, size_t n) {
+  size_t max_size = 0, max_off = 0, run = 0;
+  for (size_t i = 0; i < n; ++i) {
+    if (a[i] == b[i]) {
+      ++run;
+    } else {
+      if (run > max_size) {
+        max_size = run;
+        max_off = i - run;
+      }
+      run = 0;
+    }
+  }
+  return {max_off, max_size};
+}
+
+// block_builder.cc:162-222 (FindMaxSharedMiddle)
+static ComponentSizes FindMaxSharedMiddle(const uint8_t* lhs, size_t lhs_size,
+                                          const uint8_t* rhs,
+                                          size_t rhs_size) {
+  size_t min_length;
+  std::pair<size_t, size_t> max_shared;
+  bool from_left = true;
+  if (lhs_size == rhs_size) {
+    min_length = rhs_size;
+    max_shared = MaxSharedRun(lhs, rhs, min_length);
+  } else {
+    const uint8_t* lhs_suffix;
+    const uint8_t* rhs_suffix;
+    if (lhs_size > rhs_size) {
+      min_length = rhs_size;
+      lhs_suffix = lhs + lhs_size - min_length;
+      rhs_suffix = rhs;
+    } else {
+      min_length = lhs_size;
+      lhs_suffix = lhs;
+      rhs_suffix = rhs + rhs_size - min_length;
+    }
+    max_shared = MaxSharedRun(lhs, rhs, min_length);
+    auto from_right = MaxSharedRun(lhs_suffix, rhs_suffix, min_length);
+    if (from_right.second > max_shared.second) {
+      from_left = false;
+      max_shared = from_right;
+    }
+  }
+  ComponentSizes r;
+  if (max_shared.second == 0) {
+    r.prev_ns1 = lhs_size;
+    r.ns1 = rhs_size;
+    return r;
+  }
+  if (from_left) {
+    size_t head = max_shared.first + max_shared.second;
+    r.prev_ns1 = max_shared.first;
+    r.ns1 = max_shared.first;
+    r.middle = max_shared.second;
+    r.prev_ns2 = lhs_size - head;
+    r.ns2 = rhs_size - head;
+  } else {
+    size_t mid_plus_ns2 = min_length - max_shared.first;
+    size_t ns2 = mid_plus_ns2 - max_shared.second;
+    r.prev_ns1 = lhs_size - mid_plus_ns2;
+    r.ns1 = rhs_size - mid_plus_ns2;
+    r.middle = max_shared.second;
+    r.prev_ns2 = ns2;
+    r.ns2 = ns2;
+  }
+  return r;
+}
+
+class BlockBuilder {
+ public:
+  BlockBuilder(int restart_interval, ybg_kv_format_t fmt)
+      : restart_interval_(restart_interval), fmt_(fmt) {
+    restarts_.push_back(0);
+  }
+
+  void Reset() {
+    buf_.clear();
+    restarts_.clear();
+    restarts_.push_back(0);
+    counter_ = 0;
+    last_key_.clear();
+  }
+
+  size_t SizeEstimate() const {
+    return buf_.size() + restarts_.size() * 4 + 4;
+  }
+  bool Empty() const { return buf_.empty(); }
+
+  // block_builder.cc:348-412 (Add)
+  void Add(const uint8_t* key, size_t key_size, const uint8_t* value,
+           size_t value_size) {
+    size_t shared_prefix = 0;
+    bool delta = false;
+    if (counter_ >= restart_interval_) {
+      restarts_.push_back((uint32_t)buf_.size());
+      counter_ = 0;
+    } else {
+      delta = true;
+      size_t min_len = std::min(last_key_.size(), key_size);
+      while (shared_prefix < min_len && last_key_[shared_prefix] == key[shared_prefix])
+        ++shared_prefix;
+    }
+
+    if (fmt_ == YBG_ENC_SHARED_PREFIX) {
+      // block_builder.cc:389-399
+      size_t non_shared = key_size - shared_prefix;
+      Leb128Append(shared_prefix, &buf_);
+      Leb128Append(non_shared, &buf_);
+      Leb128Append(value_size, &buf_);
+      buf_.insert(buf_.end(), key + shared_prefix, key + key_size);
+      buf_.insert(buf_.end(), value, value + value_size);
+    } else {
+      AddThreeSharedParts(key, key_size, value, value_size, shared_prefix,
+                          delta);
+    }
+
+    last_key_.assign(key, key + key_size);
+    ++counter_;
+  }
+
+  // block_builder.cc:337-346 (Finish)
+  void Finish() {
+    for (uint32_t r : restarts_) Fixed32LEAppend(r, &buf_);
+    Fixed32LEAppend((uint32_t)restarts_.size(), &buf_);
+  }
+
+  const Buf& data() const { return buf_; }
+  size_t entries() const { return n_entries_; }
+
+ private:
+  void AddThreeSharedParts(const uint8_t* key, size_t key_size,
+                           const uint8_t* value, size_t value_size,
+                           size_t shared_prefix, bool delta) {
+    ++n_entries_;
+    size_t reuse = 0;
+    bool inc = false;
+    ComponentSizes cs;
+    cs.prev_ns1 = last_key_.size();
+    cs.ns1 = key_size;
+    if (delta) {
+      size_t min_len = std::min(last_key_.size(), key_size);
+      // block_builder.cc:224-249 (CalculateLastInternalComponentReuse)
+      if (min_len >= shared_prefix + 8) {
+        uint64_t prev8, cur8;
+        memcpy(&prev8, last_key_.data() + last_key_.size() - 8, 8);
+        memcpy(&cur8, key + key_size - 8, 8);
+        if (cur8 == prev8 + 0x100) {
+          inc = true;
+          reuse = 8;
+        } else if (cur8 == prev8) {
+          reuse = 8;
+        }
+      }
+      cs = FindMaxSharedMiddle(last_key_.data() + shared_prefix,
+                               last_key_.size() - shared_prefix - reuse,
+                               key + shared_prefix,
+                               key_size - shared_prefix - reuse);
+    } else {
+      shared_prefix = 0;
+    }
+
+    // block_builder_internal.h:100-239 (EncodeThreeSharedPartsSizes)
+    int64_t ns1_delta = (int64_t)cs.ns1 - (int64_t)cs.prev_ns1;
+    int64_t ns2_delta = (int64_t)cs.ns2 - (int64_t)cs.prev_ns2;
+    bool frequent = reuse > 0 && cs.ns1 == 1 && cs.ns2 == 1 &&
+                    ns1_delta == 0 && ns2_delta == 0;
+    uint64_t encoded_1 =
+        ((uint64_t)value_size << 2) | ((uint64_t)inc << 1) | (uint64_t)frequent;
+    Leb128Append(encoded_1, &buf_);
+    if (frequent) {
+      Leb128Append(shared_prefix, &buf_);
+    } else {
+      bool reused = cs.ns1 < key_size;  // == shared_prefix+middle+reuse > 0
+      if (reused) {
+        if (reuse > 0 && ns1_delta == 0 && (ns2_delta == 0 || ns2_delta == 1) &&
+            cs.ns1 < 8 && cs.ns2 < 4) {
+          uint8_t encoded_2 = (uint8_t)(0b01 | ((ns2_delta == 1) << 2) |
+                                        (cs.ns1 << 3) | (cs.ns2 << 6));
+          buf_.push_back(encoded_2);
+        } else {
+          uint8_t encoded_2 = (uint8_t)(0b11 | ((reuse > 0) << 2) |
+                                        ((ns1_delta != 0) << 3) |
+                                        ((cs.ns2 != 0) << 4) |
+                                        ((ns2_delta != 0) << 5));
+          buf_.push_back(encoded_2);
+          Leb128Append(cs.ns1, &buf_);
+          if (ns1_delta != 0) SVarintAppend(ns1_delta, &buf_);
+          if (cs.ns2 != 0) Leb128Append(cs.ns2, &buf_);
+          if (ns2_delta != 0) SVarintAppend(ns2_delta, &buf_);
+        }
+        Leb128Append(shared_prefix, &buf_);
+      } else {
+        if (key_size > 0 && key_size < 128) {
+          buf_.push_back((uint8_t)(key_size << 1));
+        } else {
+          buf_.push_back(0);
+          Leb128Append(key_size, &buf_);
+        }
+      }
+    }
+    buf_.insert(buf_.end(), key + shared_prefix, key + shared_prefix + cs.ns1);
+    buf_.insert(buf_.end(), key + key_size - reuse - cs.ns2,
+                key + key_size - reuse);
+    buf_.insert(buf_.end(), value, value + value_size);
+  }
+
+  int restart_interval_;
+  ybg_kv_format_t fmt_;
+  Buf buf_;
+  std::vector<uint32_t> restarts_;
+  Buf last_key_;
+  int counter_ = 0;
+  size_t n_entries_ = 0;
+};
+
+// ---------------------------------------------------------------------------
+// DocKey / value encoding for a row
+// ---------------------------------------------------------------------------
+
+static void EncodeDocKey(const ybg_schema_t* sc, const ybg_key_t* k, Buf* out) {
+  int col = 0;
+  auto append_col = [&](int c) {
+    switch ((ybg_keytype_t)sc->key_types[c]) {
+      case YBG_KT_INT64:
+        out->push_back(kInt64Byte);
+        KeyInt64Append((int64_t)k->datums[c], out);
+        break;
+      case YBG_KT_INT32:
+        out->push_back(kInt32Byte);
+        KeyInt32Append((int32_t)k->datums[c], out);
+        break;
+      case YBG_KT_STRING:
+        out->push_back(kStringByte);
+        KeyStringAppend(k->strs[c], k->str_lens[c], out);
+        break;
+    }
+  };
+  if (sc->has_hash) {
+    out->push_back(kUInt16Hash);
+    out->push_back((uint8_t)(k->hash >> 8));
+    out->push_back((uint8_t)k->hash);
+    for (int i = 0; i < sc->num_hash_cols; ++i, ++col) append_col(col);
+    out->push_back(kGroupEnd);
+  }
+  for (int i = 0; i < sc->num_range_cols; ++i, ++col) append_col(col);
+  out->push_back(kGroupEnd);
+}
+
+// Single V1 value (primitive_value.cc:1066-1125). null => tombstone 'X'.
+static void EncodeV1Value(ybg_dtype_t dt, uint64_t datum, const uint8_t* str,
+                          uint64_t str_len, int null, Buf* out) {
+  if (null) {
+    out->push_back(kTombstoneByte);
+    return;
+  }
+  switch (dt) {
+    case YBG_T_BOOL:
+      out->push_back(datum ? kTrueByte : kFalseByte);
+      break;
+    case YBG_T_INT8:
+    case YBG_T_INT16:
+    case YBG_T_INT32:
+      out->push_back(kInt32Byte);
+      BE32Append((uint32_t)(int32_t)(int64_t)datum, out);
+      break;
+    case YBG_T_INT64:
+      out->push_back(kInt64Byte);
+      BE64Append(datum, out);
+      break;
+    case YBG_T_UINT32:
+      out->push_back(0x4F);  // 'O' kUInt32
+      BE32Append((uint32_t)datum, out);
+      break;
+    case YBG_T_UINT64:
+      out->push_back(0x55);  // 'U' kUInt64
+      BE64Append(datum, out);
+      break;
+    case YBG_T_FLOAT:
+      out->push_back(kFloatByte);
+      BE32Append((uint32_t)datum, out);
+      break;
+    case YBG_T_DOUBLE:
+      out->push_back(kDoubleByte);
+      BE64Append(datum, out);
+      break;
+    case YBG_T_STRING:
+      out->push_back(kStringByte);
+      out->insert(out->end(), str, str + str_len);
+      break;
+  }
+}
+
+static size_t V2FixedSize(ybg_dtype_t dt) {
+  switch (dt) {
+    case YBG_T_BOOL: case YBG_T_INT8: return 1;
+    case YBG_T_INT16: return 2;
+    case YBG_T_INT32: case YBG_T_UINT32: case YBG_T_FLOAT: return 4;
+    case YBG_T_INT64: case YBG_T_UINT64: case YBG_T_DOUBLE: return 8;
+    default: return 0;
+  }
+}
+
+// Packed row value body. V1: packed_row.cc:453-461 (Init) + ColumnPackerV1;
+// V2: packed_row.cc:523-544 (Init) + ColumnPackerV2 + Complete :577-597.
+static void EncodePackedRow(const ybg_schema_t* sc, int version,
+                            const ybg_rowvals_t* vals, Buf* out) {
+  if (version == 1) {
+    out->push_back(kPackedV1Byte);
+    UVarintAppend(0 /*schema version*/, out);
+    // varlen end-offset array: nullable or string columns are varlen
+    // (schema_packing.cc:45-49).
+    size_t hdr_start = out->size();
+    int nvarlen = 0;
+    for (int i = 0; i < sc->num_value_cols; ++i) {
+      if (sc->value_cols[i].nullable ||
+          sc->value_cols[i].dtype == YBG_T_STRING)
+        ++nvarlen;
+    }
+    out->resize(hdr_start + (size_t)nvarlen * 4);
+    size_t body_start = out->size();
+    int vi = 0;
+    for (int i = 0; i < sc->num_value_cols; ++i) {
+      ybg_dtype_t dt = (ybg_dtype_t)sc->value_cols[i].dtype;
+      bool varlen = sc->value_cols[i].nullable || dt == YBG_T_STRING;
+      if (!vals->null[i]) {
+        // V1 NULL columns append nothing (ColumnPackerV1::DoPackValue).
+        EncodeV1Value(dt, vals->datums[i], vals->strs[i], vals->str_lens[i],
+                      0, out);
+      }
+      if (varlen) {
+        uint32_t end = (uint32_t)(out->size() - body_start);
+        memcpy(out->data() + hdr_start + (size_t)vi * 4, &end, 4);
+        ++vi;
+      }
+    }
+  } else {
+    out->push_back(kPackedV2Byte);
+    UVarintAppend(0, out);
+    bool has_null = false;
+    for (int i = 0; i < sc->num_value_cols; ++i)
+      if (vals->null[i]) has_null = true;
+    uint8_t flags = has_null ? kV2HasNullsFlag : 0;
+    out->push_back(flags);
+    if (has_null) {
+      size_t mask_start = out->size();
+      out->resize(mask_start + (size_t)((sc->num_value_cols + 7) / 8), 0);
+      for (int i = 0; i < sc->num_value_cols; ++i)
+        if (vals->null[i])
+          (*out)[mask_start + (size_t)i / 8] |= (uint8_t)(1 << (i & 7));
+    }
+    for (int i = 0; i < sc->num_value_cols; ++i) {
+      if (vals->null[i]) continue;
+      ybg_dtype_t dt = (ybg_dtype_t)sc->value_cols[i].dtype;
+      size_t fs = V2FixedSize(dt);
+      if (fs) {
+        uint64_t u = vals->datums[i];
+        const uint8_t* p = reinterpret_cast<const uint8_t*>(&u);
+        out->insert(out->end(), p, p + fs);  // raw little-endian
+      } else {
+        FieldLengthAppend((uint32_t)vals->str_lens[i], out);
+        out->insert(out->end(), vals->strs[i], vals->strs[i] + vals->str_lens[i]);
+      }
+    }
+  }
+}
+
+}  // namespace
+}  // namespace ybg
+
+// ---------------------------------------------------------------------------
+// C ABI: builder
+// ---------------------------------------------------------------------------
+
+struct ybg_builder {
+  ybg_schema_t schema;
+  ybg_kv_format_t fmt;
+  size_t block_target;
+  int restart_interval;
+  ybg::BlockBuilder bb;
+  ybg::Buf all_blocks;
+  std::vector<uint64_t> offsets;
+  uint64_t n_entries = 0;
+  ybg::Buf key_scratch, value_scratch;
+
+  ybg_builder(const ybg_schema_t* sc, int f, size_t bt, int ri)
+      : schema(*sc), fmt((ybg_kv_format_t)f), block_target(bt),
+        restart_interval(ri), bb(ri, (ybg_kv_format_t)f) {
+    offsets.push_back(0);
+  }
+
+  void FlushBlock() {
+    if (bb.Empty()) return;
+    bb.Finish();
+    all_blocks.insert(all_blocks.end(), bb.data().begin(), bb.data().end());
+    offsets.push_back(all_blocks.size());
+    bb.Reset();
+  }
+
+  void AddInternal(const uint8_t* ukey, size_t ukey_len, uint64_t seq,
+                   const uint8_t* value, size_t value_len) {
+    // BlockBasedTableBuilder flushes when the block reaches the target size
+    // BEFORE adding the next key (table_builder Add/ShouldFlush semantics).
+    if (!bb.Empty() && bb.SizeEstimate() >= block_target) FlushBlock();
+    key_scratch.assign(ukey, ukey + ukey_len);
+    ybg::Fixed64LEAppend((seq << 8) | ybg::kTypeValue, &key_scratch);
+    bb.Add(key_scratch.data(), key_scratch.size(), value, value_len);
+    ++n_entries;
+  }
+};
+
+extern "C" {
+
+ybg_builder_t* ybg_builder_create(const ybg_schema_t* schema, int kv_format,
+                                  size_t block_size_target,
+                                  int restart_interval) {
+  return new ybg_builder(schema, kv_format, block_size_target,
+                         restart_interval);
+}
+
+int ybg_builder_add_packed_row(ybg_builder_t* b, const ybg_key_t* key,
+                               uint64_t ht, uint32_t write_id, uint64_t seq,
+                               int packed_version, const ybg_rowvals_t* vals) {
+  ybg::Buf k;
+  ybg::EncodeDocKey(&b->schema, key, &k);
+  k.push_back(ybg::kHybridTimeByte);
+  ybg::DocHtAppend(ht, write_id, &k);
+  b->value_scratch.clear();
+  ybg::EncodePackedRow(&b->schema, packed_version, vals, &b->value_scratch);
+  b->AddInternal(k.data(), k.size(), seq, b->value_scratch.data(),
+                 b->value_scratch.size());
+  return 0;
+}
+
+int ybg_builder_add_column_update(ybg_builder_t* b, const ybg_key_t* key,
+                                  int value_col_idx, uint64_t ht,
+                                  uint32_t write_id, uint64_t seq,
+                                  uint64_t datum, const uint8_t* str,
+                                  uint64_t str_len, int null) {
+  ybg::Buf k;
+  ybg::EncodeDocKey(&b->schema, key, &k);
+  k.push_back(ybg::kColByte);
+  ybg::SVarintAppend(b->schema.value_cols[value_col_idx].column_id, &k);
+  k.push_back(ybg::kHybridTimeByte);
+  ybg::DocHtAppend(ht, write_id, &k);
+  b->value_scratch.clear();
+  ybg::EncodeV1Value((ybg_dtype_t)b->schema.value_cols[value_col_idx].dtype,
+                     datum, str, str_len, null, &b->value_scratch);
+  b->AddInternal(k.data(), k.size(), seq, b->value_scratch.data(),
+                 b->value_scratch.size());
+  return 0;
+}
+
+int ybg_builder_add_row_tombstone(ybg_builder_t* b, const ybg_key_t* key,
+                                  uint64_t ht, uint32_t write_id,
+                                  uint64_t seq) {
+  ybg::Buf k;
+  ybg::EncodeDocKey(&b->schema, key, &k);
+  k.push_back(ybg::kHybridTimeByte);
+  ybg::DocHtAppend(ht, write_id, &k);
+  uint8_t tomb = ybg::kTombstoneByte;
+  b->AddInternal(k.data(), k.size(), seq, &tomb, 1);
+  return 0;
+}
+
+int ybg_builder_add_raw(ybg_builder_t* b, const uint8_t* user_key,
+                        size_t user_key_len, uint64_t seq,
+                        const uint8_t* value, size_t value_len) {
+  b->AddInternal(user_key, user_key_len, seq, value, value_len);
+  return 0;
+}
+
+int ybg_builder_finish(ybg_builder_t* b, const uint8_t** data,
+                       const uint64_t** offsets, uint64_t* n_blocks,
+                       uint64_t* total_bytes, uint64_t* n_entries) {
+  b->FlushBlock();
+  *data = b->all_blocks.data();
+  *offsets = b->offsets.data();
+  *n_blocks = b->offsets.size() - 1;
+  *total_bytes = b->all_blocks.size();
+  *n_entries = b->n_entries;
+  return 0;
+}
+
+void ybg_builder_destroy(ybg_builder_t* b) { delete b; }
+
+void ybg_read_time_init(ybg_read_time_t* rt, uint64_t read_ht,
+                        uint64_t local_limit_ht, uint64_t global_limit_ht) {
+  rt->read_len =
+      (int32_t)ybg::DocHtEncode(read_ht, 0xffffffffu, rt->read);
+  rt->local_limit_len =
+      (int32_t)ybg::DocHtEncode(local_limit_ht, 0xffffffffu, rt->local_limit);
+  rt->global_limit_len =
+      (int32_t)ybg::DocHtEncode(global_limit_ht, 0xffffffffu, rt->global_limit);
+}
+
+// ---------------------------------------------------------------------------
+// Multi-threaded benchmark generator
+// ---------------------------------------------------------------------------
+
+static inline uint64_t splitmix64(uint64_t& x) {
+  x += 0x9E3779B97f4A7C15ull;
+  uint64_t z = x;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  return z ^ (z >> 31);
+}
+
+int ybg_generate(const ybg_schema_t* schema, const ybg_gen_params_t* p,
+                 uint8_t** data, uint64_t** offsets, uint64_t* n_blocks,
+                 uint64_t* total_bytes, uint64_t* n_entries) {
+  int nthreads = p->nthreads > 0 ? p->nthreads
+                                 : (int)std::thread::hardware_concurrency();
+  if (nthreads < 1) nthreads = 1;
+  uint64_t rows = p->rows;
+  if (rows < (uint64_t)nthreads * 64) nthreads = 1;
+
+  struct Shard {
+    std::unique_ptr<ybg_builder> b;
+    uint64_t rows = 0;
+  };
+  std::vector<Shard> shards((size_t)nthreads);
+  int versions = p->versions < 1 ? 1 : p->versions;
+
+  auto worker = [&](int t) {
+    uint64_t r0 = rows * (uint64_t)t / (uint64_t)nthreads;
+    uint64_t r1 = rows * (uint64_t)(t + 1) / (uint64_t)nthreads;
+    auto b = std::make_unique<ybg_builder>(schema, p->kv_format,
+                                           p->block_size ? p->block_size : 4096,
+                                           p->restart_interval
+                                               ? p->restart_interval
+                                               : 16);
+    ybg_key_t key;
+    memset(&key, 0, sizeof(key));
+    ybg_rowvals_t vals;
+    memset(&vals, 0, sizeof(vals));
+    uint8_t strbufs[YBG_MAX_COLS][64];
+    uint64_t seq = ybg::kInitialSeqno + r0 * (uint64_t)versions;
+    for (uint64_t r = r0; r < r1; ++r) {
+      // Hash walks monotonically so emission order == DocDB key order.
+      key.hash = (uint16_t)((r * 65536ull) / (rows ? rows : 1));
+      int kc = 0;
+      for (int c = 0; c < schema->num_hash_cols + schema->num_range_cols;
+           ++c, ++kc) {
+        key.datums[kc] = r;  // row ordinal as the key column value
+      }
+      uint64_t vstate = p->seed * 0x9E3779B97f4A7C15ull + r * 2654435761ull;
+      for (int i = 0; i < schema->num_value_cols; ++i) {
+        ybg_dtype_t dt = (ybg_dtype_t)schema->value_cols[i].dtype;
+        uint64_t rv = splitmix64(vstate);
+        vals.null[i] = 0;
+        if (dt == YBG_T_STRING) {
+          // 32-byte lowercase alnum
+          uint64_t s0 = rv, s1 = splitmix64(vstate);
+          for (int j = 0; j < 32; ++j) {
+            uint64_t v = (j < 16 ? s0 : s1) >> ((j & 15) * 4);
+            strbufs[i][j] = (uint8_t)('a' + (v & 15));
+          }
+          vals.strs[i] = strbufs[i];
+          vals.str_lens[i] = 32;
+        } else if (dt == YBG_T_DOUBLE) {
+          double d = (double)(rv >> 24) / (double)(1ull << 40);
+          memcpy(&vals.datums[i], &d, 8);
+        } else if (dt == YBG_T_FLOAT) {
+          float f = (float)(rv >> 40) / (float)(1ull << 24);
+          uint32_t u;
+          memcpy(&u, &f, 4);
+          vals.datums[i] = u;
+        } else {
+          vals.datums[i] = rv & ((1ull << 40) - 1);  // uniform in [0, 2^40)
+        }
+      }
+      // MVCC versions, newest first (encoded DocHybridTime sorts newest
+      // first — doc_hybrid_time.h:89-95). Only the newest version's values
+      // are generated above; older versions get distinct values derived from
+      // the version index so visibility mistakes change results.
+      for (int v = versions - 1; v >= 0; --v) {
+        // Distinct, monotonically increasing HT per row: physical micros walk
+        // up every 4096 rows, the logical component distinguishes rows within
+        // that window (hybrid_time.h:70-99). Versions are ht_step apart.
+        uint64_t ht_micros =
+            p->ht_base_micros + (uint64_t)v * p->ht_step_micros + (r >> 12);
+        uint64_t ht = (ht_micros << 12) | (r & 0xfff);
+        // Older versions carry value = newest + age*1000003 so that a
+        // visibility mistake changes aggregate results deterministically.
+        uint64_t age = (uint64_t)(versions - 1 - v);
+        ybg_rowvals_t vv = vals;
+        if (age) {
+          for (int i = 0; i < schema->num_value_cols; ++i) {
+            if ((ybg_dtype_t)schema->value_cols[i].dtype != YBG_T_STRING)
+              vv.datums[i] = vals.datums[i] + age * 1000003ull;
+          }
+        }
+        ybg_builder_add_packed_row(b.get(), &key, ht, 0, seq++,
+                                   p->packed_version, &vv);
+      }
+    }
+    shards[(size_t)t].b = std::move(b);
+    shards[(size_t)t].rows = r1 - r0;
+  };
+
+  std::vector<std::thread> threads;
+  for (int t = 0; t < nthreads; ++t) threads.emplace_back(worker, t);
+  for (auto& th : threads) th.join();
+
+  // stitch shards
+  uint64_t total = 0, nb = 0, ne = 0;
+  std::vector<std::pair<const uint8_t*, const uint64_t*>> parts;
+  std::vector<std::pair<uint64_t, uint64_t>> sizes;  // (bytes, nblocks)
+  for (auto& s : shards) {
+    const uint8_t* d;
+    const uint64_t* off;
+    uint64_t nblk, bytes, ent;
+    ybg_builder_finish(s.b.get(), &d, &off, &nblk, &bytes, &ent);
+    parts.push_back({d, off});
+    sizes.push_back({bytes, nblk});
+    total += bytes;
+    nb += nblk;
+    ne += ent;
+  }
+  uint8_t* out = (uint8_t*)malloc(total ? total : 1);
+  uint64_t* out_off = (uint64_t*)malloc((nb + 1) * sizeof(uint64_t));
+  uint64_t pos = 0, bi = 0;
+  out_off[0] = 0;
+  for (size_t s = 0; s < parts.size(); ++s) {
+    memcpy(out + pos, parts[s].first, sizes[s].first);
+    for (uint64_t k = 1; k <= sizes[s].second; ++k) {
+      out_off[bi + k] = pos + parts[s].second[k];
+    }
+    bi += sizes[s].second;
+    pos += sizes[s].first;
+  }
+  *data = out;
+  *offsets = out_off;
+  *n_blocks = nb;
+  *total_bytes = total;
+  *n_entries = ne;
+  return 0;
+}
+
+void ybg_free(void* p) { free(p); }
+
+}  // extern "C"
