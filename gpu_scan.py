@@ -1,0 +1,90 @@
+"""GpuScan — thin Python wrapper over the product C ABI
+(include/yb_gpu_scan.h). This is the PRODUCT path: it requires an MI355X
+(gfx950) and fails loudly when no HIP device or extension is present — there
+is no CPU fallback."""
+import ctypes as C
+
+import ybgpu as y
+
+
+class GpuScanError(RuntimeError):
+    pass
+
+
+def _lib():
+    lib = y.product()
+    f = lib.yb_gpu_scan_open
+    f.restype = C.c_int
+    f.argtypes = [C.POINTER(y.ScanSpec), C.POINTER(C.c_void_p)]
+    lib.yb_gpu_scan_feed_blocks.restype = C.c_int
+    lib.yb_gpu_scan_feed_blocks.argtypes = [
+        C.c_void_p, C.POINTER(C.c_uint8), C.POINTER(C.c_uint64), C.c_uint64,
+        C.c_int]
+    lib.yb_gpu_scan_execute.restype = C.c_int
+    lib.yb_gpu_scan_execute.argtypes = [C.c_void_p]
+    lib.yb_gpu_scan_wait.restype = C.c_int
+    lib.yb_gpu_scan_wait.argtypes = [C.c_void_p]
+    lib.yb_gpu_scan_aggregate.restype = C.c_int
+    lib.yb_gpu_scan_aggregate.argtypes = [C.c_void_p,
+                                          C.POINTER(y.ScanResult)]
+    lib.yb_gpu_scan_kernel_ms.restype = C.c_int
+    lib.yb_gpu_scan_kernel_ms.argtypes = [C.c_void_p, C.POINTER(C.c_double),
+                                          C.POINTER(C.c_double)]
+    lib.yb_gpu_scan_close.restype = C.c_int
+    lib.yb_gpu_scan_close.argtypes = [C.c_void_p]
+    lib.yb_gpu_last_error.restype = C.c_char_p
+    lib.yb_gpu_available.restype = C.c_int
+    return lib
+
+
+def gpu_available():
+    return bool(_lib().yb_gpu_available())
+
+
+class GpuScan:
+    def __init__(self, spec):
+        self._lib = _lib()
+        self._h = C.c_void_p()
+        self._spec = spec  # keep alive (aux pointers)
+        rc = self._lib.yb_gpu_scan_open(C.byref(spec), C.byref(self._h))
+        if rc:
+            raise GpuScanError(
+                f"yb_gpu_scan_open rc={rc}: "
+                f"{self._lib.yb_gpu_last_error().decode()}")
+
+    def _check(self, rc, what):
+        if rc:
+            raise GpuScanError(
+                f"{what} rc={rc}: {self._lib.yb_gpu_last_error().decode()}")
+
+    def feed_blocks_host(self, data, offsets, n_blocks, total_bytes):
+        del total_bytes
+        self._check(
+            self._lib.yb_gpu_scan_feed_blocks(self._h, data, offsets,
+                                              n_blocks, 0),
+            "feed_blocks")
+
+    def execute(self):
+        self._check(self._lib.yb_gpu_scan_execute(self._h), "execute")
+
+    def wait(self):
+        self._check(self._lib.yb_gpu_scan_wait(self._h), "wait")
+
+    def aggregates(self):
+        res = y.ScanResult()
+        self._check(self._lib.yb_gpu_scan_aggregate(self._h, C.byref(res)),
+                    "aggregate")
+        return res
+
+    def kernel_ms(self):
+        total = C.c_double()
+        decode = C.c_double()
+        self._check(
+            self._lib.yb_gpu_scan_kernel_ms(self._h, C.byref(total),
+                                            C.byref(decode)), "kernel_ms")
+        return total.value, decode.value
+
+    def close(self):
+        if self._h:
+            self._lib.yb_gpu_scan_close(self._h)
+            self._h = None
