@@ -1,0 +1,271 @@
+#!/usr/bin/env python3
+"""bench.py — flagship benchmark: the DocDB SST-block scan-and-filter hot
+path on MI355X (BASELINE.json metric: scanned rows/s, 100M-row filtered
+aggregate).
+
+One "step" = one pass of the hot path over the synthetic tablet set: GPU
+block-decode + MVCC visibility + 3 int64 predicates + SUM(int64)/COUNT, plus
+the cross-tablet aggregate merge (RCCL all-reduce when world_size > 1).
+
+N=1 workload = BASELINE configs[1]: 1 tablet, 100M rows, 4KB SST blocks,
+three_shared_parts encoding, packed-row V2, 3 int64 predicates + SUM(int64).
+N>1 = configs[2]: 8 tablets x 12.5M rows sharded tablet->rank (strong
+scaling), final RCCL all-reduce of the partial aggregates over xGMI.
+
+Inputs are generated on host (seed 42) and resident in HBM before the timed
+region. The CPU baseline leg times the oracle (oracle/ — test infrastructure)
+on a bounded sample on this host; it is a reported baseline, not the target.
+"""
+import argparse
+import ctypes
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+HBM_PEAK_GBS = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, file=sys.stderr, flush=True)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--rows", type=int, default=100_000_000,
+                    help="total rows across all tablets")
+    ap.add_argument("--cpu-sample-seconds", type=float, default=10.0)
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = world if world > 1 else args.gpus
+
+    import ybgpu as y
+    import gpu_scan
+
+    dist = None
+    torch = None
+    if world > 1:
+        import torch  # noqa
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group(
+            backend="nccl" if torch.cuda.is_available() else "gloo")
+        if torch.cuda.is_available():
+            torch.cuda.set_device(local_rank)
+        gpu_scan._lib().yb_gpu_set_device(local_rank)
+
+    if not gpu_scan.gpu_available():
+        log("FATAL: no HIP device — the product path has no CPU fallback")
+        sys.exit(2)
+
+    # ---- dataset ---------------------------------------------------------
+    schema = y.make_schema([y.KT_INT64],
+                           [(10 + i, y.T_INT64, 1) for i in range(4)])
+    n_tablets = 1 if n_gpus == 1 and world <= 1 else 8
+    rows_per_tablet = args.rows // n_tablets
+    my_tablets = [t for t in range(n_tablets) if t % max(world, 1) == rank]
+
+    preds = [y.Pred(0, 0, y.PRED_GT, 1 << 39, None, 0),
+             y.Pred(0, 1, y.PRED_LT, 3 << 38, None, 0),
+             y.Pred(0, 2, y.PRED_GE, 1 << 36, None, 0)]
+    aggs = [y.Agg(y.AGG_SUM_INT64, 3), y.Agg(y.AGG_COUNT_STAR, 0)]
+
+    def make_spec():
+        spec = y.ScanSpec()
+        spec.schema = schema
+        spec.kv_format = y.ENC_THREE_SHARED_PARTS
+        spec.read_time = y.read_time(1_700_000_000_000_000)
+        spec.num_preds = len(preds)
+        for i, p in enumerate(preds):
+            spec.preds[i] = p
+        spec.num_aggs = len(aggs)
+        for i, a in enumerate(aggs):
+            spec.aggs[i] = a
+        return spec
+
+    scans = []
+    total_bytes_local = 0
+    gen_t0 = time.time()
+    first_tablet_data = None
+    for t in my_tablets:
+        data, offsets, nb, total, ne = y.generate(
+            schema, rows=rows_per_tablet, seed=42 + t)
+        total_bytes_local += total
+        s = gpu_scan.GpuScan(make_spec())
+        s.feed_blocks_host(data, offsets, nb, total)
+        scans.append(s)
+        if first_tablet_data is None:
+            first_tablet_data = (data, offsets, nb, total, ne)
+        else:
+            y.product().ybg_free(data)
+            y.product().ybg_free(offsets)
+    log(f"[rank {rank}] generated+uploaded {len(my_tablets)} tablet(s), "
+        f"{total_bytes_local/1e9:.2f} GB in {time.time()-gen_t0:.1f}s "
+        f"({total_bytes_local/max(rows_per_tablet*len(my_tablets),1):.1f} "
+        f"bytes/row)")
+
+    agg_buf = None
+    if dist is not None:
+        import torch
+        agg_buf = torch.zeros(2, dtype=torch.float64,
+                              device="cuda" if torch.cuda.is_available()
+                              else "cpu")
+
+    def step():
+        for s in scans:
+            s.execute()
+        results = [s.aggregates() for s in scans]
+        total_sum = sum(r.aggs[0].value_i64 for r in results)
+        total_cnt = sum(r.aggs[1].value_i64 for r in results)
+        if dist is not None:
+            agg_buf[0] = float(total_sum)
+            agg_buf[1] = float(total_cnt)
+            dist.all_reduce(agg_buf)  # RCCL over xGMI: the one collective
+            total_sum = int(agg_buf[0].item())
+            total_cnt = int(agg_buf[1].item())
+        return total_sum, total_cnt
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+            import torch
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+
+    # ---- warmup ----------------------------------------------------------
+    for _ in range(args.warmup):
+        step()
+    barrier_sync()
+
+    # ---- timed region ----------------------------------------------------
+    t0 = time.time()
+    checksum = None
+    kernel_decode_ms = 0.0
+    kernel_total_ms = 0.0
+    for _ in range(args.steps):
+        checksum = step()
+        for s in scans:
+            tot, dec = s.kernel_ms()
+            kernel_total_ms += tot
+            kernel_decode_ms += dec
+    barrier_sync()
+    elapsed = time.time() - t0
+    if dist is not None:
+        import torch
+        e = torch.tensor([elapsed], dtype=torch.float64,
+                         device=agg_buf.device)
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    total_rows = rows_per_tablet * n_tablets
+    rows_per_s = total_rows * args.steps / elapsed
+    ms_per_step = elapsed * 1000.0 / args.steps
+
+    # roofline: dominant kernel = k_scan (block decode+filter+aggregate).
+    # achieved = algorithmic bytes per launch / avg launch duration; one
+    # launch per tablet per step, algorithmic bytes = that tablet's
+    # serialized block bytes (every block byte crosses HBM once).
+    n_launches = len(scans) * args.steps
+    avg_decode_ms = kernel_decode_ms / max(n_launches, 1)
+    bytes_per_launch = total_bytes_local / max(len(scans), 1)
+    achieved_gbs = (bytes_per_launch / (avg_decode_ms / 1e3)) / 1e9 \
+        if avg_decode_ms > 0 else 0.0
+    traffic = os.environ.get("YBG_TRAFFIC_BYTES_PER_LAUNCH")
+    traffic = float(traffic) if traffic else None
+
+    # ---- CPU baseline (oracle, rank 0, N=1 only) -------------------------
+    cpu_baseline = None
+    if rank == 0 and world <= 1 and not args.skip_cpu_baseline:
+        data, offsets, nb, total, ne = first_tablet_data
+        # calibrate on 256 blocks, then size the sample for the budget
+        osc = y.orcl_schema_from(schema)
+        ospec = y.OrclScanSpec()
+        ospec.read_time = y.orcl_read_time(1_700_000_000_000_000)
+        ospec.num_preds = len(preds)
+        for i, p in enumerate(preds):
+            ospec.preds[i] = y.OrclPred(p.is_key_col, p.col, p.op, p.datum,
+                                        p.bytes, p.bytes_len)
+        ospec.num_aggs = len(aggs)
+        for i, a in enumerate(aggs):
+            ospec.aggs[i] = y.OrclAgg(a.op, a.col)
+
+        def run_orcl(nblocks):
+            t = time.time()
+            res, _ = y.orcl_scan(data, offsets, nblocks, osc, ospec)
+            return time.time() - t, res
+
+        cal_blocks = min(nb, 256)
+        cal_t, cal_res = run_orcl(cal_blocks)
+        per_block = max(cal_t / cal_blocks, 1e-9)
+        sample_blocks = min(nb, max(cal_blocks,
+                                    int(args.cpu_sample_seconds / per_block)))
+        st, sres = run_orcl(sample_blocks)
+        sample_rows = sres.rows_scanned
+        cpu_rows_per_s = sample_rows / st
+        cpu_baseline = {
+            "value": cpu_rows_per_s,
+            "unit": "rows/s",
+            "cores": 1,
+            "kind": "port",
+            "sample": f"{sample_blocks} of {nb} blocks "
+                      f"({sample_rows} rows, {st:.1f}s, single thread)",
+        }
+        log(f"[cpu baseline] {cpu_rows_per_s/1e6:.2f} Mrows/s "
+            f"(single-thread oracle)")
+
+    if rank == 0:
+        out = {
+            "metric": "scanned_rows_per_s",
+            "value": rows_per_s,
+            "unit": "rows/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "int64",
+            "data": "synthetic",
+            "config": {
+                "workload": ("1 tablet, 100M rows, 4KB SST blocks, "
+                             "three_shared_parts, packed-row V2, 3 int64 "
+                             "predicates + SUM(int64)" if n_tablets == 1 else
+                             "8 tablets x 12.5M rows, same query, "
+                             "sharded tablet->rank + RCCL all-reduce"),
+                "rows": total_rows,
+                "tablets": n_tablets,
+                "block_size": 4096,
+                "checksum": {"sum": checksum[0], "count": checksum[1]},
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": achieved_gbs,
+                "peak": HBM_PEAK_GBS,
+                "unit": "GB/s",
+                "frac": achieved_gbs / HBM_PEAK_GBS,
+                "traffic": traffic,
+            },
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(out), flush=True)
+
+    for s in scans:
+        s.close()
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -233,6 +233,9 @@ int yb_gpu_scan_close(ybg_scan_t *s);
 /* Returns 1 if a gfx950-class HIP device is visible. */
 int yb_gpu_available(void);
 
+/* Select the HIP device subsequent handles bind to (one rank per GPU). */
+int yb_gpu_set_device(int device);
+
 /* ---- synthetic tablet generator (write path) ----------------------------- */
 
 typedef struct ybg_builder ybg_builder_t;

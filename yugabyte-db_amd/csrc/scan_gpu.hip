@@ -1405,6 +1405,11 @@ int yb_gpu_available(void) {
   return n > 0;
 }
 
+int yb_gpu_set_device(int device) {
+  HIP_TRY(hipSetDevice(device));
+  return 0;
+}
+
 int yb_gpu_scan_open(const ybg_scan_spec_t* spec, ybg_scan_t** out) {
   int ndev = 0;
   if (hipGetDeviceCount(&ndev) != hipSuccess || ndev == 0)
