@@ -1,0 +1,252 @@
+"""Shared parity scenarios: each case builds a tablet (via the product
+generator/builder) and a list of scan runs. Consumed by
+ - test_sim_parity.py  (CPU: host simulator of the device algorithm vs oracle)
+ - test_gpu_parity.py  (GPU: HIP kernels vs oracle)
+"""
+import ctypes as C
+
+import ybgpu as y
+
+SCHEMA_4I = y.make_schema([y.KT_INT64],
+                          [(10 + i, y.T_INT64, 1) for i in range(4)])
+
+_KEEP = []
+
+
+def _case(name, schema, built, runs, kv_format=y.ENC_THREE_SHARED_PARTS):
+    data, offsets, nb, total = built[0], built[1], built[2], built[3]
+    return {
+        "name": name,
+        "schema": schema,
+        "data": data,
+        "offsets": offsets,
+        "n_blocks": nb,
+        "total": total,
+        "runs": runs,
+        "kv_format": kv_format,
+    }
+
+
+def build_cases():
+    cases = []
+
+    # config #2 shape: 3 int64 predicates + aggregates
+    built = y.generate(SCHEMA_4I, rows=200_000, seed=42)
+    preds = [y.Pred(0, 0, y.PRED_GT, 1 << 39, None, 0),
+             y.Pred(0, 1, y.PRED_LT, 3 << 38, None, 0),
+             y.Pred(0, 2, y.PRED_GE, 1 << 36, None, 0)]
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 3),
+            y.Agg(y.AGG_MIN_INT64, 3), y.Agg(y.AGG_MAX_INT64, 2)]
+    cases.append(_case("config2_filtered_sum", SCHEMA_4I, built,
+                       [(1_700_000_000_000_000, preds, aggs)]))
+
+    # config #4 shape: MVCC 5 versions, read times at version boundaries
+    built = y.generate(SCHEMA_4I, rows=20_000, versions=5,
+                       ht_base_micros=1_600_000_000_000_000,
+                       ht_step_micros=1_000_000_000)
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 0)]
+    cases.append(_case("mvcc_5_versions", SCHEMA_4I, built,
+                       [(rm, (), aggs) for rm in
+                        (1_599_999_999_000_000, 1_600_000_500_000_000,
+                         1_602_000_000_500_000, 1_700_000_000_000_000)]))
+
+    # shared_prefix encoding
+    built = y.generate(SCHEMA_4I, rows=50_000, kv_format=y.ENC_SHARED_PREFIX)
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 1)]
+    cases.append(_case("shared_prefix_fmt", SCHEMA_4I, built,
+                       [(1_700_000_000_000_000, (), aggs)],
+                       kv_format=y.ENC_SHARED_PREFIX))
+
+    # packed V1
+    built = y.generate(SCHEMA_4I, rows=50_000, packed_version=1)
+    aggs = [y.Agg(y.AGG_COUNT, 2), y.Agg(y.AGG_SUM_INT64, 3)]
+    cases.append(_case("packed_v1", SCHEMA_4I, built,
+                       [(1_700_000_000_000_000, (), aggs)]))
+
+    # config #5 shape: mixed types
+    schema_mix = y.make_schema(
+        [y.KT_INT64],
+        [(10, y.T_INT64, 1), (11, y.T_DOUBLE, 1), (12, y.T_STRING, 1)])
+    built = y.generate(schema_mix, rows=100_000)
+    preds = [y.Pred(0, 0, y.PRED_GT, 1 << 38, None, 0),
+             y.Pred(0, 0, y.PRED_LT, 1 << 39, None, 0)]
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_DOUBLE, 1),
+            y.Agg(y.AGG_MIN_DOUBLE, 1), y.Agg(y.AGG_MAX_DOUBLE, 1)]
+    cases.append(_case("mixed_types", schema_mix, built,
+                       [(1_700_000_000_000_000, preds, aggs)]))
+
+    # string equality predicate
+    schema_s = y.make_schema([y.KT_INT64], [(10, y.T_STRING, 1),
+                                            (11, y.T_INT64, 1)])
+    b = y.Builder(schema_s)
+    target = b"hello-world"
+    for r in range(5000):
+        s = target if r % 7 == 0 else b"other-%05d" % r
+        b.add_packed_row(1000 + r, [(y.T_STRING, s), (y.T_INT64, r)],
+                         hash_=r // 64, key_datums=(r,))
+    buf = C.create_string_buffer(target, len(target))
+    _KEEP.append((b, buf))
+    pred = y.Pred(0, 0, y.PRED_EQ, 0, C.cast(buf, C.POINTER(C.c_uint8)),
+                  len(target))
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 1)]
+    cases.append(_case("string_eq_pred", schema_s, b.finish(),
+                       [(1_700_000_000_000_000, [pred], aggs)]))
+
+    # column updates / tombstones (docrowwiseiterator-test.cc scenarios)
+    schema_cu = y.make_schema([y.KT_INT64], [(30, y.T_INT64, 1),
+                                             (40, y.T_INT64, 1)])
+    b = y.Builder(schema_cu)
+    b.add_packed_row(1000, [(y.T_INT64, 1), (y.T_INT64, 2)], hash_=0,
+                     key_datums=(0,))
+    b.add_column_update(2000, 1, 222, hash_=0, key_datums=(0,))
+    b.add_row_tombstone(2500, hash_=1, key_datums=(1,), seq=(1 << 50) + 10)
+    b.add_packed_row(1000, [(y.T_INT64, 3), (y.T_INT64, 4)], hash_=1,
+                     key_datums=(1,), seq=(1 << 50) + 5)
+    b.add_row_tombstone(2500, hash_=2, key_datums=(2,), seq=(1 << 50) + 20)
+    b.add_packed_row(1000, [(y.T_INT64, 5), (y.T_INT64, 6)], hash_=2,
+                     key_datums=(2,), seq=(1 << 50) + 15)
+    b.add_column_update(3000, 0, 555, hash_=2, key_datums=(2,),
+                        seq=(1 << 50) + 25)
+    b.add_column_update(1000, 0, 7, hash_=3, key_datums=(3,))
+    b.add_column_update(1500, 1, 8, hash_=3, key_datums=(3,))
+    b.add_packed_row(1000, [(y.T_INT64, 9), (y.T_INT64, 10)], hash_=4,
+                     key_datums=(4,))
+    b.add_column_update(3000, 1, 333, hash_=4, key_datums=(4,),
+                        seq=(1 << 50) + 40)
+    b.add_column_update(2500, 1, None, hash_=4, key_datums=(4,),
+                        seq=(1 << 50) + 38, null=True)
+    _KEEP.append(b)
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 0),
+            y.Agg(y.AGG_SUM_INT64, 1), y.Agg(y.AGG_COUNT, 1)]
+    cases.append(_case("column_updates_tombstones", schema_cu, b.finish(),
+                       [(rm, (), aggs) for rm in (1200, 2200, 2600, 3200,
+                                                  5000)]))
+
+    # NULLs in packed rows, both versions interleaved
+    schema_n = y.make_schema([y.KT_INT64], [(10, y.T_INT64, 1),
+                                            (11, y.T_INT64, 1)])
+    b = y.Builder(schema_n)
+    for r in range(4000):
+        v0 = None if r % 3 == 0 else r
+        v1 = None if r % 5 == 0 else r * 2
+        pv = 1 if r % 2 == 0 else 2
+        b.add_packed_row(1000 + r, [(y.T_INT64, v0), (y.T_INT64, v1)],
+                         hash_=r // 64, key_datums=(r,), packed_version=pv)
+    _KEEP.append(b)
+    aggs = [y.Agg(y.AGG_COUNT, 0), y.Agg(y.AGG_COUNT, 1),
+            y.Agg(y.AGG_SUM_INT64, 0), y.Agg(y.AGG_SUM_INT64, 1)]
+    cases.append(_case("nulls_packed", schema_n, b.finish(),
+                       [(1_000_000, (), aggs)]))
+
+    # key-column predicate
+    built = y.generate(SCHEMA_4I, rows=30_000)
+    preds = [y.Pred(1, 0, y.PRED_LT, 10_000, None, 0)]
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0)]
+    cases.append(_case("key_pred", SCHEMA_4I, built,
+                       [(1_700_000_000_000_000, preds, aggs)]))
+
+    # single row + empty result
+    b = y.Builder(SCHEMA_4I)
+    b.add_packed_row(1000, [(y.T_INT64, i) for i in range(4)], hash_=7,
+                     key_datums=(123,))
+    _KEEP.append(b)
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 2)]
+    cases.append(_case("single_row", SCHEMA_4I, b.finish(),
+                       [(5000, (), aggs), (500, (), aggs)]))
+
+    # rows straddling interval and block boundaries (23 versions, 1KB blocks)
+    built = y.generate(SCHEMA_4I, rows=3_000, versions=23, block_size=1024,
+                       ht_base_micros=1_600_000_000_000_000,
+                       ht_step_micros=1_000)
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 0)]
+    cases.append(_case("spanning_rows", SCHEMA_4I, built,
+                       [(rm, (), aggs) for rm in
+                        (1_600_000_000_000_005, 1_600_000_000_010_000,
+                         1_700_000_000_000_000)]))
+
+    # bounds: lower inclusive / upper exclusive on encoded rowkey
+    # (qlexpr/ql_scanspec.h:200-267 bounds; generator rows r have
+    # hash = r*65536//rows, key col = r)
+    built = y.generate(SCHEMA_4I, rows=30_000)
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0)]
+
+    def dockey(rows, r):
+        h = (r * 65536) // rows
+        return (b"G" + h.to_bytes(2, "big") + b"I" +
+                ((r ^ (1 << 63)).to_bytes(8, "big")) + b"!!")
+
+    lo = dockey(30_000, 5_000)
+    hi = dockey(30_000, 25_000)
+    lo_buf = C.create_string_buffer(lo, len(lo))
+    hi_buf = C.create_string_buffer(hi, len(hi))
+    _KEEP.append((lo_buf, hi_buf))
+    cases.append(_case("bounded", SCHEMA_4I, built,
+                       [(1_700_000_000_000_000, (), aggs,
+                         (lo_buf, len(lo)), (hi_buf, len(hi)))]))
+    return cases
+
+
+def make_spec(case, read_micros, preds, aggs, lower=None, upper=None):
+    spec = y.ScanSpec()
+    spec.schema = case["schema"]
+    spec.kv_format = case["kv_format"]
+    spec.read_time = y.read_time(read_micros)
+    spec.num_preds = len(preds)
+    for i, p in enumerate(preds):
+        spec.preds[i] = p
+    spec.num_aggs = len(aggs)
+    for i, a in enumerate(aggs):
+        spec.aggs[i] = a
+    if lower:
+        spec.lower_bound = C.cast(lower[0], C.POINTER(C.c_uint8))
+        spec.lower_bound_len = lower[1]
+    if upper:
+        spec.upper_bound = C.cast(upper[0], C.POINTER(C.c_uint8))
+        spec.upper_bound_len = upper[1]
+    return spec
+
+
+def make_orcl_spec(read_micros, preds, aggs, lower=None, upper=None):
+    spec = y.OrclScanSpec()
+    spec.read_time = y.orcl_read_time(read_micros)
+    spec.num_preds = len(preds)
+    for i, p in enumerate(preds):
+        spec.preds[i] = y.OrclPred(p.is_key_col, p.col, p.op, p.datum,
+                                   p.bytes, p.bytes_len)
+    spec.num_aggs = len(aggs)
+    for i, a in enumerate(aggs):
+        spec.aggs[i] = y.OrclAgg(a.op, a.col)
+    if lower:
+        spec.lower_bound = C.cast(lower[0], C.POINTER(C.c_uint8))
+        spec.lower_bound_len = lower[1]
+    if upper:
+        spec.upper_bound = C.cast(upper[0], C.POINTER(C.c_uint8))
+        spec.upper_bound_len = upper[1]
+    return spec
+
+
+def run_oracle(case, read_micros, preds, aggs, lower=None, upper=None):
+    osc = y.orcl_schema_from(case["schema"])
+    ospec = make_orcl_spec(read_micros, preds, aggs, lower, upper)
+    res, _ = y.orcl_scan(case["data"], case["offsets"], case["n_blocks"],
+                         osc, ospec, kv_format=case["kv_format"])
+    return res
+
+
+def check_match(gres, ores, aggs, f64_rel=1e-12):
+    assert gres.entries_seen == ores.entries_seen, \
+        (gres.entries_seen, ores.entries_seen)
+    assert gres.rows_scanned == ores.rows_scanned, \
+        (gres.rows_scanned, ores.rows_scanned)
+    assert gres.rows_matched == ores.rows_matched, \
+        (gres.rows_matched, ores.rows_matched)
+    for i, a in enumerate(aggs):
+        assert gres.aggs[i].is_null == ores.aggs[i].is_null, i
+        if ores.aggs[i].is_null:
+            continue
+        if a.op in (y.AGG_SUM_DOUBLE, y.AGG_MIN_DOUBLE, y.AGG_MAX_DOUBLE):
+            gv, ov = gres.aggs[i].value_f64, ores.aggs[i].value_f64
+            assert abs(gv - ov) <= f64_rel * max(1.0, abs(ov)), (i, gv, ov)
+        else:
+            assert gres.aggs[i].value_i64 == ores.aggs[i].value_i64, \
+                (i, gres.aggs[i].value_i64, ores.aggs[i].value_i64)

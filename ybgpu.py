@@ -405,6 +405,21 @@ class OrclScanResult(C.Structure):
                 ("aggs", OrclAggResult * MAX_AGGS)]
 
 
+def sim_scan(spec, data, offsets, n_blocks):
+    """Run the host SIMULATOR of the GPU per-interval algorithm
+    (scan_device.h via ybg_sim_scan) — TEST INFRASTRUCTURE; sequentially
+    executes the exact device code path on CPU."""
+    lib = product()
+    f = _sig(lib, "ybg_sim_scan", C.c_int,
+             [C.POINTER(ScanSpec), C.POINTER(C.c_uint8),
+              C.POINTER(C.c_uint64), C.c_uint64, C.POINTER(ScanResult)])
+    res = ScanResult()
+    rc = f(C.byref(spec), data, offsets, n_blocks, C.byref(res))
+    if rc != 0:
+        raise RuntimeError(f"ybg_sim_scan failed rc={rc}")
+    return res
+
+
 def orcl_schema_from(schema):
     o = OrclSchema()
     o.has_hash = schema.has_hash

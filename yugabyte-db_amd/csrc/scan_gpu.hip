@@ -12,1062 +12,42 @@
 // Execution model (MI355X-first; HBM-bound integer/byte work — MFMA unused):
 //  * Parallel unit = one RESTART INTERVAL (16 entries, rocksdb/table.h:150).
 //    The per-entry key delta chain is serial inside an interval; a 100M-row
-//    tablet has ~6M intervals >> the ~10K concurrent threads per pass, so one
+//    tablet has ~6M intervals >> the ~1M threads of a full launch, so one
 //    THREAD per interval with a contiguous grid-stride mapping (adjacent
 //    lanes = adjacent intervals => a wave streams a contiguous span of the
 //    block array through L1/L2).
-//  * Reconstructed keys live in LDS (128 B/thread): dynamically indexed byte
-//    arrays would otherwise spill to scratch. The finalized row's key bytes
-//    are saved to a global per-thread scratch (L1-resident) because the
-//    in-place delta update destroys them.
+//  * Reconstructed keys live in LDS (128 B/thread); the finalized row's key
+//    bytes are saved to a per-thread global scratch (L1-resident) because
+//    the in-place delta update destroys them.
 //  * Rows straddling interval boundaries, with NO inter-workgroup sync:
 //    every interval DEFERS its first (head) row — ownership unknown — and
 //    WALKS its last (tail) row forward across interval/block boundaries.
-//    After the pass, lane l learns from lane l-1 (shfl/LDS relay) whether
+//    After the pass, lane l learns from lane l-1 (shfl / LDS relay) whether
 //    its head row was already consumed by the predecessor's walk; only
 //    intervals at workgroup boundaries (1 in 256) write a global head record
-//    + continuation flag, resolved by the final reduction.
+//    + continuation flag, folded by the final reduction.
 //  * The final single-workgroup reduction folds wave partials and boundary
 //    head records IN FIXED ORDER => results (including double SUM) are
 //    deterministic run-to-run.
+//
+// The per-interval algorithm itself lives in scan_device.h and is also
+// compiled into a host simulator (scan_host_sim.cc) for CPU-side testing.
 #include <hip/hip_runtime.h>
 
+#include <algorithm>
 #include <cstdio>
 #include <cstring>
 #include <string>
 #include <vector>
 
-#include "../../include/yb_gpu_scan.h"
+#include "scan_device.h"
 
-#define DEV __device__ __forceinline__
+using namespace ybgdev;
 
 namespace {
 
 // ---------------------------------------------------------------------------
-// Device-side spec (POD kernel arg)
-// ---------------------------------------------------------------------------
-
-struct HtLim {      // zero-padded big-endian <=16-byte encoded DocHybridTime
-  uint64_t hi, lo;  // hi = bytes [0,8), lo = [8,16)
-  uint32_t len;
-};
-
-struct DevCol {
-  int32_t id;
-  int32_t dtype;
-  int32_t v1_varlen;  // nullable || string (schema_packing.cc:45-49)
-  int32_t v1_nvb;     // varlen columns before
-  int32_t v1_off;     // offset after prev varlen column
-  int32_t v2_fixed;   // 0 varlen, else 1/2/4/8
-};
-
-struct DevPred {
-  int32_t is_key_col, col, op;
-  uint64_t datum;
-  uint32_t str_len;
-  uint32_t rhs_off;  // rhs bytes offset in aux buffer
-};
-
-struct DevAgg {
-  int32_t op, col;
-};
-
-struct DevSpec {
-  int32_t has_hash, num_hash_cols, num_range_cols;
-  int32_t key_types[YBG_MAX_KEYCOLS];
-  int32_t num_value_cols;
-  DevCol cols[YBG_MAX_COLS];
-  int32_t fmt;
-  HtLim read, local_lim, global_lim, reg_lim;
-  int32_t num_preds;
-  DevPred preds[YBG_MAX_PREDS];
-  int32_t num_aggs;
-  DevAgg aggs[YBG_MAX_AGGS];
-  uint32_t lower_len, lower_off, upper_len, upper_off;
-  int32_t v1_varlen_count;
-};
-
-struct Interval {
-  uint32_t block;
-  uint32_t start;
-  uint32_t end;
-};
-
-constexpr int kThreads = 256;
-constexpr int kKeyCap = 128;
-// wave partial: entries, scanned, matched, err, {val,cnt} x MAX_AGGS
-constexpr int kPartialStride = 4 + 2 * YBG_MAX_AGGS;
-// head record: {val,cnt} x MAX_AGGS, scanned, matched
-constexpr int kHeadStride = 2 * YBG_MAX_AGGS + 2;
-
-constexpr uint8_t kGroupEnd = 0x21, kHybridTimeByte = 0x23, kNullLow = 0x24,
-                  kFloatB = 0x43, kDoubleB = 0x44, kFalseB = 0x46,
-                  kUInt16Hash = 0x47, kInt32B = 0x48, kInt64B = 0x49,
-                  kSysColB = 0x4A, kColB = 0x4B, kStringB = 0x53,
-                  kTrueB = 0x54, kTombB = 0x58, kMergeFlagsB = 0x6B,
-                  kTtlB = 0x74, kUserTsB = 0x75, kPackedV1B = 0x7A,
-                  kPackedV2B = 0x7C;
-
-// ---------------------------------------------------------------------------
-// Byte-level device helpers
-// ---------------------------------------------------------------------------
-
-DEV uint64_t load_be64(const uint8_t* p) {
-  uint64_t v;
-  __builtin_memcpy(&v, p, 8);
-  return __builtin_bswap64(v);
-}
-DEV uint32_t load_be32(const uint8_t* p) {
-  uint32_t v;
-  __builtin_memcpy(&v, p, 4);
-  return __builtin_bswap32(v);
-}
-DEV uint64_t load_le64_u(const uint8_t* p) {
-  uint64_t v;
-  __builtin_memcpy(&v, p, 8);
-  return v;
-}
-DEV uint32_t load_le32_u(const uint8_t* p) {
-  uint32_t v;
-  __builtin_memcpy(&v, p, 4);
-  return v;
-}
-
-// LEB128 varint (rocksdb util/coding.h)
-DEV const uint8_t* leb128(const uint8_t* p, const uint8_t* limit,
-                          uint64_t* v) {
-  uint64_t result = 0;
-  int shift = 0;
-  while (p < limit && shift <= 63) {
-    uint64_t b = *p++;
-    if (b & 128) {
-      result |= (b & 127) << shift;
-    } else {
-      result |= b << shift;
-      *v = result;
-      return p;
-    }
-    shift += 7;
-  }
-  return nullptr;
-}
-
-// yb fast signed varint (util/fast_varint.cc:171-227)
-DEV const uint8_t* svarint(const uint8_t* p, const uint8_t* limit,
-                           int64_t* v) {
-  if (p >= limit) return nullptr;
-  uint32_t header = ((uint32_t)p[0] << 8) | (p + 1 < limit ? p[1] : 0);
-  uint64_t neg = -(uint64_t)((header & 0x8000u) == 0);
-  header ^= (uint32_t)neg;
-  int n = __builtin_clz((~header & 0x7fffu) | 0x20u) - 16;
-  if (p + n > limit) return nullptr;
-  uint64_t temp = 0;
-  for (int i = 0; i < n; ++i) temp = (temp << 8) | p[i];
-  uint64_t mask;
-  if (n >= 10) mask = ~0ull;
-  else if (n == 9) mask = 0x3fffffffffffffffull;
-  else mask = (1ull << (7 * n - 1)) - 1;
-  *v = (int64_t)(((temp & mask) | (~mask & neg)) - neg);
-  return p + n;
-}
-
-DEV int desc_svarint_size(const uint8_t* p, const uint8_t* limit) {
-  if (p >= limit) return 0;
-  uint32_t header = ((uint32_t)p[0] << 8) | (p + 1 < limit ? p[1] : 0);
-  uint64_t neg = -(uint64_t)((header & 0x8000u) == 0);
-  header ^= (uint32_t)neg;
-  return __builtin_clz((~header & 0x7fffu) | 0x20u) - 16;
-}
-
-// yb fast unsigned varint (util/fast_varint.cc:291-334)
-DEV const uint8_t* uvarint(const uint8_t* p, const uint8_t* limit,
-                           uint64_t* v) {
-  if (p >= limit) return nullptr;
-  uint32_t first = p[0];
-  int n = __builtin_clz((((first << 1) ^ 0x1ffu) << 22) | 1u) - 9;
-  // kUnsignedVarIntSize (fast_varint.cc:28-37): clz32((i<<1)^0x1ff)-22
-  if (p + n > limit) return nullptr;
-  if (n == 1) {
-    *v = first & 0x7f;
-    return p + 1;
-  }
-  uint64_t result = 0;
-  int i = 0;
-  if (n == 9) {
-    if (p[1] & 0x80) {
-      n = 10;
-      if (p + n > limit) return nullptr;
-      result = p[1] & 0x3f;
-      i = 2;
-    } else {
-      result = 0;
-      i = 1;
-    }
-  } else {
-    result = first & ((1u << (8 - n)) - 1);
-    i = 1;
-  }
-  for (; i < n; ++i) result = (result << 8) | p[i];
-  *v = result;
-  return p + n;
-}
-
-DEV int dht_size_from_start(const uint8_t* p, const uint8_t* limit) {
-  int off = 0;
-  for (int i = 0; i < 4; ++i) {
-    int sz = desc_svarint_size(p + off, limit);
-    if (sz == 0 || p + off + sz > limit) return 0;
-    off += sz;
-  }
-  return off;
-}
-
-template <class PtrT>
-DEV void slice_u128(PtrT p, uint32_t len, uint64_t* hi, uint64_t* lo) {
-  uint64_t h = 0, l = 0;
-  uint32_t n = len < 16 ? len : 16;
-  for (uint32_t i = 0; i < n; ++i) {
-    uint64_t b = p[i];
-    if (i < 8) h |= b << (56 - 8 * i);
-    else l |= b << (56 - 8 * (i - 8));
-  }
-  *hi = h;
-  *lo = l;
-}
-
-// memcmp + length tiebreak over zero-padded 16-byte slices (Slice::compare)
-DEV int u128_slice_cmp(uint64_t ahi, uint64_t alo, uint32_t alen,
-                       uint64_t bhi, uint64_t blo, uint32_t blen) {
-  if (ahi != bhi) return ahi < bhi ? -1 : 1;
-  if (alo != blo) return alo < blo ? -1 : 1;
-  if (alen != blen) return alen < blen ? -1 : 1;
-  return 0;
-}
-
-// ---------------------------------------------------------------------------
-// Interval table construction
-// ---------------------------------------------------------------------------
-
-__global__ void k_count_restarts(const uint8_t* __restrict__ data,
-                                 const uint64_t* __restrict__ offsets,
-                                 uint64_t n_blocks,
-                                 uint32_t* __restrict__ counts,
-                                 int* __restrict__ error) {
-  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-  for (; i < n_blocks; i += stride) {
-    uint64_t sz = offsets[i + 1] - offsets[i];
-    uint32_t nr = sz >= 8 ? load_le32_u(data + offsets[i + 1] - 4) : 0;
-    if (sz < 8 || nr == 0 || (uint64_t)nr * 4 + 4 > sz) {
-      atomicExch(error, 1);
-      counts[i] = 0;
-      continue;
-    }
-    counts[i] = nr;
-  }
-}
-
-__global__ void k_emit_intervals(const uint8_t* __restrict__ data,
-                                 const uint64_t* __restrict__ offsets,
-                                 uint64_t n_blocks,
-                                 const uint64_t* __restrict__ iv_base,
-                                 Interval* __restrict__ ivs) {
-  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-  for (; i < n_blocks; i += stride) {
-    uint64_t sz = offsets[i + 1] - offsets[i];
-    const uint8_t* blk = data + offsets[i];
-    uint32_t nr = load_le32_u(blk + sz - 4);
-    uint32_t restarts_off = (uint32_t)(sz - 4 - (uint64_t)nr * 4);
-    uint64_t base = iv_base[i];
-    for (uint32_t r = 0; r < nr; ++r) {
-      uint32_t start = load_le32_u(blk + restarts_off + 4ull * r);
-      uint32_t end = (r + 1 < nr)
-                         ? load_le32_u(blk + restarts_off + 4ull * (r + 1))
-                         : restarts_off;
-      ivs[base + r] = Interval{(uint32_t)i, start, end};
-    }
-  }
-}
-
-// ---------------------------------------------------------------------------
-// Entry decode (both formats) into the LDS key scratch
-// ---------------------------------------------------------------------------
-
-struct EntryRef {
-  const uint8_t* value;
-  uint32_t value_len;
-};
-
-// Mirrors DecodeEntryThreeSharedParts (block_internal.h:54-160) +
-// IterKey::Update (db/dbformat.h:405-476) and the shared_prefix DecodeEntry
-// (block.cc:411-436). Returns pointer past the entry or nullptr.
-DEV const uint8_t* decode_entry(int fmt, const uint8_t* p,
-                                const uint8_t* limit, uint8_t* key,
-                                uint32_t* key_len, EntryRef* out) {
-  if (fmt == YBG_ENC_SHARED_PREFIX) {
-    uint64_t shared, non_shared, value_len;
-    if (!(p = leb128(p, limit, &shared))) return nullptr;
-    if (!(p = leb128(p, limit, &non_shared))) return nullptr;
-    if (!(p = leb128(p, limit, &value_len))) return nullptr;
-    if ((uint64_t)(limit - p) < non_shared + value_len) return nullptr;
-    if (shared > *key_len || shared + non_shared > kKeyCap) return nullptr;
-    for (uint32_t i = 0; i < (uint32_t)non_shared; ++i) key[shared + i] = p[i];
-    *key_len = (uint32_t)(shared + non_shared);
-    out->value = p + non_shared;
-    out->value_len = (uint32_t)value_len;
-    return out->value + value_len;
-  }
-
-  if (limit - p < 2) return nullptr;
-  uint64_t encoded_1;
-  if (!(p = leb128(p, limit, &encoded_1))) return nullptr;
-  uint32_t value_size = (uint32_t)(encoded_1 >> 2);
-  uint64_t last8_inc = (encoded_1 & 2) << 7;
-  bool frequent = encoded_1 & 1;
-
-  uint32_t shared_prefix = 0, ns1 = 0, ns2 = 0, last8 = 0;
-  int64_t ns1_delta = 0, ns2_delta = 0;
-  bool shared_something;
-  uint64_t tmp;
-
-  if (frequent) {
-    if (!(p = leb128(p, limit, &tmp))) return nullptr;
-    shared_prefix = (uint32_t)tmp;
-    last8 = 8;
-    shared_something = true;
-    ns1 = 1;
-    ns2 = 1;
-  } else {
-    uint32_t e2 = *p++;
-    if ((e2 & 1) == 0) {
-      shared_something = false;
-      if (e2 == 0) {
-        if (!(p = leb128(p, limit, &tmp))) return nullptr;
-        ns1 = (uint32_t)tmp;
-      } else {
-        ns1 = e2 >> 1;
-      }
-    } else {
-      shared_something = true;
-      if ((e2 & 2) == 0) {
-        last8 = 8;
-        ns2_delta = (e2 >> 2) & 1;
-        ns1 = (e2 >> 3) & 7;
-        ns2 = (e2 >> 6) & 3;
-      } else {
-        last8 = (e2 & 4) ? 8 : 0;
-        if (!(p = leb128(p, limit, &tmp))) return nullptr;
-        ns1 = (uint32_t)tmp;
-        if (e2 & 8) {
-          if (!(p = svarint(p, limit, &ns1_delta))) return nullptr;
-        }
-        if (e2 & 16) {
-          if (!(p = leb128(p, limit, &tmp))) return nullptr;
-          ns2 = (uint32_t)tmp;
-        }
-        if (e2 & 32) {
-          if (!(p = svarint(p, limit, &ns2_delta))) return nullptr;
-        }
-      }
-      if (!(p = leb128(p, limit, &tmp))) return nullptr;
-      shared_prefix = (uint32_t)tmp;
-    }
-  }
-  if ((uint64_t)(limit - p) < (uint64_t)ns1 + ns2 + value_size) return nullptr;
-
-  if (!shared_something) {
-    if (ns1 > kKeyCap) return nullptr;
-    for (uint32_t i = 0; i < ns1; ++i) key[i] = p[i];
-    *key_len = ns1;
-    out->value = p + ns1;
-    out->value_len = value_size;
-    return out->value + value_size;
-  }
-
-  uint64_t prev_mid_start = (uint64_t)shared_prefix + ns1 - (uint64_t)ns1_delta;
-  uint64_t prev_ns2 = (uint64_t)ns2 - (uint64_t)ns2_delta;
-  uint64_t prev_except_mid = prev_mid_start + prev_ns2 + last8;
-  if (*key_len < prev_except_mid) return nullptr;
-  uint64_t mid = *key_len - prev_except_mid;
-  if (shared_prefix + mid + last8 == 0) return nullptr;
-
-  uint64_t new_mid_start = shared_prefix + ns1;
-  uint64_t new_ns2_start = new_mid_start + mid;
-  uint64_t new_last8_start = new_ns2_start + ns2;
-  uint64_t new_key_size = new_last8_start + last8;
-  if (new_key_size > kKeyCap) return nullptr;
-
-  uint64_t last_comp = 0;
-  if (last8) {
-    const uint8_t* q = key + *key_len - 8;
-    for (int i = 7; i >= 0; --i) last_comp = (last_comp << 8) | q[i];
-    last_comp += last8_inc;
-  }
-  if (new_mid_start != prev_mid_start && mid > 0) {
-    if (new_mid_start < prev_mid_start) {
-      for (uint64_t i = 0; i < mid; ++i)
-        key[new_mid_start + i] = key[prev_mid_start + i];
-    } else {
-      for (uint64_t i = mid; i-- > 0;)
-        key[new_mid_start + i] = key[prev_mid_start + i];
-    }
-  }
-  for (uint32_t i = 0; i < ns1; ++i) key[shared_prefix + i] = p[i];
-  for (uint32_t i = 0; i < ns2; ++i) key[new_ns2_start + i] = p[ns1 + i];
-  if (last8) {
-    uint8_t* q = key + new_last8_start;
-    for (int i = 0; i < 8; ++i) q[i] = (uint8_t)(last_comp >> (8 * i));
-  }
-  *key_len = (uint32_t)new_key_size;
-  out->value = p + ns1 + ns2;
-  out->value_len = value_size;
-  return out->value + value_size;
-}
-
-// ---------------------------------------------------------------------------
-// DocKey length (doc_key.h:40-63); key lives in LDS
-// ---------------------------------------------------------------------------
-
-DEV uint32_t dockey_len(const DevSpec& sp, const uint8_t* p, uint32_t len) {
-  uint32_t off = 0;
-  int col = 0;
-  if (sp.has_hash) {
-    if (len < 3 || p[0] != kUInt16Hash) return 0;
-    off = 3;
-  }
-  for (int group = 0; group < 2; ++group) {
-    if (group == 0 && !sp.has_hash) continue;
-    int ncols = group == 0 ? sp.num_hash_cols : sp.num_range_cols;
-    for (int i = 0; i < ncols; ++i, ++col) {
-      if (off >= len) return 0;
-      uint8_t t = p[off];
-      int kt = sp.key_types[col];
-      if (kt == YBG_KT_INT64) {
-        if (t != kInt64B || off + 9 > len) return 0;
-        off += 9;
-      } else if (kt == YBG_KT_INT32) {
-        if (t != kInt32B || off + 5 > len) return 0;
-        off += 5;
-      } else {
-        if (t != kStringB) return 0;
-        uint32_t s = off + 1;
-        for (;;) {
-          if (s + 1 >= len) return 0;
-          if (p[s] == 0) {
-            if (p[s + 1] == 0) break;
-            if (p[s + 1] != 1) return 0;
-            s += 2;
-          } else {
-            ++s;
-          }
-        }
-        off = s + 2;
-      }
-    }
-    if (off >= len || p[off] != kGroupEnd) return 0;
-    ++off;
-  }
-  return off;
-}
-
-DEV int skip_control(const uint8_t* v, uint32_t len) {
-  uint32_t off = 0;
-  if (off < len && v[off] == kMergeFlagsB) {
-    ++off;
-    uint64_t tmp;
-    const uint8_t* q = uvarint(v + off, v + len, &tmp);
-    if (!q) return -1;
-    off = (uint32_t)(q - v);
-  }
-  if (off < len && v[off] == kHybridTimeByte) {
-    ++off;
-    int sz = dht_size_from_start(v + off, v + len);
-    if (!sz) return -1;
-    off += sz;
-  }
-  if (off < len && v[off] == kTtlB) {
-    ++off;
-    int64_t tmp;
-    const uint8_t* q = svarint(v + off, v + len, &tmp);
-    if (!q) return -1;
-    off = (uint32_t)(q - v);
-  }
-  if (off < len && v[off] == kUserTsB) {
-    off += 9;
-    if (off > len) return -1;
-  }
-  return (int)off;
-}
-
-// ---------------------------------------------------------------------------
-// Streaming row context
-// ---------------------------------------------------------------------------
-
-struct Operand {
-  uint64_t datum;     // numeric bit pattern, or (str global offset)
-  uint32_t str_len;
-  bool is_null;
-};
-
-struct RowCtx {
-  bool base_seen;
-  bool found;
-  uint64_t base_ht_hi, base_ht_lo;
-  uint32_t base_ht_len;
-  int32_t cur_col;     // current column-update group (-2 liveness)
-  bool cur_col_done;
-  Operand pred_op[YBG_MAX_PREDS];
-  Operand agg_op[YBG_MAX_AGGS];
-};
-
-DEV void row_reset(RowCtx* rc, const DevSpec& sp) {
-  rc->base_seen = false;
-  rc->found = false;
-  rc->cur_col = -1;
-  rc->cur_col_done = false;
-#pragma unroll
-  for (int i = 0; i < YBG_MAX_PREDS; ++i) rc->pred_op[i].is_null = true;
-#pragma unroll
-  for (int i = 0; i < YBG_MAX_AGGS; ++i) rc->agg_op[i].is_null = true;
-}
-
-DEV void set_col_operand(RowCtx* rc, const DevSpec& sp, int col, bool is_null,
-                         uint64_t datum, uint32_t str_len) {
-#pragma unroll
-  for (int i = 0; i < YBG_MAX_PREDS; ++i) {
-    if (i < sp.num_preds && !sp.preds[i].is_key_col && sp.preds[i].col == col) {
-      rc->pred_op[i].is_null = is_null;
-      rc->pred_op[i].datum = datum;
-      rc->pred_op[i].str_len = str_len;
-    }
-  }
-#pragma unroll
-  for (int i = 0; i < YBG_MAX_AGGS; ++i) {
-    if (i < sp.num_aggs && sp.aggs[i].col == col &&
-        sp.aggs[i].op != YBG_AGG_COUNT_STAR) {
-      rc->agg_op[i].is_null = is_null;
-      rc->agg_op[i].datum = datum;
-      rc->agg_op[i].str_len = str_len;
-    }
-  }
-}
-
-// Decode one V1-encoded single value (primitive_value.cc:1066-1125).
-// Returns 0 null/tombstone, 1 applied, -1 error.
-DEV int decode_single_v1(const DevSpec& sp, const uint8_t* base, int col,
-                         const uint8_t* vp, uint32_t vlen, RowCtx* rc) {
-  if (vlen == 0) {
-    set_col_operand(rc, sp, col, true, 0, 0);
-    return 0;
-  }
-  uint8_t t = vp[0];
-  if (t == kTombB || t == kNullLow) {
-    set_col_operand(rc, sp, col, true, 0, 0);
-    return 0;
-  }
-  const DevCol& c = sp.cols[col];
-  uint64_t datum = 0;
-  uint32_t slen = 0;
-  switch (c.dtype) {
-    case YBG_T_BOOL:
-      if (t != kTrueB && t != kFalseB) return -1;
-      datum = (t == kTrueB);
-      break;
-    case YBG_T_INT8:
-    case YBG_T_INT16:
-    case YBG_T_INT32:
-      if (t != kInt32B || vlen < 5) return -1;
-      datum = (uint64_t)(int64_t)(int32_t)load_be32(vp + 1);
-      break;
-    case YBG_T_INT64:
-      if (t != kInt64B || vlen < 9) return -1;
-      datum = load_be64(vp + 1);
-      break;
-    case YBG_T_UINT64:
-      if (vlen < 9) return -1;
-      datum = load_be64(vp + 1);
-      break;
-    case YBG_T_FLOAT:
-      if (t != kFloatB || vlen < 5) return -1;
-      datum = load_be32(vp + 1);
-      break;
-    case YBG_T_UINT32:
-      if (vlen < 5) return -1;
-      datum = load_be32(vp + 1);
-      break;
-    case YBG_T_DOUBLE:
-      if (t != kDoubleB || vlen < 9) return -1;
-      datum = load_be64(vp + 1);
-      break;
-    case YBG_T_STRING:
-      if (t != kStringB) return -1;
-      datum = (uint64_t)(vp + 1 - base);
-      slen = vlen - 1;
-      break;
-    default:
-      return -1;
-  }
-  set_col_operand(rc, sp, col, false, datum, slen);
-  return 1;
-}
-
-// Packed row decode streaming into operand slots. `body` points at 'z'/'|'.
-DEV bool decode_packed(const DevSpec& sp, const uint8_t* base,
-                       const uint8_t* body, uint32_t len, RowCtx* rc) {
-  uint8_t kind = body[0];
-  uint32_t off = 1;
-  uint64_t version;
-  const uint8_t* q = uvarint(body + off, body + len, &version);
-  if (!q) return false;
-  off = (uint32_t)(q - body);
-
-  if (kind == kPackedV1B) {
-    const uint8_t* header = body + off;
-    uint32_t prefix_len = (uint32_t)sp.v1_varlen_count * 4;
-    if (off + prefix_len > len) return false;
-    const uint8_t* data = header + prefix_len;
-    uint32_t data_len = len - off - prefix_len;
-    for (int i = 0; i < sp.num_value_cols; ++i) {
-      const DevCol& c = sp.cols[i];
-      uint32_t start = (uint32_t)c.v1_off;
-      if (c.v1_nvb) start += load_le32_u(header + (c.v1_nvb - 1) * 4);
-      uint32_t end;
-      if (c.v1_varlen) {
-        end = load_le32_u(header + c.v1_nvb * 4);
-      } else {
-        uint32_t fs;
-        switch (c.dtype) {
-          case YBG_T_BOOL: fs = 1; break;
-          case YBG_T_INT8: case YBG_T_INT16: case YBG_T_INT32:
-          case YBG_T_UINT32: case YBG_T_FLOAT: fs = 5; break;
-          default: fs = 9; break;
-        }
-        end = start + fs;
-      }
-      if (end < start || end > data_len) return false;
-      int r = decode_single_v1(sp, base, i, data + start, end - start, rc);
-      if (r < 0) return false;
-    }
-    return true;
-  }
-
-  if (kind == kPackedV2B) {
-    if (off >= len) return false;
-    uint8_t flags = body[off++];
-    const uint8_t* null_mask = nullptr;
-    if (flags & 1) {  // kHasNullsFlag (packed_row.h:197)
-      null_mask = body + off;
-      off += (uint32_t)((sp.num_value_cols + 7) / 8);
-      if (off > len) return false;
-    }
-    const uint8_t* data = body + off;
-    const uint8_t* end = body + len;
-    for (int i = 0; i < sp.num_value_cols; ++i) {
-      if (null_mask && (null_mask[i >> 3] & (1 << (i & 7)))) {
-        set_col_operand(rc, sp, i, true, 0, 0);
-        continue;
-      }
-      const DevCol& c = sp.cols[i];
-      if (c.v2_fixed) {
-        if (data + c.v2_fixed > end) return false;
-        uint64_t u = 0;
-        switch (c.v2_fixed) {  // raw little-endian (value_packing_v2.cc:73-77)
-          case 1: u = data[0]; break;
-          case 2: { uint16_t x; __builtin_memcpy(&x, data, 2); u = x; break; }
-          case 4: u = load_le32_u(data); break;
-          default: u = load_le64_u(data); break;
-        }
-        if (c.dtype == YBG_T_INT8) u = (uint64_t)(int64_t)(int8_t)u;
-        else if (c.dtype == YBG_T_INT16) u = (uint64_t)(int64_t)(int16_t)u;
-        else if (c.dtype == YBG_T_INT32) u = (uint64_t)(int64_t)(int32_t)u;
-        set_col_operand(rc, sp, i, false, u, 0);
-        data += c.v2_fixed;
-      } else {
-        if (data >= end) return false;
-        uint32_t flen, consumed;
-        uint8_t b0 = *data;
-        if ((b0 & 1) == 0) {  // field length (fast_varint.cc:373-384)
-          flen = b0 >> 1;
-          consumed = 1;
-        } else {
-          flen = load_le32_u(data) >> 1;
-          consumed = 4;
-        }
-        data += consumed;
-        if (data + flen > end) return false;
-        set_col_operand(rc, sp, i, false, (uint64_t)(data - base), flen);
-        data += flen;
-      }
-    }
-    return true;
-  }
-  return false;
-}
-
-// key-column datum from saved rowkey bytes (global scratch)
-DEV bool key_col_value(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
-                       int target, uint64_t* out_datum, uint32_t* out_soff,
-                       uint32_t* out_slen) {
-  uint32_t off = sp.has_hash ? 3u : 0u;
-  int col = 0;
-  for (int group = 0; group < 2; ++group) {
-    if (group == 0 && !sp.has_hash) continue;
-    int ncols = group == 0 ? sp.num_hash_cols : sp.num_range_cols;
-    for (int k = 0; k < ncols; ++k, ++col) {
-      int kt = sp.key_types[col];
-      if (kt == YBG_KT_INT64) {
-        if (col == target) {
-          *out_datum =
-              (uint64_t)(int64_t)(load_be64(rk + off + 1) ^ 0x8000000000000000ull);
-          *out_slen = 0;
-          return true;
-        }
-        off += 9;
-      } else if (kt == YBG_KT_INT32) {
-        if (col == target) {
-          *out_datum = (uint64_t)(int64_t)(int32_t)(load_be32(rk + off + 1) ^
-                                                    0x80000000u);
-          *out_slen = 0;
-          return true;
-        }
-        off += 5;
-      } else {
-        uint32_t s = off + 1;
-        while (s + 1 < rk_len && !(rk[s] == 0 && rk[s + 1] == 0))
-          s += (rk[s] == 0) ? 2 : 1;
-        if (col == target) {
-          *out_soff = off + 1;
-          *out_slen = s - (off + 1);
-          return true;
-        }
-        off = s + 2;
-      }
-    }
-    ++off;  // group end
-  }
-  return false;
-}
-
-// Predicates on streamed operands + saved rowkey
-// (pgsql_operation.cc:602-668 typed-compare subset; NULL => filtered).
-DEV bool eval_preds(const DevSpec& sp, const uint8_t* base, const RowCtx& rc,
-                    const uint8_t* rk, uint32_t rk_len, const uint8_t* aux) {
-  for (int i = 0; i < sp.num_preds; ++i) {
-    const DevPred& pr = sp.preds[i];
-    int cmp = 0;
-    int dt;
-    uint64_t lhs_datum = 0;
-    const uint8_t* lstr = nullptr;
-    uint32_t lstr_len = 0;
-    bool key_string_escaped = false;
-    if (pr.is_key_col) {
-      uint64_t d;
-      uint32_t soff = 0, sl = 0;
-      if (!key_col_value(sp, rk, rk_len, pr.col, &d, &soff, &sl)) return false;
-      int kt = sp.key_types[pr.col];
-      if (kt == YBG_KT_STRING) {
-        lstr = rk + soff;
-        lstr_len = sl;
-        key_string_escaped = true;
-        dt = YBG_T_STRING;
-      } else {
-        lhs_datum = d;
-        dt = YBG_T_INT64;
-      }
-    } else {
-      const Operand& op = rc.pred_op[i];
-      if (op.is_null) return false;
-      dt = sp.cols[pr.col].dtype;
-      if (dt == YBG_T_STRING) {
-        lstr = base + op.datum;
-        lstr_len = op.str_len;
-      } else {
-        lhs_datum = op.datum;
-      }
-    }
-    if (dt == YBG_T_STRING) {
-      const uint8_t* rhs = aux + pr.rhs_off;
-      if (key_string_escaped) {
-        // zero-escaped lhs ('\0'->"\0\1"): unescape on the fly
-        uint32_t si = 0, ri = 0;
-        cmp = 0;
-        while (si < lstr_len && ri < pr.str_len) {
-          uint8_t cb = lstr[si];
-          si += (cb == 0) ? 2 : 1;
-          if (cb != rhs[ri]) {
-            cmp = cb < rhs[ri] ? -1 : 1;
-            break;
-          }
-          ++ri;
-        }
-        if (cmp == 0) {
-          bool le = si >= lstr_len, re = ri >= pr.str_len;
-          cmp = (le && re) ? 0 : (le ? -1 : 1);
-        }
-      } else {
-        uint32_t n = lstr_len < pr.str_len ? lstr_len : pr.str_len;
-        for (uint32_t k = 0; k < n; ++k) {
-          if (lstr[k] != rhs[k]) {
-            cmp = lstr[k] < rhs[k] ? -1 : 1;
-            break;
-          }
-        }
-        if (cmp == 0 && lstr_len != pr.str_len)
-          cmp = lstr_len < pr.str_len ? -1 : 1;
-      }
-    } else if (dt == YBG_T_DOUBLE) {
-      double a = __longlong_as_double((long long)lhs_datum);
-      double b = __longlong_as_double((long long)pr.datum);
-      cmp = a < b ? -1 : (a > b ? 1 : 0);
-    } else if (dt == YBG_T_FLOAT) {
-      float a = __uint_as_float((uint32_t)lhs_datum);
-      float b = __uint_as_float((uint32_t)pr.datum);
-      cmp = a < b ? -1 : (a > b ? 1 : 0);
-    } else {
-      int64_t a = (int64_t)lhs_datum, b = (int64_t)pr.datum;
-      cmp = a < b ? -1 : (a > b ? 1 : 0);
-    }
-    bool pass;
-    switch (pr.op) {
-      case YBG_PRED_GT: pass = cmp > 0; break;
-      case YBG_PRED_GE: pass = cmp >= 0; break;
-      case YBG_PRED_LT: pass = cmp < 0; break;
-      case YBG_PRED_LE: pass = cmp <= 0; break;
-      case YBG_PRED_EQ: pass = cmp == 0; break;
-      default: pass = cmp != 0; break;
-    }
-    if (!pass) return false;
-  }
-  return true;
-}
-
-DEV bool in_bounds(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
-                   const uint8_t* aux) {
-  if (sp.lower_len) {
-    const uint8_t* b = aux + sp.lower_off;
-    uint32_t n = rk_len < sp.lower_len ? rk_len : sp.lower_len;
-    int cmp = 0;
-    for (uint32_t i = 0; i < n; ++i)
-      if (rk[i] != b[i]) { cmp = rk[i] < b[i] ? -1 : 1; break; }
-    if (cmp == 0 && rk_len != sp.lower_len)
-      cmp = rk_len < sp.lower_len ? -1 : 1;
-    if (cmp < 0) return false;
-  }
-  if (sp.upper_len) {
-    const uint8_t* b = aux + sp.upper_off;
-    uint32_t n = rk_len < sp.upper_len ? rk_len : sp.upper_len;
-    int cmp = 0;
-    for (uint32_t i = 0; i < n; ++i)
-      if (rk[i] != b[i]) { cmp = rk[i] < b[i] ? -1 : 1; break; }
-    if (cmp == 0 && rk_len != sp.upper_len)
-      cmp = rk_len < sp.upper_len ? -1 : 1;
-    if (cmp >= 0) return false;
-  }
-  return true;
-}
-
-// Aggregate accumulate (doc_expr.cc:248-395); cnt counts non-null
-// contributions (SUM/MIN/MAX null-ness = cnt == 0).
-DEV void acc_row(const DevSpec& sp, const RowCtx& rc, uint64_t* agg_val,
-                 uint64_t* agg_cnt) {
-#pragma unroll
-  for (int g = 0; g < YBG_MAX_AGGS; ++g) {
-    if (g >= sp.num_aggs) break;
-    const DevAgg& a = sp.aggs[g];
-    bool isnull = (a.op == YBG_AGG_COUNT_STAR) ? false : rc.agg_op[g].is_null;
-    if (isnull) continue;
-    uint64_t v = rc.agg_op[g].datum;
-    switch (a.op) {
-      case YBG_AGG_COUNT_STAR:
-      case YBG_AGG_COUNT:
-        agg_val[g] += 1;
-        break;
-      case YBG_AGG_SUM_INT64:
-        agg_val[g] = (uint64_t)((int64_t)agg_val[g] + (int64_t)v);
-        break;
-      case YBG_AGG_SUM_DOUBLE: {
-        double cur = __longlong_as_double((long long)agg_val[g]) +
-                     __longlong_as_double((long long)v);
-        agg_val[g] = (uint64_t)__double_as_longlong(cur);
-        break;
-      }
-      case YBG_AGG_MIN_INT64:
-        if (agg_cnt[g] == 0 || (int64_t)v < (int64_t)agg_val[g]) agg_val[g] = v;
-        break;
-      case YBG_AGG_MAX_INT64:
-        if (agg_cnt[g] == 0 || (int64_t)v > (int64_t)agg_val[g]) agg_val[g] = v;
-        break;
-      case YBG_AGG_MIN_DOUBLE: {
-        double d = __longlong_as_double((long long)v);
-        if (agg_cnt[g] == 0 || d < __longlong_as_double((long long)agg_val[g]))
-          agg_val[g] = (uint64_t)__double_as_longlong(d);
-        break;
-      }
-      case YBG_AGG_MAX_DOUBLE: {
-        double d = __longlong_as_double((long long)v);
-        if (agg_cnt[g] == 0 || d > __longlong_as_double((long long)agg_val[g]))
-          agg_val[g] = (uint64_t)__double_as_longlong(d);
-        break;
-      }
-    }
-    agg_cnt[g] += 1;
-  }
-}
-
-// Combine one partial (val,cnt) pair into an accumulator. Fold order must be
-// kept FIXED by callers so double SUM is deterministic.
-DEV void combine1(int op, uint64_t* av, uint64_t* ac, uint64_t bv,
-                  uint64_t bc) {
-  if (bc == 0) return;
-  switch (op) {
-    case YBG_AGG_COUNT_STAR:
-    case YBG_AGG_COUNT:
-    case YBG_AGG_SUM_INT64:
-      *av = (uint64_t)((int64_t)*av + (int64_t)bv);
-      break;
-    case YBG_AGG_SUM_DOUBLE: {
-      double cur = (*ac ? __longlong_as_double((long long)*av) : 0.0) +
-                   __longlong_as_double((long long)bv);
-      *av = (uint64_t)__double_as_longlong(cur);
-      break;
-    }
-    case YBG_AGG_MIN_INT64:
-      if (*ac == 0 || (int64_t)bv < (int64_t)*av) *av = bv;
-      break;
-    case YBG_AGG_MAX_INT64:
-      if (*ac == 0 || (int64_t)bv > (int64_t)*av) *av = bv;
-      break;
-    case YBG_AGG_MIN_DOUBLE:
-      if (*ac == 0 || __longlong_as_double((long long)bv) <
-                          __longlong_as_double((long long)*av))
-        *av = bv;
-      break;
-    case YBG_AGG_MAX_DOUBLE:
-      if (*ac == 0 || __longlong_as_double((long long)bv) >
-                          __longlong_as_double((long long)*av))
-        *av = bv;
-      break;
-  }
-  *ac += bc;
-}
-
-// Combine full partial arrays in fixed order.
-DEV void agg_combine(const DevSpec& sp, uint64_t* a_val, uint64_t* a_cnt,
-                     const uint64_t* b_val, const uint64_t* b_cnt) {
-#pragma unroll
-  for (int g = 0; g < YBG_MAX_AGGS; ++g) {
-    if (g >= sp.num_aggs) break;
-    combine1(sp.aggs[g].op, &a_val[g], &a_cnt[g], b_val[g], b_cnt[g]);
-  }
-}
-
-// Visibility + row-state update for one entry. key/rowkey live in LDS.
-// Returns false on corruption.
-DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
-                       const uint8_t* key, uint32_t key_len,
-                       const uint8_t* value, uint32_t value_len,
-                       uint32_t rowkey_len, RowCtx* rc) {
-  uint32_t ukey_len = key_len - 8;
-  uint32_t ht_size = key[ukey_len - 1] & 0x1f;
-  const uint8_t* ht_enc = key + ukey_len - ht_size;
-  uint32_t prefix_len = ukey_len - ht_size - 1;
-
-  uint64_t ht_hi, ht_lo;
-  slice_u128(ht_enc, ht_size, &ht_hi, &ht_lo);
-  bool visible;
-  if (value_len > 0 && value[0] == kHybridTimeByte) {
-    // committed-txn record with intent time (:1249-1267)
-    uint64_t v_hi, v_lo;
-    slice_u128(value + 1, value_len - 1, &v_hi, &v_lo);
-    bool use_global =
-        u128_slice_cmp(v_hi, v_lo, value_len - 1, sp.local_lim.hi,
-                       sp.local_lim.lo, sp.local_lim.len) > 0;
-    const HtLim& lim = use_global ? sp.global_lim : sp.read;
-    visible = u128_slice_cmp(ht_hi, ht_lo, ht_size, lim.hi, lim.lo, lim.len) >= 0;
-    if (visible) {
-      int iht = dht_size_from_start(value + 1, value + value_len);
-      if (!iht) return false;
-      value += 1 + iht;
-      value_len -= 1 + iht;
-    }
-  } else {
-    visible = u128_slice_cmp(ht_hi, ht_lo, ht_size, sp.reg_lim.hi,
-                             sp.reg_lim.lo, sp.reg_lim.len) >= 0;
-  }
-  if (!visible) return true;
-
-  if (rowkey_len == prefix_len) {
-    if (!rc->base_seen) {
-      rc->base_seen = true;
-      rc->base_ht_hi = ht_hi;
-      rc->base_ht_lo = ht_lo;
-      rc->base_ht_len = ht_size;
-      int cf = skip_control(value, value_len);
-      if (cf < 0) return false;
-      const uint8_t* body = value + cf;
-      uint32_t body_len = value_len - cf;
-      if (body_len > 0 && (body[0] == kPackedV1B || body[0] == kPackedV2B)) {
-        if (!decode_packed(sp, base, body, body_len, rc)) return false;
-        rc->found = true;  // doc_reader.cc:1894-1900
-      }
-    }
-  } else {
-    const uint8_t* sk = key + rowkey_len;
-    uint32_t sk_len = prefix_len - rowkey_len;
-    if (sk_len < 2) return false;
-    if (sk[0] == kSysColB) {
-      if (rc->cur_col != -2) {
-        rc->cur_col = -2;
-        rc->cur_col_done = false;
-      }
-      if (!rc->cur_col_done) {
-        rc->cur_col_done = true;
-        bool newer = !rc->base_seen ||
-                     u128_slice_cmp(ht_hi, ht_lo, ht_size, rc->base_ht_hi,
-                                    rc->base_ht_lo, rc->base_ht_len) < 0;
-        if (newer) {
-          int cf = skip_control(value, value_len);
-          if (cf < 0) return false;
-          if (value_len - (uint32_t)cf > 0 && value[cf] != kTombB)
-            rc->found = true;
-        }
-      }
-    } else if (sk[0] == kColB) {
-      int64_t col_id;
-      const uint8_t* q = svarint(sk + 1, sk + sk_len, &col_id);
-      if (!q || q != sk + sk_len) return false;
-      int idx = -1;
-      for (int i = 0; i < sp.num_value_cols; ++i)
-        if (sp.cols[i].id == (int32_t)col_id) { idx = i; break; }
-      if (idx < 0) return true;
-      if (rc->cur_col != idx) {
-        rc->cur_col = idx;
-        rc->cur_col_done = false;
-      }
-      if (!rc->cur_col_done) {
-        rc->cur_col_done = true;
-        bool newer = !rc->base_seen ||
-                     u128_slice_cmp(ht_hi, ht_lo, ht_size, rc->base_ht_hi,
-                                    rc->base_ht_lo, rc->base_ht_len) < 0;
-        if (newer) {
-          int cf = skip_control(value, value_len);
-          if (cf < 0) return false;
-          int r =
-              decode_single_v1(sp, base, idx, value + cf, value_len - cf, rc);
-          if (r < 0) return false;
-          if (r > 0) rc->found = true;
-        }
-      }
-    } else {
-      return false;
-    }
-  }
-  return true;
-}
-
-// ---------------------------------------------------------------------------
-// Phase-1 scan kernel (aggregate mode)
+// Phase-1 scan kernel (aggregate mode): one thread per interval
 // ---------------------------------------------------------------------------
 
 __global__ __launch_bounds__(kThreads) void k_scan(
@@ -1080,13 +60,10 @@ __global__ __launch_bounds__(kThreads) void k_scan(
   __shared__ uint8_t key_scratch[kThreads * kKeyCap];
   __shared__ uint8_t wave_relay[kThreads / 64 + 1];
   uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
-  const uint32_t gtid =
-      blockIdx.x * kThreads + threadIdx.x;  // global thread id (< grid span)
+  const uint32_t gtid = blockIdx.x * kThreads + threadIdx.x;
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
-
   const uint64_t span = (uint64_t)gridDim.x * kThreads;
 
-  // thread-local committed accumulators (middle/tail rows + resolved heads)
   uint64_t entries = 0, scanned = 0, matched = 0, errs = 0;
   uint64_t agg_val[YBG_MAX_AGGS], agg_cnt[YBG_MAX_AGGS];
 #pragma unroll
@@ -1094,138 +71,53 @@ __global__ __launch_bounds__(kThreads) void k_scan(
 
   for (uint64_t j0 = 0; j0 < n_ivs; j0 += span) {
     const uint64_t j = j0 + gtid;
-    bool active = j < n_ivs;
-
-    // deferred head-row contribution for this interval
-    uint64_t h_val[YBG_MAX_AGGS], h_cnt[YBG_MAX_AGGS];
-#pragma unroll
-    for (int g = 0; g < YBG_MAX_AGGS; ++g) { h_val[g] = 0; h_cnt[g] = 0; }
-    uint64_t h_scanned = 0, h_matched = 0;
+    const bool active = j < n_ivs;
+    HeadOut ho;
     bool walked_next = false;
-    bool fail = false;
-
     if (active) {
-      Interval iv = ivs[j];
-      const uint8_t* blk = data + block_offsets[iv.block];
-      const uint8_t* p = blk + iv.start;
-      const uint8_t* limit = blk + iv.end;
-      uint64_t cur_iv = j;
-
-      uint32_t key_len = 0;
-      uint32_t rk_len = 0;  // saved rowkey length
-      bool row_open = false;
-      bool in_head = true;
-      RowCtx rc;
-      row_reset(&rc, sp);
-      EntryRef er;
-
-      for (;;) {
-        if (p >= limit) {
-          if (cur_iv + 1 >= n_ivs) break;
-          Interval nx = ivs[cur_iv + 1];
-          cur_iv += 1;
-          blk = data + block_offsets[nx.block];
-          p = blk + nx.start;
-          limit = blk + nx.end;
-          // first entry of a restart interval is self-contained; the carried
-          // key state is overwritten by its full-key decode.
-          continue;
-        }
-        const uint8_t* q = decode_entry(sp.fmt, p, limit, key, &key_len, &er);
-        if (!q || key_len < 10) { fail = true; break; }
-        uint32_t ukey_len = key_len - 8;
-        uint32_t ht_sz = key[ukey_len - 1] & 0x1f;
-        if (ht_sz == 0 || ukey_len < ht_sz + 2 ||
-            key[ukey_len - ht_sz - 1] != kHybridTimeByte) { fail = true; break; }
-        uint32_t prefix_len = ukey_len - ht_sz - 1;
-        uint32_t rk = dockey_len(sp, key, prefix_len);
-        if (!rk || rk > prefix_len) { fail = true; break; }
-
-        bool row_change = !row_open || (rk != rk_len);
-        if (!row_change) {
-          for (uint32_t i = 0; i < rk; ++i) {
-            if (key[i] != rk_save[i]) { row_change = true; break; }
-          }
-        }
-        if (row_change) {
-          if (row_open) {
-            // finalize previous row (doc_rowwise_iterator row boundary)
-            if (rc.found && in_bounds(sp, rk_save, rk_len, aux)) {
-              uint64_t* sc = in_head ? &h_scanned : &scanned;
-              uint64_t* ma = in_head ? &h_matched : &matched;
-              *sc += 1;
-              if (eval_preds(sp, data, rc, rk_save, rk_len, aux)) {
-                *ma += 1;
-                acc_row(sp, rc, in_head ? h_val : agg_val,
-                        in_head ? h_cnt : agg_cnt);
-              }
-            }
-            in_head = false;
-            row_open = false;
-            if (cur_iv > j) break;  // tail walk ended at a new row
-          }
-          for (uint32_t i = 0; i < rk; ++i) rk_save[i] = key[i];
-          rk_len = rk;
-          row_open = true;
-          row_reset(&rc, sp);
-        }
-        if (cur_iv == j) entries += 1;
-        else if (cur_iv == j + 1) walked_next = true;
-        if (!process_entry(sp, data, key, key_len, er.value, er.value_len,
-                           rk_len, &rc)) { fail = true; break; }
-        p = q;
+      if (!scan_one_interval(sp, data, block_offsets, ivs, n_ivs, j, aux, key,
+                             rk_save, &entries, &scanned, &matched, agg_val,
+                             agg_cnt, &ho, &walked_next)) {
+        errs += 1;
       }
-      // finalize the in-flight row (scan ended at tablet end)
-      if (!fail && row_open && rc.found &&
-          in_bounds(sp, rk_save, rk_len, aux)) {
-        uint64_t* sc = in_head ? &h_scanned : &scanned;
-        uint64_t* ma = in_head ? &h_matched : &matched;
-        *sc += 1;
-        if (eval_preds(sp, data, rc, rk_save, rk_len, aux)) {
-          *ma += 1;
-          acc_row(sp, rc, in_head ? h_val : agg_val,
-                  in_head ? h_cnt : agg_cnt);
-        }
-      }
-      if (fail) errs += 1;
+    } else {
+#pragma unroll
+      for (int g = 0; g < YBG_MAX_AGGS; ++g) { ho.val[g] = 0; ho.cnt[g] = 0; }
+      ho.scanned = ho.matched = 0;
     }
 
-    // --- resolve head ownership within the pass --------------------------
-    // lane l learns whether interval j-1's tail walk consumed j's head.
+    // resolve head ownership: lane l needs lane l-1's walked_next
     unsigned lane = threadIdx.x & 63;
     unsigned wave = threadIdx.x >> 6;
     int wn = walked_next ? 1 : 0;
     int from_prev_lane = __shfl_up(wn, 1);
     if (lane == 63) wave_relay[wave + 1] = (uint8_t)wn;
     __syncthreads();
-    bool head_consumed_by_prev;
+    bool head_consumed;
     if (threadIdx.x == 0) {
-      head_consumed_by_prev = false;  // resolved via global record instead
+      head_consumed = false;  // resolved via the global record instead
     } else if (lane == 0) {
-      head_consumed_by_prev = wave_relay[wave] != 0;
+      head_consumed = wave_relay[wave] != 0;
     } else {
-      head_consumed_by_prev = from_prev_lane != 0;
+      head_consumed = from_prev_lane != 0;
     }
     __syncthreads();
 
     if (active) {
       if (threadIdx.x == 0) {
-        // boundary interval: defer to global head record
-        uint64_t* hr = heads + (j / (kThreads)) * kHeadStride;
-        // index by j/kThreads: one record per workgroup-boundary interval
+        uint64_t* hr = heads + (j / kThreads) * kHeadStride;
 #pragma unroll
         for (int g = 0; g < YBG_MAX_AGGS; ++g) {
-          hr[2 * g] = h_val[g];
-          hr[2 * g + 1] = h_cnt[g];
+          hr[2 * g] = ho.val[g];
+          hr[2 * g + 1] = ho.cnt[g];
         }
-        hr[2 * YBG_MAX_AGGS] = h_scanned;
-        hr[2 * YBG_MAX_AGGS + 1] = h_matched;
-      } else if (!head_consumed_by_prev) {
-        scanned += h_scanned;
-        matched += h_matched;
-        agg_combine(sp, agg_val, agg_cnt, h_val, h_cnt);
+        hr[2 * YBG_MAX_AGGS] = ho.scanned;
+        hr[2 * YBG_MAX_AGGS + 1] = ho.matched;
+      } else if (!head_consumed) {
+        scanned += ho.scanned;
+        matched += ho.matched;
+        agg_combine(sp, agg_val, agg_cnt, ho.val, ho.cnt);
       }
-      // write the continuation flag for the NEXT boundary interval
       if (walked_next && ((j + 1) % kThreads) == 0 && (j + 1) < n_ivs) {
         cont_flags[(j + 1) / kThreads] = 1;
       }
@@ -1233,7 +125,7 @@ __global__ __launch_bounds__(kThreads) void k_scan(
     __syncthreads();
   }
 
-  // --- wave reduction into partials (fixed lane order => deterministic) --
+  // wave reduction into partials (fixed lane order => deterministic)
   {
     unsigned lane = threadIdx.x & 63;
 #pragma unroll
@@ -1251,10 +143,9 @@ __global__ __launch_bounds__(kThreads) void k_scan(
       __syncthreads();
       if (lane == 0) {
         uint64_t av = 0, ac = 0;
-        for (int l = 0; l < 64; ++l) {
+        for (int l = 0; l < 64; ++l)
           combine1(sp.aggs[g].op, &av, &ac, red_val[threadIdx.x + l],
                    red_cnt[threadIdx.x + l]);
-        }
         partials[wave_id * kPartialStride + 4 + 2 * g] = av;
         partials[wave_id * kPartialStride + 4 + 2 * g + 1] = ac;
       }
@@ -1286,12 +177,11 @@ __global__ __launch_bounds__(256) void k_reduce(
   __shared__ uint64_t sval[256], scnt[256], scal[256];
   const unsigned t = threadIdx.x;
 
-  // scalars (order-independent u64 sums)
   for (int s = 0; s < 4; ++s) {
     uint64_t acc = 0;
     for (uint64_t i = t; i < n_partials; i += 256)
       acc += partials[i * kPartialStride + s];
-    if (s == 1 || s == 2) {  // scanned/matched also in head records
+    if (s == 1 || s == 2) {  // scanned/matched also live in head records
       for (uint64_t i = t; i < n_heads; i += 256)
         if (cont_flags[i] == 0)
           acc += heads[i * kHeadStride + 2 * YBG_MAX_AGGS + (s - 1)];
@@ -1306,14 +196,12 @@ __global__ __launch_bounds__(256) void k_reduce(
     __syncthreads();
   }
 
-  // aggregates: fixed-chunk parallel fold, then fixed-order serial combine
   for (int g = 0; g < sp.num_aggs; ++g) {
     int op = sp.aggs[g].op;
     uint64_t av = 0, ac = 0;
-    for (uint64_t i = t; i < n_partials; i += 256) {
+    for (uint64_t i = t; i < n_partials; i += 256)
       combine1(op, &av, &ac, partials[i * kPartialStride + 4 + 2 * g],
                partials[i * kPartialStride + 4 + 2 * g + 1]);
-    }
     for (uint64_t i = t; i < n_heads; i += 256) {
       if (cont_flags[i] != 0) continue;
       combine1(op, &av, &ac, heads[i * kHeadStride + 2 * g],
@@ -1347,12 +235,15 @@ int set_err(int code, const std::string& msg) {
   return code;
 }
 
-#define HIP_TRY(x)                                                         \
-  do {                                                                     \
-    hipError_t _e = (x);                                                   \
-    if (_e != hipSuccess)                                                  \
-      return set_err(10, std::string(#x) + ": " + hipGetErrorString(_e));  \
+#define HIP_TRY(x)                                                        \
+  do {                                                                    \
+    hipError_t _e = (x);                                                  \
+    if (_e != hipSuccess)                                                 \
+      return set_err(10, std::string(#x) + ": " + hipGetErrorString(_e)); \
   } while (0)
+
+#define HIP_WARN(x) \
+  do { hipError_t _e = (x); (void)_e; } while (0)
 
 void htlim_from(const uint8_t* b, int32_t len, HtLim* out) {
   uint64_t hi = 0, lo = 0;
@@ -1373,7 +264,6 @@ struct ybg_scan {
   DevSpec dspec;
   hipStream_t stream = nullptr;
   hipEvent_t ev_start = nullptr, ev_mid = nullptr, ev_end = nullptr;
-  // device buffers
   uint8_t* d_data = nullptr;
   bool d_data_owned = false;
   uint64_t* d_offsets = nullptr;
@@ -1450,7 +340,8 @@ int yb_gpu_scan_open(const ybg_scan_spec_t* spec, ybg_scan_t** out) {
     switch (sc.value_cols[i].dtype) {
       case YBG_T_BOOL: case YBG_T_INT8: c.v2_fixed = 1; break;
       case YBG_T_INT16: c.v2_fixed = 2; break;
-      case YBG_T_INT32: case YBG_T_UINT32: case YBG_T_FLOAT: c.v2_fixed = 4; break;
+      case YBG_T_INT32: case YBG_T_UINT32: case YBG_T_FLOAT:
+        c.v2_fixed = 4; break;
       case YBG_T_STRING: c.v2_fixed = 0; break;
       default: c.v2_fixed = 8; break;
     }
@@ -1462,16 +353,15 @@ int yb_gpu_scan_open(const ybg_scan_spec_t* spec, ybg_scan_t** out) {
              &d.local_lim);
   htlim_from(spec->read_time.global_limit, spec->read_time.global_limit_len,
              &d.global_lim);
-  // regular_limit = memcmp-min(enc(read), enc(local)) == enc(max(read,local))
-  // (intent_aware_iterator.h:74-77)
   {
+    // regular_limit = memcmp-min(enc(read), enc(local)) == enc(max(read,
+    // local)) — intent_aware_iterator.h:74-77
     int rl = spec->read_time.read_len, ll = spec->read_time.local_limit_len;
     int n = rl < ll ? rl : ll;
     int cmp = memcmp(spec->read_time.local_limit, spec->read_time.read, n);
     bool local_smaller = cmp < 0 || (cmp == 0 && ll < rl);
     d.reg_lim = local_smaller ? d.local_lim : d.read;
   }
-  // aux buffer: predicate rhs strings + bounds
   d.num_preds = spec->num_preds;
   for (int i = 0; i < spec->num_preds; ++i) {
     const ybg_pred_t& p = spec->preds[i];
@@ -1535,7 +425,6 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
   HIP_TRY(hipMemcpy(s->d_offsets, offsets, (n_blocks + 1) * sizeof(uint64_t),
                     hipMemcpyHostToDevice));
 
-  // interval table: count restarts per block, host prefix-sum, emit
   uint32_t* d_counts;
   int* d_err;
   HIP_TRY(hipMalloc(&d_counts, n_blocks * sizeof(uint32_t)));
@@ -1546,16 +435,15 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
                      s->stream, s->d_data, s->d_offsets, n_blocks, d_counts,
                      d_err);
   std::vector<uint32_t> counts(n_blocks);
-  HIP_TRY(hipMemcpyAsync(counts.data(), d_counts,
-                         n_blocks * sizeof(uint32_t), hipMemcpyDeviceToHost,
-                         s->stream));
+  HIP_TRY(hipMemcpyAsync(counts.data(), d_counts, n_blocks * sizeof(uint32_t),
+                         hipMemcpyDeviceToHost, s->stream));
   int h_err = 0;
   HIP_TRY(hipMemcpyAsync(&h_err, d_err, sizeof(int), hipMemcpyDeviceToHost,
                          s->stream));
   HIP_TRY(hipStreamSynchronize(s->stream));
   if (h_err) {
-    hipFree(d_counts);
-    hipFree(d_err);
+    HIP_WARN(hipFree(d_counts));
+    HIP_WARN(hipFree(d_err));
     return set_err(3, "corrupt block trailer");
   }
   std::vector<uint64_t> base(n_blocks + 1);
@@ -1575,17 +463,16 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
                      s->stream, s->d_data, s->d_offsets, n_blocks, d_base,
                      s->d_ivs);
   HIP_TRY(hipStreamSynchronize(s->stream));
-  hipFree(d_base);
-  hipFree(d_counts);
-  hipFree(d_err);
+  HIP_WARN(hipFree(d_base));
+  HIP_WARN(hipFree(d_counts));
+  HIP_WARN(hipFree(d_err));
 
-  // launch geometry + working buffers
   uint64_t want = (s->n_ivs + kThreads - 1) / kThreads;
   s->grid = (int)std::min<uint64_t>(want, 4096);
   if (s->grid < 1) s->grid = 1;
   uint64_t span_threads = (uint64_t)s->grid * kThreads;
   s->n_partials = span_threads / 64;
-  s->n_heads = (s->n_ivs + kThreads - 1) / kThreads;  // 1 per wg-boundary iv
+  s->n_heads = (s->n_ivs + kThreads - 1) / kThreads;
   HIP_TRY(hipMalloc(&s->d_rk_save, span_threads * kKeyCap));
   HIP_TRY(hipMalloc(&s->d_partials,
                     s->n_partials * kPartialStride * sizeof(uint64_t)));
@@ -1644,9 +531,9 @@ int yb_gpu_scan_aggregate(ybg_scan_t* s, ybg_scan_result_t* out) {
       case YBG_AGG_SUM_DOUBLE:
       case YBG_AGG_MIN_DOUBLE:
       case YBG_AGG_MAX_DOUBLE: {
-        double d;
-        memcpy(&d, &r.agg_val[g], 8);
-        a.value_f64 = d;
+        double dd;
+        memcpy(&dd, &r.agg_val[g], 8);
+        a.value_f64 = dd;
         break;
       }
       default:
@@ -1679,19 +566,19 @@ int yb_gpu_scan_kernel_ms(ybg_scan_t* s, double* total_ms, double* decode_ms) {
 }
 
 int yb_gpu_scan_close(ybg_scan_t* s) {
-  if (s->d_data_owned && s->d_data) hipFree(s->d_data);
-  if (s->d_offsets) hipFree(s->d_offsets);
-  if (s->d_ivs) hipFree(s->d_ivs);
-  if (s->d_aux) hipFree(s->d_aux);
-  if (s->d_rk_save) hipFree(s->d_rk_save);
-  if (s->d_partials) hipFree(s->d_partials);
-  if (s->d_heads) hipFree(s->d_heads);
-  if (s->d_cont) hipFree(s->d_cont);
-  if (s->d_result) hipFree(s->d_result);
-  if (s->ev_start) hipEventDestroy(s->ev_start);
-  if (s->ev_mid) hipEventDestroy(s->ev_mid);
-  if (s->ev_end) hipEventDestroy(s->ev_end);
-  if (s->stream) hipStreamDestroy(s->stream);
+  if (s->d_data_owned && s->d_data) HIP_WARN(hipFree(s->d_data));
+  if (s->d_offsets) HIP_WARN(hipFree(s->d_offsets));
+  if (s->d_ivs) HIP_WARN(hipFree(s->d_ivs));
+  if (s->d_aux) HIP_WARN(hipFree(s->d_aux));
+  if (s->d_rk_save) HIP_WARN(hipFree(s->d_rk_save));
+  if (s->d_partials) HIP_WARN(hipFree(s->d_partials));
+  if (s->d_heads) HIP_WARN(hipFree(s->d_heads));
+  if (s->d_cont) HIP_WARN(hipFree(s->d_cont));
+  if (s->d_result) HIP_WARN(hipFree(s->d_result));
+  if (s->ev_start) HIP_WARN(hipEventDestroy(s->ev_start));
+  if (s->ev_mid) HIP_WARN(hipEventDestroy(s->ev_mid));
+  if (s->ev_end) HIP_WARN(hipEventDestroy(s->ev_end));
+  if (s->stream) HIP_WARN(hipStreamDestroy(s->stream));
   delete s;
   return 0;
 }
