@@ -233,9 +233,13 @@ def run_oracle(case, read_micros, preds, aggs, lower=None, upper=None):
     return res
 
 
-def check_match(gres, ores, aggs, f64_rel=1e-12):
-    assert gres.entries_seen == ores.entries_seen, \
-        (gres.entries_seen, ores.entries_seen)
+def check_match(gres, ores, aggs, f64_rel=1e-12, check_entries=True):
+    if check_entries:
+        # the oracle stops at the upper bound like the reference iterator;
+        # the GPU filters instead, so entries_seen is compared only for
+        # unbounded scans.
+        assert gres.entries_seen == ores.entries_seen, \
+            (gres.entries_seen, ores.entries_seen)
     assert gres.rows_scanned == ores.rows_scanned, \
         (gres.rows_scanned, ores.rows_scanned)
     assert gres.rows_matched == ores.rows_matched, \
