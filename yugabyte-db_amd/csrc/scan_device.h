@@ -76,6 +76,7 @@ struct DevSpec {
   DevAgg aggs[YBG_MAX_AGGS];
   uint32_t lower_len, lower_off, upper_len, upper_off;
   int32_t v1_varlen_count;
+  uint32_t value_pred_mask;  // bit i: predicate i is on a value column
 };
 
 struct Interval {
@@ -494,74 +495,122 @@ DEV int skip_control(const uint8_t* v, uint32_t len) {
   return (int)off;
 }
 
+
 // ---------------------------------------------------------------------------
-// Streaming row context
+// Streaming row context (templated on the aggregate-slot capacity NA so the
+// per-thread register state shrinks to what the query needs). Value-column
+// predicates are evaluated EAGERLY at column decode into a pass bitmask —
+// no operand values are retained; aggregate operands keep only a datum +
+// null bit per slot.
 // ---------------------------------------------------------------------------
 
-struct Operand {
-  uint64_t datum;     // numeric bit pattern, or (str global offset)
-  uint32_t str_len;
-  bool is_null;
-};
-
-struct RowCtx {
+template <int NA>
+struct RowCtxT {
   bool base_seen;
   bool found;
   uint64_t base_ht_hi, base_ht_lo;
   uint32_t base_ht_len;
-  int32_t cur_col;     // current column-update group (-2 liveness)
+  int32_t cur_col;  // current column-update group (-2 liveness, -1 none)
   bool cur_col_done;
-  Operand pred_op[YBG_MAX_PREDS];
-  Operand agg_op[YBG_MAX_AGGS];
+  uint32_t pred_pass;  // bit i: value-col predicate i passes (key preds at
+                       // finalize)
+  uint32_t agg_null;   // bit g: aggregate operand g is NULL
+  uint64_t agg_datum[NA];
 };
 
-DEV void row_reset(RowCtx* rc, const DevSpec& sp) {
+template <int NA>
+DEV void row_reset(RowCtxT<NA>* rc, const DevSpec& sp) {
   rc->base_seen = false;
   rc->found = false;
   rc->cur_col = -1;
   rc->cur_col_done = false;
-#pragma unroll
-  for (int i = 0; i < YBG_MAX_PREDS; ++i) rc->pred_op[i].is_null = true;
-#pragma unroll
-  for (int i = 0; i < YBG_MAX_AGGS; ++i) rc->agg_op[i].is_null = true;
+  rc->pred_pass = 0;
+  rc->agg_null = 0xffffffffu;
 }
 
-DEV void set_col_operand(RowCtx* rc, const DevSpec& sp, int col, bool is_null,
-                         uint64_t datum, uint32_t str_len) {
+// Value-column predicate compare (pgsql_operation.cc:602-668 typed-compare
+// subset). sptr/slen only for string columns.
+DEV bool pred_compare(const DevPred& pr, int dtype, uint64_t datum,
+                      const uint8_t* sptr, uint32_t slen, const uint8_t* aux) {
+  int cmp;
+  if (dtype == YBG_T_STRING) {
+    const uint8_t* rhs = aux + pr.rhs_off;
+    uint32_t n = slen < pr.str_len ? slen : pr.str_len;
+    cmp = 0;
+    for (uint32_t k = 0; k < n; ++k) {
+      if (sptr[k] != rhs[k]) {
+        cmp = sptr[k] < rhs[k] ? -1 : 1;
+        break;
+      }
+    }
+    if (cmp == 0 && slen != pr.str_len) cmp = slen < pr.str_len ? -1 : 1;
+  } else if (dtype == YBG_T_DOUBLE) {
+    double a = __longlong_as_double((long long)datum);
+    double b = __longlong_as_double((long long)pr.datum);
+    cmp = a < b ? -1 : (a > b ? 1 : 0);
+  } else if (dtype == YBG_T_FLOAT) {
+    float a = __uint_as_float((uint32_t)datum);
+    float b = __uint_as_float((uint32_t)pr.datum);
+    cmp = a < b ? -1 : (a > b ? 1 : 0);
+  } else {
+    int64_t a = (int64_t)datum, b = (int64_t)pr.datum;
+    cmp = a < b ? -1 : (a > b ? 1 : 0);
+  }
+  switch (pr.op) {
+    case YBG_PRED_GT: return cmp > 0;
+    case YBG_PRED_GE: return cmp >= 0;
+    case YBG_PRED_LT: return cmp < 0;
+    case YBG_PRED_LE: return cmp <= 0;
+    case YBG_PRED_EQ: return cmp == 0;
+    default: return cmp != 0;
+  }
+}
+
+// Column value arrives (from packed row or column update): evaluate the
+// predicates that reference it and stash aggregate operands.
+template <int NA>
+DEV void eval_col(const DevSpec& sp, const uint8_t* aux, RowCtxT<NA>* rc,
+                  int col, bool is_null, uint64_t datum, const uint8_t* sptr,
+                  uint32_t slen) {
 #pragma unroll
   for (int i = 0; i < YBG_MAX_PREDS; ++i) {
-    if (i < sp.num_preds && !sp.preds[i].is_key_col && sp.preds[i].col == col) {
-      rc->pred_op[i].is_null = is_null;
-      rc->pred_op[i].datum = datum;
-      rc->pred_op[i].str_len = str_len;
+    if (i < sp.num_preds && !sp.preds[i].is_key_col &&
+        sp.preds[i].col == col) {
+      bool pass =
+          !is_null && pred_compare(sp.preds[i], sp.cols[col].dtype, datum,
+                                   sptr, slen, aux);
+      rc->pred_pass = (rc->pred_pass & ~(1u << i)) | ((uint32_t)pass << i);
     }
   }
 #pragma unroll
-  for (int i = 0; i < YBG_MAX_AGGS; ++i) {
-    if (i < sp.num_aggs && sp.aggs[i].col == col &&
-        sp.aggs[i].op != YBG_AGG_COUNT_STAR) {
-      rc->agg_op[i].is_null = is_null;
-      rc->agg_op[i].datum = datum;
-      rc->agg_op[i].str_len = str_len;
+  for (int g = 0; g < NA; ++g) {
+    if (g < sp.num_aggs && sp.aggs[g].col == col &&
+        sp.aggs[g].op != YBG_AGG_COUNT_STAR) {
+      rc->agg_datum[g] = datum;
+      rc->agg_null =
+          (rc->agg_null & ~(1u << g)) | ((uint32_t)is_null << g);
     }
   }
 }
 
 // Decode one V1-encoded single value (primitive_value.cc:1066-1125).
 // Returns 0 null/tombstone, 1 applied, -1 error.
-DEV int decode_single_v1(const DevSpec& sp, const uint8_t* base, int col,
-                         const uint8_t* vp, uint32_t vlen, RowCtx* rc) {
+template <int NA>
+DEV int decode_single_v1(const DevSpec& sp, const uint8_t* base,
+                         const uint8_t* aux, int col, const uint8_t* vp,
+                         uint32_t vlen, RowCtxT<NA>* rc) {
   if (vlen == 0) {
-    set_col_operand(rc, sp, col, true, 0, 0);
+    eval_col(sp, aux, rc, col, true, 0, nullptr, 0);
     return 0;
   }
   uint8_t t = vp[0];
   if (t == kTombB || t == kNullLow) {
-    set_col_operand(rc, sp, col, true, 0, 0);
+    eval_col(sp, aux, rc, col, true, 0, nullptr, 0);
     return 0;
   }
   const DevCol& c = sp.cols[col];
   uint64_t datum = 0;
+  const uint8_t* sptr = nullptr;
   uint32_t slen = 0;
   switch (c.dtype) {
     case YBG_T_BOOL:
@@ -596,19 +645,23 @@ DEV int decode_single_v1(const DevSpec& sp, const uint8_t* base, int col,
       break;
     case YBG_T_STRING:
       if (t != kStringB) return -1;
-      datum = (uint64_t)(vp + 1 - base);
+      sptr = vp + 1;
       slen = vlen - 1;
       break;
     default:
       return -1;
   }
-  set_col_operand(rc, sp, col, false, datum, slen);
+  eval_col(sp, aux, rc, col, false, datum, sptr, slen);
+  (void)base;
   return 1;
 }
 
-// Packed row decode streaming into operand slots. `body` points at 'z'/'|'.
+// Packed row decode streaming into predicate/aggregate slots.
+// `body` points at 'z'/'|'.
+template <int NA>
 DEV bool decode_packed(const DevSpec& sp, const uint8_t* base,
-                       const uint8_t* body, uint32_t len, RowCtx* rc) {
+                       const uint8_t* aux, const uint8_t* body, uint32_t len,
+                       RowCtxT<NA>* rc) {
   uint8_t kind = body[0];
   uint32_t off = 1;
   uint64_t version;
@@ -640,7 +693,8 @@ DEV bool decode_packed(const DevSpec& sp, const uint8_t* base,
         end = start + fs;
       }
       if (end < start || end > data_len) return false;
-      int r = decode_single_v1(sp, base, i, data + start, end - start, rc);
+      int r = decode_single_v1(sp, base, aux, i, data + start, end - start,
+                               rc);
       if (r < 0) return false;
     }
     return true;
@@ -659,7 +713,7 @@ DEV bool decode_packed(const DevSpec& sp, const uint8_t* base,
     const uint8_t* end = body + len;
     for (int i = 0; i < sp.num_value_cols; ++i) {
       if (null_mask && (null_mask[i >> 3] & (1 << (i & 7)))) {
-        set_col_operand(rc, sp, i, true, 0, 0);
+        eval_col(sp, aux, rc, i, true, 0, nullptr, 0);
         continue;
       }
       const DevCol& c = sp.cols[i];
@@ -675,7 +729,7 @@ DEV bool decode_packed(const DevSpec& sp, const uint8_t* base,
         if (c.dtype == YBG_T_INT8) u = (uint64_t)(int64_t)(int8_t)u;
         else if (c.dtype == YBG_T_INT16) u = (uint64_t)(int64_t)(int16_t)u;
         else if (c.dtype == YBG_T_INT32) u = (uint64_t)(int64_t)(int32_t)u;
-        set_col_operand(rc, sp, i, false, u, 0);
+        eval_col(sp, aux, rc, i, false, u, nullptr, 0);
         data += c.v2_fixed;
       } else {
         if (data >= end) return false;
@@ -690,7 +744,7 @@ DEV bool decode_packed(const DevSpec& sp, const uint8_t* base,
         }
         data += consumed;
         if (data + flen > end) return false;
-        set_col_operand(rc, sp, i, false, (uint64_t)(data - base), flen);
+        eval_col(sp, aux, rc, i, false, 0, data, flen);
         data += flen;
       }
     }
@@ -712,8 +766,8 @@ DEV bool key_col_value(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
       int kt = sp.key_types[col];
       if (kt == YBG_KT_INT64) {
         if (col == target) {
-          *out_datum =
-              (uint64_t)(int64_t)(load_be64(rk + off + 1) ^ 0x8000000000000000ull);
+          *out_datum = (uint64_t)(int64_t)(load_be64(rk + off + 1) ^
+                                           0x8000000000000000ull);
           *out_slen = 0;
           return true;
         }
@@ -743,83 +797,39 @@ DEV bool key_col_value(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
   return false;
 }
 
-// Predicates on streamed operands + saved rowkey
-// (pgsql_operation.cc:602-668 typed-compare subset; NULL => filtered).
-DEV bool eval_preds(const DevSpec& sp, const uint8_t* base, const RowCtx& rc,
-                    const uint8_t* rk, uint32_t rk_len, const uint8_t* aux) {
-  for (int i = 0; i < sp.num_preds; ++i) {
+// Key-column predicates (evaluated at row finalize; the rowkey bytes come
+// from the saved row key). Key strings are zero-escaped in the key
+// (doc_kv_util.h:101-167): unescape on the fly.
+DEV bool eval_key_preds(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
+                        const uint8_t* aux) {
+#pragma unroll
+  for (int i = 0; i < YBG_MAX_PREDS; ++i) {
+    if (i >= sp.num_preds || !sp.preds[i].is_key_col) continue;
     const DevPred& pr = sp.preds[i];
-    int cmp = 0;
-    int dt;
-    uint64_t lhs_datum = 0;
-    const uint8_t* lstr = nullptr;
-    uint32_t lstr_len = 0;
-    bool key_string_escaped = false;
-    if (pr.is_key_col) {
-      uint64_t d;
-      uint32_t soff = 0, sl = 0;
-      if (!key_col_value(sp, rk, rk_len, pr.col, &d, &soff, &sl)) return false;
-      int kt = sp.key_types[pr.col];
-      if (kt == YBG_KT_STRING) {
-        lstr = rk + soff;
-        lstr_len = sl;
-        key_string_escaped = true;
-        dt = YBG_T_STRING;
-      } else {
-        lhs_datum = d;
-        dt = YBG_T_INT64;
-      }
-    } else {
-      const Operand& op = rc.pred_op[i];
-      if (op.is_null) return false;
-      dt = sp.cols[pr.col].dtype;
-      if (dt == YBG_T_STRING) {
-        lstr = base + op.datum;
-        lstr_len = op.str_len;
-      } else {
-        lhs_datum = op.datum;
-      }
-    }
-    if (dt == YBG_T_STRING) {
+    uint64_t d = 0;
+    uint32_t soff = 0, sl = 0;
+    if (!key_col_value(sp, rk, rk_len, pr.col, &d, &soff, &sl)) return false;
+    int cmp;
+    if (sp.key_types[pr.col] == YBG_KT_STRING) {
+      const uint8_t* lstr = rk + soff;
       const uint8_t* rhs = aux + pr.rhs_off;
-      if (key_string_escaped) {
-        // zero-escaped lhs ('\0'->"\0\1"): unescape on the fly
-        uint32_t si = 0, ri = 0;
-        cmp = 0;
-        while (si < lstr_len && ri < pr.str_len) {
-          uint8_t cb = lstr[si];
-          si += (cb == 0) ? 2 : 1;
-          if (cb != rhs[ri]) {
-            cmp = cb < rhs[ri] ? -1 : 1;
-            break;
-          }
-          ++ri;
+      uint32_t si = 0, ri = 0;
+      cmp = 0;
+      while (si < sl && ri < pr.str_len) {
+        uint8_t cb = lstr[si];
+        si += (cb == 0) ? 2 : 1;
+        if (cb != rhs[ri]) {
+          cmp = cb < rhs[ri] ? -1 : 1;
+          break;
         }
-        if (cmp == 0) {
-          bool le = si >= lstr_len, re = ri >= pr.str_len;
-          cmp = (le && re) ? 0 : (le ? -1 : 1);
-        }
-      } else {
-        uint32_t n = lstr_len < pr.str_len ? lstr_len : pr.str_len;
-        for (uint32_t k = 0; k < n; ++k) {
-          if (lstr[k] != rhs[k]) {
-            cmp = lstr[k] < rhs[k] ? -1 : 1;
-            break;
-          }
-        }
-        if (cmp == 0 && lstr_len != pr.str_len)
-          cmp = lstr_len < pr.str_len ? -1 : 1;
+        ++ri;
       }
-    } else if (dt == YBG_T_DOUBLE) {
-      double a = __longlong_as_double((long long)lhs_datum);
-      double b = __longlong_as_double((long long)pr.datum);
-      cmp = a < b ? -1 : (a > b ? 1 : 0);
-    } else if (dt == YBG_T_FLOAT) {
-      float a = __uint_as_float((uint32_t)lhs_datum);
-      float b = __uint_as_float((uint32_t)pr.datum);
-      cmp = a < b ? -1 : (a > b ? 1 : 0);
+      if (cmp == 0) {
+        bool le = si >= sl, re = ri >= pr.str_len;
+        cmp = (le && re) ? 0 : (le ? -1 : 1);
+      }
     } else {
-      int64_t a = (int64_t)lhs_datum, b = (int64_t)pr.datum;
+      int64_t a = (int64_t)d, b = (int64_t)pr.datum;
       cmp = a < b ? -1 : (a > b ? 1 : 0);
     }
     bool pass;
@@ -863,15 +873,17 @@ DEV bool in_bounds(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
 
 // Aggregate accumulate (doc_expr.cc:248-395); cnt counts non-null
 // contributions (SUM/MIN/MAX null-ness = cnt == 0).
-DEV void acc_row(const DevSpec& sp, const RowCtx& rc, uint64_t* agg_val,
+template <int NA>
+DEV void acc_row(const DevSpec& sp, const RowCtxT<NA>& rc, uint64_t* agg_val,
                  uint64_t* agg_cnt) {
 #pragma unroll
-  for (int g = 0; g < YBG_MAX_AGGS; ++g) {
-    if (g >= sp.num_aggs) break;
+  for (int g = 0; g < NA; ++g) {
+    if (g >= sp.num_aggs) continue;
     const DevAgg& a = sp.aggs[g];
-    bool isnull = (a.op == YBG_AGG_COUNT_STAR) ? false : rc.agg_op[g].is_null;
+    bool isnull =
+        (a.op == YBG_AGG_COUNT_STAR) ? false : ((rc.agg_null >> g) & 1);
     if (isnull) continue;
-    uint64_t v = rc.agg_op[g].datum;
+    uint64_t v = rc.agg_datum[g];
     switch (a.op) {
       case YBG_AGG_COUNT_STAR:
       case YBG_AGG_COUNT:
@@ -893,15 +905,15 @@ DEV void acc_row(const DevSpec& sp, const RowCtx& rc, uint64_t* agg_val,
         if (agg_cnt[g] == 0 || (int64_t)v > (int64_t)agg_val[g]) agg_val[g] = v;
         break;
       case YBG_AGG_MIN_DOUBLE: {
-        double d = __longlong_as_double((long long)v);
-        if (agg_cnt[g] == 0 || d < __longlong_as_double((long long)agg_val[g]))
-          agg_val[g] = (uint64_t)__double_as_longlong(d);
+        double dd = __longlong_as_double((long long)v);
+        if (agg_cnt[g] == 0 || dd < __longlong_as_double((long long)agg_val[g]))
+          agg_val[g] = (uint64_t)__double_as_longlong(dd);
         break;
       }
       case YBG_AGG_MAX_DOUBLE: {
-        double d = __longlong_as_double((long long)v);
-        if (agg_cnt[g] == 0 || d > __longlong_as_double((long long)agg_val[g]))
-          agg_val[g] = (uint64_t)__double_as_longlong(d);
+        double dd = __longlong_as_double((long long)v);
+        if (agg_cnt[g] == 0 || dd > __longlong_as_double((long long)agg_val[g]))
+          agg_val[g] = (uint64_t)__double_as_longlong(dd);
         break;
       }
     }
@@ -951,17 +963,19 @@ DEV void agg_combine(const DevSpec& sp, uint64_t* a_val, uint64_t* a_cnt,
                      const uint64_t* b_val, const uint64_t* b_cnt) {
 #pragma unroll
   for (int g = 0; g < YBG_MAX_AGGS; ++g) {
-    if (g >= sp.num_aggs) break;
+    if (g >= sp.num_aggs) continue;
     combine1(sp.aggs[g].op, &a_val[g], &a_cnt[g], b_val[g], b_cnt[g]);
   }
 }
 
 // Visibility + row-state update for one entry. key/rowkey live in LDS.
 // Returns false on corruption.
+template <int NA>
 DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
-                       const uint8_t* key, uint32_t key_len,
-                       const uint8_t* value, uint32_t value_len,
-                       uint32_t rowkey_len, RowCtx* rc) {
+                       const uint8_t* aux, const uint8_t* key,
+                       uint32_t key_len, const uint8_t* value,
+                       uint32_t value_len, uint32_t rowkey_len,
+                       RowCtxT<NA>* rc) {
   uint32_t ukey_len = key_len - 8;
   uint32_t ht_size = key[ukey_len - 1] & 0x1f;
   const uint8_t* ht_enc = key + ukey_len - ht_size;
@@ -971,14 +985,16 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
   slice_u128(ht_enc, ht_size, &ht_hi, &ht_lo);
   bool visible;
   if (value_len > 0 && value[0] == kHybridTimeByte) {
-    // committed-txn record with intent time (:1249-1267)
+    // committed-txn record with intent time
+    // (intent_aware_iterator.cc:1249-1267)
     uint64_t v_hi, v_lo;
     slice_u128(value + 1, value_len - 1, &v_hi, &v_lo);
     bool use_global =
         u128_slice_cmp(v_hi, v_lo, value_len - 1, sp.local_lim.hi,
                        sp.local_lim.lo, sp.local_lim.len) > 0;
     const HtLim& lim = use_global ? sp.global_lim : sp.read;
-    visible = u128_slice_cmp(ht_hi, ht_lo, ht_size, lim.hi, lim.lo, lim.len) >= 0;
+    visible =
+        u128_slice_cmp(ht_hi, ht_lo, ht_size, lim.hi, lim.lo, lim.len) >= 0;
     if (visible) {
       int iht = dht_size_from_start(value + 1, value + value_len);
       if (!iht) return false;
@@ -1002,7 +1018,7 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
       const uint8_t* body = value + cf;
       uint32_t body_len = value_len - cf;
       if (body_len > 0 && (body[0] == kPackedV1B || body[0] == kPackedV2B)) {
-        if (!decode_packed(sp, base, body, body_len, rc)) return false;
+        if (!decode_packed(sp, base, aux, body, body_len, rc)) return false;
         rc->found = true;  // doc_reader.cc:1894-1900
       }
     }
@@ -1047,8 +1063,8 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
         if (newer) {
           int cf = skip_control(value, value_len);
           if (cf < 0) return false;
-          int r =
-              decode_single_v1(sp, base, idx, value + cf, value_len - cf, rc);
+          int r = decode_single_v1(sp, base, aux, idx, value + cf,
+                                   value_len - cf, rc);
           if (r < 0) return false;
           if (r > 0) rc->found = true;
         }
@@ -1063,25 +1079,22 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
 // ---------------------------------------------------------------------------
 // One interval, the whole algorithm (head deferral + tail walk). Called by
 // the k_scan kernel (one thread per interval) and by the host simulator.
-// key: kKeyCap scratch (LDS on device); rk_save: kKeyCap row-key save.
-// Outputs: committed contributions added into (entries..agg_cnt); the head
-// row's contribution into h_*; walked_next = this interval's tail walk
-// consumed >= 1 entry of interval j+1 (== continuation flag for j+1).
-// Returns false on corrupt data.
 // ---------------------------------------------------------------------------
+template <int NA>
 struct HeadOut {
-  uint64_t val[YBG_MAX_AGGS];
-  uint64_t cnt[YBG_MAX_AGGS];
+  uint64_t val[NA];
+  uint64_t cnt[NA];
   uint64_t scanned, matched;
 };
 
+template <int NA>
 DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
                            const uint64_t* block_offsets, const Interval* ivs,
                            uint64_t n_ivs, uint64_t j, const uint8_t* aux,
                            uint8_t* key, uint8_t* rk_save, uint64_t* entries,
                            uint64_t* scanned, uint64_t* matched,
-                           uint64_t* agg_val, uint64_t* agg_cnt, HeadOut* ho,
-                           bool* walked_next_out) {
+                           uint64_t* agg_val, uint64_t* agg_cnt,
+                           HeadOut<NA>* ho, bool* walked_next_out) {
   Interval iv = ivs[j];
   const uint8_t* blk = data + block_offsets[iv.block];
   const uint8_t* p = blk + iv.start;
@@ -1089,7 +1102,7 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
   uint64_t cur_iv = j;
 
 #pragma unroll
-  for (int g = 0; g < YBG_MAX_AGGS; ++g) { ho->val[g] = 0; ho->cnt[g] = 0; }
+  for (int g = 0; g < NA; ++g) { ho->val[g] = 0; ho->cnt[g] = 0; }
   ho->scanned = 0;
   ho->matched = 0;
   bool walked_next = false;
@@ -1099,7 +1112,7 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
   uint32_t rk_len = 0;
   bool row_open = false;
   bool in_head = true;
-  RowCtx rc;
+  RowCtxT<NA> rc;
   row_reset(&rc, sp);
   EntryRef er;
 
@@ -1111,6 +1124,8 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
       blk = data + block_offsets[nx.block];
       p = blk + nx.start;
       limit = blk + nx.end;
+      // first entry of a restart interval is self-contained; the carried
+      // key state is overwritten by its full-key decode.
       continue;
     }
     const uint8_t* q = decode_entry(sp.fmt, p, limit, key, &key_len, &er);
@@ -1131,11 +1146,13 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
     }
     if (row_change) {
       if (row_open) {
+        // finalize previous row (doc_rowwise_iterator row boundary)
         if (rc.found && in_bounds(sp, rk_save, rk_len, aux)) {
           uint64_t* sc = in_head ? &ho->scanned : scanned;
           uint64_t* ma = in_head ? &ho->matched : matched;
           *sc += 1;
-          if (eval_preds(sp, data, rc, rk_save, rk_len, aux)) {
+          if ((rc.pred_pass & sp.value_pred_mask) == sp.value_pred_mask &&
+              eval_key_preds(sp, rk_save, rk_len, aux)) {
             *ma += 1;
             acc_row(sp, rc, in_head ? ho->val : agg_val,
                     in_head ? ho->cnt : agg_cnt);
@@ -1152,7 +1169,7 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
     }
     if (cur_iv == j) *entries += 1;
     else if (cur_iv == j + 1) walked_next = true;
-    if (!process_entry(sp, data, key, key_len, er.value, er.value_len,
+    if (!process_entry(sp, data, aux, key, key_len, er.value, er.value_len,
                        rk_len, &rc)) { fail = true; break; }
     p = q;
   }
@@ -1160,13 +1177,117 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
     uint64_t* sc = in_head ? &ho->scanned : scanned;
     uint64_t* ma = in_head ? &ho->matched : matched;
     *sc += 1;
-    if (eval_preds(sp, data, rc, rk_save, rk_len, aux)) {
+    if ((rc.pred_pass & sp.value_pred_mask) == sp.value_pred_mask &&
+        eval_key_preds(sp, rk_save, rk_len, aux)) {
       *ma += 1;
       acc_row(sp, rc, in_head ? ho->val : agg_val, in_head ? ho->cnt : agg_cnt);
     }
   }
   *walked_next_out = walked_next;
   return !fail;
+}
+
+// ---------------------------------------------------------------------------
+// Host-side spec translation (shared by the ABI and the host simulator).
+// ---------------------------------------------------------------------------
+inline void build_dev_spec(const ybg_scan_spec_t* spec, DevSpec* dp,
+                           unsigned char* aux, uint32_t* aux_len,
+                           uint32_t aux_cap) {
+  DevSpec& d = *dp;
+  memset(&d, 0, sizeof(d));
+  const ybg_schema_t& sc = spec->schema;
+  d.has_hash = sc.has_hash;
+  d.num_hash_cols = sc.num_hash_cols;
+  d.num_range_cols = sc.num_range_cols;
+  for (int i = 0; i < YBG_MAX_KEYCOLS; ++i) d.key_types[i] = sc.key_types[i];
+  d.num_value_cols = sc.num_value_cols;
+  int nvar = 0, off_after = 0;
+  for (int i = 0; i < sc.num_value_cols; ++i) {
+    DevCol& c = d.cols[i];
+    c.id = sc.value_cols[i].column_id;
+    c.dtype = sc.value_cols[i].dtype;
+    bool varlen =
+        sc.value_cols[i].nullable || sc.value_cols[i].dtype == YBG_T_STRING;
+    c.v1_varlen = varlen;
+    c.v1_nvb = nvar;
+    c.v1_off = off_after;
+    int v1sz;
+    switch (sc.value_cols[i].dtype) {
+      case YBG_T_BOOL: v1sz = 1; break;
+      case YBG_T_INT8: case YBG_T_INT16: case YBG_T_INT32:
+      case YBG_T_UINT32: case YBG_T_FLOAT: v1sz = 5; break;
+      default: v1sz = 9; break;
+    }
+    if (varlen) { ++nvar; off_after = 0; }
+    else off_after += v1sz;
+    switch (sc.value_cols[i].dtype) {
+      case YBG_T_BOOL: case YBG_T_INT8: c.v2_fixed = 1; break;
+      case YBG_T_INT16: c.v2_fixed = 2; break;
+      case YBG_T_INT32: case YBG_T_UINT32: case YBG_T_FLOAT:
+        c.v2_fixed = 4; break;
+      case YBG_T_STRING: c.v2_fixed = 0; break;
+      default: c.v2_fixed = 8; break;
+    }
+  }
+  d.v1_varlen_count = nvar;
+  d.fmt = spec->kv_format;
+  auto htlim = [](const uint8_t* b, int32_t len, HtLim* o) {
+    uint64_t hi = 0, lo = 0;
+    for (int i = 0; i < len && i < 16; ++i) {
+      uint64_t v = b[i];
+      if (i < 8) hi |= v << (56 - 8 * i);
+      else lo |= v << (56 - 8 * (i - 8));
+    }
+    o->hi = hi;
+    o->lo = lo;
+    o->len = (uint32_t)len;
+  };
+  htlim(spec->read_time.read, spec->read_time.read_len, &d.read);
+  htlim(spec->read_time.local_limit, spec->read_time.local_limit_len,
+        &d.local_lim);
+  htlim(spec->read_time.global_limit, spec->read_time.global_limit_len,
+        &d.global_lim);
+  {
+    // regular_limit = memcmp-min(enc(read), enc(local)) == enc(max(read,
+    // local)) — intent_aware_iterator.h:74-77
+    int rl = spec->read_time.read_len, ll = spec->read_time.local_limit_len;
+    int n = rl < ll ? rl : ll;
+    int cmp = memcmp(spec->read_time.local_limit, spec->read_time.read, n);
+    bool local_smaller = cmp < 0 || (cmp == 0 && ll < rl);
+    d.reg_lim = local_smaller ? d.local_lim : d.read;
+  }
+  uint32_t pos = 0;
+  auto put = [&](const uint8_t* b, uint64_t n) {
+    uint32_t off = pos;
+    if (b && n && pos + n <= aux_cap) {
+      memcpy(aux + pos, b, n);
+      pos += (uint32_t)n;
+    }
+    return off;
+  };
+  d.num_preds = spec->num_preds;
+  d.value_pred_mask = 0;
+  for (int i = 0; i < spec->num_preds; ++i) {
+    const ybg_pred_t& p = spec->preds[i];
+    DevPred& pr = d.preds[i];
+    pr.is_key_col = p.is_key_col;
+    pr.col = p.col;
+    pr.op = p.op;
+    pr.datum = p.datum;
+    pr.str_len = (uint32_t)p.bytes_len;
+    pr.rhs_off = put(p.bytes, p.bytes_len);
+    if (!p.is_key_col) d.value_pred_mask |= 1u << i;
+  }
+  d.num_aggs = spec->num_aggs;
+  for (int i = 0; i < spec->num_aggs; ++i) {
+    d.aggs[i].op = spec->aggs[i].op;
+    d.aggs[i].col = spec->aggs[i].col;
+  }
+  d.lower_len = (uint32_t)spec->lower_bound_len;
+  d.lower_off = put(spec->lower_bound, spec->lower_bound_len);
+  d.upper_len = (uint32_t)spec->upper_bound_len;
+  d.upper_off = put(spec->upper_bound, spec->upper_bound_len);
+  *aux_len = pos ? pos : 1;
 }
 
 }  // namespace ybgdev

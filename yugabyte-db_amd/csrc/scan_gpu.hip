@@ -50,6 +50,7 @@ namespace {
 // Phase-1 scan kernel (aggregate mode): one thread per interval
 // ---------------------------------------------------------------------------
 
+template <int NA>
 __global__ __launch_bounds__(kThreads) void k_scan(
     DevSpec sp, const uint8_t* __restrict__ data,
     const uint64_t* __restrict__ block_offsets,
@@ -72,17 +73,17 @@ __global__ __launch_bounds__(kThreads) void k_scan(
   for (uint64_t j0 = 0; j0 < n_ivs; j0 += span) {
     const uint64_t j = j0 + gtid;
     const bool active = j < n_ivs;
-    HeadOut ho;
+    HeadOut<NA> ho;
     bool walked_next = false;
     if (active) {
-      if (!scan_one_interval(sp, data, block_offsets, ivs, n_ivs, j, aux, key,
-                             rk_save, &entries, &scanned, &matched, agg_val,
-                             agg_cnt, &ho, &walked_next)) {
+      if (!scan_one_interval<NA>(sp, data, block_offsets, ivs, n_ivs, j, aux,
+                                 key, rk_save, &entries, &scanned, &matched,
+                                 agg_val, agg_cnt, &ho, &walked_next)) {
         errs += 1;
       }
     } else {
 #pragma unroll
-      for (int g = 0; g < YBG_MAX_AGGS; ++g) { ho.val[g] = 0; ho.cnt[g] = 0; }
+      for (int g = 0; g < NA; ++g) { ho.val[g] = 0; ho.cnt[g] = 0; }
       ho.scanned = ho.matched = 0;
     }
 
@@ -107,7 +108,7 @@ __global__ __launch_bounds__(kThreads) void k_scan(
       if (threadIdx.x == 0) {
         uint64_t* hr = heads + (j / kThreads) * kHeadStride;
 #pragma unroll
-        for (int g = 0; g < YBG_MAX_AGGS; ++g) {
+        for (int g = 0; g < NA; ++g) {
           hr[2 * g] = ho.val[g];
           hr[2 * g + 1] = ho.cnt[g];
         }
@@ -116,7 +117,12 @@ __global__ __launch_bounds__(kThreads) void k_scan(
       } else if (!head_consumed) {
         scanned += ho.scanned;
         matched += ho.matched;
-        agg_combine(sp, agg_val, agg_cnt, ho.val, ho.cnt);
+#pragma unroll
+        for (int g = 0; g < NA; ++g) {
+          if (g < sp.num_aggs)
+            combine1(sp.aggs[g].op, &agg_val[g], &agg_cnt[g], ho.val[g],
+                     ho.cnt[g]);
+        }
       }
       if (walked_next && ((j + 1) % kThreads) == 0 && (j + 1) < n_ivs) {
         cont_flags[(j + 1) / kThreads] = 1;
@@ -308,89 +314,11 @@ int yb_gpu_scan_open(const ybg_scan_spec_t* spec, ybg_scan_t** out) {
                    "fallback (oracle/ is test infrastructure only)");
   auto* s = new ybg_scan();
   s->spec = *spec;
-  DevSpec& d = s->dspec;
-  memset(&d, 0, sizeof(d));
-  const ybg_schema_t& sc = spec->schema;
-  d.has_hash = sc.has_hash;
-  d.num_hash_cols = sc.num_hash_cols;
-  d.num_range_cols = sc.num_range_cols;
-  for (int i = 0; i < YBG_MAX_KEYCOLS; ++i) d.key_types[i] = sc.key_types[i];
-  d.num_value_cols = sc.num_value_cols;
-  // V1 packing layout (schema_packing.cc:499-535)
-  int nvar = 0;
-  int off_after = 0;
-  for (int i = 0; i < sc.num_value_cols; ++i) {
-    DevCol& c = d.cols[i];
-    c.id = sc.value_cols[i].column_id;
-    c.dtype = sc.value_cols[i].dtype;
-    bool varlen =
-        sc.value_cols[i].nullable || sc.value_cols[i].dtype == YBG_T_STRING;
-    c.v1_varlen = varlen;
-    c.v1_nvb = nvar;
-    c.v1_off = off_after;
-    int v1sz;
-    switch (sc.value_cols[i].dtype) {
-      case YBG_T_BOOL: v1sz = 1; break;
-      case YBG_T_INT8: case YBG_T_INT16: case YBG_T_INT32:
-      case YBG_T_UINT32: case YBG_T_FLOAT: v1sz = 5; break;
-      default: v1sz = 9; break;
-    }
-    if (varlen) { ++nvar; off_after = 0; }
-    else off_after += v1sz;
-    switch (sc.value_cols[i].dtype) {
-      case YBG_T_BOOL: case YBG_T_INT8: c.v2_fixed = 1; break;
-      case YBG_T_INT16: c.v2_fixed = 2; break;
-      case YBG_T_INT32: case YBG_T_UINT32: case YBG_T_FLOAT:
-        c.v2_fixed = 4; break;
-      case YBG_T_STRING: c.v2_fixed = 0; break;
-      default: c.v2_fixed = 8; break;
-    }
-  }
-  d.v1_varlen_count = nvar;
-  d.fmt = spec->kv_format;
-  htlim_from(spec->read_time.read, spec->read_time.read_len, &d.read);
-  htlim_from(spec->read_time.local_limit, spec->read_time.local_limit_len,
-             &d.local_lim);
-  htlim_from(spec->read_time.global_limit, spec->read_time.global_limit_len,
-             &d.global_lim);
-  {
-    // regular_limit = memcmp-min(enc(read), enc(local)) == enc(max(read,
-    // local)) — intent_aware_iterator.h:74-77
-    int rl = spec->read_time.read_len, ll = spec->read_time.local_limit_len;
-    int n = rl < ll ? rl : ll;
-    int cmp = memcmp(spec->read_time.local_limit, spec->read_time.read, n);
-    bool local_smaller = cmp < 0 || (cmp == 0 && ll < rl);
-    d.reg_lim = local_smaller ? d.local_lim : d.read;
-  }
-  d.num_preds = spec->num_preds;
-  for (int i = 0; i < spec->num_preds; ++i) {
-    const ybg_pred_t& p = spec->preds[i];
-    DevPred& dp = d.preds[i];
-    dp.is_key_col = p.is_key_col;
-    dp.col = p.col;
-    dp.op = p.op;
-    dp.datum = p.datum;
-    dp.str_len = (uint32_t)p.bytes_len;
-    dp.rhs_off = (uint32_t)s->aux_host.size();
-    if (p.bytes && p.bytes_len)
-      s->aux_host.insert(s->aux_host.end(), p.bytes, p.bytes + p.bytes_len);
-  }
-  d.num_aggs = spec->num_aggs;
-  for (int i = 0; i < spec->num_aggs; ++i) {
-    d.aggs[i].op = spec->aggs[i].op;
-    d.aggs[i].col = spec->aggs[i].col;
-  }
-  d.lower_off = (uint32_t)s->aux_host.size();
-  d.lower_len = (uint32_t)spec->lower_bound_len;
-  if (spec->lower_bound && spec->lower_bound_len)
-    s->aux_host.insert(s->aux_host.end(), spec->lower_bound,
-                       spec->lower_bound + spec->lower_bound_len);
-  d.upper_off = (uint32_t)s->aux_host.size();
-  d.upper_len = (uint32_t)spec->upper_bound_len;
-  if (spec->upper_bound && spec->upper_bound_len)
-    s->aux_host.insert(s->aux_host.end(), spec->upper_bound,
-                       spec->upper_bound + spec->upper_bound_len);
-  if (s->aux_host.empty()) s->aux_host.push_back(0);
+  s->aux_host.resize(1 << 20);
+  uint32_t aux_len = 0;
+  build_dev_spec(spec, &s->dspec, s->aux_host.data(), &aux_len,
+                 (uint32_t)s->aux_host.size());
+  s->aux_host.resize(aux_len);
 
   hipError_t e;
   if ((e = hipStreamCreate(&s->stream)) != hipSuccess ||
@@ -490,10 +418,23 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
                          s->n_heads * kHeadStride * sizeof(uint64_t),
                          s->stream));
   HIP_TRY(hipEventRecord(s->ev_start, s->stream));
-  hipLaunchKernelGGL(k_scan, dim3(s->grid), dim3(kThreads), 0, s->stream,
-                     s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
-                     s->d_aux, s->d_rk_save, s->d_partials, s->d_heads,
-                     s->d_cont);
+  // dispatch on aggregate-slot capacity: smaller NA => fewer live VGPRs
+  if (s->dspec.num_aggs <= 2) {
+    hipLaunchKernelGGL(k_scan<2>, dim3(s->grid), dim3(kThreads), 0, s->stream,
+                       s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
+                       s->d_aux, s->d_rk_save, s->d_partials, s->d_heads,
+                       s->d_cont);
+  } else if (s->dspec.num_aggs <= 4) {
+    hipLaunchKernelGGL(k_scan<4>, dim3(s->grid), dim3(kThreads), 0, s->stream,
+                       s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
+                       s->d_aux, s->d_rk_save, s->d_partials, s->d_heads,
+                       s->d_cont);
+  } else {
+    hipLaunchKernelGGL(k_scan<8>, dim3(s->grid), dim3(kThreads), 0, s->stream,
+                       s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
+                       s->d_aux, s->d_rk_save, s->d_partials, s->d_heads,
+                       s->d_cont);
+  }
   HIP_TRY(hipEventRecord(s->ev_mid, s->stream));
   hipLaunchKernelGGL(k_reduce, dim3(1), dim3(256), 0, s->stream, s->dspec,
                      s->d_partials, s->n_partials, s->d_heads, s->d_cont,
