@@ -77,6 +77,8 @@ struct DevSpec {
   uint32_t lower_len, lower_off, upper_len, upper_off;
   int32_t v1_varlen_count;
   uint32_t value_pred_mask;  // bit i: predicate i is on a value column
+  uint32_t fixed_rk_len;     // encoded DocKey length when schema has no
+                             // string key columns (0 = variable)
 };
 
 struct Interval {
@@ -104,30 +106,61 @@ constexpr uint8_t kGroupEnd = 0x21, kHybridTimeByte = 0x23, kNullLow = 0x24,
 // Byte-level device helpers
 // ---------------------------------------------------------------------------
 
+// Unaligned little-endian u64 via three aligned dword loads + funnel shift.
+// The compiler emits per-byte loads for unaligned memcpy on amdgcn; this is
+// 3 loads instead of 8. May read up to 11 bytes past p — every data buffer
+// (generator output, device copy) carries >= 16 bytes of tail slack.
+DEV uint64_t load_u64_una(const uint8_t* p) {
+  uintptr_t a = (uintptr_t)p;
+  const uint32_t* q = (const uint32_t*)(a & ~(uintptr_t)3);
+  uint32_t sh = (uint32_t)(a & 3) * 8;
+  uint64_t lo = ((uint64_t)q[1] << 32) | q[0];
+  if (sh == 0) return lo;
+  return (lo >> sh) | ((uint64_t)q[2] << (64 - sh));
+}
 DEV uint64_t load_be64(const uint8_t* p) {
-  uint64_t v;
-  __builtin_memcpy(&v, p, 8);
-  return __builtin_bswap64(v);
+  return __builtin_bswap64(load_u64_una(p));
 }
 DEV uint32_t load_be32(const uint8_t* p) {
-  uint32_t v;
-  __builtin_memcpy(&v, p, 4);
-  return __builtin_bswap32(v);
+  return __builtin_bswap32((uint32_t)load_u64_una(p));
 }
-DEV uint64_t load_le64_u(const uint8_t* p) {
-  uint64_t v;
-  __builtin_memcpy(&v, p, 8);
-  return v;
-}
-DEV uint32_t load_le32_u(const uint8_t* p) {
-  uint32_t v;
-  __builtin_memcpy(&v, p, 4);
-  return v;
-}
+DEV uint64_t load_le64_u(const uint8_t* p) { return load_u64_una(p); }
+DEV uint32_t load_le32_u(const uint8_t* p) { return (uint32_t)load_u64_una(p); }
 
-// LEB128 varint (rocksdb util/coding.h)
+// LEB128 varint (rocksdb util/coding.h) — one windowed load covers the
+// 1-4 byte encodings that dominate (block headers); longer forms fall back.
 DEV const uint8_t* leb128(const uint8_t* p, const uint8_t* limit,
                           uint64_t* v) {
+  if (p >= limit) return nullptr;
+  uint64_t w = load_u64_una(p);
+  uint64_t b0 = w & 0xff;
+  if (!(b0 & 0x80)) {
+    *v = b0;
+    return p + 1;
+  }
+  if (p + 2 <= limit) {
+    uint64_t b1 = (w >> 8) & 0xff;
+    if (!(b1 & 0x80)) {
+      *v = (b0 & 0x7f) | (b1 << 7);
+      return p + 2;
+    }
+    if (p + 3 <= limit) {
+      uint64_t b2 = (w >> 16) & 0xff;
+      if (!(b2 & 0x80)) {
+        *v = (b0 & 0x7f) | ((b1 & 0x7f) << 7) | (b2 << 14);
+        return p + 3;
+      }
+      if (p + 4 <= limit) {
+        uint64_t b3 = (w >> 24) & 0xff;
+        if (!(b3 & 0x80)) {
+          *v = (b0 & 0x7f) | ((b1 & 0x7f) << 7) | ((b2 & 0x7f) << 14) |
+               (b3 << 21);
+          return p + 4;
+        }
+      }
+    }
+  }
+  // slow path (>= 5 bytes or close to limit)
   uint64_t result = 0;
   int shift = 0;
   while (p < limit && shift <= 63) {
@@ -148,24 +181,30 @@ DEV const uint8_t* leb128(const uint8_t* p, const uint8_t* limit,
 DEV const uint8_t* svarint(const uint8_t* p, const uint8_t* limit,
                            int64_t* v) {
   if (p >= limit) return nullptr;
-  uint32_t header = ((uint32_t)p[0] << 8) | (p + 1 < limit ? p[1] : 0);
+  uint64_t wbe = __builtin_bswap64(load_u64_una(p));  // bytes big-endian
+  uint32_t header = (uint32_t)(wbe >> 48);
   uint64_t neg = -(uint64_t)((header & 0x8000u) == 0);
   header ^= (uint32_t)neg;
   int n = __builtin_clz((~header & 0x7fffu) | 0x20u) - 16;
   if (p + n > limit) return nullptr;
+  if (n <= 8) {
+    uint64_t temp = wbe >> (8 * (8 - n));
+    uint64_t mask = (1ull << (7 * n - 1)) - 1;
+    *v = (int64_t)(((temp & mask) | (~mask & neg)) - neg);
+    return p + n;
+  }
   uint64_t temp = 0;
   for (int i = 0; i < n; ++i) temp = (temp << 8) | p[i];
-  uint64_t mask;
-  if (n >= 10) mask = ~0ull;
-  else if (n == 9) mask = 0x3fffffffffffffffull;
-  else mask = (1ull << (7 * n - 1)) - 1;
+  uint64_t mask = (n >= 10) ? ~0ull : 0x3fffffffffffffffull;
   *v = (int64_t)(((temp & mask) | (~mask & neg)) - neg);
   return p + n;
 }
 
 DEV int desc_svarint_size(const uint8_t* p, const uint8_t* limit) {
   if (p >= limit) return 0;
-  uint32_t header = ((uint32_t)p[0] << 8) | (p + 1 < limit ? p[1] : 0);
+  uint32_t b0 = p[0];
+  uint32_t b1 = (p + 1 < limit) ? p[1] : 0;
+  uint32_t header = (b0 << 8) | b1;
   uint64_t neg = -(uint64_t)((header & 0x8000u) == 0);
   header ^= (uint32_t)neg;
   return __builtin_clz((~header & 0x7fffu) | 0x20u) - 16;
@@ -183,6 +222,18 @@ DEV const uint8_t* uvarint(const uint8_t* p, const uint8_t* limit,
   if (n == 1) {
     *v = first & 0x7f;
     return p + 1;
+  }
+  if (n <= 8) {
+    uint64_t wbe = __builtin_bswap64(load_u64_una(p));
+    uint64_t masked =
+        (wbe >> (8 * (8 - n))) & ((1ull << (8 * n - n)) - 1);
+    // top n-1 bits of the first byte are the size prefix; keep 8-n bits
+    uint64_t first_bits = (uint64_t)(first & ((1u << (8 - n)) - 1))
+                          << (8 * (n - 1));
+    uint64_t tail = (wbe >> (8 * (8 - n))) & ((n > 1) ? ((1ull << (8 * (n - 1))) - 1) : 0);
+    *v = first_bits | tail;
+    (void)masked;
+    return p + n;
   }
   uint64_t result = 0;
   int i = 0;
@@ -292,25 +343,47 @@ __global__ void k_emit_intervals(const uint8_t* __restrict__ data,
 struct EntryRef {
   const uint8_t* value;
   uint32_t value_len;
+  uint32_t shared;  // shared-prefix bytes reused from the previous key
 };
 
 // Mirrors DecodeEntryThreeSharedParts (block_internal.h:54-160) +
 // IterKey::Update (db/dbformat.h:405-476) and the shared_prefix DecodeEntry
 // (block.cc:411-436). Returns pointer past the entry or nullptr.
+//
+// Key representation: bytes [0, key_len-8) live in the LDS array `key`; the
+// trailing 8-byte internal component (seqno<<8|type, dbformat.h:84-110) is
+// cached in *last8 (little-endian). Nothing on the read path parses the
+// internal component, and the dominant "last8 reused (+0x100)" case then
+// performs no LDS traffic for it. For the rare paths that materialize tail
+// bytes into LDS (restarts, no-reuse deltas), the register is reloaded from
+// LDS afterwards; stale tail bytes in LDS are never read (the middle-copy
+// consults *last8 for source positions inside the previous key's tail).
 DEV const uint8_t* decode_entry(int fmt, const uint8_t* p,
                                 const uint8_t* limit, uint8_t* key,
-                                uint32_t* key_len, EntryRef* out) {
+                                uint32_t* key_len, uint64_t* last8,
+                                EntryRef* out) {
   if (fmt == YBG_ENC_SHARED_PREFIX) {
     uint64_t shared, non_shared, value_len;
     if (!(p = leb128(p, limit, &shared))) return nullptr;
     if (!(p = leb128(p, limit, &non_shared))) return nullptr;
     if (!(p = leb128(p, limit, &value_len))) return nullptr;
     if ((uint64_t)(limit - p) < non_shared + value_len) return nullptr;
-    if (shared > *key_len || shared + non_shared > kKeyCap) return nullptr;
+    uint64_t new_len = shared + non_shared;
+    if (shared > *key_len || new_len > kKeyCap || new_len < 9)
+      return nullptr;
+    uint64_t prev_len = *key_len;
+    // ensure LDS holds prev tail bytes the shared prefix reaches into
+    if (prev_len >= 8 && shared > prev_len - 8) {
+      uint64_t old8 = *last8;
+      for (int i = 0; i < 8; ++i)
+        key[prev_len - 8 + i] = (uint8_t)(old8 >> (8 * i));
+    }
     for (uint32_t i = 0; i < (uint32_t)non_shared; ++i) key[shared + i] = p[i];
-    *key_len = (uint32_t)(shared + non_shared);
+    *key_len = (uint32_t)new_len;
+    *last8 = load_u64_una(key + new_len - 8);
     out->value = p + non_shared;
     out->value_len = (uint32_t)value_len;
+    out->shared = (uint32_t)shared;
     return out->value + value_len;
   }
 
@@ -321,7 +394,7 @@ DEV const uint8_t* decode_entry(int fmt, const uint8_t* p,
   uint64_t last8_inc = (encoded_1 & 2) << 7;
   bool frequent = encoded_1 & 1;
 
-  uint32_t shared_prefix = 0, ns1 = 0, ns2 = 0, last8 = 0;
+  uint32_t shared_prefix = 0, ns1 = 0, ns2 = 0, reuse8 = 0;
   int64_t ns1_delta = 0, ns2_delta = 0;
   bool shared_something;
   uint64_t tmp;
@@ -329,7 +402,7 @@ DEV const uint8_t* decode_entry(int fmt, const uint8_t* p,
   if (frequent) {
     if (!(p = leb128(p, limit, &tmp))) return nullptr;
     shared_prefix = (uint32_t)tmp;
-    last8 = 8;
+    reuse8 = 8;
     shared_something = true;
     ns1 = 1;
     ns2 = 1;
@@ -346,12 +419,12 @@ DEV const uint8_t* decode_entry(int fmt, const uint8_t* p,
     } else {
       shared_something = true;
       if ((e2 & 2) == 0) {
-        last8 = 8;
+        reuse8 = 8;
         ns2_delta = (e2 >> 2) & 1;
         ns1 = (e2 >> 3) & 7;
         ns2 = (e2 >> 6) & 3;
       } else {
-        last8 = (e2 & 4) ? 8 : 0;
+        reuse8 = (e2 & 4) ? 8 : 0;
         if (!(p = leb128(p, limit, &tmp))) return nullptr;
         ns1 = (uint32_t)tmp;
         if (e2 & 8) {
@@ -372,51 +445,74 @@ DEV const uint8_t* decode_entry(int fmt, const uint8_t* p,
   if ((uint64_t)(limit - p) < (uint64_t)ns1 + ns2 + value_size) return nullptr;
 
   if (!shared_something) {
-    if (ns1 > kKeyCap) return nullptr;
+    // restart / full key inline (block.cc:311-317)
+    if (ns1 > kKeyCap || ns1 < 9) return nullptr;
     for (uint32_t i = 0; i < ns1; ++i) key[i] = p[i];
     *key_len = ns1;
+    *last8 = load_u64_una(key + ns1 - 8);
     out->value = p + ns1;
     out->value_len = value_size;
+    out->shared = 0;
     return out->value + value_size;
   }
 
   uint64_t prev_mid_start = (uint64_t)shared_prefix + ns1 - (uint64_t)ns1_delta;
   uint64_t prev_ns2 = (uint64_t)ns2 - (uint64_t)ns2_delta;
-  uint64_t prev_except_mid = prev_mid_start + prev_ns2 + last8;
+  uint64_t prev_except_mid = prev_mid_start + prev_ns2 + reuse8;
   if (*key_len < prev_except_mid) return nullptr;
   uint64_t mid = *key_len - prev_except_mid;
-  if (shared_prefix + mid + last8 == 0) return nullptr;
+  if (shared_prefix + mid + reuse8 == 0) return nullptr;
 
   uint64_t new_mid_start = shared_prefix + ns1;
   uint64_t new_ns2_start = new_mid_start + mid;
   uint64_t new_last8_start = new_ns2_start + ns2;
-  uint64_t new_key_size = new_last8_start + last8;
-  if (new_key_size > kKeyCap) return nullptr;
+  uint64_t new_key_size = new_last8_start + reuse8;
+  if (new_key_size > kKeyCap || new_key_size < 9) return nullptr;
+  uint64_t prev_len = *key_len;
 
-  uint64_t last_comp = 0;
-  if (last8) {
-    const uint8_t* q = key + *key_len - 8;
-    for (int i = 7; i >= 0; --i) last_comp = (last_comp << 8) | q[i];
-    last_comp += last8_inc;
-  }
-  if (new_mid_start != prev_mid_start && mid > 0) {
-    if (new_mid_start < prev_mid_start) {
-      for (uint64_t i = 0; i < mid; ++i)
-        key[new_mid_start + i] = key[prev_mid_start + i];
-    } else {
-      for (uint64_t i = mid; i-- > 0;)
-        key[new_mid_start + i] = key[prev_mid_start + i];
+  if (reuse8) {
+    // FAST PATH: the internal component is carried in the register; only
+    // body bytes move. The middle source region always lies before the
+    // previous tail (prev_except_mid accounting), so plain LDS moves apply.
+    if (new_mid_start != prev_mid_start && mid > 0) {
+      if (new_mid_start < prev_mid_start) {
+        for (uint64_t i = 0; i < mid; ++i)
+          key[new_mid_start + i] = key[prev_mid_start + i];
+      } else {
+        for (uint64_t i = mid; i-- > 0;)
+          key[new_mid_start + i] = key[prev_mid_start + i];
+      }
     }
-  }
-  for (uint32_t i = 0; i < ns1; ++i) key[shared_prefix + i] = p[i];
-  for (uint32_t i = 0; i < ns2; ++i) key[new_ns2_start + i] = p[ns1 + i];
-  if (last8) {
-    uint8_t* q = key + new_last8_start;
-    for (int i = 0; i < 8; ++i) q[i] = (uint8_t)(last_comp >> (8 * i));
+    for (uint32_t i = 0; i < ns1; ++i) key[shared_prefix + i] = p[i];
+    for (uint32_t i = 0; i < ns2; ++i) key[new_ns2_start + i] = p[ns1 + i];
+    *last8 += last8_inc;
+  } else {
+    // No tail reuse (rare): the retained prefix or the shared middle may
+    // reach into the previous key's register-resident tail. Materialize the
+    // old tail into LDS first, then proceed with plain LDS moves and reload
+    // the register from the new tail bytes.
+    if (prev_len >= 8) {
+      uint64_t old8 = *last8;
+      for (int i = 0; i < 8; ++i)
+        key[prev_len - 8 + i] = (uint8_t)(old8 >> (8 * i));
+    }
+    if (new_mid_start != prev_mid_start && mid > 0) {
+      if (new_mid_start < prev_mid_start) {
+        for (uint64_t i = 0; i < mid; ++i)
+          key[new_mid_start + i] = key[prev_mid_start + i];
+      } else {
+        for (uint64_t i = mid; i-- > 0;)
+          key[new_mid_start + i] = key[prev_mid_start + i];
+      }
+    }
+    for (uint32_t i = 0; i < ns1; ++i) key[shared_prefix + i] = p[i];
+    for (uint32_t i = 0; i < ns2; ++i) key[new_ns2_start + i] = p[ns1 + i];
+    *last8 = load_u64_una(key + new_key_size - 8);
   }
   *key_len = (uint32_t)new_key_size;
   out->value = p + ns1 + ns2;
   out->value_len = value_size;
+  out->shared = shared_prefix;
   return out->value + value_size;
 }
 
@@ -1109,6 +1205,7 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
   bool fail = false;
 
   uint32_t key_len = 0;
+  uint64_t reg_last8 = 0;
   uint32_t rk_len = 0;
   bool row_open = false;
   bool in_head = true;
@@ -1128,19 +1225,30 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
       // key state is overwritten by its full-key decode.
       continue;
     }
-    const uint8_t* q = decode_entry(sp.fmt, p, limit, key, &key_len, &er);
+    const uint8_t* q =
+        decode_entry(sp.fmt, p, limit, key, &key_len, &reg_last8, &er);
     if (!q || key_len < 10) { fail = true; break; }
     uint32_t ukey_len = key_len - 8;
     uint32_t ht_sz = key[ukey_len - 1] & 0x1f;
     if (ht_sz == 0 || ukey_len < ht_sz + 2 ||
         key[ukey_len - ht_sz - 1] != kHybridTimeByte) { fail = true; break; }
     uint32_t prefix_len = ukey_len - ht_sz - 1;
-    uint32_t rk = dockey_len(sp, key, prefix_len);
-    if (!rk || rk > prefix_len) { fail = true; break; }
+    uint32_t rk;
+    if (sp.fixed_rk_len) {
+      rk = sp.fixed_rk_len;
+      if (rk > prefix_len) { fail = true; break; }
+    } else {
+      rk = dockey_len(sp, key, prefix_len);
+      if (!rk || rk > prefix_len) { fail = true; break; }
+    }
 
+    // Row-change detection: bytes [0, er.shared) are identical to the
+    // previous entry's key by construction (delta encoding), and while the
+    // row is open the previous entry's rowkey == rk_save, so only the tail
+    // [er.shared, rk) needs comparing.
     bool row_change = !row_open || (rk != rk_len);
-    if (!row_change) {
-      for (uint32_t i = 0; i < rk; ++i) {
+    if (!row_change && er.shared < rk) {
+      for (uint32_t i = er.shared; i < rk; ++i) {
         if (key[i] != rk_save[i]) { row_change = true; break; }
       }
     }
@@ -1287,6 +1395,21 @@ inline void build_dev_spec(const ybg_scan_spec_t* spec, DevSpec* dp,
   d.lower_off = put(spec->lower_bound, spec->lower_bound_len);
   d.upper_len = (uint32_t)spec->upper_bound_len;
   d.upper_off = put(spec->upper_bound, spec->upper_bound_len);
+  // fixed rowkey length fast path (doc_key.h:40-63 layout) when no string
+  // key columns
+  {
+    bool fixed = true;
+    // 'G' + hash16 + hashed cols + '!' + range cols + '!'  (with hash)
+    // range cols + '!'                                     (without)
+    uint32_t L = sc.has_hash ? 5u : 1u;
+    int nk = (sc.has_hash ? sc.num_hash_cols : 0) + sc.num_range_cols;
+    for (int i = 0; i < nk; ++i) {
+      if (sc.key_types[i] == YBG_KT_INT64) L += 9;
+      else if (sc.key_types[i] == YBG_KT_INT32) L += 5;
+      else fixed = false;
+    }
+    d.fixed_rk_len = fixed ? L : 0;
+  }
   *aux_len = pos ? pos : 1;
 }
 

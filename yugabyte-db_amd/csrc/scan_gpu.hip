@@ -344,7 +344,9 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
     s->d_data = const_cast<uint8_t*>(blocks);
     s->d_data_owned = false;
   } else {
-    HIP_TRY(hipMalloc(&s->d_data, s->total_bytes));
+    // +16 bytes tail slack for windowed loads (scan_device.h load_u64_una)
+    HIP_TRY(hipMalloc(&s->d_data, s->total_bytes + 16));
+    HIP_TRY(hipMemset(s->d_data + s->total_bytes, 0, 16));
     HIP_TRY(hipMemcpy(s->d_data, blocks, s->total_bytes,
                       hipMemcpyHostToDevice));
     s->d_data_owned = true;
