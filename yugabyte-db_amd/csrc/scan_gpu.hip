@@ -36,6 +36,7 @@
 
 #include <algorithm>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <string>
 #include <vector>
@@ -398,7 +399,12 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
   HIP_WARN(hipFree(d_err));
 
   uint64_t want = (s->n_ivs + kThreads - 1) / kThreads;
-  s->grid = (int)std::min<uint64_t>(want, 4096);
+  uint64_t cap = 4096;
+  if (const char* g = getenv("YBG_GRID")) {
+    long v = atol(g);
+    if (v > 0) cap = (uint64_t)v;
+  }
+  s->grid = (int)std::min<uint64_t>(want, cap);
   if (s->grid < 1) s->grid = 1;
   uint64_t span_threads = (uint64_t)s->grid * kThreads;
   s->n_partials = span_threads / 64;
