@@ -289,6 +289,71 @@ DEV int u128_slice_cmp(uint64_t ahi, uint64_t alo, uint32_t alen,
 }
 
 // ---------------------------------------------------------------------------
+// Register-window stream reader. The scan's decode is a strictly forward
+// byte stream; parsing through per-byte global loads makes every entry a
+// chain of dependent ~200-900-cycle memory ops. Rdr keeps 32 bytes in four
+// registers; peeks are funnel shifts (VALU), consumes shift the window and
+// refill with ALIGNED 8-byte loads whose results are not needed until >=16
+// bytes later — the latency hides behind parsing. Buffers must carry >= 48
+// bytes of tail slack (generator/device allocations do).
+// ---------------------------------------------------------------------------
+struct Rdr {
+  uint64_t q0, q1, q2, q3;  // aligned u64s: [base, base+32)
+  const uint8_t* base;      // 8-byte aligned
+  uint32_t k;               // logical position = base + k, k in [0,8)
+
+  DEV void init(const uint8_t* p) {
+    base = (const uint8_t*)((uintptr_t)p & ~(uintptr_t)7);
+    k = (uint32_t)((uintptr_t)p & 7);
+    const uint64_t* q = (const uint64_t*)base;
+    q0 = q[0];
+    q1 = q[1];
+    q2 = q[2];
+    q3 = q[3];
+  }
+  DEV const uint8_t* pos() const { return base + k; }
+  // 8 bytes at the current position, little-endian
+  DEV uint64_t peek8() const {
+    if (k == 0) return q0;
+    return (q0 >> (8 * k)) | (q1 << (64 - 8 * k));
+  }
+  // 8 bytes at position + off (off + k must be < 25)
+  DEV uint64_t peek8_at(uint32_t off) const {
+    uint32_t kk = k + off;
+    uint64_t lo, hi;
+    if (kk >= 16) { lo = q2; hi = q3; kk -= 16; }
+    else if (kk >= 8) { lo = q1; hi = q2; kk -= 8; }
+    else { lo = q0; hi = q1; }
+    if (kk == 0) return lo;
+    return (lo >> (8 * kk)) | (hi << (64 - 8 * kk));
+  }
+  DEV void consume(uint32_t n) {  // n <= 16
+    k += n;
+    while (k >= 8) {
+      q0 = q1;
+      q1 = q2;
+      q2 = q3;
+      base += 8;
+      q3 = ((const uint64_t*)base)[3];
+      k -= 8;
+    }
+  }
+  DEV void skip(uint32_t n) {
+    if (n <= 16) consume(n);
+    else init(pos() + n);
+  }
+  DEV void seek(const uint8_t* p) {
+    if (p == pos()) return;
+    uintptr_t d = (uintptr_t)p - (uintptr_t)base;
+    if (d < 8) {  // within current low word
+      k = (uint32_t)d;
+      return;
+    }
+    init(p);
+  }
+};
+
+// ---------------------------------------------------------------------------
 // Interval table construction
 // ---------------------------------------------------------------------------
 
@@ -358,10 +423,10 @@ struct EntryRef {
 // bytes into LDS (restarts, no-reuse deltas), the register is reloaded from
 // LDS afterwards; stale tail bytes in LDS are never read (the middle-copy
 // consults *last8 for source positions inside the previous key's tail).
-DEV const uint8_t* decode_entry(int fmt, const uint8_t* p,
-                                const uint8_t* limit, uint8_t* key,
-                                uint32_t* key_len, uint64_t* last8,
-                                EntryRef* out) {
+DEV const uint8_t* decode_entry_ptr(int fmt, const uint8_t* p,
+                                    const uint8_t* limit, uint8_t* key,
+                                    uint32_t* key_len, uint64_t* last8,
+                                    EntryRef* out) {
   if (fmt == YBG_ENC_SHARED_PREFIX) {
     uint64_t shared, non_shared, value_len;
     if (!(p = leb128(p, limit, &shared))) return nullptr;
@@ -514,6 +579,107 @@ DEV const uint8_t* decode_entry(int fmt, const uint8_t* p,
   out->value_len = value_size;
   out->shared = shared_prefix;
   return out->value + value_size;
+}
+
+// Rdr-based entry decode: fast path for the dominant three_shared_parts
+// encodings (frequent case and case 2.1.1, block_builder_internal.h:139-183)
+// — the whole header plus the <=10 non-shared key bytes are parsed from the
+// register window with no per-byte memory chain, and in these cases the
+// shared middle never moves (ns1_delta == 0), so only ns1+ns2 LDS bytes are
+// written. Everything else delegates to decode_entry_ptr. On success the
+// reader is positioned AT THE VALUE START (value not consumed).
+DEV const uint8_t* decode_entry(int fmt, Rdr* rdr, const uint8_t* limit,
+                                uint8_t* key, uint32_t* key_len,
+                                uint64_t* last8, EntryRef* out) {
+  const uint8_t* p = rdr->pos();
+  if (fmt == YBG_ENC_THREE_SHARED_PARTS && limit - p >= 8) {
+    uint64_t w = rdr->peek8();
+    uint32_t b0 = (uint32_t)(w & 0xff);
+    uint64_t e1;
+    uint32_t e1len;
+    bool ok = true;
+    if (!(b0 & 0x80)) {
+      e1 = b0;
+      e1len = 1;
+    } else {
+      uint32_t b1 = (uint32_t)((w >> 8) & 0xff);
+      if (!(b1 & 0x80)) {
+        e1 = (b0 & 0x7f) | ((uint64_t)b1 << 7);
+        e1len = 2;
+      } else {
+        e1 = 0;
+        e1len = 0;
+        ok = false;
+      }
+    }
+    if (ok) {
+      uint32_t value_size = (uint32_t)(e1 >> 2);
+      uint64_t inc = (e1 & 2) << 7;
+      uint32_t ns1, ns2, d2 = 0, hl;
+      bool fast = false;
+      if (e1 & 1) {  // frequent: <sp>
+        uint32_t spb = (uint32_t)((w >> (8 * e1len)) & 0xff);
+        if (!(spb & 0x80)) {
+          ns1 = 1;
+          ns2 = 1;
+          hl = e1len + 1;
+          fast = true;
+          out->shared = spb;
+        }
+      } else {
+        uint32_t e2 = (uint32_t)((w >> (8 * e1len)) & 0xff);
+        if ((e2 & 3) == 1) {  // case 2.1.1: reuse, ns1<8, ns2<4, d1=0
+          uint32_t spb = (uint32_t)((w >> (8 * (e1len + 1))) & 0xff);
+          if (!(spb & 0x80)) {
+            d2 = (e2 >> 2) & 1;
+            ns1 = (e2 >> 3) & 7;
+            ns2 = (e2 >> 6) & 3;
+            hl = e1len + 2;
+            fast = true;
+            out->shared = spb;
+          }
+        }
+      }
+      if (fast) {
+        uint32_t sp = out->shared;
+        uint64_t prev_len = *key_len;
+        uint64_t prev_ns2 = (uint64_t)ns2 - d2;
+        uint64_t prev_except = (uint64_t)sp + ns1 + prev_ns2 + 8;
+        uint64_t total = (uint64_t)hl + ns1 + ns2 + value_size;
+        if (prev_len >= prev_except && (uint64_t)(limit - p) >= total) {
+          uint64_t mid = prev_len - prev_except;
+          uint64_t new_ns2_start = sp + ns1 + mid;
+          uint64_t new_len = new_ns2_start + ns2 + 8;
+          if (new_len <= kKeyCap) {
+            // key bytes [hl, hl+ns1+ns2) — within the first 16 window bytes
+            uint64_t w2 = rdr->peek8_at(8);
+            for (uint32_t i = 0; i < ns1; ++i) {
+              uint32_t j = hl + i;
+              uint64_t src = j < 8 ? w : w2;
+              key[sp + i] = (uint8_t)(src >> (8 * (j & 7)));
+            }
+            for (uint32_t i = 0; i < ns2; ++i) {
+              uint32_t j = hl + ns1 + i;
+              uint64_t src = j < 8 ? w : w2;
+              key[new_ns2_start + i] = (uint8_t)(src >> (8 * (j & 7)));
+            }
+            *last8 += inc;
+            *key_len = (uint32_t)new_len;
+            rdr->consume(hl + ns1 + ns2);
+            out->value = rdr->pos();
+            out->value_len = value_size;
+            return out->value + value_size;
+          }
+        }
+      }
+    }
+  }
+  // general path
+  const uint8_t* q =
+      decode_entry_ptr(fmt, p, limit, key, key_len, last8, out);
+  if (!q) return nullptr;
+  rdr->seek(out->value);
+  return q;
 }
 
 // ---------------------------------------------------------------------------
@@ -1064,14 +1230,68 @@ DEV void agg_combine(const DevSpec& sp, uint64_t* a_val, uint64_t* a_cnt,
   }
 }
 
+// Rdr-based packed-row V2 decode fast path: schema version < 128, no null
+// mask (the dominant shape); anything else falls back to the pointer
+// decoder. Assumes rdr is positioned at the 'kPackedRowV2' byte; consumes
+// the whole body on success.
+template <int NA>
+DEV bool decode_packed_v2_rdr(const DevSpec& sp, const uint8_t* base,
+                              const uint8_t* aux, Rdr* rdr, uint32_t len,
+                              RowCtxT<NA>* rc, bool* done) {
+  uint64_t w = rdr->peek8();
+  uint32_t ver = (uint32_t)((w >> 8) & 0xff);
+  uint32_t flags = (uint32_t)((w >> 16) & 0xff);
+  if ((ver & 0x80) || flags != 0) {
+    *done = false;  // caller falls back
+    return true;
+  }
+  *done = true;
+  const uint8_t* end = rdr->pos() + len;
+  rdr->consume(3);
+  for (int i = 0; i < sp.num_value_cols; ++i) {
+    const DevCol& c = sp.cols[i];
+    if (c.v2_fixed) {
+      if (rdr->pos() + c.v2_fixed > end) return false;
+      uint64_t u = rdr->peek8();
+      switch (c.v2_fixed) {
+        case 1: u &= 0xff; break;
+        case 2: u &= 0xffff; break;
+        case 4: u &= 0xffffffffull; break;
+        default: break;
+      }
+      if (c.dtype == YBG_T_INT8) u = (uint64_t)(int64_t)(int8_t)u;
+      else if (c.dtype == YBG_T_INT16) u = (uint64_t)(int64_t)(int16_t)u;
+      else if (c.dtype == YBG_T_INT32) u = (uint64_t)(int64_t)(int32_t)u;
+      eval_col(sp, aux, rc, i, false, u, nullptr, 0);
+      rdr->consume((uint32_t)c.v2_fixed);
+    } else {
+      if (rdr->pos() >= end) return false;
+      uint64_t w0 = rdr->peek8();
+      uint32_t flen;
+      if ((w0 & 1) == 0) {
+        flen = (uint32_t)((w0 & 0xff) >> 1);
+        rdr->consume(1);
+      } else {
+        flen = (uint32_t)((w0 & 0xffffffffull) >> 1);
+        rdr->consume(4);
+      }
+      if (rdr->pos() + flen > end) return false;
+      eval_col(sp, aux, rc, i, false, 0, rdr->pos(), flen);
+      rdr->skip(flen);
+    }
+  }
+  (void)base;
+  return true;
+}
+
 // Visibility + row-state update for one entry. key/rowkey live in LDS.
-// Returns false on corruption.
+// rdr is positioned at the value start. Returns false on corruption.
 template <int NA>
 DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
                        const uint8_t* aux, const uint8_t* key,
                        uint32_t key_len, const uint8_t* value,
                        uint32_t value_len, uint32_t rowkey_len,
-                       RowCtxT<NA>* rc) {
+                       RowCtxT<NA>* rc, Rdr* rdr) {
   uint32_t ukey_len = key_len - 8;
   uint32_t ht_size = key[ukey_len - 1] & 0x1f;
   const uint8_t* ht_enc = key + ukey_len - ht_size;
@@ -1079,8 +1299,9 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
 
   uint64_t ht_hi, ht_lo;
   slice_u128(ht_enc, ht_size, &ht_hi, &ht_lo);
+  uint32_t vb0 = value_len > 0 ? (uint32_t)(rdr->peek8() & 0xff) : 0u;
   bool visible;
-  if (value_len > 0 && value[0] == kHybridTimeByte) {
+  if (value_len > 0 && vb0 == kHybridTimeByte) {
     // committed-txn record with intent time
     // (intent_aware_iterator.cc:1249-1267)
     uint64_t v_hi, v_lo;
@@ -1109,6 +1330,17 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
       rc->base_ht_hi = ht_hi;
       rc->base_ht_lo = ht_lo;
       rc->base_ht_len = ht_size;
+      if (value_len > 0 && vb0 == kPackedV2B && value == rdr->pos()) {
+        // dominant shape: packed V2, no control fields — register-window
+        // decode; falls back when a null mask / big schema version appears
+        bool done;
+        if (!decode_packed_v2_rdr(sp, base, aux, rdr, value_len, rc, &done))
+          return false;
+        if (done) {
+          rc->found = true;
+          return true;
+        }
+      }
       int cf = skip_control(value, value_len);
       if (cf < 0) return false;
       const uint8_t* body = value + cf;
@@ -1196,6 +1428,8 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
   const uint8_t* p = blk + iv.start;
   const uint8_t* limit = blk + iv.end;
   uint64_t cur_iv = j;
+  Rdr rdr;
+  rdr.init(p);
 
 #pragma unroll
   for (int g = 0; g < NA; ++g) { ho->val[g] = 0; ho->cnt[g] = 0; }
@@ -1221,12 +1455,13 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
       blk = data + block_offsets[nx.block];
       p = blk + nx.start;
       limit = blk + nx.end;
+      rdr.init(p);
       // first entry of a restart interval is self-contained; the carried
       // key state is overwritten by its full-key decode.
       continue;
     }
     const uint8_t* q =
-        decode_entry(sp.fmt, p, limit, key, &key_len, &reg_last8, &er);
+        decode_entry(sp.fmt, &rdr, limit, key, &key_len, &reg_last8, &er);
     if (!q || key_len < 10) { fail = true; break; }
     uint32_t ukey_len = key_len - 8;
     uint32_t ht_sz = key[ukey_len - 1] & 0x1f;
@@ -1278,7 +1513,8 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
     if (cur_iv == j) *entries += 1;
     else if (cur_iv == j + 1) walked_next = true;
     if (!process_entry(sp, data, aux, key, key_len, er.value, er.value_len,
-                       rk_len, &rc)) { fail = true; break; }
+                       rk_len, &rc, &rdr)) { fail = true; break; }
+    rdr.seek(q);
     p = q;
   }
   if (!fail && row_open && rc.found && in_bounds(sp, rk_save, rk_len, aux)) {
