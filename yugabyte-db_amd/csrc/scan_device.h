@@ -79,6 +79,8 @@ struct DevSpec {
   uint32_t value_pred_mask;  // bit i: predicate i is on a value column
   uint32_t fixed_rk_len;     // encoded DocKey length when schema has no
                              // string key columns (0 = variable)
+  int32_t need_rowkey;       // bounds / key predicates / row emission need
+                             // the finalized row's key bytes (rk_save)
 };
 
 struct Interval {
@@ -268,14 +270,22 @@ DEV int dht_size_from_start(const uint8_t* p, const uint8_t* limit) {
 
 template <class PtrT>
 DEV void slice_u128(PtrT p, uint32_t len, uint64_t* hi, uint64_t* lo) {
-  uint64_t h = 0, l = 0;
   uint32_t n = len < 16 ? len : 16;
-  for (uint32_t i = 0; i < n; ++i) {
-    uint64_t b = p[i];
-    if (i < 8) h |= b << (56 - 8 * i);
-    else l |= b << (56 - 8 * (i - 8));
+  uint64_t h = __builtin_bswap64(load_u64_una(&p[0]));
+  if (n < 8) {
+    uint32_t sh = 8 * (8 - n);
+    h = (h >> sh) << sh;
+    *hi = h;
+    *lo = 0;
+    return;
   }
   *hi = h;
+  uint64_t l = 0;
+  if (n > 8) {
+    l = __builtin_bswap64(load_u64_una(&p[8]));
+    uint32_t sh = 8 * (16 - n);
+    l = (l >> sh) << sh;
+  }
   *lo = l;
 }
 
@@ -423,10 +433,18 @@ struct EntryRef {
 // bytes into LDS (restarts, no-reuse deltas), the register is reloaded from
 // LDS afterwards; stale tail bytes in LDS are never read (the middle-copy
 // consults *last8 for source positions inside the previous key's tail).
+// rkb/changed: when rkb > 0, any write that modifies a key byte at a
+// position < rkb sets *changed — used for row-change detection without a
+// separate saved-rowkey compare (valid for fixed-length rowkeys).
 DEV const uint8_t* decode_entry_ptr(int fmt, const uint8_t* p,
                                     const uint8_t* limit, uint8_t* key,
                                     uint32_t* key_len, uint64_t* last8,
+                                    uint32_t rkb, bool* changed,
                                     EntryRef* out) {
+  auto wr = [&](uint64_t pos, uint8_t b) {
+    if (pos < rkb && key[pos] != b) *changed = true;
+    key[pos] = b;
+  };
   if (fmt == YBG_ENC_SHARED_PREFIX) {
     uint64_t shared, non_shared, value_len;
     if (!(p = leb128(p, limit, &shared))) return nullptr;
@@ -441,9 +459,10 @@ DEV const uint8_t* decode_entry_ptr(int fmt, const uint8_t* p,
     if (prev_len >= 8 && shared > prev_len - 8) {
       uint64_t old8 = *last8;
       for (int i = 0; i < 8; ++i)
-        key[prev_len - 8 + i] = (uint8_t)(old8 >> (8 * i));
+        wr(prev_len - 8 + i, (uint8_t)(old8 >> (8 * i)));
     }
-    for (uint32_t i = 0; i < (uint32_t)non_shared; ++i) key[shared + i] = p[i];
+    for (uint32_t i = 0; i < (uint32_t)non_shared; ++i)
+      wr(shared + i, p[i]);
     *key_len = (uint32_t)new_len;
     *last8 = load_u64_una(key + new_len - 8);
     out->value = p + non_shared;
@@ -512,7 +531,7 @@ DEV const uint8_t* decode_entry_ptr(int fmt, const uint8_t* p,
   if (!shared_something) {
     // restart / full key inline (block.cc:311-317)
     if (ns1 > kKeyCap || ns1 < 9) return nullptr;
-    for (uint32_t i = 0; i < ns1; ++i) key[i] = p[i];
+    for (uint32_t i = 0; i < ns1; ++i) wr(i, p[i]);
     *key_len = ns1;
     *last8 = load_u64_una(key + ns1 - 8);
     out->value = p + ns1;
@@ -542,14 +561,14 @@ DEV const uint8_t* decode_entry_ptr(int fmt, const uint8_t* p,
     if (new_mid_start != prev_mid_start && mid > 0) {
       if (new_mid_start < prev_mid_start) {
         for (uint64_t i = 0; i < mid; ++i)
-          key[new_mid_start + i] = key[prev_mid_start + i];
+          wr(new_mid_start + i, key[prev_mid_start + i]);
       } else {
         for (uint64_t i = mid; i-- > 0;)
-          key[new_mid_start + i] = key[prev_mid_start + i];
+          wr(new_mid_start + i, key[prev_mid_start + i]);
       }
     }
-    for (uint32_t i = 0; i < ns1; ++i) key[shared_prefix + i] = p[i];
-    for (uint32_t i = 0; i < ns2; ++i) key[new_ns2_start + i] = p[ns1 + i];
+    for (uint32_t i = 0; i < ns1; ++i) wr(shared_prefix + i, p[i]);
+    for (uint32_t i = 0; i < ns2; ++i) wr(new_ns2_start + i, p[ns1 + i]);
     *last8 += last8_inc;
   } else {
     // No tail reuse (rare): the retained prefix or the shared middle may
@@ -559,19 +578,19 @@ DEV const uint8_t* decode_entry_ptr(int fmt, const uint8_t* p,
     if (prev_len >= 8) {
       uint64_t old8 = *last8;
       for (int i = 0; i < 8; ++i)
-        key[prev_len - 8 + i] = (uint8_t)(old8 >> (8 * i));
+        wr(prev_len - 8 + i, (uint8_t)(old8 >> (8 * i)));
     }
     if (new_mid_start != prev_mid_start && mid > 0) {
       if (new_mid_start < prev_mid_start) {
         for (uint64_t i = 0; i < mid; ++i)
-          key[new_mid_start + i] = key[prev_mid_start + i];
+          wr(new_mid_start + i, key[prev_mid_start + i]);
       } else {
         for (uint64_t i = mid; i-- > 0;)
-          key[new_mid_start + i] = key[prev_mid_start + i];
+          wr(new_mid_start + i, key[prev_mid_start + i]);
       }
     }
-    for (uint32_t i = 0; i < ns1; ++i) key[shared_prefix + i] = p[i];
-    for (uint32_t i = 0; i < ns2; ++i) key[new_ns2_start + i] = p[ns1 + i];
+    for (uint32_t i = 0; i < ns1; ++i) wr(shared_prefix + i, p[i]);
+    for (uint32_t i = 0; i < ns2; ++i) wr(new_ns2_start + i, p[ns1 + i]);
     *last8 = load_u64_una(key + new_key_size - 8);
   }
   *key_len = (uint32_t)new_key_size;
@@ -590,7 +609,8 @@ DEV const uint8_t* decode_entry_ptr(int fmt, const uint8_t* p,
 // reader is positioned AT THE VALUE START (value not consumed).
 DEV const uint8_t* decode_entry(int fmt, Rdr* rdr, const uint8_t* limit,
                                 uint8_t* key, uint32_t* key_len,
-                                uint64_t* last8, EntryRef* out) {
+                                uint64_t* last8, uint32_t rkb, bool* changed,
+                                EntryRef* out) {
   const uint8_t* p = rdr->pos();
   if (fmt == YBG_ENC_THREE_SHARED_PARTS && limit - p >= 8) {
     uint64_t w = rdr->peek8();
@@ -656,12 +676,17 @@ DEV const uint8_t* decode_entry(int fmt, Rdr* rdr, const uint8_t* limit,
             for (uint32_t i = 0; i < ns1; ++i) {
               uint32_t j = hl + i;
               uint64_t src = j < 8 ? w : w2;
-              key[sp + i] = (uint8_t)(src >> (8 * (j & 7)));
+              uint8_t b = (uint8_t)(src >> (8 * (j & 7)));
+              if (sp + i < rkb && key[sp + i] != b) *changed = true;
+              key[sp + i] = b;
             }
             for (uint32_t i = 0; i < ns2; ++i) {
               uint32_t j = hl + ns1 + i;
               uint64_t src = j < 8 ? w : w2;
-              key[new_ns2_start + i] = (uint8_t)(src >> (8 * (j & 7)));
+              uint8_t b = (uint8_t)(src >> (8 * (j & 7)));
+              if (new_ns2_start + i < rkb && key[new_ns2_start + i] != b)
+                *changed = true;
+              key[new_ns2_start + i] = b;
             }
             *last8 += inc;
             *key_len = (uint32_t)new_len;
@@ -676,7 +701,7 @@ DEV const uint8_t* decode_entry(int fmt, Rdr* rdr, const uint8_t* limit,
   }
   // general path
   const uint8_t* q =
-      decode_entry_ptr(fmt, p, limit, key, key_len, last8, out);
+      decode_entry_ptr(fmt, p, limit, key, key_len, last8, rkb, changed, out);
   if (!q) return nullptr;
   rdr->seek(out->value);
   return q;
@@ -1446,6 +1471,11 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
   RowCtxT<NA> rc;
   row_reset(&rc, sp);
   EntryRef er;
+  // Fixed-rowkey fast mode: row changes are detected inside the decoder
+  // (compare-on-write below rkb); rk_save is not maintained (finalize needs
+  // no rowkey bytes when no bounds/key predicates are set).
+  const uint32_t rkb =
+      (sp.fixed_rk_len && !sp.need_rowkey) ? sp.fixed_rk_len : 0;
 
   for (;;) {
     if (p >= limit) {
@@ -1460,8 +1490,9 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
       // key state is overwritten by its full-key decode.
       continue;
     }
-    const uint8_t* q =
-        decode_entry(sp.fmt, &rdr, limit, key, &key_len, &reg_last8, &er);
+    bool kchg = false;
+    const uint8_t* q = decode_entry(sp.fmt, &rdr, limit, key, &key_len,
+                                    &reg_last8, rkb, &kchg, &er);
     if (!q || key_len < 10) { fail = true; break; }
     uint32_t ukey_len = key_len - 8;
     uint32_t ht_sz = key[ukey_len - 1] & 0x1f;
@@ -1477,25 +1508,31 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
       if (!rk || rk > prefix_len) { fail = true; break; }
     }
 
-    // Row-change detection: bytes [0, er.shared) are identical to the
-    // previous entry's key by construction (delta encoding), and while the
-    // row is open the previous entry's rowkey == rk_save, so only the tail
-    // [er.shared, rk) needs comparing.
-    bool row_change = !row_open || (rk != rk_len);
-    if (!row_change && er.shared < rk) {
-      for (uint32_t i = er.shared; i < rk; ++i) {
-        if (key[i] != rk_save[i]) { row_change = true; break; }
+    bool row_change;
+    if (rkb) {
+      row_change = !row_open || kchg;
+    } else {
+      // Bytes [0, er.shared) are identical to the previous entry's key by
+      // construction (delta encoding), and while the row is open the
+      // previous entry's rowkey == rk_save, so only [er.shared, rk)
+      // needs comparing.
+      row_change = !row_open || (rk != rk_len);
+      if (!row_change && er.shared < rk) {
+        for (uint32_t i = er.shared; i < rk; ++i) {
+          if (key[i] != rk_save[i]) { row_change = true; break; }
+        }
       }
     }
     if (row_change) {
       if (row_open) {
         // finalize previous row (doc_rowwise_iterator row boundary)
-        if (rc.found && in_bounds(sp, rk_save, rk_len, aux)) {
+        const uint8_t* rkp = rkb ? key : rk_save;
+        if (rc.found && in_bounds(sp, rkp, rk_len, aux)) {
           uint64_t* sc = in_head ? &ho->scanned : scanned;
           uint64_t* ma = in_head ? &ho->matched : matched;
           *sc += 1;
           if ((rc.pred_pass & sp.value_pred_mask) == sp.value_pred_mask &&
-              eval_key_preds(sp, rk_save, rk_len, aux)) {
+              eval_key_preds(sp, rkp, rk_len, aux)) {
             *ma += 1;
             acc_row(sp, rc, in_head ? ho->val : agg_val,
                     in_head ? ho->cnt : agg_cnt);
@@ -1505,7 +1542,9 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
         row_open = false;
         if (cur_iv > j) break;  // tail walk ended at a new row
       }
-      for (uint32_t i = 0; i < rk; ++i) rk_save[i] = key[i];
+      if (!rkb) {
+        for (uint32_t i = 0; i < rk; ++i) rk_save[i] = key[i];
+      }
       rk_len = rk;
       row_open = true;
       row_reset(&rc, sp);
@@ -1517,12 +1556,13 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
     rdr.seek(q);
     p = q;
   }
-  if (!fail && row_open && rc.found && in_bounds(sp, rk_save, rk_len, aux)) {
+  const uint8_t* rkp_end = rkb ? key : rk_save;
+  if (!fail && row_open && rc.found && in_bounds(sp, rkp_end, rk_len, aux)) {
     uint64_t* sc = in_head ? &ho->scanned : scanned;
     uint64_t* ma = in_head ? &ho->matched : matched;
     *sc += 1;
     if ((rc.pred_pass & sp.value_pred_mask) == sp.value_pred_mask &&
-        eval_key_preds(sp, rk_save, rk_len, aux)) {
+        eval_key_preds(sp, rkp_end, rk_len, aux)) {
       *ma += 1;
       acc_row(sp, rc, in_head ? ho->val : agg_val, in_head ? ho->cnt : agg_cnt);
     }
@@ -1631,6 +1671,12 @@ inline void build_dev_spec(const ybg_scan_spec_t* spec, DevSpec* dp,
   d.lower_off = put(spec->lower_bound, spec->lower_bound_len);
   d.upper_len = (uint32_t)spec->upper_bound_len;
   d.upper_off = put(spec->upper_bound, spec->upper_bound_len);
+  {
+    bool kp = false;
+    for (int i = 0; i < spec->num_preds; ++i)
+      if (spec->preds[i].is_key_col) kp = true;
+    d.need_rowkey = kp || spec->lower_bound_len || spec->upper_bound_len;
+  }
   // fixed rowkey length fast path (doc_key.h:40-63 layout) when no string
   // key columns
   {

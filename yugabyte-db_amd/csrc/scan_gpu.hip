@@ -52,7 +52,7 @@ namespace {
 // ---------------------------------------------------------------------------
 
 template <int NA>
-__global__ __launch_bounds__(kThreads) void k_scan(
+__global__ __launch_bounds__(kThreads, 4) void k_scan(
     DevSpec sp, const uint8_t* __restrict__ data,
     const uint64_t* __restrict__ block_offsets,
     const Interval* __restrict__ ivs, uint64_t n_ivs,
