@@ -51,8 +51,8 @@ namespace {
 // Phase-1 scan kernel (aggregate mode): one thread per interval
 // ---------------------------------------------------------------------------
 
-template <int NA>
-__global__ __launch_bounds__(kThreads, 4) void k_scan(
+template <int NA, int WPS>
+__global__ __launch_bounds__(kThreads, WPS) void k_scan(
     DevSpec sp, const uint8_t* __restrict__ data,
     const uint64_t* __restrict__ block_offsets,
     const Interval* __restrict__ ivs, uint64_t n_ivs,
@@ -426,22 +426,36 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
                          s->n_heads * kHeadStride * sizeof(uint64_t),
                          s->stream));
   HIP_TRY(hipEventRecord(s->ev_start, s->stream));
-  // dispatch on aggregate-slot capacity: smaller NA => fewer live VGPRs
-  if (s->dspec.num_aggs <= 2) {
-    hipLaunchKernelGGL(k_scan<2>, dim3(s->grid), dim3(kThreads), 0, s->stream,
+  // dispatch on aggregate-slot capacity (register footprint) and the
+  // waves-per-SIMD occupancy bound (YBG_WPS for tuning, default 4)
+  int wps = 4;
+  if (const char* e = getenv("YBG_WPS")) {
+    long v = atol(e);
+    if (v >= 2 && v <= 6) wps = (int)v;
+  }
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(s->grid), dim3(kThreads), 0, s->stream,
                        s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
                        s->d_aux, s->d_rk_save, s->d_partials, s->d_heads,
                        s->d_cont);
-  } else if (s->dspec.num_aggs <= 4) {
-    hipLaunchKernelGGL(k_scan<4>, dim3(s->grid), dim3(kThreads), 0, s->stream,
-                       s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
-                       s->d_aux, s->d_rk_save, s->d_partials, s->d_heads,
-                       s->d_cont);
-  } else {
-    hipLaunchKernelGGL(k_scan<8>, dim3(s->grid), dim3(kThreads), 0, s->stream,
-                       s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
-                       s->d_aux, s->d_rk_save, s->d_partials, s->d_heads,
-                       s->d_cont);
+  };
+  int na = s->dspec.num_aggs <= 2 ? 2 : (s->dspec.num_aggs <= 4 ? 4 : 8);
+  switch (na * 10 + wps) {
+    case 22: launch(k_scan<2, 2>); break;
+    case 23: launch(k_scan<2, 3>); break;
+    case 25: launch(k_scan<2, 5>); break;
+    case 26: launch(k_scan<2, 6>); break;
+    case 42: launch(k_scan<4, 2>); break;
+    case 43: launch(k_scan<4, 3>); break;
+    case 45: launch(k_scan<4, 5>); break;
+    case 46: launch(k_scan<4, 6>); break;
+    case 82: launch(k_scan<8, 2>); break;
+    case 83: launch(k_scan<8, 3>); break;
+    case 85: launch(k_scan<8, 5>); break;
+    case 86: launch(k_scan<8, 6>); break;
+    case 44: launch(k_scan<4, 4>); break;
+    case 84: launch(k_scan<8, 4>); break;
+    default: launch(k_scan<2, 4>); break;
   }
   HIP_TRY(hipEventRecord(s->ev_mid, s->stream));
   hipLaunchKernelGGL(k_reduce, dim3(1), dim3(256), 0, s->stream, s->dspec,
