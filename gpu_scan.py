@@ -27,6 +27,20 @@ def _lib():
     lib.yb_gpu_scan_aggregate.restype = C.c_int
     lib.yb_gpu_scan_aggregate.argtypes = [C.c_void_p,
                                           C.POINTER(y.ScanResult)]
+    lib.yb_gpu_scan_next_batch.restype = C.c_int
+    lib.yb_gpu_scan_next_batch.argtypes = [C.c_void_p,
+                                           C.POINTER(y.RowBatch)]
+    lib.yb_host_iter_open.restype = C.c_void_p
+    lib.yb_host_iter_open.argtypes = [C.POINTER(y.ScanSpec),
+                                      C.POINTER(C.c_uint8),
+                                      C.POINTER(C.c_uint64), C.c_uint64]
+    lib.yb_host_iter_next.restype = C.c_int
+    lib.yb_host_iter_next.argtypes = [C.c_void_p, C.POINTER(C.c_uint64),
+                                      C.POINTER(C.c_uint64),
+                                      C.POINTER(C.c_uint32),
+                                      C.POINTER(C.POINTER(C.c_uint8))]
+    lib.yb_host_iter_close.restype = None
+    lib.yb_host_iter_close.argtypes = [C.c_void_p]
     lib.yb_gpu_scan_kernel_ms.restype = C.c_int
     lib.yb_gpu_scan_kernel_ms.argtypes = [C.c_void_p, C.POINTER(C.c_double),
                                           C.POINTER(C.c_double)]
@@ -75,6 +89,16 @@ class GpuScan:
         self._check(self._lib.yb_gpu_scan_aggregate(self._h, C.byref(res)),
                     "aggregate")
         return res
+
+    def batch_rows(self):
+        """Materialize matching rows (next_batch) and decode into python
+        tuples sorted by sort_key."""
+        b = y.RowBatch()
+        self._check(self._lib.yb_gpu_scan_next_batch(self._h, C.byref(b)),
+                    "next_batch")
+        return y.decode_batch_rows(self._spec.schema, b.n_rows, b.sort_key,
+                                   b.key_datums, b.datums, b.null_masks,
+                                   b.varlen)
 
     def kernel_ms(self):
         total = C.c_double()

@@ -420,6 +420,69 @@ def sim_scan(spec, data, offsets, n_blocks):
     return res
 
 
+def sim_emit(spec, data, offsets, n_blocks, row_cap=1 << 20,
+             varlen_cap=1 << 24):
+    """Host-simulator row emission (exact device emit path, serial) —
+    TEST INFRASTRUCTURE. Returns list of rows sorted by sort_key:
+    (key_datums tuple, value tuple with None for NULL / bytes for strings)."""
+    lib = product()
+    f = _sig(lib, "ybg_sim_emit", C.c_int,
+             [C.POINTER(ScanSpec), C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
+              C.c_uint64, C.c_uint64, C.POINTER(C.c_uint64),
+              C.POINTER(C.c_uint64), C.POINTER(C.c_uint64),
+              C.POINTER(C.c_uint32), C.POINTER(C.c_uint8), C.c_uint64,
+              C.POINTER(C.c_uint64), C.POINTER(C.c_uint64)])
+    nk = spec.schema.num_hash_cols + spec.schema.num_range_cols
+    nc = spec.schema.num_value_cols
+    sort_key = (C.c_uint64 * row_cap)()
+    key_datums = (C.c_uint64 * (row_cap * max(nk, 1)))()
+    datums = (C.c_uint64 * (row_cap * max(nc, 1)))()
+    null_masks = (C.c_uint32 * row_cap)()
+    varlen = (C.c_uint8 * varlen_cap)()
+    n_rows = C.c_uint64()
+    vl = C.c_uint64()
+    rc = f(C.byref(spec), data, offsets, n_blocks, row_cap, sort_key,
+           key_datums, datums, null_masks, varlen, varlen_cap,
+           C.byref(n_rows), C.byref(vl))
+    if rc != 0:
+        raise RuntimeError(f"ybg_sim_emit rc={rc}")
+    return decode_batch_rows(spec.schema, n_rows.value, sort_key, key_datums,
+                             datums, null_masks, varlen)
+
+
+def decode_batch_rows(schema, n_rows, sort_key, key_datums, datums,
+                      null_masks, varlen):
+    """Decode emitted rows into python tuples, sorted by sort_key."""
+    nk = schema.num_hash_cols + schema.num_range_cols
+    nc = schema.num_value_cols
+    rows = []
+    for r in range(n_rows):
+        kd = []
+        for c in range(nk):
+            d = key_datums[r * max(nk, 1) + c]
+            if schema.key_types[c] == KT_STRING:
+                off = d & ((1 << 40) - 1)
+                ln = d >> 40
+                kd.append(bytes(varlen[off:off + ln]))
+            else:
+                kd.append(d)
+        vals = []
+        nm = null_masks[r]
+        for c in range(nc):
+            if (nm >> c) & 1:
+                vals.append(None)
+            elif schema.value_cols[c].dtype == T_STRING:
+                d = datums[r * max(nc, 1) + c]
+                off = d & ((1 << 40) - 1)
+                ln = d >> 40
+                vals.append(bytes(varlen[off:off + ln]))
+            else:
+                vals.append(datums[r * max(nc, 1) + c])
+        rows.append((sort_key[r], tuple(kd), tuple(vals)))
+    rows.sort(key=lambda t: t[0])
+    return [(kd, vals) for (_, kd, vals) in rows]
+
+
 def orcl_schema_from(schema):
     o = OrclSchema()
     o.has_hash = schema.has_hash
@@ -480,7 +543,13 @@ def orcl_scan(data, offsets, n_blocks, schema, spec, kv_format=ENC_THREE_SHARED_
                 else:
                     vals.append(r.datums[i])
             nk = sc.num_hash_cols + sc.num_range_cols
-            rows.append((tuple(r.key_datums[i] for i in range(nk)), tuple(vals)))
+            kd = []
+            for i in range(nk):
+                if sc.key_types[i] == KT_STRING:
+                    kd.append(C.string_at(r.key_str[i], r.key_str_len[i]))
+                else:
+                    kd.append(r.key_datums[i])
+            rows.append((tuple(kd), tuple(vals)))
             return 0
 
         cbf = ORCL_ROW_CB(cb)

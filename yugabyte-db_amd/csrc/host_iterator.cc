@@ -1,0 +1,112 @@
+// yugabyte-db_amd/csrc/host_iterator.cc — see host_iterator.h.
+#include "host_iterator.h"
+
+#include <algorithm>
+#include <cstring>
+
+namespace ybg {
+
+GpuDocRowwiseIterator::GpuDocRowwiseIterator(const ybg_scan_spec_t& spec)
+    : spec_(spec) {
+  spec_.emit_rows = 1;
+  open_rc_ = yb_gpu_scan_open(&spec_, &handle_);
+}
+
+GpuDocRowwiseIterator::~GpuDocRowwiseIterator() {
+  if (handle_) yb_gpu_scan_close(handle_);
+}
+
+int GpuDocRowwiseIterator::FeedBlocks(const uint8_t* blocks,
+                                      const uint64_t* offsets,
+                                      uint64_t n_blocks,
+                                      bool device_resident) {
+  if (open_rc_) return open_rc_;
+  return yb_gpu_scan_feed_blocks(handle_, blocks, offsets, n_blocks,
+                                 device_resident ? 1 : 0);
+}
+
+int GpuDocRowwiseIterator::MaterializeBatch() {
+  int rc = yb_gpu_scan_next_batch(handle_, &batch_);
+  if (rc) return rc;
+  // Restore tablet key order: the kernels emit rows in completion order
+  // with a scan-position sort key (interval, entry offset).
+  order_.resize(batch_.n_rows);
+  for (uint64_t i = 0; i < batch_.n_rows; ++i) order_[i] = i;
+  const uint64_t* sk = batch_.sort_key;
+  std::sort(order_.begin(), order_.end(),
+            [sk](uint64_t a, uint64_t b) { return sk[a] < sk[b]; });
+  pos_ = 0;
+  batch_ready_ = true;
+  return 0;
+}
+
+int GpuDocRowwiseIterator::PgFetchNext(PgRow* row) {
+  if (open_rc_) return -open_rc_;
+  if (!batch_ready_) {
+    int rc = MaterializeBatch();
+    if (rc) return -rc;
+  }
+  if (pos_ >= batch_.n_rows) return 0;
+  uint64_t r = order_[pos_++];
+  uint64_t nk = batch_.n_key_cols, nc = batch_.n_value_cols;
+  for (uint64_t c = 0; c < nk; ++c)
+    row->key_datums[c] = batch_.key_datums[r * nk + c];
+  for (uint64_t c = 0; c < nc; ++c)
+    row->datums[c] = batch_.datums[r * nc + c];
+  row->null_mask = batch_.null_masks[r];
+  row->varlen = batch_.varlen;
+  return 1;
+}
+
+int GpuDocRowwiseIterator::Aggregate(ybg_scan_result_t* out) {
+  if (open_rc_) return open_rc_;
+  int rc = yb_gpu_scan_execute(handle_);
+  if (rc) return rc;
+  return yb_gpu_scan_aggregate(handle_, out);
+}
+
+int GpuDocRowwiseIterator::PagingState(uint8_t* key_out, size_t cap,
+                                       size_t* len) {
+  return yb_gpu_scan_paging_state(handle_, key_out, cap, len);
+}
+
+const char* GpuDocRowwiseIterator::LastError() const {
+  return yb_gpu_last_error();
+}
+
+}  // namespace ybg
+
+// ---------------------------------------------------------------------------
+// C wrapper so the test suite can drive the C++ adapter row-at-a-time.
+// ---------------------------------------------------------------------------
+extern "C" {
+
+void* yb_host_iter_open(const ybg_scan_spec_t* spec, const uint8_t* blocks,
+                        const uint64_t* offsets, uint64_t n_blocks) {
+  auto* it = new ybg::GpuDocRowwiseIterator(*spec);
+  if (it->FeedBlocks(blocks, offsets, n_blocks, false) != 0) {
+    delete it;
+    return nullptr;
+  }
+  return it;
+}
+
+// Returns 1 and fills the row arrays, 0 at end, <0 on error.
+int yb_host_iter_next(void* h, uint64_t* key_datums, uint64_t* datums,
+                      uint32_t* null_mask, const uint8_t** varlen) {
+  auto* it = static_cast<ybg::GpuDocRowwiseIterator*>(h);
+  ybg::PgRow row;
+  int rc = it->PgFetchNext(&row);
+  if (rc != 1) return rc;
+  memcpy(key_datums, row.key_datums, sizeof(row.key_datums));
+  memcpy(datums, row.datums, sizeof(row.datums));
+  *null_mask = row.null_mask;
+  *varlen = row.varlen;
+  return 1;
+}
+
+void yb_host_iter_close(void* h) {
+  delete static_cast<ybg::GpuDocRowwiseIterator*>(h);
+}
+
+}  // extern "C"

@@ -30,6 +30,12 @@ static inline float __uint_as_float(unsigned v) {
   memcpy(&f, &v, 4);
   return f;
 }
+static inline unsigned long long atomicAdd(unsigned long long* p,
+                                           unsigned long long v) {
+  unsigned long long old = *p;
+  *p += v;
+  return old;
+}
 #endif
 
 namespace ybgdev {
@@ -791,6 +797,23 @@ DEV int skip_control(const uint8_t* v, uint32_t len) {
 // null bit per slot.
 // ---------------------------------------------------------------------------
 
+// Row materialization target (next_batch / PgFetchNext path). All pointers
+// device-global (host in the simulator).
+struct EmitCtx {
+  uint64_t* sort_key;    // [row_cap]
+  uint64_t* key_datums;  // [row_cap * nk]
+  uint64_t* datums;      // [row_cap * nc]; strings: (len<<40)|varlen offset
+  uint32_t* null_masks;  // [row_cap]
+  uint8_t* varlen;
+  uint64_t varlen_cap;
+  unsigned long long* row_counter;
+  unsigned long long* varlen_counter;
+  uint64_t row_cap;
+  int nk, nc;
+  const uint32_t* head_consumed;  // [n_ivs] from the flags pre-pass
+  unsigned long long* overflow;   // set nonzero when a cap was exceeded
+};
+
 template <int NA>
 struct RowCtxT {
   bool base_seen;
@@ -803,6 +826,11 @@ struct RowCtxT {
                        // finalize)
   uint32_t agg_null;   // bit g: aggregate operand g is NULL
   uint64_t agg_datum[NA];
+  // emit mode only (nullptr otherwise): per-thread row buffer in LDS
+  uint64_t* emit_datums;  // [nc]: numeric datum or (global str address)
+  uint32_t* emit_lens;    // [nc]: string length (numeric: 0)
+  uint32_t emit_null;     // bit i: column i NULL
+  uint32_t emit_str;      // bit i: column i holds a string address
 };
 
 template <int NA>
@@ -813,6 +841,8 @@ DEV void row_reset(RowCtxT<NA>* rc, const DevSpec& sp) {
   rc->cur_col_done = false;
   rc->pred_pass = 0;
   rc->agg_null = 0xffffffffu;
+  rc->emit_null = 0xffffffffu;
+  rc->emit_str = 0;
 }
 
 // Value-column predicate compare (pgsql_operation.cc:602-668 typed-compare
@@ -859,6 +889,20 @@ template <int NA>
 DEV void eval_col(const DevSpec& sp, const uint8_t* aux, RowCtxT<NA>* rc,
                   int col, bool is_null, uint64_t datum, const uint8_t* sptr,
                   uint32_t slen) {
+  if (rc->emit_datums) {
+    rc->emit_null = (rc->emit_null & ~(1u << col)) | ((uint32_t)is_null << col);
+    if (!is_null) {
+      if (sptr) {
+        rc->emit_datums[col] = (uint64_t)(uintptr_t)sptr;
+        rc->emit_lens[col] = slen;
+        rc->emit_str |= 1u << col;
+      } else {
+        rc->emit_datums[col] = datum;
+        rc->emit_lens[col] = 0;
+        rc->emit_str &= ~(1u << col);
+      }
+    }
+  }
 #pragma unroll
   for (int i = 0; i < YBG_MAX_PREDS; ++i) {
     if (i < sp.num_preds && !sp.preds[i].is_key_col &&
@@ -1440,14 +1484,76 @@ struct HeadOut {
   uint64_t scanned, matched;
 };
 
+// Write one materialized row (PgTableRow analog — dockv/pg_row.h:91-179).
+// rk bytes = the finalized row's key (rk_save), used for key-column datums.
 template <int NA>
+DEV void emit_row(const DevSpec& sp, EmitCtx* ec, const RowCtxT<NA>& rc,
+                  const uint8_t* rk, uint32_t rk_len, uint64_t sort_key) {
+  unsigned long long slot = atomicAdd(ec->row_counter, 1ull);
+  if (slot >= ec->row_cap) {
+    atomicAdd(ec->overflow, 1ull);
+    return;
+  }
+  ec->sort_key[slot] = sort_key;
+  int nk = ec->nk;
+  for (int c = 0; c < nk; ++c) {
+    uint64_t d = 0;
+    uint32_t soff = 0, sl = 0;
+    if (key_col_value(sp, rk, rk_len, c, &d, &soff, &sl)) {
+      if (sp.key_types[c] == YBG_KT_STRING) {
+        // unescape into the varlen heap ('\0\1' -> '\0',
+        // doc_kv_util.h:101-167)
+        unsigned long long off = atomicAdd(ec->varlen_counter, sl);
+        if (off + sl > ec->varlen_cap) {
+          atomicAdd(ec->overflow, 1ull);
+          d = 0;
+        } else {
+          uint32_t o = 0;
+          for (uint32_t i = 0; i < sl; ++i) {
+            uint8_t b = rk[soff + i];
+            ec->varlen[off + o++] = b;
+            if (b == 0) ++i;  // skip escape byte
+          }
+          d = ((uint64_t)o << 40) | off;
+        }
+      }
+      ec->key_datums[slot * nk + c] = d;
+    }
+  }
+  int nc = ec->nc;
+  for (int c = 0; c < nc; ++c) {
+    uint64_t d = 0;
+    if (!((rc.emit_null >> c) & 1)) {
+      d = rc.emit_datums[c];
+      if ((rc.emit_str >> c) & 1) {
+        uint32_t sl = rc.emit_lens[c];
+        unsigned long long off = atomicAdd(ec->varlen_counter, sl);
+        if (off + sl > ec->varlen_cap) {
+          atomicAdd(ec->overflow, 1ull);
+          d = 0;
+        } else {
+          const uint8_t* src = (const uint8_t*)(uintptr_t)d;
+          for (uint32_t i = 0; i < sl; ++i) ec->varlen[off + i] = src[i];
+          d = ((uint64_t)sl << 40) | off;
+        }
+      }
+    }
+    ec->datums[slot * nc + c] = d;
+  }
+  ec->null_masks[slot] = rc.emit_null;
+}
+
+template <int NA, bool EMIT = false>
 DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
                            const uint64_t* block_offsets, const Interval* ivs,
                            uint64_t n_ivs, uint64_t j, const uint8_t* aux,
                            uint8_t* key, uint8_t* rk_save, uint64_t* entries,
                            uint64_t* scanned, uint64_t* matched,
                            uint64_t* agg_val, uint64_t* agg_cnt,
-                           HeadOut<NA>* ho, bool* walked_next_out) {
+                           HeadOut<NA>* ho, bool* walked_next_out,
+                           EmitCtx* ec = nullptr,
+                           uint64_t* emit_datums = nullptr,
+                           uint32_t* emit_lens = nullptr) {
   Interval iv = ivs[j];
   const uint8_t* blk = data + block_offsets[iv.block];
   const uint8_t* p = blk + iv.start;
@@ -1469,8 +1575,11 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
   bool row_open = false;
   bool in_head = true;
   RowCtxT<NA> rc;
+  rc.emit_datums = EMIT ? emit_datums : nullptr;
+  rc.emit_lens = EMIT ? emit_lens : nullptr;
   row_reset(&rc, sp);
   EntryRef er;
+  uint64_t row_sort_key = 0;
   // Fixed-rowkey fast mode: row changes are detected inside the decoder
   // (compare-on-write below rkb); rk_save is not maintained (finalize needs
   // no rowkey bytes when no bounds/key predicates are set).
@@ -1526,7 +1635,7 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
     if (row_change) {
       if (row_open) {
         // finalize previous row (doc_rowwise_iterator row boundary)
-        const uint8_t* rkp = rkb ? key : rk_save;
+        const uint8_t* rkp = (rkb && !EMIT) ? key : rk_save;
         if (rc.found && in_bounds(sp, rkp, rk_len, aux)) {
           uint64_t* sc = in_head ? &ho->scanned : scanned;
           uint64_t* ma = in_head ? &ho->matched : matched;
@@ -1536,17 +1645,21 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
             *ma += 1;
             acc_row(sp, rc, in_head ? ho->val : agg_val,
                     in_head ? ho->cnt : agg_cnt);
+            if (EMIT && (!in_head || ec->head_consumed[j] == 0))
+              emit_row(sp, ec, rc, rkp, rk_len, row_sort_key);
           }
         }
         in_head = false;
         row_open = false;
         if (cur_iv > j) break;  // tail walk ended at a new row
       }
-      if (!rkb) {
+      if (!rkb || EMIT) {
         for (uint32_t i = 0; i < rk; ++i) rk_save[i] = key[i];
       }
       rk_len = rk;
       row_open = true;
+      if (EMIT)  // scan position: (interval, entry offset within block)
+        row_sort_key = ((uint64_t)cur_iv << 16) | (uint64_t)(p - blk);
       row_reset(&rc, sp);
     }
     if (cur_iv == j) *entries += 1;
@@ -1556,7 +1669,7 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
     rdr.seek(q);
     p = q;
   }
-  const uint8_t* rkp_end = rkb ? key : rk_save;
+  const uint8_t* rkp_end = (rkb && !EMIT) ? key : rk_save;
   if (!fail && row_open && rc.found && in_bounds(sp, rkp_end, rk_len, aux)) {
     uint64_t* sc = in_head ? &ho->scanned : scanned;
     uint64_t* ma = in_head ? &ho->matched : matched;
@@ -1565,6 +1678,8 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
         eval_key_preds(sp, rkp_end, rk_len, aux)) {
       *ma += 1;
       acc_row(sp, rc, in_head ? ho->val : agg_val, in_head ? ho->cnt : agg_cnt);
+      if (EMIT && (!in_head || ec->head_consumed[j] == 0))
+        emit_row(sp, ec, rc, rkp_end, rk_len, row_sort_key);
     }
   }
   *walked_next_out = walked_next;
@@ -1675,7 +1790,8 @@ inline void build_dev_spec(const ybg_scan_spec_t* spec, DevSpec* dp,
     bool kp = false;
     for (int i = 0; i < spec->num_preds; ++i)
       if (spec->preds[i].is_key_col) kp = true;
-    d.need_rowkey = kp || spec->lower_bound_len || spec->upper_bound_len;
+    d.need_rowkey = kp || spec->lower_bound_len || spec->upper_bound_len ||
+                    spec->emit_rows;
   }
   // fixed rowkey length fast path (doc_key.h:40-63 layout) when no string
   // key columns

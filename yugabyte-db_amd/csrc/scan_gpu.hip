@@ -58,7 +58,7 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
     const Interval* __restrict__ ivs, uint64_t n_ivs,
     const uint8_t* __restrict__ aux, uint8_t* __restrict__ rk_save_buf,
     uint64_t* __restrict__ partials, uint64_t* __restrict__ heads,
-    uint32_t* __restrict__ cont_flags) {
+    uint32_t* __restrict__ cont_flags, int write_all_flags) {
   __shared__ uint8_t key_scratch[kThreads * kKeyCap];
   __shared__ uint8_t wave_relay[kThreads / 64 + 1];
   uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
@@ -125,7 +125,10 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
                      ho.cnt[g]);
         }
       }
-      if (walked_next && ((j + 1) % kThreads) == 0 && (j + 1) < n_ivs) {
+      if (write_all_flags) {
+        if (walked_next && (j + 1) < n_ivs) cont_flags[j + 1] = 1;
+      } else if (walked_next && ((j + 1) % kThreads) == 0 &&
+                 (j + 1) < n_ivs) {
         cont_flags[(j + 1) / kThreads] = 1;
       }
     }
@@ -163,6 +166,45 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
       partials[wave_id * kPartialStride + 1] = scanned;
       partials[wave_id * kPartialStride + 2] = matched;
       partials[wave_id * kPartialStride + 3] = errs;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Row-emission kernel (next_batch): a flags pre-pass (k_scan with
+// write_all_flags) resolves head-row ownership for every interval; this
+// kernel then re-walks intervals and materializes matching rows. 128-thread
+// workgroups: each thread carries key scratch + a row buffer in LDS.
+// ---------------------------------------------------------------------------
+
+constexpr int kEmitThreads = 128;
+
+__global__ __launch_bounds__(kEmitThreads) void k_emit(
+    DevSpec sp, const uint8_t* __restrict__ data,
+    const uint64_t* __restrict__ block_offsets,
+    const Interval* __restrict__ ivs, uint64_t n_ivs,
+    const uint8_t* __restrict__ aux, uint8_t* __restrict__ rk_save_buf,
+    EmitCtx ec, unsigned long long* __restrict__ err_counter) {
+  __shared__ uint8_t key_scratch[kEmitThreads * kKeyCap];
+  __shared__ uint64_t rowbuf[kEmitThreads * YBG_MAX_COLS];
+  __shared__ uint32_t lenbuf[kEmitThreads * YBG_MAX_COLS];
+  uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
+  uint64_t* rb = rowbuf + (size_t)threadIdx.x * YBG_MAX_COLS;
+  uint32_t* lb = lenbuf + (size_t)threadIdx.x * YBG_MAX_COLS;
+  const uint32_t gtid = blockIdx.x * kEmitThreads + threadIdx.x;
+  uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
+  const uint64_t span = (uint64_t)gridDim.x * kEmitThreads;
+
+  uint64_t entries = 0, scanned = 0, matched = 0;
+  uint64_t agg_val[2] = {0, 0}, agg_cnt[2] = {0, 0};
+  for (uint64_t j = gtid; j < n_ivs; j += span) {
+    HeadOut<2> ho;
+    bool wn = false;
+    if (!scan_one_interval<2, true>(sp, data, block_offsets, ivs, n_ivs, j,
+                                    aux, key, rk_save, &entries, &scanned,
+                                    &matched, agg_val, agg_cnt, &ho, &wn,
+                                    &ec, rb, lb)) {
+      atomicAdd(err_counter, 1ull);
     }
   }
 }
@@ -290,6 +332,19 @@ struct ybg_scan {
   bool executed = false;
   double last_total_ms = 0, last_decode_ms = 0;
   std::vector<uint8_t> aux_host;
+  // emit (next_batch) buffers
+  uint32_t* d_flags_all = nullptr;
+  uint64_t emit_row_cap = 0;
+  uint64_t* d_em_sort = nullptr;
+  uint64_t* d_em_key = nullptr;
+  uint64_t* d_em_dat = nullptr;
+  uint32_t* d_em_null = nullptr;
+  uint8_t* d_em_varlen = nullptr;
+  uint64_t em_varlen_cap = 0;
+  unsigned long long* d_em_counters = nullptr;  // rows, varlen, overflow, err
+  std::vector<uint64_t> h_sort, h_key, h_dat;
+  std::vector<uint32_t> h_null;
+  std::vector<uint8_t> h_varlen;
 };
 
 extern "C" {
@@ -437,7 +492,7 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
     hipLaunchKernelGGL(kern, dim3(s->grid), dim3(kThreads), 0, s->stream,
                        s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
                        s->d_aux, s->d_rk_save, s->d_partials, s->d_heads,
-                       s->d_cont);
+                       s->d_cont, 0);
   };
   int na = s->dspec.num_aggs <= 2 ? 2 : (s->dspec.num_aggs <= 4 ? 4 : 8);
   switch (na * 10 + wps) {
@@ -508,9 +563,101 @@ int yb_gpu_scan_aggregate(ybg_scan_t* s, ybg_scan_result_t* out) {
 }
 
 int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
-  (void)s;
-  (void)out;
-  return set_err(7, "row materialization batch not yet implemented");
+  if (!s->d_data) return set_err(4, "feed_blocks not called");
+  int nk = s->spec.schema.num_hash_cols + s->spec.schema.num_range_cols;
+  int nc = s->spec.schema.num_value_cols;
+  if (!s->d_flags_all) {
+    s->emit_row_cap = std::min<uint64_t>(s->n_ivs * 16ull, 32ull << 20);
+    s->em_varlen_cap =
+        std::min<uint64_t>(s->total_bytes + (1ull << 20), 1ull << 30);
+    HIP_TRY(hipMalloc(&s->d_flags_all, s->n_ivs * sizeof(uint32_t)));
+    HIP_TRY(hipMalloc(&s->d_em_sort, s->emit_row_cap * 8));
+    HIP_TRY(hipMalloc(&s->d_em_key, s->emit_row_cap * 8 * (nk ? nk : 1)));
+    HIP_TRY(hipMalloc(&s->d_em_dat, s->emit_row_cap * 8 * (nc ? nc : 1)));
+    HIP_TRY(hipMalloc(&s->d_em_null, s->emit_row_cap * 4));
+    HIP_TRY(hipMalloc(&s->d_em_varlen, s->em_varlen_cap));
+    HIP_TRY(hipMalloc(&s->d_em_counters, 4 * sizeof(unsigned long long)));
+  }
+  HIP_TRY(hipMemsetAsync(s->d_flags_all, 0, s->n_ivs * sizeof(uint32_t),
+                         s->stream));
+  HIP_TRY(hipMemsetAsync(s->d_em_counters, 0, 4 * sizeof(unsigned long long),
+                         s->stream));
+  HIP_TRY(hipMemsetAsync(s->d_heads, 0,
+                         s->n_heads * kHeadStride * sizeof(uint64_t),
+                         s->stream));
+  // flags pre-pass: resolves head-row ownership for EVERY interval
+  auto flags_kernel = k_scan<2, 4>;
+  hipLaunchKernelGGL(flags_kernel, dim3(s->grid), dim3(kThreads), 0,
+                     s->stream, s->dspec, s->d_data, s->d_offsets, s->d_ivs,
+                     s->n_ivs, s->d_aux, s->d_rk_save, s->d_partials,
+                     s->d_heads, s->d_flags_all, 1);
+  EmitCtx ec;
+  ec.sort_key = s->d_em_sort;
+  ec.key_datums = s->d_em_key;
+  ec.datums = s->d_em_dat;
+  ec.null_masks = s->d_em_null;
+  ec.varlen = s->d_em_varlen;
+  ec.varlen_cap = s->em_varlen_cap;
+  ec.row_counter = s->d_em_counters + 0;
+  ec.varlen_counter = s->d_em_counters + 1;
+  ec.overflow = s->d_em_counters + 2;
+  ec.row_cap = s->emit_row_cap;
+  ec.nk = nk;
+  ec.nc = nc;
+  ec.head_consumed = s->d_flags_all;
+  int egrid = (int)std::min<uint64_t>(
+      (s->n_ivs + kEmitThreads - 1) / kEmitThreads, 8192);
+  // emit threads use their own rk_save area sized for the emit grid
+  uint64_t need_rk = (uint64_t)egrid * kEmitThreads * kKeyCap;
+  uint64_t have_rk = (uint64_t)s->grid * kThreads * kKeyCap;
+  uint8_t* rk_area = s->d_rk_save;
+  uint8_t* rk_extra = nullptr;
+  if (need_rk > have_rk) {
+    HIP_TRY(hipMalloc(&rk_extra, need_rk));
+    rk_area = rk_extra;
+  }
+  hipLaunchKernelGGL(k_emit, dim3(egrid), dim3(kEmitThreads), 0, s->stream,
+                     s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
+                     s->d_aux, rk_area, ec, s->d_em_counters + 3);
+  unsigned long long ctr[4];
+  HIP_TRY(hipMemcpyAsync(ctr, s->d_em_counters,
+                         4 * sizeof(unsigned long long),
+                         hipMemcpyDeviceToHost, s->stream));
+  HIP_TRY(hipStreamSynchronize(s->stream));
+  if (rk_extra) HIP_WARN(hipFree(rk_extra));
+  if (ctr[3]) return set_err(6, "corrupt entries encountered during scan");
+  if (ctr[2]) return set_err(8, "row batch capacity exceeded");
+  uint64_t n_rows = ctr[0];
+  uint64_t vl = ctr[1];
+  s->h_sort.resize(n_rows);
+  s->h_key.resize(n_rows * (nk ? nk : 1));
+  s->h_dat.resize(n_rows * (nc ? nc : 1));
+  s->h_null.resize(n_rows);
+  s->h_varlen.resize(vl ? vl : 1);
+  if (n_rows) {
+    HIP_TRY(hipMemcpy(s->h_sort.data(), s->d_em_sort, n_rows * 8,
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(s->h_key.data(), s->d_em_key, n_rows * 8 * (nk ? nk : 1),
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(s->h_dat.data(), s->d_em_dat, n_rows * 8 * (nc ? nc : 1),
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(s->h_null.data(), s->d_em_null, n_rows * 4,
+                      hipMemcpyDeviceToHost));
+  }
+  if (vl)
+    HIP_TRY(hipMemcpy(s->h_varlen.data(), s->d_em_varlen, vl,
+                      hipMemcpyDeviceToHost));
+  memset(out, 0, sizeof(*out));
+  out->n_rows = n_rows;
+  out->n_key_cols = (uint64_t)nk;
+  out->n_value_cols = (uint64_t)nc;
+  out->sort_key = s->h_sort.data();
+  out->key_datums = s->h_key.data();
+  out->datums = s->h_dat.data();
+  out->null_masks = s->h_null.data();
+  out->varlen = s->h_varlen.data();
+  out->varlen_size = vl;
+  return 0;
 }
 
 int yb_gpu_scan_paging_state(ybg_scan_t* s, uint8_t* key_out, size_t cap,
@@ -538,6 +685,13 @@ int yb_gpu_scan_close(ybg_scan_t* s) {
   if (s->d_heads) HIP_WARN(hipFree(s->d_heads));
   if (s->d_cont) HIP_WARN(hipFree(s->d_cont));
   if (s->d_result) HIP_WARN(hipFree(s->d_result));
+  if (s->d_flags_all) HIP_WARN(hipFree(s->d_flags_all));
+  if (s->d_em_sort) HIP_WARN(hipFree(s->d_em_sort));
+  if (s->d_em_key) HIP_WARN(hipFree(s->d_em_key));
+  if (s->d_em_dat) HIP_WARN(hipFree(s->d_em_dat));
+  if (s->d_em_null) HIP_WARN(hipFree(s->d_em_null));
+  if (s->d_em_varlen) HIP_WARN(hipFree(s->d_em_varlen));
+  if (s->d_em_counters) HIP_WARN(hipFree(s->d_em_counters));
   if (s->ev_start) HIP_WARN(hipEventDestroy(s->ev_start));
   if (s->ev_mid) HIP_WARN(hipEventDestroy(s->ev_mid));
   if (s->ev_end) HIP_WARN(hipEventDestroy(s->ev_end));

@@ -91,4 +91,85 @@ int ybg_sim_scan(const ybg_scan_spec_t* spec, const uint8_t* data,
   return 0;
 }
 
+// Sequential emit-mode simulator: pass 1 computes walked flags, pass 2
+// re-runs the exact device emit path. Caller provides output arrays.
+int ybg_sim_emit(const ybg_scan_spec_t* spec, const uint8_t* data,
+                 const uint64_t* offsets, uint64_t n_blocks,
+                 uint64_t row_cap, uint64_t* sort_key, uint64_t* key_datums,
+                 uint64_t* datums, uint32_t* null_masks, uint8_t* varlen,
+                 uint64_t varlen_cap, uint64_t* n_rows_out,
+                 uint64_t* varlen_out) {
+  DevSpec d;
+  std::vector<unsigned char> aux(1 << 20);
+  uint32_t aux_len = 0;
+  ybg_scan_spec_t spec2 = *spec;
+  spec2.emit_rows = 1;
+  build_dev_spec(&spec2, &d, aux.data(), &aux_len, (uint32_t)aux.size());
+
+  std::vector<Interval> ivs;
+  for (uint64_t b = 0; b < n_blocks; ++b) {
+    uint64_t sz = offsets[b + 1] - offsets[b];
+    if (sz < 8) return 3;
+    const uint8_t* blk = data + offsets[b];
+    uint32_t nr = load_le32_u(blk + sz - 4);
+    if (nr == 0 || (uint64_t)nr * 4 + 4 > sz) return 3;
+    uint32_t restarts_off = (uint32_t)(sz - 4 - (uint64_t)nr * 4);
+    for (uint32_t r = 0; r < nr; ++r) {
+      uint32_t start = load_le32_u(blk + restarts_off + 4ull * r);
+      uint32_t end = (r + 1 < nr)
+                         ? load_le32_u(blk + restarts_off + 4ull * (r + 1))
+                         : restarts_off;
+      ivs.push_back(Interval{(uint32_t)b, start, end});
+    }
+  }
+  uint64_t n_ivs = ivs.size();
+  uint64_t entries = 0, scanned = 0, matched = 0;
+  uint64_t agg_val[YBG_MAX_AGGS] = {0}, agg_cnt[YBG_MAX_AGGS] = {0};
+  std::vector<uint8_t> walked(n_ivs, 0);
+  std::vector<uint32_t> head_consumed(n_ivs ? n_ivs : 1, 0);
+  uint8_t key[kKeyCap], rk_save[kKeyCap];
+  HeadOut<YBG_MAX_AGGS> ho;
+  for (uint64_t j = 0; j < n_ivs; ++j) {
+    bool wn = false;
+    if (!scan_one_interval<YBG_MAX_AGGS>(d, data, offsets, ivs.data(), n_ivs,
+                                         j, aux.data(), key, rk_save,
+                                         &entries, &scanned, &matched,
+                                         agg_val, agg_cnt, &ho, &wn))
+      return 6;
+    walked[j] = wn ? 1 : 0;
+  }
+  for (uint64_t j = 1; j < n_ivs; ++j) head_consumed[j] = walked[j - 1];
+
+  unsigned long long row_counter = 0, varlen_counter = 0, overflow = 0;
+  EmitCtx ec;
+  ec.sort_key = sort_key;
+  ec.key_datums = key_datums;
+  ec.datums = datums;
+  ec.null_masks = null_masks;
+  ec.varlen = varlen;
+  ec.varlen_cap = varlen_cap;
+  ec.row_counter = &row_counter;
+  ec.varlen_counter = &varlen_counter;
+  ec.overflow = &overflow;
+  ec.row_cap = row_cap;
+  ec.nk = spec->schema.num_hash_cols + spec->schema.num_range_cols;
+  ec.nc = spec->schema.num_value_cols;
+  ec.head_consumed = head_consumed.data();
+  uint64_t rowbuf[YBG_MAX_COLS];
+  uint32_t lenbuf[YBG_MAX_COLS];
+  for (uint64_t j = 0; j < n_ivs; ++j) {
+    bool wn = false;
+    HeadOut<YBG_MAX_AGGS> ho2;
+    if (!scan_one_interval<YBG_MAX_AGGS, true>(
+            d, data, offsets, ivs.data(), n_ivs, j, aux.data(), key, rk_save,
+            &entries, &scanned, &matched, agg_val, agg_cnt, &ho2, &wn, &ec,
+            rowbuf, lenbuf))
+      return 6;
+  }
+  if (overflow) return 8;
+  *n_rows_out = row_counter;
+  *varlen_out = varlen_counter;
+  return 0;
+}
+
 }  // extern "C"
