@@ -41,6 +41,11 @@ def _lib():
                                       C.POINTER(C.POINTER(C.c_uint8))]
     lib.yb_host_iter_close.restype = None
     lib.yb_host_iter_close.argtypes = [C.c_void_p]
+    lib.yb_gpu_scan_group_aggregate.restype = C.c_int
+    lib.yb_gpu_scan_group_aggregate.argtypes = [
+        C.c_void_p, C.POINTER(C.c_uint64), C.POINTER(C.c_int64),
+        C.POINTER(C.c_uint64), C.POINTER(C.c_uint8), C.c_uint64, C.c_uint64,
+        C.POINTER(C.c_uint64)]
     lib.yb_gpu_scan_kernel_ms.restype = C.c_int
     lib.yb_gpu_scan_kernel_ms.argtypes = [C.c_void_p, C.POINTER(C.c_double),
                                           C.POINTER(C.c_double)]
@@ -99,6 +104,21 @@ class GpuScan:
         return y.decode_batch_rows(self._spec.schema, b.n_rows, b.sort_key,
                                    b.key_datums, b.datums, b.null_masks,
                                    b.varlen)
+
+    def group_aggregate(self, cap=1 << 20, key_bytes_cap=1 << 24):
+        """GROUP BY partial aggregates (spec.group_col must be set)."""
+        keys = (C.c_uint64 * cap)()
+        vals = (C.c_int64 * (cap * y.MAX_AGGS))()
+        cnts = (C.c_uint64 * (cap * y.MAX_AGGS))()
+        kb = (C.c_uint8 * key_bytes_cap)()
+        n = C.c_uint64()
+        self._check(
+            self._lib.yb_gpu_scan_group_aggregate(
+                self._h, keys, vals, cnts, kb, key_bytes_cap, cap,
+                C.byref(n)), "group_aggregate")
+        return y._decode_groups(self._spec.schema, self._spec.group_col - 1,
+                                n.value, keys, vals, cnts, kb,
+                                self._spec.num_aggs, self._spec.aggs)
 
     def kernel_ms(self):
         total = C.c_double()

@@ -151,6 +151,9 @@ typedef struct {
   const uint8_t *upper_bound; uint64_t upper_bound_len; /* excl. */
   int32_t emit_rows;  /* 1: materialize matching rows (next_batch) */
   uint64_t row_limit; /* 0 = unlimited; else paging after this many rows */
+  /* GROUP BY (config #5 "GROUP-BY-key partial aggregates"):
+   * 0 = plain aggregates, else 1 + value-column index to group on. */
+  int32_t group_col;
 } ybg_scan_spec_t;
 
 /* ---- scan handle --------------------------------------------------------- */
@@ -196,6 +199,18 @@ typedef struct {
 
 /* Fetch aggregate results (implies wait). */
 int yb_gpu_scan_aggregate(ybg_scan_t *s, ybg_scan_result_t *out);
+
+/* GROUP BY (spec.group_col >= 0): run the grouped scan and return the
+ * per-group partials. keys[g]: numeric group key datum, or for string group
+ * columns (len<<40)|offset into key_bytes. vals/cnts: [n_groups * num_aggs]
+ * (value bit pattern + non-null contribution count; cnt==0 <=> NULL).
+ * Returns 0 and *n_groups on success; error if cap exceeded. Integer
+ * aggregates are exact; grouped double SUM uses device atomics and is only
+ * reproducible up to summation order. */
+int yb_gpu_scan_group_aggregate(ybg_scan_t *s, uint64_t *keys,
+                                int64_t *vals, uint64_t *cnts,
+                                uint8_t *key_bytes, uint64_t key_bytes_cap,
+                                uint64_t cap, uint64_t *n_groups);
 
 /* Materialized row batch (PgTableRow analog — dockv/pg_row.h:91-179).
  * Row order within the batch is by (sort_key) = the row's position in the

@@ -155,3 +155,29 @@ def test_gpu_host_iterator_ordered(cases):
     _, want = y.orcl_scan(case["data"], case["offsets"], case["n_blocks"],
                           osc, ospec, collect_rows=True)
     assert rows == want
+
+
+def test_gpu_group_by():
+    """GROUP BY on device (hash-table partial aggregates) vs oracle —
+    integer aggregates exact, grouped double SUM within tolerance, string
+    and NULL group keys."""
+    import test_group_parity as tg
+    gpu_scan = _gpu()
+    for dataset, gcol in ((tg._dataset_int_groups, 0),
+                          (tg._dataset_str_groups, 0)):
+        schema, built, _b = dataset()
+        data, offsets, nb, total, ne = built
+        if dataset is tg._dataset_int_groups:
+            aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 1),
+                    y.Agg(y.AGG_MIN_INT64, 1), y.Agg(y.AGG_MAX_INT64, 1),
+                    y.Agg(y.AGG_SUM_DOUBLE, 2), y.Agg(y.AGG_COUNT, 1)]
+        else:
+            aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 1)]
+        spec, ospec = tg._specs(schema, gcol, aggs)
+        s = gpu_scan.GpuScan(spec)
+        s.feed_blocks_host(data, offsets, nb, total)
+        got = s.group_aggregate()
+        s.close()
+        osc = y.orcl_schema_from(schema)
+        want = y.orcl_group(data, offsets, nb, osc, ospec, gcol)
+        tg.check_groups(got, want, aggs)

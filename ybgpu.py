@@ -77,6 +77,7 @@ class ScanSpec(C.Structure):
         ("upper_bound", C.POINTER(C.c_uint8)), ("upper_bound_len", C.c_uint64),
         ("emit_rows", C.c_int32),
         ("row_limit", C.c_uint64),
+        ("group_col", C.c_int32),  # 0 = none, else 1 + value column index
     ]
 
 
@@ -481,6 +482,97 @@ def decode_batch_rows(schema, n_rows, sort_key, key_datums, datums,
         rows.append((sort_key[r], tuple(kd), tuple(vals)))
     rows.sort(key=lambda t: t[0])
     return [(kd, vals) for (_, kd, vals) in rows]
+
+
+def _decode_groups(schema, group_col, n, keys, vals, cnts, key_bytes,
+                   num_aggs, aggs):
+    """Decode group arrays into {key: (vals...)}; key None for the NULL
+    group; doubles decoded from bit patterns."""
+    import struct
+    out = {}
+    is_str = schema.value_cols[group_col].dtype == T_STRING
+    for g in range(n):
+        kv = keys[g]
+        if kv == (1 << 64) - 1:
+            k = None
+        elif is_str:
+            off, ln = kv & ((1 << 40) - 1), kv >> 40
+            k = bytes(key_bytes[off:off + ln])
+        else:
+            k = kv
+        row = []
+        for a in range(num_aggs):
+            c = cnts[g * MAX_AGGS + a]
+            v = vals[g * MAX_AGGS + a]
+            if c == 0:
+                row.append(None)
+            elif aggs[a].op == AGG_SUM_DOUBLE:
+                row.append(struct.unpack("<d", struct.pack("<q", v))[0])
+            else:
+                row.append(v)
+        out[k] = tuple(row)
+    return out
+
+
+def sim_group(spec, data, offsets, n_blocks, cap=1 << 20,
+              key_bytes_cap=1 << 24):
+    """Host-simulator GROUP BY — TEST INFRASTRUCTURE."""
+    lib = product()
+    f = _sig(lib, "ybg_sim_group", C.c_int,
+             [C.POINTER(ScanSpec), C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
+              C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(C.c_int64),
+              C.POINTER(C.c_uint64), C.POINTER(C.c_uint8), C.c_uint64,
+              C.c_uint64, C.POINTER(C.c_uint64)])
+    keys = (C.c_uint64 * cap)()
+    vals = (C.c_int64 * (cap * MAX_AGGS))()
+    cnts = (C.c_uint64 * (cap * MAX_AGGS))()
+    kb = (C.c_uint8 * key_bytes_cap)()
+    n = C.c_uint64()
+    rc = f(C.byref(spec), data, offsets, n_blocks, keys, vals, cnts, kb,
+           key_bytes_cap, cap, C.byref(n))
+    if rc != 0:
+        raise RuntimeError(f"ybg_sim_group rc={rc}")
+    return _decode_groups(spec.schema, spec.group_col - 1, n.value, keys,
+                          vals, cnts, kb, spec.num_aggs, spec.aggs)
+
+
+def orcl_group(data, offsets, n_blocks, schema, spec, group_col,
+               kv_format=ENC_THREE_SHARED_PARTS, cap=1 << 20,
+               key_bytes_cap=1 << 24, num_aggs=None, aggs=None):
+    """Oracle GROUP BY — TEST INFRASTRUCTURE."""
+    lib = oracle()
+    f = _sig(lib, "orcl_group_scan", C.c_int,
+             [C.POINTER(C.POINTER(C.c_uint8)), C.POINTER(C.c_size_t),
+              C.c_size_t, C.c_int, C.POINTER(OrclSchema),
+              C.POINTER(OrclScanSpec), C.c_int, C.POINTER(C.c_uint64),
+              C.POINTER(C.c_int64), C.POINTER(C.c_uint64),
+              C.POINTER(C.c_uint8), C.c_size_t, C.c_size_t,
+              C.POINTER(C.c_size_t)])
+    base = C.cast(data, C.c_void_p).value
+    blocks = (C.POINTER(C.c_uint8) * n_blocks)()
+    sizes = (C.c_size_t * n_blocks)()
+    for i in range(n_blocks):
+        blocks[i] = C.cast(base + offsets[i], C.POINTER(C.c_uint8))
+        sizes[i] = offsets[i + 1] - offsets[i]
+    keys = (C.c_uint64 * cap)()
+    vals = (C.c_int64 * (cap * MAX_AGGS))()
+    cnts = (C.c_uint64 * (cap * MAX_AGGS))()
+    kb = (C.c_uint8 * key_bytes_cap)()
+    n = C.c_size_t()
+    osc = schema
+    rc = f(blocks, sizes, n_blocks, kv_format, C.byref(osc), C.byref(spec),
+           group_col, keys, vals, cnts, kb, key_bytes_cap, cap, C.byref(n))
+    if rc != 0:
+        raise RuntimeError(f"orcl_group_scan rc={rc}")
+
+    class _S:
+        pass
+
+    # adapt orcl schema for the decoder
+    sch = _S()
+    sch.value_cols = osc.value_cols
+    return _decode_groups(sch, group_col, n.value, keys, vals, cnts, kb,
+                          spec.num_aggs, spec.aggs)
 
 
 def orcl_schema_from(schema):

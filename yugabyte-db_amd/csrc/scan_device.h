@@ -36,6 +36,71 @@ static inline unsigned long long atomicAdd(unsigned long long* p,
   *p += v;
   return old;
 }
+static inline unsigned atomicCAS(unsigned* p, unsigned cmp, unsigned val) {
+  unsigned old = *p;
+  if (old == cmp) *p = val;
+  return old;
+}
+static inline double ybg_atomic_add_f64(double* p, double v) {
+  double old = *p;
+  *p += v;
+  return old;
+}
+static inline long long ybg_atomic_min_i64(long long* p, long long v) {
+  long long old = *p;
+  if (v < old) *p = v;
+  return old;
+}
+static inline long long ybg_atomic_max_i64(long long* p, long long v) {
+  long long old = *p;
+  if (v > old) *p = v;
+  return old;
+}
+static inline unsigned long long ybg_atomic_exch_u64(unsigned long long* p,
+                                                     unsigned long long v) {
+  unsigned long long old = *p;
+  *p = v;
+  return old;
+}
+static inline unsigned long long ybg_atomic_load_u64(unsigned long long* p) {
+  return *p;
+}
+static inline unsigned ybg_atomic_load_u32(unsigned* p) { return *p; }
+static inline void ybg_atomic_store_rel_u32(unsigned* p, unsigned v) {
+  *p = v;
+}
+static inline unsigned ybg_atomic_load_acq_u32(unsigned* p) { return *p; }
+#else
+// device wrappers (agent-scope L2 atomics; no L1 staleness)
+__device__ __forceinline__ double ybg_atomic_add_f64(double* p, double v) {
+  return atomicAdd(p, v);
+}
+__device__ __forceinline__ long long ybg_atomic_min_i64(long long* p,
+                                                        long long v) {
+  return atomicMin(p, v);
+}
+__device__ __forceinline__ long long ybg_atomic_max_i64(long long* p,
+                                                        long long v) {
+  return atomicMax(p, v);
+}
+__device__ __forceinline__ unsigned long long ybg_atomic_exch_u64(
+    unsigned long long* p, unsigned long long v) {
+  return atomicExch(p, v);
+}
+__device__ __forceinline__ unsigned long long ybg_atomic_load_u64(
+    unsigned long long* p) {
+  return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+__device__ __forceinline__ unsigned ybg_atomic_load_u32(unsigned* p) {
+  return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+__device__ __forceinline__ void ybg_atomic_store_rel_u32(unsigned* p,
+                                                         unsigned v) {
+  __hip_atomic_store(p, v, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+}
+__device__ __forceinline__ unsigned ybg_atomic_load_acq_u32(unsigned* p) {
+  return __hip_atomic_load(p, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT);
+}
 #endif
 
 namespace ybgdev {
@@ -87,6 +152,7 @@ struct DevSpec {
                              // string key columns (0 = variable)
   int32_t need_rowkey;       // bounds / key predicates / row emission need
                              // the finalized row's key bytes (rk_save)
+  int32_t group_col;         // value column to GROUP BY, -1 = none
 };
 
 struct Interval {
@@ -831,6 +897,10 @@ struct RowCtxT {
   uint32_t* emit_lens;    // [nc]: string length (numeric: 0)
   uint32_t emit_null;     // bit i: column i NULL
   uint32_t emit_str;      // bit i: column i holds a string address
+  // GROUP BY operand (captured when sp.group_col matches)
+  uint64_t grp_datum;     // numeric datum / global string address
+  uint32_t grp_len;       // string length (0 numeric)
+  bool grp_null;
 };
 
 template <int NA>
@@ -843,6 +913,8 @@ DEV void row_reset(RowCtxT<NA>* rc, const DevSpec& sp) {
   rc->agg_null = 0xffffffffu;
   rc->emit_null = 0xffffffffu;
   rc->emit_str = 0;
+  rc->grp_null = true;
+  rc->grp_len = 0;
 }
 
 // Value-column predicate compare (pgsql_operation.cc:602-668 typed-compare
@@ -889,6 +961,11 @@ template <int NA>
 DEV void eval_col(const DevSpec& sp, const uint8_t* aux, RowCtxT<NA>* rc,
                   int col, bool is_null, uint64_t datum, const uint8_t* sptr,
                   uint32_t slen) {
+  if (col == sp.group_col) {
+    rc->grp_null = is_null;
+    rc->grp_datum = sptr ? (uint64_t)(uintptr_t)sptr : datum;
+    rc->grp_len = sptr ? slen : 0;
+  }
   if (rc->emit_datums) {
     rc->emit_null = (rc->emit_null & ~(1u << col)) | ((uint32_t)is_null << col);
     if (!is_null) {
@@ -1484,6 +1561,149 @@ struct HeadOut {
   uint64_t scanned, matched;
 };
 
+// ---------------------------------------------------------------------------
+// GROUP BY partial aggregates (config #5). Open-addressing hash table in
+// device memory: state[] 0=empty / 2=claiming / 1=ready, gkey[] the group
+// key (numeric datum, or for string group columns (len<<40)|offset into the
+// immutable block data — exemplar bytes are compared, never copied). Probe
+// loops always complete an iteration before retrying a claiming slot so
+// divergent lanes reconverge (no intra-wave spin deadlock). Integer
+// aggregate updates are atomic adds/min/max (exact, order-independent);
+// grouped double SUM uses atomic f64 adds (order nondeterministic,
+// documented).
+// ---------------------------------------------------------------------------
+struct GroupCtx {
+  unsigned long long* gkey;  // [cap + 1]; slot cap = the NULL-key group
+  unsigned* state;           // [cap + 1]
+  long long* vals;           // [(cap + 1) * YBG_MAX_AGGS]
+  unsigned long long* cnts;  // [(cap + 1) * YBG_MAX_AGGS]
+  uint64_t cap;              // power of two
+  unsigned long long* overflow;
+  const uint8_t* data;       // block data base (string exemplars)
+};
+
+DEV uint64_t grp_hash_bytes(const uint8_t* p, uint32_t len) {
+  uint64_t h = 1469598103934665603ull;
+  for (uint32_t i = 0; i < len; ++i) {
+    h ^= p[i];
+    h *= 1099511628211ull;
+  }
+  return h;
+}
+
+DEV uint64_t grp_mix(uint64_t x) {
+  x ^= x >> 33;
+  x *= 0xff51afd7ed558ccdull;
+  x ^= x >> 33;
+  x *= 0xc4ceb9fe1a85ec53ull;
+  x ^= x >> 33;
+  return x;
+}
+
+// Returns slot index or ~0ull on table overflow. kv: numeric datum, or for
+// strings (len<<40)|(address - data).
+DEV uint64_t grp_find_or_insert(const GroupCtx& gc, uint64_t kv,
+                                bool is_str) {
+  uint64_t h;
+  const uint8_t* sp1 = nullptr;
+  uint32_t sl = 0;
+  if (is_str) {
+    sl = (uint32_t)(kv >> 40);
+    sp1 = gc.data + (kv & ((1ull << 40) - 1));
+    h = grp_hash_bytes(sp1, sl);
+  } else {
+    h = grp_mix(kv);
+  }
+  uint64_t mask = gc.cap - 1;
+  uint64_t probe = 0;
+  for (uint64_t iters = 0; probe < gc.cap && iters < gc.cap * 64; ++iters) {
+    uint64_t i = (h + probe) & mask;
+    unsigned st = ybg_atomic_load_u32(&gc.state[i]);
+    if (st == 0) {
+      st = atomicCAS(&gc.state[i], 0u, 2u);
+      if (st == 0) {
+        ybg_atomic_exch_u64(&gc.gkey[i], kv);
+        ybg_atomic_store_rel_u32(&gc.state[i], 1u);
+        return i;
+      }
+    }
+    if (st == 1) {
+      st = ybg_atomic_load_acq_u32(&gc.state[i]);
+      uint64_t k2 = ybg_atomic_load_u64(&gc.gkey[i]);
+      bool match;
+      if (is_str) {
+        uint32_t l2 = (uint32_t)(k2 >> 40);
+        match = (l2 == sl);
+        if (match) {
+          const uint8_t* p2 = gc.data + (k2 & ((1ull << 40) - 1));
+          for (uint32_t b = 0; b < sl; ++b)
+            if (p2[b] != sp1[b]) { match = false; break; }
+        }
+      } else {
+        match = (k2 == kv);
+      }
+      if (match) return i;
+      ++probe;
+    }
+    // st == 2: another lane is publishing this slot; retry same probe
+  }
+  return ~0ull;
+}
+
+template <int NA>
+DEV void group_accum(const DevSpec& sp, const GroupCtx& gc,
+                     const RowCtxT<NA>& rc) {
+  uint64_t slot;
+  if (rc.grp_null) {
+    slot = gc.cap;  // the NULL-key group
+    if (ybg_atomic_load_u32(&gc.state[slot]) != 1)
+      ybg_atomic_store_rel_u32(&gc.state[slot], 1u);
+  } else {
+    bool is_str = rc.grp_len != 0;
+    uint64_t kv = is_str ? (((uint64_t)rc.grp_len << 40) |
+                            (rc.grp_datum - (uint64_t)(uintptr_t)gc.data))
+                         : rc.grp_datum;
+    slot = grp_find_or_insert(gc, kv, is_str);
+    if (slot == ~0ull) {
+      atomicAdd(gc.overflow, 1ull);
+      return;
+    }
+  }
+  long long* v = gc.vals + slot * YBG_MAX_AGGS;
+  unsigned long long* c = gc.cnts + slot * YBG_MAX_AGGS;
+#pragma unroll
+  for (int g = 0; g < NA; ++g) {
+    if (g >= sp.num_aggs) continue;
+    int op = sp.aggs[g].op;
+    bool isnull =
+        (op == YBG_AGG_COUNT_STAR) ? false : ((rc.agg_null >> g) & 1);
+    if (isnull) continue;
+    uint64_t d = rc.agg_datum[g];
+    switch (op) {
+      case YBG_AGG_COUNT_STAR:
+      case YBG_AGG_COUNT:
+        atomicAdd((unsigned long long*)&v[g], 1ull);
+        break;
+      case YBG_AGG_SUM_INT64:
+        atomicAdd((unsigned long long*)&v[g], d);
+        break;
+      case YBG_AGG_SUM_DOUBLE:
+        ybg_atomic_add_f64((double*)&v[g],
+                           __longlong_as_double((long long)d));
+        break;
+      case YBG_AGG_MIN_INT64:
+        ybg_atomic_min_i64(&v[g], (long long)d);
+        break;
+      case YBG_AGG_MAX_INT64:
+        ybg_atomic_max_i64(&v[g], (long long)d);
+        break;
+      default:
+        break;  // MIN/MAX double unsupported in group mode (open() rejects)
+    }
+    atomicAdd(&c[g], 1ull);
+  }
+}
+
 // Write one materialized row (PgTableRow analog — dockv/pg_row.h:91-179).
 // rk bytes = the finalized row's key (rk_save), used for key-column datums.
 template <int NA>
@@ -1543,7 +1763,7 @@ DEV void emit_row(const DevSpec& sp, EmitCtx* ec, const RowCtxT<NA>& rc,
   ec->null_masks[slot] = rc.emit_null;
 }
 
-template <int NA, bool EMIT = false>
+template <int NA, bool EMIT = false, bool GROUP = false>
 DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
                            const uint64_t* block_offsets, const Interval* ivs,
                            uint64_t n_ivs, uint64_t j, const uint8_t* aux,
@@ -1553,7 +1773,9 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
                            HeadOut<NA>* ho, bool* walked_next_out,
                            EmitCtx* ec = nullptr,
                            uint64_t* emit_datums = nullptr,
-                           uint32_t* emit_lens = nullptr) {
+                           uint32_t* emit_lens = nullptr,
+                           const GroupCtx* gc = nullptr,
+                           const uint32_t* head_flags = nullptr) {
   Interval iv = ivs[j];
   const uint8_t* blk = data + block_offsets[iv.block];
   const uint8_t* p = blk + iv.start;
@@ -1647,6 +1869,8 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
                     in_head ? ho->cnt : agg_cnt);
             if (EMIT && (!in_head || ec->head_consumed[j] == 0))
               emit_row(sp, ec, rc, rkp, rk_len, row_sort_key);
+            if (GROUP && (!in_head || head_flags[j] == 0))
+              group_accum(sp, *gc, rc);
           }
         }
         in_head = false;
@@ -1680,6 +1904,8 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
       acc_row(sp, rc, in_head ? ho->val : agg_val, in_head ? ho->cnt : agg_cnt);
       if (EMIT && (!in_head || ec->head_consumed[j] == 0))
         emit_row(sp, ec, rc, rkp_end, rk_len, row_sort_key);
+      if (GROUP && (!in_head || head_flags[j] == 0))
+        group_accum(sp, *gc, rc);
     }
   }
   *walked_next_out = walked_next;
@@ -1782,6 +2008,7 @@ inline void build_dev_spec(const ybg_scan_spec_t* spec, DevSpec* dp,
     d.aggs[i].op = spec->aggs[i].op;
     d.aggs[i].col = spec->aggs[i].col;
   }
+  d.group_col = spec->group_col > 0 ? spec->group_col - 1 : -1;
   d.lower_len = (uint32_t)spec->lower_bound_len;
   d.lower_off = put(spec->lower_bound, spec->lower_bound_len);
   d.upper_len = (uint32_t)spec->upper_bound_len;

@@ -210,6 +210,98 @@ __global__ __launch_bounds__(kEmitThreads) void k_emit(
 }
 
 // ---------------------------------------------------------------------------
+// GROUP BY kernels: init table, grouped scan (after a flags pre-pass),
+// export compacted groups (string exemplar bytes copied to a device heap).
+// ---------------------------------------------------------------------------
+
+__global__ void k_group_init(DevSpec sp, GroupCtx gc) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  uint64_t n = gc.cap + 1;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    gc.state[i] = 0;
+    gc.gkey[i] = 0;
+    for (int g = 0; g < sp.num_aggs; ++g) {
+      long long init = 0;
+      if (sp.aggs[g].op == YBG_AGG_MIN_INT64) init = 0x7fffffffffffffffll;
+      else if (sp.aggs[g].op == YBG_AGG_MAX_INT64)
+        init = (long long)0x8000000000000000ll;
+      gc.vals[i * YBG_MAX_AGGS + g] = init;
+      gc.cnts[i * YBG_MAX_AGGS + g] = 0;
+    }
+  }
+}
+
+__global__ __launch_bounds__(kEmitThreads) void k_group(
+    DevSpec sp, const uint8_t* __restrict__ data,
+    const uint64_t* __restrict__ block_offsets,
+    const Interval* __restrict__ ivs, uint64_t n_ivs,
+    const uint8_t* __restrict__ aux, uint8_t* __restrict__ rk_save_buf,
+    GroupCtx gc, const uint32_t* __restrict__ head_flags,
+    unsigned long long* __restrict__ err_counter) {
+  __shared__ uint8_t key_scratch[kEmitThreads * kKeyCap];
+  uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
+  const uint32_t gtid = blockIdx.x * kEmitThreads + threadIdx.x;
+  uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
+  const uint64_t span = (uint64_t)gridDim.x * kEmitThreads;
+  uint64_t entries = 0, scanned = 0, matched = 0;
+  uint64_t agg_val[4] = {0}, agg_cnt[4] = {0};
+  for (uint64_t j = gtid; j < n_ivs; j += span) {
+    HeadOut<4> ho;
+    bool wn = false;
+    if (!scan_one_interval<4, false, true>(
+            sp, data, block_offsets, ivs, n_ivs, j, aux, key, rk_save,
+            &entries, &scanned, &matched, agg_val, agg_cnt, &ho, &wn,
+            nullptr, nullptr, nullptr, &gc, head_flags)) {
+      atomicAdd(err_counter, 1ull);
+    }
+  }
+}
+
+__global__ void k_group_export(DevSpec sp, GroupCtx gc,
+                               uint64_t* __restrict__ out_keys,
+                               long long* __restrict__ out_vals,
+                               unsigned long long* __restrict__ out_cnts,
+                               uint8_t* __restrict__ out_bytes,
+                               uint64_t bytes_cap, uint64_t out_cap,
+                               unsigned long long* __restrict__ counters) {
+  // counters[0] = groups, counters[1] = bytes, counters[2] = overflow
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  uint64_t n = gc.cap + 1;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  bool grp_is_str = sp.group_col >= 0 &&
+                    sp.cols[sp.group_col].dtype == YBG_T_STRING;
+  for (; i < n; i += stride) {
+    if (gc.state[i] != 1) continue;
+    unsigned long long slot = atomicAdd(&counters[0], 1ull);
+    if (slot >= out_cap) {
+      atomicAdd(&counters[2], 1ull);
+      continue;
+    }
+    uint64_t kv = gc.gkey[i];
+    if (i == gc.cap) {
+      kv = ~0ull;  // the NULL-key group marker
+    } else if (grp_is_str) {
+      uint32_t len = (uint32_t)(kv >> 40);
+      unsigned long long off = atomicAdd(&counters[1], len);
+      if (off + len > bytes_cap) {
+        atomicAdd(&counters[2], 1ull);
+        kv = 0;
+      } else {
+        const uint8_t* src = gc.data + (kv & ((1ull << 40) - 1));
+        for (uint32_t b = 0; b < len; ++b) out_bytes[off + b] = src[b];
+        kv = ((uint64_t)len << 40) | off;
+      }
+    }
+    out_keys[slot] = kv;
+    for (int g = 0; g < sp.num_aggs; ++g) {
+      out_vals[slot * YBG_MAX_AGGS + g] = gc.vals[i * YBG_MAX_AGGS + g];
+      out_cnts[slot * YBG_MAX_AGGS + g] = gc.cnts[i * YBG_MAX_AGGS + g];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Final reduction: single workgroup, fixed order => deterministic
 // ---------------------------------------------------------------------------
 
@@ -345,6 +437,14 @@ struct ybg_scan {
   std::vector<uint64_t> h_sort, h_key, h_dat;
   std::vector<uint32_t> h_null;
   std::vector<uint8_t> h_varlen;
+  // group-by buffers
+  GroupCtx gc = {};
+  uint64_t group_cap = 0;
+  uint64_t* d_gk_out = nullptr;
+  long long* d_gv_out = nullptr;
+  unsigned long long* d_gc_out = nullptr;
+  uint8_t* d_gb_out = nullptr;
+  unsigned long long* d_g_counters = nullptr;
 };
 
 extern "C" {
@@ -660,6 +760,99 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
   return 0;
 }
 
+int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
+                                int64_t* vals, uint64_t* cnts,
+                                uint8_t* key_bytes, uint64_t key_bytes_cap,
+                                uint64_t cap, uint64_t* n_groups) {
+  if (!s->d_data) return set_err(4, "feed_blocks not called");
+  if (s->dspec.group_col < 0)
+    return set_err(9, "spec.group_col not set");
+  for (int g = 0; g < s->dspec.num_aggs; ++g) {
+    int op = s->dspec.aggs[g].op;
+    if (op == YBG_AGG_MIN_DOUBLE || op == YBG_AGG_MAX_DOUBLE)
+      return set_err(9, "grouped MIN/MAX over double not supported yet");
+  }
+  if (!s->gc.gkey) {
+    s->group_cap = 1ull << 20;  // 1M groups
+    HIP_TRY(hipMalloc(&s->gc.gkey,
+                      (s->group_cap + 1) * sizeof(unsigned long long)));
+    HIP_TRY(hipMalloc(&s->gc.state, (s->group_cap + 1) * sizeof(unsigned)));
+    HIP_TRY(hipMalloc(&s->gc.vals, (s->group_cap + 1) * YBG_MAX_AGGS * 8));
+    HIP_TRY(hipMalloc(&s->gc.cnts, (s->group_cap + 1) * YBG_MAX_AGGS * 8));
+    HIP_TRY(hipMalloc(&s->gc.overflow, sizeof(unsigned long long)));
+    HIP_TRY(hipMalloc(&s->d_g_counters, 3 * sizeof(unsigned long long)));
+    s->gc.cap = s->group_cap;
+    s->gc.data = s->d_data;
+    if (!s->d_flags_all)
+      HIP_TRY(hipMalloc(&s->d_flags_all, s->n_ivs * sizeof(uint32_t)));
+  }
+  HIP_TRY(hipMemsetAsync(s->gc.overflow, 0, 8, s->stream));
+  HIP_TRY(hipMemsetAsync(s->d_g_counters, 0, 24, s->stream));
+  HIP_TRY(hipMemsetAsync(s->d_flags_all, 0, s->n_ivs * sizeof(uint32_t),
+                         s->stream));
+  HIP_TRY(hipMemsetAsync(s->d_heads, 0,
+                         s->n_heads * kHeadStride * sizeof(uint64_t),
+                         s->stream));
+  hipLaunchKernelGGL(k_group_init, dim3(512), dim3(256), 0, s->stream,
+                     s->dspec, s->gc);
+  // flags pre-pass resolves head-row ownership for every interval
+  auto flags_kernel2 = k_scan<2, 4>;
+  hipLaunchKernelGGL(flags_kernel2, dim3(s->grid), dim3(kThreads), 0,
+                     s->stream, s->dspec, s->d_data, s->d_offsets, s->d_ivs,
+                     s->n_ivs, s->d_aux, s->d_rk_save, s->d_partials,
+                     s->d_heads, s->d_flags_all, 1);
+  int egrid = (int)std::min<uint64_t>(
+      (s->n_ivs + kEmitThreads - 1) / kEmitThreads, 8192);
+  uint64_t need_rk = (uint64_t)egrid * kEmitThreads * kKeyCap;
+  uint64_t have_rk = (uint64_t)s->grid * kThreads * kKeyCap;
+  uint8_t* rk_area = s->d_rk_save;
+  uint8_t* rk_extra = nullptr;
+  if (need_rk > have_rk) {
+    HIP_TRY(hipMalloc(&rk_extra, need_rk));
+    rk_area = rk_extra;
+  }
+  unsigned long long* err_ctr = s->d_g_counters + 2;  // reuse overflow slot?
+  // use gc.overflow for decode errors separate from export overflow
+  hipLaunchKernelGGL(k_group, dim3(egrid), dim3(kEmitThreads), 0, s->stream,
+                     s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
+                     s->d_aux, rk_area, s->gc, s->d_flags_all, s->gc.overflow);
+  (void)err_ctr;
+  // export buffers sized to caller caps
+  if (!s->d_gk_out) {
+    HIP_TRY(hipMalloc(&s->d_gk_out, cap * 8));
+    HIP_TRY(hipMalloc(&s->d_gv_out, cap * YBG_MAX_AGGS * 8));
+    HIP_TRY(hipMalloc(&s->d_gc_out, cap * YBG_MAX_AGGS * 8));
+    HIP_TRY(hipMalloc(&s->d_gb_out, key_bytes_cap ? key_bytes_cap : 1));
+  }
+  hipLaunchKernelGGL(k_group_export, dim3(512), dim3(256), 0, s->stream,
+                     s->dspec, s->gc, s->d_gk_out, s->d_gv_out, s->d_gc_out,
+                     s->d_gb_out, key_bytes_cap, cap, s->d_g_counters);
+  unsigned long long ctr[3];
+  unsigned long long errs = 0;
+  HIP_TRY(hipMemcpyAsync(ctr, s->d_g_counters, 24, hipMemcpyDeviceToHost,
+                         s->stream));
+  HIP_TRY(hipMemcpyAsync(&errs, s->gc.overflow, 8, hipMemcpyDeviceToHost,
+                         s->stream));
+  HIP_TRY(hipStreamSynchronize(s->stream));
+  if (rk_extra) HIP_WARN(hipFree(rk_extra));
+  if (errs) return set_err(6, "corrupt entries or group table overflow");
+  if (ctr[2]) return set_err(8, "group output capacity exceeded");
+  uint64_t ng = ctr[0];
+  *n_groups = ng;
+  if (ng) {
+    HIP_TRY(hipMemcpy(keys, s->d_gk_out, ng * 8, hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(vals, s->d_gv_out, ng * YBG_MAX_AGGS * 8,
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(cnts, s->d_gc_out, ng * YBG_MAX_AGGS * 8,
+                      hipMemcpyDeviceToHost));
+  }
+  if (ctr[1] && key_bytes)
+    HIP_TRY(hipMemcpy(key_bytes, s->d_gb_out,
+                      std::min<uint64_t>(ctr[1], key_bytes_cap),
+                      hipMemcpyDeviceToHost));
+  return 0;
+}
+
 int yb_gpu_scan_paging_state(ybg_scan_t* s, uint8_t* key_out, size_t cap,
                              size_t* len_out) {
   (void)s;
@@ -692,6 +885,16 @@ int yb_gpu_scan_close(ybg_scan_t* s) {
   if (s->d_em_null) HIP_WARN(hipFree(s->d_em_null));
   if (s->d_em_varlen) HIP_WARN(hipFree(s->d_em_varlen));
   if (s->d_em_counters) HIP_WARN(hipFree(s->d_em_counters));
+  if (s->gc.gkey) HIP_WARN(hipFree(s->gc.gkey));
+  if (s->gc.state) HIP_WARN(hipFree(s->gc.state));
+  if (s->gc.vals) HIP_WARN(hipFree(s->gc.vals));
+  if (s->gc.cnts) HIP_WARN(hipFree(s->gc.cnts));
+  if (s->gc.overflow) HIP_WARN(hipFree(s->gc.overflow));
+  if (s->d_g_counters) HIP_WARN(hipFree(s->d_g_counters));
+  if (s->d_gk_out) HIP_WARN(hipFree(s->d_gk_out));
+  if (s->d_gv_out) HIP_WARN(hipFree(s->d_gv_out));
+  if (s->d_gc_out) HIP_WARN(hipFree(s->d_gc_out));
+  if (s->d_gb_out) HIP_WARN(hipFree(s->d_gb_out));
   if (s->ev_start) HIP_WARN(hipEventDestroy(s->ev_start));
   if (s->ev_mid) HIP_WARN(hipEventDestroy(s->ev_mid));
   if (s->ev_end) HIP_WARN(hipEventDestroy(s->ev_end));

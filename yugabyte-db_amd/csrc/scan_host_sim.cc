@@ -172,4 +172,110 @@ int ybg_sim_emit(const ybg_scan_spec_t* spec, const uint8_t* data,
   return 0;
 }
 
+// Sequential GROUP BY simulator: flags pass + grouped pass + export,
+// running the exact device code path.
+int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
+                  const uint64_t* offsets, uint64_t n_blocks,
+                  uint64_t* keys_out, long long* vals_out,
+                  unsigned long long* cnts_out, uint8_t* key_bytes_out,
+                  uint64_t key_bytes_cap, uint64_t cap, uint64_t* n_out) {
+  DevSpec d;
+  std::vector<unsigned char> aux(1 << 20);
+  uint32_t aux_len = 0;
+  build_dev_spec(spec, &d, aux.data(), &aux_len, (uint32_t)aux.size());
+  if (d.group_col < 0) return 9;
+
+  std::vector<Interval> ivs;
+  for (uint64_t b = 0; b < n_blocks; ++b) {
+    uint64_t sz = offsets[b + 1] - offsets[b];
+    if (sz < 8) return 3;
+    const uint8_t* blk = data + offsets[b];
+    uint32_t nr = load_le32_u(blk + sz - 4);
+    if (nr == 0 || (uint64_t)nr * 4 + 4 > sz) return 3;
+    uint32_t restarts_off = (uint32_t)(sz - 4 - (uint64_t)nr * 4);
+    for (uint32_t r = 0; r < nr; ++r) {
+      uint32_t start = load_le32_u(blk + restarts_off + 4ull * r);
+      uint32_t end = (r + 1 < nr)
+                         ? load_le32_u(blk + restarts_off + 4ull * (r + 1))
+                         : restarts_off;
+      ivs.push_back(Interval{(uint32_t)b, start, end});
+    }
+  }
+  uint64_t n_ivs = ivs.size();
+  uint64_t entries = 0, scanned = 0, matched = 0;
+  uint64_t agg_val[YBG_MAX_AGGS] = {0}, agg_cnt[YBG_MAX_AGGS] = {0};
+  std::vector<uint8_t> walked(n_ivs, 0);
+  std::vector<uint32_t> head_flags(n_ivs ? n_ivs : 1, 0);
+  uint8_t key[kKeyCap], rk_save[kKeyCap];
+  for (uint64_t j = 0; j < n_ivs; ++j) {
+    bool wn = false;
+    HeadOut<YBG_MAX_AGGS> ho;
+    if (!scan_one_interval<YBG_MAX_AGGS>(d, data, offsets, ivs.data(), n_ivs,
+                                         j, aux.data(), key, rk_save,
+                                         &entries, &scanned, &matched,
+                                         agg_val, agg_cnt, &ho, &wn))
+      return 6;
+    walked[j] = wn ? 1 : 0;
+  }
+  for (uint64_t j = 1; j < n_ivs; ++j) head_flags[j] = walked[j - 1];
+
+  uint64_t gcap = 1ull << 18;
+  std::vector<unsigned long long> gkey(gcap + 1, 0);
+  std::vector<unsigned> state(gcap + 1, 0);
+  std::vector<long long> vals((gcap + 1) * YBG_MAX_AGGS, 0);
+  std::vector<unsigned long long> cnts((gcap + 1) * YBG_MAX_AGGS, 0);
+  for (uint64_t i = 0; i <= gcap; ++i)
+    for (int g = 0; g < d.num_aggs; ++g) {
+      if (d.aggs[g].op == YBG_AGG_MIN_INT64)
+        vals[i * YBG_MAX_AGGS + g] = 0x7fffffffffffffffll;
+      else if (d.aggs[g].op == YBG_AGG_MAX_INT64)
+        vals[i * YBG_MAX_AGGS + g] = (long long)0x8000000000000000ll;
+    }
+  unsigned long long overflow = 0;
+  GroupCtx gc;
+  gc.gkey = gkey.data();
+  gc.state = state.data();
+  gc.vals = vals.data();
+  gc.cnts = cnts.data();
+  gc.cap = gcap;
+  gc.overflow = &overflow;
+  gc.data = data;
+  for (uint64_t j = 0; j < n_ivs; ++j) {
+    bool wn = false;
+    HeadOut<YBG_MAX_AGGS> ho;
+    if (!scan_one_interval<YBG_MAX_AGGS, false, true>(
+            d, data, offsets, ivs.data(), n_ivs, j, aux.data(), key, rk_save,
+            &entries, &scanned, &matched, agg_val, agg_cnt, &ho, &wn,
+            nullptr, nullptr, nullptr, &gc, head_flags.data()))
+      return 6;
+  }
+  if (overflow) return 8;
+  bool grp_is_str =
+      d.cols[d.group_col].dtype == YBG_T_STRING;
+  uint64_t n = 0, boff = 0;
+  for (uint64_t i = 0; i <= gcap; ++i) {
+    if (state[i] != 1) continue;
+    if (n >= cap) return 8;
+    uint64_t kv = gkey[i];
+    if (i == gcap) {
+      kv = ~0ull;
+    } else if (grp_is_str) {
+      uint32_t len = (uint32_t)(kv >> 40);
+      if (boff + len > key_bytes_cap) return 8;
+      const uint8_t* src = data + (kv & ((1ull << 40) - 1));
+      memcpy(key_bytes_out + boff, src, len);
+      kv = ((uint64_t)len << 40) | boff;
+      boff += len;
+    }
+    keys_out[n] = kv;
+    for (int g = 0; g < YBG_MAX_AGGS; ++g) {
+      vals_out[n * YBG_MAX_AGGS + g] = vals[i * YBG_MAX_AGGS + g];
+      cnts_out[n * YBG_MAX_AGGS + g] = cnts[i * YBG_MAX_AGGS + g];
+    }
+    ++n;
+  }
+  *n_out = n;
+  return 0;
+}
+
 }  // extern "C"
