@@ -232,6 +232,7 @@ __global__ void k_group_init(DevSpec sp, GroupCtx gc) {
   }
 }
 
+template <int NA>
 __global__ __launch_bounds__(kEmitThreads) void k_group(
     DevSpec sp, const uint8_t* __restrict__ data,
     const uint64_t* __restrict__ block_offsets,
@@ -245,11 +246,11 @@ __global__ __launch_bounds__(kEmitThreads) void k_group(
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
   const uint64_t span = (uint64_t)gridDim.x * kEmitThreads;
   uint64_t entries = 0, scanned = 0, matched = 0;
-  uint64_t agg_val[4] = {0}, agg_cnt[4] = {0};
+  uint64_t agg_val[NA] = {0}, agg_cnt[NA] = {0};
   for (uint64_t j = gtid; j < n_ivs; j += span) {
-    HeadOut<4> ho;
+    HeadOut<NA> ho;
     bool wn = false;
-    if (!scan_one_interval<4, false, true>(
+    if (!scan_one_interval<NA, false, true>(
             sp, data, block_offsets, ivs, n_ivs, j, aux, key, rk_save,
             &entries, &scanned, &matched, agg_val, agg_cnt, &ho, &wn,
             nullptr, nullptr, nullptr, &gc, head_flags)) {
@@ -813,9 +814,19 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
   }
   unsigned long long* err_ctr = s->d_g_counters + 2;  // reuse overflow slot?
   // use gc.overflow for decode errors separate from export overflow
-  hipLaunchKernelGGL(k_group, dim3(egrid), dim3(kEmitThreads), 0, s->stream,
-                     s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
-                     s->d_aux, rk_area, s->gc, s->d_flags_all, s->gc.overflow);
+  if (s->dspec.num_aggs <= 4) {
+    auto kg = k_group<4>;
+    hipLaunchKernelGGL(kg, dim3(egrid), dim3(kEmitThreads), 0, s->stream,
+                       s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
+                       s->d_aux, rk_area, s->gc, s->d_flags_all,
+                       s->gc.overflow);
+  } else {
+    auto kg = k_group<8>;
+    hipLaunchKernelGGL(kg, dim3(egrid), dim3(kEmitThreads), 0, s->stream,
+                       s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
+                       s->d_aux, rk_area, s->gc, s->d_flags_all,
+                       s->gc.overflow);
+  }
   (void)err_ctr;
   // export buffers sized to caller caps
   if (!s->d_gk_out) {
