@@ -10,14 +10,8 @@
 
 #ifndef YBG_DEV_QUAL
 #define YBG_DEV_QUAL __device__ __forceinline__
-#define YBG_DEV_COLD __device__ __attribute__((noinline))
-#else
-#define YBG_DEV_COLD YBG_DEV_QUAL
 #endif
 #define DEV YBG_DEV_QUAL
-// Cold paths are kept out of line on device so their register demand does
-// not inflate the hot loop's allocation (spill elimination).
-#define DEVCOLD YBG_DEV_COLD
 
 #ifdef YBG_HOST_SIM
 // host shims for the HIP bit-cast intrinsics
@@ -514,7 +508,7 @@ struct EntryRef {
 // rkb/changed: when rkb > 0, any write that modifies a key byte at a
 // position < rkb sets *changed — used for row-change detection without a
 // separate saved-rowkey compare (valid for fixed-length rowkeys).
-DEVCOLD const uint8_t* decode_entry_ptr(int fmt, const uint8_t* p,
+DEV const uint8_t* decode_entry_ptr(int fmt, const uint8_t* p,
                                     const uint8_t* limit, uint8_t* key,
                                     uint32_t* key_len, uint64_t* last8,
                                     uint32_t rkb, bool* changed,
@@ -789,8 +783,7 @@ DEV const uint8_t* decode_entry(int fmt, Rdr* rdr, const uint8_t* limit,
 // DocKey length (doc_key.h:40-63); key lives in LDS
 // ---------------------------------------------------------------------------
 
-DEVCOLD uint32_t dockey_len(const DevSpec& sp, const uint8_t* p,
-                            uint32_t len) {
+DEV uint32_t dockey_len(const DevSpec& sp, const uint8_t* p, uint32_t len) {
   uint32_t off = 0;
   int col = 0;
   if (sp.has_hash) {
@@ -1074,9 +1067,9 @@ DEV int decode_single_v1(const DevSpec& sp, const uint8_t* base,
 // Packed row decode streaming into predicate/aggregate slots.
 // `body` points at 'z'/'|'.
 template <int NA>
-DEVCOLD bool decode_packed(const DevSpec& sp, const uint8_t* base,
-                           const uint8_t* aux, const uint8_t* body,
-                           uint32_t len, RowCtxT<NA>* rc) {
+DEV bool decode_packed(const DevSpec& sp, const uint8_t* base,
+                       const uint8_t* aux, const uint8_t* body, uint32_t len,
+                       RowCtxT<NA>* rc) {
   uint8_t kind = body[0];
   uint32_t off = 1;
   uint64_t version;
@@ -1215,7 +1208,7 @@ DEV bool key_col_value(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
 // Key-column predicates (evaluated at row finalize; the rowkey bytes come
 // from the saved row key). Key strings are zero-escaped in the key
 // (doc_kv_util.h:101-167): unescape on the fly.
-DEVCOLD bool eval_key_preds(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
+DEV bool eval_key_preds(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
                         const uint8_t* aux) {
 #pragma unroll
   for (int i = 0; i < YBG_MAX_PREDS; ++i) {
