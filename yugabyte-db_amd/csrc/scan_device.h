@@ -705,133 +705,66 @@ DEV const uint8_t* decode_entry(int fmt, Rdr* rdr, const uint8_t* limit,
       }
     }
     if (ok) {
-      // window-parse the size header for the shapes whose fields are all
-      // single bytes: frequent, case 2.1.1 and small case 2.1.2 with last8
-      // reuse (block_builder_internal.h:139-226). Everything else falls
-      // back to the pointer decoder.
       uint32_t value_size = (uint32_t)(e1 >> 2);
       uint64_t inc = (e1 & 2) << 7;
-      uint32_t ns1 = 0, ns2 = 0, hl = 0, sp = 0;
-      int32_t d1 = 0, d2 = 0;
+      uint32_t ns1, ns2, d2 = 0, hl;
       bool fast = false;
-      auto wbyte = [&](uint32_t j) -> uint32_t {
-        return (uint32_t)((w >> (8 * j)) & 0xff);
-      };
-      if (e1 & 1) {  // frequent
-        uint32_t spb = wbyte(e1len);
+      if (e1 & 1) {  // frequent: <sp>
+        uint32_t spb = (uint32_t)((w >> (8 * e1len)) & 0xff);
         if (!(spb & 0x80)) {
-          sp = spb;
           ns1 = 1;
           ns2 = 1;
           hl = e1len + 1;
           fast = true;
+          out->shared = spb;
         }
       } else {
-        uint32_t e2 = wbyte(e1len);
-        if ((e2 & 3) == 1) {  // 2.1.1: reuse, ns1<8, ns2<4, d1=0
-          uint32_t spb = wbyte(e1len + 1);
+        uint32_t e2 = (uint32_t)((w >> (8 * e1len)) & 0xff);
+        if ((e2 & 3) == 1) {  // case 2.1.1: reuse, ns1<8, ns2<4, d1=0
+          uint32_t spb = (uint32_t)((w >> (8 * (e1len + 1))) & 0xff);
           if (!(spb & 0x80)) {
-            d2 = (int32_t)((e2 >> 2) & 1);
+            d2 = (e2 >> 2) & 1;
             ns1 = (e2 >> 3) & 7;
             ns2 = (e2 >> 6) & 3;
-            sp = spb;
             hl = e1len + 2;
             fast = true;
-          }
-        } else if ((e2 & 3) == 3 && (e2 & 4) && e1len + 6 <= 8) {
-          // 2.1.2 with last8 reuse; accept when every present field is one
-          // byte (sizes < 128, deltas |d| <= 63) and fits the first window
-          uint32_t q2 = e1len + 1;
-          uint32_t v = wbyte(q2);
-          bool good = !(v & 0x80);
-          ns1 = v & 0x7f;
-          ++q2;
-          if (good && (e2 & 8)) {  // d1 svarint, 1 byte iff top bits 10/01
-            v = wbyte(q2);
-            uint32_t top = v & 0xC0;
-            if (top == 0x80) d1 = (int32_t)(v & 0x3f);
-            else if (top == 0x40) d1 = -(int32_t)((~v) & 0x3f);
-            else good = false;
-            ++q2;
-          }
-          if (good && (e2 & 16)) {  // ns2 leb, 1 byte
-            v = wbyte(q2);
-            if (v & 0x80) good = false;
-            ns2 = v & 0x7f;
-            ++q2;
-          }
-          if (good && (e2 & 32)) {  // d2 svarint, 1 byte
-            v = wbyte(q2);
-            uint32_t top = v & 0xC0;
-            if (top == 0x80) d2 = (int32_t)(v & 0x3f);
-            else if (top == 0x40) d2 = -(int32_t)((~v) & 0x3f);
-            else good = false;
-            ++q2;
-          }
-          if (good && q2 < 8) {  // shared prefix leb, 1 byte
-            v = wbyte(q2);
-            if (!(v & 0x80)) {
-              sp = v;
-              hl = q2 + 1;
-              fast = true;
-            }
+            out->shared = spb;
           }
         }
       }
-      if (fast && ns1 + ns2 <= 16 && hl + ns1 + ns2 <= 24) {
+      if (fast) {
+        uint32_t sp = out->shared;
         uint64_t prev_len = *key_len;
-        uint64_t prev_ns1 = (uint64_t)ns1 - d1;
         uint64_t prev_ns2 = (uint64_t)ns2 - d2;
-        uint64_t prev_mid_start = (uint64_t)sp + prev_ns1;
-        uint64_t prev_except = prev_mid_start + prev_ns2 + 8;
+        uint64_t prev_except = (uint64_t)sp + ns1 + prev_ns2 + 8;
         uint64_t total = (uint64_t)hl + ns1 + ns2 + value_size;
         if (prev_len >= prev_except && (uint64_t)(limit - p) >= total) {
           uint64_t mid = prev_len - prev_except;
-          uint64_t new_mid_start = sp + ns1;
-          uint64_t new_ns2_start = new_mid_start + mid;
+          uint64_t new_ns2_start = sp + ns1 + mid;
           uint64_t new_len = new_ns2_start + ns2 + 8;
           if (new_len <= kKeyCap) {
-            // middle move (ns1_delta != 0); source always before prev tail
-            if (new_mid_start != prev_mid_start && mid > 0) {
-              if (new_mid_start < prev_mid_start) {
-                for (uint64_t i = 0; i < mid; ++i) {
-                  uint8_t b = key[prev_mid_start + i];
-                  uint64_t dst = new_mid_start + i;
-                  if (dst < rkb && key[dst] != b) *changed = true;
-                  key[dst] = b;
-                }
-              } else {
-                for (uint64_t i = mid; i-- > 0;) {
-                  uint8_t b = key[prev_mid_start + i];
-                  uint64_t dst = new_mid_start + i;
-                  if (dst < rkb && key[dst] != b) *changed = true;
-                  key[dst] = b;
-                }
-              }
-            }
+            // key bytes [hl, hl+ns1+ns2) — within the first 16 window bytes
             uint64_t w2 = rdr->peek8_at(8);
-            uint64_t w3 = (hl + ns1 + ns2 > 16) ? rdr->peek8_at(16) : 0;
-            auto kb = [&](uint32_t j) -> uint8_t {
-              uint64_t src = j < 8 ? w : (j < 16 ? w2 : w3);
-              return (uint8_t)(src >> (8 * (j & 7)));
-            };
             for (uint32_t i = 0; i < ns1; ++i) {
-              uint8_t b = kb(hl + i);
+              uint32_t j = hl + i;
+              uint64_t src = j < 8 ? w : w2;
+              uint8_t b = (uint8_t)(src >> (8 * (j & 7)));
               if (sp + i < rkb && key[sp + i] != b) *changed = true;
               key[sp + i] = b;
             }
             for (uint32_t i = 0; i < ns2; ++i) {
-              uint8_t b = kb(hl + ns1 + i);
-              uint64_t dst = new_ns2_start + i;
-              if (dst < rkb && key[dst] != b) *changed = true;
-              key[dst] = b;
+              uint32_t j = hl + ns1 + i;
+              uint64_t src = j < 8 ? w : w2;
+              uint8_t b = (uint8_t)(src >> (8 * (j & 7)));
+              if (new_ns2_start + i < rkb && key[new_ns2_start + i] != b)
+                *changed = true;
+              key[new_ns2_start + i] = b;
             }
             *last8 += inc;
             *key_len = (uint32_t)new_len;
             rdr->consume(hl + ns1 + ns2);
             out->value = rdr->pos();
             out->value_len = value_size;
-            out->shared = sp;
             return out->value + value_size;
           }
         }
