@@ -957,8 +957,8 @@ DEV bool pred_compare(const DevPred& pr, int dtype, uint64_t datum,
 
 // Column value arrives (from packed row or column update): evaluate the
 // predicates that reference it and stash aggregate operands.
-template <int NA, int NP, typename RC>
-DEV void eval_col(const DevSpec& sp, const uint8_t* aux, RC* rc,
+template <int NA>
+DEV void eval_col(const DevSpec& sp, const uint8_t* aux, RowCtxT<NA>* rc,
                   int col, bool is_null, uint64_t datum, const uint8_t* sptr,
                   uint32_t slen) {
   if (col == sp.group_col) {
@@ -981,7 +981,7 @@ DEV void eval_col(const DevSpec& sp, const uint8_t* aux, RC* rc,
     }
   }
 #pragma unroll
-  for (int i = 0; i < NP; ++i) {
+  for (int i = 0; i < YBG_MAX_PREDS; ++i) {
     if (i < sp.num_preds && !sp.preds[i].is_key_col &&
         sp.preds[i].col == col) {
       bool pass =
@@ -1003,17 +1003,17 @@ DEV void eval_col(const DevSpec& sp, const uint8_t* aux, RC* rc,
 
 // Decode one V1-encoded single value (primitive_value.cc:1066-1125).
 // Returns 0 null/tombstone, 1 applied, -1 error.
-template <int NA, int NP = YBG_MAX_PREDS>
+template <int NA>
 DEV int decode_single_v1(const DevSpec& sp, const uint8_t* base,
                          const uint8_t* aux, int col, const uint8_t* vp,
                          uint32_t vlen, RowCtxT<NA>* rc) {
   if (vlen == 0) {
-    eval_col<NA, NP>(sp, aux, rc, col, true, 0, nullptr, 0);
+    eval_col(sp, aux, rc, col, true, 0, nullptr, 0);
     return 0;
   }
   uint8_t t = vp[0];
   if (t == kTombB || t == kNullLow) {
-    eval_col<NA, NP>(sp, aux, rc, col, true, 0, nullptr, 0);
+    eval_col(sp, aux, rc, col, true, 0, nullptr, 0);
     return 0;
   }
   const DevCol& c = sp.cols[col];
@@ -1059,14 +1059,14 @@ DEV int decode_single_v1(const DevSpec& sp, const uint8_t* base,
     default:
       return -1;
   }
-  eval_col<NA, NP>(sp, aux, rc, col, false, datum, sptr, slen);
+  eval_col(sp, aux, rc, col, false, datum, sptr, slen);
   (void)base;
   return 1;
 }
 
 // Packed row decode streaming into predicate/aggregate slots.
 // `body` points at 'z'/'|'.
-template <int NA, int NP = YBG_MAX_PREDS>
+template <int NA>
 DEV bool decode_packed(const DevSpec& sp, const uint8_t* base,
                        const uint8_t* aux, const uint8_t* body, uint32_t len,
                        RowCtxT<NA>* rc) {
@@ -1101,8 +1101,8 @@ DEV bool decode_packed(const DevSpec& sp, const uint8_t* base,
         end = start + fs;
       }
       if (end < start || end > data_len) return false;
-      int r = decode_single_v1<NA, NP>(sp, base, aux, i, data + start,
-                                       end - start, rc);
+      int r = decode_single_v1(sp, base, aux, i, data + start, end - start,
+                               rc);
       if (r < 0) return false;
     }
     return true;
@@ -1121,7 +1121,7 @@ DEV bool decode_packed(const DevSpec& sp, const uint8_t* base,
     const uint8_t* end = body + len;
     for (int i = 0; i < sp.num_value_cols; ++i) {
       if (null_mask && (null_mask[i >> 3] & (1 << (i & 7)))) {
-        eval_col<NA, NP>(sp, aux, rc, i, true, 0, nullptr, 0);
+        eval_col(sp, aux, rc, i, true, 0, nullptr, 0);
         continue;
       }
       const DevCol& c = sp.cols[i];
@@ -1137,7 +1137,7 @@ DEV bool decode_packed(const DevSpec& sp, const uint8_t* base,
         if (c.dtype == YBG_T_INT8) u = (uint64_t)(int64_t)(int8_t)u;
         else if (c.dtype == YBG_T_INT16) u = (uint64_t)(int64_t)(int16_t)u;
         else if (c.dtype == YBG_T_INT32) u = (uint64_t)(int64_t)(int32_t)u;
-        eval_col<NA, NP>(sp, aux, rc, i, false, u, nullptr, 0);
+        eval_col(sp, aux, rc, i, false, u, nullptr, 0);
         data += c.v2_fixed;
       } else {
         if (data >= end) return false;
@@ -1152,7 +1152,7 @@ DEV bool decode_packed(const DevSpec& sp, const uint8_t* base,
         }
         data += consumed;
         if (data + flen > end) return false;
-        eval_col<NA, NP>(sp, aux, rc, i, false, 0, data, flen);
+        eval_col(sp, aux, rc, i, false, 0, data, flen);
         data += flen;
       }
     }
@@ -1380,7 +1380,7 @@ DEV void agg_combine(const DevSpec& sp, uint64_t* a_val, uint64_t* a_cnt,
 // mask (the dominant shape); anything else falls back to the pointer
 // decoder. Assumes rdr is positioned at the 'kPackedRowV2' byte; consumes
 // the whole body on success.
-template <int NA, int NP = YBG_MAX_PREDS>
+template <int NA>
 DEV bool decode_packed_v2_rdr(const DevSpec& sp, const uint8_t* base,
                               const uint8_t* aux, Rdr* rdr, uint32_t len,
                               RowCtxT<NA>* rc, bool* done) {
@@ -1408,7 +1408,7 @@ DEV bool decode_packed_v2_rdr(const DevSpec& sp, const uint8_t* base,
       if (c.dtype == YBG_T_INT8) u = (uint64_t)(int64_t)(int8_t)u;
       else if (c.dtype == YBG_T_INT16) u = (uint64_t)(int64_t)(int16_t)u;
       else if (c.dtype == YBG_T_INT32) u = (uint64_t)(int64_t)(int32_t)u;
-      eval_col<NA, NP>(sp, aux, rc, i, false, u, nullptr, 0);
+      eval_col(sp, aux, rc, i, false, u, nullptr, 0);
       rdr->consume((uint32_t)c.v2_fixed);
     } else {
       if (rdr->pos() >= end) return false;
@@ -1422,7 +1422,7 @@ DEV bool decode_packed_v2_rdr(const DevSpec& sp, const uint8_t* base,
         rdr->consume(4);
       }
       if (rdr->pos() + flen > end) return false;
-      eval_col<NA, NP>(sp, aux, rc, i, false, 0, rdr->pos(), flen);
+      eval_col(sp, aux, rc, i, false, 0, rdr->pos(), flen);
       rdr->skip(flen);
     }
   }
@@ -1432,7 +1432,7 @@ DEV bool decode_packed_v2_rdr(const DevSpec& sp, const uint8_t* base,
 
 // Visibility + row-state update for one entry. key/rowkey live in LDS.
 // rdr is positioned at the value start. Returns false on corruption.
-template <int NA, int NP = YBG_MAX_PREDS>
+template <int NA>
 DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
                        const uint8_t* aux, const uint8_t* key,
                        uint32_t key_len, const uint8_t* value,
@@ -1480,8 +1480,7 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
         // dominant shape: packed V2, no control fields — register-window
         // decode; falls back when a null mask / big schema version appears
         bool done;
-        if (!decode_packed_v2_rdr<NA, NP>(sp, base, aux, rdr, value_len,
-                                          rc, &done))
+        if (!decode_packed_v2_rdr(sp, base, aux, rdr, value_len, rc, &done))
           return false;
         if (done) {
           rc->found = true;
@@ -1493,8 +1492,7 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
       const uint8_t* body = value + cf;
       uint32_t body_len = value_len - cf;
       if (body_len > 0 && (body[0] == kPackedV1B || body[0] == kPackedV2B)) {
-        if (!decode_packed<NA, NP>(sp, base, aux, body, body_len, rc))
-          return false;
+        if (!decode_packed(sp, base, aux, body, body_len, rc)) return false;
         rc->found = true;  // doc_reader.cc:1894-1900
       }
     }
@@ -1539,8 +1537,8 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
         if (newer) {
           int cf = skip_control(value, value_len);
           if (cf < 0) return false;
-          int r = decode_single_v1<NA, NP>(sp, base, aux, idx, value + cf,
-                                           value_len - cf, rc);
+          int r = decode_single_v1(sp, base, aux, idx, value + cf,
+                                   value_len - cf, rc);
           if (r < 0) return false;
           if (r > 0) rc->found = true;
         }
@@ -1765,8 +1763,7 @@ DEV void emit_row(const DevSpec& sp, EmitCtx* ec, const RowCtxT<NA>& rc,
   ec->null_masks[slot] = rc.emit_null;
 }
 
-template <int NA, bool EMIT = false, bool GROUP = false,
-          int NP = YBG_MAX_PREDS>
+template <int NA, bool EMIT = false, bool GROUP = false>
 DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
                            const uint64_t* block_offsets, const Interval* ivs,
                            uint64_t n_ivs, uint64_t j, const uint8_t* aux,
@@ -1891,11 +1888,8 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
     }
     if (cur_iv == j) *entries += 1;
     else if (cur_iv == j + 1) walked_next = true;
-    if (!process_entry<NA, NP>(sp, data, aux, key, key_len, er.value,
-                               er.value_len, rk_len, &rc, &rdr)) {
-      fail = true;
-      break;
-    }
+    if (!process_entry(sp, data, aux, key, key_len, er.value, er.value_len,
+                       rk_len, &rc, &rdr)) { fail = true; break; }
     rdr.seek(q);
     p = q;
   }

@@ -51,7 +51,7 @@ namespace {
 // Phase-1 scan kernel (aggregate mode): one thread per interval
 // ---------------------------------------------------------------------------
 
-template <int NA, int WPS, int NP = YBG_MAX_PREDS>
+template <int NA, int WPS>
 __global__ __launch_bounds__(kThreads, WPS) void k_scan(
     DevSpec sp, const uint8_t* __restrict__ data,
     const uint64_t* __restrict__ block_offsets,
@@ -77,10 +77,9 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
     HeadOut<NA> ho;
     bool walked_next = false;
     if (active) {
-      if (!scan_one_interval<NA, false, false, NP>(
-              sp, data, block_offsets, ivs, n_ivs, j, aux, key, rk_save,
-              &entries, &scanned, &matched, agg_val, agg_cnt, &ho,
-              &walked_next)) {
+      if (!scan_one_interval<NA>(sp, data, block_offsets, ivs, n_ivs, j, aux,
+                                 key, rk_save, &entries, &scanned, &matched,
+                                 agg_val, agg_cnt, &ho, &walked_next)) {
         errs += 1;
       }
     } else {
@@ -588,7 +587,7 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
   int wps = 4;
   if (const char* e = getenv("YBG_WPS")) {
     long v = atol(e);
-    if (v == 3 || v == 4) wps = (int)v;
+    if (v >= 2 && v <= 6) wps = (int)v;
   }
   auto launch = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(s->grid), dim3(kThreads), 0, s->stream,
@@ -597,20 +596,22 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
                        s->d_cont, 0);
   };
   int na = s->dspec.num_aggs <= 2 ? 2 : (s->dspec.num_aggs <= 4 ? 4 : 8);
-  int np = s->dspec.num_preds <= 4 ? 4 : 8;
-  switch (na * 100 + np * 10 + wps) {
-    case 243: launch(k_scan<2, 3, 4>); break;
-    case 244: launch(k_scan<2, 4, 4>); break;
-    case 283: launch(k_scan<2, 3, 8>); break;
-    case 284: launch(k_scan<2, 4, 8>); break;
-    case 443: launch(k_scan<4, 3, 4>); break;
-    case 444: launch(k_scan<4, 4, 4>); break;
-    case 483: launch(k_scan<4, 3, 8>); break;
-    case 484: launch(k_scan<4, 4, 8>); break;
-    case 843: launch(k_scan<8, 3, 4>); break;
-    case 844: launch(k_scan<8, 4, 4>); break;
-    case 883: launch(k_scan<8, 3, 8>); break;
-    default: launch(k_scan<8, 4, 8>); break;
+  switch (na * 10 + wps) {
+    case 22: launch(k_scan<2, 2>); break;
+    case 23: launch(k_scan<2, 3>); break;
+    case 25: launch(k_scan<2, 5>); break;
+    case 26: launch(k_scan<2, 6>); break;
+    case 42: launch(k_scan<4, 2>); break;
+    case 43: launch(k_scan<4, 3>); break;
+    case 45: launch(k_scan<4, 5>); break;
+    case 46: launch(k_scan<4, 6>); break;
+    case 82: launch(k_scan<8, 2>); break;
+    case 83: launch(k_scan<8, 3>); break;
+    case 85: launch(k_scan<8, 5>); break;
+    case 86: launch(k_scan<8, 6>); break;
+    case 44: launch(k_scan<4, 4>); break;
+    case 84: launch(k_scan<8, 4>); break;
+    default: launch(k_scan<2, 4>); break;
   }
   HIP_TRY(hipEventRecord(s->ev_mid, s->stream));
   hipLaunchKernelGGL(k_reduce, dim3(1), dim3(256), 0, s->stream, s->dspec,
@@ -686,7 +687,7 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
                          s->n_heads * kHeadStride * sizeof(uint64_t),
                          s->stream));
   // flags pre-pass: resolves head-row ownership for EVERY interval
-  auto flags_kernel = k_scan<2, 4, 8>;
+  auto flags_kernel = k_scan<2, 4>;
   hipLaunchKernelGGL(flags_kernel, dim3(s->grid), dim3(kThreads), 0,
                      s->stream, s->dspec, s->d_data, s->d_offsets, s->d_ivs,
                      s->n_ivs, s->d_aux, s->d_rk_save, s->d_partials,
@@ -796,7 +797,7 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
   hipLaunchKernelGGL(k_group_init, dim3(512), dim3(256), 0, s->stream,
                      s->dspec, s->gc);
   // flags pre-pass resolves head-row ownership for every interval
-  auto flags_kernel2 = k_scan<2, 4, 8>;
+  auto flags_kernel2 = k_scan<2, 4>;
   hipLaunchKernelGGL(flags_kernel2, dim3(s->grid), dim3(kThreads), 0,
                      s->stream, s->dspec, s->d_data, s->d_offsets, s->d_ivs,
                      s->n_ivs, s->d_aux, s->d_rk_save, s->d_partials,
