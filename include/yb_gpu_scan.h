@@ -225,6 +225,7 @@ typedef struct {
   const uint64_t *datums;     /* [n_rows * n_value_cols]; string cols: offset
                                  into varlen heap (hi32 = len) */
   const uint32_t *null_masks; /* [n_rows] bit i = value col i NULL */
+  const uint16_t *hashes;     /* [n_rows] kUInt16Hash prefix (hash schemas) */
   const uint8_t *varlen;      /* varlen heap */
   uint64_t varlen_size;
 } ybg_row_batch_t;
@@ -337,6 +338,11 @@ int ybg_generate(const ybg_schema_t *schema, const ybg_gen_params_t *p,
                  uint64_t *total_bytes, uint64_t *n_entries);
 
 void ybg_free(void *p);
+
+/* Encode a DocKey (doc_key.h:40-63) from key-column values (paging-state
+ * serialization). Returns encoded length, 0 if cap too small. */
+size_t ybg_encode_dockey(const ybg_schema_t *schema, const ybg_key_t *key,
+                         uint8_t *out, size_t cap);
 
 #ifdef __cplusplus
 }

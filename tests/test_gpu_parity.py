@@ -181,3 +181,57 @@ def test_gpu_group_by():
         osc = y.orcl_schema_from(schema)
         want = y.orcl_group(data, offsets, nb, osc, ospec, gcol)
         tg.check_groups(got, want, aggs)
+
+
+def test_gpu_paging_resume(cases):
+    """row_limit + paging state: a limited scan returns the first page in
+    key order plus a resumable DocKey; a second scan with that key as the
+    inclusive lower bound returns the rest (pgsql_operation.cc:2796-2806,
+    2908-2922 semantics)."""
+    import ctypes as C
+    from parity_cases import make_spec, make_orcl_spec
+    gpu_scan = _gpu()
+    lib = gpu_scan._lib()
+    lib.yb_host_iter_paging_state.restype = C.c_int
+    lib.yb_host_iter_paging_state.argtypes = [
+        C.c_void_p, C.POINTER(C.c_uint8), C.c_size_t, C.POINTER(C.c_size_t)]
+    case = [c for c in cases if c["name"] == "nulls_packed"][0]
+    read_micros = case["runs"][0][0]
+    sc = case["schema"]
+    nk = sc.num_hash_cols + sc.num_range_cols
+
+    def drain(spec):
+        h = lib.yb_host_iter_open(C.byref(spec), case["data"],
+                                  case["offsets"], case["n_blocks"])
+        assert h
+        kd = (C.c_uint64 * y.MAX_KEYCOLS)()
+        vd = (C.c_uint64 * y.MAX_COLS)()
+        nm = C.c_uint32()
+        vl = C.POINTER(C.c_uint8)()
+        rows = []
+        while lib.yb_host_iter_next(h, kd, vd, C.byref(nm), C.byref(vl)) == 1:
+            vals = tuple(None if (nm.value >> c) & 1 else vd[c]
+                         for c in range(sc.num_value_cols))
+            rows.append((tuple(kd[i] for i in range(nk)), vals))
+        pk = (C.c_uint8 * 64)()
+        pl = C.c_size_t()
+        assert lib.yb_host_iter_paging_state(h, pk, 64, C.byref(pl)) == 0
+        lib.yb_host_iter_close(h)
+        return rows, bytes(pk[:pl.value])
+
+    spec1 = make_spec(case, read_micros, (), ())
+    spec1.row_limit = 1500
+    page1, pkey = drain(spec1)
+    assert len(page1) == 1500 and len(pkey) > 0
+
+    buf = C.create_string_buffer(pkey, len(pkey))
+    spec2 = make_spec(case, read_micros, (), (),
+                      lower=(buf, len(pkey)))
+    page2, pkey2 = drain(spec2)
+    assert pkey2 == b""  # second page unlimited: scan complete
+
+    osc = y.orcl_schema_from(sc)
+    ospec = make_orcl_spec(read_micros, (), ())
+    _, want = y.orcl_scan(case["data"], case["offsets"], case["n_blocks"],
+                          osc, ospec, collect_rows=True)
+    assert page1 + page2 == want

@@ -99,6 +99,7 @@ class RowBatch(C.Structure):
         ("key_datums", C.POINTER(C.c_uint64)),
         ("datums", C.POINTER(C.c_uint64)),
         ("null_masks", C.POINTER(C.c_uint32)),
+        ("hashes", C.POINTER(C.c_uint16)),
         ("varlen", C.POINTER(C.c_uint8)),
         ("varlen_size", C.c_uint64),
     ]

@@ -9,6 +9,7 @@ namespace ybg {
 GpuDocRowwiseIterator::GpuDocRowwiseIterator(const ybg_scan_spec_t& spec)
     : spec_(spec) {
   spec_.emit_rows = 1;
+  limit_ = spec.row_limit;
   open_rc_ = yb_gpu_scan_open(&spec_, &handle_);
 }
 
@@ -47,6 +48,7 @@ int GpuDocRowwiseIterator::PgFetchNext(PgRow* row) {
     if (rc) return -rc;
   }
   if (pos_ >= batch_.n_rows) return 0;
+  if (limit_ && pos_ >= limit_) return 0;  // page exhausted (row_limit)
   uint64_t r = order_[pos_++];
   uint64_t nk = batch_.n_key_cols, nc = batch_.n_value_cols;
   for (uint64_t c = 0; c < nk; ++c)
@@ -65,9 +67,34 @@ int GpuDocRowwiseIterator::Aggregate(ybg_scan_result_t* out) {
   return yb_gpu_scan_aggregate(handle_, out);
 }
 
+// Resumable position (pgsql_operation.cc:2796-2806, 2908-2922): the encoded
+// DocKey of the first undelivered row; a follow-up scan resumes with it as
+// the inclusive lower bound. len = 0 when the scan is complete.
 int GpuDocRowwiseIterator::PagingState(uint8_t* key_out, size_t cap,
                                        size_t* len) {
-  return yb_gpu_scan_paging_state(handle_, key_out, cap, len);
+  *len = 0;
+  if (!batch_ready_ || !limit_ || limit_ >= batch_.n_rows) return 0;
+  uint64_t next = order_[limit_];
+  ybg_key_t k = {};
+  const ybg_schema_t& sc = spec_.schema;
+  int nk = sc.num_hash_cols + sc.num_range_cols;
+  std::vector<std::vector<uint8_t>> strs((size_t)nk);
+  for (int c = 0; c < nk; ++c) {
+    uint64_t d = batch_.key_datums[next * (uint64_t)nk + c];
+    if (sc.key_types[c] == YBG_KT_STRING) {
+      uint64_t off = d & ((1ull << 40) - 1);
+      uint32_t ln = (uint32_t)(d >> 40);
+      strs[c].assign(batch_.varlen + off, batch_.varlen + off + ln);
+      k.strs[c] = strs[c].data();
+      k.str_lens[c] = ln;
+    } else {
+      k.datums[c] = d;
+    }
+  }
+  if (sc.has_hash) k.hash = batch_.hashes[next];
+  size_t n = ybg_encode_dockey(&sc, &k, key_out, cap);
+  *len = n;
+  return 0;
 }
 
 const char* GpuDocRowwiseIterator::LastError() const {
@@ -103,6 +130,12 @@ int yb_host_iter_next(void* h, uint64_t* key_datums, uint64_t* datums,
   *null_mask = row.null_mask;
   *varlen = row.varlen;
   return 1;
+}
+
+int yb_host_iter_paging_state(void* h, uint8_t* key_out, size_t cap,
+                              size_t* len) {
+  return static_cast<ybg::GpuDocRowwiseIterator*>(h)->PagingState(key_out,
+                                                                  cap, len);
 }
 
 void yb_host_iter_close(void* h) {

@@ -54,6 +54,7 @@ class GpuDocRowwiseIterator {
   ybg_row_batch_t batch_ = {};
   std::vector<uint64_t> order_;  // row indices sorted by sort_key
   uint64_t pos_ = 0;
+  uint64_t limit_ = 0;  // rows to deliver this page (0 = all)
   bool batch_ready_ = false;
   int open_rc_ = 0;
 };

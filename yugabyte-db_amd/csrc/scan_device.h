@@ -870,6 +870,7 @@ struct EmitCtx {
   uint64_t* key_datums;  // [row_cap * nk]
   uint64_t* datums;      // [row_cap * nc]; strings: (len<<40)|varlen offset
   uint32_t* null_masks;  // [row_cap]
+  uint16_t* hashes;      // [row_cap] kUInt16Hash prefix (doc_key.h:54)
   uint8_t* varlen;
   uint64_t varlen_cap;
   unsigned long long* row_counter;
@@ -1761,6 +1762,8 @@ DEV void emit_row(const DevSpec& sp, EmitCtx* ec, const RowCtxT<NA>& rc,
     ec->datums[slot * nc + c] = d;
   }
   ec->null_masks[slot] = rc.emit_null;
+  ec->hashes[slot] =
+      sp.has_hash ? (uint16_t)(((uint16_t)rk[1] << 8) | rk[2]) : 0;
 }
 
 template <int NA, bool EMIT = false, bool GROUP = false>

@@ -432,11 +432,13 @@ struct ybg_scan {
   uint64_t* d_em_key = nullptr;
   uint64_t* d_em_dat = nullptr;
   uint32_t* d_em_null = nullptr;
+  uint16_t* d_em_hash = nullptr;
   uint8_t* d_em_varlen = nullptr;
   uint64_t em_varlen_cap = 0;
   unsigned long long* d_em_counters = nullptr;  // rows, varlen, overflow, err
   std::vector<uint64_t> h_sort, h_key, h_dat;
   std::vector<uint32_t> h_null;
+  std::vector<uint16_t> h_hash;
   std::vector<uint8_t> h_varlen;
   // group-by buffers
   GroupCtx gc = {};
@@ -676,6 +678,7 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
     HIP_TRY(hipMalloc(&s->d_em_key, s->emit_row_cap * 8 * (nk ? nk : 1)));
     HIP_TRY(hipMalloc(&s->d_em_dat, s->emit_row_cap * 8 * (nc ? nc : 1)));
     HIP_TRY(hipMalloc(&s->d_em_null, s->emit_row_cap * 4));
+    HIP_TRY(hipMalloc(&s->d_em_hash, s->emit_row_cap * 2));
     HIP_TRY(hipMalloc(&s->d_em_varlen, s->em_varlen_cap));
     HIP_TRY(hipMalloc(&s->d_em_counters, 4 * sizeof(unsigned long long)));
   }
@@ -697,6 +700,7 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
   ec.key_datums = s->d_em_key;
   ec.datums = s->d_em_dat;
   ec.null_masks = s->d_em_null;
+  ec.hashes = s->d_em_hash;
   ec.varlen = s->d_em_varlen;
   ec.varlen_cap = s->em_varlen_cap;
   ec.row_counter = s->d_em_counters + 0;
@@ -734,6 +738,7 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
   s->h_key.resize(n_rows * (nk ? nk : 1));
   s->h_dat.resize(n_rows * (nc ? nc : 1));
   s->h_null.resize(n_rows);
+  s->h_hash.resize(n_rows);
   s->h_varlen.resize(vl ? vl : 1);
   if (n_rows) {
     HIP_TRY(hipMemcpy(s->h_sort.data(), s->d_em_sort, n_rows * 8,
@@ -743,6 +748,8 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
     HIP_TRY(hipMemcpy(s->h_dat.data(), s->d_em_dat, n_rows * 8 * (nc ? nc : 1),
                       hipMemcpyDeviceToHost));
     HIP_TRY(hipMemcpy(s->h_null.data(), s->d_em_null, n_rows * 4,
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(s->h_hash.data(), s->d_em_hash, n_rows * 2,
                       hipMemcpyDeviceToHost));
   }
   if (vl)
@@ -756,6 +763,7 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
   out->key_datums = s->h_key.data();
   out->datums = s->h_dat.data();
   out->null_masks = s->h_null.data();
+  out->hashes = s->h_hash.data();
   out->varlen = s->h_varlen.data();
   out->varlen_size = vl;
   return 0;
@@ -894,6 +902,7 @@ int yb_gpu_scan_close(ybg_scan_t* s) {
   if (s->d_em_key) HIP_WARN(hipFree(s->d_em_key));
   if (s->d_em_dat) HIP_WARN(hipFree(s->d_em_dat));
   if (s->d_em_null) HIP_WARN(hipFree(s->d_em_null));
+  if (s->d_em_hash) HIP_WARN(hipFree(s->d_em_hash));
   if (s->d_em_varlen) HIP_WARN(hipFree(s->d_em_varlen));
   if (s->d_em_counters) HIP_WARN(hipFree(s->d_em_counters));
   if (s->gc.gkey) HIP_WARN(hipFree(s->gc.gkey));

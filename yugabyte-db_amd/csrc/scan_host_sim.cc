@@ -99,6 +99,7 @@ int ybg_sim_emit(const ybg_scan_spec_t* spec, const uint8_t* data,
                  uint64_t* datums, uint32_t* null_masks, uint8_t* varlen,
                  uint64_t varlen_cap, uint64_t* n_rows_out,
                  uint64_t* varlen_out) {
+  std::vector<uint16_t> hashes(row_cap);
   DevSpec d;
   std::vector<unsigned char> aux(1 << 20);
   uint32_t aux_len = 0;
@@ -146,6 +147,7 @@ int ybg_sim_emit(const ybg_scan_spec_t* spec, const uint8_t* data,
   ec.key_datums = key_datums;
   ec.datums = datums;
   ec.null_masks = null_masks;
+  ec.hashes = hashes.data();
   ec.varlen = varlen;
   ec.varlen_cap = varlen_cap;
   ec.row_counter = &row_counter;

@@ -681,4 +681,17 @@ int ybg_generate(const ybg_schema_t* schema, const ybg_gen_params_t* p,
 
 void ybg_free(void* p) { free(p); }
 
+// Encode a DocKey (doc_key.h:40-63) from key-column datums — used by the
+// host adapter to serialize the resumable paging position
+// (pgsql_operation.cc:2796-2806: the next row's key becomes the paging
+// state). Returns encoded length.
+size_t ybg_encode_dockey(const ybg_schema_t* schema, const ybg_key_t* key,
+                         uint8_t* out, size_t cap) {
+  ybg::Buf b;
+  ybg::EncodeDocKey(schema, key, &b);
+  if (b.size() > cap) return 0;
+  memcpy(out, b.data(), b.size());
+  return b.size();
+}
+
 }  // extern "C"
