@@ -47,18 +47,25 @@ using namespace ybgdev;
 
 namespace {
 
+// The spec lives in constant memory: its fields are read throughout the hot
+// loop, and passing it by value made the compiler hold/spill hundreds of
+// SGPRs per wave (PMC showed ~19.6 GB of scratch write traffic per 100M-row
+// dispatch from the spill save/restore around the interval loop).
+__constant__ DevSpec c_spec;
+
 // ---------------------------------------------------------------------------
 // Phase-1 scan kernel (aggregate mode): one thread per interval
 // ---------------------------------------------------------------------------
 
 template <int NA, int WPS>
 __global__ __launch_bounds__(kThreads, WPS) void k_scan(
-    DevSpec sp, const uint8_t* __restrict__ data,
+    const uint8_t* __restrict__ data,
     const uint64_t* __restrict__ block_offsets,
     const Interval* __restrict__ ivs, uint64_t n_ivs,
     const uint8_t* __restrict__ aux, uint8_t* __restrict__ rk_save_buf,
     uint64_t* __restrict__ partials, uint64_t* __restrict__ heads,
     uint32_t* __restrict__ cont_flags, int write_all_flags) {
+  const DevSpec& sp = c_spec;
   __shared__ uint8_t key_scratch[kThreads * kKeyCap];
   __shared__ uint8_t wave_relay[kThreads / 64 + 1];
   uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
@@ -591,9 +598,12 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
     long v = atol(e);
     if (v >= 2 && v <= 6) wps = (int)v;
   }
+  HIP_TRY(hipMemcpyToSymbolAsync(HIP_SYMBOL(c_spec), &s->dspec,
+                                 sizeof(DevSpec), 0, hipMemcpyHostToDevice,
+                                 s->stream));
   auto launch = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(s->grid), dim3(kThreads), 0, s->stream,
-                       s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
+                       s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
                        s->d_aux, s->d_rk_save, s->d_partials, s->d_heads,
                        s->d_cont, 0);
   };
@@ -690,9 +700,12 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
                          s->n_heads * kHeadStride * sizeof(uint64_t),
                          s->stream));
   // flags pre-pass: resolves head-row ownership for EVERY interval
+  HIP_TRY(hipMemcpyToSymbolAsync(HIP_SYMBOL(c_spec), &s->dspec,
+                                 sizeof(DevSpec), 0, hipMemcpyHostToDevice,
+                                 s->stream));
   auto flags_kernel = k_scan<2, 4>;
   hipLaunchKernelGGL(flags_kernel, dim3(s->grid), dim3(kThreads), 0,
-                     s->stream, s->dspec, s->d_data, s->d_offsets, s->d_ivs,
+                     s->stream, s->d_data, s->d_offsets, s->d_ivs,
                      s->n_ivs, s->d_aux, s->d_rk_save, s->d_partials,
                      s->d_heads, s->d_flags_all, 1);
   EmitCtx ec;
@@ -805,9 +818,12 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
   hipLaunchKernelGGL(k_group_init, dim3(512), dim3(256), 0, s->stream,
                      s->dspec, s->gc);
   // flags pre-pass resolves head-row ownership for every interval
+  HIP_TRY(hipMemcpyToSymbolAsync(HIP_SYMBOL(c_spec), &s->dspec,
+                                 sizeof(DevSpec), 0, hipMemcpyHostToDevice,
+                                 s->stream));
   auto flags_kernel2 = k_scan<2, 4>;
   hipLaunchKernelGGL(flags_kernel2, dim3(s->grid), dim3(kThreads), 0,
-                     s->stream, s->dspec, s->d_data, s->d_offsets, s->d_ivs,
+                     s->stream, s->d_data, s->d_offsets, s->d_ivs,
                      s->n_ivs, s->d_aux, s->d_rk_save, s->d_partials,
                      s->d_heads, s->d_flags_all, 1);
   int egrid = (int)std::min<uint64_t>(
