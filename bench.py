@@ -40,6 +40,11 @@ def main():
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--rows", type=int, default=100_000_000,
                     help="total rows across all tablets")
+    ap.add_argument("--workload", choices=["filtersum", "mvcc", "groupby"],
+                    default="filtersum",
+                    help="filtersum = BASELINE configs[1]/[2]; mvcc = "
+                         "configs[3] (5 versions/row, COUNT); groupby = "
+                         "configs[4]-shaped grouped aggregate")
     ap.add_argument("--cpu-sample-seconds", type=float, default=10.0)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
@@ -70,21 +75,38 @@ def main():
         sys.exit(2)
 
     # ---- dataset ---------------------------------------------------------
-    schema = y.make_schema([y.KT_INT64],
-                           [(10 + i, y.T_INT64, 1) for i in range(4)])
+    versions = 1
+    group_col = 0
+    if args.workload == "filtersum":
+        schema = y.make_schema([y.KT_INT64],
+                               [(10 + i, y.T_INT64, 1) for i in range(4)])
+        preds = [y.Pred(0, 0, y.PRED_GT, 1 << 39, None, 0),
+                 y.Pred(0, 1, y.PRED_LT, 3 << 38, None, 0),
+                 y.Pred(0, 2, y.PRED_GE, 1 << 36, None, 0)]
+        aggs = [y.Agg(y.AGG_SUM_INT64, 3), y.Agg(y.AGG_COUNT_STAR, 0)]
+    elif args.workload == "mvcc":
+        # BASELINE configs[3]: 5 versions/row, visibility + COUNT
+        schema = y.make_schema([y.KT_INT64],
+                               [(10 + i, y.T_INT64, 1) for i in range(4)])
+        preds = []
+        aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 0)]
+        versions = 5
+    else:  # groupby — BASELINE configs[4]-shaped (mixed row, grouped)
+        schema = y.make_schema(
+            [y.KT_INT64],
+            [(10, y.T_INT64, 1), (11, y.T_INT64, 1), (12, y.T_DOUBLE, 1)])
+        preds = [y.Pred(0, 1, y.PRED_GT, 1 << 38, None, 0)]
+        aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 1)]
+        group_col = 1  # 1 + value col 0
     n_tablets = 1 if n_gpus == 1 and world <= 1 else 8
     rows_per_tablet = args.rows // n_tablets
     my_tablets = [t for t in range(n_tablets) if t % max(world, 1) == rank]
-
-    preds = [y.Pred(0, 0, y.PRED_GT, 1 << 39, None, 0),
-             y.Pred(0, 1, y.PRED_LT, 3 << 38, None, 0),
-             y.Pred(0, 2, y.PRED_GE, 1 << 36, None, 0)]
-    aggs = [y.Agg(y.AGG_SUM_INT64, 3), y.Agg(y.AGG_COUNT_STAR, 0)]
 
     def make_spec():
         spec = y.ScanSpec()
         spec.schema = schema
         spec.kv_format = y.ENC_THREE_SHARED_PARTS
+        spec.group_col = group_col
         spec.read_time = y.read_time(1_700_000_000_000_000)
         spec.num_preds = len(preds)
         for i, p in enumerate(preds):
@@ -100,7 +122,9 @@ def main():
     first_tablet_data = None
     for t in my_tablets:
         data, offsets, nb, total, ne = y.generate(
-            schema, rows=rows_per_tablet, seed=42 + t)
+            schema, rows=rows_per_tablet, seed=42 + t, versions=versions,
+            ht_base_micros=1_600_000_000_000_000,
+            ht_step_micros=1_000_000 if versions > 1 else 1000)
         total_bytes_local += total
         s = gpu_scan.GpuScan(make_spec())
         s.feed_blocks_host(data, offsets, nb, total)
@@ -123,6 +147,24 @@ def main():
                               else "cpu")
 
     def step():
+        if group_col:
+            # grouped partial aggregates; the cross-tablet merge of the
+            # partial tables happens host-side (SURVEY §8e all-gather +
+            # merge; single-node benchmark merges locally)
+            total_sum = 0
+            total_cnt = 0
+            for s in scans:
+                groups = s.group_aggregate()
+                for k, vals in groups.items():
+                    total_cnt += vals[0] or 0
+                    total_sum += vals[1] or 0
+            if dist is not None:
+                agg_buf[0] = float(total_sum)
+                agg_buf[1] = float(total_cnt)
+                dist.all_reduce(agg_buf)
+                total_sum = int(agg_buf[0].item())
+                total_cnt = int(agg_buf[1].item())
+            return total_sum, total_cnt
         for s in scans:
             s.execute()
         results = [s.aggregates() for s in scans]
@@ -182,7 +224,23 @@ def main():
     achieved_gbs = (bytes_per_launch / (avg_decode_ms / 1e3)) / 1e9 \
         if avg_decode_ms > 0 else 0.0
     traffic = os.environ.get("YBG_TRAFFIC_BYTES_PER_LAUNCH")
-    traffic = float(traffic) if traffic else None
+    if traffic:
+        traffic = float(traffic)
+    else:
+        # committed PMC measurement for this exact workload, if present
+        # (profiles/traffic_calibration.json; see profiles/README.md for the
+        # collection commands and the gfx950 FETCH_SIZE caveat)
+        try:
+            cal = json.load(open(os.path.join(
+                os.path.dirname(os.path.abspath(__file__)), "profiles",
+                "traffic_calibration.json")))
+            if (cal.get("workload") == args.workload
+                    and cal.get("rows") == args.rows):
+                traffic = float(cal["traffic_bytes_per_launch"])
+            else:
+                traffic = None
+        except Exception:
+            traffic = None
 
     # ---- CPU baseline (oracle, rank 0, N=1 only) -------------------------
     cpu_baseline = None
@@ -239,11 +297,17 @@ def main():
             "dtype": "int64",
             "data": "synthetic",
             "config": {
-                "workload": ("1 tablet, 100M rows, 4KB SST blocks, "
-                             "three_shared_parts, packed-row V2, 3 int64 "
-                             "predicates + SUM(int64)" if n_tablets == 1 else
-                             "8 tablets x 12.5M rows, same query, "
-                             "sharded tablet->rank + RCCL all-reduce"),
+                "workload": (
+                    {"filtersum":
+                         ("1 tablet, 100M rows, 4KB SST blocks, "
+                          "three_shared_parts, packed-row V2, 3 int64 "
+                          "predicates + SUM(int64)" if n_tablets == 1 else
+                          "8 tablets x 12.5M rows, same query, sharded "
+                          "tablet->rank + RCCL all-reduce"),
+                     "mvcc": "100M rows x 5 versions, read-at-HT visibility "
+                             "+ COUNT",
+                     "groupby": "mixed-type rows, predicate + GROUP-BY-key "
+                                "partial aggregates"}[args.workload]),
                 "rows": total_rows,
                 "tablets": n_tablets,
                 "block_size": 4096,
