@@ -124,7 +124,8 @@ def main():
         data, offsets, nb, total, ne = y.generate(
             schema, rows=rows_per_tablet, seed=42 + t, versions=versions,
             ht_base_micros=1_600_000_000_000_000,
-            ht_step_micros=1_000_000 if versions > 1 else 1000)
+            ht_step_micros=1_000_000 if versions > 1 else 1000,
+            group_mod=65536 if group_col else 0)
         total_bytes_local += total
         s = gpu_scan.GpuScan(make_spec())
         s.feed_blocks_host(data, offsets, nb, total)
@@ -311,7 +312,7 @@ def main():
                 "rows": total_rows,
                 "tablets": n_tablets,
                 "block_size": 4096,
-                "checksum": {"sum": checksum[0], "count": checksum[1]},
+                "checksum": {"agg0": checksum[0], "agg1": checksum[1]},
             },
             "roofline": {
                 "bound": "hbm",

@@ -331,6 +331,9 @@ typedef struct {
   uint64_t ht_base_micros; /* first version's HT physical micros */
   uint64_t ht_step_micros; /* HT increment between versions */
   int nthreads;            /* 0 = hw concurrency */
+  uint64_t group_mod;      /* >0: value column 0 becomes (value %% group_mod)
+                              — bounded group-key cardinality for GROUP BY
+                              benchmarks */
 } ybg_gen_params_t;
 
 int ybg_generate(const ybg_schema_t *schema, const ybg_gen_params_t *p,

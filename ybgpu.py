@@ -131,6 +131,7 @@ class GenParams(C.Structure):
         ("versions", C.c_int32),
         ("ht_base_micros", C.c_uint64), ("ht_step_micros", C.c_uint64),
         ("nthreads", C.c_int32),
+        ("group_mod", C.c_uint64),
     ]
 
 
@@ -189,7 +190,7 @@ def read_time(read_micros, local_micros=None, global_micros=None, lib=None):
 def generate(schema, rows, seed=42, packed_version=2,
              kv_format=ENC_THREE_SHARED_PARTS, block_size=4096,
              restart_interval=16, versions=1, ht_base_micros=1_600_000_000_000_000,
-             ht_step_micros=1000, nthreads=0):
+             ht_step_micros=1000, nthreads=0, group_mod=0):
     """Run the multithreaded dataset generator. Returns (data, offsets,
     n_blocks, total_bytes, n_entries); data/offsets are ctypes pointers owned
     by the caller (freed via ybg_free at process exit — we keep them)."""
@@ -201,7 +202,7 @@ def generate(schema, rows, seed=42, packed_version=2,
               C.POINTER(C.c_uint64)])
     p = GenParams(rows, seed, packed_version, kv_format, block_size,
                   restart_interval, versions, ht_base_micros, ht_step_micros,
-                  nthreads)
+                  nthreads, group_mod)
     data = C.POINTER(C.c_uint8)()
     offsets = C.POINTER(C.c_uint64)()
     n_blocks = C.c_uint64()

@@ -608,6 +608,7 @@ int ybg_generate(const ybg_schema_t* schema, const ybg_gen_params_t* p,
           vals.datums[i] = u;
         } else {
           vals.datums[i] = rv & ((1ull << 40) - 1);  // uniform in [0, 2^40)
+          if (i == 0 && p->group_mod) vals.datums[i] %= p->group_mod;
         }
       }
       // MVCC versions, newest first (encoded DocHybridTime sorts newest
