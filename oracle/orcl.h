@@ -171,6 +171,8 @@ typedef enum {
 typedef struct {
   int32_t column_id;     /* value column id (subkey encoding uses this) */
   orcl_dtype_t dtype;
+  int32_t nullable;      /* V1 packing: nullable columns are varlen
+                            (schema_packing.cc:45-49) */
 } orcl_value_col_t;
 
 #define ORCL_MAX_COLS 32

@@ -610,11 +610,10 @@ int orcl_scan(const uint8_t *const *blocks, const size_t *sizes, size_t nblocks,
   for (int g = 0; g < spec->num_aggs; ++g) result->aggs[g].is_null = 1;
 
   packing_t pk;
-  /* nullable flags: all value columns nullable unless caller says otherwise
-   * via schema — we default nullable=1 (matches YSQL tables where non-key
-   * columns are nullable unless NOT NULL). Callers needing NOT NULL fixed
-   * columns use orcl_scan_ex. */
-  packing_init(&pk, schema, NULL);
+  int nullable[ORCL_MAX_COLS];
+  for (int i = 0; i < schema->num_value_cols; ++i)
+    nullable[i] = schema->value_cols[i].nullable;
+  packing_init(&pk, schema, nullable);
 
   scan_ctx_t cx = {.pk = &pk, .spec = spec, .res = result,
                    .cb = row_cb, .cb_arg = cb_arg, .row_seq = 0};

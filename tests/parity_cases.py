@@ -164,6 +164,44 @@ def build_cases():
                         (1_600_000_000_000_005, 1_600_000_000_010_000,
                          1_700_000_000_000_000)]))
 
+    # V1 packed rows with NON-NULLABLE fixed columns (truly fixed layout,
+    # schema_packing.cc:45-49: only nullable/string columns are varlen)
+    schema_nn = y.make_schema(
+        [y.KT_INT64],
+        [(10, y.T_INT64, 0), (11, y.T_INT32, 0), (12, y.T_STRING, 1),
+         (13, y.T_INT64, 1)])
+    b = y.Builder(schema_nn)
+    for r in range(3000):
+        b.add_packed_row(1000 + r,
+                         [(y.T_INT64, r * 3), (y.T_INT32, r % 1000),
+                          (y.T_STRING, b"s%04d" % (r % 50)),
+                          (y.T_INT64, None if r % 4 == 0 else r)],
+                         hash_=r // 64, key_datums=(r,),
+                         packed_version=1)
+    _KEEP.append(b)
+    aggs = [y.Agg(y.AGG_SUM_INT64, 0), y.Agg(y.AGG_SUM_INT64, 1),
+            y.Agg(y.AGG_COUNT, 3)]
+    cases.append(_case("v1_nonnullable_fixed", schema_nn, b.finish(),
+                       [(1_000_000, (), aggs)]))
+
+    # non-default restart interval and the reference's default 32KB blocks
+    built = y.generate(SCHEMA_4I, rows=40_000, restart_interval=4,
+                       block_size=32768)
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 2)]
+    cases.append(_case("restart4_block32k", SCHEMA_4I, built,
+                       [(1_700_000_000_000_000, (), aggs)]))
+
+    # shared_prefix encoding + MVCC versions together
+    built = y.generate(SCHEMA_4I, rows=5_000, versions=3,
+                       kv_format=y.ENC_SHARED_PREFIX,
+                       ht_base_micros=1_600_000_000_000_000,
+                       ht_step_micros=1_000_000_000)
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 0)]
+    cases.append(_case("shared_prefix_mvcc", SCHEMA_4I, built,
+                       [(rm, (), aggs) for rm in
+                        (1_600_000_500_000_000, 1_700_000_000_000_000)],
+                       kv_format=y.ENC_SHARED_PREFIX))
+
     # bounds: lower inclusive / upper exclusive on encoded rowkey
     # (qlexpr/ql_scanspec.h:200-267 bounds; generator rows r have
     # hash = r*65536//rows, key col = r)

@@ -358,7 +358,8 @@ ORCL_ROW_CB = C.CFUNCTYPE(C.c_int, C.POINTER(OrclRow), C.c_void_p)
 
 
 class OrclValueCol(C.Structure):
-    _fields_ = [("column_id", C.c_int32), ("dtype", C.c_int32)]
+    _fields_ = [("column_id", C.c_int32), ("dtype", C.c_int32),
+                ("nullable", C.c_int32)]
 
 
 class OrclSchema(C.Structure):
@@ -587,7 +588,8 @@ def orcl_schema_from(schema):
     o.num_value_cols = schema.num_value_cols
     for i in range(schema.num_value_cols):
         o.value_cols[i] = OrclValueCol(schema.value_cols[i].column_id,
-                                       schema.value_cols[i].dtype)
+                                       schema.value_cols[i].dtype,
+                                       schema.value_cols[i].nullable)
     return o
 
 
