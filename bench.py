@@ -59,16 +59,21 @@ def main():
 
     dist = None
     torch = None
+    backend = None
     if world > 1:
         import torch  # noqa
         import torch.distributed as dist_mod
         dist = dist_mod
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        dist.init_process_group(
-            backend="nccl" if torch.cuda.is_available() else "gloo")
-        if torch.cuda.is_available():
-            torch.cuda.set_device(local_rank)
-        gpu_scan._lib().yb_gpu_set_device(local_rank)
+        backend = os.environ.get(
+            "YBG_DIST_BACKEND",
+            "nccl" if torch.cuda.is_available() else "gloo")
+        dist.init_process_group(backend=backend)
+        ndev = torch.cuda.device_count() if torch.cuda.is_available() else 0
+        dev = local_rank % ndev if ndev else 0
+        if ndev:
+            torch.cuda.set_device(dev)
+        gpu_scan._lib().yb_gpu_set_device(dev)
 
     if not gpu_scan.gpu_available():
         log("FATAL: no HIP device — the product path has no CPU fallback")
@@ -143,9 +148,9 @@ def main():
     agg_buf = None
     if dist is not None:
         import torch
-        agg_buf = torch.zeros(2, dtype=torch.float64,
-                              device="cuda" if torch.cuda.is_available()
-                              else "cpu")
+        agg_buf = torch.zeros(
+            2, dtype=torch.int64,
+            device="cuda" if backend == "nccl" else "cpu")
 
     def step():
         if group_col:
@@ -160,8 +165,8 @@ def main():
                     total_cnt += vals[0] or 0
                     total_sum += vals[1] or 0
             if dist is not None:
-                agg_buf[0] = float(total_sum)
-                agg_buf[1] = float(total_cnt)
+                agg_buf[0] = total_sum
+                agg_buf[1] = total_cnt
                 dist.all_reduce(agg_buf)
                 total_sum = int(agg_buf[0].item())
                 total_cnt = int(agg_buf[1].item())
@@ -172,8 +177,8 @@ def main():
         total_sum = sum(r.aggs[0].value_i64 for r in results)
         total_cnt = sum(r.aggs[1].value_i64 for r in results)
         if dist is not None:
-            agg_buf[0] = float(total_sum)
-            agg_buf[1] = float(total_cnt)
+            agg_buf[0] = total_sum
+            agg_buf[1] = total_cnt
             dist.all_reduce(agg_buf)  # RCCL over xGMI: the one collective
             total_sum = int(agg_buf[0].item())
             total_cnt = int(agg_buf[1].item())
@@ -207,7 +212,7 @@ def main():
     if dist is not None:
         import torch
         e = torch.tensor([elapsed], dtype=torch.float64,
-                         device=agg_buf.device)
+                         device="cuda" if backend == "nccl" else "cpu")
         dist.all_reduce(e, op=dist.ReduceOp.MAX)
         elapsed = float(e.item())
 
