@@ -153,8 +153,6 @@ struct DevSpec {
   int32_t need_rowkey;       // bounds / key predicates / row emission need
                              // the finalized row's key bytes (rk_save)
   int32_t group_col;         // value column to GROUP BY, -1 = none
-  int32_t ablate;            // perf diagnosis only (YBG_ABLATE): 1 = decode
-                             // only, 2 = decode+visibility; results invalid
 };
 
 struct Interval {
@@ -1450,13 +1448,6 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
   slice_u128(ht_enc, ht_size, &ht_hi, &ht_lo);
   uint32_t vb0 = value_len > 0 ? (uint32_t)(rdr->peek8() & 0xff) : 0u;
   bool visible;
-  if (sp.ablate == 2) {
-    // visibility computed, value processing skipped (diagnosis only)
-    visible = u128_slice_cmp(ht_hi, ht_lo, ht_size, sp.reg_lim.hi,
-                             sp.reg_lim.lo, sp.reg_lim.len) >= 0;
-    rc->found |= visible;
-    return true;
-  }
   if (value_len > 0 && vb0 == kHybridTimeByte) {
     // committed-txn record with intent time
     // (intent_aware_iterator.cc:1249-1267)
@@ -1900,8 +1891,7 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
     }
     if (cur_iv == j) *entries += 1;
     else if (cur_iv == j + 1) walked_next = true;
-    if (sp.ablate != 1 &&
-        !process_entry(sp, data, aux, key, key_len, er.value, er.value_len,
+    if (!process_entry(sp, data, aux, key, key_len, er.value, er.value_len,
                        rk_len, &rc, &rdr)) { fail = true; break; }
     rdr.seek(q);
     p = q;

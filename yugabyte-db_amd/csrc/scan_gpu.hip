@@ -585,8 +585,6 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
 
 int yb_gpu_scan_execute(ybg_scan_t* s) {
   if (!s->d_data) return set_err(4, "feed_blocks not called");
-  s->dspec.ablate = 0;
-  if (const char* a = getenv("YBG_ABLATE")) s->dspec.ablate = atoi(a);
   HIP_TRY(hipMemsetAsync(s->d_cont, 0, s->n_heads * sizeof(uint32_t),
                          s->stream));
   HIP_TRY(hipMemsetAsync(s->d_heads, 0,
