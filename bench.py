@@ -156,14 +156,18 @@ def main():
         if group_col:
             # grouped partial aggregates; the cross-tablet merge of the
             # partial tables happens host-side (SURVEY §8e all-gather +
-            # merge; single-node benchmark merges locally)
+            # merge; single-node benchmark reduces the partials directly)
+            import numpy as np
             total_sum = 0
             total_cnt = 0
             for s in scans:
-                groups = s.group_aggregate()
-                for k, vals in groups.items():
-                    total_cnt += vals[0] or 0
-                    total_sum += vals[1] or 0
+                _keys, vals, cnts, _kb, n = s.group_aggregate_raw()
+                if n:
+                    va = np.frombuffer(vals, dtype=np.int64,
+                                       count=n * y.MAX_AGGS)
+                    va = va.reshape(n, y.MAX_AGGS)
+                    total_cnt += int(va[:, 0].sum())
+                    total_sum += int(va[:, 1].sum())
             if dist is not None:
                 agg_buf[0] = total_sum
                 agg_buf[1] = total_cnt

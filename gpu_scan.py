@@ -105,8 +105,9 @@ class GpuScan:
                                    b.key_datums, b.datums, b.null_masks,
                                    b.varlen)
 
-    def group_aggregate(self, cap=1 << 20, key_bytes_cap=1 << 24):
-        """GROUP BY partial aggregates (spec.group_col must be set)."""
+    def group_aggregate_raw(self, cap=1 << 20, key_bytes_cap=1 << 24):
+        """GROUP BY partial aggregates: raw ctypes arrays
+        (keys, vals, cnts, key_bytes, n_groups)."""
         keys = (C.c_uint64 * cap)()
         vals = (C.c_int64 * (cap * y.MAX_AGGS))()
         cnts = (C.c_uint64 * (cap * y.MAX_AGGS))()
@@ -116,8 +117,13 @@ class GpuScan:
             self._lib.yb_gpu_scan_group_aggregate(
                 self._h, keys, vals, cnts, kb, key_bytes_cap, cap,
                 C.byref(n)), "group_aggregate")
+        return keys, vals, cnts, kb, n.value
+
+    def group_aggregate(self, cap=1 << 20, key_bytes_cap=1 << 24):
+        """GROUP BY partial aggregates (spec.group_col must be set)."""
+        keys, vals, cnts, kb, n = self.group_aggregate_raw(cap, key_bytes_cap)
         return y._decode_groups(self._spec.schema, self._spec.group_col - 1,
-                                n.value, keys, vals, cnts, kb,
+                                n, keys, vals, cnts, kb,
                                 self._spec.num_aggs, self._spec.aggs)
 
     def kernel_ms(self):
