@@ -441,6 +441,7 @@ static int pred_eval(const orcl_pred_t *pr, const orcl_schema_t *sc,
 /* doc_expr.cc:248-395 aggregate semantics. */
 static void agg_update(orcl_agg_result_t *a, const orcl_agg_t *spec,
                        const orcl_schema_t *sc, const orcl_row_t *row) {
+  (void)sc;
   int col = spec->col;
   int isnull = spec->op == ORCL_AGG_COUNT_STAR
                    ? 0

@@ -607,8 +607,10 @@ def orcl_read_time(read_micros, local_micros=None, global_micros=None,
 
 
 def orcl_scan(data, offsets, n_blocks, schema, spec, kv_format=ENC_THREE_SHARED_PARTS,
-              collect_rows=False):
+              collect_rows=False, block_lo=0):
     """Run the oracle over a block array (as returned by generate()/Builder).
+    `block_lo` starts the scan at a later block (contiguous sub-range; rows
+    straddling the cut are attributed per-range — measurement use only).
     Returns (OrclScanResult, rows or None). TEST INFRASTRUCTURE ONLY."""
     lib = oracle()
     f = _sig(lib, "orcl_scan", C.c_int,
@@ -620,8 +622,8 @@ def orcl_scan(data, offsets, n_blocks, schema, spec, kv_format=ENC_THREE_SHARED_
     blocks = (C.POINTER(C.c_uint8) * n_blocks)()
     sizes = (C.c_size_t * n_blocks)()
     for i in range(n_blocks):
-        blocks[i] = C.cast(base + offsets[i], C.POINTER(C.c_uint8))
-        sizes[i] = offsets[i + 1] - offsets[i]
+        blocks[i] = C.cast(base + offsets[block_lo + i], C.POINTER(C.c_uint8))
+        sizes[i] = offsets[block_lo + i + 1] - offsets[block_lo + i]
     res = OrclScanResult()
     rows = [] if collect_rows else None
 
