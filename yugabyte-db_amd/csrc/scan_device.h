@@ -1862,14 +1862,25 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
         // finalize previous row (doc_rowwise_iterator row boundary)
         const uint8_t* rkp = (rkb && !EMIT) ? key : rk_save;
         if (rc.found && in_bounds(sp, rkp, rk_len, aux)) {
-          uint64_t* sc = in_head ? &ho->scanned : scanned;
-          uint64_t* ma = in_head ? &ho->matched : matched;
-          *sc += 1;
-          if ((rc.pred_pass & sp.value_pred_mask) == sp.value_pred_mask &&
-              eval_key_preds(sp, rkp, rk_len, aux)) {
-            *ma += 1;
-            acc_row(sp, rc, in_head ? ho->val : agg_val,
-                    in_head ? ho->cnt : agg_cnt);
+          // explicit in_head/main branches (never a runtime-selected
+          // pointer): both destinations must stay register-promotable
+          bool hit = (rc.pred_pass & sp.value_pred_mask) ==
+                         sp.value_pred_mask &&
+                     eval_key_preds(sp, rkp, rk_len, aux);
+          if (in_head) {
+            ho->scanned += 1;
+            if (hit) {
+              ho->matched += 1;
+              acc_row(sp, rc, ho->val, ho->cnt);
+            }
+          } else {
+            *scanned += 1;
+            if (hit) {
+              *matched += 1;
+              acc_row(sp, rc, agg_val, agg_cnt);
+            }
+          }
+          if (hit) {
             if (EMIT && (!in_head || ec->head_consumed[j] == 0))
               emit_row(sp, ec, rc, rkp, rk_len, row_sort_key);
             if (GROUP && (!in_head || head_flags[j] == 0))
@@ -1898,13 +1909,22 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
   }
   const uint8_t* rkp_end = (rkb && !EMIT) ? key : rk_save;
   if (!fail && row_open && rc.found && in_bounds(sp, rkp_end, rk_len, aux)) {
-    uint64_t* sc = in_head ? &ho->scanned : scanned;
-    uint64_t* ma = in_head ? &ho->matched : matched;
-    *sc += 1;
-    if ((rc.pred_pass & sp.value_pred_mask) == sp.value_pred_mask &&
-        eval_key_preds(sp, rkp_end, rk_len, aux)) {
-      *ma += 1;
-      acc_row(sp, rc, in_head ? ho->val : agg_val, in_head ? ho->cnt : agg_cnt);
+    bool hit = (rc.pred_pass & sp.value_pred_mask) == sp.value_pred_mask &&
+               eval_key_preds(sp, rkp_end, rk_len, aux);
+    if (in_head) {
+      ho->scanned += 1;
+      if (hit) {
+        ho->matched += 1;
+        acc_row(sp, rc, ho->val, ho->cnt);
+      }
+    } else {
+      *scanned += 1;
+      if (hit) {
+        *matched += 1;
+        acc_row(sp, rc, agg_val, agg_cnt);
+      }
+    }
+    if (hit) {
       if (EMIT && (!in_head || ec->head_consumed[j] == 0))
         emit_row(sp, ec, rc, rkp_end, rk_len, row_sort_key);
       if (GROUP && (!in_head || head_flags[j] == 0))

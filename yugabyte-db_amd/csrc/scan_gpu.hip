@@ -74,9 +74,13 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
   const uint64_t span = (uint64_t)gridDim.x * kThreads;
 
   uint64_t entries = 0, scanned = 0, matched = 0, errs = 0;
-  uint64_t agg_val[YBG_MAX_AGGS], agg_cnt[YBG_MAX_AGGS];
+  // NA-sized and only ever constant-indexed (all loops over them unrolled):
+  // any runtime index would force these accumulators into scratch memory,
+  // and with them every per-row aggregate update in the hot loop (measured
+  // as ~19.6 GB/dispatch of scratch write traffic before this was fixed).
+  uint64_t agg_val[NA], agg_cnt[NA];
 #pragma unroll
-  for (int g = 0; g < YBG_MAX_AGGS; ++g) { agg_val[g] = 0; agg_cnt[g] = 0; }
+  for (int g = 0; g < NA; ++g) { agg_val[g] = 0; agg_cnt[g] = 0; }
 
   for (uint64_t j0 = 0; j0 < n_ivs; j0 += span) {
     const uint64_t j = j0 + gtid;
@@ -154,7 +158,9 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
     }
     __shared__ uint64_t red_val[kThreads], red_cnt[kThreads];
     uint64_t wave_id = ((uint64_t)blockIdx.x * kThreads + threadIdx.x) >> 6;
-    for (int g = 0; g < sp.num_aggs; ++g) {
+#pragma unroll
+    for (int g = 0; g < NA; ++g) {
+      if (g >= sp.num_aggs) continue;
       red_val[threadIdx.x] = agg_val[g];
       red_cnt[threadIdx.x] = agg_cnt[g];
       __syncthreads();
@@ -394,18 +400,6 @@ int set_err(int code, const std::string& msg) {
 #define HIP_WARN(x) \
   do { hipError_t _e = (x); (void)_e; } while (0)
 
-void htlim_from(const uint8_t* b, int32_t len, HtLim* out) {
-  uint64_t hi = 0, lo = 0;
-  for (int i = 0; i < len && i < 16; ++i) {
-    uint64_t v = b[i];
-    if (i < 8) hi |= v << (56 - 8 * i);
-    else lo |= v << (56 - 8 * (i - 8));
-  }
-  out->hi = hi;
-  out->lo = lo;
-  out->len = (uint32_t)len;
-}
-
 }  // namespace
 
 struct ybg_scan {
@@ -592,8 +586,10 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
                          s->stream));
   HIP_TRY(hipEventRecord(s->ev_start, s->stream));
   // dispatch on aggregate-slot capacity (register footprint) and the
-  // waves-per-SIMD occupancy bound (YBG_WPS for tuning, default 4)
-  int wps = 4;
+  // waves-per-SIMD occupancy bound (YBG_WPS for tuning, default 3: after the
+  // accumulator register promotion, the WPS=4 build only fits 4 waves/SIMD
+  // by spilling VGPRs in the hot loop and measures ~30% slower)
+  int wps = 3;
   if (const char* e = getenv("YBG_WPS")) {
     long v = atol(e);
     if (v >= 2 && v <= 6) wps = (int)v;
@@ -703,7 +699,7 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
   HIP_TRY(hipMemcpyToSymbolAsync(HIP_SYMBOL(c_spec), &s->dspec,
                                  sizeof(DevSpec), 0, hipMemcpyHostToDevice,
                                  s->stream));
-  auto flags_kernel = k_scan<2, 4>;
+  auto flags_kernel = k_scan<2, 3>;
   hipLaunchKernelGGL(flags_kernel, dim3(s->grid), dim3(kThreads), 0,
                      s->stream, s->d_data, s->d_offsets, s->d_ivs,
                      s->n_ivs, s->d_aux, s->d_rk_save, s->d_partials,
@@ -821,7 +817,7 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
   HIP_TRY(hipMemcpyToSymbolAsync(HIP_SYMBOL(c_spec), &s->dspec,
                                  sizeof(DevSpec), 0, hipMemcpyHostToDevice,
                                  s->stream));
-  auto flags_kernel2 = k_scan<2, 4>;
+  auto flags_kernel2 = k_scan<2, 3>;
   hipLaunchKernelGGL(flags_kernel2, dim3(s->grid), dim3(kThreads), 0,
                      s->stream, s->d_data, s->d_offsets, s->d_ivs,
                      s->n_ivs, s->d_aux, s->d_rk_save, s->d_partials,
