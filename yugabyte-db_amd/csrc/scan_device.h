@@ -134,6 +134,15 @@ struct DevAgg {
   int32_t op, col;
 };
 
+// Compact predicate record for the hot eval path: one 16-byte load fetches
+// everything a typed compare needs. For string predicates, datum packs
+// (str_len << 32) | rhs_off (aux-buffer offset of the RHS bytes).
+struct PredC {
+  uint64_t datum;
+  uint32_t opdt;  // op | (dtype << 8)
+  uint32_t pad_;
+};
+
 struct DevSpec {
   int32_t has_hash, num_hash_cols, num_range_cols;
   int32_t key_types[YBG_MAX_KEYCOLS];
@@ -153,7 +162,33 @@ struct DevSpec {
   int32_t need_rowkey;       // bounds / key predicates / row emission need
                              // the finalized row's key bytes (rk_save)
   int32_t group_col;         // value column to GROUP BY, -1 = none
+
+  // --- Hot-path compaction (host-precomputed). The generic arrays above
+  // stay for the cold paths (key predicates, V1 decode); the hot loop reads
+  // only these, so the compiler keeps a handful of uniform values live
+  // instead of every slot of every array (measured: the full-unroll
+  // eval_col held ~600 spilled SGPRs of hoisted spec fields).
+  //
+  // col_act[i] — one action word per value column:
+  //   bits 0-7   mask of predicates evaluated on this column
+  //   bits 8-15  mask of aggregate slots fed by this column
+  //   bit  16    this is the GROUP BY column
+  //   bits 17-20 column dtype (YBG_T_*)
+  //   bits 21-24 packed-V2 fixed width (0 = varlen, else 1/2/4/8)
+  uint32_t col_act[YBG_MAX_COLS];
+  PredC predc[YBG_MAX_PREDS];
+  int32_t agg_op[YBG_MAX_AGGS];
+  uint32_t key_pred_mask;  // mask of predicates on key columns
 };
+
+constexpr uint32_t kActPredM = 0xffu;
+constexpr uint32_t kActAggShift = 8, kActAggM = 0xffu;
+constexpr uint32_t kActGroup = 1u << 16;
+constexpr uint32_t kActDtShift = 17, kActDtM = 0xfu;
+constexpr uint32_t kActV2Shift = 21, kActV2M = 0xfu;
+
+#define YBG_UNLIKELY(x) __builtin_expect(!!(x), 0)
+#define YBG_LIKELY(x) __builtin_expect(!!(x), 1)
 
 struct Interval {
   uint32_t block;
@@ -929,12 +964,18 @@ DEV void row_reset(RowCtxT<NA>* rc, const DevSpec& sp) {
 
 // Value-column predicate compare (pgsql_operation.cc:602-668 typed-compare
 // subset). sptr/slen only for string columns.
-DEV bool pred_compare(const DevPred& pr, int dtype, uint64_t datum,
-                      const uint8_t* sptr, uint32_t slen, const uint8_t* aux) {
+DEV bool pred_compare(const PredC& pr, uint64_t datum, const uint8_t* sptr,
+                      uint32_t slen, const uint8_t* aux) {
+  int dtype = (int)(pr.opdt >> 8);
   int cmp;
-  if (dtype == YBG_T_STRING) {
-    const uint8_t* rhs = aux + pr.rhs_off;
-    uint32_t n = slen < pr.str_len ? slen : pr.str_len;
+  if (YBG_LIKELY(dtype != YBG_T_STRING && dtype != YBG_T_DOUBLE &&
+                 dtype != YBG_T_FLOAT)) {
+    int64_t a = (int64_t)datum, b = (int64_t)pr.datum;
+    cmp = a < b ? -1 : (a > b ? 1 : 0);
+  } else if (dtype == YBG_T_STRING) {
+    const uint8_t* rhs = aux + (uint32_t)pr.datum;
+    uint32_t rlen = (uint32_t)(pr.datum >> 32);
+    uint32_t n = slen < rlen ? slen : rlen;
     cmp = 0;
     for (uint32_t k = 0; k < n; ++k) {
       if (sptr[k] != rhs[k]) {
@@ -942,20 +983,17 @@ DEV bool pred_compare(const DevPred& pr, int dtype, uint64_t datum,
         break;
       }
     }
-    if (cmp == 0 && slen != pr.str_len) cmp = slen < pr.str_len ? -1 : 1;
+    if (cmp == 0 && slen != rlen) cmp = slen < rlen ? -1 : 1;
   } else if (dtype == YBG_T_DOUBLE) {
     double a = __longlong_as_double((long long)datum);
     double b = __longlong_as_double((long long)pr.datum);
     cmp = a < b ? -1 : (a > b ? 1 : 0);
-  } else if (dtype == YBG_T_FLOAT) {
+  } else {
     float a = __uint_as_float((uint32_t)datum);
     float b = __uint_as_float((uint32_t)pr.datum);
     cmp = a < b ? -1 : (a > b ? 1 : 0);
-  } else {
-    int64_t a = (int64_t)datum, b = (int64_t)pr.datum;
-    cmp = a < b ? -1 : (a > b ? 1 : 0);
   }
-  switch (pr.op) {
+  switch (pr.opdt & 0xff) {
     case YBG_PRED_GT: return cmp > 0;
     case YBG_PRED_GE: return cmp >= 0;
     case YBG_PRED_LT: return cmp < 0;
@@ -971,7 +1009,8 @@ template <int NA>
 DEV void eval_col(const DevSpec& sp, const uint8_t* aux, RowCtxT<NA>* rc,
                   int col, bool is_null, uint64_t datum, const uint8_t* sptr,
                   uint32_t slen) {
-  if (col == sp.group_col) {
+  const uint32_t act = sp.col_act[col];
+  if (YBG_UNLIKELY(act & kActGroup)) {
     rc->grp_null = is_null;
     rc->grp_datum = sptr ? (uint64_t)(uintptr_t)sptr : datum;
     rc->grp_len = sptr ? slen : 0;
@@ -990,23 +1029,23 @@ DEV void eval_col(const DevSpec& sp, const uint8_t* aux, RowCtxT<NA>* rc,
       }
     }
   }
-#pragma unroll
-  for (int i = 0; i < YBG_MAX_PREDS; ++i) {
-    if (i < sp.num_preds && !sp.preds[i].is_key_col &&
-        sp.preds[i].col == col) {
-      bool pass =
-          !is_null && pred_compare(sp.preds[i], sp.cols[col].dtype, datum,
-                                   sptr, slen, aux);
-      rc->pred_pass = (rc->pred_pass & ~(1u << i)) | ((uint32_t)pass << i);
-    }
+  // predicates on this column: iterate set bits only (the full
+  // MAX_PREDS unroll kept every slot's fields live as uniform registers)
+  uint32_t pm = act & kActPredM;
+  while (pm) {
+    int i = __builtin_ctz(pm);
+    pm &= pm - 1;
+    bool pass = !is_null && pred_compare(sp.predc[i], datum, sptr, slen, aux);
+    rc->pred_pass = (rc->pred_pass & ~(1u << i)) | ((uint32_t)pass << i);
   }
+  uint32_t am = (act >> kActAggShift) & kActAggM;
+  if (am) {
 #pragma unroll
-  for (int g = 0; g < NA; ++g) {
-    if (g < sp.num_aggs && sp.aggs[g].col == col &&
-        sp.aggs[g].op != YBG_AGG_COUNT_STAR) {
-      rc->agg_datum[g] = datum;
-      rc->agg_null =
-          (rc->agg_null & ~(1u << g)) | ((uint32_t)is_null << g);
+    for (int g = 0; g < NA; ++g) {
+      if (am & (1u << g)) {
+        rc->agg_datum[g] = datum;
+        rc->agg_null = (rc->agg_null & ~(1u << g)) | ((uint32_t)is_null << g);
+      }
     }
   }
 }
@@ -1220,9 +1259,10 @@ DEV bool key_col_value(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
 // (doc_kv_util.h:101-167): unescape on the fly.
 DEV bool eval_key_preds(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
                         const uint8_t* aux) {
-#pragma unroll
-  for (int i = 0; i < YBG_MAX_PREDS; ++i) {
-    if (i >= sp.num_preds || !sp.preds[i].is_key_col) continue;
+  uint32_t kp = sp.key_pred_mask;
+  while (kp) {
+    int i = __builtin_ctz(kp);
+    kp &= kp - 1;
     const DevPred& pr = sp.preds[i];
     uint64_t d = 0;
     uint32_t soff = 0, sl = 0;
@@ -1297,12 +1337,12 @@ DEV void acc_row(const DevSpec& sp, const RowCtxT<NA>& rc, uint64_t* agg_val,
 #pragma unroll
   for (int g = 0; g < NA; ++g) {
     if (g >= sp.num_aggs) continue;
-    const DevAgg& a = sp.aggs[g];
+    const int op = sp.agg_op[g];
     bool isnull =
-        (a.op == YBG_AGG_COUNT_STAR) ? false : ((rc.agg_null >> g) & 1);
+        (op == YBG_AGG_COUNT_STAR) ? false : ((rc.agg_null >> g) & 1);
     if (isnull) continue;
     uint64_t v = rc.agg_datum[g];
-    switch (a.op) {
+    switch (op) {
       case YBG_AGG_COUNT_STAR:
       case YBG_AGG_COUNT:
         agg_val[g] += 1;
@@ -1382,7 +1422,7 @@ DEV void agg_combine(const DevSpec& sp, uint64_t* a_val, uint64_t* a_cnt,
 #pragma unroll
   for (int g = 0; g < YBG_MAX_AGGS; ++g) {
     if (g >= sp.num_aggs) continue;
-    combine1(sp.aggs[g].op, &a_val[g], &a_cnt[g], b_val[g], b_cnt[g]);
+    combine1(sp.agg_op[g], &a_val[g], &a_cnt[g], b_val[g], b_cnt[g]);
   }
 }
 
@@ -1405,21 +1445,23 @@ DEV bool decode_packed_v2_rdr(const DevSpec& sp, const uint8_t* base,
   const uint8_t* end = rdr->pos() + len;
   rdr->consume(3);
   for (int i = 0; i < sp.num_value_cols; ++i) {
-    const DevCol& c = sp.cols[i];
-    if (c.v2_fixed) {
-      if (rdr->pos() + c.v2_fixed > end) return false;
+    const uint32_t act = sp.col_act[i];
+    const uint32_t fw = (act >> kActV2Shift) & kActV2M;
+    if (fw) {
+      if (YBG_UNLIKELY(rdr->pos() + fw > end)) return false;
       uint64_t u = rdr->peek8();
-      switch (c.v2_fixed) {
+      switch (fw) {
         case 1: u &= 0xff; break;
         case 2: u &= 0xffff; break;
         case 4: u &= 0xffffffffull; break;
         default: break;
       }
-      if (c.dtype == YBG_T_INT8) u = (uint64_t)(int64_t)(int8_t)u;
-      else if (c.dtype == YBG_T_INT16) u = (uint64_t)(int64_t)(int16_t)u;
-      else if (c.dtype == YBG_T_INT32) u = (uint64_t)(int64_t)(int32_t)u;
+      const uint32_t dt = (act >> kActDtShift) & kActDtM;
+      if (dt == YBG_T_INT8) u = (uint64_t)(int64_t)(int8_t)u;
+      else if (dt == YBG_T_INT16) u = (uint64_t)(int64_t)(int16_t)u;
+      else if (dt == YBG_T_INT32) u = (uint64_t)(int64_t)(int32_t)u;
       eval_col(sp, aux, rc, i, false, u, nullptr, 0);
-      rdr->consume((uint32_t)c.v2_fixed);
+      rdr->consume(fw);
     } else {
       if (rdr->pos() >= end) return false;
       uint64_t w0 = rdr->peek8();
@@ -2039,8 +2081,39 @@ inline void build_dev_spec(const ybg_scan_spec_t* spec, DevSpec* dp,
   for (int i = 0; i < spec->num_aggs; ++i) {
     d.aggs[i].op = spec->aggs[i].op;
     d.aggs[i].col = spec->aggs[i].col;
+    d.agg_op[i] = spec->aggs[i].op;
   }
   d.group_col = spec->group_col > 0 ? spec->group_col - 1 : -1;
+  // hot-path compaction: per-column action words + compact predicate records
+  d.key_pred_mask = 0;
+  for (int i = 0; i < spec->num_preds; ++i) {
+    const DevPred& pr = d.preds[i];
+    PredC& pc = d.predc[i];
+    if (pr.is_key_col) {
+      d.key_pred_mask |= 1u << i;
+      pc.datum = 0;
+      pc.opdt = 0;
+      continue;
+    }
+    int dt = d.cols[pr.col].dtype;
+    pc.opdt = (uint32_t)pr.op | ((uint32_t)dt << 8);
+    pc.datum = (dt == YBG_T_STRING)
+                   ? (((uint64_t)pr.str_len << 32) | pr.rhs_off)
+                   : pr.datum;
+    pc.pad_ = 0;
+  }
+  for (int c = 0; c < sc.num_value_cols; ++c) {
+    uint32_t act = 0;
+    for (int i = 0; i < spec->num_preds; ++i)
+      if (!d.preds[i].is_key_col && d.preds[i].col == c) act |= 1u << i;
+    for (int g = 0; g < spec->num_aggs; ++g)
+      if (d.aggs[g].col == c && d.aggs[g].op != YBG_AGG_COUNT_STAR)
+        act |= 1u << (kActAggShift + g);
+    if (d.group_col == c) act |= kActGroup;
+    act |= ((uint32_t)d.cols[c].dtype & kActDtM) << kActDtShift;
+    act |= ((uint32_t)d.cols[c].v2_fixed & kActV2M) << kActV2Shift;
+    d.col_act[c] = act;
+  }
   d.lower_len = (uint32_t)spec->lower_bound_len;
   d.lower_off = put(spec->lower_bound, spec->lower_bound_len);
   d.upper_len = (uint32_t)spec->upper_bound_len;

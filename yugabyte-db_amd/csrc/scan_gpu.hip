@@ -132,7 +132,7 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
 #pragma unroll
         for (int g = 0; g < NA; ++g) {
           if (g < sp.num_aggs)
-            combine1(sp.aggs[g].op, &agg_val[g], &agg_cnt[g], ho.val[g],
+            combine1(sp.agg_op[g], &agg_val[g], &agg_cnt[g], ho.val[g],
                      ho.cnt[g]);
         }
       }
@@ -167,7 +167,7 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
       if (lane == 0) {
         uint64_t av = 0, ac = 0;
         for (int l = 0; l < 64; ++l)
-          combine1(sp.aggs[g].op, &av, &ac, red_val[threadIdx.x + l],
+          combine1(sp.agg_op[g], &av, &ac, red_val[threadIdx.x + l],
                    red_cnt[threadIdx.x + l]);
         partials[wave_id * kPartialStride + 4 + 2 * g] = av;
         partials[wave_id * kPartialStride + 4 + 2 * g + 1] = ac;
