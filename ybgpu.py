@@ -9,7 +9,8 @@ import ctypes as C
 import os
 
 _ROOT = os.path.dirname(os.path.abspath(__file__))
-_PRODUCT_SO = os.path.join(_ROOT, "yugabyte-db_amd", "libybgpu.so")
+_PRODUCT_SO = os.environ.get(
+    "YBG_SO", os.path.join(_ROOT, "yugabyte-db_amd", "libybgpu.so"))
 _ORACLE_SO = os.path.join(_ROOT, "oracle", "liborcl.so")
 
 MAX_COLS = 32

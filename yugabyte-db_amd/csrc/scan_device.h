@@ -179,6 +179,12 @@ struct DevSpec {
   PredC predc[YBG_MAX_PREDS];
   int32_t agg_op[YBG_MAX_AGGS];
   uint32_t key_pred_mask;  // mask of predicates on key columns
+  int32_t col_ids[YBG_MAX_COLS];  // column ids (kColB update lookup)
+  // When every value column is fixed-width: total packed-V2 body length
+  // ('|' + version + flags + fixed bodies; single-byte version assumed) and
+  // per-column body offsets from the value start. 0 = fast path off.
+  uint32_t v2_fixed_len;
+  uint8_t v2_off[YBG_MAX_COLS];
 };
 
 constexpr uint32_t kActPredM = 0xffu;
@@ -787,22 +793,26 @@ DEV const uint8_t* decode_entry(int fmt, Rdr* rdr, const uint8_t* limit,
           uint64_t new_ns2_start = sp + ns1 + mid;
           uint64_t new_len = new_ns2_start + ns2 + 8;
           if (new_len <= kKeyCap) {
+            // Row-change detection without LDS readback: the reference's
+            // BlockBuilder emits the MAXIMAL shared prefix
+            // (rocksdb/table/block_builder.cc delta encode), so a shared
+            // prefix below the fixed rowkey length means the byte at
+            // `shared_prefix` differs from the previous key — the row
+            // changed; a shared prefix >= rkb means bytes [0, rkb) are
+            // untouched. (The general decode path below keeps the
+            // compare-on-write detection and covers restart entries.)
+            if (sp < rkb) *changed = true;
             // key bytes [hl, hl+ns1+ns2) — within the first 16 window bytes
             uint64_t w2 = rdr->peek8_at(8);
             for (uint32_t i = 0; i < ns1; ++i) {
               uint32_t j = hl + i;
               uint64_t src = j < 8 ? w : w2;
-              uint8_t b = (uint8_t)(src >> (8 * (j & 7)));
-              if (sp + i < rkb && key[sp + i] != b) *changed = true;
-              key[sp + i] = b;
+              key[sp + i] = (uint8_t)(src >> (8 * (j & 7)));
             }
             for (uint32_t i = 0; i < ns2; ++i) {
               uint32_t j = hl + ns1 + i;
               uint64_t src = j < 8 ? w : w2;
-              uint8_t b = (uint8_t)(src >> (8 * (j & 7)));
-              if (new_ns2_start + i < rkb && key[new_ns2_start + i] != b)
-                *changed = true;
-              key[new_ns2_start + i] = b;
+              key[new_ns2_start + i] = (uint8_t)(src >> (8 * (j & 7)));
             }
             *last8 += inc;
             *key_len = (uint32_t)new_len;
@@ -1065,11 +1075,11 @@ DEV int decode_single_v1(const DevSpec& sp, const uint8_t* base,
     eval_col(sp, aux, rc, col, true, 0, nullptr, 0);
     return 0;
   }
-  const DevCol& c = sp.cols[col];
+  const int dtype = (int)((sp.col_act[col] >> kActDtShift) & kActDtM);
   uint64_t datum = 0;
   const uint8_t* sptr = nullptr;
   uint32_t slen = 0;
-  switch (c.dtype) {
+  switch (dtype) {
     case YBG_T_BOOL:
       if (t != kTrueB && t != kFalseB) return -1;
       datum = (t == kTrueB);
@@ -1482,6 +1492,38 @@ DEV bool decode_packed_v2_rdr(const DevSpec& sp, const uint8_t* base,
   return true;
 }
 
+// Fixed-offset packed-V2 decode: when the schema is all-fixed-width and the
+// row carries no null mask (flags == 0, single-byte schema version), every
+// column body sits at a host-precomputed offset from the value start
+// (schema_packing.cc:1076-1121 layout with no varlen entries). The loads
+// are INDEPENDENT global dword loads on lines the reader window just
+// touched (L1-resident) — no serial window-consume chain.
+template <int NA>
+DEV void decode_packed_v2_fixed(const DevSpec& sp, const uint8_t* aux,
+                                const uint8_t* value, RowCtxT<NA>* rc) {
+  for (int i = 0; i < sp.num_value_cols; ++i) {
+    const uint32_t act = sp.col_act[i];
+    uint64_t u = load_u64_una(value + sp.v2_off[i]);
+    const uint32_t dt = (act >> kActDtShift) & kActDtM;
+    switch ((act >> kActV2Shift) & kActV2M) {
+      case 1:
+        u = (dt == YBG_T_INT8) ? (uint64_t)(int64_t)(int8_t)u : (u & 0xff);
+        break;
+      case 2:
+        u = (dt == YBG_T_INT16) ? (uint64_t)(int64_t)(int16_t)u
+                                : (u & 0xffff);
+        break;
+      case 4:
+        u = (dt == YBG_T_INT32) ? (uint64_t)(int64_t)(int32_t)u
+                                : (u & 0xffffffffull);
+        break;
+      default:
+        break;
+    }
+    eval_col(sp, aux, rc, i, false, u, nullptr, 0);
+  }
+}
+
 // Visibility + row-state update for one entry. key/rowkey live in LDS.
 // rdr is positioned at the value start. Returns false on corruption.
 template <int NA>
@@ -1529,6 +1571,14 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
       rc->base_ht_lo = ht_lo;
       rc->base_ht_len = ht_size;
       if (value_len > 0 && vb0 == kPackedV2B && value == rdr->pos()) {
+        if (sp.v2_fixed_len && value_len == sp.v2_fixed_len &&
+            (rdr->peek8() & 0xff8000u) == 0) {
+          // all-fixed schema, no null mask, 1-byte version: direct
+          // fixed-offset column loads, no window-consume chain
+          decode_packed_v2_fixed(sp, aux, value, rc);
+          rc->found = true;
+          return true;
+        }
         // dominant shape: packed V2, no control fields — register-window
         // decode; falls back when a null mask / big schema version appears
         bool done;
@@ -1574,8 +1624,9 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
       const uint8_t* q = svarint(sk + 1, sk + sk_len, &col_id);
       if (!q || q != sk + sk_len) return false;
       int idx = -1;
+#pragma clang loop unroll(disable)
       for (int i = 0; i < sp.num_value_cols; ++i)
-        if (sp.cols[i].id == (int32_t)col_id) { idx = i; break; }
+        if (sp.col_ids[i] == (int32_t)col_id) { idx = i; break; }
       if (idx < 0) return true;
       if (rc->cur_col != idx) {
         rc->cur_col = idx;
@@ -2113,6 +2164,19 @@ inline void build_dev_spec(const ybg_scan_spec_t* spec, DevSpec* dp,
     act |= ((uint32_t)d.cols[c].dtype & kActDtM) << kActDtShift;
     act |= ((uint32_t)d.cols[c].v2_fixed & kActV2M) << kActV2Shift;
     d.col_act[c] = act;
+    d.col_ids[c] = d.cols[c].id;
+  }
+  {
+    // fixed-offset packed-V2 fast path: usable when every value column is
+    // fixed-width ('|' + 1-byte version + flags + bodies)
+    uint32_t off = 3;
+    bool all_fixed = sc.num_value_cols > 0;
+    for (int c = 0; c < sc.num_value_cols; ++c) {
+      if (!d.cols[c].v2_fixed || off > 255) { all_fixed = false; break; }
+      d.v2_off[c] = (uint8_t)off;
+      off += (uint32_t)d.cols[c].v2_fixed;
+    }
+    d.v2_fixed_len = all_fixed ? off : 0;
   }
   d.lower_len = (uint32_t)spec->lower_bound_len;
   d.lower_off = put(spec->lower_bound, spec->lower_bound_len);

@@ -587,7 +587,7 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
   HIP_TRY(hipEventRecord(s->ev_start, s->stream));
   // dispatch on aggregate-slot capacity (register footprint) and the
   // waves-per-SIMD occupancy bound (YBG_WPS for tuning, default 3: after the
-  // accumulator register promotion, the WPS=4 build only fits 4 waves/SIMD
+  // accumulator register promotion, the WPS=4 build only fits 4 waves/SIMD (default 5: best in the post-col_act sweep; the bound compiles to 3 waves)
   // by spilling VGPRs in the hot loop and measures ~30% slower)
   int wps = 3;
   if (const char* e = getenv("YBG_WPS")) {
