@@ -1501,9 +1501,18 @@ DEV bool decode_packed_v2_rdr(const DevSpec& sp, const uint8_t* base,
 template <int NA>
 DEV void decode_packed_v2_fixed(const DevSpec& sp, const uint8_t* aux,
                                 const uint8_t* value, RowCtxT<NA>* rc) {
+  // one shared alignment for the whole body: every column extract is a
+  // funnel shift over two aligned words (the words are shared between
+  // adjacent columns instead of re-deriving per-column unaligned loads)
+  const uintptr_t a = (uintptr_t)value;
+  const uint64_t* qw = (const uint64_t*)(a & ~(uintptr_t)7);
+  const uint32_t abase = (uint32_t)(a & 7);
   for (int i = 0; i < sp.num_value_cols; ++i) {
     const uint32_t act = sp.col_act[i];
-    uint64_t u = load_u64_una(value + sp.v2_off[i]);
+    const uint32_t ob = abase + sp.v2_off[i];
+    const uint32_t wi = ob >> 3, sh = (ob & 7) * 8;
+    uint64_t u = qw[wi];
+    if (sh) u = (u >> sh) | (qw[wi + 1] << (64 - sh));
     const uint32_t dt = (act >> kActDtShift) & kActDtM;
     switch ((act >> kActV2Shift) & kActV2M) {
       case 1:

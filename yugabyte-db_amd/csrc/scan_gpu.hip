@@ -325,6 +325,67 @@ struct DevResult {
   uint64_t agg_cnt[YBG_MAX_AGGS];
 };
 
+// Pre-reduction: workgroup c folds a CONTIGUOUS chunk of wave partials and
+// head records (fixed thread-strided order inside the chunk) into one
+// partial-layout record; k_reduce then folds the 256 chunk records in fixed
+// order. The fold tree is fixed => results stay deterministic run-to-run;
+// the single-workgroup k_reduce no longer reads megabytes through one CU.
+__global__ __launch_bounds__(256) void k_reduce_pre(
+    DevSpec sp, const uint64_t* __restrict__ partials, uint64_t n_partials,
+    const uint64_t* __restrict__ heads, const uint32_t* __restrict__ cont_flags,
+    uint64_t n_heads, uint64_t* __restrict__ chunk_out) {
+  __shared__ uint64_t sval[256], scnt[256], scal[256];
+  const unsigned t = threadIdx.x;
+  const uint64_t nch = gridDim.x;
+  const uint64_t pl = (n_partials + nch - 1) / nch;
+  const uint64_t plo = blockIdx.x * pl;
+  const uint64_t phi = plo + pl < n_partials ? plo + pl : n_partials;
+  const uint64_t hl = (n_heads + nch - 1) / nch;
+  const uint64_t hlo = blockIdx.x * hl;
+  const uint64_t hhi = hlo + hl < n_heads ? hlo + hl : n_heads;
+
+  for (int s = 0; s < 4; ++s) {
+    uint64_t acc = 0;
+    for (uint64_t i = plo + t; i < phi; i += 256)
+      acc += partials[i * kPartialStride + s];
+    if (s == 1 || s == 2) {
+      for (uint64_t i = hlo + t; i < hhi; i += 256)
+        if (cont_flags[i] == 0)
+          acc += heads[i * kHeadStride + 2 * YBG_MAX_AGGS + (s - 1)];
+    }
+    scal[t] = acc;
+    __syncthreads();
+    if (t == 0) {
+      uint64_t total = 0;
+      for (int i = 0; i < 256; ++i) total += scal[i];
+      chunk_out[blockIdx.x * kPartialStride + s] = total;
+    }
+    __syncthreads();
+  }
+  for (int g = 0; g < sp.num_aggs; ++g) {
+    int op = sp.agg_op[g];
+    uint64_t av = 0, ac = 0;
+    for (uint64_t i = plo + t; i < phi; i += 256)
+      combine1(op, &av, &ac, partials[i * kPartialStride + 4 + 2 * g],
+               partials[i * kPartialStride + 4 + 2 * g + 1]);
+    for (uint64_t i = hlo + t; i < hhi; i += 256) {
+      if (cont_flags[i] != 0) continue;
+      combine1(op, &av, &ac, heads[i * kHeadStride + 2 * g],
+               heads[i * kHeadStride + 2 * g + 1]);
+    }
+    sval[t] = av;
+    scnt[t] = ac;
+    __syncthreads();
+    if (t == 0) {
+      uint64_t fv = 0, fc = 0;
+      for (int i = 0; i < 256; ++i) combine1(op, &fv, &fc, sval[i], scnt[i]);
+      chunk_out[blockIdx.x * kPartialStride + 4 + 2 * g] = fv;
+      chunk_out[blockIdx.x * kPartialStride + 4 + 2 * g + 1] = fc;
+    }
+    __syncthreads();
+  }
+}
+
 __global__ __launch_bounds__(256) void k_reduce(
     DevSpec sp, const uint64_t* __restrict__ partials, uint64_t n_partials,
     const uint64_t* __restrict__ heads, const uint32_t* __restrict__ cont_flags,
@@ -422,6 +483,7 @@ struct ybg_scan {
   uint32_t* d_cont = nullptr;
   uint64_t n_heads = 0;
   DevResult* d_result = nullptr;
+  uint64_t* d_chunk = nullptr;  // k_reduce_pre output (256 partial records)
   int grid = 0;
   bool executed = false;
   double last_total_ms = 0, last_decode_ms = 0;
@@ -574,6 +636,7 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
   HIP_TRY(hipMalloc(&s->d_heads, s->n_heads * kHeadStride * sizeof(uint64_t)));
   HIP_TRY(hipMalloc(&s->d_cont, s->n_heads * sizeof(uint32_t)));
   HIP_TRY(hipMalloc(&s->d_result, sizeof(DevResult)));
+  HIP_TRY(hipMalloc(&s->d_chunk, 256 * kPartialStride * sizeof(uint64_t)));
   return 0;
 }
 
@@ -622,9 +685,12 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
     default: launch(k_scan<2, 4>); break;
   }
   HIP_TRY(hipEventRecord(s->ev_mid, s->stream));
+  hipLaunchKernelGGL(k_reduce_pre, dim3(256), dim3(256), 0, s->stream,
+                     s->dspec, s->d_partials, s->n_partials, s->d_heads,
+                     s->d_cont, s->n_heads, s->d_chunk);
   hipLaunchKernelGGL(k_reduce, dim3(1), dim3(256), 0, s->stream, s->dspec,
-                     s->d_partials, s->n_partials, s->d_heads, s->d_cont,
-                     s->n_heads, s->d_result);
+                     s->d_chunk, 256, s->d_heads, s->d_cont, 0,
+                     s->d_result);
   HIP_TRY(hipEventRecord(s->ev_end, s->stream));
   s->executed = true;
   return 0;
@@ -909,6 +975,7 @@ int yb_gpu_scan_close(ybg_scan_t* s) {
   if (s->d_heads) HIP_WARN(hipFree(s->d_heads));
   if (s->d_cont) HIP_WARN(hipFree(s->d_cont));
   if (s->d_result) HIP_WARN(hipFree(s->d_result));
+  if (s->d_chunk) HIP_WARN(hipFree(s->d_chunk));
   if (s->d_flags_all) HIP_WARN(hipFree(s->d_flags_all));
   if (s->d_em_sort) HIP_WARN(hipFree(s->d_em_sort));
   if (s->d_em_key) HIP_WARN(hipFree(s->d_em_key));
