@@ -1583,7 +1583,11 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
         if (sp.v2_fixed_len && value_len == sp.v2_fixed_len &&
             (rdr->peek8() & 0xff8000u) == 0) {
           // all-fixed schema, no null mask, 1-byte version: direct
-          // fixed-offset column loads, no window-consume chain
+          // fixed-offset column loads, no window-consume chain. Reposition
+          // the reader FIRST: the next entry's window loads issue here and
+          // land while the columns below decode (they are otherwise the
+          // first thing the next iteration stalls on).
+          rdr->seek(value + value_len);
           decode_packed_v2_fixed(sp, aux, value, rc);
           rc->found = true;
           return true;

@@ -566,9 +566,10 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
     s->d_data = const_cast<uint8_t*>(blocks);
     s->d_data_owned = false;
   } else {
-    // +16 bytes tail slack for windowed loads (scan_device.h load_u64_una)
-    HIP_TRY(hipMalloc(&s->d_data, s->total_bytes + 16));
-    HIP_TRY(hipMemset(s->d_data + s->total_bytes, 0, 16));
+    // 48 bytes tail slack: Rdr window init can read up to 32 bytes past
+    // the position and load_u64_una up to 11 (scan_device.h contract)
+    HIP_TRY(hipMalloc(&s->d_data, s->total_bytes + 48));
+    HIP_TRY(hipMemset(s->d_data + s->total_bytes, 0, 48));
     HIP_TRY(hipMemcpy(s->d_data, blocks, s->total_bytes,
                       hipMemcpyHostToDevice));
     s->d_data_owned = true;

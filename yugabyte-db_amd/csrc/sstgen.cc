@@ -508,12 +508,12 @@ int ybg_builder_finish(ybg_builder_t* b, const uint8_t** data,
                        const uint64_t** offsets, uint64_t* n_blocks,
                        uint64_t* total_bytes, uint64_t* n_entries) {
   b->FlushBlock();
-  // 16 bytes of tail slack: the scan path's windowed loads
-  // (scan_device.h load_u64_una) may read past the last block.
+  // 48 bytes of tail slack: the scan path's window init can read up to
+  // 32 bytes past the position (scan_device.h Rdr contract).
   uint64_t sz = b->all_blocks.size();
-  b->all_blocks.resize(sz + 16, 0);
+  b->all_blocks.resize(sz + 48, 0);
   b->all_blocks.resize(sz);
-  b->all_blocks.reserve(sz + 16);
+  b->all_blocks.reserve(sz + 48);
   *data = b->all_blocks.data();
   *offsets = b->offsets.data();
   *n_blocks = b->offsets.size() - 1;
@@ -659,8 +659,8 @@ int ybg_generate(const ybg_schema_t* schema, const ybg_gen_params_t* p,
     nb += nblk;
     ne += ent;
   }
-  // +16 bytes tail slack for the scan path's windowed loads
-  uint8_t* out = (uint8_t*)calloc(1, (total ? total : 1) + 16);
+  // +48 bytes tail slack for the scan path's windowed loads
+  uint8_t* out = (uint8_t*)calloc(1, (total ? total : 1) + 48);
   uint64_t* out_off = (uint64_t*)malloc((nb + 1) * sizeof(uint64_t));
   uint64_t pos = 0, bi = 0;
   out_off[0] = 0;
