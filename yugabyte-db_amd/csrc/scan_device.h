@@ -653,7 +653,28 @@ DEV const uint8_t* decode_entry_ptr(int fmt, const uint8_t* p,
   if (!shared_something) {
     // restart / full key inline (block.cc:311-317)
     if (ns1 > kKeyCap || ns1 < 9) return nullptr;
-    for (uint32_t i = 0; i < ns1; ++i) wr(i, p[i]);
+    if (ns1 <= kKeyCap - 8) {
+      // u64-chunk copy (key slots are 8-aligned, cap leaves write slack);
+      // row-change compare runs over the first rkb bytes up front
+      if (rkb) {
+        bool ch = false;
+        uint32_t full = rkb & ~7u;
+        for (uint32_t i = 0; i < full; i += 8)
+          if (load_u64_una(&key[i]) != load_u64_una(p + i)) ch = true;
+        if (rkb & 7) {
+          uint64_t m = (1ull << (8 * (rkb & 7))) - 1;
+          if (((load_u64_una(&key[full]) ^ load_u64_una(p + full)) & m) != 0)
+            ch = true;
+        }
+        if (ch) *changed = true;
+      }
+      for (uint32_t i = 0; i < ns1; i += 8) {
+        uint64_t w8 = load_u64_una(p + i);
+        __builtin_memcpy(&key[i], &w8, 8);
+      }
+    } else {
+      for (uint32_t i = 0; i < ns1; ++i) wr(i, p[i]);
+    }
     *key_len = ns1;
     *last8 = load_u64_una(key + ns1 - 8);
     out->value = p + ns1;

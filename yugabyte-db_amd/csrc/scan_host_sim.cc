@@ -49,7 +49,8 @@ int ybg_sim_scan(const ybg_scan_spec_t* spec, const uint8_t* data,
   uint64_t agg_val[YBG_MAX_AGGS] = {0}, agg_cnt[YBG_MAX_AGGS] = {0};
   std::vector<HeadOut<YBG_MAX_AGGS>> heads(n_ivs);
   std::vector<uint8_t> walked(n_ivs, 0);
-  uint8_t key[kKeyCap], rk_save[kKeyCap];
+  alignas(8) uint8_t key[kKeyCap];
+  alignas(8) uint8_t rk_save[kKeyCap];
   for (uint64_t j = 0; j < n_ivs; ++j) {
     bool wn = false;
     if (!scan_one_interval<YBG_MAX_AGGS>(d, data, offsets, ivs.data(), n_ivs,
@@ -128,7 +129,8 @@ int ybg_sim_emit(const ybg_scan_spec_t* spec, const uint8_t* data,
   uint64_t agg_val[YBG_MAX_AGGS] = {0}, agg_cnt[YBG_MAX_AGGS] = {0};
   std::vector<uint8_t> walked(n_ivs, 0);
   std::vector<uint32_t> head_consumed(n_ivs ? n_ivs : 1, 0);
-  uint8_t key[kKeyCap], rk_save[kKeyCap];
+  alignas(8) uint8_t key[kKeyCap];
+  alignas(8) uint8_t rk_save[kKeyCap];
   HeadOut<YBG_MAX_AGGS> ho;
   for (uint64_t j = 0; j < n_ivs; ++j) {
     bool wn = false;
@@ -208,7 +210,8 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
   uint64_t agg_val[YBG_MAX_AGGS] = {0}, agg_cnt[YBG_MAX_AGGS] = {0};
   std::vector<uint8_t> walked(n_ivs, 0);
   std::vector<uint32_t> head_flags(n_ivs ? n_ivs : 1, 0);
-  uint8_t key[kKeyCap], rk_save[kKeyCap];
+  alignas(8) uint8_t key[kKeyCap];
+  alignas(8) uint8_t rk_save[kKeyCap];
   for (uint64_t j = 0; j < n_ivs; ++j) {
     bool wn = false;
     HeadOut<YBG_MAX_AGGS> ho;
