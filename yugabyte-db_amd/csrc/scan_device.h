@@ -402,6 +402,31 @@ DEV void slice_u128(PtrT p, uint32_t len, uint64_t* hi, uint64_t* lo) {
   *lo = l;
 }
 
+// Register-resident user-key tail: big-endian bytes
+// [ukey_len-16, ukey_len) as (thi, tlo) — thi = bytes [-16,-8), tlo =
+// [-8,0). The DocHybridTime suffix, its size byte and the kHybridTime
+// marker all live in this window for every key the benchmark configs
+// produce, so per-entry visibility runs on registers; the fast decode path
+// patches the window in place and every other path invalidates it (rebuilt
+// from LDS once per invalidation). Requires ukey_len >= 16.
+DEV void tail_from_lds(const uint8_t* key, uint32_t ukey_len, uint64_t* thi,
+                       uint64_t* tlo) {
+  *thi = __builtin_bswap64(load_u64_una(key + ukey_len - 16));
+  *tlo = __builtin_bswap64(load_u64_una(key + ukey_len - 8));
+}
+
+// patch byte at absolute key position pos (window-relative guard inside)
+DEV void tail_patch(uint64_t* thi, uint64_t* tlo, uint32_t ukey_len,
+                    uint32_t pos, uint8_t b) {
+  int32_t off = (int32_t)pos - (int32_t)(ukey_len - 16);
+  if (off < 0) return;
+  uint32_t sh = 8 * (7 - ((uint32_t)off & 7));
+  if (off < 8)
+    *thi = (*thi & ~(0xffull << sh)) | ((uint64_t)b << sh);
+  else
+    *tlo = (*tlo & ~(0xffull << sh)) | ((uint64_t)b << sh);
+}
+
 // memcmp + length tiebreak over zero-padded 16-byte slices (Slice::compare)
 DEV int u128_slice_cmp(uint64_t ahi, uint64_t alo, uint32_t alen,
                        uint64_t bhi, uint64_t blo, uint32_t blen) {
@@ -753,7 +778,8 @@ DEV const uint8_t* decode_entry_ptr(int fmt, const uint8_t* p,
 DEV const uint8_t* decode_entry(int fmt, Rdr* rdr, const uint8_t* limit,
                                 uint8_t* key, uint32_t* key_len,
                                 uint64_t* last8, uint32_t rkb, bool* changed,
-                                EntryRef* out) {
+                                EntryRef* out, uint64_t* thi, uint64_t* tlo,
+                                bool* tail_valid) {
   const uint8_t* p = rdr->pos();
   if (fmt == YBG_ENC_THREE_SHARED_PARTS && limit - p >= 8) {
     uint64_t w = rdr->peek8();
@@ -823,17 +849,28 @@ DEV const uint8_t* decode_entry(int fmt, Rdr* rdr, const uint8_t* limit,
             // untouched. (The general decode path below keeps the
             // compare-on-write detection and covers restart entries.)
             if (sp < rkb) *changed = true;
+            // register tail maintenance: d2 lengthens the key by one byte
+            // (window slides by one); the written bytes are patched below
+            const uint32_t new_ukey = (uint32_t)new_len - 8;
+            if (d2) {
+              *thi = (*thi << 8) | (*tlo >> 56);
+              *tlo <<= 8;
+            }
             // key bytes [hl, hl+ns1+ns2) — within the first 16 window bytes
             uint64_t w2 = rdr->peek8_at(8);
             for (uint32_t i = 0; i < ns1; ++i) {
               uint32_t j = hl + i;
               uint64_t src = j < 8 ? w : w2;
-              key[sp + i] = (uint8_t)(src >> (8 * (j & 7)));
+              uint8_t b = (uint8_t)(src >> (8 * (j & 7)));
+              key[sp + i] = b;
+              tail_patch(thi, tlo, new_ukey, sp + i, b);
             }
             for (uint32_t i = 0; i < ns2; ++i) {
               uint32_t j = hl + ns1 + i;
               uint64_t src = j < 8 ? w : w2;
-              key[new_ns2_start + i] = (uint8_t)(src >> (8 * (j & 7)));
+              uint8_t b = (uint8_t)(src >> (8 * (j & 7)));
+              key[new_ns2_start + i] = b;
+              tail_patch(thi, tlo, new_ukey, new_ns2_start + i, b);
             }
             *last8 += inc;
             *key_len = (uint32_t)new_len;
@@ -846,10 +883,11 @@ DEV const uint8_t* decode_entry(int fmt, Rdr* rdr, const uint8_t* limit,
       }
     }
   }
-  // general path
+  // general path: the register tail no longer mirrors LDS
   const uint8_t* q =
       decode_entry_ptr(fmt, p, limit, key, key_len, last8, rkb, changed, out);
   if (!q) return nullptr;
+  *tail_valid = false;
   rdr->seek(out->value);
   return q;
 }
@@ -1561,14 +1599,10 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
                        const uint8_t* aux, const uint8_t* key,
                        uint32_t key_len, const uint8_t* value,
                        uint32_t value_len, uint32_t rowkey_len,
-                       RowCtxT<NA>* rc, Rdr* rdr) {
+                       RowCtxT<NA>* rc, Rdr* rdr, uint32_t ht_size,
+                       uint64_t ht_hi, uint64_t ht_lo) {
   uint32_t ukey_len = key_len - 8;
-  uint32_t ht_size = key[ukey_len - 1] & 0x1f;
-  const uint8_t* ht_enc = key + ukey_len - ht_size;
   uint32_t prefix_len = ukey_len - ht_size - 1;
-
-  uint64_t ht_hi, ht_lo;
-  slice_u128(ht_enc, ht_size, &ht_hi, &ht_lo);
   uint32_t vb0 = value_len > 0 ? (uint32_t)(rdr->peek8() & 0xff) : 0u;
   bool visible;
   if (value_len > 0 && vb0 == kHybridTimeByte) {
@@ -1932,6 +1966,8 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
 
   uint32_t key_len = 0;
   uint64_t reg_last8 = 0;
+  uint64_t thi = 0, tlo = 0;  // register user-key tail (see tail_from_lds)
+  bool tail_ok = false;
   uint32_t rk_len = 0;
   bool row_open = false;
   bool in_head = true;
@@ -1962,12 +1998,43 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
     }
     bool kchg = false;
     const uint8_t* q = decode_entry(sp.fmt, &rdr, limit, key, &key_len,
-                                    &reg_last8, rkb, &kchg, &er);
+                                    &reg_last8, rkb, &kchg, &er, &thi, &tlo,
+                                    &tail_ok);
     if (!q || key_len < 10) { fail = true; break; }
     uint32_t ukey_len = key_len - 8;
-    uint32_t ht_sz = key[ukey_len - 1] & 0x1f;
-    if (ht_sz == 0 || ukey_len < ht_sz + 2 ||
-        key[ukey_len - ht_sz - 1] != kHybridTimeByte) { fail = true; break; }
+    uint32_t ht_sz;
+    uint64_t ht_hi, ht_lo;
+    if (YBG_LIKELY(ukey_len >= 16)) {
+      if (!tail_ok) {
+        tail_from_lds(key, ukey_len, &thi, &tlo);
+        tail_ok = true;
+      }
+      ht_sz = (uint32_t)tlo & 0x1f;
+      uint32_t mb = ht_sz + 1;  // marker distance from the key end
+      uint32_t marker;
+      if (mb <= 8) marker = (uint32_t)(tlo >> (8 * (mb - 1))) & 0xff;
+      else if (mb <= 16) marker = (uint32_t)(thi >> (8 * (mb - 9))) & 0xff;
+      else marker = key[ukey_len - mb];  // ht_sz == 16 only
+      if (ht_sz == 0 || ukey_len < ht_sz + 2 ||
+          marker != kHybridTimeByte) { fail = true; break; }
+      // visibility slice = (thi:tlo) << 8*(16-ht_sz): the top ht_sz bytes
+      // are the encoded DocHybridTime, the shifted-in bits are zero — the
+      // exact zero-padded slice_u128 layout
+      uint32_t s = 8 * (16 - ht_sz);
+      if (s == 0) { ht_hi = thi; ht_lo = tlo; }
+      else if (s < 64) {
+        ht_hi = (thi << s) | (tlo >> (64 - s));
+        ht_lo = tlo << s;
+      } else {
+        ht_hi = tlo << (s - 64);
+        ht_lo = 0;
+      }
+    } else {
+      ht_sz = key[ukey_len - 1] & 0x1f;
+      if (ht_sz == 0 || ukey_len < ht_sz + 2 ||
+          key[ukey_len - ht_sz - 1] != kHybridTimeByte) { fail = true; break; }
+      slice_u128(key + ukey_len - ht_sz, ht_sz, &ht_hi, &ht_lo);
+    }
     uint32_t prefix_len = ukey_len - ht_sz - 1;
     uint32_t rk;
     if (sp.fixed_rk_len) {
@@ -2039,7 +2106,10 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
     if (cur_iv == j) *entries += 1;
     else if (cur_iv == j + 1) walked_next = true;
     if (!process_entry(sp, data, aux, key, key_len, er.value, er.value_len,
-                       rk_len, &rc, &rdr)) { fail = true; break; }
+                       rk_len, &rc, &rdr, ht_sz, ht_hi, ht_lo)) {
+      fail = true;
+      break;
+    }
     rdr.seek(q);
     p = q;
   }
