@@ -415,16 +415,16 @@ DEV void tail_from_lds(const uint8_t* key, uint32_t ukey_len, uint64_t* thi,
   *tlo = __builtin_bswap64(load_u64_una(key + ukey_len - 8));
 }
 
-// patch byte at absolute key position pos (window-relative guard inside)
+// patch byte at absolute key position pos (window-relative, branchless)
 DEV void tail_patch(uint64_t* thi, uint64_t* tlo, uint32_t ukey_len,
                     uint32_t pos, uint8_t b) {
   int32_t off = (int32_t)pos - (int32_t)(ukey_len - 16);
-  if (off < 0) return;
   uint32_t sh = 8 * (7 - ((uint32_t)off & 7));
-  if (off < 8)
-    *thi = (*thi & ~(0xffull << sh)) | ((uint64_t)b << sh);
-  else
-    *tlo = (*tlo & ~(0xffull << sh)) | ((uint64_t)b << sh);
+  uint64_t m = off >= 0 ? (0xffull << sh) : 0;
+  uint64_t v = (uint64_t)b << sh;
+  bool hi = off < 8;
+  *thi = (*thi & ~(hi ? m : 0)) | (hi ? (v & m) : 0);
+  *tlo = (*tlo & ~(hi ? 0 : m)) | (hi ? 0 : (v & m));
 }
 
 // memcmp + length tiebreak over zero-padded 16-byte slices (Slice::compare)
@@ -2103,8 +2103,8 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
         row_sort_key = ((uint64_t)cur_iv << 16) | (uint64_t)(p - blk);
       row_reset(&rc, sp);
     }
-    if (cur_iv == j) *entries += 1;
-    else if (cur_iv == j + 1) walked_next = true;
+    *entries += (cur_iv == j);
+    walked_next |= (cur_iv == j + 1);
     if (!process_entry(sp, data, aux, key, key_len, er.value, er.value_len,
                        rk_len, &rc, &rdr, ht_sz, ht_hi, ht_lo)) {
       fail = true;
