@@ -621,7 +621,7 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
   HIP_WARN(hipFree(d_err));
 
   uint64_t want = (s->n_ivs + kThreads - 1) / kThreads;
-  uint64_t cap = 4096;
+  uint64_t cap = 8192;
   if (const char* g = getenv("YBG_GRID")) {
     long v = atol(g);
     if (v > 0) cap = (uint64_t)v;
