@@ -1729,7 +1729,7 @@ template <int NA>
 struct HeadOut {
   uint64_t val[NA];
   uint64_t cnt[NA];
-  uint64_t scanned, matched;
+  uint32_t scanned, matched;
 };
 
 // ---------------------------------------------------------------------------
@@ -1940,8 +1940,8 @@ template <int NA, bool EMIT = false, bool GROUP = false>
 DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
                            const uint64_t* block_offsets, const Interval* ivs,
                            uint64_t n_ivs, uint64_t j, const uint8_t* aux,
-                           uint8_t* key, uint8_t* rk_save, uint64_t* entries,
-                           uint64_t* scanned, uint64_t* matched,
+                           uint8_t* key, uint8_t* rk_save, uint32_t* entries,
+                           uint32_t* scanned, uint32_t* matched,
                            uint64_t* agg_val, uint64_t* agg_cnt,
                            HeadOut<NA>* ho, bool* walked_next_out,
                            EmitCtx* ec = nullptr,

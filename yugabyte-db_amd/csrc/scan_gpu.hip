@@ -73,7 +73,7 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
   const uint64_t span = (uint64_t)gridDim.x * kThreads;
 
-  uint64_t entries = 0, scanned = 0, matched = 0, errs = 0;
+  uint32_t entries = 0, scanned = 0, matched = 0, errs = 0;
   // NA-sized and only ever constant-indexed (all loops over them unrolled):
   // any runtime index would force these accumulators into scratch memory,
   // and with them every per-row aggregate update in the hot loop (measured
@@ -208,7 +208,7 @@ __global__ __launch_bounds__(kEmitThreads) void k_emit(
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
   const uint64_t span = (uint64_t)gridDim.x * kEmitThreads;
 
-  uint64_t entries = 0, scanned = 0, matched = 0;
+  uint32_t entries = 0, scanned = 0, matched = 0;
   uint64_t agg_val[2] = {0, 0}, agg_cnt[2] = {0, 0};
   for (uint64_t j = gtid; j < n_ivs; j += span) {
     HeadOut<2> ho;
@@ -258,7 +258,7 @@ __global__ __launch_bounds__(kEmitThreads) void k_group(
   const uint32_t gtid = blockIdx.x * kEmitThreads + threadIdx.x;
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
   const uint64_t span = (uint64_t)gridDim.x * kEmitThreads;
-  uint64_t entries = 0, scanned = 0, matched = 0;
+  uint32_t entries = 0, scanned = 0, matched = 0;
   uint64_t agg_val[NA] = {0}, agg_cnt[NA] = {0};
   for (uint64_t j = gtid; j < n_ivs; j += span) {
     HeadOut<NA> ho;

@@ -53,11 +53,15 @@ int ybg_sim_scan(const ybg_scan_spec_t* spec, const uint8_t* data,
   alignas(8) uint8_t rk_save[kKeyCap];
   for (uint64_t j = 0; j < n_ivs; ++j) {
     bool wn = false;
+    uint32_t e32 = 0, s32 = 0, m32 = 0;
     if (!scan_one_interval<YBG_MAX_AGGS>(d, data, offsets, ivs.data(), n_ivs,
                                          j, aux.data(), key, rk_save,
-                                         &entries, &scanned, &matched,
+                                         &e32, &s32, &m32,
                                          agg_val, agg_cnt, &heads[j], &wn))
       return 6;
+    entries += e32;
+    scanned += s32;
+    matched += m32;
     walked[j] = wn ? 1 : 0;
   }
   // head ownership resolution (shfl relay / cont flags, serial equivalent)
@@ -125,7 +129,7 @@ int ybg_sim_emit(const ybg_scan_spec_t* spec, const uint8_t* data,
     }
   }
   uint64_t n_ivs = ivs.size();
-  uint64_t entries = 0, scanned = 0, matched = 0;
+  uint32_t entries = 0, scanned = 0, matched = 0;
   uint64_t agg_val[YBG_MAX_AGGS] = {0}, agg_cnt[YBG_MAX_AGGS] = {0};
   std::vector<uint8_t> walked(n_ivs, 0);
   std::vector<uint32_t> head_consumed(n_ivs ? n_ivs : 1, 0);
@@ -206,7 +210,7 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
     }
   }
   uint64_t n_ivs = ivs.size();
-  uint64_t entries = 0, scanned = 0, matched = 0;
+  uint32_t entries = 0, scanned = 0, matched = 0;
   uint64_t agg_val[YBG_MAX_AGGS] = {0}, agg_cnt[YBG_MAX_AGGS] = {0};
   std::vector<uint8_t> walked(n_ivs, 0);
   std::vector<uint32_t> head_flags(n_ivs ? n_ivs : 1, 0);
