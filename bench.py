@@ -244,11 +244,13 @@ def main():
             cal = json.load(open(os.path.join(
                 os.path.dirname(os.path.abspath(__file__)), "profiles",
                 "traffic_calibration.json")))
-            if (cal.get("workload") == args.workload
-                    and cal.get("rows") == args.rows):
-                traffic = float(cal["traffic_bytes_per_launch"])
-            else:
-                traffic = None
+            entries = cal if isinstance(cal, list) else [cal]
+            traffic = None
+            for c in entries:
+                if (c.get("workload") == args.workload
+                        and c.get("rows") == args.rows):
+                    traffic = float(c["traffic_bytes_per_launch"])
+                    break
         except Exception:
             traffic = None
 
