@@ -193,11 +193,12 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
 constexpr int kEmitThreads = 128;
 
 __global__ __launch_bounds__(kEmitThreads) void k_emit(
-    DevSpec sp, const uint8_t* __restrict__ data,
+    const uint8_t* __restrict__ data,
     const uint64_t* __restrict__ block_offsets,
     const Interval* __restrict__ ivs, uint64_t n_ivs,
     const uint8_t* __restrict__ aux, uint8_t* __restrict__ rk_save_buf,
     EmitCtx ec, unsigned long long* __restrict__ err_counter) {
+  const DevSpec& sp = c_spec;
   __shared__ uint8_t key_scratch[kEmitThreads * kKeyCap];
   __shared__ uint64_t rowbuf[kEmitThreads * YBG_MAX_COLS];
   __shared__ uint32_t lenbuf[kEmitThreads * YBG_MAX_COLS];
@@ -247,12 +248,13 @@ __global__ void k_group_init(DevSpec sp, GroupCtx gc) {
 
 template <int NA>
 __global__ __launch_bounds__(kEmitThreads) void k_group(
-    DevSpec sp, const uint8_t* __restrict__ data,
+    const uint8_t* __restrict__ data,
     const uint64_t* __restrict__ block_offsets,
     const Interval* __restrict__ ivs, uint64_t n_ivs,
     const uint8_t* __restrict__ aux, uint8_t* __restrict__ rk_save_buf,
     GroupCtx gc, const uint32_t* __restrict__ head_flags,
     unsigned long long* __restrict__ err_counter) {
+  const DevSpec& sp = c_spec;
   __shared__ uint8_t key_scratch[kEmitThreads * kKeyCap];
   uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
   const uint32_t gtid = blockIdx.x * kEmitThreads + threadIdx.x;
@@ -798,7 +800,7 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
     rk_area = rk_extra;
   }
   hipLaunchKernelGGL(k_emit, dim3(egrid), dim3(kEmitThreads), 0, s->stream,
-                     s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
+                     s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
                      s->d_aux, rk_area, ec, s->d_em_counters + 3);
   unsigned long long ctr[4];
   HIP_TRY(hipMemcpyAsync(ctr, s->d_em_counters,
@@ -904,13 +906,13 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
   if (s->dspec.num_aggs <= 4) {
     auto kg = k_group<4>;
     hipLaunchKernelGGL(kg, dim3(egrid), dim3(kEmitThreads), 0, s->stream,
-                       s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
+                       s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
                        s->d_aux, rk_area, s->gc, s->d_flags_all,
                        s->gc.overflow);
   } else {
     auto kg = k_group<8>;
     hipLaunchKernelGGL(kg, dim3(egrid), dim3(kEmitThreads), 0, s->stream,
-                       s->dspec, s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
+                       s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
                        s->d_aux, rk_area, s->gc, s->d_flags_all,
                        s->gc.overflow);
   }
