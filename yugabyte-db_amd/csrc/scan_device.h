@@ -1799,7 +1799,10 @@ DEV uint64_t grp_find_or_insert(const GroupCtx& gc, uint64_t kv,
       }
     }
     if (st == 1) {
-      st = ybg_atomic_load_acq_u32(&gc.state[i]);
+      // no acquire reload: gkey is read with a device-scope atomic load
+      // (coherence-point direct) and the branch on st==1 orders it after
+      // the state observation; the per-row L1 invalidate an acquire load
+      // implies costs more than the whole probe
       uint64_t k2 = ybg_atomic_load_u64(&gc.gkey[i]);
       bool match;
       if (is_str) {
