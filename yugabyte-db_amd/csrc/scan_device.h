@@ -1824,6 +1824,19 @@ DEV uint64_t grp_find_or_insert(const GroupCtx& gc, uint64_t kv,
   return ~0ull;
 }
 
+// Deferred head-row record for the single-pass GROUP kernel: the head
+// row's raw operands (ownership unknown until the lane relay / cont-flag
+// resolution). agg_datum is fixed-size for a stable memory layout; the
+// unrolled NA loops keep register copies constant-indexed.
+struct GroupHead {
+  uint64_t grp_datum;
+  uint64_t agg_datum[YBG_MAX_AGGS];
+  uint32_t grp_len;
+  uint32_t agg_null;
+  uint32_t grp_null;
+  uint32_t hit;
+};
+
 template <int NA>
 DEV void group_accum(const DevSpec& sp, const GroupCtx& gc,
                      const RowCtxT<NA>& rc) {
@@ -1873,6 +1886,62 @@ DEV void group_accum(const DevSpec& sp, const GroupCtx& gc,
         break;
       default:
         break;  // MIN/MAX double unsupported in group mode (open() rejects)
+    }
+    atomicAdd(&c[g], 1ull);
+  }
+}
+
+// Same accumulate from a deferred head record (the heads-resolution pass
+// and the in-kernel not-consumed head path).
+template <int NA>
+DEV void group_accum_rec(const DevSpec& sp, const GroupCtx& gc,
+                         const GroupHead& r) {
+  uint64_t slot;
+  if (r.grp_null) {
+    slot = gc.cap;
+    if (ybg_atomic_load_u32(&gc.state[slot]) != 1)
+      ybg_atomic_store_rel_u32(&gc.state[slot], 1u);
+  } else {
+    bool is_str = r.grp_len != 0;
+    uint64_t kv = is_str ? (((uint64_t)r.grp_len << 40) |
+                            (r.grp_datum - (uint64_t)(uintptr_t)gc.data))
+                         : r.grp_datum;
+    slot = grp_find_or_insert(gc, kv, is_str);
+    if (slot == ~0ull) {
+      atomicAdd(gc.overflow, 1ull);
+      return;
+    }
+  }
+  long long* v = gc.vals + slot * YBG_MAX_AGGS;
+  unsigned long long* c = gc.cnts + slot * YBG_MAX_AGGS;
+#pragma unroll
+  for (int g = 0; g < NA; ++g) {
+    if (g >= sp.num_aggs) continue;
+    int op = sp.aggs[g].op;
+    bool isnull =
+        (op == YBG_AGG_COUNT_STAR) ? false : ((r.agg_null >> g) & 1);
+    if (isnull) continue;
+    uint64_t d = r.agg_datum[g];
+    switch (op) {
+      case YBG_AGG_COUNT_STAR:
+      case YBG_AGG_COUNT:
+        atomicAdd((unsigned long long*)&v[g], 1ull);
+        break;
+      case YBG_AGG_SUM_INT64:
+        atomicAdd((unsigned long long*)&v[g], d);
+        break;
+      case YBG_AGG_SUM_DOUBLE:
+        ybg_atomic_add_f64((double*)&v[g],
+                           __longlong_as_double((long long)d));
+        break;
+      case YBG_AGG_MIN_INT64:
+        ybg_atomic_min_i64(&v[g], (long long)d);
+        break;
+      case YBG_AGG_MAX_INT64:
+        ybg_atomic_max_i64(&v[g], (long long)d);
+        break;
+      default:
+        break;
     }
     atomicAdd(&c[g], 1ull);
   }
@@ -1951,7 +2020,8 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
                            uint64_t* emit_datums = nullptr,
                            uint32_t* emit_lens = nullptr,
                            const GroupCtx* gc = nullptr,
-                           const uint32_t* head_flags = nullptr) {
+                           const uint32_t* head_flags = nullptr,
+                           GroupHead* gh_out = nullptr) {
   Interval iv = ivs[j];
   const uint8_t* blk = data + block_offsets[iv.block];
   const uint8_t* p = blk + iv.start;
@@ -2089,8 +2159,23 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
           if (hit) {
             if (EMIT && (!in_head || ec->head_consumed[j] == 0))
               emit_row(sp, ec, rc, rkp, rk_len, row_sort_key);
-            if (GROUP && (!in_head || head_flags[j] == 0))
-              group_accum(sp, *gc, rc);
+            if (GROUP) {
+              if (!in_head) {
+                group_accum(sp, *gc, rc);
+              } else if (gh_out) {
+                // defer the head row (ownership unknown): raw operands
+                gh_out->grp_datum = rc.grp_datum;
+                gh_out->grp_len = rc.grp_len;
+                gh_out->grp_null = rc.grp_null ? 1u : 0u;
+                gh_out->agg_null = rc.agg_null;
+#pragma unroll
+                for (int g = 0; g < NA; ++g)
+                  gh_out->agg_datum[g] = rc.agg_datum[g];
+                gh_out->hit = 1;
+              } else if (head_flags && head_flags[j] == 0) {
+                group_accum(sp, *gc, rc);
+              }
+            }
           }
         }
         in_head = false;
@@ -2136,8 +2221,22 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
     if (hit) {
       if (EMIT && (!in_head || ec->head_consumed[j] == 0))
         emit_row(sp, ec, rc, rkp_end, rk_len, row_sort_key);
-      if (GROUP && (!in_head || head_flags[j] == 0))
-        group_accum(sp, *gc, rc);
+      if (GROUP) {
+        if (!in_head) {
+          group_accum(sp, *gc, rc);
+        } else if (gh_out) {
+          gh_out->grp_datum = rc.grp_datum;
+          gh_out->grp_len = rc.grp_len;
+          gh_out->grp_null = rc.grp_null ? 1u : 0u;
+          gh_out->agg_null = rc.agg_null;
+#pragma unroll
+          for (int g = 0; g < NA; ++g)
+            gh_out->agg_datum[g] = rc.agg_datum[g];
+          gh_out->hit = 1;
+        } else if (head_flags && head_flags[j] == 0) {
+          group_accum(sp, *gc, rc);
+        }
+      }
     }
   }
   *walked_next_out = walked_next;

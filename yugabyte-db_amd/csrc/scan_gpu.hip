@@ -246,32 +246,85 @@ __global__ void k_group_init(DevSpec sp, GroupCtx gc) {
   }
 }
 
+// Single-pass GROUP BY: same head-deferral protocol as k_scan — no flags
+// pre-pass. Non-head rows accumulate into the device hash table directly;
+// each interval's head row is deferred as a GroupHead record, applied
+// in-kernel once the lane relay resolves ownership, with workgroup-first
+// intervals going through the global record + cont-flag path
+// (k_group_heads folds those).
 template <int NA>
-__global__ __launch_bounds__(kEmitThreads) void k_group(
+__global__ __launch_bounds__(kThreads, 3) void k_group(
     const uint8_t* __restrict__ data,
     const uint64_t* __restrict__ block_offsets,
     const Interval* __restrict__ ivs, uint64_t n_ivs,
     const uint8_t* __restrict__ aux, uint8_t* __restrict__ rk_save_buf,
-    GroupCtx gc, const uint32_t* __restrict__ head_flags,
+    GroupCtx gc, GroupHead* __restrict__ gheads,
+    uint32_t* __restrict__ cont_flags,
     unsigned long long* __restrict__ err_counter) {
   const DevSpec& sp = c_spec;
-  __shared__ uint8_t key_scratch[kEmitThreads * kKeyCap];
+  __shared__ uint8_t key_scratch[kThreads * kKeyCap];
+  __shared__ uint8_t wave_relay[kThreads / 64 + 1];
   uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
-  const uint32_t gtid = blockIdx.x * kEmitThreads + threadIdx.x;
+  const uint32_t gtid = blockIdx.x * kThreads + threadIdx.x;
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
-  const uint64_t span = (uint64_t)gridDim.x * kEmitThreads;
+  const uint64_t span = (uint64_t)gridDim.x * kThreads;
   uint32_t entries = 0, scanned = 0, matched = 0;
   uint64_t agg_val[NA] = {0}, agg_cnt[NA] = {0};
-  for (uint64_t j = gtid; j < n_ivs; j += span) {
+
+  for (uint64_t j0 = 0; j0 < n_ivs; j0 += span) {
+    const uint64_t j = j0 + gtid;
+    const bool active = j < n_ivs;
     HeadOut<NA> ho;
-    bool wn = false;
-    if (!scan_one_interval<NA, false, true>(
-            sp, data, block_offsets, ivs, n_ivs, j, aux, key, rk_save,
-            &entries, &scanned, &matched, agg_val, agg_cnt, &ho, &wn,
-            nullptr, nullptr, nullptr, &gc, head_flags)) {
-      atomicAdd(err_counter, 1ull);
+    GroupHead gh;
+    gh.hit = 0;
+    bool walked_next = false;
+    if (active) {
+      if (!scan_one_interval<NA, false, true>(
+              sp, data, block_offsets, ivs, n_ivs, j, aux, key, rk_save,
+              &entries, &scanned, &matched, agg_val, agg_cnt, &ho,
+              &walked_next, nullptr, nullptr, nullptr, &gc, nullptr, &gh)) {
+        atomicAdd(err_counter, 1ull);
+      }
     }
+    unsigned lane = threadIdx.x & 63;
+    unsigned wave = threadIdx.x >> 6;
+    int wn = walked_next ? 1 : 0;
+    int from_prev_lane = __shfl_up(wn, 1);
+    if (lane == 63) wave_relay[wave + 1] = (uint8_t)wn;
+    __syncthreads();
+    bool head_consumed;
+    if (threadIdx.x == 0) {
+      head_consumed = false;  // resolved via the global record instead
+    } else if (lane == 0) {
+      head_consumed = wave_relay[wave] != 0;
+    } else {
+      head_consumed = from_prev_lane != 0;
+    }
+    __syncthreads();
+    if (active) {
+      if (threadIdx.x == 0) {
+        gheads[j / kThreads] = gh;
+      } else if (!head_consumed && gh.hit) {
+        group_accum_rec<NA>(sp, gc, gh);
+      }
+      if (walked_next && ((j + 1) % kThreads) == 0 && (j + 1) < n_ivs)
+        cont_flags[(j + 1) / kThreads] = 1;
+    }
+    __syncthreads();
   }
+}
+
+template <int NA>
+__global__ void k_group_heads(GroupCtx gc,
+                              const GroupHead* __restrict__ gheads,
+                              const uint32_t* __restrict__ cont_flags,
+                              uint64_t n_heads) {
+  const DevSpec& sp = c_spec;
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (; i < n_heads; i += stride)
+    if (cont_flags[i] == 0 && gheads[i].hit)
+      group_accum_rec<NA>(sp, gc, gheads[i]);
 }
 
 __global__ void k_group_export(DevSpec sp, GroupCtx gc,
@@ -507,6 +560,7 @@ struct ybg_scan {
   std::vector<uint8_t> h_varlen;
   // group-by buffers
   GroupCtx gc = {};
+  GroupHead* d_gheads = nullptr;
   uint64_t group_cap = 0;
   uint64_t* d_gk_out = nullptr;
   long long* d_gv_out = nullptr;
@@ -870,53 +924,37 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
     HIP_TRY(hipMalloc(&s->d_g_counters, 3 * sizeof(unsigned long long)));
     s->gc.cap = s->group_cap;
     s->gc.data = s->d_data;
-    if (!s->d_flags_all)
-      HIP_TRY(hipMalloc(&s->d_flags_all, s->n_ivs * sizeof(uint32_t)));
+    HIP_TRY(hipMalloc(&s->d_gheads, s->n_heads * sizeof(GroupHead)));
   }
   HIP_TRY(hipMemsetAsync(s->gc.overflow, 0, 8, s->stream));
   HIP_TRY(hipMemsetAsync(s->d_g_counters, 0, 24, s->stream));
-  HIP_TRY(hipMemsetAsync(s->d_flags_all, 0, s->n_ivs * sizeof(uint32_t),
+  HIP_TRY(hipMemsetAsync(s->d_gheads, 0, s->n_heads * sizeof(GroupHead),
                          s->stream));
-  HIP_TRY(hipMemsetAsync(s->d_heads, 0,
-                         s->n_heads * kHeadStride * sizeof(uint64_t),
+  HIP_TRY(hipMemsetAsync(s->d_cont, 0, s->n_heads * sizeof(uint32_t),
                          s->stream));
   hipLaunchKernelGGL(k_group_init, dim3(512), dim3(256), 0, s->stream,
                      s->dspec, s->gc);
-  // flags pre-pass resolves head-row ownership for every interval
   HIP_TRY(hipMemcpyToSymbolAsync(HIP_SYMBOL(c_spec), &s->dspec,
                                  sizeof(DevSpec), 0, hipMemcpyHostToDevice,
                                  s->stream));
-  auto flags_kernel2 = k_scan<2, 3>;
-  hipLaunchKernelGGL(flags_kernel2, dim3(s->grid), dim3(kThreads), 0,
-                     s->stream, s->d_data, s->d_offsets, s->d_ivs,
-                     s->n_ivs, s->d_aux, s->d_rk_save, s->d_partials,
-                     s->d_heads, s->d_flags_all, 1);
-  int egrid = (int)std::min<uint64_t>(
-      (s->n_ivs + kEmitThreads - 1) / kEmitThreads, 8192);
-  uint64_t need_rk = (uint64_t)egrid * kEmitThreads * kKeyCap;
-  uint64_t have_rk = (uint64_t)s->grid * kThreads * kKeyCap;
-  uint8_t* rk_area = s->d_rk_save;
-  uint8_t* rk_extra = nullptr;
-  if (need_rk > have_rk) {
-    HIP_TRY(hipMalloc(&rk_extra, need_rk));
-    rk_area = rk_extra;
-  }
-  unsigned long long* err_ctr = s->d_g_counters + 2;  // reuse overflow slot?
-  // use gc.overflow for decode errors separate from export overflow
+  // single fused pass: head-deferral protocol identical to k_scan
+  int hgrid = (int)std::min<uint64_t>(
+      (s->n_ivs + kThreads - 1) / kThreads, 512);
   if (s->dspec.num_aggs <= 4) {
-    auto kg = k_group<4>;
-    hipLaunchKernelGGL(kg, dim3(egrid), dim3(kEmitThreads), 0, s->stream,
-                       s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
-                       s->d_aux, rk_area, s->gc, s->d_flags_all,
-                       s->gc.overflow);
+    hipLaunchKernelGGL(k_group<4>, dim3(s->grid), dim3(kThreads), 0,
+                       s->stream, s->d_data, s->d_offsets, s->d_ivs,
+                       s->n_ivs, s->d_aux, s->d_rk_save, s->gc, s->d_gheads,
+                       s->d_cont, s->gc.overflow);
+    hipLaunchKernelGGL(k_group_heads<4>, dim3(hgrid), dim3(kThreads), 0,
+                       s->stream, s->gc, s->d_gheads, s->d_cont, s->n_heads);
   } else {
-    auto kg = k_group<8>;
-    hipLaunchKernelGGL(kg, dim3(egrid), dim3(kEmitThreads), 0, s->stream,
-                       s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
-                       s->d_aux, rk_area, s->gc, s->d_flags_all,
-                       s->gc.overflow);
+    hipLaunchKernelGGL(k_group<8>, dim3(s->grid), dim3(kThreads), 0,
+                       s->stream, s->d_data, s->d_offsets, s->d_ivs,
+                       s->n_ivs, s->d_aux, s->d_rk_save, s->gc, s->d_gheads,
+                       s->d_cont, s->gc.overflow);
+    hipLaunchKernelGGL(k_group_heads<8>, dim3(hgrid), dim3(kThreads), 0,
+                       s->stream, s->gc, s->d_gheads, s->d_cont, s->n_heads);
   }
-  (void)err_ctr;
   // export buffers sized to caller caps
   if (!s->d_gk_out) {
     HIP_TRY(hipMalloc(&s->d_gk_out, cap * 8));
@@ -934,7 +972,6 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
   HIP_TRY(hipMemcpyAsync(&errs, s->gc.overflow, 8, hipMemcpyDeviceToHost,
                          s->stream));
   HIP_TRY(hipStreamSynchronize(s->stream));
-  if (rk_extra) HIP_WARN(hipFree(rk_extra));
   if (errs) return set_err(6, "corrupt entries or group table overflow");
   if (ctr[2]) return set_err(8, "group output capacity exceeded");
   uint64_t ng = ctr[0];
@@ -992,6 +1029,7 @@ int yb_gpu_scan_close(ybg_scan_t* s) {
   if (s->gc.vals) HIP_WARN(hipFree(s->gc.vals));
   if (s->gc.cnts) HIP_WARN(hipFree(s->gc.cnts));
   if (s->gc.overflow) HIP_WARN(hipFree(s->gc.overflow));
+  if (s->d_gheads) HIP_WARN(hipFree(s->d_gheads));
   if (s->d_g_counters) HIP_WARN(hipFree(s->d_g_counters));
   if (s->d_gk_out) HIP_WARN(hipFree(s->d_gk_out));
   if (s->d_gv_out) HIP_WARN(hipFree(s->d_gv_out));

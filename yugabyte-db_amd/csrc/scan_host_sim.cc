@@ -213,20 +213,9 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
   uint32_t entries = 0, scanned = 0, matched = 0;
   uint64_t agg_val[YBG_MAX_AGGS] = {0}, agg_cnt[YBG_MAX_AGGS] = {0};
   std::vector<uint8_t> walked(n_ivs, 0);
-  std::vector<uint32_t> head_flags(n_ivs ? n_ivs : 1, 0);
+  std::vector<GroupHead> gheads(n_ivs ? n_ivs : 1);
   alignas(8) uint8_t key[kKeyCap];
   alignas(8) uint8_t rk_save[kKeyCap];
-  for (uint64_t j = 0; j < n_ivs; ++j) {
-    bool wn = false;
-    HeadOut<YBG_MAX_AGGS> ho;
-    if (!scan_one_interval<YBG_MAX_AGGS>(d, data, offsets, ivs.data(), n_ivs,
-                                         j, aux.data(), key, rk_save,
-                                         &entries, &scanned, &matched,
-                                         agg_val, agg_cnt, &ho, &wn))
-      return 6;
-    walked[j] = wn ? 1 : 0;
-  }
-  for (uint64_t j = 1; j < n_ivs; ++j) head_flags[j] = walked[j - 1];
 
   uint64_t gcap = 1ull << 18;
   std::vector<unsigned long long> gkey(gcap + 1, 0);
@@ -252,11 +241,19 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
   for (uint64_t j = 0; j < n_ivs; ++j) {
     bool wn = false;
     HeadOut<YBG_MAX_AGGS> ho;
+    gheads[j].hit = 0;
     if (!scan_one_interval<YBG_MAX_AGGS, false, true>(
             d, data, offsets, ivs.data(), n_ivs, j, aux.data(), key, rk_save,
             &entries, &scanned, &matched, agg_val, agg_cnt, &ho, &wn,
-            nullptr, nullptr, nullptr, &gc, head_flags.data()))
+            nullptr, nullptr, nullptr, &gc, nullptr, &gheads[j]))
       return 6;
+    walked[j] = wn ? 1 : 0;
+  }
+  // head-ownership resolution (single-pass protocol, serial equivalent)
+  for (uint64_t j = 0; j < n_ivs; ++j) {
+    bool consumed = j > 0 && walked[j - 1];
+    if (!consumed && gheads[j].hit)
+      group_accum_rec<YBG_MAX_AGGS>(d, gc, gheads[j]);
   }
   if (overflow) return 8;
   bool grp_is_str =
