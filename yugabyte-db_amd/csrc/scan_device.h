@@ -998,8 +998,13 @@ template <int NA>
 struct RowCtxT {
   bool base_seen;
   bool found;
-  uint64_t base_ht_hi, base_ht_lo;
-  uint32_t base_ht_len;
+  // The packed row's write time, needed ONLY when a later column-update
+  // entry must be ordered against it (doc_reader.cc:1847-1869). Written
+  // once per row but read on the cold update path only — kept in LDS (the
+  // kernel passes a per-thread slot) instead of registers, where the
+  // allocator was spilling it to HBM-backed scratch (~2.4 GB of write
+  // traffic per 100M-row dispatch).
+  uint64_t* bht;  // [3]: hi, lo, len
   int32_t cur_col;  // current column-update group (-2 liveness, -1 none)
   bool cur_col_done;
   uint32_t pred_pass;  // bit i: value-col predicate i passes (key preds at
@@ -1631,9 +1636,9 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
   if (rowkey_len == prefix_len) {
     if (!rc->base_seen) {
       rc->base_seen = true;
-      rc->base_ht_hi = ht_hi;
-      rc->base_ht_lo = ht_lo;
-      rc->base_ht_len = ht_size;
+      rc->bht[0] = ht_hi;
+      rc->bht[1] = ht_lo;
+      rc->bht[2] = ht_size;
       if (value_len > 0 && vb0 == kPackedV2B && value == rdr->pos()) {
         if (sp.v2_fixed_len && value_len == sp.v2_fixed_len &&
             (rdr->peek8() & 0xff8000u) == 0) {
@@ -1678,8 +1683,8 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
       if (!rc->cur_col_done) {
         rc->cur_col_done = true;
         bool newer = !rc->base_seen ||
-                     u128_slice_cmp(ht_hi, ht_lo, ht_size, rc->base_ht_hi,
-                                    rc->base_ht_lo, rc->base_ht_len) < 0;
+                     u128_slice_cmp(ht_hi, ht_lo, ht_size, rc->bht[0],
+                                    rc->bht[1], (uint32_t)rc->bht[2]) < 0;
         if (newer) {
           int cf = skip_control(value, value_len);
           if (cf < 0) return false;
@@ -1703,8 +1708,8 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
       if (!rc->cur_col_done) {
         rc->cur_col_done = true;
         bool newer = !rc->base_seen ||
-                     u128_slice_cmp(ht_hi, ht_lo, ht_size, rc->base_ht_hi,
-                                    rc->base_ht_lo, rc->base_ht_len) < 0;
+                     u128_slice_cmp(ht_hi, ht_lo, ht_size, rc->bht[0],
+                                    rc->bht[1], (uint32_t)rc->bht[2]) < 0;
         if (newer) {
           int cf = skip_control(value, value_len);
           if (cf < 0) return false;
@@ -2012,7 +2017,8 @@ template <int NA, bool EMIT = false, bool GROUP = false>
 DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
                            const uint64_t* block_offsets, const Interval* ivs,
                            uint64_t n_ivs, uint64_t j, const uint8_t* aux,
-                           uint8_t* key, uint8_t* rk_save, uint32_t* entries,
+                           uint8_t* key, uint8_t* rk_save, uint64_t* bht,
+                           uint32_t* entries,
                            uint32_t* scanned, uint32_t* matched,
                            uint64_t* agg_val, uint64_t* agg_cnt,
                            HeadOut<NA>* ho, bool* walked_next_out,
@@ -2045,6 +2051,7 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
   bool row_open = false;
   bool in_head = true;
   RowCtxT<NA> rc;
+  rc.bht = bht;
   rc.emit_datums = EMIT ? emit_datums : nullptr;
   rc.emit_lens = EMIT ? emit_lens : nullptr;
   row_reset(&rc, sp);

@@ -68,7 +68,9 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
   const DevSpec& sp = c_spec;
   __shared__ uint8_t key_scratch[kThreads * kKeyCap];
   __shared__ uint8_t wave_relay[kThreads / 64 + 1];
+  __shared__ uint64_t bht_scratch[kThreads * 3];
   uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
+  uint64_t* bht = bht_scratch + (size_t)threadIdx.x * 3;
   const uint32_t gtid = blockIdx.x * kThreads + threadIdx.x;
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
   const uint64_t span = (uint64_t)gridDim.x * kThreads;
@@ -89,8 +91,9 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
     bool walked_next = false;
     if (active) {
       if (!scan_one_interval<NA>(sp, data, block_offsets, ivs, n_ivs, j, aux,
-                                 key, rk_save, &entries, &scanned, &matched,
-                                 agg_val, agg_cnt, &ho, &walked_next)) {
+                                 key, rk_save, bht, &entries, &scanned,
+                                 &matched, agg_val, agg_cnt, &ho,
+                                 &walked_next)) {
         errs += 1;
       }
     } else {
@@ -202,6 +205,8 @@ __global__ __launch_bounds__(kEmitThreads) void k_emit(
   __shared__ uint8_t key_scratch[kEmitThreads * kKeyCap];
   __shared__ uint64_t rowbuf[kEmitThreads * YBG_MAX_COLS];
   __shared__ uint32_t lenbuf[kEmitThreads * YBG_MAX_COLS];
+  __shared__ uint64_t bht_scratch[kEmitThreads * 3];
+  uint64_t* bht = bht_scratch + (size_t)threadIdx.x * 3;
   uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
   uint64_t* rb = rowbuf + (size_t)threadIdx.x * YBG_MAX_COLS;
   uint32_t* lb = lenbuf + (size_t)threadIdx.x * YBG_MAX_COLS;
@@ -215,9 +220,9 @@ __global__ __launch_bounds__(kEmitThreads) void k_emit(
     HeadOut<2> ho;
     bool wn = false;
     if (!scan_one_interval<2, true>(sp, data, block_offsets, ivs, n_ivs, j,
-                                    aux, key, rk_save, &entries, &scanned,
-                                    &matched, agg_val, agg_cnt, &ho, &wn,
-                                    &ec, rb, lb)) {
+                                    aux, key, rk_save, bht, &entries,
+                                    &scanned, &matched, agg_val, agg_cnt,
+                                    &ho, &wn, &ec, rb, lb)) {
       atomicAdd(err_counter, 1ull);
     }
   }
@@ -264,6 +269,8 @@ __global__ __launch_bounds__(kThreads, 3) void k_group(
   const DevSpec& sp = c_spec;
   __shared__ uint8_t key_scratch[kThreads * kKeyCap];
   __shared__ uint8_t wave_relay[kThreads / 64 + 1];
+  __shared__ uint64_t bht_scratch[kThreads * 3];
+  uint64_t* bht = bht_scratch + (size_t)threadIdx.x * 3;
   uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
   const uint32_t gtid = blockIdx.x * kThreads + threadIdx.x;
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
@@ -281,7 +288,7 @@ __global__ __launch_bounds__(kThreads, 3) void k_group(
     if (active) {
       if (!scan_one_interval<NA, false, true>(
               sp, data, block_offsets, ivs, n_ivs, j, aux, key, rk_save,
-              &entries, &scanned, &matched, agg_val, agg_cnt, &ho,
+              bht, &entries, &scanned, &matched, agg_val, agg_cnt, &ho,
               &walked_next, nullptr, nullptr, nullptr, &gc, nullptr, &gh)) {
         atomicAdd(err_counter, 1ull);
       }

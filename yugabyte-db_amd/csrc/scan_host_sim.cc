@@ -51,11 +51,12 @@ int ybg_sim_scan(const ybg_scan_spec_t* spec, const uint8_t* data,
   std::vector<uint8_t> walked(n_ivs, 0);
   alignas(8) uint8_t key[kKeyCap];
   alignas(8) uint8_t rk_save[kKeyCap];
+  uint64_t bht[3] = {0, 0, 0};
   for (uint64_t j = 0; j < n_ivs; ++j) {
     bool wn = false;
     uint32_t e32 = 0, s32 = 0, m32 = 0;
     if (!scan_one_interval<YBG_MAX_AGGS>(d, data, offsets, ivs.data(), n_ivs,
-                                         j, aux.data(), key, rk_save,
+                                         j, aux.data(), key, rk_save, bht,
                                          &e32, &s32, &m32,
                                          agg_val, agg_cnt, &heads[j], &wn))
       return 6;
@@ -135,11 +136,12 @@ int ybg_sim_emit(const ybg_scan_spec_t* spec, const uint8_t* data,
   std::vector<uint32_t> head_consumed(n_ivs ? n_ivs : 1, 0);
   alignas(8) uint8_t key[kKeyCap];
   alignas(8) uint8_t rk_save[kKeyCap];
+  uint64_t bht[3] = {0, 0, 0};
   HeadOut<YBG_MAX_AGGS> ho;
   for (uint64_t j = 0; j < n_ivs; ++j) {
     bool wn = false;
     if (!scan_one_interval<YBG_MAX_AGGS>(d, data, offsets, ivs.data(), n_ivs,
-                                         j, aux.data(), key, rk_save,
+                                         j, aux.data(), key, rk_save, bht,
                                          &entries, &scanned, &matched,
                                          agg_val, agg_cnt, &ho, &wn))
       return 6;
@@ -170,8 +172,8 @@ int ybg_sim_emit(const ybg_scan_spec_t* spec, const uint8_t* data,
     HeadOut<YBG_MAX_AGGS> ho2;
     if (!scan_one_interval<YBG_MAX_AGGS, true>(
             d, data, offsets, ivs.data(), n_ivs, j, aux.data(), key, rk_save,
-            &entries, &scanned, &matched, agg_val, agg_cnt, &ho2, &wn, &ec,
-            rowbuf, lenbuf))
+            bht, &entries, &scanned, &matched, agg_val, agg_cnt, &ho2, &wn,
+            &ec, rowbuf, lenbuf))
       return 6;
   }
   if (overflow) return 8;
@@ -216,6 +218,7 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
   std::vector<GroupHead> gheads(n_ivs ? n_ivs : 1);
   alignas(8) uint8_t key[kKeyCap];
   alignas(8) uint8_t rk_save[kKeyCap];
+  uint64_t bht[3] = {0, 0, 0};
 
   uint64_t gcap = 1ull << 18;
   std::vector<unsigned long long> gkey(gcap + 1, 0);
@@ -244,7 +247,7 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
     gheads[j].hit = 0;
     if (!scan_one_interval<YBG_MAX_AGGS, false, true>(
             d, data, offsets, ivs.data(), n_ivs, j, aux.data(), key, rk_save,
-            &entries, &scanned, &matched, agg_val, agg_cnt, &ho, &wn,
+            bht, &entries, &scanned, &matched, agg_val, agg_cnt, &ho, &wn,
             nullptr, nullptr, nullptr, &gc, nullptr, &gheads[j]))
       return 6;
     walked[j] = wn ? 1 : 0;
