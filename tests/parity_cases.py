@@ -138,6 +138,37 @@ def build_cases():
     cases.append(_case("nulls_packed", schema_n, b.finish(),
                        [(1_000_000, (), aggs)]))
 
+    # wide all-fixed schema (32 int64 columns): the fixed-offset packed-V2
+    # fast path at its column-count and offset bounds (v2_off up to 251)
+    schema_w = y.make_schema([y.KT_INT64],
+                             [(100 + i, y.T_INT64, 0) for i in range(32)])
+    b = y.Builder(schema_w)
+    for r in range(1500):
+        b.add_packed_row(1000 + r,
+                         [(y.T_INT64, r * 31 + c) for c in range(32)],
+                         hash_=r // 64, key_datums=(r,), packed_version=2)
+    _KEEP.append(b)
+    aggs = [y.Agg(y.AGG_SUM_INT64, 0), y.Agg(y.AGG_SUM_INT64, 31),
+            y.Agg(y.AGG_COUNT_STAR, 0)]
+    preds = [y.Pred(0, 16, y.PRED_GT, 20_000, None, 0)]
+    cases.append(_case("wide_fixed_32cols", schema_w, b.finish(),
+                       [(1_000_000, (), aggs), (1_000_000, preds, aggs)]))
+
+    # wide nullable rows: per-row null masks force the window/pointer
+    # decoders off the fixed-offset path for exactly the rows with nulls
+    schema_wn = y.make_schema([y.KT_INT64],
+                              [(100 + i, y.T_INT64, 1) for i in range(12)])
+    b = y.Builder(schema_wn)
+    for r in range(3000):
+        vals = [(y.T_INT64, None if (r + c) % 7 == 0 else r + c)
+                for c in range(12)]
+        b.add_packed_row(1000 + r, vals, hash_=r // 64, key_datums=(r,),
+                         packed_version=2)
+    _KEEP.append(b)
+    aggs = [y.Agg(y.AGG_COUNT, 3), y.Agg(y.AGG_SUM_INT64, 11)]
+    cases.append(_case("wide_nullable_mixed", schema_wn, b.finish(),
+                       [(1_000_000, (), aggs)]))
+
     # key-column predicate
     built = y.generate(SCHEMA_4I, rows=30_000)
     preds = [y.Pred(1, 0, y.PRED_LT, 10_000, None, 0)]
