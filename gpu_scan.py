@@ -83,6 +83,16 @@ class GpuScan:
                                               n_blocks, 0),
             "feed_blocks")
 
+    def feed_sst(self, file_ptr, size, verify=True):
+        """Feed a complete BlockBasedTable SST file (footer + index block
+        parsed host-side; block checksums verified when verify)."""
+        import ctypes as C
+        f = self._lib.yb_gpu_scan_feed_sst
+        f.restype = C.c_int
+        f.argtypes = [C.c_void_p, C.POINTER(C.c_uint8), C.c_uint64, C.c_int]
+        self._check(f(self._h, file_ptr, size, 1 if verify else 0),
+                    "feed_sst")
+
     def execute(self):
         self._check(self._lib.yb_gpu_scan_execute(self._h), "execute")
 

@@ -177,6 +177,15 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t *s, const uint8_t *blocks,
                             const uint64_t *offsets, uint64_t n_blocks,
                             int device);
 
+/* Feed a complete BlockBasedTable SST file (host memory): parses the
+ * version-2 footer + index block (rocksdb/table/format.cc:59-155,
+ * block_based_table_builder.cc:658-700), optionally verifies every data
+ * block's masked-crc32c trailer, strips trailers and feeds the data
+ * blocks. kNoCompression blocks only (the reference-supported uncompressed
+ * configuration, docdb_rocksdb_util.cc:200-221). */
+int yb_gpu_scan_feed_sst(ybg_scan_t *s, const uint8_t *file, uint64_t size,
+                         int verify_checksums);
+
 /* Run the scan asynchronously on the handle's stream. */
 int yb_gpu_scan_execute(ybg_scan_t *s);
 
@@ -311,6 +320,19 @@ int ybg_builder_add_raw(ybg_builder_t *b, const uint8_t *user_key,
 int ybg_builder_finish(ybg_builder_t *b, const uint8_t **data,
                        const uint64_t **offsets, uint64_t *n_blocks,
                        uint64_t *total_bytes, uint64_t *n_entries);
+
+/* Finish as a complete SST FILE (data blocks + [type|masked-crc32c]
+ * trailers + empty metaindex + shared-prefix index block + version-2
+ * footer). Pointer valid until destroy. */
+int ybg_builder_finish_sst(ybg_builder_t *b, const uint8_t **data,
+                           uint64_t *total_bytes, uint64_t *n_blocks,
+                           uint64_t *n_entries);
+
+/* Parse an SST file and report its data-block (offset, size) handles —
+ * the exact parser yb_gpu_scan_feed_sst uses (exposed for CPU tests). */
+int ybg_sst_index(const uint8_t *file, uint64_t size, int verify,
+                  uint64_t *offsets, uint64_t *sizes, uint64_t cap,
+                  uint64_t *n_blocks);
 
 void ybg_builder_destroy(ybg_builder_t *b);
 

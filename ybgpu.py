@@ -337,6 +337,46 @@ class Builder:
         assert rc == 0
         return data, offsets, n_blocks.value, total.value, n_entries.value
 
+    def finish_sst(self):
+        """Finish as a complete BlockBasedTable SST file; returns
+        (data_ptr, total_bytes, n_blocks, n_entries)."""
+        lib = product()
+        f = _sig(lib, "ybg_builder_finish_sst", C.c_int,
+                 [C.c_void_p, C.POINTER(C.POINTER(C.c_uint8)),
+                  C.POINTER(C.c_uint64), C.POINTER(C.c_uint64),
+                  C.POINTER(C.c_uint64)])
+        data = C.POINTER(C.c_uint8)()
+        total = C.c_uint64()
+        n_blocks = C.c_uint64()
+        n_entries = C.c_uint64()
+        rc = f(self._h, C.byref(data), C.byref(total), C.byref(n_blocks),
+               C.byref(n_entries))
+        assert rc == 0
+        return data, total.value, n_blocks.value, n_entries.value
+
+
+def sst_index(file_ptr, size=None, verify=True, cap=1 << 20):
+    """Parse an SST file's footer + index: data-block (offset, size) lists.
+    Raises RuntimeError on corrupt input. file_ptr: ctypes pointer or
+    bytes."""
+    lib = product()
+    f = _sig(lib, "ybg_sst_index", C.c_int,
+             [C.POINTER(C.c_uint8), C.c_uint64, C.c_int,
+              C.POINTER(C.c_uint64), C.POINTER(C.c_uint64), C.c_uint64,
+              C.POINTER(C.c_uint64)])
+    if isinstance(file_ptr, (bytes, bytearray)):
+        buf = (C.c_uint8 * len(file_ptr)).from_buffer_copy(file_ptr)
+        file_ptr = C.cast(buf, C.POINTER(C.c_uint8))
+        size = len(buf)
+    offs = (C.c_uint64 * cap)()
+    szs = (C.c_uint64 * cap)()
+    n = C.c_uint64()
+    rc = f(file_ptr, size, 1 if verify else 0, offs, szs, cap, C.byref(n))
+    if rc:
+        err = _sig(lib, "yb_gpu_last_error", C.c_char_p, [])
+        raise RuntimeError("sst_index: rc=%d %s" % (rc, err().decode()))
+    return list(offs[:n.value]), list(szs[:n.value])
+
 
 # ---------------------------------------------------------------------------
 # Oracle wrappers (TEST INFRASTRUCTURE ONLY)

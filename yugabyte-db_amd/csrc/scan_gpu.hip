@@ -42,6 +42,7 @@
 #include <vector>
 
 #include "scan_device.h"
+#include "sst_internal.h"
 
 using namespace ybgdev;
 
@@ -702,6 +703,48 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
   HIP_TRY(hipMalloc(&s->d_result, sizeof(DevResult)));
   HIP_TRY(hipMalloc(&s->d_chunk, 256 * kPartialStride * sizeof(uint64_t)));
   return 0;
+}
+
+// Parse an SST file and report the data-block handles (the same parser
+// the feed path uses; exposed for CPU-side tests — errors via
+// yb_gpu_last_error).
+int ybg_sst_index(const uint8_t* file, uint64_t size, int verify,
+                  uint64_t* offsets, uint64_t* sizes, uint64_t cap,
+                  uint64_t* n_blocks) {
+  std::vector<uint64_t> offs, szs;
+  std::string err;
+  int rc = ybsst::parse_sst(file, size, verify, &offs, &szs, &err);
+  if (rc) return set_err(rc, err);
+  *n_blocks = offs.size();
+  for (uint64_t i = 0; i < offs.size() && i < cap; ++i) {
+    offsets[i] = offs[i];
+    sizes[i] = szs[i];
+  }
+  return 0;
+}
+
+// Feed a complete SST file: parse footer/index (sst_format.cc), verify
+// block checksums, strip the per-block trailers into the concatenated
+// block layout and hand over to the existing feed path.
+int yb_gpu_scan_feed_sst(ybg_scan_t* s, const uint8_t* file, uint64_t size,
+                         int verify_checksums) {
+  std::vector<uint64_t> offs, szs;
+  std::string err;
+  int rc = ybsst::parse_sst(file, size, verify_checksums, &offs, &szs, &err);
+  if (rc) return set_err(rc, err);
+  if (offs.empty()) return set_err(3, "SST file holds no data blocks");
+  uint64_t total = 0;
+  for (uint64_t sz : szs) total += sz;
+  std::vector<uint8_t> blocks;
+  blocks.reserve(total);
+  std::vector<uint64_t> boff(offs.size() + 1);
+  boff[0] = 0;
+  for (size_t i = 0; i < offs.size(); ++i) {
+    blocks.insert(blocks.end(), file + offs[i], file + offs[i] + szs[i]);
+    boff[i + 1] = blocks.size();
+  }
+  return yb_gpu_scan_feed_blocks(s, blocks.data(), boff.data(), offs.size(),
+                                 0);
 }
 
 int yb_gpu_scan_execute(ybg_scan_t* s) {

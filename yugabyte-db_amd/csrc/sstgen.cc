@@ -9,6 +9,7 @@
 //   Single V1 values       src/yb/dockv/primitive_value.cc:1066-1125
 #include "../../include/yb_gpu_scan.h"
 #include "codec.h"
+#include "sst_internal.h"
 
 #include <algorithm>
 #include <atomic>
@@ -161,6 +162,7 @@ class BlockBuilder {
 
   const Buf& data() const { return buf_; }
   size_t entries() const { return n_entries_; }
+  const Buf& last_key() const { return last_key_; }
 
  private:
   void AddThreeSharedParts(const uint8_t* key, size_t key_size,
@@ -414,6 +416,8 @@ struct ybg_builder {
   ybg::BlockBuilder bb;
   ybg::Buf all_blocks;
   std::vector<uint64_t> offsets;
+  std::vector<ybg::Buf> block_last_keys;
+  ybg::Buf sst_file;
   uint64_t n_entries = 0;
   ybg::Buf key_scratch, value_scratch;
 
@@ -425,6 +429,7 @@ struct ybg_builder {
 
   void FlushBlock() {
     if (bb.Empty()) return;
+    block_last_keys.push_back(bb.last_key());
     bb.Finish();
     all_blocks.insert(all_blocks.end(), bb.data().begin(), bb.data().end());
     offsets.push_back(all_blocks.size());
@@ -518,6 +523,81 @@ int ybg_builder_finish(ybg_builder_t* b, const uint8_t** data,
   *offsets = b->offsets.data();
   *n_blocks = b->offsets.size() - 1;
   *total_bytes = b->all_blocks.size();
+  *n_entries = b->n_entries;
+  return 0;
+}
+
+// Assemble a complete BlockBasedTable SST file: data blocks with
+// [type|crc32c] trailers, an empty metaindex block, a shared-prefix index
+// block (restart interval 1) whose values are BlockHandles, and the
+// version-2 footer. Format citations: sst_internal.h.
+int ybg_builder_finish_sst(ybg_builder_t* b, const uint8_t** data,
+                           uint64_t* total_bytes, uint64_t* n_blocks,
+                           uint64_t* n_entries) {
+  using namespace ybsst;
+  b->FlushBlock();
+  ybg::Buf& f = b->sst_file;
+  f.clear();
+  uint64_t nb = b->offsets.size() - 1;
+  auto append_block = [&](const uint8_t* p, uint64_t n, uint64_t* h_off,
+                          uint64_t* h_sz) {
+    *h_off = f.size();
+    *h_sz = n;
+    f.insert(f.end(), p, p + n);
+    uint8_t type = 0;  // kNoCompression
+    uint32_t crc = crc32c_extend(crc32c_value(p, n), &type, 1);
+    f.push_back(type);
+    ybg::Fixed32LEAppend(crc32c_mask(crc), &f);
+  };
+  std::vector<uint64_t> h_off(nb), h_sz(nb);
+  for (uint64_t i = 0; i < nb; ++i)
+    append_block(b->all_blocks.data() + b->offsets[i],
+                 b->offsets[i + 1] - b->offsets[i], &h_off[i], &h_sz[i]);
+  // empty metaindex block: restart array only (block_builder.cc Finish)
+  uint64_t mi_off, mi_sz;
+  {
+    ybg::Buf mi;
+    ybg::Fixed32LEAppend(0, &mi);  // restart[0]
+    ybg::Fixed32LEAppend(1, &mi);  // num_restarts
+    append_block(mi.data(), mi.size(), &mi_off, &mi_sz);
+  }
+  // index block: key = the block's last internal key, value = BlockHandle
+  uint64_t ix_off, ix_sz;
+  {
+    ybg::BlockBuilder ib(1, YBG_ENC_SHARED_PREFIX);
+    ybg::Buf hv;
+    for (uint64_t i = 0; i < nb; ++i) {
+      hv.clear();
+      ybg::Leb128Append(h_off[i], &hv);
+      ybg::Leb128Append(h_sz[i], &hv);
+      const ybg::Buf& lk = b->block_last_keys[i];
+      ib.Add(lk.data(), lk.size(), hv.data(), hv.size());
+    }
+    ib.Finish();
+    append_block(ib.data().data(), ib.data().size(), &ix_off, &ix_sz);
+  }
+  // version-2 footer (format.cc:129-155): checksum byte, two handles,
+  // zero padding to 41 bytes, version, magic
+  {
+    size_t base = f.size();
+    f.push_back(1);  // kCRC32c
+    ybg::Leb128Append(mi_off, &f);
+    ybg::Leb128Append(mi_sz, &f);
+    ybg::Leb128Append(ix_off, &f);
+    ybg::Leb128Append(ix_sz, &f);
+    f.resize(base + 41, 0);
+    ybg::Fixed32LEAppend(2, &f);  // version
+    ybg::Fixed32LEAppend((uint32_t)(kBlockBasedTableMagic & 0xffffffffu), &f);
+    ybg::Fixed32LEAppend((uint32_t)(kBlockBasedTableMagic >> 32), &f);
+  }
+  // tail slack for the scan path's windowed loads (Rdr contract)
+  uint64_t sz = f.size();
+  f.resize(sz + 48, 0);
+  f.resize(sz);
+  f.reserve(sz + 48);
+  *data = f.data();
+  *total_bytes = sz;
+  *n_blocks = nb;
   *n_entries = b->n_entries;
   return 0;
 }
