@@ -109,6 +109,14 @@ void ybg_read_time_init(ybg_read_time_t *rt, uint64_t read_ht,
 typedef enum {
   YBG_PRED_GT = 0, YBG_PRED_GE, YBG_PRED_LT, YBG_PRED_LE,
   YBG_PRED_EQ, YBG_PRED_NE,
+  /* IN-list over a NUMERIC column: bytes = n x 8-byte little-endian datum
+   * bit patterns (the column dtype's representation), bytes_len = 8n.
+   * This is the scan-side equivalent of the reference's hybrid-scan
+   * option filters (docdb/hybrid_scan_choices.h:43-60, IN extraction
+   * qlexpr/ql_scanspec.cc:323-346): on a brute-force bandwidth-bound GPU
+   * scan the discrete options become a filter, not a seek plan. String
+   * columns are rejected at open(). */
+  YBG_PRED_IN,
 } ybg_pred_op_t;
 
 typedef struct {

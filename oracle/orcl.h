@@ -208,6 +208,9 @@ void orcl_read_time_init(orcl_read_time_t *rt, uint64_t read_ht,
 typedef enum {
   ORCL_PRED_GT = 0, ORCL_PRED_GE, ORCL_PRED_LT, ORCL_PRED_LE,
   ORCL_PRED_EQ, ORCL_PRED_NE,
+  /* IN list over a numeric column: bytes = n x 8-byte LE datum patterns
+   * (hybrid_scan_choices.h:43-60; ql_scanspec.cc:323-346) */
+  ORCL_PRED_IN,
 } orcl_pred_op_t;
 
 typedef struct {

@@ -425,6 +425,30 @@ static int pred_eval(const orcl_pred_t *pr, const orcl_schema_t *sc,
       rhs_i = (int64_t)pr->datum;
     }
   }
+  if (pr->op == ORCL_PRED_IN) {
+    /* membership over the option list (numeric columns only) */
+    size_t n = pr->bytes_len / 8, i;
+    if (numeric == 0) return 0;
+    for (i = 0; i < n; ++i) {
+      uint64_t raw;
+      memcpy(&raw, pr->bytes + 8 * i, 8);
+      if (numeric == 1) {
+        if (lhs_i == (int64_t)raw) return 1;
+      } else {
+        double b;
+        if (sc->value_cols[pr->col].dtype == ORCL_T_FLOAT) {
+          uint32_t bu = (uint32_t)raw;
+          float bf;
+          memcpy(&bf, &bu, 4);
+          b = bf;
+        } else {
+          memcpy(&b, &raw, 8);
+        }
+        if (lhs_f == b) return 1;
+      }
+    }
+    return 0;
+  }
   if (numeric == 1) cmp = lhs_i < rhs_i ? -1 : (lhs_i > rhs_i ? 1 : 0);
   else if (numeric == 2) cmp = lhs_f < rhs_f ? -1 : (lhs_f > rhs_f ? 1 : 0);
   switch (pr->op) {
@@ -434,6 +458,7 @@ static int pred_eval(const orcl_pred_t *pr, const orcl_schema_t *sc,
     case ORCL_PRED_LE: return cmp <= 0;
     case ORCL_PRED_EQ: return cmp == 0;
     case ORCL_PRED_NE: return cmp != 0;
+    case ORCL_PRED_IN: return 0; /* handled above */
   }
   return 0;
 }

@@ -176,6 +176,36 @@ def build_cases():
     cases.append(_case("key_pred", SCHEMA_4I, built,
                        [(1_700_000_000_000_000, preds, aggs)]))
 
+    # IN-list predicates (hybrid-scan option filters): value column and key
+    # column membership over packed 8-byte LE datum lists
+    import ctypes as _C
+    import struct as _struct
+    built = y.generate(SCHEMA_4I, rows=25_000)
+    # real col-0 datums (oracle row collection) + junk -> a list that both
+    # matches and misses
+    _osc = y.orcl_schema_from(SCHEMA_4I)
+    _ospec = make_orcl_spec(1_700_000_000_000_000, (), [])
+    _, _rows = y.orcl_scan(built[0], built[1], built[2], _osc, _ospec,
+                           collect_rows=True)
+    vals = sorted({int(r[1][0]) for r in _rows[:4000:97]
+                   if r[1][0] is not None} | {0, 12345678901234})
+    inbuf1 = _struct.pack("<%dQ" % len(vals), *[v & (2**64 - 1) for v in vals])
+    a1 = (_C.c_uint8 * len(inbuf1)).from_buffer_copy(inbuf1)
+    keys = list(range(100, 24_000, 311))
+    inbuf2 = _struct.pack("<%dq" % len(keys), *keys)
+    a2 = (_C.c_uint8 * len(inbuf2)).from_buffer_copy(inbuf2)
+    _KEEP.append(a1)
+    _KEEP.append(a2)
+    preds_v = [y.Pred(0, 0, y.PRED_IN, 0, a1, len(inbuf1))]
+    preds_k = [y.Pred(1, 0, y.PRED_IN, 0, a2, len(inbuf2))]
+    preds_both = [y.Pred(0, 0, y.PRED_IN, 0, a1, len(inbuf1)),
+                  y.Pred(1, 0, y.PRED_IN, 0, a2, len(inbuf2))]
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 3)]
+    cases.append(_case("in_list_preds", SCHEMA_4I, built,
+                       [(1_700_000_000_000_000, preds_v, aggs),
+                        (1_700_000_000_000_000, preds_k, aggs),
+                        (1_700_000_000_000_000, preds_both, aggs)]))
+
     # single row + empty result
     b = y.Builder(SCHEMA_4I)
     b.add_packed_row(1000, [(y.T_INT64, i) for i in range(4)], hash_=7,

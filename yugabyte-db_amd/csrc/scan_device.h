@@ -1042,6 +1042,27 @@ DEV bool pred_compare(const PredC& pr, uint64_t datum, const uint8_t* sptr,
                       uint32_t slen, const uint8_t* aux) {
   int dtype = (int)(pr.opdt >> 8);
   int cmp;
+  if (YBG_UNLIKELY((pr.opdt & 0xff) == YBG_PRED_IN)) {
+    // membership over the option list (hybrid_scan_choices.h:43-60);
+    // numeric columns only (open() rejects strings). datum packs
+    // (count << 32) | aux offset of the 8-byte LE option patterns.
+    uint32_t n = (uint32_t)(pr.datum >> 32);
+    const uint8_t* lst = aux + (uint32_t)pr.datum;
+    for (uint32_t i = 0; i < n; ++i) {
+      uint64_t rv = load_u64_una(lst + 8ull * i);
+      if (dtype == YBG_T_DOUBLE) {
+        if (__longlong_as_double((long long)datum) ==
+            __longlong_as_double((long long)rv))
+          return true;
+      } else if (dtype == YBG_T_FLOAT) {
+        if (__uint_as_float((uint32_t)datum) == __uint_as_float((uint32_t)rv))
+          return true;
+      } else {
+        if (datum == rv) return true;
+      }
+    }
+    return false;
+  }
   if (YBG_LIKELY(dtype != YBG_T_STRING && dtype != YBG_T_DOUBLE &&
                  dtype != YBG_T_FLOAT)) {
     int64_t a = (int64_t)datum, b = (int64_t)pr.datum;
@@ -1360,6 +1381,13 @@ DEV bool eval_key_preds(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
         bool le = si >= sl, re = ri >= pr.str_len;
         cmp = (le && re) ? 0 : (le ? -1 : 1);
       }
+    } else if (pr.op == YBG_PRED_IN) {
+      uint32_t n = pr.str_len / 8;
+      bool hit = false;
+      for (uint32_t k2 = 0; k2 < n; ++k2)
+        if (d == load_u64_una(aux + pr.rhs_off + 8ull * k2)) hit = true;
+      if (!hit) return false;
+      continue;
     } else {
       int64_t a = (int64_t)d, b = (int64_t)pr.datum;
       cmp = a < b ? -1 : (a > b ? 1 : 0);
@@ -2361,9 +2389,12 @@ inline void build_dev_spec(const ybg_scan_spec_t* spec, DevSpec* dp,
     }
     int dt = d.cols[pr.col].dtype;
     pc.opdt = (uint32_t)pr.op | ((uint32_t)dt << 8);
-    pc.datum = (dt == YBG_T_STRING)
-                   ? (((uint64_t)pr.str_len << 32) | pr.rhs_off)
-                   : pr.datum;
+    if (pr.op == YBG_PRED_IN)
+      pc.datum = (((uint64_t)(pr.str_len / 8) << 32) | pr.rhs_off);
+    else if (dt == YBG_T_STRING)
+      pc.datum = (((uint64_t)pr.str_len << 32) | pr.rhs_off);
+    else
+      pc.datum = pr.datum;
     pc.pad_ = 0;
   }
   for (int c = 0; c < sc.num_value_cols; ++c) {

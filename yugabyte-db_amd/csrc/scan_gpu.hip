@@ -598,6 +598,16 @@ int yb_gpu_scan_open(const ybg_scan_spec_t* spec, ybg_scan_t** out) {
     return set_err(2,
                    "no HIP device visible — the GPU product path has no CPU "
                    "fallback (oracle/ is test infrastructure only)");
+  for (int i = 0; i < spec->num_preds; ++i) {
+    const ybg_pred_t& p = spec->preds[i];
+    if (p.op != YBG_PRED_IN) continue;
+    bool is_str = p.is_key_col
+                      ? spec->schema.key_types[p.col] == YBG_KT_STRING
+                      : spec->schema.value_cols[p.col].dtype == YBG_T_STRING;
+    if (is_str) return set_err(9, "IN over string columns not supported");
+    if (p.bytes_len % 8 || !p.bytes)
+      return set_err(9, "IN list must be n x 8-byte datum patterns");
+  }
   auto* s = new ybg_scan();
   s->spec = *spec;
   s->aux_host.resize(1 << 20);
