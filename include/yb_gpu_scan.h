@@ -251,9 +251,11 @@ typedef struct {
  * Valid until the next execute/close on this handle. */
 int yb_gpu_scan_next_batch(ybg_scan_t *s, ybg_row_batch_t *out);
 
-/* Resumable position after row_limit was hit: encoded DocKey of the next
- * undelivered row (pgsql_operation.cc:2796-2806, 2908-2922). Returns length,
- * 0 if the scan is complete. */
+/* Resumable position (pgsql_operation.cc:2796-2806, 2908-2922): for an
+ * unlimited scan reports length 0 (complete). row_limit paging requires
+ * delivered-row ORDER, which the batch ABI leaves to the row-at-a-time
+ * adapter — use yb_host_iter_paging_state (host_iterator.h); calling this
+ * on a limited scan returns an error directing there. */
 int yb_gpu_scan_paging_state(ybg_scan_t *s, uint8_t *key_out, size_t cap,
                              size_t *len_out);
 

@@ -1052,9 +1052,16 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
 
 int yb_gpu_scan_paging_state(ybg_scan_t* s, uint8_t* key_out, size_t cap,
                              size_t* len_out) {
-  (void)s;
   (void)key_out;
   (void)cap;
+  // Paging needs delivered-row ordering, which the batch ABI leaves to the
+  // row-at-a-time adapter (GpuDocRowwiseIterator::PagingState — the
+  // GetSubDocKey analog, ql_rowwise_iterator_interface.h:32-97). An
+  // unlimited scan is simply complete.
+  if (s->spec.row_limit != 0)
+    return set_err(9,
+                   "row_limit paging state is tracked by the iterator "
+                   "adapter (yb_host_iter_paging_state)");
   *len_out = 0;
   return 0;
 }
