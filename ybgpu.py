@@ -337,20 +337,21 @@ class Builder:
         assert rc == 0
         return data, offsets, n_blocks.value, total.value, n_entries.value
 
-    def finish_sst(self):
+    def finish_sst(self, compression=0):
         """Finish as a complete BlockBasedTable SST file; returns
-        (data_ptr, total_bytes, n_blocks, n_entries)."""
+        (data_ptr, total_bytes, n_blocks, n_entries). compression: 0 none,
+        1 snappy (blocks that do not shrink stay uncompressed)."""
         lib = product()
-        f = _sig(lib, "ybg_builder_finish_sst", C.c_int,
-                 [C.c_void_p, C.POINTER(C.POINTER(C.c_uint8)),
+        f = _sig(lib, "ybg_builder_finish_sst2", C.c_int,
+                 [C.c_void_p, C.c_int, C.POINTER(C.POINTER(C.c_uint8)),
                   C.POINTER(C.c_uint64), C.POINTER(C.c_uint64),
                   C.POINTER(C.c_uint64)])
         data = C.POINTER(C.c_uint8)()
         total = C.c_uint64()
         n_blocks = C.c_uint64()
         n_entries = C.c_uint64()
-        rc = f(self._h, C.byref(data), C.byref(total), C.byref(n_blocks),
-               C.byref(n_entries))
+        rc = f(self._h, compression, C.byref(data), C.byref(total),
+               C.byref(n_blocks), C.byref(n_entries))
         assert rc == 0
         return data, total.value, n_blocks.value, n_entries.value
 
@@ -699,3 +700,31 @@ def orcl_scan(data, offsets, n_blocks, schema, spec, kv_format=ENC_THREE_SHARED_
     if rc != 0:
         raise RuntimeError(f"orcl_scan failed rc={rc}")
     return res, rows
+
+
+def snappy_compress(data):
+    """Host snappy compressor (generator codec; tests)."""
+    lib = product()
+    f = _sig(lib, "ybg_snappy_compress", C.c_int64,
+             [C.POINTER(C.c_uint8), C.c_uint64, C.POINTER(C.c_uint8),
+              C.c_uint64])
+    src = (C.c_uint8 * len(data)).from_buffer_copy(data)
+    cap = len(data) * 2 + 64
+    dst = (C.c_uint8 * cap)()
+    n = f(src, len(data), dst, cap)
+    assert n > 0
+    return bytes(dst[:n])
+
+
+def snappy_uncompress(data, cap):
+    """Host snappy decompressor (same code the GPU kernel runs)."""
+    lib = product()
+    f = _sig(lib, "ybg_snappy_uncompress", C.c_int64,
+             [C.POINTER(C.c_uint8), C.c_uint64, C.POINTER(C.c_uint8),
+              C.c_uint64])
+    src = (C.c_uint8 * len(data)).from_buffer_copy(data)
+    dst = (C.c_uint8 * cap)()
+    n = f(src, len(data), dst, cap)
+    if n < 0:
+        raise RuntimeError("snappy_uncompress failed")
+    return bytes(dst[:n])

@@ -43,6 +43,7 @@
 
 #include "scan_device.h"
 #include "sst_internal.h"
+#include "snappy_dev.h"
 
 using namespace ybgdev;
 
@@ -499,6 +500,37 @@ __global__ __launch_bounds__(256) void k_reduce(
   }
 }
 
+// Per-block decompression at feed time: one thread per data block (~1M
+// blocks >> threads), serial snappy decode per thread (snappy_dev.h).
+// type 0 blocks copy through in u64 chunks.
+__global__ __launch_bounds__(256) void k_snappy(
+    const uint8_t* __restrict__ src_blob,
+    const uint64_t* __restrict__ src_off,
+    const uint64_t* __restrict__ src_len,
+    const uint64_t* __restrict__ dst_off, const uint8_t* __restrict__ types,
+    uint64_t n_blocks, uint8_t* __restrict__ out, uint64_t out_cap,
+    unsigned long long* __restrict__ err) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (; i < n_blocks; i += stride) {
+    const uint8_t* src = src_blob + src_off[i];
+    uint64_t n = src_len[i];
+    uint8_t* dst = out + dst_off[i];
+    uint64_t cap = out_cap - dst_off[i];
+    if (types[i] == 1) {
+      if (ybsnappy::snappy_uncompress(src, n, dst, cap) < 0)
+        atomicAdd(err, 1ull);
+    } else {
+      for (uint64_t b = 0; b + 8 <= n; b += 8) {
+        uint64_t w;
+        __builtin_memcpy(&w, src + b, 8);
+        __builtin_memcpy(dst + b, &w, 8);
+      }
+      for (uint64_t b = n & ~7ull; b < n; ++b) dst[b] = src[b];
+    }
+  }
+}
+
 }  // namespace
 
 // ===========================================================================
@@ -722,8 +754,9 @@ int ybg_sst_index(const uint8_t* file, uint64_t size, int verify,
                   uint64_t* offsets, uint64_t* sizes, uint64_t cap,
                   uint64_t* n_blocks) {
   std::vector<uint64_t> offs, szs;
+  std::vector<uint8_t> typs;
   std::string err;
-  int rc = ybsst::parse_sst(file, size, verify, &offs, &szs, &err);
+  int rc = ybsst::parse_sst(file, size, verify, &offs, &szs, &typs, &err);
   if (rc) return set_err(rc, err);
   *n_blocks = offs.size();
   for (uint64_t i = 0; i < offs.size() && i < cap; ++i) {
@@ -739,22 +772,97 @@ int ybg_sst_index(const uint8_t* file, uint64_t size, int verify,
 int yb_gpu_scan_feed_sst(ybg_scan_t* s, const uint8_t* file, uint64_t size,
                          int verify_checksums) {
   std::vector<uint64_t> offs, szs;
+  std::vector<uint8_t> typs;
   std::string err;
-  int rc = ybsst::parse_sst(file, size, verify_checksums, &offs, &szs, &err);
+  int rc = ybsst::parse_sst(file, size, verify_checksums, &offs, &szs, &typs,
+                            &err);
   if (rc) return set_err(rc, err);
   if (offs.empty()) return set_err(3, "SST file holds no data blocks");
-  uint64_t total = 0;
-  for (uint64_t sz : szs) total += sz;
-  std::vector<uint8_t> blocks;
-  blocks.reserve(total);
-  std::vector<uint64_t> boff(offs.size() + 1);
-  boff[0] = 0;
-  for (size_t i = 0; i < offs.size(); ++i) {
-    blocks.insert(blocks.end(), file + offs[i], file + offs[i] + szs[i]);
-    boff[i + 1] = blocks.size();
+  bool any_comp = false;
+  for (uint8_t t : typs) any_comp |= (t == 1);
+  if (!any_comp) {
+    uint64_t total = 0;
+    for (uint64_t sz : szs) total += sz;
+    std::vector<uint8_t> blocks;
+    blocks.reserve(total);
+    std::vector<uint64_t> boff(offs.size() + 1);
+    boff[0] = 0;
+    for (size_t i = 0; i < offs.size(); ++i) {
+      blocks.insert(blocks.end(), file + offs[i], file + offs[i] + szs[i]);
+      boff[i + 1] = blocks.size();
+    }
+    return yb_gpu_scan_feed_blocks(s, blocks.data(), boff.data(),
+                                   offs.size(), 0);
   }
-  return yb_gpu_scan_feed_blocks(s, blocks.data(), boff.data(), offs.size(),
-                                 0);
+  // On-GPU decompression: upload the data-block region once, decompress
+  // every block into its final slot (uncompressed lengths come from each
+  // snappy preamble), then the normal device-resident feed path runs.
+  uint64_t n = offs.size();
+  std::vector<uint64_t> un_len(n), boff(n + 1);
+  boff[0] = 0;
+  for (uint64_t i = 0; i < n; ++i) {
+    if (typs[i] == 1) {
+      const uint8_t* p = file + offs[i];
+      const uint8_t* lim = p + szs[i];
+      uint64_t ulen = 0;
+      int shift = 0;
+      for (;;) {
+        if (p >= lim || shift > 28)
+          return set_err(3, "corrupt snappy preamble");
+        uint8_t b = *p++;
+        ulen |= (uint64_t)(b & 0x7f) << shift;
+        if (!(b & 0x80)) break;
+        shift += 7;
+      }
+      un_len[i] = ulen;
+    } else {
+      un_len[i] = szs[i];
+    }
+    boff[i + 1] = boff[i] + un_len[i];
+  }
+  uint64_t total_un = boff[n];
+  uint64_t blob_len = offs[n - 1] + szs[n - 1];  // blocks start at offset 0
+  uint8_t* d_blob = nullptr;
+  uint64_t* d_meta = nullptr;  // src_off | src_len | dst_off
+  uint8_t* d_types = nullptr;
+  uint8_t* d_out = nullptr;
+  unsigned long long* d_err = nullptr;
+  HIP_TRY(hipMalloc(&d_blob, blob_len));
+  HIP_TRY(hipMemcpy(d_blob, file, blob_len, hipMemcpyHostToDevice));
+  HIP_TRY(hipMalloc(&d_meta, 3 * n * sizeof(uint64_t)));
+  HIP_TRY(hipMemcpy(d_meta, offs.data(), n * 8, hipMemcpyHostToDevice));
+  HIP_TRY(hipMemcpy(d_meta + n, szs.data(), n * 8, hipMemcpyHostToDevice));
+  HIP_TRY(hipMemcpy(d_meta + 2 * n, boff.data(), n * 8,
+                    hipMemcpyHostToDevice));
+  HIP_TRY(hipMalloc(&d_types, n));
+  HIP_TRY(hipMemcpy(d_types, typs.data(), n, hipMemcpyHostToDevice));
+  HIP_TRY(hipMalloc(&d_out, total_un + 48));
+  HIP_TRY(hipMemset(d_out + total_un, 0, 48));
+  HIP_TRY(hipMalloc(&d_err, 8));
+  HIP_TRY(hipMemset(d_err, 0, 8));
+  int sgrid = (int)std::min<uint64_t>((n + 255) / 256, 4096);
+  hipLaunchKernelGGL(k_snappy, dim3(sgrid), dim3(256), 0, s->stream, d_blob,
+                     d_meta, d_meta + n, d_meta + 2 * n, d_types, n, d_out,
+                     total_un, d_err);
+  unsigned long long h_err = 0;
+  HIP_TRY(hipMemcpyAsync(&h_err, d_err, 8, hipMemcpyDeviceToHost,
+                         s->stream));
+  HIP_TRY(hipStreamSynchronize(s->stream));
+  HIP_WARN(hipFree(d_blob));
+  HIP_WARN(hipFree(d_meta));
+  HIP_WARN(hipFree(d_types));
+  HIP_WARN(hipFree(d_err));
+  if (h_err) {
+    HIP_WARN(hipFree(d_out));
+    return set_err(3, "snappy block decompression failed");
+  }
+  rc = yb_gpu_scan_feed_blocks(s, d_out, boff.data(), n, 1);
+  if (rc) {
+    HIP_WARN(hipFree(d_out));
+    return rc;
+  }
+  s->d_data_owned = true;  // the decompressed copy is ours to free
+  return 0;
 }
 
 int yb_gpu_scan_execute(ybg_scan_t* s) {

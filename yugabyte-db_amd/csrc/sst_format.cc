@@ -37,9 +37,10 @@ const uint8_t* get_varint64(const uint8_t* p, const uint8_t* limit,
 
 int parse_sst(const uint8_t* file, uint64_t size, int verify,
               std::vector<uint64_t>* offsets, std::vector<uint64_t>* sizes,
-              std::string* err) {
+              std::vector<uint8_t>* types, std::string* err) {
   offsets->clear();
   sizes->clear();
+  types->clear();
   if (size < kNewFooterLen) {
     *err = "file too small for an SST footer";
     return 3;
@@ -132,16 +133,17 @@ int parse_sst(const uint8_t* file, uint64_t size, int verify,
     }
     offsets->push_back(b_off);
     sizes->push_back(b_sz);
+    types->push_back(file[b_off + b_sz]);
   }
-  if (verify) {
-    for (size_t i = 0; i < offsets->size(); ++i) {
-      const uint8_t* b = file + (*offsets)[i];
-      uint64_t n = (*sizes)[i];
-      uint8_t type = b[n];
-      if (type != 0) {
-        *err = "compressed data block not supported (kNoCompression only)";
-        return 3;
-      }
+  for (size_t i = 0; i < offsets->size(); ++i) {
+    const uint8_t* b = file + (*offsets)[i];
+    uint64_t n = (*sizes)[i];
+    uint8_t type = (*types)[i];
+    if (type > 1) {
+      *err = "unsupported block compression type (none/snappy only)";
+      return 3;
+    }
+    if (verify) {
       uint32_t crc = crc32c_extend(crc32c_value(b, n), &type, 1);
       if (crc32c_mask(crc) != fixed32(b + n + 1)) {
         *err = "data block checksum mismatch";

@@ -65,11 +65,12 @@ inline uint32_t crc32c_mask(uint32_t crc) {
 }
 
 // Parse an SST file: footer -> index block -> data-block handles.
-// On success fills (offset, size) per data block, in file order. verify=1
-// checks every data block's crc32c trailer and kNoCompression type.
+// On success fills (offset, size, trailer type) per data block, in file
+// order. verify=1 checks every data block's crc32c trailer; types other
+// than kNoCompression(0) / kSnappyCompression(1) are rejected.
 int parse_sst(const uint8_t* file, uint64_t size, int verify,
               std::vector<uint64_t>* offsets, std::vector<uint64_t>* sizes,
-              std::string* err);
+              std::vector<uint8_t>* types, std::string* err);
 
 }  // namespace ybsst
 

@@ -10,6 +10,7 @@
 #include "../../include/yb_gpu_scan.h"
 #include "codec.h"
 #include "sst_internal.h"
+#include "snappy_dev.h"
 
 #include <algorithm>
 #include <atomic>
@@ -531,28 +532,45 @@ int ybg_builder_finish(ybg_builder_t* b, const uint8_t** data,
 // [type|crc32c] trailers, an empty metaindex block, a shared-prefix index
 // block (restart interval 1) whose values are BlockHandles, and the
 // version-2 footer. Format citations: sst_internal.h.
-int ybg_builder_finish_sst(ybg_builder_t* b, const uint8_t** data,
-                           uint64_t* total_bytes, uint64_t* n_blocks,
-                           uint64_t* n_entries) {
+int ybg_builder_finish_sst2(ybg_builder_t* b, int compression,
+                            const uint8_t** data, uint64_t* total_bytes,
+                            uint64_t* n_blocks, uint64_t* n_entries) {
   using namespace ybsst;
   b->FlushBlock();
   ybg::Buf& f = b->sst_file;
   f.clear();
   uint64_t nb = b->offsets.size() - 1;
-  auto append_block = [&](const uint8_t* p, uint64_t n, uint64_t* h_off,
-                          uint64_t* h_sz) {
+  auto append_block_t = [&](const uint8_t* p, uint64_t n, uint8_t type,
+                            uint64_t* h_off, uint64_t* h_sz) {
     *h_off = f.size();
     *h_sz = n;
     f.insert(f.end(), p, p + n);
-    uint8_t type = 0;  // kNoCompression
     uint32_t crc = crc32c_extend(crc32c_value(p, n), &type, 1);
     f.push_back(type);
     ybg::Fixed32LEAppend(crc32c_mask(crc), &f);
   };
+  auto append_block = [&](const uint8_t* p, uint64_t n, uint64_t* h_off,
+                          uint64_t* h_sz) {
+    append_block_t(p, n, 0 /* kNoCompression */, h_off, h_sz);
+  };
   std::vector<uint64_t> h_off(nb), h_sz(nb);
-  for (uint64_t i = 0; i < nb; ++i)
-    append_block(b->all_blocks.data() + b->offsets[i],
-                 b->offsets[i + 1] - b->offsets[i], &h_off[i], &h_sz[i]);
+  ybg::Buf cbuf;
+  for (uint64_t i = 0; i < nb; ++i) {
+    const uint8_t* bp = b->all_blocks.data() + b->offsets[i];
+    uint64_t bn = b->offsets[i + 1] - b->offsets[i];
+    if (compression == 1 /* kSnappyCompression */) {
+      cbuf.resize(bn + bn / 2 + 32);
+      int64_t cn = ybsnappy::snappy_compress(bp, bn, cbuf.data(),
+                                             cbuf.size());
+      // like the reference, keep the block uncompressed when compression
+      // does not shrink it (CompressBlock GoodCompressionRatio)
+      if (cn > 0 && (uint64_t)cn < bn) {
+        append_block_t(cbuf.data(), (uint64_t)cn, 1, &h_off[i], &h_sz[i]);
+        continue;
+      }
+    }
+    append_block(bp, bn, &h_off[i], &h_sz[i]);
+  }
   // empty metaindex block: restart array only (block_builder.cc Finish)
   uint64_t mi_off, mi_sz;
   {
@@ -600,6 +618,23 @@ int ybg_builder_finish_sst(ybg_builder_t* b, const uint8_t** data,
   *n_blocks = nb;
   *n_entries = b->n_entries;
   return 0;
+}
+
+int ybg_builder_finish_sst(ybg_builder_t* b, const uint8_t** data,
+                           uint64_t* total_bytes, uint64_t* n_blocks,
+                           uint64_t* n_entries) {
+  return ybg_builder_finish_sst2(b, 0, data, total_bytes, n_blocks,
+                                 n_entries);
+}
+
+// Host snappy codec exports (tests pin the codec by round-trip)
+int64_t ybg_snappy_compress(const uint8_t* src, uint64_t n, uint8_t* dst,
+                            uint64_t cap) {
+  return ybsnappy::snappy_compress(src, n, dst, cap);
+}
+int64_t ybg_snappy_uncompress(const uint8_t* src, uint64_t n, uint8_t* dst,
+                              uint64_t cap) {
+  return ybsnappy::snappy_uncompress(src, n, dst, cap);
 }
 
 void ybg_builder_destroy(ybg_builder_t* b) { delete b; }
