@@ -32,26 +32,42 @@ constexpr size_t kBlockTrailerSize = 5;
 constexpr size_t kNewFooterLen = 53;  // 1 + 2*20 + 4 + 8
 constexpr uint32_t kCrcMaskDelta = 0xa282ead8ul;
 
-// crc32c (Castagnoli, reflected poly 0x82F63B78), bytewise table.
-inline const uint32_t* crc32c_table() {
-  static uint32_t t[256];
+// crc32c (Castagnoli, reflected poly 0x82F63B78), slicing-by-8 tables —
+// the bytewise version measured ~0.4 GB/s on the host verify path, which
+// dominated compressed-SST feed time.
+inline const uint32_t (*crc32c_tables())[256] {
+  static uint32_t t[8][256];
   static bool init = false;
   if (!init) {
     for (uint32_t i = 0; i < 256; ++i) {
       uint32_t c = i;
       for (int k = 0; k < 8; ++k)
         c = (c & 1) ? (0x82F63B78u ^ (c >> 1)) : (c >> 1);
-      t[i] = c;
+      t[0][i] = c;
     }
+    for (int s = 1; s < 8; ++s)
+      for (uint32_t i = 0; i < 256; ++i)
+        t[s][i] = t[0][t[s - 1][i] & 0xff] ^ (t[s - 1][i] >> 8);
     init = true;
   }
   return t;
 }
 
 inline uint32_t crc32c_extend(uint32_t crc, const uint8_t* p, size_t n) {
-  const uint32_t* t = crc32c_table();
+  const uint32_t(*t)[256] = crc32c_tables();
   uint32_t c = crc ^ 0xffffffffu;
-  for (size_t i = 0; i < n; ++i) c = t[(c ^ p[i]) & 0xff] ^ (c >> 8);
+  while (n >= 8) {
+    uint32_t lo, hi;
+    __builtin_memcpy(&lo, p, 4);
+    __builtin_memcpy(&hi, p + 4, 4);
+    lo ^= c;
+    c = t[7][lo & 0xff] ^ t[6][(lo >> 8) & 0xff] ^ t[5][(lo >> 16) & 0xff] ^
+        t[4][lo >> 24] ^ t[3][hi & 0xff] ^ t[2][(hi >> 8) & 0xff] ^
+        t[1][(hi >> 16) & 0xff] ^ t[0][hi >> 24];
+    p += 8;
+    n -= 8;
+  }
+  for (size_t i = 0; i < n; ++i) c = t[0][(c ^ p[i]) & 0xff] ^ (c >> 8);
   return c ^ 0xffffffffu;
 }
 
