@@ -1920,7 +1920,10 @@ DEV void group_accum(const DevSpec& sp, const GroupCtx& gc,
       default:
         break;  // MIN/MAX double unsupported in group mode (open() rejects)
     }
-    atomicAdd(&c[g], 1ull);
+    // COUNT ops carry their contribution count in the value itself; the
+    // export reconstructs cnt = val for them (one fewer atomic per row)
+    if (op != YBG_AGG_COUNT_STAR && op != YBG_AGG_COUNT)
+      atomicAdd(&c[g], 1ull);
   }
 }
 
@@ -1976,7 +1979,8 @@ DEV void group_accum_rec(const DevSpec& sp, const GroupCtx& gc,
       default:
         break;
     }
-    atomicAdd(&c[g], 1ull);
+    if (op != YBG_AGG_COUNT_STAR && op != YBG_AGG_COUNT)
+      atomicAdd(&c[g], 1ull);
   }
 }
 

@@ -373,8 +373,14 @@ __global__ void k_group_export(DevSpec sp, GroupCtx gc,
     }
     out_keys[slot] = kv;
     for (int g = 0; g < sp.num_aggs; ++g) {
-      out_vals[slot * YBG_MAX_AGGS + g] = gc.vals[i * YBG_MAX_AGGS + g];
-      out_cnts[slot * YBG_MAX_AGGS + g] = gc.cnts[i * YBG_MAX_AGGS + g];
+      int op = sp.agg_op[g];
+      long long v = gc.vals[i * YBG_MAX_AGGS + g];
+      out_vals[slot * YBG_MAX_AGGS + g] = v;
+      // COUNT ops skip the per-row cnt atomic: cnt == val by definition
+      out_cnts[slot * YBG_MAX_AGGS + g] =
+          (op == YBG_AGG_COUNT_STAR || op == YBG_AGG_COUNT)
+              ? (unsigned long long)v
+              : gc.cnts[i * YBG_MAX_AGGS + g];
     }
   }
 }

@@ -278,8 +278,13 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
     }
     keys_out[n] = kv;
     for (int g = 0; g < YBG_MAX_AGGS; ++g) {
-      vals_out[n * YBG_MAX_AGGS + g] = vals[i * YBG_MAX_AGGS + g];
-      cnts_out[n * YBG_MAX_AGGS + g] = cnts[i * YBG_MAX_AGGS + g];
+      int op = g < d.num_aggs ? d.agg_op[g] : -1;
+      long long v = vals[i * YBG_MAX_AGGS + g];
+      vals_out[n * YBG_MAX_AGGS + g] = v;
+      cnts_out[n * YBG_MAX_AGGS + g] =
+          (op == YBG_AGG_COUNT_STAR || op == YBG_AGG_COUNT)
+              ? (unsigned long long)v
+              : cnts[i * YBG_MAX_AGGS + g];
     }
     ++n;
   }
