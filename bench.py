@@ -82,6 +82,7 @@ def main():
     # ---- dataset ---------------------------------------------------------
     versions = 1
     group_col = 0
+    read_micros = 1_700_000_000_000_000
     if args.workload == "filtersum":
         schema = y.make_schema([y.KT_INT64],
                                [(10 + i, y.T_INT64, 1) for i in range(4)])
@@ -96,6 +97,11 @@ def main():
         preds = []
         aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 0)]
         versions = 5
+        # read BETWEEN versions 2 and 3 (versions at ht_base + k*1e6 us):
+        # the two newer versions per row are in the future and must be
+        # skipped (SkipFutureRecords semantics), the newest visible one is
+        # deduped — the visibility work config 4 names, not just dedup
+        read_micros = 1_600_000_002_500_000
     else:  # groupby — BASELINE configs[4]-shaped (mixed row, grouped)
         schema = y.make_schema(
             [y.KT_INT64],
@@ -112,7 +118,7 @@ def main():
         spec.schema = schema
         spec.kv_format = y.ENC_THREE_SHARED_PARTS
         spec.group_col = group_col
-        spec.read_time = y.read_time(1_700_000_000_000_000)
+        spec.read_time = y.read_time(read_micros)
         spec.num_preds = len(preds)
         for i, p in enumerate(preds):
             spec.preds[i] = p
@@ -261,7 +267,7 @@ def main():
         # calibrate on 256 blocks, then size the sample for the budget
         osc = y.orcl_schema_from(schema)
         ospec = y.OrclScanSpec()
-        ospec.read_time = y.orcl_read_time(1_700_000_000_000_000)
+        ospec.read_time = y.orcl_read_time(read_micros)
         ospec.num_preds = len(preds)
         for i, p in enumerate(preds):
             ospec.preds[i] = y.OrclPred(p.is_key_col, p.col, p.op, p.datum,
