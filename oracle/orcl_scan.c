@@ -34,6 +34,23 @@
 #define V2_HAS_NULLS_FLAG 1  /* src/yb/dockv/packed_row.h:197 */
 
 /* lexicographic slice compare, rocksdb Slice::compare semantics */
+/* escape-aware compare: lhs = zero-escaped key bytes (00 -> 00 01,
+ * doc_kv_util.h:101-167), rhs = plain bytes */
+static int key_slice_cmp(const uint8_t *e, size_t elen, const uint8_t *b,
+                         size_t blen) {
+  size_t si = 0, ri = 0;
+  while (si < elen && ri < blen) {
+    uint8_t cb = e[si];
+    si += (cb == 0) ? 2 : 1;
+    if (cb != b[ri]) return cb < b[ri] ? -1 : 1;
+    ++ri;
+  }
+  {
+    int le = si >= elen, re = ri >= blen;
+    return (le && re) ? 0 : (le ? -1 : 1);
+  }
+}
+
 static int slice_cmp(const uint8_t *a, size_t alen, const uint8_t *b,
                      size_t blen) {
   size_t n = alen < blen ? alen : blen;
@@ -392,8 +409,10 @@ static int pred_eval(const orcl_pred_t *pr, const orcl_schema_t *sc,
   if (pr->is_key_col) {
     orcl_keytype_t kt = sc->key_types[pr->col];
     if (kt == ORCL_KT_STRING) {
-      cmp = slice_cmp(row->key_str[pr->col], row->key_str_len[pr->col],
-                      pr->bytes, pr->bytes_len);
+      /* key_str points at the zero-ESCAPED bytes inside the encoded key
+       * (doc_kv_util.h:101-167); compare unescaping on the fly */
+      cmp = key_slice_cmp(row->key_str[pr->col], row->key_str_len[pr->col],
+                          pr->bytes, pr->bytes_len);
       numeric = 0;
     } else {
       lhs_i = (int64_t)row->key_datums[pr->col];
@@ -426,9 +445,34 @@ static int pred_eval(const orcl_pred_t *pr, const orcl_schema_t *sc,
     }
   }
   if (pr->op == ORCL_PRED_IN) {
-    /* membership over the option list (numeric columns only) */
+    /* membership over the option list */
     size_t n = pr->bytes_len / 8, i;
-    if (numeric == 0) return 0;
+    if (numeric == 0) {
+      /* string options: [u32 LE length][bytes] records */
+      const uint8_t *lp;
+      size_t ll;
+      if (pr->is_key_col) {
+        lp = row->key_str[pr->col];
+        ll = row->key_str_len[pr->col];
+      } else {
+        lp = row->strp[pr->col];
+        ll = row->strlen_[pr->col];
+      }
+      const uint8_t *q = pr->bytes, *qe = pr->bytes + pr->bytes_len;
+      while (q + 4 <= qe) {
+        uint32_t ol;
+        memcpy(&ol, q, 4);
+        q += 4;
+        if (q + ol > qe) return 0;
+        if (pr->is_key_col) {
+          if (key_slice_cmp(lp, ll, q, ol) == 0) return 1;
+        } else if (ol == ll && memcmp(q, lp, ol) == 0) {
+          return 1;
+        }
+        q += ol;
+      }
+      return 0;
+    }
     for (i = 0; i < n; ++i) {
       uint64_t raw;
       memcpy(&raw, pr->bytes + 8 * i, 8);

@@ -206,6 +206,36 @@ def build_cases():
                         (1_700_000_000_000_000, preds_k, aggs),
                         (1_700_000_000_000_000, preds_both, aggs)]))
 
+    # string IN options: value column (plain bytes) and string KEY column
+    # (zero-escaped in the rowkey — the escape-aware compare path)
+    schema_si = y.make_schema([y.KT_INT64, y.KT_STRING],
+                              [(10, y.T_STRING, 1), (11, y.T_INT64, 1)],
+                              num_hash_cols=1)
+    b = y.Builder(schema_si)
+    for r in range(4000):
+        sval = b"val-%04d" % (r % 50)
+        skey = (b"k\x00z-%03d" % (r % 97)) if r % 3 == 0 else \
+            (b"key-%04d" % r)
+        b.add_packed_row(1000 + r, [(y.T_STRING, sval), (y.T_INT64, r)],
+                         hash_=r // 64, key_datums=(r, 0),
+                         key_strs=(None, skey))
+    _KEEP.append(b)
+
+    def _strlist(opts):
+        out = b"".join(_struct.pack("<I", len(o)) + o for o in opts)
+        arr = (_C.c_uint8 * len(out)).from_buffer_copy(out)
+        _KEEP.append(arr)
+        return arr, len(out)
+
+    a3, l3 = _strlist([b"val-0003", b"val-0017", b"no-such"])
+    a4, l4 = _strlist([b"k\x00z-005", b"key-0100", b"absent"])
+    preds_sv = [y.Pred(0, 0, y.PRED_IN, 0, a3, l3)]
+    preds_sk = [y.Pred(1, 1, y.PRED_IN, 0, a4, l4)]
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 1)]
+    cases.append(_case("in_list_string", schema_si, b.finish(),
+                       [(1_700_000_000_000_000, preds_sv, aggs),
+                        (1_700_000_000_000_000, preds_sk, aggs)]))
+
     # single row + empty result
     b = y.Builder(SCHEMA_4I)
     b.add_packed_row(1000, [(y.T_INT64, i) for i in range(4)], hash_=7,

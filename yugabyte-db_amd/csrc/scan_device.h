@@ -1043,11 +1043,24 @@ DEV bool pred_compare(const PredC& pr, uint64_t datum, const uint8_t* sptr,
   int dtype = (int)(pr.opdt >> 8);
   int cmp;
   if (YBG_UNLIKELY((pr.opdt & 0xff) == YBG_PRED_IN)) {
-    // membership over the option list (hybrid_scan_choices.h:43-60);
-    // numeric columns only (open() rejects strings). datum packs
-    // (count << 32) | aux offset of the 8-byte LE option patterns.
+    // membership over the option list (hybrid_scan_choices.h:43-60).
+    // datum packs (count << 32) | aux offset. Numeric options are 8-byte
+    // LE patterns; string options are [u32 LE length][bytes] records.
     uint32_t n = (uint32_t)(pr.datum >> 32);
     const uint8_t* lst = aux + (uint32_t)pr.datum;
+    if (dtype == YBG_T_STRING) {
+      for (uint32_t i = 0; i < n; ++i) {
+        uint32_t ol = (uint32_t)load_u64_una(lst) & 0xffffffffu;
+        const uint8_t* ob = lst + 4;
+        lst += 4 + ol;
+        if (ol != slen) continue;
+        bool eq = true;
+        for (uint32_t k = 0; k < ol; ++k)
+          if (sptr[k] != ob[k]) { eq = false; break; }
+        if (eq) return true;
+      }
+      return false;
+    }
     for (uint32_t i = 0; i < n; ++i) {
       uint64_t rv = load_u64_una(lst + 8ull * i);
       if (dtype == YBG_T_DOUBLE) {
@@ -1363,6 +1376,31 @@ DEV bool eval_key_preds(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
     uint32_t soff = 0, sl = 0;
     if (!key_col_value(sp, rk, rk_len, pr.col, &d, &soff, &sl)) return false;
     int cmp;
+    if (sp.key_types[pr.col] == YBG_KT_STRING && pr.op == YBG_PRED_IN) {
+      // option membership over [u32 len][bytes] records; the key bytes
+      // are zero-escaped in the rowkey (doc_kv_util.h:101-167) so each
+      // candidate compare unescapes on the fly like the EQ path
+      const uint8_t* lst = aux + pr.rhs_off;
+      const uint8_t* lend = lst + pr.str_len;
+      bool hit = false;
+      while (lst + 4 <= lend && !hit) {
+        uint32_t ol = (uint32_t)load_u64_una(lst) & 0xffffffffu;
+        const uint8_t* ob = lst + 4;
+        lst += 4 + ol;
+        const uint8_t* lstr = rk + soff;
+        uint32_t si = 0, ri = 0;
+        bool eq = true;
+        while (si < sl && ri < ol) {
+          uint8_t cb = lstr[si];
+          si += (cb == 0) ? 2 : 1;
+          if (cb != ob[ri]) { eq = false; break; }
+          ++ri;
+        }
+        if (eq && si >= sl && ri >= ol) hit = true;
+      }
+      if (!hit) return false;
+      continue;
+    }
     if (sp.key_types[pr.col] == YBG_KT_STRING) {
       const uint8_t* lstr = rk + soff;
       const uint8_t* rhs = aux + pr.rhs_off;
@@ -2393,9 +2431,23 @@ inline void build_dev_spec(const ybg_scan_spec_t* spec, DevSpec* dp,
     }
     int dt = d.cols[pr.col].dtype;
     pc.opdt = (uint32_t)pr.op | ((uint32_t)dt << 8);
-    if (pr.op == YBG_PRED_IN)
-      pc.datum = (((uint64_t)(pr.str_len / 8) << 32) | pr.rhs_off);
-    else if (dt == YBG_T_STRING)
+    if (pr.op == YBG_PRED_IN) {
+      uint64_t cnt;
+      if (dt == YBG_T_STRING) {
+        // [u32 len][bytes] records
+        cnt = 0;
+        uint32_t o = 0;
+        while (o + 4 <= pr.str_len) {
+          uint32_t ol;
+          memcpy(&ol, aux + pr.rhs_off + o, 4);
+          o += 4 + ol;
+          ++cnt;
+        }
+      } else {
+        cnt = pr.str_len / 8;
+      }
+      pc.datum = ((cnt << 32) | pr.rhs_off);
+    } else if (dt == YBG_T_STRING)
       pc.datum = (((uint64_t)pr.str_len << 32) | pr.rhs_off);
     else
       pc.datum = pr.datum;

@@ -642,9 +642,21 @@ int yb_gpu_scan_open(const ybg_scan_spec_t* spec, ybg_scan_t** out) {
     bool is_str = p.is_key_col
                       ? spec->schema.key_types[p.col] == YBG_KT_STRING
                       : spec->schema.value_cols[p.col].dtype == YBG_T_STRING;
-    if (is_str) return set_err(9, "IN over string columns not supported");
-    if (p.bytes_len % 8 || !p.bytes)
+    if (!p.bytes)
+      return set_err(9, "IN list missing");
+    if (is_str) {
+      // [u32 LE length][bytes] records, exactly covering bytes_len
+      uint64_t o = 0;
+      while (o + 4 <= p.bytes_len) {
+        uint32_t ol;
+        memcpy(&ol, p.bytes + o, 4);
+        o += 4ull + ol;
+      }
+      if (o != p.bytes_len)
+        return set_err(9, "malformed string IN list");
+    } else if (p.bytes_len % 8) {
       return set_err(9, "IN list must be n x 8-byte datum patterns");
+    }
   }
   auto* s = new ybg_scan();
   s->spec = *spec;
