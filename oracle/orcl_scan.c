@@ -666,6 +666,23 @@ static int finalize_row(scan_ctx_t *cx, row_state_t *st) {
     agg_update(&cx->res->aggs[g], &spec->aggs[g], sc, &row);
   }
   if (cx->cb) {
+    /* emitted rows expose DECODED key strings: unescape the zero-encoded
+     * rowkey bytes (doc_kv_util.h:101-167) into per-column scratch. The
+     * predicate/bound paths above keep the escaped in-key form. */
+    static uint8_t key_unesc[ORCL_MAX_KEYCOLS][256];
+    int nk = sc->num_hash_cols + sc->num_range_cols;
+    for (int kc = 0; kc < nk; ++kc) {
+      if (sc->key_types[kc] != ORCL_KT_STRING || !row.key_str[kc]) continue;
+      const uint8_t *e = row.key_str[kc];
+      size_t elen = row.key_str_len[kc], o = 0;
+      for (size_t si = 0; si < elen && o < sizeof(key_unesc[0]); ) {
+        uint8_t cb = e[si];
+        key_unesc[kc][o++] = cb;
+        si += (cb == 0) ? 2 : 1;
+      }
+      row.key_str[kc] = key_unesc[kc];
+      row.key_str_len[kc] = (uint32_t)o;
+    }
     row.seq_in_scan = cx->row_seq++;
     if (cx->cb(&row, cx->cb_arg)) return 1;
   }
