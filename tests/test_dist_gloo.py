@@ -89,3 +89,90 @@ def test_tablet_sharding_allreduce():
         tot_cnt += res.aggs[1].value_i64
 
     assert got == (tot_sum, tot_cnt)
+
+
+def _group_worker(rank, world, port, out_q):
+    sys.path.insert(0, ROOT)
+    import torch.distributed as dist
+    import ybgpu as y
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    schema = y.make_schema(
+        [y.KT_INT64],
+        [(10, y.T_INT64, 1), (11, y.T_INT64, 1)])
+    my_tablets = [t for t in range(8) if t % world == rank]
+    partial = {}
+    for t in my_tablets:
+        data, offsets, nb, total, ne = y.generate(
+            schema, rows=1500, seed=42 + t, nthreads=1, group_mod=64)
+        spec = y.ScanSpec()
+        spec.schema = schema
+        spec.kv_format = y.ENC_THREE_SHARED_PARTS
+        spec.read_time = y.read_time(1_700_000_000_000_000)
+        spec.group_col = 1
+        spec.num_aggs = 2
+        spec.aggs[0] = y.Agg(y.AGG_COUNT_STAR, 0)
+        spec.aggs[1] = y.Agg(y.AGG_SUM_INT64, 1)
+        for key, row in y.sim_group(spec, data, offsets, nb).items():
+            # row = [COUNT, SUM] (None when no contribution)
+            c0 = row[0] or 0
+            s1 = row[1] or 0
+            if key in partial:
+                partial[key] = (partial[key][0] + c0, partial[key][1] + s1)
+            else:
+                partial[key] = (c0, s1)
+
+    # the cross-tablet GROUP-TABLE merge: all_gather the partial tables,
+    # fold in rank order (SURVEY §8e — config 5's collective)
+    gathered = [None] * world
+    dist.all_gather_object(gathered, partial)
+    if rank == 0:
+        merged = {}
+        for part in gathered:
+            for key, (c0, s1) in part.items():
+                mc, ms = merged.get(key, (0, 0))
+                merged[key] = (mc + c0, ms + s1)
+        out_q.put(merged)
+    dist.destroy_process_group()
+
+
+def test_group_table_merge_across_ranks():
+    """Config-5 semantics at N>1: per-rank GROUP BY partial tables merged
+    across ranks equal the oracle's grouped result over all 8 tablets."""
+    import ybgpu as y
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_group_worker, args=(r, 2, 29523, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=300)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    schema = y.make_schema(
+        [y.KT_INT64],
+        [(10, y.T_INT64, 1), (11, y.T_INT64, 1)])
+    want = {}
+    for t in range(8):
+        data, offsets, nb, total, ne = y.generate(
+            schema, rows=1500, seed=42 + t, nthreads=1, group_mod=64)
+        osc = y.orcl_schema_from(schema)
+        ospec = y.OrclScanSpec()
+        ospec.read_time = y.orcl_read_time(1_700_000_000_000_000)
+        ospec.num_aggs = 2
+        ospec.aggs[0] = y.OrclAgg(y.AGG_COUNT_STAR, 0)
+        ospec.aggs[1] = y.OrclAgg(y.AGG_SUM_INT64, 1)
+        for key, row in y.orcl_group(
+                data, offsets, nb, osc, ospec, group_col=0,
+                num_aggs=2, aggs=ospec.aggs).items():
+            c0 = row[0] or 0
+            s1 = row[1] or 0
+            wc, ws = want.get(key, (0, 0))
+            want[key] = (wc + c0, ws + s1)
+    assert got == want
