@@ -117,6 +117,13 @@ typedef enum {
    * scan the discrete options become a filter, not a seek plan. String
    * columns are rejected at open(). */
   YBG_PRED_IN,
+  /* Tuple membership over MULTIPLE numeric KEY columns — the reference's
+   * multi-column option groups ((r1,r3) IN ((1,3),(5,6)) ...),
+   * docdb/hybrid_scan_choices.h:43-77. bytes layout:
+   *   [u32 LE ncols][u32 LE key-col index x ncols]
+   *   [tuples: ncols x 8-byte LE datum patterns each]
+   * is_key_col must be 1; col is ignored. */
+  YBG_PRED_IN_TUPLE,
 } ybg_pred_op_t;
 
 typedef struct {

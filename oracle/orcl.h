@@ -211,6 +211,10 @@ typedef enum {
   /* IN list over a numeric column: bytes = n x 8-byte LE datum patterns
    * (hybrid_scan_choices.h:43-60; ql_scanspec.cc:323-346) */
   ORCL_PRED_IN,
+  /* tuple membership over multiple numeric key columns
+   * (hybrid_scan_choices.h:43-77); bytes = [u32 ncols][u32 colidx x n]
+   * [tuples of n x 8-byte LE datums] */
+  ORCL_PRED_IN_TUPLE,
 } orcl_pred_op_t;
 
 typedef struct {

@@ -406,6 +406,31 @@ static int pred_eval(const orcl_pred_t *pr, const orcl_schema_t *sc,
   int64_t lhs_i = 0, rhs_i = 0;
   int numeric = 1;
   int cmp;
+  if (pr->op == ORCL_PRED_IN_TUPLE) {
+    /* multi-column option group (hybrid_scan_choices.h:43-77) */
+    uint32_t nc, t, c;
+    const uint8_t *cols, *tup;
+    size_t tup_sz, ntup;
+    if (pr->bytes_len < 4) return 0;
+    memcpy(&nc, pr->bytes, 4);
+    cols = pr->bytes + 4;
+    tup = cols + 4ull * nc;
+    tup_sz = 8ull * nc;
+    if (!tup_sz || pr->bytes_len < 4 + 4ull * nc) return 0;
+    ntup = (pr->bytes_len - 4 - 4ull * nc) / tup_sz;
+    for (t = 0; t < ntup; ++t) {
+      int all = 1;
+      for (c = 0; c < nc && all; ++c) {
+        uint32_t ci;
+        uint64_t want;
+        memcpy(&ci, cols + 4ull * c, 4);
+        memcpy(&want, tup + t * tup_sz + 8ull * c, 8);
+        if (row->key_datums[ci] != want) all = 0;
+      }
+      if (all) return 1;
+    }
+    return 0;
+  }
   if (pr->is_key_col) {
     orcl_keytype_t kt = sc->key_types[pr->col];
     if (kt == ORCL_KT_STRING) {

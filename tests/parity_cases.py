@@ -236,6 +236,26 @@ def build_cases():
                        [(1_700_000_000_000_000, preds_sv, aggs),
                         (1_700_000_000_000_000, preds_sk, aggs)]))
 
+    # multi-column tuple options ((k0,k1) IN ((3,103),(7,257),(3,9999)) —
+    # the reference's multi-column option groups)
+    schema_t2 = y.make_schema([y.KT_INT64, y.KT_INT64],
+                              [(10, y.T_INT64, 1)], num_hash_cols=1)
+    b = y.Builder(schema_t2)
+    for r in range(4000):
+        b.add_packed_row(1000 + r, [(y.T_INT64, r * 3)],
+                         hash_=r // 64, key_datums=(r % 50, r))
+    _KEEP.append(b)
+    tuples = [(3, 103), (7, 257), (3, 9999)]
+    tb = _struct.pack("<I", 2) + _struct.pack("<II", 0, 1)
+    for t2 in tuples:
+        tb += _struct.pack("<QQ", t2[0] & (2**64 - 1), t2[1] & (2**64 - 1))
+    at = (_C.c_uint8 * len(tb)).from_buffer_copy(tb)
+    _KEEP.append(at)
+    preds_t = [y.Pred(1, 0, y.PRED_IN_TUPLE, 0, at, len(tb))]
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 0)]
+    cases.append(_case("in_tuple_keycols", schema_t2, b.finish(),
+                       [(1_700_000_000_000_000, preds_t, aggs)]))
+
     # single row + empty result
     b = y.Builder(SCHEMA_4I)
     b.add_packed_row(1000, [(y.T_INT64, i) for i in range(4)], hash_=7,

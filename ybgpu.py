@@ -24,7 +24,8 @@ T_BOOL, T_INT8, T_INT16, T_INT32, T_INT64 = 0, 1, 2, 3, 4
 T_UINT32, T_UINT64, T_FLOAT, T_DOUBLE, T_STRING = 5, 6, 7, 8, 9
 KT_INT64, KT_INT32, KT_STRING = 0, 1, 2
 ENC_SHARED_PREFIX, ENC_THREE_SHARED_PARTS = 0, 1
-PRED_GT, PRED_GE, PRED_LT, PRED_LE, PRED_EQ, PRED_NE, PRED_IN = range(7)
+PRED_GT, PRED_GE, PRED_LT, PRED_LE, PRED_EQ, PRED_NE, PRED_IN, \
+    PRED_IN_TUPLE = range(8)
 (AGG_COUNT, AGG_COUNT_STAR, AGG_SUM_INT64, AGG_SUM_DOUBLE,
  AGG_MIN_INT64, AGG_MAX_INT64, AGG_MIN_DOUBLE, AGG_MAX_DOUBLE) = range(8)
 

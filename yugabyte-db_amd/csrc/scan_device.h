@@ -1372,6 +1372,33 @@ DEV bool eval_key_preds(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
     int i = __builtin_ctz(kp);
     kp &= kp - 1;
     const DevPred& pr = sp.preds[i];
+    if (YBG_UNLIKELY(pr.op == YBG_PRED_IN_TUPLE)) {
+      // multi-column option group (hybrid_scan_choices.h:43-77):
+      // [u32 ncols][u32 colidx x n][tuples of n x u64]
+      const uint8_t* q = aux + pr.rhs_off;
+      uint32_t nc = (uint32_t)load_u64_una(q) & 0xffffffffu;
+      const uint8_t* cols = q + 4;
+      const uint8_t* tup = cols + 4ull * nc;
+      uint32_t tup_sz = 8 * nc;
+      uint32_t ntup =
+          tup_sz ? (uint32_t)((pr.str_len - 4 - 4ull * nc) / tup_sz) : 0;
+      bool hit = false;
+      for (uint32_t t = 0; t < ntup && !hit; ++t) {
+        bool all = true;
+        for (uint32_t c = 0; c < nc && all; ++c) {
+          uint32_t ci = (uint32_t)load_u64_una(cols + 4ull * c) & 0xffffffffu;
+          uint64_t dv = 0;
+          uint32_t so = 0, sn = 0;
+          if (!key_col_value(sp, rk, rk_len, (int)ci, &dv, &so, &sn))
+            return false;
+          if (dv != load_u64_una(tup + (uint64_t)t * tup_sz + 8ull * c))
+            all = false;
+        }
+        hit = all;
+      }
+      if (!hit) return false;
+      continue;
+    }
     uint64_t d = 0;
     uint32_t soff = 0, sl = 0;
     if (!key_col_value(sp, rk, rk_len, pr.col, &d, &soff, &sl)) return false;

@@ -638,6 +638,25 @@ int yb_gpu_scan_open(const ybg_scan_spec_t* spec, ybg_scan_t** out) {
                    "fallback (oracle/ is test infrastructure only)");
   for (int i = 0; i < spec->num_preds; ++i) {
     const ybg_pred_t& p = spec->preds[i];
+    if (p.op == YBG_PRED_IN_TUPLE) {
+      if (!p.is_key_col || !p.bytes || p.bytes_len < 4)
+        return set_err(9, "IN_TUPLE requires key columns and a tuple list");
+      uint32_t nc;
+      memcpy(&nc, p.bytes, 4);
+      int nk = spec->schema.num_hash_cols + spec->schema.num_range_cols;
+      if (nc == 0 || nc > (uint32_t)nk ||
+          p.bytes_len < 4 + 4ull * nc ||
+          (p.bytes_len - 4 - 4ull * nc) % (8ull * nc))
+        return set_err(9, "malformed IN_TUPLE list");
+      for (uint32_t c = 0; c < nc; ++c) {
+        uint32_t ci;
+        memcpy(&ci, p.bytes + 4 + 4ull * c, 4);
+        if (ci >= (uint32_t)nk ||
+            spec->schema.key_types[ci] == YBG_KT_STRING)
+          return set_err(9, "IN_TUPLE columns must be numeric key columns");
+      }
+      continue;
+    }
     if (p.op != YBG_PRED_IN) continue;
     bool is_str = p.is_key_col
                       ? spec->schema.key_types[p.col] == YBG_KT_STRING
