@@ -6,6 +6,8 @@ tombstones, column updates, NULLs, string columns, both KV encodings,
 restart intervals 1/4/16, small blocks (rows straddling intervals and
 blocks), typed/IN predicates and read-time sweeps."""
 import ctypes as C
+
+import pytest
 import random
 import struct
 
@@ -149,4 +151,44 @@ def test_fuzz_sim_vs_oracle():
             except AssertionError as e:
                 raise AssertionError(
                     f"fuzz iter {it} run {run} (rows={rows}, "
+                    f"dtypes={dtypes}, read={read}): {e}") from e
+
+
+@pytest.mark.gpu
+def test_fuzz_gpu_vs_oracle():
+    """The same randomized differential check against the REAL kernels:
+    12 random cases x 2 specs, GPU scan vs oracle."""
+    from gpu_scan import GpuScan
+
+    rng = random.Random(77)
+    for it in range(12):
+        schema, kv_format, built, rows, dtypes = _random_case(rng)
+        data, offsets, nb, total, ne = built
+        for run in range(2):
+            read, preds, aggs = _random_spec(rng, len(dtypes), dtypes, rows)
+            spec = y.ScanSpec()
+            spec.schema = schema
+            spec.kv_format = kv_format
+            spec.read_time = y.read_time(read)
+            spec.num_preds = len(preds)
+            for i, p in enumerate(preds):
+                spec.preds[i] = p
+            spec.num_aggs = len(aggs)
+            for i, a in enumerate(aggs):
+                spec.aggs[i] = a
+            s = GpuScan(spec)
+            s.feed_blocks_host(data, offsets, nb, total)
+            s.execute()
+            gres = s.aggregates()
+            s.close()
+
+            osc = y.orcl_schema_from(schema)
+            ospec = make_orcl_spec(read, preds, aggs)
+            ores, _ = y.orcl_scan(data, offsets, nb, osc, ospec,
+                                  kv_format=kv_format)
+            try:
+                check_match(gres, ores, aggs)
+            except AssertionError as e:
+                raise AssertionError(
+                    f"gpu fuzz iter {it} run {run} (rows={rows}, "
                     f"dtypes={dtypes}, read={read}): {e}") from e
