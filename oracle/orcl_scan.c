@@ -527,7 +527,8 @@ static int pred_eval(const orcl_pred_t *pr, const orcl_schema_t *sc,
     case ORCL_PRED_LE: return cmp <= 0;
     case ORCL_PRED_EQ: return cmp == 0;
     case ORCL_PRED_NE: return cmp != 0;
-    case ORCL_PRED_IN: return 0; /* handled above */
+    case ORCL_PRED_IN: return 0;       /* handled above */
+    case ORCL_PRED_IN_TUPLE: return 0; /* handled above */
   }
   return 0;
 }
