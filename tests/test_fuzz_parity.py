@@ -192,3 +192,68 @@ def test_fuzz_gpu_vs_oracle():
                 raise AssertionError(
                     f"gpu fuzz iter {it} run {run} (rows={rows}, "
                     f"dtypes={dtypes}, read={read}): {e}") from e
+
+
+def test_fuzz_emit_vs_oracle_rows():
+    """Row materialization (the PgFetchNext surface) on random cases: the
+    simulator's emitted rows vs the oracle's collected rows."""
+    rng = random.Random(5150)
+    for it in range(12):
+        schema, kv_format, built, rows, dtypes = _random_case(rng)
+        data, offsets, nb, total, ne = built
+        read, preds, _ = _random_spec(rng, len(dtypes), dtypes, rows)
+        spec = y.ScanSpec()
+        spec.schema = schema
+        spec.kv_format = kv_format
+        spec.read_time = y.read_time(read)
+        spec.num_preds = len(preds)
+        for i, p in enumerate(preds):
+            spec.preds[i] = p
+        got = y.sim_emit(spec, data, offsets, nb)
+
+        osc = y.orcl_schema_from(schema)
+        ospec = make_orcl_spec(read, preds, ())
+        _, want = y.orcl_scan(data, offsets, nb, osc, ospec,
+                              kv_format=kv_format, collect_rows=True)
+        assert len(got) == len(want), (it, len(got), len(want))
+        for g, w in zip(got, want):
+            assert g == w, (it, g, w)
+
+
+def test_fuzz_group_vs_oracle():
+    """GROUP BY partial aggregates on random cases (int64 group keys,
+    COUNT/SUM/MIN/MAX int64)."""
+    rng = random.Random(8086)
+    for it in range(12):
+        schema, kv_format, built, rows, dtypes = _random_case(rng)
+        int_cols = [i for i, d in enumerate(dtypes)
+                    if d in (y.T_INT64, y.T_INT32)]
+        if not int_cols:
+            continue
+        gcol = rng.choice(int_cols)
+        data, offsets, nb, total, ne = built
+        read, preds, _ = _random_spec(rng, len(dtypes), dtypes, rows)
+        aggs = [y.Agg(y.AGG_COUNT_STAR, 0),
+                y.Agg(rng.choice([y.AGG_SUM_INT64, y.AGG_MIN_INT64,
+                                  y.AGG_MAX_INT64]), gcol)]
+        spec = y.ScanSpec()
+        spec.schema = schema
+        spec.kv_format = kv_format
+        spec.read_time = y.read_time(read)
+        spec.group_col = 1 + gcol
+        spec.num_preds = len(preds)
+        for i, p in enumerate(preds):
+            spec.preds[i] = p
+        spec.num_aggs = len(aggs)
+        for i, a in enumerate(aggs):
+            spec.aggs[i] = a
+        got = y.sim_group(spec, data, offsets, nb)
+
+        osc = y.orcl_schema_from(schema)
+        ospec = make_orcl_spec(read, preds, aggs)
+        want = y.orcl_group(data, offsets, nb, osc, ospec, gcol,
+                            kv_format=kv_format, num_aggs=2,
+                            aggs=ospec.aggs)
+        assert got == want, (it, gcol,
+                             {k: got[k] for k in list(got)[:3]},
+                             {k: want[k] for k in list(want)[:3]})
