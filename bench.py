@@ -19,6 +19,14 @@ on a bounded sample on this host; it is a reported baseline, not the target.
 import argparse
 import ctypes
 import json
+
+
+def _wrap_i64(v):
+    """Two's-complement wrap to int64 — the per-tablet partial SUMs are
+    already wrapped C int64s; their PYTHON sum can exceed the int64 range
+    the all-reduce tensor holds. Wrapping keeps rank-side math identical
+    to the device/C semantics."""
+    return ((v + (1 << 63)) & ((1 << 64) - 1)) - (1 << 63)
 import os
 import sys
 import time
@@ -175,24 +183,24 @@ def main():
                     total_cnt += int(va[:, 0].sum())
                     total_sum += int(va[:, 1].sum())
             if dist is not None:
-                agg_buf[0] = total_sum
+                agg_buf[0] = _wrap_i64(total_sum)
                 agg_buf[1] = total_cnt
                 dist.all_reduce(agg_buf)
                 total_sum = int(agg_buf[0].item())
                 total_cnt = int(agg_buf[1].item())
-            return total_sum, total_cnt
+            return _wrap_i64(total_sum), total_cnt
         for s in scans:
             s.execute()
         results = [s.aggregates() for s in scans]
         total_sum = sum(r.aggs[0].value_i64 for r in results)
         total_cnt = sum(r.aggs[1].value_i64 for r in results)
         if dist is not None:
-            agg_buf[0] = total_sum
+            agg_buf[0] = _wrap_i64(total_sum)
             agg_buf[1] = total_cnt
             dist.all_reduce(agg_buf)  # RCCL over xGMI: the one collective
             total_sum = int(agg_buf[0].item())
             total_cnt = int(agg_buf[1].item())
-        return total_sum, total_cnt
+        return _wrap_i64(total_sum), total_cnt
 
     def barrier_sync():
         if dist is not None:
