@@ -522,7 +522,8 @@ __global__ __launch_bounds__(256) void k_snappy(
     const uint8_t* src = src_blob + src_off[i];
     uint64_t n = src_len[i];
     uint8_t* dst = out + dst_off[i];
-    uint64_t cap = out_cap - dst_off[i];
+    uint64_t cap = dst_off[i + 1] - dst_off[i];  // exact slot, never the
+                                                 // neighbour's bytes
     if (types[i] == 1) {
       if (ybsnappy::snappy_uncompress(src, n, dst, cap) < 0)
         atomicAdd(err, 1ull);
@@ -683,7 +684,9 @@ int yb_gpu_scan_open(const ybg_scan_spec_t* spec, ybg_scan_t** out) {
   uint32_t aux_len = 0;
   build_dev_spec(spec, &s->dspec, s->aux_host.data(), &aux_len,
                  (uint32_t)s->aux_host.size());
-  s->aux_host.resize(aux_len);
+  // 16 bytes of tail slack: the device predicate paths read the aux
+  // buffer through load_u64_una (up to 11 bytes past a position)
+  s->aux_host.resize(aux_len + 16, 0);
 
   hipError_t e;
   if ((e = hipStreamCreate(&s->stream)) != hipSuccess ||
@@ -858,7 +861,12 @@ int yb_gpu_scan_feed_sst(ybg_scan_t* s, const uint8_t* file, uint64_t size,
     boff[i + 1] = boff[i] + un_len[i];
   }
   uint64_t total_un = boff[n];
-  uint64_t blob_len = offs[n - 1] + szs[n - 1];  // blocks start at offset 0
+  uint64_t blob_len = 0;  // cover every listed block + its trailer
+  for (uint64_t i = 0; i < n; ++i) {
+    uint64_t e2 = offs[i] + szs[i] + ybsst::kBlockTrailerSize;
+    if (e2 > blob_len) blob_len = e2;
+  }
+  if (blob_len > size) blob_len = size;
   uint8_t* d_blob = nullptr;
   uint64_t* d_meta = nullptr;  // src_off | src_len | dst_off
   uint8_t* d_types = nullptr;
@@ -866,10 +874,10 @@ int yb_gpu_scan_feed_sst(ybg_scan_t* s, const uint8_t* file, uint64_t size,
   unsigned long long* d_err = nullptr;
   HIP_TRY(hipMalloc(&d_blob, blob_len));
   HIP_TRY(hipMemcpy(d_blob, file, blob_len, hipMemcpyHostToDevice));
-  HIP_TRY(hipMalloc(&d_meta, 3 * n * sizeof(uint64_t)));
+  HIP_TRY(hipMalloc(&d_meta, (3 * n + 1) * sizeof(uint64_t)));
   HIP_TRY(hipMemcpy(d_meta, offs.data(), n * 8, hipMemcpyHostToDevice));
   HIP_TRY(hipMemcpy(d_meta + n, szs.data(), n * 8, hipMemcpyHostToDevice));
-  HIP_TRY(hipMemcpy(d_meta + 2 * n, boff.data(), n * 8,
+  HIP_TRY(hipMemcpy(d_meta + 2 * n, boff.data(), (n + 1) * 8,
                     hipMemcpyHostToDevice));
   HIP_TRY(hipMalloc(&d_types, n));
   HIP_TRY(hipMemcpy(d_types, typs.data(), n, hipMemcpyHostToDevice));

@@ -23,7 +23,7 @@ int ybg_sim_scan(const ybg_scan_spec_t* spec, const uint8_t* data,
   std::vector<unsigned char> aux(1 << 20);
   uint32_t aux_len = 0;
   build_dev_spec(spec, &d, aux.data(), &aux_len, (uint32_t)aux.size());
-  aux.resize(aux_len);
+  aux.resize(aux_len + 16, 0);  // windowed-load tail slack (see ABI open)
 
   // ---- interval table (k_count_restarts + k_emit_intervals, serial) ----
   std::vector<Interval> ivs;
