@@ -235,3 +235,63 @@ def test_gpu_paging_resume(cases):
     _, want = y.orcl_scan(case["data"], case["offsets"], case["n_blocks"],
                           osc, ospec, collect_rows=True)
     assert page1 + page2 == want
+
+
+@pytest.mark.gpu
+def test_gpu_seek_tuple(cases):
+    """GetTupleId / SeekTuple (the ybctid surface,
+    ql_rowwise_iterator_interface.h:62-71): fetch forward, capture a row's
+    tuple id, seek back to it and re-fetch the identical row; seeking a
+    nonexistent ybctid reports not-found."""
+    import ctypes as C
+    from parity_cases import make_spec
+    gpu_scan = _gpu()
+    lib = gpu_scan._lib()
+    lib.yb_host_iter_tuple_id.restype = C.c_int
+    lib.yb_host_iter_tuple_id.argtypes = [
+        C.c_void_p, C.POINTER(C.c_uint8), C.c_size_t, C.POINTER(C.c_size_t)]
+    lib.yb_host_iter_seek_tuple.restype = C.c_int
+    lib.yb_host_iter_seek_tuple.argtypes = [
+        C.c_void_p, C.POINTER(C.c_uint8), C.c_size_t]
+    case = [c for c in cases if c["name"] == "nulls_packed"][0]
+    read_micros = case["runs"][0][0]
+    sc = case["schema"]
+    nk = sc.num_hash_cols + sc.num_range_cols
+
+    spec = make_spec(case, read_micros, (), ())
+    h = lib.yb_host_iter_open(C.byref(spec), case["data"], case["offsets"],
+                              case["n_blocks"])
+    assert h
+    kd = (C.c_uint64 * y.MAX_KEYCOLS)()
+    vd = (C.c_uint64 * y.MAX_COLS)()
+    nm = C.c_uint32()
+    vl = C.POINTER(C.c_uint8)()
+
+    def fetch():
+        rc = lib.yb_host_iter_next(h, kd, vd, C.byref(nm), C.byref(vl))
+        assert rc == 1
+        return (tuple(kd[i] for i in range(nk)),
+                tuple(None if (nm.value >> c) & 1 else vd[c]
+                      for c in range(sc.num_value_cols)))
+
+    seen = [fetch() for _ in range(500)]
+    tk = (C.c_uint8 * 64)()
+    tl = C.c_size_t()
+    assert lib.yb_host_iter_tuple_id(h, tk, 64, C.byref(tl)) == 0
+    assert tl.value > 0
+    tid500 = bytes(tk[:tl.value])
+
+    for _ in range(200):
+        fetch()
+
+    # seek back to row 500's ybctid; the next fetch must reproduce it
+    buf = (C.c_uint8 * len(tid500)).from_buffer_copy(tid500)
+    assert lib.yb_host_iter_seek_tuple(h, buf, len(tid500)) == 0
+    assert fetch() == seen[499]
+
+    # nonexistent ybctid -> not found (flip a datum byte deep in the key)
+    bad = bytearray(tid500)
+    bad[-3] ^= 0x7F
+    bbuf = (C.c_uint8 * len(bad)).from_buffer_copy(bytes(bad))
+    assert lib.yb_host_iter_seek_tuple(h, bbuf, len(bad)) == 1
+    lib.yb_host_iter_close(h)

@@ -70,17 +70,14 @@ int GpuDocRowwiseIterator::Aggregate(ybg_scan_result_t* out) {
 // Resumable position (pgsql_operation.cc:2796-2806, 2908-2922): the encoded
 // DocKey of the first undelivered row; a follow-up scan resumes with it as
 // the inclusive lower bound. len = 0 when the scan is complete.
-int GpuDocRowwiseIterator::PagingState(uint8_t* key_out, size_t cap,
-                                       size_t* len) {
-  *len = 0;
-  if (!batch_ready_ || !limit_ || limit_ >= batch_.n_rows) return 0;
-  uint64_t next = order_[limit_];
+size_t GpuDocRowwiseIterator::EncodeRowKey(uint64_t row, uint8_t* out,
+                                           size_t cap) {
   ybg_key_t k = {};
   const ybg_schema_t& sc = spec_.schema;
   int nk = sc.num_hash_cols + sc.num_range_cols;
   std::vector<std::vector<uint8_t>> strs((size_t)nk);
   for (int c = 0; c < nk; ++c) {
-    uint64_t d = batch_.key_datums[next * (uint64_t)nk + c];
+    uint64_t d = batch_.key_datums[row * (uint64_t)nk + c];
     if (sc.key_types[c] == YBG_KT_STRING) {
       uint64_t off = d & ((1ull << 40) - 1);
       uint32_t ln = (uint32_t)(d >> 40);
@@ -91,9 +88,53 @@ int GpuDocRowwiseIterator::PagingState(uint8_t* key_out, size_t cap,
       k.datums[c] = d;
     }
   }
-  if (sc.has_hash) k.hash = batch_.hashes[next];
-  size_t n = ybg_encode_dockey(&sc, &k, key_out, cap);
-  *len = n;
+  if (sc.has_hash) k.hash = batch_.hashes[row];
+  return ybg_encode_dockey(&sc, &k, out, cap);
+}
+
+int GpuDocRowwiseIterator::PagingState(uint8_t* key_out, size_t cap,
+                                       size_t* len) {
+  *len = 0;
+  if (!batch_ready_ || !limit_ || limit_ >= batch_.n_rows) return 0;
+  *len = EncodeRowKey(order_[limit_], key_out, cap);
+  return 0;
+}
+
+// ~ YQLRowwiseIteratorIf::GetTupleId (ql_rowwise_iterator_interface.h:
+// 62-66): the ybctid of the row the last PgFetchNext returned — its
+// encoded DocKey.
+int GpuDocRowwiseIterator::GetTupleId(uint8_t* key_out, size_t cap,
+                                      size_t* len) {
+  *len = 0;
+  if (!batch_ready_ || pos_ == 0 || pos_ > batch_.n_rows) return -1;
+  *len = EncodeRowKey(order_[pos_ - 1], key_out, cap);
+  return *len ? 0 : -1;
+}
+
+// ~ YQLRowwiseIteratorIf::SeekTuple (ql_rowwise_iterator_interface.h:
+// 68-71): position so the next PgFetchNext returns the row with this
+// exact encoded DocKey (ybctid). Returns 0 positioned, 1 not found.
+int GpuDocRowwiseIterator::SeekTuple(const uint8_t* dockey, size_t len) {
+  if (open_rc_) return -open_rc_;
+  if (!batch_ready_) {
+    int rc = MaterializeBatch();
+    if (rc) return -rc;
+  }
+  // rows are sorted by tablet key order: binary search on encoded keys
+  uint8_t buf[256];
+  uint64_t lo = 0, hi = batch_.n_rows;
+  while (lo < hi) {
+    uint64_t mid = (lo + hi) / 2;
+    size_t n = EncodeRowKey(order_[mid], buf, sizeof(buf));
+    int cmp = memcmp(buf, dockey, n < len ? n : len);
+    if (cmp == 0) cmp = (n < len) ? -1 : (n > len ? 1 : 0);
+    if (cmp < 0) lo = mid + 1;
+    else hi = mid;
+  }
+  if (lo >= batch_.n_rows) return 1;
+  size_t n = EncodeRowKey(order_[lo], buf, sizeof(buf));
+  if (n != len || memcmp(buf, dockey, len) != 0) return 1;
+  pos_ = lo;
   return 0;
 }
 
@@ -136,6 +177,16 @@ int yb_host_iter_paging_state(void* h, uint8_t* key_out, size_t cap,
                               size_t* len) {
   return static_cast<ybg::GpuDocRowwiseIterator*>(h)->PagingState(key_out,
                                                                   cap, len);
+}
+
+int yb_host_iter_tuple_id(void* h, uint8_t* key_out, size_t cap,
+                          size_t* len) {
+  return static_cast<ybg::GpuDocRowwiseIterator*>(h)->GetTupleId(key_out,
+                                                                 cap, len);
+}
+
+int yb_host_iter_seek_tuple(void* h, const uint8_t* dockey, size_t len) {
+  return static_cast<ybg::GpuDocRowwiseIterator*>(h)->SeekTuple(dockey, len);
 }
 
 void yb_host_iter_close(void* h) {

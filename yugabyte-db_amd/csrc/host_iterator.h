@@ -44,10 +44,16 @@ class GpuDocRowwiseIterator {
   // ~ GetSubDocKey paging position (pgsql_operation.cc:2908-2922)
   int PagingState(uint8_t* key_out, size_t cap, size_t* len);
 
+  // ~ GetTupleId / SeekTuple — the ybctid surface
+  // (ql_rowwise_iterator_interface.h:62-71)
+  int GetTupleId(uint8_t* key_out, size_t cap, size_t* len);
+  int SeekTuple(const uint8_t* dockey, size_t len);
+
   const char* LastError() const;
 
  private:
   int MaterializeBatch();
+  size_t EncodeRowKey(uint64_t row, uint8_t* out, size_t cap);
 
   ybg_scan_t* handle_ = nullptr;
   ybg_scan_spec_t spec_;
