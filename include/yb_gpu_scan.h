@@ -219,6 +219,14 @@ typedef struct {
   uint64_t rows_matched;  /* rows passing predicates */
   uint64_t entries_seen;  /* KV entries decoded */
   ybg_agg_result_t aggs[YBG_MAX_AGGS];
+  /* Read-restart data (GetReadRestartData analog,
+   * intent_aware_iterator.cc:1400-1410): when local_limit > read and a
+   * visible record committed in (read, local_limit], the ENCODED
+   * DocHybridTime of the newest such record (the max seen commit time,
+   * smallest encoded bytes). restart_ht_len == 0 means no restart. */
+  uint8_t restart_ht[YBG_MAX_HT];
+  uint32_t restart_ht_len;
+  uint32_t pad2_;
 } ybg_scan_result_t;
 
 /* Fetch aggregate results (implies wait). */

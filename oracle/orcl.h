@@ -285,6 +285,12 @@ typedef struct {
   uint64_t rows_matched;    /* rows passing predicates */
   uint64_t entries_seen;    /* KV entries decoded */
   orcl_agg_result_t aggs[ORCL_MAX_AGGS];
+  /* read-restart data (intent_aware_iterator.cc:815-827, 1400-1410):
+   * encoded DocHybridTime of the newest visible record with commit time
+   * in (read, local_limit]; len 0 = no restart needed */
+  uint8_t restart_ht[ORCL_MAX_HT_SIZE];
+  uint32_t restart_ht_len;
+  uint32_t pad2_;
 } orcl_scan_result_t;
 
 /* Row callback for non-aggregate scans (parity tests). Return 0 to continue. */

@@ -790,6 +790,21 @@ int orcl_scan(const uint8_t *const *blocks, const size_t *sizes, size_t nblocks,
       }
       if (!visible) continue;
 
+      /* --- read-restart tracking (UpdateMaxSeenHt,
+       * intent_aware_iterator.cc:815-827; GetReadRestartData :1400-1410):
+       * a visible record with commit > read (encoded bytes BELOW
+       * encoded(read)) is a restart candidate; keep the MIN encoded form
+       * (= max commit time). */
+      if (slice_cmp(ht_enc, ht_size, rt->read, rt->read_len) < 0) {
+        if (result->restart_ht_len == 0 ||
+            slice_cmp(ht_enc, ht_size, result->restart_ht,
+                      result->restart_ht_len) < 0) {
+          size_t n = ht_size < ORCL_MAX_HT_SIZE ? ht_size : ORCL_MAX_HT_SIZE;
+          memcpy(result->restart_ht, ht_enc, n);
+          result->restart_ht_len = (uint32_t)n;
+        }
+      }
+
       /* --- row grouping: split DocKey from subkeys (InitIterKey,
        * doc_rowwise_iterator.cc:363-404) */
       size_t dockey_len = parse_dockey(schema, ukey, prefix_len, NULL);

@@ -90,7 +90,9 @@ class AggResult(C.Structure):
 
 class ScanResult(C.Structure):
     _fields_ = [("rows_scanned", C.c_uint64), ("rows_matched", C.c_uint64),
-                ("entries_seen", C.c_uint64), ("aggs", AggResult * MAX_AGGS)]
+                ("entries_seen", C.c_uint64), ("aggs", AggResult * MAX_AGGS),
+                ("restart_ht", C.c_uint8 * MAX_HT),
+                ("restart_ht_len", C.c_uint32), ("pad2_", C.c_uint32)]
 
 
 class RowBatch(C.Structure):
@@ -449,7 +451,9 @@ class OrclAggResult(C.Structure):
 class OrclScanResult(C.Structure):
     _fields_ = [("rows_scanned", C.c_uint64), ("rows_matched", C.c_uint64),
                 ("entries_seen", C.c_uint64),
-                ("aggs", OrclAggResult * MAX_AGGS)]
+                ("aggs", OrclAggResult * MAX_AGGS),
+                ("restart_ht", C.c_uint8 * MAX_HT),
+                ("restart_ht_len", C.c_uint32), ("pad2_", C.c_uint32)]
 
 
 def sim_scan(spec, data, offsets, n_blocks):

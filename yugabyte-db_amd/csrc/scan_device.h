@@ -185,6 +185,11 @@ struct DevSpec {
   // per-column body offsets from the value start. 0 = fast path off.
   uint32_t v2_fixed_len;
   uint8_t v2_off[YBG_MAX_COLS];
+  // Read-restart tracking (intent_aware_iterator.cc:815-827 UpdateMaxSeenHt
+  // + GetReadRestartData :1400-1410): only possible when local_limit >
+  // read — a visible record committed in (read, local_limit] forces a
+  // restart. 0 disables all per-entry work.
+  int32_t track_restart;
 };
 
 constexpr uint32_t kActPredM = 0xffu;
@@ -204,8 +209,10 @@ struct Interval {
 
 constexpr int kThreads = 256;
 constexpr int kKeyCap = 128;
-// wave partial: entries, scanned, matched, err, {val,cnt} x MAX_AGGS
-constexpr int kPartialStride = 4 + 2 * YBG_MAX_AGGS;
+// wave partial: entries, scanned, matched, err, {val,cnt} x MAX_AGGS,
+// restart-min {hi, lo, len} (encoded-HT MIN among visible restart
+// candidates; len 0 = none)
+constexpr int kPartialStride = 4 + 2 * YBG_MAX_AGGS + 3;
 // head record: {val,cnt} x MAX_AGGS, scanned, matched
 constexpr int kHeadStride = 2 * YBG_MAX_AGGS + 2;
 
@@ -1725,6 +1732,21 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
                              sp.reg_lim.lo, sp.reg_lim.len) >= 0;
   }
   if (!visible) return true;
+  if (YBG_UNLIKELY(sp.track_restart)) {
+    // visible record with commit > read (encoded bytes BELOW enc(read)):
+    // a restart candidate — keep the MIN encoded (max commit time). The
+    // slot lives beside bht in LDS (rc->bht[3..5]).
+    if (u128_slice_cmp(ht_hi, ht_lo, ht_size, sp.read.hi, sp.read.lo,
+                       sp.read.len) < 0) {
+      uint64_t* rr = rc->bht + 3;
+      if (rr[2] == 0 || u128_slice_cmp(ht_hi, ht_lo, ht_size, rr[0], rr[1],
+                                       (uint32_t)rr[2]) < 0) {
+        rr[0] = ht_hi;
+        rr[1] = ht_lo;
+        rr[2] = ht_size;
+      }
+    }
+  }
 
   if (rowkey_len == prefix_len) {
     if (!rc->base_seen) {
@@ -2415,6 +2437,9 @@ inline void build_dev_spec(const ybg_scan_spec_t* spec, DevSpec* dp,
     int cmp = memcmp(spec->read_time.local_limit, spec->read_time.read, n);
     bool local_smaller = cmp < 0 || (cmp == 0 && ll < rl);
     d.reg_lim = local_smaller ? d.local_lim : d.read;
+    // restart possible iff regular_limit != read (local_limit > read):
+    // encoded order reversed, so local_smaller means local > read
+    d.track_restart = local_smaller ? 1 : 0;
   }
   uint32_t pos = 0;
   auto put = [&](const uint8_t* b, uint64_t n) {

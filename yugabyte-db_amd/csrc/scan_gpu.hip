@@ -70,9 +70,10 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
   const DevSpec& sp = c_spec;
   __shared__ uint8_t key_scratch[kThreads * kKeyCap];
   __shared__ uint8_t wave_relay[kThreads / 64 + 1];
-  __shared__ uint64_t bht_scratch[kThreads * 3];
+  __shared__ uint64_t bht_scratch[kThreads * 6];
   uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
-  uint64_t* bht = bht_scratch + (size_t)threadIdx.x * 3;
+  uint64_t* bht = bht_scratch + (size_t)threadIdx.x * 6;
+  bht[5] = 0;  // restart-min slot empty (bht[3..5] = {hi, lo, len})
   const uint32_t gtid = blockIdx.x * kThreads + threadIdx.x;
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
   const uint64_t span = (uint64_t)gridDim.x * kThreads;
@@ -184,6 +185,22 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
       partials[wave_id * kPartialStride + 1] = scanned;
       partials[wave_id * kPartialStride + 2] = matched;
       partials[wave_id * kPartialStride + 3] = errs;
+      // restart-min across the wave (cold: len==0 for every lane unless
+      // track_restart saw candidates)
+      uint64_t mh = 0, ml = 0, mn = 0;
+      for (int l = 0; l < 64; ++l) {
+        const uint64_t* rr = bht_scratch + (size_t)(threadIdx.x + l) * 6 + 3;
+        if (rr[2] == 0) continue;
+        if (mn == 0 || u128_slice_cmp(rr[0], rr[1], (uint32_t)rr[2], mh, ml,
+                                      (uint32_t)mn) < 0) {
+          mh = rr[0];
+          ml = rr[1];
+          mn = rr[2];
+        }
+      }
+      partials[wave_id * kPartialStride + 4 + 2 * YBG_MAX_AGGS] = mh;
+      partials[wave_id * kPartialStride + 4 + 2 * YBG_MAX_AGGS + 1] = ml;
+      partials[wave_id * kPartialStride + 4 + 2 * YBG_MAX_AGGS + 2] = mn;
     }
   }
 }
@@ -207,8 +224,9 @@ __global__ __launch_bounds__(kEmitThreads) void k_emit(
   __shared__ uint8_t key_scratch[kEmitThreads * kKeyCap];
   __shared__ uint64_t rowbuf[kEmitThreads * YBG_MAX_COLS];
   __shared__ uint32_t lenbuf[kEmitThreads * YBG_MAX_COLS];
-  __shared__ uint64_t bht_scratch[kEmitThreads * 3];
-  uint64_t* bht = bht_scratch + (size_t)threadIdx.x * 3;
+  __shared__ uint64_t bht_scratch[kEmitThreads * 6];
+  uint64_t* bht = bht_scratch + (size_t)threadIdx.x * 6;
+  bht[5] = 0;
   uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
   uint64_t* rb = rowbuf + (size_t)threadIdx.x * YBG_MAX_COLS;
   uint32_t* lb = lenbuf + (size_t)threadIdx.x * YBG_MAX_COLS;
@@ -393,6 +411,7 @@ struct DevResult {
   uint64_t entries, scanned, matched, errs;
   uint64_t agg_val[YBG_MAX_AGGS];
   uint64_t agg_cnt[YBG_MAX_AGGS];
+  uint64_t restart_hi, restart_lo, restart_len;
 };
 
 // Pre-reduction: workgroup c folds a CONTIGUOUS chunk of wave partials and
@@ -454,6 +473,42 @@ __global__ __launch_bounds__(256) void k_reduce_pre(
     }
     __syncthreads();
   }
+  {
+    // restart-min fold (encoded-HT MIN; len 0 = none)
+    const int rb = 4 + 2 * YBG_MAX_AGGS;
+    uint64_t mh = 0, ml = 0, mn = 0;
+    for (uint64_t i = plo + t; i < phi; i += 256) {
+      uint64_t n = partials[i * kPartialStride + rb + 2];
+      if (n == 0) continue;
+      uint64_t h = partials[i * kPartialStride + rb];
+      uint64_t l = partials[i * kPartialStride + rb + 1];
+      if (mn == 0 ||
+          u128_slice_cmp(h, l, (uint32_t)n, mh, ml, (uint32_t)mn) < 0) {
+        mh = h;
+        ml = l;
+        mn = n;
+      }
+    }
+    sval[t] = mh;
+    scnt[t] = ml;
+    scal[t] = mn;
+    __syncthreads();
+    if (t == 0) {
+      uint64_t fh = 0, fl = 0, fn = 0;
+      for (int i = 0; i < 256; ++i) {
+        if (scal[i] == 0) continue;
+        if (fn == 0 || u128_slice_cmp(sval[i], scnt[i], (uint32_t)scal[i],
+                                      fh, fl, (uint32_t)fn) < 0) {
+          fh = sval[i];
+          fl = scnt[i];
+          fn = scal[i];
+        }
+      }
+      chunk_out[blockIdx.x * kPartialStride + rb] = fh;
+      chunk_out[blockIdx.x * kPartialStride + rb + 1] = fl;
+      chunk_out[blockIdx.x * kPartialStride + rb + 2] = fn;
+    }
+  }
 }
 
 __global__ __launch_bounds__(256) void k_reduce(
@@ -503,6 +558,42 @@ __global__ __launch_bounds__(256) void k_reduce(
       out->agg_cnt[g] = fc;
     }
     __syncthreads();
+  }
+  {
+    // restart-min over the chunk records (len 0 = none)
+    const int rb = 4 + 2 * YBG_MAX_AGGS;
+    uint64_t mh = 0, ml = 0, mn = 0;
+    for (uint64_t i = t; i < n_partials; i += 256) {
+      uint64_t n = partials[i * kPartialStride + rb + 2];
+      if (n == 0) continue;
+      uint64_t h = partials[i * kPartialStride + rb];
+      uint64_t l = partials[i * kPartialStride + rb + 1];
+      if (mn == 0 ||
+          u128_slice_cmp(h, l, (uint32_t)n, mh, ml, (uint32_t)mn) < 0) {
+        mh = h;
+        ml = l;
+        mn = n;
+      }
+    }
+    sval[t] = mh;
+    scnt[t] = ml;
+    scal[t] = mn;
+    __syncthreads();
+    if (t == 0) {
+      uint64_t fh = 0, fl = 0, fn = 0;
+      for (int i = 0; i < 256; ++i) {
+        if (scal[i] == 0) continue;
+        if (fn == 0 || u128_slice_cmp(sval[i], scnt[i], (uint32_t)scal[i],
+                                      fh, fl, (uint32_t)fn) < 0) {
+          fh = sval[i];
+          fl = scnt[i];
+          fn = scal[i];
+        }
+      }
+      out->restart_hi = fh;
+      out->restart_lo = fl;
+      out->restart_len = fn;
+    }
   }
 }
 
@@ -987,6 +1078,15 @@ int yb_gpu_scan_aggregate(ybg_scan_t* s, ybg_scan_result_t* out) {
   out->entries_seen = r.entries;
   out->rows_scanned = r.scanned;
   out->rows_matched = r.matched;
+  if (r.restart_len) {
+    uint32_t n = (uint32_t)r.restart_len;
+    if (n > YBG_MAX_HT) n = YBG_MAX_HT;
+    for (uint32_t i = 0; i < n; ++i)
+      out->restart_ht[i] = (uint8_t)(
+          (i < 8 ? r.restart_hi >> (56 - 8 * i)
+                 : r.restart_lo >> (56 - 8 * (i - 8))) & 0xff);
+    out->restart_ht_len = n;
+  }
   for (int g = 0; g < s->spec.num_aggs; ++g) {
     ybg_agg_result_t& a = out->aggs[g];
     a.is_null = (r.agg_cnt[g] == 0);

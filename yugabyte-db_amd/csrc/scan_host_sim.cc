@@ -51,7 +51,7 @@ int ybg_sim_scan(const ybg_scan_spec_t* spec, const uint8_t* data,
   std::vector<uint8_t> walked(n_ivs, 0);
   alignas(8) uint8_t key[kKeyCap];
   alignas(8) uint8_t rk_save[kKeyCap];
-  uint64_t bht[3] = {0, 0, 0};
+  uint64_t bht[6] = {0, 0, 0, 0, 0, 0};
   for (uint64_t j = 0; j < n_ivs; ++j) {
     bool wn = false;
     uint32_t e32 = 0, s32 = 0, m32 = 0;
@@ -78,6 +78,15 @@ int ybg_sim_scan(const ybg_scan_spec_t* spec, const uint8_t* data,
   out->entries_seen = entries;
   out->rows_scanned = scanned;
   out->rows_matched = matched;
+  if (bht[5]) {  // restart-min slot {hi, lo, len} (process_entry tracking)
+    uint32_t n = (uint32_t)bht[5];
+    if (n > YBG_MAX_HT) n = YBG_MAX_HT;
+    for (uint32_t i = 0; i < n; ++i)
+      out->restart_ht[i] = (uint8_t)(
+          (i < 8 ? bht[3] >> (56 - 8 * i) : bht[4] >> (56 - 8 * (i - 8))) &
+          0xff);
+    out->restart_ht_len = n;
+  }
   for (int g = 0; g < d.num_aggs; ++g) {
     out->aggs[g].is_null = (agg_cnt[g] == 0);
     switch (d.aggs[g].op) {
@@ -136,7 +145,7 @@ int ybg_sim_emit(const ybg_scan_spec_t* spec, const uint8_t* data,
   std::vector<uint32_t> head_consumed(n_ivs ? n_ivs : 1, 0);
   alignas(8) uint8_t key[kKeyCap];
   alignas(8) uint8_t rk_save[kKeyCap];
-  uint64_t bht[3] = {0, 0, 0};
+  uint64_t bht[6] = {0, 0, 0, 0, 0, 0};
   HeadOut<YBG_MAX_AGGS> ho;
   for (uint64_t j = 0; j < n_ivs; ++j) {
     bool wn = false;
@@ -218,7 +227,7 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
   std::vector<GroupHead> gheads(n_ivs ? n_ivs : 1);
   alignas(8) uint8_t key[kKeyCap];
   alignas(8) uint8_t rk_save[kKeyCap];
-  uint64_t bht[3] = {0, 0, 0};
+  uint64_t bht[6] = {0, 0, 0, 0, 0, 0};
 
   uint64_t gcap = 1ull << 18;
   std::vector<unsigned long long> gkey(gcap + 1, 0);
