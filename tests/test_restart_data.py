@@ -100,3 +100,28 @@ def test_restart_window_gpu():
         assert bytes(gres.restart_ht[:gres.restart_ht_len]) == \
             bytes(sres.restart_ht[:sres.restart_ht_len])
         assert bool(gres.restart_ht_len) == expect_restart, (read, local)
+
+
+def test_restart_window_fuzz():
+    """Random read/local/global windows over random multi-version tablets:
+    sim and oracle must agree byte-exactly on the restart data."""
+    import random
+    rng = random.Random(31415)
+    for it in range(15):
+        schema = y.make_schema([y.KT_INT64], [(10, y.T_INT64, 1)])
+        b = y.Builder(schema)
+        seq = 1 << 50
+        rows = rng.randint(40, 400)
+        for r in range(rows):
+            hts = sorted(rng.sample(range(1000, 9000), rng.randint(1, 4)),
+                         reverse=True)
+            for ht in hts:
+                seq += 1
+                b.add_packed_row(ht, [(y.T_INT64, r)], hash_=r // 64,
+                                 key_datums=(r,), seq=seq)
+        built = b.finish()
+        for _ in range(4):
+            read = rng.randint(500, 9500)
+            local = read + rng.choice([0, 0, rng.randint(1, 4000)])
+            glob = local + rng.randint(0, 2000)
+            _run(schema, built, read, local, glob)
