@@ -467,23 +467,29 @@ struct Rdr {
     q3 = q[3];
   }
   DEV const uint8_t* pos() const { return base + k; }
+  // All window selection below is ARITHMETIC on values (mask/ternary on
+  // u64s), never an if/else over which member to read: the optimizer was
+  // turning the branchy form into a dynamically-indexed q[] array in
+  // scratch memory — every peek/refill became an HBM-backed scratch
+  // round-trip (~5 GB/dispatch of WRITE_SIZE on the 100M-row bench).
   // 8 bytes at the current position, little-endian
   DEV uint64_t peek8() const {
     uint32_t kk = k;
-    uint64_t lo = q0, hi = q1;
-    if (kk >= 8) { lo = q1; hi = q2; kk -= 8; }
-    if (kk == 0) return lo;
-    return (lo >> (8 * kk)) | (hi << (64 - 8 * kk));
+    uint64_t m1 = (uint64_t)0 - (uint64_t)(kk >= 8);
+    uint64_t lo = (q0 & ~m1) | (q1 & m1);
+    uint64_t hi = (q1 & ~m1) | (q2 & m1);
+    uint32_t sh = 8 * (kk & 7);
+    return sh ? ((lo >> sh) | (hi << (64 - sh))) : lo;
   }
   // 8 bytes at position + off (off + k must be < 25)
   DEV uint64_t peek8_at(uint32_t off) const {
     uint32_t kk = k + off;
-    uint64_t lo, hi;
-    if (kk >= 16) { lo = q2; hi = q3; kk -= 16; }
-    else if (kk >= 8) { lo = q1; hi = q2; kk -= 8; }
-    else { lo = q0; hi = q1; }
-    if (kk == 0) return lo;
-    return (lo >> (8 * kk)) | (hi << (64 - 8 * kk));
+    uint64_t m1 = (uint64_t)0 - (uint64_t)(kk >= 8);
+    uint64_t m2 = (uint64_t)0 - (uint64_t)(kk >= 16);
+    uint64_t lo = (q0 & ~m1) | (q1 & (m1 & ~m2)) | (q2 & m2);
+    uint64_t hi = (q1 & ~m1) | (q2 & (m1 & ~m2)) | (q3 & m2);
+    uint32_t sh = 8 * (kk & 7);
+    return sh ? ((lo >> sh) | (hi << (64 - sh))) : lo;
   }
   DEV void consume(uint32_t n) {  // n <= 16
     // refill lazily in 16-byte steps: the two adjacent u64 loads hit the
