@@ -46,6 +46,9 @@ def _lib():
         C.c_void_p, C.POINTER(C.c_uint64), C.POINTER(C.c_int64),
         C.POINTER(C.c_uint64), C.POINTER(C.c_uint8), C.c_uint64, C.c_uint64,
         C.POINTER(C.c_uint64)]
+    lib.yb_gpu_scan_restart_data.restype = C.c_int
+    lib.yb_gpu_scan_restart_data.argtypes = [
+        C.c_void_p, C.POINTER(C.c_uint8), C.POINTER(C.c_uint32)]
     lib.yb_gpu_scan_kernel_ms.restype = C.c_int
     lib.yb_gpu_scan_kernel_ms.argtypes = [C.c_void_p, C.POINTER(C.c_double),
                                           C.POINTER(C.c_double)]
@@ -143,6 +146,17 @@ class GpuScan:
         return y._decode_groups(self._spec.schema, self._spec.group_col - 1,
                                 n, keys, vals, cnts, kb,
                                 self._spec.num_aggs, self._spec.aggs)
+
+    def restart_data(self):
+        """Read-restart data of the last execute/group_aggregate
+        (GetReadRestartData analog): encoded DocHybridTime bytes, b'' when
+        no restart is needed."""
+        ht = (C.c_uint8 * y.MAX_HT)()
+        ln = C.c_uint32()
+        self._check(
+            self._lib.yb_gpu_scan_restart_data(self._h, ht, C.byref(ln)),
+            "restart_data")
+        return bytes(ht[:ln.value])
 
     def kernel_ms(self):
         total = C.c_double()

@@ -109,13 +109,16 @@ void ybg_read_time_init(ybg_read_time_t *rt, uint64_t read_ht,
 typedef enum {
   YBG_PRED_GT = 0, YBG_PRED_GE, YBG_PRED_LT, YBG_PRED_LE,
   YBG_PRED_EQ, YBG_PRED_NE,
-  /* IN-list over a NUMERIC column: bytes = n x 8-byte little-endian datum
-   * bit patterns (the column dtype's representation), bytes_len = 8n.
+  /* IN-list membership. NUMERIC column: bytes = n x 8-byte little-endian
+   * datum bit patterns (the column dtype's representation), bytes_len =
+   * 8n. STRING column (value or key): bytes = [u32 LE length][raw bytes]
+   * records back to back, bytes_len covering them exactly (key-column
+   * options are given UNESCAPED; the kernels compare against the
+   * zero-escaped rowkey form on the fly, doc_kv_util.h:101-167).
    * This is the scan-side equivalent of the reference's hybrid-scan
    * option filters (docdb/hybrid_scan_choices.h:43-60, IN extraction
    * qlexpr/ql_scanspec.cc:323-346): on a brute-force bandwidth-bound GPU
-   * scan the discrete options become a filter, not a seek plan. String
-   * columns are rejected at open(). */
+   * scan the discrete options become a filter, not a seek plan. */
   YBG_PRED_IN,
   /* Tuple membership over MULTIPLE numeric KEY columns — the reference's
    * multi-column option groups ((r1,r3) IN ((1,3),(5,6)) ...),
@@ -265,6 +268,17 @@ typedef struct {
 /* Fetch the materialized matching rows (host-visible; implies wait).
  * Valid until the next execute/close on this handle. */
 int yb_gpu_scan_next_batch(ybg_scan_t *s, ybg_row_batch_t *out);
+
+/* Read-restart data of the last execute/group_aggregate
+ * (GetReadRestartData analog — intent_aware_iterator.cc:1400-1410): the
+ * encoded DocHybridTime (smallest encoded = max commit time) of a visible
+ * record committed in (read, local_limit], i.e. the time the statement
+ * must restart at. *len_out == 0 means no restart. The plain aggregate
+ * path also reports this in ybg_scan_result_t.restart_ht; this entry
+ * point exists for the GROUP BY path, whose result shape has no restart
+ * field. ht_out must hold YBG_MAX_HT bytes. */
+int yb_gpu_scan_restart_data(ybg_scan_t *s, uint8_t *ht_out,
+                             uint32_t *len_out);
 
 /* Resumable position (pgsql_operation.cc:2796-2806, 2908-2922): for an
  * unlimited scan reports length 0 (complete). row_limit paging requires

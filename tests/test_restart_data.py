@@ -125,3 +125,79 @@ def test_restart_window_fuzz():
             local = read + rng.choice([0, 0, rng.randint(1, 4000)])
             glob = local + rng.randint(0, 2000)
             _run(schema, built, read, local, glob)
+
+
+def _grouped_tablet():
+    # same 3-version rows, but with a second (group) column
+    schema = y.make_schema([y.KT_INT64],
+                          [(10, y.T_INT64, 1), (11, y.T_INT64, 1)])
+    b = y.Builder(schema)
+    seq = 1 << 50
+    for r in range(200):
+        for ht in (3000, 2000, 1000):
+            seq += 1
+            b.add_packed_row(ht, [(y.T_INT64, r % 7), (y.T_INT64, r)],
+                             hash_=r // 64, key_datums=(r,), seq=seq)
+    return schema, b.finish()
+
+
+def test_group_scan_reports_restart():
+    """GROUP BY scans must report read-restart data like plain scans
+    (restart tracking is query-independent: it depends only on record
+    visibility vs the read window). The sim group path shares the device
+    code; its restart bytes must equal the plain scan's."""
+    schema, built = _grouped_tablet()
+    for (read, local, glob), expect in (
+            ((1500, 2500, 3500), True),
+            ((1500, 1500, 3500), False),
+            ((5000, 6000, 7000), False)):
+        spec = y.ScanSpec()
+        spec.schema = schema
+        spec.kv_format = y.ENC_THREE_SHARED_PARTS
+        spec.read_time = y.read_time(read, local, glob)
+        spec.group_col = 1  # group by column id 10 (1-based slot)
+        spec.num_aggs = 1
+        spec.aggs[0] = y.Agg(y.AGG_COUNT_STAR, 0)
+        groups, restart = y.sim_group(spec, built[0], built[1], built[2],
+                                      return_restart=True)
+        assert sum(v[0] for v in groups.values()) == 200
+        spec2 = y.ScanSpec()
+        spec2.schema = schema
+        spec2.kv_format = y.ENC_THREE_SHARED_PARTS
+        spec2.read_time = y.read_time(read, local, glob)
+        spec2.num_aggs = 1
+        spec2.aggs[0] = y.Agg(y.AGG_COUNT_STAR, 0)
+        sres = y.sim_scan(spec2, built[0], built[1], built[2])
+        want = bytes(sres.restart_ht[:sres.restart_ht_len])
+        assert restart == want, (restart.hex(), want.hex())
+        assert bool(restart) == expect
+
+
+@pytest.mark.gpu
+def test_group_scan_restart_gpu():
+    """k_group's restart-min wave fold + the k_reduce fold into DevResult,
+    surfaced through yb_gpu_scan_restart_data (the bht sizing fix: group
+    kernels share the 6-slot LDS layout with k_scan)."""
+    from gpu_scan import GpuScan
+    schema, built = _grouped_tablet()
+    for (read, local, glob), expect in (
+            ((1500, 2500, 3500), True),
+            ((1500, 1500, 3500), False),
+            ((5000, 6000, 7000), False)):
+        spec = y.ScanSpec()
+        spec.schema = schema
+        spec.kv_format = y.ENC_THREE_SHARED_PARTS
+        spec.read_time = y.read_time(read, local, glob)
+        spec.group_col = 1
+        spec.num_aggs = 1
+        spec.aggs[0] = y.Agg(y.AGG_COUNT_STAR, 0)
+        s = GpuScan(spec)
+        s.feed_blocks_host(built[0], built[1], built[2], built[3])
+        groups = s.group_aggregate()
+        restart = s.restart_data()
+        s.close()
+        sgroups, want = y.sim_group(spec, built[0], built[1], built[2],
+                                    return_restart=True)
+        assert groups == sgroups
+        assert restart == want
+        assert bool(restart) == expect

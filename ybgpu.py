@@ -565,25 +565,31 @@ def _decode_groups(schema, group_col, n, keys, vals, cnts, key_bytes,
 
 
 def sim_group(spec, data, offsets, n_blocks, cap=1 << 20,
-              key_bytes_cap=1 << 24):
+              key_bytes_cap=1 << 24, return_restart=False):
     """Host-simulator GROUP BY — TEST INFRASTRUCTURE."""
     lib = product()
     f = _sig(lib, "ybg_sim_group", C.c_int,
              [C.POINTER(ScanSpec), C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
               C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(C.c_int64),
               C.POINTER(C.c_uint64), C.POINTER(C.c_uint8), C.c_uint64,
-              C.c_uint64, C.POINTER(C.c_uint64)])
+              C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(C.c_uint8),
+              C.POINTER(C.c_uint32)])
     keys = (C.c_uint64 * cap)()
     vals = (C.c_int64 * (cap * MAX_AGGS))()
     cnts = (C.c_uint64 * (cap * MAX_AGGS))()
     kb = (C.c_uint8 * key_bytes_cap)()
     n = C.c_uint64()
+    rht = (C.c_uint8 * MAX_HT)()
+    rlen = C.c_uint32()
     rc = f(C.byref(spec), data, offsets, n_blocks, keys, vals, cnts, kb,
-           key_bytes_cap, cap, C.byref(n))
+           key_bytes_cap, cap, C.byref(n), rht, C.byref(rlen))
     if rc != 0:
         raise RuntimeError(f"ybg_sim_group rc={rc}")
-    return _decode_groups(spec.schema, spec.group_col - 1, n.value, keys,
-                          vals, cnts, kb, spec.num_aggs, spec.aggs)
+    groups = _decode_groups(spec.schema, spec.group_col - 1, n.value, keys,
+                            vals, cnts, kb, spec.num_aggs, spec.aggs)
+    if return_restart:
+        return groups, bytes(rht[:rlen.value])
+    return groups
 
 
 def orcl_group(data, offsets, n_blocks, schema, spec, group_col,

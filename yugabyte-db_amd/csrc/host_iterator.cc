@@ -92,11 +92,25 @@ size_t GpuDocRowwiseIterator::EncodeRowKey(uint64_t row, uint8_t* out,
   return ybg_encode_dockey(&sc, &k, out, cap);
 }
 
+size_t GpuDocRowwiseIterator::EncodeRowKeyDyn(uint64_t row) {
+  if (keybuf_.size() < 256) keybuf_.resize(256);
+  for (;;) {
+    size_t n = EncodeRowKey(row, keybuf_.data(), keybuf_.size());
+    if (n) return n;  // 0 = capacity exceeded (ybg_encode_dockey contract)
+    if (keybuf_.size() >= (1u << 20)) return 0;  // corrupt, not just long
+    keybuf_.resize(keybuf_.size() * 2);
+  }
+}
+
 int GpuDocRowwiseIterator::PagingState(uint8_t* key_out, size_t cap,
                                        size_t* len) {
   *len = 0;
   if (!batch_ready_ || !limit_ || limit_ >= batch_.n_rows) return 0;
-  *len = EncodeRowKey(order_[limit_], key_out, cap);
+  size_t n = EncodeRowKeyDyn(order_[limit_]);
+  if (!n) return -1;        // row key failed to encode (corrupt state)
+  if (n > cap) return -1;   // explicit error, never a silent empty state
+  memcpy(key_out, keybuf_.data(), n);
+  *len = n;
   return 0;
 }
 
@@ -107,8 +121,11 @@ int GpuDocRowwiseIterator::GetTupleId(uint8_t* key_out, size_t cap,
                                       size_t* len) {
   *len = 0;
   if (!batch_ready_ || pos_ == 0 || pos_ > batch_.n_rows) return -1;
-  *len = EncodeRowKey(order_[pos_ - 1], key_out, cap);
-  return *len ? 0 : -1;
+  size_t n = EncodeRowKeyDyn(order_[pos_ - 1]);
+  if (!n || n > cap) return -1;
+  memcpy(key_out, keybuf_.data(), n);
+  *len = n;
+  return 0;
 }
 
 // ~ YQLRowwiseIteratorIf::SeekTuple (ql_rowwise_iterator_interface.h:
@@ -121,19 +138,21 @@ int GpuDocRowwiseIterator::SeekTuple(const uint8_t* dockey, size_t len) {
     if (rc) return -rc;
   }
   // rows are sorted by tablet key order: binary search on encoded keys
-  uint8_t buf[256];
+  // (keys encode into the growable keybuf_ — long string key columns must
+  // not truncate, or the search order breaks)
   uint64_t lo = 0, hi = batch_.n_rows;
   while (lo < hi) {
     uint64_t mid = (lo + hi) / 2;
-    size_t n = EncodeRowKey(order_[mid], buf, sizeof(buf));
-    int cmp = memcmp(buf, dockey, n < len ? n : len);
+    size_t n = EncodeRowKeyDyn(order_[mid]);
+    if (!n) return -1;
+    int cmp = memcmp(keybuf_.data(), dockey, n < len ? n : len);
     if (cmp == 0) cmp = (n < len) ? -1 : (n > len ? 1 : 0);
     if (cmp < 0) lo = mid + 1;
     else hi = mid;
   }
   if (lo >= batch_.n_rows) return 1;
-  size_t n = EncodeRowKey(order_[lo], buf, sizeof(buf));
-  if (n != len || memcmp(buf, dockey, len) != 0) return 1;
+  size_t n = EncodeRowKeyDyn(order_[lo]);
+  if (n != len || memcmp(keybuf_.data(), dockey, len) != 0) return 1;
   pos_ = lo;
   return 0;
 }

@@ -54,11 +54,15 @@ class GpuDocRowwiseIterator {
  private:
   int MaterializeBatch();
   size_t EncodeRowKey(uint64_t row, uint8_t* out, size_t cap);
+  // Encode into the growable keybuf_ (no fixed cap: string key columns
+  // make encoded DocKeys unbounded). Returns the length, 0 on error.
+  size_t EncodeRowKeyDyn(uint64_t row);
 
   ybg_scan_t* handle_ = nullptr;
   ybg_scan_spec_t spec_;
   ybg_row_batch_t batch_ = {};
   std::vector<uint64_t> order_;  // row indices sorted by sort_key
+  std::vector<uint8_t> keybuf_;  // EncodeRowKeyDyn scratch
   uint64_t pos_ = 0;
   uint64_t limit_ = 0;  // rows to deliver this page (0 = all)
   bool batch_ready_ = false;

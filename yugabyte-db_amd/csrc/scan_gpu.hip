@@ -285,12 +285,18 @@ __global__ __launch_bounds__(kThreads, 3) void k_group(
     const uint8_t* __restrict__ aux, uint8_t* __restrict__ rk_save_buf,
     GroupCtx gc, GroupHead* __restrict__ gheads,
     uint32_t* __restrict__ cont_flags,
-    unsigned long long* __restrict__ err_counter) {
+    unsigned long long* __restrict__ err_counter,
+    uint64_t* __restrict__ partials) {
   const DevSpec& sp = c_spec;
   __shared__ uint8_t key_scratch[kThreads * kKeyCap];
   __shared__ uint8_t wave_relay[kThreads / 64 + 1];
-  __shared__ uint64_t bht_scratch[kThreads * 3];
-  uint64_t* bht = bht_scratch + (size_t)threadIdx.x * 3;
+  // 6 slots/thread: bht[0..2] packed-row write time, bht[3..5] the
+  // restart-min candidate process_entry tracks when sp.track_restart
+  // (a 3-slot array here clobbered the neighbour's write-time slots and
+  // overran the shared array for threadIdx kThreads-1)
+  __shared__ uint64_t bht_scratch[kThreads * 6];
+  uint64_t* bht = bht_scratch + (size_t)threadIdx.x * 6;
+  bht[5] = 0;
   uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
   const uint32_t gtid = blockIdx.x * kThreads + threadIdx.x;
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
@@ -338,6 +344,27 @@ __global__ __launch_bounds__(kThreads, 3) void k_group(
         cont_flags[(j + 1) / kThreads] = 1;
     }
     __syncthreads();
+  }
+  // read-restart fold, as in k_scan: wave minimum of the encoded-HT
+  // restart candidates into the partial record's restart slots (the other
+  // slots were zeroed host-side; k_reduce folds these into DevResult so
+  // group scans report GetReadRestartData like plain scans)
+  if (sp.track_restart && (threadIdx.x & 63) == 0) {
+    uint64_t wave_id = ((uint64_t)blockIdx.x * kThreads + threadIdx.x) >> 6;
+    uint64_t mh = 0, ml = 0, mn = 0;
+    for (int l = 0; l < 64; ++l) {
+      const uint64_t* rr = bht_scratch + (size_t)(threadIdx.x + l) * 6 + 3;
+      if (rr[2] == 0) continue;
+      if (mn == 0 || u128_slice_cmp(rr[0], rr[1], (uint32_t)rr[2], mh, ml,
+                                    (uint32_t)mn) < 0) {
+        mh = rr[0];
+        ml = rr[1];
+        mn = rr[2];
+      }
+    }
+    partials[wave_id * kPartialStride + 4 + 2 * YBG_MAX_AGGS] = mh;
+    partials[wave_id * kPartialStride + 4 + 2 * YBG_MAX_AGGS + 1] = ml;
+    partials[wave_id * kPartialStride + 4 + 2 * YBG_MAX_AGGS + 2] = mn;
   }
 }
 
@@ -1253,20 +1280,39 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
   // single fused pass: head-deferral protocol identical to k_scan
   int hgrid = (int)std::min<uint64_t>(
       (s->n_ivs + kThreads - 1) / kThreads, 512);
+  if (s->dspec.track_restart) {
+    // k_group writes only the restart slots of each wave partial; zero the
+    // rest so the k_reduce fold below sees clean records
+    HIP_TRY(hipMemsetAsync(s->d_partials, 0,
+                           s->n_partials * kPartialStride * sizeof(uint64_t),
+                           s->stream));
+  }
   if (s->dspec.num_aggs <= 4) {
     hipLaunchKernelGGL(k_group<4>, dim3(s->grid), dim3(kThreads), 0,
                        s->stream, s->d_data, s->d_offsets, s->d_ivs,
                        s->n_ivs, s->d_aux, s->d_rk_save, s->gc, s->d_gheads,
-                       s->d_cont, s->gc.overflow);
+                       s->d_cont, s->gc.overflow, s->d_partials);
     hipLaunchKernelGGL(k_group_heads<4>, dim3(hgrid), dim3(kThreads), 0,
                        s->stream, s->gc, s->d_gheads, s->d_cont, s->n_heads);
   } else {
     hipLaunchKernelGGL(k_group<8>, dim3(s->grid), dim3(kThreads), 0,
                        s->stream, s->d_data, s->d_offsets, s->d_ivs,
                        s->n_ivs, s->d_aux, s->d_rk_save, s->gc, s->d_gheads,
-                       s->d_cont, s->gc.overflow);
+                       s->d_cont, s->gc.overflow, s->d_partials);
     hipLaunchKernelGGL(k_group_heads<8>, dim3(hgrid), dim3(kThreads), 0,
                        s->stream, s->gc, s->d_gheads, s->d_cont, s->n_heads);
+  }
+  if (s->dspec.track_restart) {
+    // fold the wave restart minima into DevResult (num_aggs = 0 spec: only
+    // the restart slots matter; n_heads = 0 skips the head records)
+    DevSpec rsp = s->dspec;
+    rsp.num_aggs = 0;
+    hipLaunchKernelGGL(k_reduce, dim3(1), dim3(256), 0, s->stream, rsp,
+                       s->d_partials, s->n_partials, s->d_heads, s->d_cont,
+                       0, s->d_result);
+  } else {
+    // restart impossible (local_limit == read): restart_data reports none
+    HIP_TRY(hipMemsetAsync(s->d_result, 0, sizeof(DevResult), s->stream));
   }
   // export buffers sized to caller caps
   if (!s->d_gk_out) {
@@ -1300,6 +1346,27 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
     HIP_TRY(hipMemcpy(key_bytes, s->d_gb_out,
                       std::min<uint64_t>(ctr[1], key_bytes_cap),
                       hipMemcpyDeviceToHost));
+  return 0;
+}
+
+// Read-restart data of the last execute/group_aggregate on this handle
+// (GetReadRestartData analog, intent_aware_iterator.cc:1400-1410): the
+// encoded DocHybridTime of the smallest-encoded (max commit time) visible
+// record past the read time, or len 0 when no restart is needed. The
+// aggregate path also surfaces this in ybg_scan_result_t; this entry point
+// serves the GROUP BY path, whose result shape has no restart field.
+int yb_gpu_scan_restart_data(ybg_scan_t* s, uint8_t* ht_out,
+                             uint32_t* len_out) {
+  HIP_TRY(hipStreamSynchronize(s->stream));
+  DevResult r;
+  HIP_TRY(hipMemcpy(&r, s->d_result, sizeof(r), hipMemcpyDeviceToHost));
+  uint32_t n = (uint32_t)r.restart_len;
+  if (n > YBG_MAX_HT) n = YBG_MAX_HT;
+  for (uint32_t i = 0; i < n; ++i)
+    ht_out[i] = (uint8_t)((i < 8 ? r.restart_hi >> (56 - 8 * i)
+                                 : r.restart_lo >> (56 - 8 * (i - 8))) &
+                          0xff);
+  *len_out = n;
   return 0;
 }
 

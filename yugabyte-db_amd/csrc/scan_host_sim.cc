@@ -197,7 +197,8 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
                   const uint64_t* offsets, uint64_t n_blocks,
                   uint64_t* keys_out, long long* vals_out,
                   unsigned long long* cnts_out, uint8_t* key_bytes_out,
-                  uint64_t key_bytes_cap, uint64_t cap, uint64_t* n_out) {
+                  uint64_t key_bytes_cap, uint64_t cap, uint64_t* n_out,
+                  uint8_t* restart_out, uint32_t* restart_len_out) {
   DevSpec d;
   std::vector<unsigned char> aux(1 << 20);
   uint32_t aux_len = 0;
@@ -268,6 +269,18 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
       group_accum_rec<YBG_MAX_AGGS>(d, gc, gheads[j]);
   }
   if (overflow) return 8;
+  if (restart_len_out) {  // restart-min slot, as ybg_sim_scan reports it
+    *restart_len_out = 0;
+    if (bht[5] && restart_out) {
+      uint32_t rn = (uint32_t)bht[5];
+      if (rn > YBG_MAX_HT) rn = YBG_MAX_HT;
+      for (uint32_t i = 0; i < rn; ++i)
+        restart_out[i] = (uint8_t)(
+            (i < 8 ? bht[3] >> (56 - 8 * i) : bht[4] >> (56 - 8 * (i - 8))) &
+            0xff);
+      *restart_len_out = rn;
+    }
+  }
   bool grp_is_str =
       d.cols[d.group_col].dtype == YBG_T_STRING;
   uint64_t n = 0, boff = 0;

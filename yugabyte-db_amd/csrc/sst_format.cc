@@ -76,7 +76,9 @@ int parse_sst(const uint8_t* file, uint64_t size, int verify,
     *err = "corrupt footer block handles";
     return 3;
   }
-  if (ix_off + ix_sz + kBlockTrailerSize > size) {
+  // overflow-safe range check: crafted varint handles can wrap uint64
+  if (ix_off > size || ix_sz > size - ix_off ||
+      kBlockTrailerSize > size - ix_off - ix_sz) {
     *err = "index handle out of range";
     return 3;
   }
@@ -127,7 +129,8 @@ int parse_sst(const uint8_t* file, uint64_t size, int verify,
       return 3;
     }
     e += vlen;
-    if (b_off + b_sz + kBlockTrailerSize > size) {
+    if (b_off > size || b_sz > size - b_off ||
+        kBlockTrailerSize > size - b_off - b_sz) {
       *err = "data block handle out of range";
       return 3;
     }
