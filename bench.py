@@ -41,6 +41,19 @@ def log(msg):
         print(msg, file=sys.stderr, flush=True)
 
 
+# all-cores CPU-baseline worker state (module level: fork workers inherit
+# the tablet copy-on-write; each worker scans a contiguous block chunk)
+_CPU_G = {}
+
+
+def _cpu_worker(rng):
+    import ybgpu as y
+    lo, hi = rng
+    res, _ = y.orcl_scan(_CPU_G["data"], _CPU_G["offsets"], hi - lo,
+                         _CPU_G["sc"], _CPU_G["spec"], block_lo=lo)
+    return res.rows_scanned, res.rows_matched
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -289,6 +302,7 @@ def main():
             res, _ = y.orcl_scan(data, offsets, nblocks, osc, ospec)
             return time.time() - t, res
 
+        # single-thread sample (secondary figure + per-block calibration)
         cal_blocks = min(nb, 256)
         cal_t, cal_res = run_orcl(cal_blocks)
         per_block = max(cal_t / cal_blocks, 1e-9)
@@ -296,17 +310,47 @@ def main():
                                     int(args.cpu_sample_seconds / per_block)))
         st, sres = run_orcl(sample_blocks)
         sample_rows = sres.rows_scanned
-        cpu_rows_per_s = sample_rows / st
+        cpu_1t_rows_per_s = sample_rows / st
+
+        # all-cores oracle on the SAME tablet in the SAME run (north-star
+        # measurement clause): one persistent fork pool, each worker owns a
+        # contiguous block chunk, warm-up pass touches each worker's OWN
+        # chunk (first-touch faults out of the timed region), then the
+        # whole tablet is scanned in parallel; best wall of 3 passes.
+        import multiprocessing as _mp
+        nproc = os.cpu_count() or 1
+        nproc = min(nproc, nb)
+        cuts = [nb * i // nproc for i in range(nproc + 1)]
+        ranges = [(cuts[i], cuts[i + 1]) for i in range(nproc)
+                  if cuts[i + 1] > cuts[i]]
+        _CPU_G.update(data=data, offsets=offsets, sc=osc, spec=ospec)
+        ctx = _mp.get_context("fork")
+        with ctx.Pool(len(ranges)) as pool:
+            pool.map(_cpu_worker, ranges)  # warm-up over own chunks
+            best = None
+            for _ in range(3):
+                t0 = time.time()
+                out_w = pool.map(_cpu_worker, ranges)
+                wall = time.time() - t0
+                if best is None or wall < best[0]:
+                    best = (wall, out_w)
+        wall, out_w = best
+        par_rows = sum(r[0] for r in out_w)
+        cpu_rows_per_s = par_rows / wall
         cpu_baseline = {
             "value": cpu_rows_per_s,
             "unit": "rows/s",
-            "cores": 1,
+            "cores": len(ranges),
             "kind": "port",
-            "sample": f"{sample_blocks} of {nb} blocks "
-                      f"({sample_rows} rows, {st:.1f}s, single thread)",
+            "sample": f"full tablet ({par_rows} rows) x best-of-3 parallel "
+                      f"passes, {wall:.2f}s wall; single-thread "
+                      f"{cpu_1t_rows_per_s/1e6:.1f} Mrows/s over "
+                      f"{sample_blocks} blocks ({st:.1f}s)",
+            "single_thread_value": cpu_1t_rows_per_s,
         }
-        log(f"[cpu baseline] {cpu_rows_per_s/1e6:.2f} Mrows/s "
-            f"(single-thread oracle)")
+        log(f"[cpu baseline] {cpu_rows_per_s/1e6:.1f} Mrows/s on "
+            f"{len(ranges)} cores; {cpu_1t_rows_per_s/1e6:.2f} Mrows/s "
+            f"single-thread")
 
     if rank == 0:
         out = {
