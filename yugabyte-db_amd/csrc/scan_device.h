@@ -2138,6 +2138,14 @@ DEV void emit_row(const DevSpec& sp, EmitCtx* ec, const RowCtxT<NA>& rc,
       sp.has_hash ? (uint16_t)(((uint16_t)rk[1] << 8) | rk[2]) : 0;
 }
 
+// Scan the interval range [j, j_hi) as ONE continuous stream (j_hi = j+1
+// is the classic one-interval form). The batch DEFERS its first (head)
+// row and WALKS its last row into interval j_hi; interior interval
+// boundaries are just stream positions (contiguous intervals of one block
+// continue without re-initializing the reader). iv_flags, when non-null
+// (the emit flags pre-pass), receives per-INTERVAL head-consumption:
+// iv_flags[i] = 1 iff interval i's first row is a continuation of a row
+// from interval i-1 — exactly what k_emit's per-interval workers need.
 template <int NA, bool EMIT = false, bool GROUP = false>
 DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
                            const uint64_t* block_offsets, const Interval* ivs,
@@ -2152,12 +2160,16 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
                            uint32_t* emit_lens = nullptr,
                            const GroupCtx* gc = nullptr,
                            const uint32_t* head_flags = nullptr,
-                           GroupHead* gh_out = nullptr) {
+                           GroupHead* gh_out = nullptr,
+                           uint64_t j_hi_in = 0,
+                           uint32_t* iv_flags = nullptr) {
+  const uint64_t j_hi = j_hi_in ? j_hi_in : j + 1;
   Interval iv = ivs[j];
   const uint8_t* blk = data + block_offsets[iv.block];
   const uint8_t* p = blk + iv.start;
   const uint8_t* limit = blk + iv.end;
   uint64_t cur_iv = j;
+  bool crossed = false;  // next decoded entry is the first of its interval
   Rdr rdr;
   rdr.init(p);
 
@@ -2193,10 +2205,17 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
       if (cur_iv + 1 >= n_ivs) break;
       Interval nx = ivs[cur_iv + 1];
       cur_iv += 1;
-      blk = data + block_offsets[nx.block];
-      p = blk + nx.start;
-      limit = blk + nx.end;
-      rdr.init(p);
+      crossed = true;
+      const uint8_t* nblk = data + block_offsets[nx.block];
+      const uint8_t* np = nblk + nx.start;
+      limit = nblk + nx.end;
+      if (np != p) {  // non-contiguous (block crossing): reposition
+        blk = nblk;
+        p = np;
+        rdr.init(p);
+      } else {
+        blk = nblk;
+      }
       // first entry of a restart interval is self-contained; the carried
       // key state is overwritten by its full-key decode.
       continue;
@@ -2312,7 +2331,7 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
         }
         in_head = false;
         row_open = false;
-        if (cur_iv > j) break;  // tail walk ended at a new row
+        if (cur_iv >= j_hi) break;  // tail walk ended at a new row
       }
       if (!rkb || EMIT) {
         for (uint32_t i = 0; i < rk; ++i) rk_save[i] = key[i];
@@ -2323,8 +2342,10 @@ DEV bool scan_one_interval(const DevSpec& sp, const uint8_t* data,
         row_sort_key = ((uint64_t)cur_iv << 16) | (uint64_t)(p - blk);
       row_reset(&rc, sp);
     }
-    *entries += (cur_iv == j);
-    walked_next |= (cur_iv == j + 1);
+    if (iv_flags && crossed && !row_change) iv_flags[cur_iv] = 1;
+    crossed = false;
+    *entries += (cur_iv < j_hi);
+    walked_next |= (cur_iv == j_hi);
     if (!process_entry(sp, data, aux, key, key_len, er.value, er.value_len,
                        rk_len, &rc, &rdr, ht_sz, ht_hi, ht_lo)) {
       fail = true;

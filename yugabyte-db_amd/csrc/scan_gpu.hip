@@ -66,7 +66,7 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
     const Interval* __restrict__ ivs, uint64_t n_ivs,
     const uint8_t* __restrict__ aux, uint8_t* __restrict__ rk_save_buf,
     uint64_t* __restrict__ partials, uint64_t* __restrict__ heads,
-    uint32_t* __restrict__ cont_flags, int write_all_flags) {
+    uint32_t* __restrict__ cont_flags, int write_all_flags, uint64_t ivb) {
   const DevSpec& sp = c_spec;
   __shared__ uint8_t key_scratch[kThreads * kKeyCap];
   __shared__ uint8_t wave_relay[kThreads / 64 + 1];
@@ -77,6 +77,9 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
   const uint32_t gtid = blockIdx.x * kThreads + threadIdx.x;
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
   const uint64_t span = (uint64_t)gridDim.x * kThreads;
+  // batch = ivb CONSECUTIVE intervals scanned as one stream by one thread
+  // (head deferral once per batch; the relay below runs on batch indices)
+  const uint64_t n_batches = (n_ivs + ivb - 1) / ivb;
 
   uint32_t entries = 0, scanned = 0, matched = 0, errs = 0;
   // NA-sized and only ever constant-indexed (all loops over them unrolled):
@@ -87,16 +90,21 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
 #pragma unroll
   for (int g = 0; g < NA; ++g) { agg_val[g] = 0; agg_cnt[g] = 0; }
 
-  for (uint64_t j0 = 0; j0 < n_ivs; j0 += span) {
+  for (uint64_t j0 = 0; j0 < n_batches; j0 += span) {
     const uint64_t j = j0 + gtid;
-    const bool active = j < n_ivs;
+    const bool active = j < n_batches;
     HeadOut<NA> ho;
     bool walked_next = false;
     if (active) {
-      if (!scan_one_interval<NA>(sp, data, block_offsets, ivs, n_ivs, j, aux,
-                                 key, rk_save, bht, &entries, &scanned,
+      const uint64_t lo = j * ivb;
+      uint64_t hi = lo + ivb;
+      if (hi > n_ivs) hi = n_ivs;
+      if (!scan_one_interval<NA>(sp, data, block_offsets, ivs, n_ivs, lo,
+                                 aux, key, rk_save, bht, &entries, &scanned,
                                  &matched, agg_val, agg_cnt, &ho,
-                                 &walked_next)) {
+                                 &walked_next, nullptr, nullptr, nullptr,
+                                 nullptr, nullptr, nullptr, hi,
+                                 write_all_flags ? cont_flags : nullptr)) {
         errs += 1;
       }
     } else {
@@ -142,10 +150,11 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
                      ho.cnt[g]);
         }
       }
-      if (write_all_flags) {
-        if (walked_next && (j + 1) < n_ivs) cont_flags[j + 1] = 1;
-      } else if (walked_next && ((j + 1) % kThreads) == 0 &&
-                 (j + 1) < n_ivs) {
+      // write_all_flags: the per-interval flags are written inside the
+      // walk (iv_flags); only the batch-level workgroup-boundary flag is
+      // recorded here for the head-record fold.
+      if (!write_all_flags && walked_next && ((j + 1) % kThreads) == 0 &&
+          (j + 1) < n_batches) {
         cont_flags[(j + 1) / kThreads] = 1;
       }
     }
@@ -286,7 +295,7 @@ __global__ __launch_bounds__(kThreads, 3) void k_group(
     GroupCtx gc, GroupHead* __restrict__ gheads,
     uint32_t* __restrict__ cont_flags,
     unsigned long long* __restrict__ err_counter,
-    uint64_t* __restrict__ partials) {
+    uint64_t* __restrict__ partials, uint64_t ivb) {
   const DevSpec& sp = c_spec;
   __shared__ uint8_t key_scratch[kThreads * kKeyCap];
   __shared__ uint8_t wave_relay[kThreads / 64 + 1];
@@ -301,21 +310,26 @@ __global__ __launch_bounds__(kThreads, 3) void k_group(
   const uint32_t gtid = blockIdx.x * kThreads + threadIdx.x;
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
   const uint64_t span = (uint64_t)gridDim.x * kThreads;
+  const uint64_t n_batches = (n_ivs + ivb - 1) / ivb;
   uint32_t entries = 0, scanned = 0, matched = 0;
   uint64_t agg_val[NA] = {0}, agg_cnt[NA] = {0};
 
-  for (uint64_t j0 = 0; j0 < n_ivs; j0 += span) {
+  for (uint64_t j0 = 0; j0 < n_batches; j0 += span) {
     const uint64_t j = j0 + gtid;
-    const bool active = j < n_ivs;
+    const bool active = j < n_batches;
     HeadOut<NA> ho;
     GroupHead gh;
     gh.hit = 0;
     bool walked_next = false;
     if (active) {
+      const uint64_t lo = j * ivb;
+      uint64_t hi = lo + ivb;
+      if (hi > n_ivs) hi = n_ivs;
       if (!scan_one_interval<NA, false, true>(
-              sp, data, block_offsets, ivs, n_ivs, j, aux, key, rk_save,
+              sp, data, block_offsets, ivs, n_ivs, lo, aux, key, rk_save,
               bht, &entries, &scanned, &matched, agg_val, agg_cnt, &ho,
-              &walked_next, nullptr, nullptr, nullptr, &gc, nullptr, &gh)) {
+              &walked_next, nullptr, nullptr, nullptr, &gc, nullptr, &gh,
+              hi)) {
         atomicAdd(err_counter, 1ull);
       }
     }
@@ -340,7 +354,7 @@ __global__ __launch_bounds__(kThreads, 3) void k_group(
       } else if (!head_consumed && gh.hit) {
         group_accum_rec<NA>(sp, gc, gh);
       }
-      if (walked_next && ((j + 1) % kThreads) == 0 && (j + 1) < n_ivs)
+      if (walked_next && ((j + 1) % kThreads) == 0 && (j + 1) < n_batches)
         cont_flags[(j + 1) / kThreads] = 1;
     }
     __syncthreads();
@@ -693,6 +707,8 @@ struct ybg_scan {
   uint64_t* d_offsets = nullptr;
   Interval* d_ivs = nullptr;
   uint64_t n_ivs = 0;
+  uint64_t ivb = 1;       // intervals per batch (one thread's stream)
+  uint64_t n_batches = 0;
   uint64_t n_blocks = 0;
   uint64_t total_bytes = 0;
   uint8_t* d_aux = nullptr;
@@ -884,7 +900,13 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
   HIP_WARN(hipFree(d_counts));
   HIP_WARN(hipFree(d_err));
 
-  uint64_t want = (s->n_ivs + kThreads - 1) / kThreads;
+  s->ivb = 1;
+  if (const char* e = getenv("YBG_IVB")) {
+    long v = atol(e);
+    if (v >= 1 && v <= 4096) s->ivb = (uint64_t)v;
+  }
+  s->n_batches = (s->n_ivs + s->ivb - 1) / s->ivb;
+  uint64_t want = (s->n_batches + kThreads - 1) / kThreads;
   uint64_t cap = 8192;
   if (const char* g = getenv("YBG_GRID")) {
     long v = atol(g);
@@ -894,7 +916,7 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
   if (s->grid < 1) s->grid = 1;
   uint64_t span_threads = (uint64_t)s->grid * kThreads;
   s->n_partials = span_threads / 64;
-  s->n_heads = (s->n_ivs + kThreads - 1) / kThreads;
+  s->n_heads = (s->n_batches + kThreads - 1) / kThreads;
   HIP_TRY(hipMalloc(&s->d_rk_save, span_threads * kKeyCap));
   HIP_TRY(hipMalloc(&s->d_partials,
                     s->n_partials * kPartialStride * sizeof(uint64_t)));
@@ -1052,7 +1074,7 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
     hipLaunchKernelGGL(kern, dim3(s->grid), dim3(kThreads), 0, s->stream,
                        s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
                        s->d_aux, s->d_rk_save, s->d_partials, s->d_heads,
-                       s->d_cont, 0);
+                       s->d_cont, 0, s->ivb);
   };
   int na = s->dspec.num_aggs <= 2 ? 2 : (s->dspec.num_aggs <= 4 ? 4 : 8);
   switch (na * 10 + wps) {
@@ -1166,7 +1188,7 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
   hipLaunchKernelGGL(flags_kernel, dim3(s->grid), dim3(kThreads), 0,
                      s->stream, s->d_data, s->d_offsets, s->d_ivs,
                      s->n_ivs, s->d_aux, s->d_rk_save, s->d_partials,
-                     s->d_heads, s->d_flags_all, 1);
+                     s->d_heads, s->d_flags_all, 1, s->ivb);
   EmitCtx ec;
   ec.sort_key = s->d_em_sort;
   ec.key_datums = s->d_em_key;
@@ -1291,14 +1313,14 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
     hipLaunchKernelGGL(k_group<4>, dim3(s->grid), dim3(kThreads), 0,
                        s->stream, s->d_data, s->d_offsets, s->d_ivs,
                        s->n_ivs, s->d_aux, s->d_rk_save, s->gc, s->d_gheads,
-                       s->d_cont, s->gc.overflow, s->d_partials);
+                       s->d_cont, s->gc.overflow, s->d_partials, s->ivb);
     hipLaunchKernelGGL(k_group_heads<4>, dim3(hgrid), dim3(kThreads), 0,
                        s->stream, s->gc, s->d_gheads, s->d_cont, s->n_heads);
   } else {
     hipLaunchKernelGGL(k_group<8>, dim3(s->grid), dim3(kThreads), 0,
                        s->stream, s->d_data, s->d_offsets, s->d_ivs,
                        s->n_ivs, s->d_aux, s->d_rk_save, s->gc, s->d_gheads,
-                       s->d_cont, s->gc.overflow, s->d_partials);
+                       s->d_cont, s->gc.overflow, s->d_partials, s->ivb);
     hipLaunchKernelGGL(k_group_heads<8>, dim3(hgrid), dim3(kThreads), 0,
                        s->stream, s->gc, s->d_gheads, s->d_cont, s->n_heads);
   }

@@ -8,9 +8,22 @@
 #define YBG_DEV_QUAL inline
 #include "scan_device.h"
 
+#include <cstdlib>
 #include <vector>
 
 using namespace ybgdev;
+
+namespace {
+// Intervals per batch, as the kernels read it (YBG_IVB): the sim loops
+// over the same batch decomposition so C > 1 is parity-testable on CPU.
+uint64_t sim_ivb() {
+  if (const char* e = getenv("YBG_IVB")) {
+    long v = atol(e);
+    if (v >= 1 && v <= 4096) return (uint64_t)v;
+  }
+  return 1;
+}
+}  // namespace
 
 extern "C" {
 
@@ -44,34 +57,40 @@ int ybg_sim_scan(const ybg_scan_spec_t* spec, const uint8_t* data,
   }
   uint64_t n_ivs = ivs.size();
 
-  // ---- per-interval scan (k_scan body, serial) ----
+  // ---- per-batch scan (k_scan body, serial) ----
+  const uint64_t ivb = sim_ivb();
+  const uint64_t n_b = (n_ivs + ivb - 1) / ivb;
   uint64_t entries = 0, scanned = 0, matched = 0;
   uint64_t agg_val[YBG_MAX_AGGS] = {0}, agg_cnt[YBG_MAX_AGGS] = {0};
-  std::vector<HeadOut<YBG_MAX_AGGS>> heads(n_ivs);
-  std::vector<uint8_t> walked(n_ivs, 0);
+  std::vector<HeadOut<YBG_MAX_AGGS>> heads(n_b);
+  std::vector<uint8_t> walked(n_b, 0);
   alignas(8) uint8_t key[kKeyCap];
   alignas(8) uint8_t rk_save[kKeyCap];
   uint64_t bht[6] = {0, 0, 0, 0, 0, 0};
-  for (uint64_t j = 0; j < n_ivs; ++j) {
+  for (uint64_t b = 0; b < n_b; ++b) {
     bool wn = false;
     uint32_t e32 = 0, s32 = 0, m32 = 0;
+    uint64_t lo = b * ivb;
+    uint64_t hi = lo + ivb < n_ivs ? lo + ivb : n_ivs;
     if (!scan_one_interval<YBG_MAX_AGGS>(d, data, offsets, ivs.data(), n_ivs,
-                                         j, aux.data(), key, rk_save, bht,
+                                         lo, aux.data(), key, rk_save, bht,
                                          &e32, &s32, &m32,
-                                         agg_val, agg_cnt, &heads[j], &wn))
+                                         agg_val, agg_cnt, &heads[b], &wn,
+                                         nullptr, nullptr, nullptr, nullptr,
+                                         nullptr, nullptr, hi))
       return 6;
     entries += e32;
     scanned += s32;
     matched += m32;
-    walked[j] = wn ? 1 : 0;
+    walked[b] = wn ? 1 : 0;
   }
   // head ownership resolution (shfl relay / cont flags, serial equivalent)
-  for (uint64_t j = 0; j < n_ivs; ++j) {
-    bool consumed = j > 0 && walked[j - 1];
+  for (uint64_t b = 0; b < n_b; ++b) {
+    bool consumed = b > 0 && walked[b - 1];
     if (consumed) continue;
-    scanned += heads[j].scanned;
-    matched += heads[j].matched;
-    agg_combine(d, agg_val, agg_cnt, heads[j].val, heads[j].cnt);
+    scanned += heads[b].scanned;
+    matched += heads[b].matched;
+    agg_combine(d, agg_val, agg_cnt, heads[b].val, heads[b].cnt);
   }
 
   memset(out, 0, sizeof(*out));
@@ -139,24 +158,30 @@ int ybg_sim_emit(const ybg_scan_spec_t* spec, const uint8_t* data,
     }
   }
   uint64_t n_ivs = ivs.size();
+  const uint64_t ivb = sim_ivb();
+  const uint64_t n_b = (n_ivs + ivb - 1) / ivb;
   uint32_t entries = 0, scanned = 0, matched = 0;
   uint64_t agg_val[YBG_MAX_AGGS] = {0}, agg_cnt[YBG_MAX_AGGS] = {0};
-  std::vector<uint8_t> walked(n_ivs, 0);
   std::vector<uint32_t> head_consumed(n_ivs ? n_ivs : 1, 0);
   alignas(8) uint8_t key[kKeyCap];
   alignas(8) uint8_t rk_save[kKeyCap];
   uint64_t bht[6] = {0, 0, 0, 0, 0, 0};
   HeadOut<YBG_MAX_AGGS> ho;
-  for (uint64_t j = 0; j < n_ivs; ++j) {
+  // flags pre-pass (write_all_flags form): per-interval head consumption
+  // is written inside the walk via iv_flags
+  for (uint64_t b = 0; b < n_b; ++b) {
     bool wn = false;
+    uint64_t lo = b * ivb;
+    uint64_t hi = lo + ivb < n_ivs ? lo + ivb : n_ivs;
     if (!scan_one_interval<YBG_MAX_AGGS>(d, data, offsets, ivs.data(), n_ivs,
-                                         j, aux.data(), key, rk_save, bht,
+                                         lo, aux.data(), key, rk_save, bht,
                                          &entries, &scanned, &matched,
-                                         agg_val, agg_cnt, &ho, &wn))
+                                         agg_val, agg_cnt, &ho, &wn,
+                                         nullptr, nullptr, nullptr, nullptr,
+                                         nullptr, nullptr, hi,
+                                         head_consumed.data()))
       return 6;
-    walked[j] = wn ? 1 : 0;
   }
-  for (uint64_t j = 1; j < n_ivs; ++j) head_consumed[j] = walked[j - 1];
 
   unsigned long long row_counter = 0, varlen_counter = 0, overflow = 0;
   EmitCtx ec;
@@ -222,10 +247,12 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
     }
   }
   uint64_t n_ivs = ivs.size();
+  const uint64_t ivb = sim_ivb();
+  const uint64_t n_b = (n_ivs + ivb - 1) / ivb;
   uint32_t entries = 0, scanned = 0, matched = 0;
   uint64_t agg_val[YBG_MAX_AGGS] = {0}, agg_cnt[YBG_MAX_AGGS] = {0};
-  std::vector<uint8_t> walked(n_ivs, 0);
-  std::vector<GroupHead> gheads(n_ivs ? n_ivs : 1);
+  std::vector<uint8_t> walked(n_b, 0);
+  std::vector<GroupHead> gheads(n_b ? n_b : 1);
   alignas(8) uint8_t key[kKeyCap];
   alignas(8) uint8_t rk_save[kKeyCap];
   uint64_t bht[6] = {0, 0, 0, 0, 0, 0};
@@ -251,22 +278,24 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
   gc.cap = gcap;
   gc.overflow = &overflow;
   gc.data = data;
-  for (uint64_t j = 0; j < n_ivs; ++j) {
+  for (uint64_t b = 0; b < n_b; ++b) {
     bool wn = false;
     HeadOut<YBG_MAX_AGGS> ho;
-    gheads[j].hit = 0;
+    gheads[b].hit = 0;
+    uint64_t lo = b * ivb;
+    uint64_t hi = lo + ivb < n_ivs ? lo + ivb : n_ivs;
     if (!scan_one_interval<YBG_MAX_AGGS, false, true>(
-            d, data, offsets, ivs.data(), n_ivs, j, aux.data(), key, rk_save,
+            d, data, offsets, ivs.data(), n_ivs, lo, aux.data(), key, rk_save,
             bht, &entries, &scanned, &matched, agg_val, agg_cnt, &ho, &wn,
-            nullptr, nullptr, nullptr, &gc, nullptr, &gheads[j]))
+            nullptr, nullptr, nullptr, &gc, nullptr, &gheads[b], hi))
       return 6;
-    walked[j] = wn ? 1 : 0;
+    walked[b] = wn ? 1 : 0;
   }
   // head-ownership resolution (single-pass protocol, serial equivalent)
-  for (uint64_t j = 0; j < n_ivs; ++j) {
-    bool consumed = j > 0 && walked[j - 1];
-    if (!consumed && gheads[j].hit)
-      group_accum_rec<YBG_MAX_AGGS>(d, gc, gheads[j]);
+  for (uint64_t b = 0; b < n_b; ++b) {
+    bool consumed = b > 0 && walked[b - 1];
+    if (!consumed && gheads[b].hit)
+      group_accum_rec<YBG_MAX_AGGS>(d, gc, gheads[b]);
   }
   if (overflow) return 8;
   if (restart_len_out) {  // restart-min slot, as ybg_sim_scan reports it
