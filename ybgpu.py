@@ -471,6 +471,25 @@ def sim_scan(spec, data, offsets, n_blocks):
     return res
 
 
+def sim_scan_fast(spec, data, offsets, n_blocks):
+    """Host simulator of the SPECIALIZED fast batch scanner
+    (scan_batch_fast + general fallback per aborted batch) — TEST
+    INFRASTRUCTURE. Returns (ScanResult, n_fallback_batches); raises on
+    ineligible specs (rc 9)."""
+    lib = product()
+    f = _sig(lib, "ybg_sim_scan_fast", C.c_int,
+             [C.POINTER(ScanSpec), C.POINTER(C.c_uint8),
+              C.POINTER(C.c_uint64), C.c_uint64, C.POINTER(ScanResult),
+              C.POINTER(C.c_uint64)])
+    res = ScanResult()
+    nf = C.c_uint64()
+    rc = f(C.byref(spec), data, offsets, n_blocks, C.byref(res),
+           C.byref(nf))
+    if rc != 0:
+        raise RuntimeError(f"ybg_sim_scan_fast failed rc={rc}")
+    return res, nf.value
+
+
 def sim_emit(spec, data, offsets, n_blocks, row_cap=1 << 20,
              varlen_cap=1 << 24):
     """Host-simulator row emission (exact device emit path, serial) —

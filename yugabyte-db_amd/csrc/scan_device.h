@@ -781,20 +781,23 @@ DEV const uint8_t* decode_entry_ptr(int fmt, const uint8_t* p,
   return out->value + value_size;
 }
 
-// Rdr-based entry decode: fast path for the dominant three_shared_parts
-// encodings (frequent case and case 2.1.1, block_builder_internal.h:139-183)
-// — the whole header plus the <=10 non-shared key bytes are parsed from the
+// Fast-entry decode core for the dominant three_shared_parts encodings
+// (frequent case and case 2.1.1, block_builder_internal.h:139-183) — the
+// whole header plus the <=10 non-shared key bytes are parsed from the
 // register window with no per-byte memory chain, and in these cases the
-// shared middle never moves (ns1_delta == 0), so only ns1+ns2 LDS bytes are
-// written. Everything else delegates to decode_entry_ptr. On success the
-// reader is positioned AT THE VALUE START (value not consumed).
-DEV const uint8_t* decode_entry(int fmt, Rdr* rdr, const uint8_t* limit,
-                                uint8_t* key, uint32_t* key_len,
-                                uint64_t* last8, uint32_t rkb, bool* changed,
-                                EntryRef* out, uint64_t* thi, uint64_t* tlo,
-                                bool* tail_valid) {
+// shared middle never moves (ns1_delta == 0), so only ns1+ns2 LDS bytes
+// are written. Returns the pointer past the entry (reader at the value
+// start) or nullptr when the entry is NOT one of the fast forms (state
+// untouched; caller falls back — or, in the specialized kernel, aborts
+// the batch). Shared verbatim by decode_entry and k_scan_fast.
+DEV const uint8_t* decode_entry_fast(Rdr* rdr, const uint8_t* limit,
+                                     uint8_t* key, uint32_t* key_len,
+                                     uint64_t* last8, uint32_t rkb,
+                                     bool* changed, EntryRef* out,
+                                     uint64_t* thi, uint64_t* tlo) {
   const uint8_t* p = rdr->pos();
-  if (fmt == YBG_ENC_THREE_SHARED_PARTS && limit - p >= 8) {
+  if (limit - p < 8) return nullptr;
+  {
     uint64_t w = rdr->peek8();
     uint32_t b0 = (uint32_t)(w & 0xff);
     uint64_t e1;
@@ -896,7 +899,23 @@ DEV const uint8_t* decode_entry(int fmt, Rdr* rdr, const uint8_t* limit,
       }
     }
   }
+  return nullptr;
+}
+
+// Full entry decode: fast core first, general fallback. On success the
+// reader is positioned AT THE VALUE START (value not consumed).
+DEV const uint8_t* decode_entry(int fmt, Rdr* rdr, const uint8_t* limit,
+                                uint8_t* key, uint32_t* key_len,
+                                uint64_t* last8, uint32_t rkb, bool* changed,
+                                EntryRef* out, uint64_t* thi, uint64_t* tlo,
+                                bool* tail_valid) {
+  if (fmt == YBG_ENC_THREE_SHARED_PARTS) {
+    const uint8_t* q = decode_entry_fast(rdr, limit, key, key_len, last8,
+                                         rkb, changed, out, thi, tlo);
+    if (q) return q;
+  }
   // general path: the register tail no longer mirrors LDS
+  const uint8_t* p = rdr->pos();
   const uint8_t* q =
       decode_entry_ptr(fmt, p, limit, key, key_len, last8, rkb, changed, out);
   if (!q) return nullptr;
@@ -2136,6 +2155,524 @@ DEV void emit_row(const DevSpec& sp, EmitCtx* ec, const RowCtxT<NA>& rc,
   ec->null_masks[slot] = rc.emit_null;
   ec->hashes[slot] =
       sp.has_hash ? (uint16_t)(((uint16_t)rk[1] << 8) | rk[2]) : 0;
+}
+
+// ---------------------------------------------------------------------------
+// Specialized fast batch scanner (k_scan_fast): the dominant shape —
+// fixed-length rowkeys, packed-V2 all-fixed-width values, frequent/2.1.1
+// entry encodings, value-column typed compares, <= NA aggregates — as a
+// SMALL straight-line loop. The general scan_one_interval below compiles
+// to a ~380K-instruction mega-kernel whose issue slots drown in phi-copies,
+// exec-mask bookkeeping and SGPR-spill lane shuffles; this loop carries a
+// fraction of the live state so the compiler emits none of that. Anything
+// outside the shape ABORTS the batch (return 0) with no side effects on
+// the accumulators; the caller re-runs the batch through the general path.
+// Semantics are pinned against scan_one_interval by sim + GPU parity tests.
+// ---------------------------------------------------------------------------
+
+// Accumulate ONE row's aggregate operand (acc_row's per-slot body).
+DEV void combine_datum(int op, uint64_t* val, uint64_t* cnt, uint64_t v) {
+  switch (op) {
+    case YBG_AGG_COUNT_STAR:
+    case YBG_AGG_COUNT:
+      *val += 1;
+      break;
+    case YBG_AGG_SUM_INT64:
+      *val = (uint64_t)((int64_t)*val + (int64_t)v);
+      break;
+    case YBG_AGG_SUM_DOUBLE: {
+      double cur = __longlong_as_double((long long)*val) +
+                   __longlong_as_double((long long)v);
+      *val = (uint64_t)__double_as_longlong(cur);
+      break;
+    }
+    case YBG_AGG_MIN_INT64:
+      if (*cnt == 0 || (int64_t)v < (int64_t)*val) *val = v;
+      break;
+    case YBG_AGG_MAX_INT64:
+      if (*cnt == 0 || (int64_t)v > (int64_t)*val) *val = v;
+      break;
+    case YBG_AGG_MIN_DOUBLE: {
+      double dd = __longlong_as_double((long long)v);
+      if (*cnt == 0 || dd < __longlong_as_double((long long)*val))
+        *val = (uint64_t)__double_as_longlong(dd);
+      break;
+    }
+    case YBG_AGG_MAX_DOUBLE: {
+      double dd = __longlong_as_double((long long)v);
+      if (*cnt == 0 || dd > __longlong_as_double((long long)*val))
+        *val = (uint64_t)__double_as_longlong(dd);
+      break;
+    }
+  }
+  *cnt += 1;
+}
+
+// Lean typed compare (pred_compare's non-string, non-IN subset).
+DEV bool pred_cmp_fast(const PredC& pr, uint64_t datum) {
+  const int dtype = (int)(pr.opdt >> 8);
+  int cmp;
+  if (YBG_LIKELY(dtype != YBG_T_DOUBLE && dtype != YBG_T_FLOAT)) {
+    int64_t a = (int64_t)datum, b = (int64_t)pr.datum;
+    cmp = a < b ? -1 : (a > b ? 1 : 0);
+  } else if (dtype == YBG_T_DOUBLE) {
+    double a = __longlong_as_double((long long)datum);
+    double b = __longlong_as_double((long long)pr.datum);
+    cmp = a < b ? -1 : (a > b ? 1 : 0);
+  } else {
+    float a = __uint_as_float((uint32_t)datum);
+    float b = __uint_as_float((uint32_t)pr.datum);
+    cmp = a < b ? -1 : (a > b ? 1 : 0);
+  }
+  switch (pr.opdt & 0xff) {
+    case YBG_PRED_GT: return cmp > 0;
+    case YBG_PRED_GE: return cmp >= 0;
+    case YBG_PRED_LT: return cmp < 0;
+    case YBG_PRED_LE: return cmp <= 0;
+    case YBG_PRED_EQ: return cmp == 0;
+    default: return cmp != 0;
+  }
+}
+
+// Host-side: may this spec take the fast kernel at all?
+inline bool fast_eligible(const DevSpec& d) {
+  if (d.fmt != YBG_ENC_THREE_SHARED_PARTS) return false;
+  if (!d.fixed_rk_len || d.need_rowkey || d.group_col >= 0) return false;
+  if (!d.v2_fixed_len || d.num_value_cols <= 0) return false;
+  if (d.num_aggs > 2) return false;
+  for (int i = 0; i < d.num_preds; ++i) {
+    if (d.preds[i].is_key_col) return false;
+    if (d.preds[i].op > YBG_PRED_NE) return false;  // no IN / IN_TUPLE
+  }
+  return true;
+}
+
+#ifndef YBG_FABORT
+#define YBG_FABORT(n) return 0
+#endif
+
+// Returns 1 = batch done (accumulators updated), 0 = abort (accumulators
+// and ho untouched except bht[3..5] restart candidates, which are benign
+// duplicates under the MIN fold when the batch is retried).
+template <int NA>
+DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
+                        const uint64_t* block_offsets, const Interval* ivs,
+                        uint64_t n_ivs, uint64_t j_lo, uint64_t j_hi,
+                        uint8_t* key, uint64_t* bht,
+                        uint32_t* entries, uint32_t* scanned,
+                        uint32_t* matched, uint64_t* agg_val,
+                        uint64_t* agg_cnt, HeadOut<NA>* ho,
+                        bool* walked_next_out) {
+  Interval iv = ivs[j_lo];
+  const uint8_t* blk = data + block_offsets[iv.block];
+  const uint8_t* p = blk + iv.start;
+  const uint8_t* limit = blk + iv.end;
+  uint64_t cur_iv = j_lo;
+  Rdr rdr;
+  rdr.init(p);
+  bool at_restart = true;  // interval starts are restart points
+
+#pragma unroll
+  for (int g = 0; g < NA; ++g) { ho->val[g] = 0; ho->cnt[g] = 0; }
+  ho->scanned = 0;
+  ho->matched = 0;
+
+  // batch-local accumulators: folded into the caller's only on success
+  uint32_t e_b = 0, s_b = 0, m_b = 0;
+  uint64_t av_b[NA], ac_b[NA];
+#pragma unroll
+  for (int g = 0; g < NA; ++g) { av_b[g] = 0; ac_b[g] = 0; }
+
+  uint32_t key_len = 0;
+  uint64_t last8 = 0;
+  uint64_t thi = 0, tlo = 0;
+  bool row_open = false, in_head = true, walked = false;
+  bool base_seen = false, found = false;
+  uint32_t pred_pass = 0, agg_null = 0xffffffffu;
+  uint64_t agg_datum[NA];
+#pragma unroll
+  for (int g = 0; g < NA; ++g) agg_datum[g] = 0;
+  const uint32_t rkb = sp.fixed_rk_len;
+  EntryRef er;
+
+  for (;;) {
+    if (p >= limit) {
+      if (cur_iv + 1 >= n_ivs) break;
+      Interval nx = ivs[cur_iv + 1];
+      cur_iv += 1;
+      at_restart = true;
+      const uint8_t* nblk = data + block_offsets[nx.block];
+      const uint8_t* np = nblk + nx.start;
+      limit = nblk + nx.end;
+      blk = nblk;
+      if (np != p) {
+        p = np;
+        rdr.init(p);
+      }
+      continue;
+    }
+    bool kchg;
+    const uint8_t* q = nullptr;
+    if (YBG_UNLIKELY(at_restart)) {
+      // compact restart decode: three_shared_parts no-reuse form
+      // (block_builder_internal.h case 2.0; full key inline). Keys longer
+      // than 127 B (e2 == 0 -> varint ns1) or shorter than 24 B (register
+      // tail needs ukey >= 16) abort to the general path.
+      at_restart = false;
+      if (limit - p < 3) YBG_FABORT(1);
+      uint64_t w = load_u64_una(p);
+      uint32_t b0 = (uint32_t)(w & 0xff);
+      uint64_t e1;
+      uint32_t e1len;
+      if (!(b0 & 0x80)) {
+        e1 = b0;
+        e1len = 1;
+      } else {
+        uint32_t b1 = (uint32_t)((w >> 8) & 0xff);
+        if (b1 & 0x80) YBG_FABORT(2);
+        e1 = (b0 & 0x7f) | ((uint64_t)b1 << 7);
+        e1len = 2;
+      }
+      uint32_t value_size = (uint32_t)(e1 >> 2);
+      if (e1 & 1) YBG_FABORT(3);  // frequent case never starts a restart
+      uint32_t e2 = (uint32_t)((w >> (8 * e1len)) & 0xff);
+      if (e2 & 1) YBG_FABORT(4);  // reuse forms: not a self-contained restart
+      uint32_t ns1 = e2 >> 1;
+      if (ns1 < 24 || ns1 > kKeyCap - 8) YBG_FABORT(5);
+      uint32_t hl = e1len + 1;
+      if ((uint64_t)(limit - p) < (uint64_t)hl + ns1 + value_size) YBG_FABORT(6);
+      const uint8_t* kp2 = p + hl;
+      // row-change detection: compare the first rkb bytes before copying
+      kchg = !row_open;
+      if (row_open) {
+        uint32_t full = rkb & ~7u;
+        for (uint32_t i = 0; i < full; i += 8)
+          if (load_u64_una(&key[i]) != load_u64_una(kp2 + i)) kchg = true;
+        if (rkb & 7) {
+          uint64_t m = (1ull << (8 * (rkb & 7))) - 1;
+          if (((load_u64_una(&key[full]) ^ load_u64_una(kp2 + full)) & m))
+            kchg = true;
+        }
+      }
+      for (uint32_t i = 0; i < ns1; i += 8) {
+        uint64_t w8 = load_u64_una(kp2 + i);
+        __builtin_memcpy(&key[i], &w8, 8);
+      }
+      key_len = ns1;
+      last8 = load_u64_una(key + ns1 - 8);
+      tail_from_lds(key, ns1 - 8, &thi, &tlo);
+      er.value = kp2 + ns1;
+      er.value_len = value_size;
+      er.shared = 0;
+      q = er.value + value_size;
+      rdr.seek(er.value);
+    } else {
+      // Unified STABLE-LENGTH delta decode: frequent, 2.1.1 (d2 == 0) and
+      // the general form with last8 reuse and zero deltas
+      // (block_builder_internal.h:100-239). Zero deltas mean prev_ns2 ==
+      // ns2 and prev_mid_start == new_mid_start: the key LENGTH and every
+      // byte position are unchanged, the shared middle never moves, and
+      // the only changed bytes are [sp, sp+ns1) and the ns2 range. Bytes
+      // below the fixed rowkey length go to LDS (the restart row-compare
+      // reads them); bytes inside the register tail window are patched in
+      // registers; bytes between rkb and the tail window need no store at
+      // all (nothing reads them before the next restart overwrites).
+      if (limit - p < 8) YBG_FABORT(7);
+      uint64_t w = rdr.peek8();
+      uint32_t b0 = (uint32_t)(w & 0xff);
+      uint64_t e1;
+      uint32_t e1len;
+      if (!(b0 & 0x80)) {
+        e1 = b0;
+        e1len = 1;
+      } else {
+        uint32_t b1 = (uint32_t)((w >> 8) & 0xff);
+        if (b1 & 0x80) YBG_FABORT(8);
+        e1 = (b0 & 0x7f) | ((uint64_t)b1 << 7);
+        e1len = 2;
+      }
+      const uint32_t value_size = (uint32_t)(e1 >> 2);
+      const uint64_t inc = (e1 & 2) << 7;
+      uint32_t ns1, ns2, hl, sp;
+      uint64_t w2 = rdr.peek8_at(8);
+      auto hdr_byte = [&](uint32_t j) -> uint32_t {
+        return (uint32_t)((j < 8 ? (w >> (8 * j)) : (w2 >> (8 * (j - 8)))) &
+                          0xff);
+      };
+      if (e1 & 1) {  // frequent
+        sp = hdr_byte(e1len);
+        if (sp & 0x80) YBG_FABORT(9);
+        ns1 = 1;
+        ns2 = 1;
+        hl = e1len + 1;
+      } else {
+        uint32_t e2 = hdr_byte(e1len);
+        if ((e2 & 3) == 1) {  // case 2.1.1; d2 != 0 changes the length
+          if (e2 & 4) YBG_FABORT(10);
+          ns1 = (e2 >> 3) & 7;
+          ns2 = (e2 >> 6) & 3;
+          sp = hdr_byte(e1len + 1);
+          if (sp & 0x80) YBG_FABORT(11);
+          hl = e1len + 2;
+        } else if ((e2 & 7) == 7 && !(e2 & 40)) {
+          // general form: reuse8 set, ns1_delta and ns2_delta absent
+          uint32_t o = e1len + 1;
+          ns1 = hdr_byte(o);
+          if (ns1 & 0x80) YBG_FABORT(12);
+          ++o;
+          ns2 = 0;
+          if (e2 & 16) {
+            ns2 = hdr_byte(o);
+            if (ns2 & 0x80) YBG_FABORT(13);
+            ++o;
+          }
+          sp = hdr_byte(o);
+          if (sp & 0x80) YBG_FABORT(14);
+          hl = o + 1;
+        } else {
+          YBG_FABORT(15);  // restart mid-interval / delta forms: general path
+        }
+      }
+      const uint32_t nb = hl + ns1 + ns2;
+      if (nb > 24) YBG_FABORT(16);  // window peek limit
+      const uint64_t prev_len = key_len;
+      const uint64_t prev_except = (uint64_t)sp + ns1 + ns2 + 8;
+      if (prev_len < prev_except ||
+          (uint64_t)(limit - p) < (uint64_t)nb + value_size)
+        YBG_FABORT(17);
+      // stable length: new_len == prev_len; ns2 range stays in place
+      const uint32_t ukey = (uint32_t)prev_len - 8;
+      const uint32_t ns2s = (uint32_t)(prev_len - ns2 - 8);
+      uint64_t w3 = rdr.peek8_at(16);
+      auto body_byte = [&](uint32_t j) -> uint8_t {
+        uint64_t src = j < 8 ? w : (j < 16 ? w2 : w3);
+        return (uint8_t)(src >> (8 * (j & 7)));
+      };
+      // a ns2 range dipping below rkb implies sp < rkb (row change), so
+      // kchg is already correct; those bytes just need LDS writes too
+      kchg = sp < rkb || !row_open;
+      {
+        // changed bytes: LDS only below rkb (the restart row-compare
+        // source), register tail inside the window, nothing in between
+        uint32_t lds_n = sp < rkb ? (rkb - sp < ns1 ? rkb - sp : ns1) : 0;
+        for (uint32_t i = 0; i < lds_n; ++i)
+          key[sp + i] = body_byte(hl + i);
+        uint32_t lds2 = ns2s < rkb ? (rkb - ns2s < ns2 ? rkb - ns2s : ns2)
+                                   : 0;
+        for (uint32_t i = 0; i < lds2; ++i)
+          key[ns2s + i] = body_byte(hl + ns1 + i);
+        for (uint32_t i = 0; i < ns1; ++i)
+          tail_patch(&thi, &tlo, ukey, sp + i, body_byte(hl + i));
+        for (uint32_t i = 0; i < ns2; ++i)
+          tail_patch(&thi, &tlo, ukey, ns2s + i, body_byte(hl + ns1 + i));
+      }
+      last8 += inc;
+      rdr.consume(nb);
+      er.value = rdr.pos();
+      er.value_len = value_size;
+      q = er.value + value_size;
+    }
+    const uint32_t ukey_len = key_len - 8;
+    if (ukey_len < rkb + 2 || ukey_len < 16) YBG_FABORT(19);
+    const uint32_t ht_sz = (uint32_t)tlo & 0x1f;
+    // ht_sz >= 16: the marker byte lies outside the register tail window
+    // (LDS would be stale on this path) — general path handles it
+    if (ht_sz == 0 || ht_sz >= 16) YBG_FABORT(20);
+    const uint32_t mb = ht_sz + 1;
+    uint32_t marker;
+    if (mb <= 8) marker = (uint32_t)(tlo >> (8 * (mb - 1))) & 0xff;
+    else marker = (uint32_t)(thi >> (8 * (mb - 9))) & 0xff;
+    if (marker != kHybridTimeByte) YBG_FABORT(21);
+    // subkey entries (column updates / liveness) take the general path
+    if (ukey_len != rkb + ht_sz + 1) YBG_FABORT(22);
+    uint64_t ht_hi, ht_lo;
+    {
+      uint32_t s = 8 * (16 - ht_sz);
+      if (s == 0) { ht_hi = thi; ht_lo = tlo; }
+      else if (s < 64) {
+        ht_hi = (thi << s) | (tlo >> (64 - s));
+        ht_lo = tlo << s;
+      } else {
+        ht_hi = tlo << (s - 64);
+        ht_lo = 0;
+      }
+    }
+    // row boundary
+    if (kchg) {
+      if (row_open) {
+        if (found) {
+          bool hit = (pred_pass & sp.value_pred_mask) == sp.value_pred_mask;
+          if (in_head) {
+            ho->scanned += 1;
+            if (hit) {
+              ho->matched += 1;
+#pragma unroll
+              for (int g = 0; g < NA; ++g) {
+                if (g >= sp.num_aggs) continue;
+                bool nul = (sp.agg_op[g] != YBG_AGG_COUNT_STAR) &&
+                           ((agg_null >> g) & 1);
+                if (!nul) {
+                  combine_datum(sp.agg_op[g], &ho->val[g], &ho->cnt[g],
+                                agg_datum[g]);
+                }
+              }
+            }
+          } else {
+            s_b += 1;
+            if (hit) {
+              m_b += 1;
+#pragma unroll
+              for (int g = 0; g < NA; ++g) {
+                if (g >= sp.num_aggs) continue;
+                bool nul = (sp.agg_op[g] != YBG_AGG_COUNT_STAR) &&
+                           ((agg_null >> g) & 1);
+                if (!nul)
+                  combine_datum(sp.agg_op[g], &av_b[g], &ac_b[g],
+                                agg_datum[g]);
+              }
+            }
+          }
+        }
+        in_head = false;
+        row_open = false;
+        if (cur_iv >= j_hi) break;
+      }
+      row_open = true;
+      base_seen = false;
+      found = false;
+      pred_pass = 0;
+      agg_null = 0xffffffffu;
+    }
+    e_b += (cur_iv < j_hi);
+    walked |= (cur_iv == j_hi);
+    // visibility (no intent-prefixed values in the fast shape)
+    const uint32_t vb0 =
+        er.value_len ? (uint32_t)(rdr.peek8() & 0xff) : 0u;
+    if (YBG_UNLIKELY(vb0 == kHybridTimeByte)) YBG_FABORT(23);
+    const bool visible =
+        u128_slice_cmp(ht_hi, ht_lo, ht_sz, sp.reg_lim.hi, sp.reg_lim.lo,
+                       sp.reg_lim.len) >= 0;
+    if (visible) {
+      if (YBG_UNLIKELY(sp.track_restart)) {
+        if (u128_slice_cmp(ht_hi, ht_lo, ht_sz, sp.read.hi, sp.read.lo,
+                           sp.read.len) < 0) {
+          uint64_t* rr = bht + 3;
+          if (rr[2] == 0 || u128_slice_cmp(ht_hi, ht_lo, ht_sz, rr[0],
+                                           rr[1], (uint32_t)rr[2]) < 0) {
+            rr[0] = ht_hi;
+            rr[1] = ht_lo;
+            rr[2] = ht_sz;
+          }
+        }
+      }
+      if (!base_seen) {
+        base_seen = true;
+        if (YBG_LIKELY(er.value_len == sp.v2_fixed_len && vb0 == kPackedV2B &&
+                       (rdr.peek8() & 0xff8000u) == 0)) {
+          const uint8_t* value = rdr.pos();
+          rdr.seek(value + er.value_len);  // next window loads issue now
+          // fixed-offset column extraction (decode_packed_v2_fixed, lean
+          // eval: typed compares + aggregate capture only)
+          const uintptr_t a = (uintptr_t)value;
+          const uint64_t* qw = (const uint64_t*)(a & ~(uintptr_t)7);
+          const uint32_t abase = (uint32_t)(a & 7);
+          for (int i = 0; i < sp.num_value_cols; ++i) {
+            const uint32_t act = sp.col_act[i];
+            const uint32_t ob = abase + sp.v2_off[i];
+            const uint32_t wi = ob >> 3, sh = (ob & 7) * 8;
+            uint64_t u = qw[wi];
+            if (sh) u = (u >> sh) | (qw[wi + 1] << (64 - sh));
+            const uint32_t dt = (act >> kActDtShift) & kActDtM;
+            switch ((act >> kActV2Shift) & kActV2M) {
+              case 1:
+                u = (dt == YBG_T_INT8) ? (uint64_t)(int64_t)(int8_t)u
+                                       : (u & 0xff);
+                break;
+              case 2:
+                u = (dt == YBG_T_INT16) ? (uint64_t)(int64_t)(int16_t)u
+                                        : (u & 0xffff);
+                break;
+              case 4:
+                u = (dt == YBG_T_INT32) ? (uint64_t)(int64_t)(int32_t)u
+                                        : (u & 0xffffffffull);
+                break;
+              default:
+                break;
+            }
+            uint32_t pm = act & kActPredM;
+            while (pm) {
+              int pi = __builtin_ctz(pm);
+              pm &= pm - 1;
+              bool pass = pred_cmp_fast(sp.predc[pi], u);
+              pred_pass =
+                  (pred_pass & ~(1u << pi)) | ((uint32_t)pass << pi);
+            }
+            uint32_t am = (act >> kActAggShift) & kActAggM;
+            if (am) {
+#pragma unroll
+              for (int g = 0; g < NA; ++g) {
+                if (am & (1u << g)) {
+                  agg_datum[g] = u;
+                  agg_null &= ~(1u << g);
+                }
+              }
+            }
+          }
+          found = true;
+          p = q;
+          continue;  // reader already past the value
+        } else if (er.value_len == 1 && vb0 == kTombB) {
+          // row tombstone: row exists but is deleted (found stays false)
+        } else {
+          YBG_FABORT(24);  // control fields / V1 / null-mask V2: general path
+        }
+      }
+    }
+    rdr.seek(q);
+    p = q;
+  }
+  // final open row
+  if (row_open && found) {
+    bool hit = (pred_pass & sp.value_pred_mask) == sp.value_pred_mask;
+    if (in_head) {
+      ho->scanned += 1;
+      if (hit) {
+        ho->matched += 1;
+#pragma unroll
+        for (int g = 0; g < NA; ++g) {
+          if (g >= sp.num_aggs) continue;
+          bool nul = (sp.agg_op[g] != YBG_AGG_COUNT_STAR) &&
+                     ((agg_null >> g) & 1);
+          if (!nul)
+            combine_datum(sp.agg_op[g], &ho->val[g], &ho->cnt[g],
+                          agg_datum[g]);
+        }
+      }
+    } else {
+      s_b += 1;
+      if (hit) {
+        m_b += 1;
+#pragma unroll
+        for (int g = 0; g < NA; ++g) {
+          if (g >= sp.num_aggs) continue;
+          bool nul = (sp.agg_op[g] != YBG_AGG_COUNT_STAR) &&
+                     ((agg_null >> g) & 1);
+          if (!nul)
+            combine_datum(sp.agg_op[g], &av_b[g], &ac_b[g], agg_datum[g]);
+        }
+      }
+    }
+  }
+  *entries += e_b;
+  *scanned += s_b;
+  *matched += m_b;
+#pragma unroll
+  for (int g = 0; g < NA; ++g) {
+    if (g < sp.num_aggs)
+      combine1(sp.agg_op[g], &agg_val[g], &agg_cnt[g], av_b[g], ac_b[g]);
+  }
+  *walked_next_out = walked;
+  return 1;
 }
 
 // Scan the interval range [j, j_hi) as ONE continuous stream (j_hi = j+1

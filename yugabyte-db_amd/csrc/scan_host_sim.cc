@@ -4,6 +4,9 @@
 // continuation-flag resolution. TEST INFRASTRUCTURE: lets the CPU test suite
 // pin the device algorithm against the oracle without a GPU. Exported as
 // ybg_sim_scan from the product library (never used on the product path).
+extern "C" unsigned long long ybg_fast_abort_hist[32];
+extern "C" unsigned long long ybg_fast_abort_hist[32] = {0};
+#define YBG_FABORT(n) do { ++ybg_fast_abort_hist[n]; return 0; } while (0)
 #define YBG_HOST_SIM 1
 #define YBG_DEV_QUAL inline
 #include "scan_device.h"
@@ -122,6 +125,119 @@ int ybg_sim_scan(const ybg_scan_spec_t* spec, const uint8_t* data,
         break;
     }
   }
+  return 0;
+}
+
+// Fast-path simulator: runs scan_batch_fast per batch with the general
+// scan_one_interval as the per-batch fallback (the retry protocol the GPU
+// dispatch uses). n_fallback_out reports how many batches aborted — tests
+// assert 0 on eligible data and bit-exact results either way.
+int ybg_sim_scan_fast(const ybg_scan_spec_t* spec, const uint8_t* data,
+                      const uint64_t* offsets, uint64_t n_blocks,
+                      ybg_scan_result_t* out, uint64_t* n_fallback_out) {
+  DevSpec d;
+  std::vector<unsigned char> aux(1 << 20);
+  uint32_t aux_len = 0;
+  build_dev_spec(spec, &d, aux.data(), &aux_len, (uint32_t)aux.size());
+  aux.resize(aux_len + 16, 0);
+  if (!fast_eligible(d)) return 9;
+
+  std::vector<Interval> ivs;
+  for (uint64_t b = 0; b < n_blocks; ++b) {
+    uint64_t sz = offsets[b + 1] - offsets[b];
+    if (sz < 8) return 3;
+    const uint8_t* blk = data + offsets[b];
+    uint32_t nr = load_le32_u(blk + sz - 4);
+    if (nr == 0 || (uint64_t)nr * 4 + 4 > sz) return 3;
+    uint32_t restarts_off = (uint32_t)(sz - 4 - (uint64_t)nr * 4);
+    for (uint32_t r = 0; r < nr; ++r) {
+      uint32_t start = load_le32_u(blk + restarts_off + 4ull * r);
+      uint32_t end = (r + 1 < nr)
+                         ? load_le32_u(blk + restarts_off + 4ull * (r + 1))
+                         : restarts_off;
+      ivs.push_back(Interval{(uint32_t)b, start, end});
+    }
+  }
+  uint64_t n_ivs = ivs.size();
+  const uint64_t ivb = sim_ivb();
+  const uint64_t n_b = (n_ivs + ivb - 1) / ivb;
+  uint64_t entries = 0, scanned = 0, matched = 0, fallbacks = 0;
+  uint64_t agg_val[2] = {0, 0}, agg_cnt[2] = {0, 0};
+  std::vector<HeadOut<2>> heads(n_b);
+  std::vector<uint8_t> walked(n_b, 0);
+  alignas(8) uint8_t key[kKeyCap];
+  alignas(8) uint8_t rk_save[kKeyCap];
+  uint64_t bht[6] = {0, 0, 0, 0, 0, 0};
+  for (uint64_t b = 0; b < n_b; ++b) {
+    bool wn = false;
+    uint32_t e32 = 0, s32 = 0, m32 = 0;
+    uint64_t lo = b * ivb;
+    uint64_t hi = lo + ivb < n_ivs ? lo + ivb : n_ivs;
+    int rc = scan_batch_fast<2>(d, data, offsets, ivs.data(), n_ivs, lo, hi,
+                                key, bht, &e32, &s32, &m32, agg_val, agg_cnt,
+                                &heads[b], &wn);
+    if (!rc) {
+      ++fallbacks;
+      uint64_t av8[YBG_MAX_AGGS] = {0}, ac8[YBG_MAX_AGGS] = {0};
+      HeadOut<YBG_MAX_AGGS> ho8;
+      e32 = s32 = m32 = 0;
+      if (!scan_one_interval<YBG_MAX_AGGS>(
+              d, data, offsets, ivs.data(), n_ivs, lo, aux.data(), key,
+              rk_save, bht, &e32, &s32, &m32, av8, ac8, &ho8, &wn, nullptr,
+              nullptr, nullptr, nullptr, nullptr, nullptr, hi))
+        return 6;
+      for (int g = 0; g < 2; ++g) {
+        combine1(d.agg_op[g], &agg_val[g], &agg_cnt[g], av8[g], ac8[g]);
+        heads[b].val[g] = ho8.val[g];
+        heads[b].cnt[g] = ho8.cnt[g];
+      }
+      heads[b].scanned = ho8.scanned;
+      heads[b].matched = ho8.matched;
+    }
+    entries += e32;
+    scanned += s32;
+    matched += m32;
+    walked[b] = wn ? 1 : 0;
+  }
+  for (uint64_t b = 0; b < n_b; ++b) {
+    bool consumed = b > 0 && walked[b - 1];
+    if (consumed) continue;
+    scanned += heads[b].scanned;
+    matched += heads[b].matched;
+    for (int g = 0; g < d.num_aggs && g < 2; ++g)
+      combine1(d.agg_op[g], &agg_val[g], &agg_cnt[g], heads[b].val[g],
+               heads[b].cnt[g]);
+  }
+  memset(out, 0, sizeof(*out));
+  out->entries_seen = entries;
+  out->rows_scanned = scanned;
+  out->rows_matched = matched;
+  if (bht[5]) {
+    uint32_t n = (uint32_t)bht[5];
+    if (n > YBG_MAX_HT) n = YBG_MAX_HT;
+    for (uint32_t i = 0; i < n; ++i)
+      out->restart_ht[i] = (uint8_t)(
+          (i < 8 ? bht[3] >> (56 - 8 * i) : bht[4] >> (56 - 8 * (i - 8))) &
+          0xff);
+    out->restart_ht_len = n;
+  }
+  for (int g = 0; g < d.num_aggs; ++g) {
+    out->aggs[g].is_null = (agg_cnt[g] == 0);
+    switch (d.aggs[g].op) {
+      case YBG_AGG_SUM_DOUBLE:
+      case YBG_AGG_MIN_DOUBLE:
+      case YBG_AGG_MAX_DOUBLE: {
+        double dd;
+        memcpy(&dd, &agg_val[g], 8);
+        out->aggs[g].value_f64 = dd;
+        break;
+      }
+      default:
+        out->aggs[g].value_i64 = (int64_t)agg_val[g];
+        break;
+    }
+  }
+  if (n_fallback_out) *n_fallback_out = fallbacks;
   return 0;
 }
 
