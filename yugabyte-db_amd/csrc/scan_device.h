@@ -2258,7 +2258,7 @@ template <int NA>
 DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
                         const uint64_t* block_offsets, const Interval* ivs,
                         uint64_t n_ivs, uint64_t j_lo, uint64_t j_hi,
-                        uint8_t* key, uint64_t* bht,
+                        uint8_t* key, uint64_t* rmin,
                         uint32_t* entries, uint32_t* scanned,
                         uint32_t* matched, uint64_t* agg_val,
                         uint64_t* agg_cnt, HeadOut<NA>* ho,
@@ -2556,7 +2556,7 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
       if (YBG_UNLIKELY(sp.track_restart)) {
         if (u128_slice_cmp(ht_hi, ht_lo, ht_sz, sp.read.hi, sp.read.lo,
                            sp.read.len) < 0) {
-          uint64_t* rr = bht + 3;
+          uint64_t* rr = rmin;
           if (rr[2] == 0 || u128_slice_cmp(ht_hi, ht_lo, ht_sz, rr[0],
                                            rr[1], (uint32_t)rr[2]) < 0) {
             rr[0] = ht_hi;

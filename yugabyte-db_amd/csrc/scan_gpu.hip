@@ -59,17 +59,26 @@ __constant__ DevSpec c_spec;
 // Phase-1 scan kernel (aggregate mode): one thread per interval
 // ---------------------------------------------------------------------------
 
+// Head resolution is GLOBAL: every batch writes its deferred head record
+// (compact 2*NA+2 u64 stride) and its walked flag; k_reduce_pre folds
+// head records whose predecessor batch did not walk into them. No
+// relay, no syncthreads, and batches may be processed by ANY kernel in
+// any launch (the fast kernel + its retry pass rely on this).
+// retry_ids/retry_n, when set, make the kernel process only the listed
+// batches (the fast kernel's aborts).
 template <int NA, int WPS>
 __global__ __launch_bounds__(kThreads, WPS) void k_scan(
     const uint8_t* __restrict__ data,
     const uint64_t* __restrict__ block_offsets,
     const Interval* __restrict__ ivs, uint64_t n_ivs,
     const uint8_t* __restrict__ aux, uint8_t* __restrict__ rk_save_buf,
-    uint64_t* __restrict__ partials, uint64_t* __restrict__ heads,
-    uint32_t* __restrict__ cont_flags, int write_all_flags, uint64_t ivb) {
+    uint64_t* __restrict__ partials, uint64_t* __restrict__ bheads,
+    uint8_t* __restrict__ walked, uint32_t* __restrict__ iv_flags,
+    int write_all_flags, uint64_t ivb,
+    const uint64_t* __restrict__ retry_ids,
+    const unsigned long long* __restrict__ retry_n) {
   const DevSpec& sp = c_spec;
   __shared__ uint8_t key_scratch[kThreads * kKeyCap];
-  __shared__ uint8_t wave_relay[kThreads / 64 + 1];
   __shared__ uint64_t bht_scratch[kThreads * 6];
   uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
   uint64_t* bht = bht_scratch + (size_t)threadIdx.x * 6;
@@ -78,8 +87,9 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
   const uint64_t span = (uint64_t)gridDim.x * kThreads;
   // batch = ivb CONSECUTIVE intervals scanned as one stream by one thread
-  // (head deferral once per batch; the relay below runs on batch indices)
   const uint64_t n_batches = (n_ivs + ivb - 1) / ivb;
+  const uint64_t n_work = retry_ids ? *retry_n : n_batches;
+  constexpr int kHS = 2 * NA + 2;  // bheads stride
 
   uint32_t entries = 0, scanned = 0, matched = 0, errs = 0;
   // NA-sized and only ever constant-indexed (all loops over them unrolled):
@@ -90,75 +100,30 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
 #pragma unroll
   for (int g = 0; g < NA; ++g) { agg_val[g] = 0; agg_cnt[g] = 0; }
 
-  for (uint64_t j0 = 0; j0 < n_batches; j0 += span) {
-    const uint64_t j = j0 + gtid;
-    const bool active = j < n_batches;
+  for (uint64_t i0 = gtid; i0 < n_work; i0 += span) {
+    const uint64_t j = retry_ids ? retry_ids[i0] : i0;
     HeadOut<NA> ho;
     bool walked_next = false;
-    if (active) {
-      const uint64_t lo = j * ivb;
-      uint64_t hi = lo + ivb;
-      if (hi > n_ivs) hi = n_ivs;
-      if (!scan_one_interval<NA>(sp, data, block_offsets, ivs, n_ivs, lo,
-                                 aux, key, rk_save, bht, &entries, &scanned,
-                                 &matched, agg_val, agg_cnt, &ho,
-                                 &walked_next, nullptr, nullptr, nullptr,
-                                 nullptr, nullptr, nullptr, hi,
-                                 write_all_flags ? cont_flags : nullptr)) {
-        errs += 1;
-      }
-    } else {
-#pragma unroll
-      for (int g = 0; g < NA; ++g) { ho.val[g] = 0; ho.cnt[g] = 0; }
-      ho.scanned = ho.matched = 0;
+    const uint64_t lo = j * ivb;
+    uint64_t hi = lo + ivb;
+    if (hi > n_ivs) hi = n_ivs;
+    if (!scan_one_interval<NA>(sp, data, block_offsets, ivs, n_ivs, lo,
+                               aux, key, rk_save, bht, &entries, &scanned,
+                               &matched, agg_val, agg_cnt, &ho,
+                               &walked_next, nullptr, nullptr, nullptr,
+                               nullptr, nullptr, nullptr, hi,
+                               write_all_flags ? iv_flags : nullptr)) {
+      errs += 1;
     }
-
-    // resolve head ownership: lane l needs lane l-1's walked_next
-    unsigned lane = threadIdx.x & 63;
-    unsigned wave = threadIdx.x >> 6;
-    int wn = walked_next ? 1 : 0;
-    int from_prev_lane = __shfl_up(wn, 1);
-    if (lane == 63) wave_relay[wave + 1] = (uint8_t)wn;
-    __syncthreads();
-    bool head_consumed;
-    if (threadIdx.x == 0) {
-      head_consumed = false;  // resolved via the global record instead
-    } else if (lane == 0) {
-      head_consumed = wave_relay[wave] != 0;
-    } else {
-      head_consumed = from_prev_lane != 0;
-    }
-    __syncthreads();
-
-    if (active) {
-      if (threadIdx.x == 0) {
-        uint64_t* hr = heads + (j / kThreads) * kHeadStride;
+    uint64_t* hr = bheads + j * kHS;
 #pragma unroll
-        for (int g = 0; g < NA; ++g) {
-          hr[2 * g] = ho.val[g];
-          hr[2 * g + 1] = ho.cnt[g];
-        }
-        hr[2 * YBG_MAX_AGGS] = ho.scanned;
-        hr[2 * YBG_MAX_AGGS + 1] = ho.matched;
-      } else if (!head_consumed) {
-        scanned += ho.scanned;
-        matched += ho.matched;
-#pragma unroll
-        for (int g = 0; g < NA; ++g) {
-          if (g < sp.num_aggs)
-            combine1(sp.agg_op[g], &agg_val[g], &agg_cnt[g], ho.val[g],
-                     ho.cnt[g]);
-        }
-      }
-      // write_all_flags: the per-interval flags are written inside the
-      // walk (iv_flags); only the batch-level workgroup-boundary flag is
-      // recorded here for the head-record fold.
-      if (!write_all_flags && walked_next && ((j + 1) % kThreads) == 0 &&
-          (j + 1) < n_batches) {
-        cont_flags[(j + 1) / kThreads] = 1;
-      }
+    for (int g = 0; g < NA; ++g) {
+      hr[2 * g] = ho.val[g];
+      hr[2 * g + 1] = ho.cnt[g];
     }
-    __syncthreads();
+    hr[2 * NA] = ho.scanned;
+    hr[2 * NA + 1] = ho.matched;
+    walked[j] = walked_next ? 1 : 0;
   }
 
   // wave reduction into partials (fixed lane order => deterministic)
@@ -199,6 +164,116 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
       uint64_t mh = 0, ml = 0, mn = 0;
       for (int l = 0; l < 64; ++l) {
         const uint64_t* rr = bht_scratch + (size_t)(threadIdx.x + l) * 6 + 3;
+        if (rr[2] == 0) continue;
+        if (mn == 0 || u128_slice_cmp(rr[0], rr[1], (uint32_t)rr[2], mh, ml,
+                                      (uint32_t)mn) < 0) {
+          mh = rr[0];
+          ml = rr[1];
+          mn = rr[2];
+        }
+      }
+      partials[wave_id * kPartialStride + 4 + 2 * YBG_MAX_AGGS] = mh;
+      partials[wave_id * kPartialStride + 4 + 2 * YBG_MAX_AGGS + 1] = ml;
+      partials[wave_id * kPartialStride + 4 + 2 * YBG_MAX_AGGS + 2] = mn;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Specialized fast kernel (scan_batch_fast body): NA = 2, small LDS
+// footprint (80 B of rowkey scratch + 3 restart slots per thread), no
+// rk_save, no relay. Batches outside the fast shape are appended to the
+// retry list and re-run by k_scan in retry mode; head resolution is the
+// same global bheads/walked protocol.
+// ---------------------------------------------------------------------------
+constexpr int kFastKeyCap = 80;  // fixed rowkeys: <= 5 + 8 cols x 9
+
+template <int WPS>
+__global__ __launch_bounds__(kThreads, WPS) void k_scan_fast(
+    const uint8_t* __restrict__ data,
+    const uint64_t* __restrict__ block_offsets,
+    const Interval* __restrict__ ivs, uint64_t n_ivs,
+    uint64_t* __restrict__ partials, uint64_t* __restrict__ bheads,
+    uint8_t* __restrict__ walked, uint64_t ivb,
+    uint64_t* __restrict__ retry_ids,
+    unsigned long long* __restrict__ retry_n) {
+  const DevSpec& sp = c_spec;
+  __shared__ uint8_t key_scratch[kThreads * kFastKeyCap];
+  __shared__ uint64_t rmin_scratch[kThreads * 3];
+  uint8_t* key = key_scratch + (size_t)threadIdx.x * kFastKeyCap;
+  uint64_t* rmin = rmin_scratch + (size_t)threadIdx.x * 3;
+  rmin[2] = 0;
+  const uint32_t gtid = blockIdx.x * kThreads + threadIdx.x;
+  const uint64_t span = (uint64_t)gridDim.x * kThreads;
+  const uint64_t n_batches = (n_ivs + ivb - 1) / ivb;
+  constexpr int NA = 2;
+  constexpr int kHS = 2 * NA + 2;
+
+  uint32_t entries = 0, scanned = 0, matched = 0;
+  uint64_t agg_val[NA], agg_cnt[NA];
+#pragma unroll
+  for (int g = 0; g < NA; ++g) { agg_val[g] = 0; agg_cnt[g] = 0; }
+
+  for (uint64_t j = gtid; j < n_batches; j += span) {
+    HeadOut<NA> ho;
+    bool walked_next = false;
+    const uint64_t lo = j * ivb;
+    uint64_t hi = lo + ivb;
+    if (hi > n_ivs) hi = n_ivs;
+    if (!scan_batch_fast<NA>(sp, data, block_offsets, ivs, n_ivs, lo, hi,
+                             key, rmin, &entries, &scanned, &matched,
+                             agg_val, agg_cnt, &ho, &walked_next)) {
+      unsigned long long slot = atomicAdd(retry_n, 1ull);
+      retry_ids[slot] = j;
+      continue;
+    }
+    uint64_t* hr = bheads + j * kHS;
+#pragma unroll
+    for (int g = 0; g < NA; ++g) {
+      hr[2 * g] = ho.val[g];
+      hr[2 * g + 1] = ho.cnt[g];
+    }
+    hr[2 * NA] = ho.scanned;
+    hr[2 * NA + 1] = ho.matched;
+    walked[j] = walked_next ? 1 : 0;
+  }
+
+  // wave reduction into partials (fixed lane order => deterministic)
+  {
+    unsigned lane = threadIdx.x & 63;
+    uint32_t errs = 0;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      entries += __shfl_down(entries, off);
+      scanned += __shfl_down(scanned, off);
+      matched += __shfl_down(matched, off);
+    }
+    __shared__ uint64_t red_val[kThreads], red_cnt[kThreads];
+    uint64_t wave_id = ((uint64_t)blockIdx.x * kThreads + threadIdx.x) >> 6;
+#pragma unroll
+    for (int g = 0; g < NA; ++g) {
+      if (g >= sp.num_aggs) continue;
+      red_val[threadIdx.x] = agg_val[g];
+      red_cnt[threadIdx.x] = agg_cnt[g];
+      __syncthreads();
+      if (lane == 0) {
+        uint64_t av = 0, ac = 0;
+        for (int l = 0; l < 64; ++l)
+          combine1(sp.agg_op[g], &av, &ac, red_val[threadIdx.x + l],
+                   red_cnt[threadIdx.x + l]);
+        partials[wave_id * kPartialStride + 4 + 2 * g] = av;
+        partials[wave_id * kPartialStride + 4 + 2 * g + 1] = ac;
+      }
+      __syncthreads();
+    }
+    if (lane == 0) {
+      partials[wave_id * kPartialStride + 0] = entries;
+      partials[wave_id * kPartialStride + 1] = scanned;
+      partials[wave_id * kPartialStride + 2] = matched;
+      partials[wave_id * kPartialStride + 3] = errs;
+      uint64_t mh = 0, ml = 0, mn = 0;
+      for (int l = 0; l < 64; ++l) {
+        const uint64_t* rr = rmin_scratch + (size_t)(threadIdx.x + l) * 3;
         if (rr[2] == 0) continue;
         if (mn == 0 || u128_slice_cmp(rr[0], rr[1], (uint32_t)rr[2], mh, ml,
                                       (uint32_t)mn) < 0) {
@@ -460,19 +535,21 @@ struct DevResult {
 // partial-layout record; k_reduce then folds the 256 chunk records in fixed
 // order. The fold tree is fixed => results stay deterministic run-to-run;
 // the single-workgroup k_reduce no longer reads megabytes through one CU.
+// hstride = the bheads per-batch record stride (2 * dispatch-NA + 2);
+// a batch's head record folds in iff its predecessor did not walk into it.
 __global__ __launch_bounds__(256) void k_reduce_pre(
     DevSpec sp, const uint64_t* __restrict__ partials, uint64_t n_partials,
-    const uint64_t* __restrict__ heads, const uint32_t* __restrict__ cont_flags,
-    uint64_t n_heads, uint64_t* __restrict__ chunk_out) {
+    const uint64_t* __restrict__ bheads, const uint8_t* __restrict__ walked,
+    uint64_t n_batches, int hstride, uint64_t* __restrict__ chunk_out) {
   __shared__ uint64_t sval[256], scnt[256], scal[256];
   const unsigned t = threadIdx.x;
   const uint64_t nch = gridDim.x;
   const uint64_t pl = (n_partials + nch - 1) / nch;
   const uint64_t plo = blockIdx.x * pl;
   const uint64_t phi = plo + pl < n_partials ? plo + pl : n_partials;
-  const uint64_t hl = (n_heads + nch - 1) / nch;
+  const uint64_t hl = (n_batches + nch - 1) / nch;
   const uint64_t hlo = blockIdx.x * hl;
-  const uint64_t hhi = hlo + hl < n_heads ? hlo + hl : n_heads;
+  const uint64_t hhi = hlo + hl < n_batches ? hlo + hl : n_batches;
 
   for (int s = 0; s < 4; ++s) {
     uint64_t acc = 0;
@@ -480,8 +557,8 @@ __global__ __launch_bounds__(256) void k_reduce_pre(
       acc += partials[i * kPartialStride + s];
     if (s == 1 || s == 2) {
       for (uint64_t i = hlo + t; i < hhi; i += 256)
-        if (cont_flags[i] == 0)
-          acc += heads[i * kHeadStride + 2 * YBG_MAX_AGGS + (s - 1)];
+        if (i == 0 || walked[i - 1] == 0)
+          acc += bheads[i * hstride + hstride - 2 + (s - 1)];
     }
     scal[t] = acc;
     __syncthreads();
@@ -499,9 +576,9 @@ __global__ __launch_bounds__(256) void k_reduce_pre(
       combine1(op, &av, &ac, partials[i * kPartialStride + 4 + 2 * g],
                partials[i * kPartialStride + 4 + 2 * g + 1]);
     for (uint64_t i = hlo + t; i < hhi; i += 256) {
-      if (cont_flags[i] != 0) continue;
-      combine1(op, &av, &ac, heads[i * kHeadStride + 2 * g],
-               heads[i * kHeadStride + 2 * g + 1]);
+      if (i > 0 && walked[i - 1] != 0) continue;
+      combine1(op, &av, &ac, bheads[i * hstride + 2 * g],
+               bheads[i * hstride + 2 * g + 1]);
     }
     sval[t] = av;
     scnt[t] = ac;
@@ -713,11 +790,17 @@ struct ybg_scan {
   uint64_t total_bytes = 0;
   uint8_t* d_aux = nullptr;
   uint8_t* d_rk_save = nullptr;
-  uint64_t* d_partials = nullptr;
+  uint64_t* d_partials = nullptr;  // 2 halves: main launch + retry launch
   uint64_t n_partials = 0;
-  uint64_t* d_heads = nullptr;
-  uint32_t* d_cont = nullptr;
-  uint64_t n_heads = 0;
+  uint64_t* d_heads = nullptr;   // per-batch head records (stride hstride)
+  uint8_t* d_walked = nullptr;   // per-batch walked-into-next flag
+  uint64_t* d_retry = nullptr;   // fast-kernel aborted batch ids
+  unsigned long long* d_retry_n = nullptr;
+  uint32_t* d_cont = nullptr;    // k_group relay flags (per 256 batches)
+  uint64_t n_heads = 0;          // = n_batches
+  uint64_t n_gheads = 0;         // = ceil(n_batches / kThreads)
+  int na_cap = 2;
+  int hstride = 6;
   DevResult* d_result = nullptr;
   uint64_t* d_chunk = nullptr;  // k_reduce_pre output (256 partial records)
   int grid = 0;
@@ -919,12 +1002,22 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
   if (s->grid < 1) s->grid = 1;
   uint64_t span_threads = (uint64_t)s->grid * kThreads;
   s->n_partials = span_threads / 64;
-  s->n_heads = (s->n_batches + kThreads - 1) / kThreads;
+  s->n_heads = s->n_batches;  // per-batch head records (global resolution)
+  s->n_gheads = (s->n_batches + kThreads - 1) / kThreads;
+  s->na_cap = s->dspec.num_aggs <= 2 ? 2 : (s->dspec.num_aggs <= 4 ? 4 : 8);
+  s->hstride = 2 * s->na_cap + 2;
   HIP_TRY(hipMalloc(&s->d_rk_save, span_threads * kKeyCap));
   HIP_TRY(hipMalloc(&s->d_partials,
-                    s->n_partials * kPartialStride * sizeof(uint64_t)));
-  HIP_TRY(hipMalloc(&s->d_heads, s->n_heads * kHeadStride * sizeof(uint64_t)));
-  HIP_TRY(hipMalloc(&s->d_cont, s->n_heads * sizeof(uint32_t)));
+                    2 * s->n_partials * kPartialStride * sizeof(uint64_t)));
+  HIP_TRY(hipMalloc(&s->d_heads,
+                    s->n_heads * s->hstride * sizeof(uint64_t)));
+  HIP_TRY(hipMalloc(&s->d_walked, s->n_heads ? s->n_heads : 1));
+  HIP_TRY(hipMalloc(&s->d_retry,
+                    (s->n_heads ? s->n_heads : 1) * sizeof(uint64_t)));
+  HIP_TRY(hipMalloc(&s->d_retry_n, sizeof(unsigned long long)));
+  HIP_TRY(hipMalloc(&s->d_cont,
+                    ((s->n_batches + kThreads - 1) / kThreads) *
+                        sizeof(uint32_t)));
   HIP_TRY(hipMalloc(&s->d_result, sizeof(DevResult)));
   HIP_TRY(hipMalloc(&s->d_chunk, 256 * kPartialStride * sizeof(uint64_t)));
   return 0;
@@ -1055,16 +1148,15 @@ int yb_gpu_scan_feed_sst(ybg_scan_t* s, const uint8_t* file, uint64_t size,
 
 int yb_gpu_scan_execute(ybg_scan_t* s) {
   if (!s->d_data) return set_err(4, "feed_blocks not called");
-  HIP_TRY(hipMemsetAsync(s->d_cont, 0, s->n_heads * sizeof(uint32_t),
-                         s->stream));
-  HIP_TRY(hipMemsetAsync(s->d_heads, 0,
-                         s->n_heads * kHeadStride * sizeof(uint64_t),
-                         s->stream));
+  // the retry half of the partials is only partially covered by the retry
+  // launch (or not at all on the general dispatch): zero it so the fold
+  // sees clean records
+  HIP_TRY(hipMemsetAsync(
+      s->d_partials + s->n_partials * kPartialStride, 0,
+      s->n_partials * kPartialStride * sizeof(uint64_t), s->stream));
   HIP_TRY(hipEventRecord(s->ev_start, s->stream));
   // dispatch on aggregate-slot capacity (register footprint) and the
-  // waves-per-SIMD occupancy bound (YBG_WPS for tuning, default 3: after the
-  // accumulator register promotion, the WPS=4 build only fits 4 waves/SIMD (default 5: best in the post-col_act sweep; the bound compiles to 3 waves)
-  // by spilling VGPRs in the hot loop and measures ~30% slower)
+  // waves-per-SIMD occupancy bound (YBG_WPS for tuning; default 3 measured)
   int wps = 3;
   if (const char* e = getenv("YBG_WPS")) {
     long v = atol(e);
@@ -1073,34 +1165,71 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
   HIP_TRY(hipMemcpyToSymbolAsync(HIP_SYMBOL(c_spec), &s->dspec,
                                  sizeof(DevSpec), 0, hipMemcpyHostToDevice,
                                  s->stream));
-  auto launch = [&](auto kern) {
-    hipLaunchKernelGGL(kern, dim3(s->grid), dim3(kThreads), 0, s->stream,
+  int na = s->na_cap;
+  bool usefast = fast_eligible(s->dspec) && na == 2;
+  if (const char* e = getenv("YBG_FAST")) usefast = usefast && atoi(e) != 0;
+  if (usefast) {
+    HIP_TRY(hipMemsetAsync(s->d_retry_n, 0, sizeof(unsigned long long),
+                           s->stream));
+    auto launchf = [&](auto kern) {
+      hipLaunchKernelGGL(kern, dim3(s->grid), dim3(kThreads), 0, s->stream,
+                         s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
+                         s->d_partials, s->d_heads, s->d_walked, s->ivb,
+                         s->d_retry, s->d_retry_n);
+    };
+    int fwps = wps;
+    if (const char* e = getenv("YBG_FWPS")) {
+      long v = atol(e);
+      if (v >= 2 && v <= 8) fwps = (int)v;
+    }
+    switch (fwps) {
+      case 2: launchf(k_scan_fast<2>); break;
+      case 4: launchf(k_scan_fast<4>); break;
+      case 5: launchf(k_scan_fast<5>); break;
+      case 6: launchf(k_scan_fast<6>); break;
+      default: launchf(k_scan_fast<3>); break;
+    }
+    // retry pass: the general kernel over the aborted batch list, into
+    // the second partials half
+    int rg = std::min(s->grid, 1024);
+    auto retry_kernel = k_scan<2, 3>;
+    hipLaunchKernelGGL(retry_kernel, dim3(rg), dim3(kThreads), 0, s->stream,
                        s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
-                       s->d_aux, s->d_rk_save, s->d_partials, s->d_heads,
-                       s->d_cont, 0, s->ivb);
-  };
-  int na = s->dspec.num_aggs <= 2 ? 2 : (s->dspec.num_aggs <= 4 ? 4 : 8);
-  switch (na * 10 + wps) {
-    case 22: launch(k_scan<2, 2>); break;
-    case 23: launch(k_scan<2, 3>); break;
-    case 25: launch(k_scan<2, 5>); break;
-    case 26: launch(k_scan<2, 6>); break;
-    case 42: launch(k_scan<4, 2>); break;
-    case 43: launch(k_scan<4, 3>); break;
-    case 45: launch(k_scan<4, 5>); break;
-    case 46: launch(k_scan<4, 6>); break;
-    case 82: launch(k_scan<8, 2>); break;
-    case 83: launch(k_scan<8, 3>); break;
-    case 85: launch(k_scan<8, 5>); break;
-    case 86: launch(k_scan<8, 6>); break;
-    case 44: launch(k_scan<4, 4>); break;
-    case 84: launch(k_scan<8, 4>); break;
-    default: launch(k_scan<2, 4>); break;
+                       s->d_aux, s->d_rk_save,
+                       s->d_partials + s->n_partials * kPartialStride,
+                       s->d_heads, s->d_walked, nullptr, 0, s->ivb,
+                       s->d_retry, s->d_retry_n);
+  } else {
+    auto launch = [&](auto kern) {
+      hipLaunchKernelGGL(kern, dim3(s->grid), dim3(kThreads), 0, s->stream,
+                         s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
+                         s->d_aux, s->d_rk_save, s->d_partials, s->d_heads,
+                         s->d_walked, nullptr, 0, s->ivb,
+                         (const uint64_t*)nullptr,
+                         (const unsigned long long*)nullptr);
+    };
+    switch (na * 10 + wps) {
+      case 22: launch(k_scan<2, 2>); break;
+      case 23: launch(k_scan<2, 3>); break;
+      case 25: launch(k_scan<2, 5>); break;
+      case 26: launch(k_scan<2, 6>); break;
+      case 42: launch(k_scan<4, 2>); break;
+      case 43: launch(k_scan<4, 3>); break;
+      case 45: launch(k_scan<4, 5>); break;
+      case 46: launch(k_scan<4, 6>); break;
+      case 82: launch(k_scan<8, 2>); break;
+      case 83: launch(k_scan<8, 3>); break;
+      case 85: launch(k_scan<8, 5>); break;
+      case 86: launch(k_scan<8, 6>); break;
+      case 44: launch(k_scan<4, 4>); break;
+      case 84: launch(k_scan<8, 4>); break;
+      default: launch(k_scan<2, 4>); break;
+    }
   }
   HIP_TRY(hipEventRecord(s->ev_mid, s->stream));
   hipLaunchKernelGGL(k_reduce_pre, dim3(256), dim3(256), 0, s->stream,
-                     s->dspec, s->d_partials, s->n_partials, s->d_heads,
-                     s->d_cont, s->n_heads, s->d_chunk);
+                     s->dspec, s->d_partials, 2 * s->n_partials, s->d_heads,
+                     s->d_walked, s->n_heads, s->hstride, s->d_chunk);
   hipLaunchKernelGGL(k_reduce, dim3(1), dim3(256), 0, s->stream, s->dspec,
                      s->d_chunk, 256, s->d_heads, s->d_cont, 0,
                      s->d_result);
@@ -1180,10 +1309,9 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
                          s->stream));
   HIP_TRY(hipMemsetAsync(s->d_em_counters, 0, 4 * sizeof(unsigned long long),
                          s->stream));
-  HIP_TRY(hipMemsetAsync(s->d_heads, 0,
-                         s->n_heads * kHeadStride * sizeof(uint64_t),
-                         s->stream));
   // flags pre-pass: resolves head-row ownership for EVERY interval
+  // (iv_flags written inside the walk; the bheads/walked side effects of
+  // this pass are scratch — nothing folds them)
   HIP_TRY(hipMemcpyToSymbolAsync(HIP_SYMBOL(c_spec), &s->dspec,
                                  sizeof(DevSpec), 0, hipMemcpyHostToDevice,
                                  s->stream));
@@ -1191,7 +1319,9 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
   hipLaunchKernelGGL(flags_kernel, dim3(s->grid), dim3(kThreads), 0,
                      s->stream, s->d_data, s->d_offsets, s->d_ivs,
                      s->n_ivs, s->d_aux, s->d_rk_save, s->d_partials,
-                     s->d_heads, s->d_flags_all, 1, s->ivb);
+                     s->d_heads, s->d_walked, s->d_flags_all, 1, s->ivb,
+                     (const uint64_t*)nullptr,
+                     (const unsigned long long*)nullptr);
   EmitCtx ec;
   ec.sort_key = s->d_em_sort;
   ec.key_datums = s->d_em_key;
@@ -1289,13 +1419,13 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
     HIP_TRY(hipMalloc(&s->d_g_counters, 3 * sizeof(unsigned long long)));
     s->gc.cap = s->group_cap;
     s->gc.data = s->d_data;
-    HIP_TRY(hipMalloc(&s->d_gheads, s->n_heads * sizeof(GroupHead)));
+    HIP_TRY(hipMalloc(&s->d_gheads, s->n_gheads * sizeof(GroupHead)));
   }
   HIP_TRY(hipMemsetAsync(s->gc.overflow, 0, 8, s->stream));
   HIP_TRY(hipMemsetAsync(s->d_g_counters, 0, 24, s->stream));
-  HIP_TRY(hipMemsetAsync(s->d_gheads, 0, s->n_heads * sizeof(GroupHead),
+  HIP_TRY(hipMemsetAsync(s->d_gheads, 0, s->n_gheads * sizeof(GroupHead),
                          s->stream));
-  HIP_TRY(hipMemsetAsync(s->d_cont, 0, s->n_heads * sizeof(uint32_t),
+  HIP_TRY(hipMemsetAsync(s->d_cont, 0, s->n_gheads * sizeof(uint32_t),
                          s->stream));
   hipLaunchKernelGGL(k_group_init, dim3(512), dim3(256), 0, s->stream,
                      s->dspec, s->gc);
@@ -1318,14 +1448,16 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
                        s->n_ivs, s->d_aux, s->d_rk_save, s->gc, s->d_gheads,
                        s->d_cont, s->gc.overflow, s->d_partials, s->ivb);
     hipLaunchKernelGGL(k_group_heads<4>, dim3(hgrid), dim3(kThreads), 0,
-                       s->stream, s->gc, s->d_gheads, s->d_cont, s->n_heads);
+                       s->stream, s->gc, s->d_gheads, s->d_cont,
+                       s->n_gheads);
   } else {
     hipLaunchKernelGGL(k_group<8>, dim3(s->grid), dim3(kThreads), 0,
                        s->stream, s->d_data, s->d_offsets, s->d_ivs,
                        s->n_ivs, s->d_aux, s->d_rk_save, s->gc, s->d_gheads,
                        s->d_cont, s->gc.overflow, s->d_partials, s->ivb);
     hipLaunchKernelGGL(k_group_heads<8>, dim3(hgrid), dim3(kThreads), 0,
-                       s->stream, s->gc, s->d_gheads, s->d_cont, s->n_heads);
+                       s->stream, s->gc, s->d_gheads, s->d_cont,
+                       s->n_gheads);
   }
   if (s->dspec.track_restart) {
     // fold the wave restart minima into DevResult (num_aggs = 0 spec: only

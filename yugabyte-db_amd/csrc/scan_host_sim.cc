@@ -174,8 +174,8 @@ int ybg_sim_scan_fast(const ybg_scan_spec_t* spec, const uint8_t* data,
     uint64_t lo = b * ivb;
     uint64_t hi = lo + ivb < n_ivs ? lo + ivb : n_ivs;
     int rc = scan_batch_fast<2>(d, data, offsets, ivs.data(), n_ivs, lo, hi,
-                                key, bht, &e32, &s32, &m32, agg_val, agg_cnt,
-                                &heads[b], &wn);
+                                key, bht + 3, &e32, &s32, &m32, agg_val,
+                                agg_cnt, &heads[b], &wn);
     if (!rc) {
       ++fallbacks;
       uint64_t av8[YBG_MAX_AGGS] = {0}, ac8[YBG_MAX_AGGS] = {0};
