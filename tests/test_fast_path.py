@@ -215,3 +215,31 @@ def test_fast_fuzz_vs_general():
             ref = _res(y.sim_scan(spec, built[0], built[1], built[2]))
             fast, nf = y.sim_scan_fast(spec, built[0], built[1], built[2])
             assert _res(fast) == ref, (it, read, local)
+
+
+def test_fast_deep_window_dht_tails():
+    """Regression: entries with header+ns1+ns2 > 16 bytes peek up to 24
+    bytes into the reader window; without k-normalization (Rdr::align8)
+    the deep peeks silently wrapped and corrupted the patched DocHybridTime
+    tail — invisible at far-future read times, visible at mid-sweep reads.
+    Wide values (4 int64 cols -> 2-byte e1) force the deep form at every
+    window alignment."""
+    schema = y.make_schema([y.KT_INT64],
+                           [(10 + i, y.T_INT64, 1) for i in range(4)])
+    b = y.Builder(schema)
+    seq = 1 << 50
+    for r in range(6000):
+        for ht in (3000, 2000, 1000):
+            seq += 1
+            b.add_packed_row(ht, [(y.T_INT64, r * 4 + i) for i in range(4)],
+                             hash_=r // 512, key_datums=(r,), seq=seq)
+    built = b.finish()
+    for read, local in ((2500, 3500), (1500, 1500), (900, 2600)):
+        spec = _spec(schema, read, local, local + 1000,
+                     preds=[y.Pred(0, 2, y.PRED_GE, 0, None, 0)],
+                     aggs=[y.Agg(y.AGG_COUNT_STAR, 0),
+                           y.Agg(y.AGG_SUM_INT64, 3)])
+        ref = _res(y.sim_scan(spec, built[0], built[1], built[2]))
+        fast, nf = y.sim_scan_fast(spec, built[0], built[1], built[2])
+        assert _res(fast) == ref, (read, local)
+        assert nf == 0

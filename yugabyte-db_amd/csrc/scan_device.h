@@ -481,7 +481,21 @@ struct Rdr {
     uint32_t sh = 8 * (kk & 7);
     return sh ? ((lo >> sh) | (hi << (64 - sh))) : lo;
   }
-  // 8 bytes at position + off (off + k must be < 25)
+  // Slide the window so k < 8 (one aligned load): callers that peek deep
+  // (peek8_at(16), body bytes up to 24 ahead) must run this first — the
+  // window holds [base, base+32) and deep peeks at large k silently wrap.
+  DEV void align8() {
+    if (k >= 8) {
+      q0 = q1;
+      q1 = q2;
+      q2 = q3;
+      base += 8;
+      q3 = ((const uint64_t*)base)[3];
+      k -= 8;
+    }
+  }
+  // 8 bytes at position + off (k + off + 7 must be < 32, i.e. call
+  // align8() first when off can reach 16)
   DEV uint64_t peek8_at(uint32_t off) const {
     uint32_t kk = k + off;
     uint64_t m1 = (uint64_t)0 - (uint64_t)(kk >= 8);
@@ -2378,6 +2392,7 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
       // registers; bytes between rkb and the tail window need no store at
       // all (nothing reads them before the next restart overwrites).
       if (limit - p < 8) YBG_FABORT(7);
+      rdr.align8();  // body bytes peek up to 24 ahead of the position
       uint64_t w = rdr.peek8();
       uint32_t b0 = (uint32_t)(w & 0xff);
       uint64_t e1;
