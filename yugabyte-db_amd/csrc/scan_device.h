@@ -2184,6 +2184,47 @@ DEV void emit_row(const DevSpec& sp, EmitCtx* ec, const RowCtxT<NA>& rc,
 // Semantics are pinned against scan_one_interval by sim + GPU parity tests.
 // ---------------------------------------------------------------------------
 
+DEV void u128_shr(uint64_t* hi, uint64_t* lo, uint32_t s) {
+  if (s >= 64) {
+    *lo = s >= 128 ? 0 : (*hi >> (s - 64));
+    *hi = 0;
+  } else if (s) {
+    *lo = (*lo >> s) | (*hi << (64 - s));
+    *hi >>= s;
+  }
+}
+
+DEV void u128_shl(uint64_t* hi, uint64_t* lo, uint32_t s) {
+  if (s >= 64) {
+    *hi = s >= 128 ? 0 : (*lo << (s - 64));
+    *lo = 0;
+  } else if (s) {
+    *hi = (*hi << s) | (*lo >> (64 - s));
+    *lo <<= s;
+  }
+}
+
+// Splice n (1..16) big-endian bytes (sv_hi:sv_lo, first byte at sv_hi's
+// MSB) into the 16-byte register tail at SIGNED tail byte offset o
+// (o = key_pos - (ukey_len - 16)); bytes falling outside [0,16) drop out.
+// Replaces a per-byte tail_patch loop (~18 instructions per byte) with
+// ~30 instructions per RANGE.
+DEV void tail_splice(uint64_t* thi, uint64_t* tlo, int32_t o, uint32_t n,
+                     uint64_t sv_hi, uint64_t sv_lo) {
+  if (o <= -(int32_t)n || o >= 16) return;
+  uint64_t mh = n >= 8 ? ~0ull : (~0ull << (64 - 8 * n));
+  uint64_t ml = n <= 8 ? 0ull : (n >= 16 ? ~0ull : (~0ull << (128 - 8 * n)));
+  if (o >= 0) {
+    u128_shr(&sv_hi, &sv_lo, 8 * (uint32_t)o);
+    u128_shr(&mh, &ml, 8 * (uint32_t)o);
+  } else {
+    u128_shl(&sv_hi, &sv_lo, 8 * (uint32_t)(-o));
+    u128_shl(&mh, &ml, 8 * (uint32_t)(-o));
+  }
+  *thi = (*thi & ~mh) | (sv_hi & mh);
+  *tlo = (*tlo & ~ml) | (sv_lo & ml);
+}
+
 // Accumulate ONE row's aggregate operand (acc_row's per-slot body).
 DEV void combine_datum(int op, uint64_t* val, uint64_t* cnt, uint64_t v) {
   switch (op) {
@@ -2449,7 +2490,8 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
         }
       }
       const uint32_t nb = hl + ns1 + ns2;
-      if (nb > 24) YBG_FABORT(16);  // window peek limit
+      // splice-source widths: ns1 from a 16-byte load pair, ns2 from one
+      if (nb > 24 || ns1 > 16 || ns2 > 8) YBG_FABORT(16);
       const uint64_t prev_len = key_len;
       const uint64_t prev_except = (uint64_t)sp + ns1 + ns2 + 8;
       if (prev_len < prev_except ||
@@ -2458,28 +2500,31 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
       // stable length: new_len == prev_len; ns2 range stays in place
       const uint32_t ukey = (uint32_t)prev_len - 8;
       const uint32_t ns2s = (uint32_t)(prev_len - ns2 - 8);
-      uint64_t w3 = rdr.peek8_at(16);
-      auto body_byte = [&](uint32_t j) -> uint8_t {
-        uint64_t src = j < 8 ? w : (j < 16 ? w2 : w3);
-        return (uint8_t)(src >> (8 * (j & 7)));
-      };
       // a ns2 range dipping below rkb implies sp < rkb (row change), so
       // kchg is already correct; those bytes just need LDS writes too
       kchg = sp < rkb || !row_open;
       {
         // changed bytes: LDS only below rkb (the restart row-compare
-        // source), register tail inside the window, nothing in between
+        // source), register tail inside the window, nothing in between.
+        // Splice sources come straight from the entry bytes (L1-hot, the
+        // window just streamed them) — no deep window peeks.
         uint32_t lds_n = sp < rkb ? (rkb - sp < ns1 ? rkb - sp : ns1) : 0;
         for (uint32_t i = 0; i < lds_n; ++i)
-          key[sp + i] = body_byte(hl + i);
+          key[sp + i] = p[hl + i];
         uint32_t lds2 = ns2s < rkb ? (rkb - ns2s < ns2 ? rkb - ns2s : ns2)
                                    : 0;
         for (uint32_t i = 0; i < lds2; ++i)
-          key[ns2s + i] = body_byte(hl + ns1 + i);
-        for (uint32_t i = 0; i < ns1; ++i)
-          tail_patch(&thi, &tlo, ukey, sp + i, body_byte(hl + i));
-        for (uint32_t i = 0; i < ns2; ++i)
-          tail_patch(&thi, &tlo, ukey, ns2s + i, body_byte(hl + ns1 + i));
+          key[ns2s + i] = p[hl + ns1 + i];
+        // register-tail update: one range splice per changed range
+        const int32_t T = (int32_t)ukey - 16;
+        uint64_t s1h = __builtin_bswap64(load_u64_una(p + hl));
+        uint64_t s1l =
+            ns1 > 8 ? __builtin_bswap64(load_u64_una(p + hl + 8)) : 0;
+        tail_splice(&thi, &tlo, (int32_t)sp - T, ns1, s1h, s1l);
+        if (ns2) {
+          uint64_t s2h = __builtin_bswap64(load_u64_una(p + hl + ns1));
+          tail_splice(&thi, &tlo, (int32_t)ns2s - T, ns2, s2h, 0);
+        }
       }
       last8 += inc;
       rdr.consume(nb);
