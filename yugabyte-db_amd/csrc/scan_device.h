@@ -13,6 +13,11 @@
 #endif
 #define DEV YBG_DEV_QUAL
 
+// tail-update strategy for the fast scanner (see scan_batch_fast)
+#ifndef YBG_TAILMODE
+#define YBG_TAILMODE 1
+#endif
+
 #ifdef YBG_HOST_SIM
 // host shims for the HIP bit-cast intrinsics
 static inline double __longlong_as_double(long long v) {
@@ -2506,8 +2511,31 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
       {
         // changed bytes: LDS only below rkb (the restart row-compare
         // source), register tail inside the window, nothing in between.
-        // Splice sources come straight from the entry bytes (L1-hot, the
-        // window just streamed them) — no deep window peeks.
+        // YBG_TAILMODE selects the tail-update strategy (A/B-measured):
+        //   0 per-byte tail_patch from window bytes (needs align8 + w3)
+        //   1 range splice sourced by direct loads from the entry bytes
+        //   2 range splice sourced by window peeks
+#if YBG_TAILMODE == 0
+        rdr.align8();
+        w = rdr.peek8();
+        w2 = rdr.peek8_at(8);
+        uint64_t w3 = rdr.peek8_at(16);
+        auto body_byte = [&](uint32_t j) -> uint8_t {
+          uint64_t src = j < 8 ? w : (j < 16 ? w2 : w3);
+          return (uint8_t)(src >> (8 * (j & 7)));
+        };
+        uint32_t lds_n = sp < rkb ? (rkb - sp < ns1 ? rkb - sp : ns1) : 0;
+        for (uint32_t i = 0; i < lds_n; ++i)
+          key[sp + i] = body_byte(hl + i);
+        uint32_t lds2 = ns2s < rkb ? (rkb - ns2s < ns2 ? rkb - ns2s : ns2)
+                                   : 0;
+        for (uint32_t i = 0; i < lds2; ++i)
+          key[ns2s + i] = body_byte(hl + ns1 + i);
+        for (uint32_t i = 0; i < ns1; ++i)
+          tail_patch(&thi, &tlo, ukey, sp + i, body_byte(hl + i));
+        for (uint32_t i = 0; i < ns2; ++i)
+          tail_patch(&thi, &tlo, ukey, ns2s + i, body_byte(hl + ns1 + i));
+#else
         uint32_t lds_n = sp < rkb ? (rkb - sp < ns1 ? rkb - sp : ns1) : 0;
         for (uint32_t i = 0; i < lds_n; ++i)
           key[sp + i] = p[hl + i];
@@ -2515,16 +2543,27 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
                                    : 0;
         for (uint32_t i = 0; i < lds2; ++i)
           key[ns2s + i] = p[hl + ns1 + i];
-        // register-tail update: one range splice per changed range
         const int32_t T = (int32_t)ukey - 16;
+#if YBG_TAILMODE == 2
+        rdr.align8();
+        uint64_t s1h = __builtin_bswap64(rdr.peek8_at(hl));
+        uint64_t s1l =
+            ns1 > 8 ? __builtin_bswap64(rdr.peek8_at(hl + 8)) : 0;
+        uint64_t s2h =
+            ns2 ? __builtin_bswap64(hl + ns1 <= 17
+                                        ? rdr.peek8_at(hl + ns1)
+                                        : load_u64_una(p + hl + ns1))
+                : 0;
+#else
         uint64_t s1h = __builtin_bswap64(load_u64_una(p + hl));
         uint64_t s1l =
             ns1 > 8 ? __builtin_bswap64(load_u64_una(p + hl + 8)) : 0;
+        uint64_t s2h =
+            ns2 ? __builtin_bswap64(load_u64_una(p + hl + ns1)) : 0;
+#endif
         tail_splice(&thi, &tlo, (int32_t)sp - T, ns1, s1h, s1l);
-        if (ns2) {
-          uint64_t s2h = __builtin_bswap64(load_u64_una(p + hl + ns1));
-          tail_splice(&thi, &tlo, (int32_t)ns2s - T, ns2, s2h, 0);
-        }
+        if (ns2) tail_splice(&thi, &tlo, (int32_t)ns2s - T, ns2, s2h, 0);
+#endif
       }
       last8 += inc;
       rdr.consume(nb);
