@@ -15,7 +15,7 @@
 
 // tail-update strategy for the fast scanner (see scan_batch_fast)
 #ifndef YBG_TAILMODE
-#define YBG_TAILMODE 1
+#define YBG_TAILMODE 0
 #endif
 
 #ifdef YBG_HOST_SIM
@@ -524,6 +524,15 @@ struct Rdr {
       const uint64_t* q = (const uint64_t*)base;
       q2 = q[2];
       q3 = q[3];
+#if defined(YBG_PF) && defined(__HIP_DEVICE_COMPILE__)
+      // sliding L2 prefetch: one extra independent load per 64 consumed
+      // bytes, ~192 B ahead of the stream (device buffers carry 256 B of
+      // tail slack for this). The asm keeps the otherwise-dead load.
+      if (((uintptr_t)base & 63) == 0) {
+        uint64_t pf = q[24];
+        asm volatile("" ::"v"(pf));
+      }
+#endif
       k -= 16;
     }
   }

@@ -929,10 +929,11 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
     s->d_data = const_cast<uint8_t*>(blocks);
     s->d_data_owned = false;
   } else {
-    // 48 bytes tail slack: Rdr window init can read up to 32 bytes past
-    // the position and load_u64_una up to 11 (scan_device.h contract)
-    HIP_TRY(hipMalloc(&s->d_data, s->total_bytes + 48));
-    HIP_TRY(hipMemset(s->d_data + s->total_bytes, 0, 48));
+    // 256 bytes tail slack: Rdr window init can read up to 32 bytes past
+    // the position, load_u64_una up to 11, and the YBG_PF prefetch build
+    // up to ~200 (scan_device.h contract)
+    HIP_TRY(hipMalloc(&s->d_data, s->total_bytes + 256));
+    HIP_TRY(hipMemset(s->d_data + s->total_bytes, 0, 256));
     HIP_TRY(hipMemcpy(s->d_data, blocks, s->total_bytes,
                       hipMemcpyHostToDevice));
     s->d_data_owned = true;
@@ -1117,8 +1118,8 @@ int yb_gpu_scan_feed_sst(ybg_scan_t* s, const uint8_t* file, uint64_t size,
                     hipMemcpyHostToDevice));
   HIP_TRY(hipMalloc(&d_types, n));
   HIP_TRY(hipMemcpy(d_types, typs.data(), n, hipMemcpyHostToDevice));
-  HIP_TRY(hipMalloc(&d_out, total_un + 48));
-  HIP_TRY(hipMemset(d_out + total_un, 0, 48));
+  HIP_TRY(hipMalloc(&d_out, total_un + 256));
+  HIP_TRY(hipMemset(d_out + total_un, 0, 256));
   HIP_TRY(hipMalloc(&d_err, 8));
   HIP_TRY(hipMemset(d_err, 0, 8));
   int sgrid = (int)std::min<uint64_t>((n + 255) / 256, 4096);
