@@ -524,7 +524,7 @@ struct Rdr {
       const uint64_t* q = (const uint64_t*)base;
       q2 = q[2];
       q3 = q[3];
-#if defined(YBG_PF) && defined(__HIP_DEVICE_COMPILE__)
+#if !defined(YBG_NO_PF) && defined(__HIP_DEVICE_COMPILE__)
       // sliding L2 prefetch: one extra independent load per 64 consumed
       // bytes, ~192 B ahead of the stream (device buffers carry 256 B of
       // tail slack for this). The asm keeps the otherwise-dead load.
