@@ -86,6 +86,19 @@ class GpuScan:
                                               n_blocks, 0),
             "feed_blocks")
 
+    def feed_blocks_intents(self, data, offsets, n_blocks, intents_blob,
+                            blob_len, txns, n_txns):
+        """Feed regular blocks + an intent stream (resolve + merge + feed;
+        yb_gpu_scan_feed_blocks_intents)."""
+        f = self._lib.yb_gpu_scan_feed_blocks_intents
+        f.restype = C.c_int
+        f.argtypes = [C.c_void_p, C.POINTER(C.c_uint8),
+                      C.POINTER(C.c_uint64), C.c_uint64,
+                      C.POINTER(C.c_uint8), C.c_uint64,
+                      C.POINTER(y.TxnStatus), C.c_uint32]
+        self._check(f(self._h, data, offsets, n_blocks, intents_blob,
+                      blob_len, txns, n_txns), "feed_blocks_intents")
+
     def feed_sst(self, file_ptr, size, verify=True):
         """Feed a complete BlockBasedTable SST file (footer + index block
         parsed host-side; block checksums verified when verify)."""

@@ -375,6 +375,74 @@ int ybg_sst_index(const uint8_t *file, uint64_t size, int verify,
 
 void ybg_builder_destroy(ybg_builder_t *b);
 
+/* ---- Intents-DB merge (docdb/intent_aware_iterator.cc:983-1011
+ * ProcessIntent + DecodeStrongWriteIntent; transaction status resolution
+ * docdb/transaction_status_cache.cc) ----------------------------------
+ *
+ * The reference scans TWO sorted streams: the regular DB and the intents
+ * DB holding provisional (in-flight transaction) records; each intent is
+ * resolved against the transaction status cache, and a committed intent
+ * behaves as a regular record positioned at its COMMIT hybrid time whose
+ * value carries the intent WRITE time as a kHybridTime control prefix
+ * (the committed-intent visibility rule, :1249-1267). Aborted and
+ * still-pending foreign intents are invisible.
+ *
+ * This implementation resolves and merges at FEED time: committed intents
+ * are synthesized into regular-format records and merge-rebuilt into the
+ * affected data blocks (intents are sparse; untouched blocks are reused
+ * byte-identical), so every scan then exercises the same committed-intent
+ * visibility rule on device. The intent stream is accepted in its
+ * post-DecodeStrongWriteIntent form: user key (DocKey [+ subkey], no HT
+ * suffix) + write DocHybridTime fields + provisional value body. The
+ * intents-DB KEY encoding (intent type sets, reverse txn index) is not
+ * transcribed — it is consumed before this boundary. */
+typedef struct {
+  uint32_t txn_id;
+  int32_t status;      /* 0 pending, 1 committed, 2 aborted */
+  uint64_t commit_ht;  /* micros<<12|logical, valid when committed */
+} ybg_txn_status_t;
+
+/* Intent-stream builder (test/ingest helper). Records may be added in any
+ * order; resolution sorts. Value forms mirror the regular builder. */
+typedef struct ybg_intents ybg_intents_t;
+ybg_intents_t *ybg_intents_create(const ybg_schema_t *schema);
+int ybg_intents_add_packed_row(ybg_intents_t *it, const ybg_key_t *key,
+                               uint64_t write_ht, uint32_t write_id,
+                               uint32_t txn_id, int packed_version,
+                               const ybg_rowvals_t *vals);
+int ybg_intents_add_column_update(ybg_intents_t *it, const ybg_key_t *key,
+                                  int value_col_idx, uint64_t write_ht,
+                                  uint32_t write_id, uint32_t txn_id,
+                                  uint64_t datum, const uint8_t *str,
+                                  uint64_t str_len, int null);
+int ybg_intents_add_row_tombstone(ybg_intents_t *it, const ybg_key_t *key,
+                                  uint64_t write_ht, uint32_t write_id,
+                                  uint32_t txn_id);
+int ybg_intents_data(ybg_intents_t *it, const uint8_t **blob,
+                     uint64_t *len);
+void ybg_intents_destroy(ybg_intents_t *it);
+
+/* Resolve the intent stream against the status table and merge committed
+ * intents into the data blocks (host-side; affected blocks are decoded,
+ * merge-inserted in internal-key order and re-encoded with the standard
+ * BlockBuilder; untouched blocks copy through). Outputs are malloc'd —
+ * free with ybg_free. */
+int ybg_merge_intents(const uint8_t *blocks, const uint64_t *offsets,
+                      uint64_t n_blocks, int kv_format,
+                      const uint8_t *intents, uint64_t intents_len,
+                      const ybg_txn_status_t *txns, uint32_t n_txns,
+                      uint8_t **out_blocks, uint64_t **out_offsets,
+                      uint64_t *out_n_blocks, uint64_t *out_total);
+
+/* Feed regular blocks + an intent stream: resolve + merge + feed. */
+int yb_gpu_scan_feed_blocks_intents(ybg_scan_t *s, const uint8_t *blocks,
+                                    const uint64_t *offsets,
+                                    uint64_t n_blocks,
+                                    const uint8_t *intents,
+                                    uint64_t intents_len,
+                                    const ybg_txn_status_t *txns,
+                                    uint32_t n_txns);
+
 /* One-call multi-threaded benchmark dataset generator.
  * Produces `rows` rows of the given schema in one tablet: hash prefix walks
  * 0..65535 monotonically, one int64 range key column ascending, packed-row

@@ -309,6 +309,31 @@ int orcl_scan(const uint8_t *const *blocks, const size_t *sizes, size_t nblocks,
               const orcl_scan_spec_t *spec, orcl_scan_result_t *result,
               orcl_row_cb row_cb, void *cb_arg);
 
+/* ---- intents-DB merge oracle (intent_aware_iterator.cc:983-1011
+ * ProcessIntent; transaction_status_cache.cc) -------------------------
+ * Runtime TWO-STREAM merge: the regular block stream and the resolved
+ * intent stream are consumed in internal-key order through the same
+ * visibility/row pipeline. Committed intents enter at their COMMIT
+ * DocHybridTime with the intent WRITE time as the value's kHybridTime
+ * prefix; pending and aborted intents are invisible. (The GPU product
+ * path merges at feed time instead — the two independent algorithms are
+ * cross-checked by the parity tests.) Intent blob layout per record:
+ * [u32 txn_id][u32 write_id][u64 write_ht][u32 klen][u32 vlen]
+ * [user key without HT suffix][value body]. */
+typedef struct {
+  uint32_t txn_id;
+  int32_t status; /* 0 pending, 1 committed, 2 aborted */
+  uint64_t commit_ht;
+} orcl_txn_status_t;
+
+int orcl_scan_intents(const uint8_t *const *blocks, const size_t *sizes,
+                      size_t nblocks, orcl_kv_format_t fmt,
+                      const orcl_schema_t *schema,
+                      const orcl_scan_spec_t *spec, const uint8_t *intents,
+                      size_t intents_len, const orcl_txn_status_t *txns,
+                      uint32_t n_txns, orcl_scan_result_t *result,
+                      orcl_row_cb row_cb, void *cb_arg);
+
 #ifdef __cplusplus
 }
 #endif

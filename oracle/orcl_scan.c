@@ -9,6 +9,7 @@
  * TEST INFRASTRUCTURE ONLY (see orcl.h header comment).
  */
 #include "orcl.h"
+#include <stdlib.h>
 #include <string.h>
 
 /* value/key entry type bytes — src/yb/dockv/value_type.h */
@@ -715,6 +716,137 @@ static int finalize_row(scan_ctx_t *cx, row_state_t *st) {
   return 0;
 }
 
+/* One merged-stream entry through the full visibility / row-assembly /
+ * restart-tracking pipeline (the body of the orcl_scan loop, shared with
+ * the intents-merging driver below). Returns 0 = continue, 1 = stop early
+ * (row callback), -1 = error. */
+static int scan_feed_entry(const orcl_schema_t *schema,
+                           const orcl_scan_spec_t *spec,
+                           orcl_scan_result_t *res, scan_ctx_t *cx,
+                           row_state_t *st, const uint8_t *reg_limit,
+                           size_t reg_limit_len, const uint8_t *key,
+                           size_t key_len, const uint8_t *value,
+                           size_t value_len) {
+  const orcl_read_time_t *rt = &spec->read_time;
+  res->entries_seen++;
+  /* internal key = user_key ‖ fixed64(seq<<8|type) — dbformat.h:84-110 */
+  if (key_len < 9) return -1;
+  const uint8_t *ukey = key;
+  size_t ukey_len = key_len - 8;
+  size_t ht_size = orcl_dht_encoded_size_from_end(ukey, ukey_len);
+  if (!ht_size || ukey_len < ht_size + 1) return -1;
+  const uint8_t *ht_enc = ukey + ukey_len - ht_size;
+  if (ukey[ukey_len - ht_size - 1] != VT_HYBRID_TIME) return -1;
+  size_t prefix_len = ukey_len - ht_size - 1; /* DocKey ‖ [subkey] */
+
+  /* --- visibility (SkipFutureRecords, intent_aware_iterator.cc:1223-1317) */
+
+  int visible;
+  if (value_len > 0 && value[0] == VT_HYBRID_TIME) {
+    /* committed-txn record with intent time (:1249-1267) */
+    const uint8_t *v1 = value + 1;
+    size_t v1_len = value_len - 1;
+    const uint8_t *max_allowed;
+    size_t max_allowed_len;
+    if (slice_cmp(v1, v1_len, rt->local_limit, rt->local_limit_len) > 0) {
+      max_allowed = rt->global_limit;
+      max_allowed_len = rt->global_limit_len;
+    } else {
+      max_allowed = rt->read;
+      max_allowed_len = rt->read_len;
+    }
+    visible = slice_cmp(ht_enc, ht_size, max_allowed, max_allowed_len) >= 0;
+    if (visible) {
+      size_t iht = orcl_dht_size_from_start(v1, v1_len);
+      if (!iht) return -1;
+      value = v1 + iht;
+      value_len = v1_len - iht;
+    }
+  } else {
+    visible = slice_cmp(ht_enc, ht_size, reg_limit, reg_limit_len) >= 0;
+  }
+  if (!visible) return 0;
+
+  /* --- read-restart tracking (UpdateMaxSeenHt,
+   * intent_aware_iterator.cc:815-827; GetReadRestartData :1400-1410):
+   * a visible record with commit > read (encoded bytes BELOW
+   * encoded(read)) is a restart candidate; keep the MIN encoded form
+   * (= max commit time). */
+  if (slice_cmp(ht_enc, ht_size, rt->read, rt->read_len) < 0) {
+    if (res->restart_ht_len == 0 ||
+        slice_cmp(ht_enc, ht_size, res->restart_ht,
+                  res->restart_ht_len) < 0) {
+      size_t n = ht_size < ORCL_MAX_HT_SIZE ? ht_size : ORCL_MAX_HT_SIZE;
+      memcpy(res->restart_ht, ht_enc, n);
+      res->restart_ht_len = (uint32_t)n;
+    }
+  }
+
+  /* --- row grouping: split DocKey from subkeys (InitIterKey,
+   * doc_rowwise_iterator.cc:363-404) */
+  size_t dockey_len = parse_dockey(schema, ukey, prefix_len, NULL);
+  if (!dockey_len || dockey_len > prefix_len) return -1;
+
+  if (!st->active || st->rowkey_len != dockey_len ||
+      memcmp(st->rowkey, ukey, dockey_len) != 0) {
+    int fr = finalize_row(cx, st);
+    if (fr) return fr > 0 ? 1 : -1;
+    memcpy(st->rowkey, ukey, dockey_len);
+    st->rowkey_len = dockey_len;
+    st->active = 1;
+    st->base_seen = 0;
+    st->col_seen_mask = 0;
+    st->liveness_seen = 0;
+  }
+
+  if (dockey_len == prefix_len) {
+    /* bare row entry (packed row / tombstone) */
+    if (!st->base_seen) {
+      st->base_seen = 1;
+      st->base_value = value;
+      st->base_value_len = value_len;
+      memcpy(st->base_ht, ht_enc, ht_size);
+      st->base_ht_len = ht_size;
+    }
+  } else {
+    /* column subkey: 'K' svarint(column_id) or 'J' svarint(id) —
+     * key_bytes.cc:54-56 (AppendColumnId) */
+    const uint8_t *sk = ukey + dockey_len;
+    size_t sk_len = prefix_len - dockey_len;
+    if (sk_len < 2) return -1;
+    int64_t col_id;
+    size_t sz = orcl_svarint_decode(sk + 1, sk_len - 1, &col_id);
+    if (!sz || 1 + sz != sk_len) return -1;
+    if (sk[0] == VT_SYS_COL) {
+      if (!st->liveness_seen) {
+        st->liveness_seen = 1;
+        st->liveness_value = value;
+        st->liveness_value_len = value_len;
+        memcpy(st->liveness_ht, ht_enc, ht_size);
+        st->liveness_ht_len = ht_size;
+      }
+    } else if (sk[0] == VT_COL) {
+      int idx = -1;
+      for (int i = 0; i < schema->num_value_cols; ++i) {
+        if (schema->value_cols[i].column_id == (int32_t)col_id) {
+          idx = i;
+          break;
+        }
+      }
+      if (idx >= 0 && !(st->col_seen_mask & (1u << idx))) {
+        st->col_seen_mask |= 1u << idx;
+        st->col_value[idx] = value;
+        st->col_value_len[idx] = value_len;
+        memcpy(st->col_ht[idx], ht_enc, ht_size);
+        st->col_ht_len[idx] = ht_size;
+      }
+    } else {
+      return -1;
+    }
+  }
+  return 0;
+}
+
 int orcl_scan(const uint8_t *const *blocks, const size_t *sizes, size_t nblocks,
               orcl_kv_format_t fmt, const orcl_schema_t *schema,
               const orcl_scan_spec_t *spec, orcl_scan_result_t *result,
@@ -750,127 +882,174 @@ int orcl_scan(const uint8_t *const *blocks, const size_t *sizes, size_t nblocks,
     if (orcl_block_iter_init(&it, blocks[b], sizes[b], fmt)) return -1;
     int r;
     while ((r = orcl_block_iter_next(&it)) == 1) {
-      result->entries_seen++;
-      /* internal key = user_key ‖ fixed64(seq<<8|type) — dbformat.h:84-110 */
-      if (it.key_len < 9) return -1;
-      const uint8_t *ukey = it.key;
-      size_t ukey_len = it.key_len - 8;
-      size_t ht_size = orcl_dht_encoded_size_from_end(ukey, ukey_len);
-      if (!ht_size || ukey_len < ht_size + 1) return -1;
-      const uint8_t *ht_enc = ukey + ukey_len - ht_size;
-      if (ukey[ukey_len - ht_size - 1] != VT_HYBRID_TIME) return -1;
-      size_t prefix_len = ukey_len - ht_size - 1; /* DocKey ‖ [subkey] */
-
-      /* --- visibility (SkipFutureRecords, intent_aware_iterator.cc:1223-1317) */
-      const uint8_t *value = it.value;
-      size_t value_len = it.value_len;
-      int visible;
-      if (value_len > 0 && value[0] == VT_HYBRID_TIME) {
-        /* committed-txn record with intent time (:1249-1267) */
-        const uint8_t *v1 = value + 1;
-        size_t v1_len = value_len - 1;
-        const uint8_t *max_allowed;
-        size_t max_allowed_len;
-        if (slice_cmp(v1, v1_len, rt->local_limit, rt->local_limit_len) > 0) {
-          max_allowed = rt->global_limit;
-          max_allowed_len = rt->global_limit_len;
-        } else {
-          max_allowed = rt->read;
-          max_allowed_len = rt->read_len;
-        }
-        visible = slice_cmp(ht_enc, ht_size, max_allowed, max_allowed_len) >= 0;
-        if (visible) {
-          size_t iht = orcl_dht_size_from_start(v1, v1_len);
-          if (!iht) return -1;
-          value = v1 + iht;
-          value_len = v1_len - iht;
-        }
-      } else {
-        visible = slice_cmp(ht_enc, ht_size, reg_limit, reg_limit_len) >= 0;
-      }
-      if (!visible) continue;
-
-      /* --- read-restart tracking (UpdateMaxSeenHt,
-       * intent_aware_iterator.cc:815-827; GetReadRestartData :1400-1410):
-       * a visible record with commit > read (encoded bytes BELOW
-       * encoded(read)) is a restart candidate; keep the MIN encoded form
-       * (= max commit time). */
-      if (slice_cmp(ht_enc, ht_size, rt->read, rt->read_len) < 0) {
-        if (result->restart_ht_len == 0 ||
-            slice_cmp(ht_enc, ht_size, result->restart_ht,
-                      result->restart_ht_len) < 0) {
-          size_t n = ht_size < ORCL_MAX_HT_SIZE ? ht_size : ORCL_MAX_HT_SIZE;
-          memcpy(result->restart_ht, ht_enc, n);
-          result->restart_ht_len = (uint32_t)n;
-        }
-      }
-
-      /* --- row grouping: split DocKey from subkeys (InitIterKey,
-       * doc_rowwise_iterator.cc:363-404) */
-      size_t dockey_len = parse_dockey(schema, ukey, prefix_len, NULL);
-      if (!dockey_len || dockey_len > prefix_len) return -1;
-
-      if (!st.active || st.rowkey_len != dockey_len ||
-          memcmp(st.rowkey, ukey, dockey_len) != 0) {
-        int fr = finalize_row(&cx, &st);
-        if (fr) return fr > 0 ? 0 : -1;
-        memcpy(st.rowkey, ukey, dockey_len);
-        st.rowkey_len = dockey_len;
-        st.active = 1;
-        st.base_seen = 0;
-        st.col_seen_mask = 0;
-        st.liveness_seen = 0;
-      }
-
-      if (dockey_len == prefix_len) {
-        /* bare row entry (packed row / tombstone) */
-        if (!st.base_seen) {
-          st.base_seen = 1;
-          st.base_value = value;
-          st.base_value_len = value_len;
-          memcpy(st.base_ht, ht_enc, ht_size);
-          st.base_ht_len = ht_size;
-        }
-      } else {
-        /* column subkey: 'K' svarint(column_id) or 'J' svarint(id) —
-         * key_bytes.cc:54-56 (AppendColumnId) */
-        const uint8_t *sk = ukey + dockey_len;
-        size_t sk_len = prefix_len - dockey_len;
-        if (sk_len < 2) return -1;
-        int64_t col_id;
-        size_t sz = orcl_svarint_decode(sk + 1, sk_len - 1, &col_id);
-        if (!sz || 1 + sz != sk_len) return -1;
-        if (sk[0] == VT_SYS_COL) {
-          if (!st.liveness_seen) {
-            st.liveness_seen = 1;
-            st.liveness_value = value;
-            st.liveness_value_len = value_len;
-            memcpy(st.liveness_ht, ht_enc, ht_size);
-            st.liveness_ht_len = ht_size;
-          }
-        } else if (sk[0] == VT_COL) {
-          int idx = -1;
-          for (int i = 0; i < schema->num_value_cols; ++i) {
-            if (schema->value_cols[i].column_id == (int32_t)col_id) {
-              idx = i;
-              break;
-            }
-          }
-          if (idx >= 0 && !(st.col_seen_mask & (1u << idx))) {
-            st.col_seen_mask |= 1u << idx;
-            st.col_value[idx] = value;
-            st.col_value_len[idx] = value_len;
-            memcpy(st.col_ht[idx], ht_enc, ht_size);
-            st.col_ht_len[idx] = ht_size;
-          }
-        } else {
-          return -1;
-        }
-      }
+      int fe = scan_feed_entry(schema, spec, result, &cx, &st, reg_limit,
+                               reg_limit_len, it.key, it.key_len, it.value,
+                               it.value_len);
+      if (fe) return fe > 0 ? 0 : -1;
     }
     if (r < 0) return -1;
   }
   int fr = finalize_row(&cx, &st);
   if (fr < 0) return -1;
   return 0;
+}
+
+
+/* ---------------------------------------------------------------------------
+ * Intents-DB merge oracle: see orcl.h. The regular stream and the resolved
+ * intent records are merged in rocksdb internal-key order (user key asc,
+ * seqno desc — db/dbformat.h:84-110) and fed one at a time through
+ * scan_feed_entry — the same pipeline orcl_scan uses.
+ * ------------------------------------------------------------------------- */
+
+typedef struct {
+  uint8_t *key;
+  size_t key_len;
+  uint8_t *value;
+  size_t value_len;
+} resolved_intent_t;
+
+static int ikey_cmp(const uint8_t *a, size_t alen, const uint8_t *b,
+                    size_t blen) {
+  size_t ua = alen - 8, ub = blen - 8;
+  size_t n = ua < ub ? ua : ub;
+  int c = memcmp(a, b, n);
+  if (c) return c;
+  if (ua != ub) return ua < ub ? -1 : 1;
+  uint64_t sa, sb;
+  memcpy(&sa, a + ua, 8);
+  memcpy(&sb, b + ub, 8);
+  return sa > sb ? -1 : (sa < sb ? 1 : 0); /* seq DESC */
+}
+
+static int resolved_cmp_qsort(const void *pa, const void *pb) {
+  const resolved_intent_t *a = (const resolved_intent_t *)pa;
+  const resolved_intent_t *b = (const resolved_intent_t *)pb;
+  return ikey_cmp(a->key, a->key_len, b->key, b->key_len);
+}
+
+int orcl_scan_intents(const uint8_t *const *blocks, const size_t *sizes,
+                      size_t nblocks, orcl_kv_format_t fmt,
+                      const orcl_schema_t *schema,
+                      const orcl_scan_spec_t *spec, const uint8_t *intents,
+                      size_t intents_len, const orcl_txn_status_t *txns,
+                      uint32_t n_txns, orcl_scan_result_t *result,
+                      orcl_row_cb row_cb, void *cb_arg) {
+  memset(result, 0, sizeof(*result));
+  for (int g = 0; g < spec->num_aggs; ++g) result->aggs[g].is_null = 1;
+
+  /* resolve (ProcessIntent + status-cache lookup) */
+  resolved_intent_t *res = NULL;
+  size_t n_res = 0, cap_res = 0;
+  int rc = -1;
+  {
+    size_t o = 0, idx = 0;
+    while (o < intents_len) {
+      uint32_t txn_id, write_id, klen, vlen;
+      uint64_t wht;
+      if (o + 24 > intents_len) goto done;
+      memcpy(&txn_id, intents + o, 4);
+      memcpy(&write_id, intents + o + 4, 4);
+      memcpy(&wht, intents + o + 8, 8);
+      memcpy(&klen, intents + o + 16, 4);
+      memcpy(&vlen, intents + o + 20, 4);
+      o += 24;
+      if (o + klen + vlen > intents_len) goto done;
+      const uint8_t *kp = intents + o;
+      const uint8_t *vp = intents + o + klen;
+      o += (size_t)klen + vlen;
+      const orcl_txn_status_t *ts = NULL;
+      for (uint32_t t = 0; t < n_txns; ++t)
+        if (txns[t].txn_id == txn_id) {
+          ts = &txns[t];
+          break;
+        }
+      if (!ts) goto done;
+      if (ts->status != 1) continue; /* pending / aborted: invisible */
+      if (n_res == cap_res) {
+        cap_res = cap_res ? cap_res * 2 : 16;
+        res = (resolved_intent_t *)realloc(res,
+                                           cap_res * sizeof(*res));
+      }
+      resolved_intent_t *r = &res[n_res++];
+      /* internal key: user key + '#' + DHT(commit, write_id) + seq desc */
+      r->key = (uint8_t *)malloc((size_t)klen + 1 + 16 + 8);
+      memcpy(r->key, kp, klen);
+      size_t kl = klen;
+      r->key[kl++] = VT_HYBRID_TIME;
+      kl += orcl_dht_encode(ts->commit_ht, write_id, r->key + kl);
+      uint64_t seq = (((uint64_t)1 << 55) + idx) << 8 | 0x1 /*kTypeValue*/;
+      memcpy(r->key + kl, &seq, 8);
+      kl += 8;
+      r->key_len = kl;
+      /* value: kHybridTime + DHT(write time) + provisional body */
+      r->value = (uint8_t *)malloc((size_t)vlen + 1 + 16);
+      r->value[0] = VT_HYBRID_TIME;
+      size_t vl = 1 + orcl_dht_encode(wht, write_id, r->value + 1);
+      memcpy(r->value + vl, vp, vlen);
+      r->value_len = vl + vlen;
+      ++idx;
+    }
+  }
+  if (n_res > 1) qsort(res, n_res, sizeof(*res), resolved_cmp_qsort);
+
+  {
+    packing_t pk;
+    int nullable[ORCL_MAX_COLS];
+    for (int i = 0; i < schema->num_value_cols; ++i)
+      nullable[i] = schema->value_cols[i].nullable;
+    packing_init(&pk, schema, nullable);
+    scan_ctx_t cx = {.pk = &pk, .spec = spec, .res = result,
+                     .cb = row_cb, .cb_arg = cb_arg, .row_seq = 0};
+    const orcl_read_time_t *rt = &spec->read_time;
+    const uint8_t *reg_limit = rt->read;
+    size_t reg_limit_len = rt->read_len;
+    if (slice_cmp(rt->local_limit, rt->local_limit_len, rt->read,
+                  rt->read_len) < 0) {
+      reg_limit = rt->local_limit;
+      reg_limit_len = rt->local_limit_len;
+    }
+    row_state_t st;
+    memset(&st, 0, sizeof(st));
+    size_t ii = 0;
+    orcl_block_iter_t it;
+    for (size_t b = 0; b < nblocks; ++b) {
+      if (orcl_block_iter_init(&it, blocks[b], sizes[b], fmt)) goto done;
+      int r2;
+      while ((r2 = orcl_block_iter_next(&it)) == 1) {
+        /* flush intents ordered before this regular entry */
+        while (ii < n_res &&
+               ikey_cmp(res[ii].key, res[ii].key_len, it.key, it.key_len) <
+                   0) {
+          int fe = scan_feed_entry(schema, spec, result, &cx, &st,
+                                   reg_limit, reg_limit_len, res[ii].key,
+                                   res[ii].key_len, res[ii].value,
+                                   res[ii].value_len);
+          if (fe) { rc = fe > 0 ? 0 : -1; goto done; }
+          ++ii;
+        }
+        int fe = scan_feed_entry(schema, spec, result, &cx, &st, reg_limit,
+                                 reg_limit_len, it.key, it.key_len,
+                                 it.value, it.value_len);
+        if (fe) { rc = fe > 0 ? 0 : -1; goto done; }
+      }
+      if (r2 < 0) goto done;
+    }
+    while (ii < n_res) {
+      int fe = scan_feed_entry(schema, spec, result, &cx, &st, reg_limit,
+                               reg_limit_len, res[ii].key, res[ii].key_len,
+                               res[ii].value, res[ii].value_len);
+      if (fe) { rc = fe > 0 ? 0 : -1; goto done; }
+      ++ii;
+    }
+    int fr = finalize_row(&cx, &st);
+    rc = fr < 0 ? -1 : 0;
+  }
+done:
+  for (size_t i = 0; i < n_res; ++i) {
+    free(res[i].key);
+    free(res[i].value);
+  }
+  free(res);
+  return rc;
 }

@@ -758,3 +758,188 @@ def snappy_uncompress(data, cap):
     if n < 0:
         raise RuntimeError("snappy_uncompress failed")
     return bytes(dst[:n])
+
+
+class TxnStatus(C.Structure):
+    """ybg_txn_status_t / orcl_txn_status_t (same layout)."""
+    _fields_ = [("txn_id", C.c_uint32), ("status", C.c_int32),
+                ("commit_ht", C.c_uint64)]
+
+
+TXN_PENDING, TXN_COMMITTED, TXN_ABORTED = 0, 1, 2
+
+
+def make_txns(table):
+    """table: dict txn_id -> ("committed", commit_ht_micros, logical?) |
+    "pending" | "aborted". Returns a ctypes TxnStatus array."""
+    arr = (TxnStatus * max(len(table), 1))()
+    for i, (tid, st) in enumerate(sorted(table.items())):
+        arr[i].txn_id = tid
+        if st == "pending":
+            arr[i].status = TXN_PENDING
+        elif st == "aborted":
+            arr[i].status = TXN_ABORTED
+        else:
+            arr[i].status = TXN_COMMITTED
+            ht = st[1] if isinstance(st, tuple) else st
+            arr[i].commit_ht = ht << 12
+    return arr, len(table)
+
+
+class Intents:
+    """Intent-stream builder (provisional records of in-flight
+    transactions; post-DecodeStrongWriteIntent form — see
+    include/yb_gpu_scan.h)."""
+
+    def __init__(self, schema):
+        lib = product()
+        self._lib = lib
+        self._schema = schema
+        create = _sig(lib, "ybg_intents_create", C.c_void_p,
+                      [C.POINTER(Schema)])
+        self._h = create(C.byref(schema))
+        self._add_packed = _sig(lib, "ybg_intents_add_packed_row", C.c_int,
+                                [C.c_void_p, C.POINTER(Key), C.c_uint64,
+                                 C.c_uint32, C.c_uint32, C.c_int,
+                                 C.POINTER(RowVals)])
+        self._add_col = _sig(lib, "ybg_intents_add_column_update", C.c_int,
+                             [C.c_void_p, C.POINTER(Key), C.c_int,
+                              C.c_uint64, C.c_uint32, C.c_uint32,
+                              C.c_uint64, C.POINTER(C.c_uint8), C.c_uint64,
+                              C.c_int])
+        self._add_tomb = _sig(lib, "ybg_intents_add_row_tombstone", C.c_int,
+                              [C.c_void_p, C.POINTER(Key), C.c_uint64,
+                               C.c_uint32, C.c_uint32])
+        self._data = _sig(lib, "ybg_intents_data", C.c_int,
+                          [C.c_void_p, C.POINTER(C.POINTER(C.c_uint8)),
+                           C.POINTER(C.c_uint64)])
+        self._keepalive = []
+
+    def _key(self, hash_=0, datums=(), strs=()):
+        k = Key()
+        k.hash = hash_
+        for i, d in enumerate(datums):
+            k.datums[i] = d & 0xFFFFFFFFFFFFFFFF
+        for i, s in enumerate(strs):
+            if s is not None:
+                buf = C.create_string_buffer(s, len(s))
+                self._keepalive.append(buf)
+                k.strs[i] = C.cast(buf, C.POINTER(C.c_uint8))
+                k.str_lens[i] = len(s)
+        return k
+
+    def add_packed_row(self, write_ht_micros, txn_id, values, hash_=0,
+                       key_datums=(), key_strs=(), write_id=0,
+                       packed_version=2, logical=0):
+        import struct
+        k = self._key(hash_, key_datums, key_strs)
+        v = RowVals()
+        for i, (dt, val) in enumerate(values):
+            if val is None:
+                v.null[i] = 1
+            elif dt == T_DOUBLE:
+                v.datums[i] = struct.unpack("<Q", struct.pack("<d", val))[0]
+            elif dt == T_FLOAT:
+                v.datums[i] = struct.unpack("<I", struct.pack("<f", val))[0]
+            elif dt == T_STRING:
+                buf = C.create_string_buffer(val, len(val))
+                self._keepalive.append(buf)
+                v.strs[i] = C.cast(buf, C.POINTER(C.c_uint8))
+                v.str_lens[i] = len(val)
+            else:
+                v.datums[i] = val & 0xFFFFFFFFFFFFFFFF
+        rc = self._add_packed(self._h, C.byref(k),
+                              (write_ht_micros << 12) | logical, write_id,
+                              txn_id, packed_version, C.byref(v))
+        assert rc == 0
+
+    def add_column_update(self, write_ht_micros, txn_id, col_idx, value,
+                          hash_=0, key_datums=(), key_strs=(), write_id=0,
+                          null=False, logical=0):
+        import struct
+        k = self._key(hash_, key_datums, key_strs)
+        datum = 0
+        sp = None
+        slen = 0
+        dt = self._schema.value_cols[col_idx].dtype
+        if value is None:
+            null = True
+        elif dt == T_DOUBLE:
+            datum = struct.unpack("<Q", struct.pack("<d", value))[0]
+        elif dt == T_STRING:
+            buf = C.create_string_buffer(value, len(value))
+            self._keepalive.append(buf)
+            sp = C.cast(buf, C.POINTER(C.c_uint8))
+            slen = len(value)
+        else:
+            datum = value & 0xFFFFFFFFFFFFFFFF
+        rc = self._add_col(self._h, C.byref(k), col_idx,
+                           (write_ht_micros << 12) | logical, write_id,
+                           txn_id, datum, sp, slen, 1 if null else 0)
+        assert rc == 0
+
+    def add_row_tombstone(self, write_ht_micros, txn_id, hash_=0,
+                          key_datums=(), key_strs=(), write_id=0,
+                          logical=0):
+        k = self._key(hash_, key_datums, key_strs)
+        rc = self._add_tomb(self._h, C.byref(k),
+                            (write_ht_micros << 12) | logical, write_id,
+                            txn_id)
+        assert rc == 0
+
+    def blob(self):
+        p = C.POINTER(C.c_uint8)()
+        n = C.c_uint64()
+        rc = self._data(self._h, C.byref(p), C.byref(n))
+        assert rc == 0
+        return p, n.value
+
+
+def merge_intents(data, offsets, n_blocks, intents_blob, blob_len, txns,
+                  n_txns, kv_format=ENC_THREE_SHARED_PARTS):
+    """Product-side feed-time resolve+merge (ybg_merge_intents). Returns
+    (merged_data, merged_offsets, n_blocks, total) as ctypes buffers the
+    caller can pass to feed/sim APIs (freed at process exit)."""
+    lib = product()
+    f = _sig(lib, "ybg_merge_intents", C.c_int,
+             [C.POINTER(C.c_uint8), C.POINTER(C.c_uint64), C.c_uint64,
+              C.c_int, C.POINTER(C.c_uint8), C.c_uint64,
+              C.POINTER(TxnStatus), C.c_uint32,
+              C.POINTER(C.POINTER(C.c_uint8)),
+              C.POINTER(C.POINTER(C.c_uint64)), C.POINTER(C.c_uint64),
+              C.POINTER(C.c_uint64)])
+    ob = C.POINTER(C.c_uint8)()
+    oo = C.POINTER(C.c_uint64)()
+    on = C.c_uint64()
+    ot = C.c_uint64()
+    rc = f(data, offsets, n_blocks, kv_format, intents_blob, blob_len, txns,
+           n_txns, C.byref(ob), C.byref(oo), C.byref(on), C.byref(ot))
+    if rc != 0:
+        raise RuntimeError(f"ybg_merge_intents rc={rc}")
+    return ob, oo, on.value, ot.value
+
+
+def orcl_scan_intents(data, offsets, n_blocks, schema, spec, intents_blob,
+                      blob_len, txns, n_txns,
+                      kv_format=ENC_THREE_SHARED_PARTS):
+    """Oracle runtime two-stream merge scan — TEST INFRASTRUCTURE."""
+    lib = oracle()
+    f = _sig(lib, "orcl_scan_intents", C.c_int,
+             [C.POINTER(C.POINTER(C.c_uint8)), C.POINTER(C.c_size_t),
+              C.c_size_t, C.c_int, C.POINTER(OrclSchema),
+              C.POINTER(OrclScanSpec), C.POINTER(C.c_uint8), C.c_size_t,
+              C.POINTER(TxnStatus), C.c_uint32,
+              C.POINTER(OrclScanResult), ORCL_ROW_CB, C.c_void_p])
+    base = C.cast(data, C.c_void_p).value
+    blocks = (C.POINTER(C.c_uint8) * n_blocks)()
+    sizes = (C.c_size_t * n_blocks)()
+    for i in range(n_blocks):
+        blocks[i] = C.cast(base + offsets[i], C.POINTER(C.c_uint8))
+        sizes[i] = offsets[i + 1] - offsets[i]
+    res = OrclScanResult()
+    rc = f(blocks, sizes, n_blocks, kv_format, C.byref(schema),
+           C.byref(spec), intents_blob, blob_len, txns, n_txns,
+           C.byref(res), C.cast(None, ORCL_ROW_CB), None)
+    if rc != 0:
+        raise RuntimeError(f"orcl_scan_intents rc={rc}")
+    return res

@@ -1024,6 +1024,31 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
   return 0;
 }
 
+// Feed regular blocks together with an intent stream: resolve the
+// transaction statuses, merge committed intents into the affected blocks
+// (ybg_merge_intents, sstgen.cc) and feed the merged tablet. The scan
+// then applies the committed-intent visibility rule on device
+// (intent_aware_iterator.cc:1249-1267).
+int yb_gpu_scan_feed_blocks_intents(ybg_scan_t* s, const uint8_t* blocks,
+                                    const uint64_t* offsets,
+                                    uint64_t n_blocks,
+                                    const uint8_t* intents,
+                                    uint64_t intents_len,
+                                    const ybg_txn_status_t* txns,
+                                    uint32_t n_txns) {
+  uint8_t* mb = nullptr;
+  uint64_t* mo = nullptr;
+  uint64_t mn = 0, mt = 0;
+  int rc = ybg_merge_intents(blocks, offsets, n_blocks, s->spec.kv_format,
+                             intents, intents_len, txns, n_txns, &mb, &mo,
+                             &mn, &mt);
+  if (rc) return set_err(rc, "intent resolve/merge failed");
+  rc = yb_gpu_scan_feed_blocks(s, mb, mo, mn, 0);
+  ybg_free(mb);
+  ybg_free(mo);
+  return rc;
+}
+
 // Parse an SST file and report the data-block handles (the same parser
 // the feed path uses; exposed for CPU-side tests — errors via
 // yb_gpu_last_error).

@@ -811,3 +811,402 @@ size_t ybg_encode_dockey(const ybg_schema_t* schema, const ybg_key_t* key,
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// Intents-DB merge (intent_aware_iterator.cc:983-1011 ProcessIntent +
+// DecodeStrongWriteIntent; transaction_status_cache.cc). Committed intents
+// become regular-format records at their COMMIT DocHybridTime whose value
+// carries the intent WRITE time as the kHybridTime control prefix
+// (:1249-1267); they are merge-rebuilt into the affected data blocks in
+// internal-key order. See include/yb_gpu_scan.h for the boundary notes.
+// ---------------------------------------------------------------------------
+
+namespace ybg {
+namespace {
+
+// General data-block entry decoder (host): both KV encodings with every
+// delta form the BlockBuilder above can emit (rocksdb/table/block.cc:
+// 287-454, block_internal.h:54-160). last_key carries the reconstruction
+// state across entries.
+const uint8_t* HostLeb128(const uint8_t* p, const uint8_t* lim, uint64_t* v) {
+  uint64_t r = 0;
+  int sh = 0;
+  while (p < lim && sh <= 63) {
+    uint8_t b = *p++;
+    if (b & 0x80) {
+      r |= (uint64_t)(b & 0x7f) << sh;
+    } else {
+      *v = r | ((uint64_t)b << sh);
+      return p;
+    }
+    sh += 7;
+  }
+  return nullptr;
+}
+
+const uint8_t* HostSvarint(const uint8_t* p, const uint8_t* lim, int64_t* v) {
+  if (p >= lim) return nullptr;
+  uint32_t b0 = p[0];
+  uint32_t b1 = p + 1 < lim ? p[1] : 0;
+  uint32_t header = (b0 << 8) | b1;
+  bool neg = !(header & 0x8000);
+  if (neg) header = (uint16_t)~header;
+  int n = 0;
+  for (uint32_t m = 0x4000; m && (header & m); m >>= 1) ++n;
+  ++n;
+  if (p + n > lim) return nullptr;
+  uint64_t temp = 0;
+  for (int i = 0; i < n; ++i)
+    temp = (temp << 8) | (uint8_t)(neg ? ~p[i] : p[i]);
+  uint64_t mask = n >= 10 ? ~0ull
+                          : ((1ull << (7 * n - 1)) - 1);
+  temp &= mask;
+  *v = neg ? -(int64_t)temp : (int64_t)temp;
+  return p + n;
+}
+
+const uint8_t* DecodeEntryHost(int fmt, const uint8_t* p, const uint8_t* lim,
+                               Buf* last_key, Buf* value) {
+  if (fmt == YBG_ENC_SHARED_PREFIX) {
+    uint64_t shared, non_shared, vlen;
+    if (!(p = HostLeb128(p, lim, &shared))) return nullptr;
+    if (!(p = HostLeb128(p, lim, &non_shared))) return nullptr;
+    if (!(p = HostLeb128(p, lim, &vlen))) return nullptr;
+    if ((uint64_t)(lim - p) < non_shared + vlen ||
+        shared > last_key->size())
+      return nullptr;
+    last_key->resize(shared);
+    last_key->insert(last_key->end(), p, p + non_shared);
+    p += non_shared;
+    value->assign(p, p + vlen);
+    return p + vlen;
+  }
+  uint64_t e1;
+  if (!(p = HostLeb128(p, lim, &e1))) return nullptr;
+  uint64_t vlen = e1 >> 2;
+  bool inc = e1 & 2;
+  uint64_t sp = 0, ns1 = 0, ns2 = 0, reuse = 0;
+  int64_t d1 = 0, d2 = 0;
+  bool shared_something;
+  if (e1 & 1) {  // frequent
+    if (!(p = HostLeb128(p, lim, &sp))) return nullptr;
+    ns1 = 1;
+    ns2 = 1;
+    reuse = 8;
+    shared_something = true;
+  } else {
+    if (p >= lim) return nullptr;
+    uint8_t e2 = *p++;
+    if (!(e2 & 1)) {
+      shared_something = false;
+      if (e2 == 0) {
+        if (!(p = HostLeb128(p, lim, &ns1))) return nullptr;
+      } else {
+        ns1 = e2 >> 1;
+      }
+    } else {
+      shared_something = true;
+      if (!(e2 & 2)) {
+        reuse = 8;
+        d2 = (e2 >> 2) & 1;
+        ns1 = (e2 >> 3) & 7;
+        ns2 = (e2 >> 6) & 3;
+      } else {
+        reuse = (e2 & 4) ? 8 : 0;
+        if (!(p = HostLeb128(p, lim, &ns1))) return nullptr;
+        if (e2 & 8) {
+          if (!(p = HostSvarint(p, lim, &d1))) return nullptr;
+        }
+        if (e2 & 16) {
+          if (!(p = HostLeb128(p, lim, &ns2))) return nullptr;
+        }
+        if (e2 & 32) {
+          if (!(p = HostSvarint(p, lim, &d2))) return nullptr;
+        }
+      }
+      if (!(p = HostLeb128(p, lim, &sp))) return nullptr;
+    }
+  }
+  if ((uint64_t)(lim - p) < ns1 + ns2 + vlen) return nullptr;
+  if (!shared_something) {
+    last_key->assign(p, p + ns1);
+    p += ns1;
+  } else {
+    uint64_t prev_mid_start = sp + ns1 - (uint64_t)d1;
+    uint64_t prev_ns2 = ns2 - (uint64_t)d2;
+    uint64_t prev_except = prev_mid_start + prev_ns2 + reuse;
+    if (last_key->size() < prev_except) return nullptr;
+    uint64_t mid = last_key->size() - prev_except;
+    Buf nk;
+    nk.reserve(sp + ns1 + mid + ns2 + reuse);
+    nk.insert(nk.end(), last_key->begin(), last_key->begin() + sp);
+    nk.insert(nk.end(), p, p + ns1);
+    nk.insert(nk.end(), last_key->begin() + prev_mid_start,
+              last_key->begin() + prev_mid_start + mid);
+    nk.insert(nk.end(), p + ns1, p + ns1 + ns2);
+    if (reuse) {
+      uint64_t l8;
+      memcpy(&l8, last_key->data() + last_key->size() - 8, 8);
+      if (inc) l8 += 0x100;
+      size_t at = nk.size();
+      nk.resize(at + 8);
+      memcpy(nk.data() + at, &l8, 8);
+    }
+    p += ns1 + ns2;
+    *last_key = std::move(nk);
+  }
+  value->assign(p, p + vlen);
+  return p + vlen;
+}
+
+int DecodeBlockHost(const uint8_t* blk, size_t size, int fmt,
+                    std::vector<std::pair<Buf, Buf>>* out) {
+  if (size < 8) return 3;
+  uint32_t nr;
+  memcpy(&nr, blk + size - 4, 4);
+  if (nr == 0 || (uint64_t)nr * 4 + 4 > size) return 3;
+  const uint8_t* lim = blk + size - 4 - (uint64_t)nr * 4;
+  const uint8_t* p = blk;
+  Buf key, val;
+  while (p < lim) {
+    p = DecodeEntryHost(fmt, p, lim, &key, &val);
+    if (!p) return 3;
+    out->emplace_back(key, val);
+  }
+  return 0;
+}
+
+// rocksdb internal-key order (dbformat.h:84-110): user key asc, seq DESC
+int IKeyCmp(const Buf& a, const Buf& b) {
+  size_t ua = a.size() - 8, ub = b.size() - 8;
+  size_t n = ua < ub ? ua : ub;
+  int c = memcmp(a.data(), b.data(), n);
+  if (c) return c;
+  if (ua != ub) return ua < ub ? -1 : 1;
+  uint64_t sa, sb;
+  memcpy(&sa, a.data() + ua, 8);
+  memcpy(&sb, b.data() + ub, 8);
+  return sa > sb ? -1 : (sa < sb ? 1 : 0);
+}
+
+struct IntentRec {
+  uint32_t txn_id, write_id;
+  uint64_t write_ht;
+  Buf key_prefix;  // user key without the '#' + DHT suffix
+  Buf value_body;
+};
+
+// synthetic seqnos above the generator's 1<<50 range: equal-user-key ties
+// order the resolved intent first (newest-wins sees it as newer)
+constexpr uint64_t kIntentSeqBase = 1ull << 55;
+
+}  // namespace
+}  // namespace ybg
+
+extern "C" {
+
+struct ybg_intents {
+  ybg_schema_t schema;
+  std::vector<ybg::IntentRec> recs;
+  ybg::Buf blob;
+};
+
+ybg_intents_t* ybg_intents_create(const ybg_schema_t* schema) {
+  auto* it = new ybg_intents();
+  it->schema = *schema;
+  return it;
+}
+
+int ybg_intents_add_packed_row(ybg_intents_t* it, const ybg_key_t* key,
+                               uint64_t write_ht, uint32_t write_id,
+                               uint32_t txn_id, int packed_version,
+                               const ybg_rowvals_t* vals) {
+  ybg::IntentRec r;
+  r.txn_id = txn_id;
+  r.write_id = write_id;
+  r.write_ht = write_ht;
+  ybg::EncodeDocKey(&it->schema, key, &r.key_prefix);
+  ybg::EncodePackedRow(&it->schema, packed_version, vals, &r.value_body);
+  it->recs.push_back(std::move(r));
+  return 0;
+}
+
+int ybg_intents_add_column_update(ybg_intents_t* it, const ybg_key_t* key,
+                                  int value_col_idx, uint64_t write_ht,
+                                  uint32_t write_id, uint32_t txn_id,
+                                  uint64_t datum, const uint8_t* str,
+                                  uint64_t str_len, int null) {
+  ybg::IntentRec r;
+  r.txn_id = txn_id;
+  r.write_id = write_id;
+  r.write_ht = write_ht;
+  ybg::EncodeDocKey(&it->schema, key, &r.key_prefix);
+  r.key_prefix.push_back(ybg::kColByte);
+  ybg::SVarintAppend(it->schema.value_cols[value_col_idx].column_id,
+                     &r.key_prefix);
+  ybg::EncodeV1Value((ybg_dtype_t)it->schema.value_cols[value_col_idx].dtype,
+                     datum, str, str_len, null, &r.value_body);
+  it->recs.push_back(std::move(r));
+  return 0;
+}
+
+int ybg_intents_add_row_tombstone(ybg_intents_t* it, const ybg_key_t* key,
+                                  uint64_t write_ht, uint32_t write_id,
+                                  uint32_t txn_id) {
+  ybg::IntentRec r;
+  r.txn_id = txn_id;
+  r.write_id = write_id;
+  r.write_ht = write_ht;
+  ybg::EncodeDocKey(&it->schema, key, &r.key_prefix);
+  r.value_body.push_back(ybg::kTombstoneByte);
+  it->recs.push_back(std::move(r));
+  return 0;
+}
+
+int ybg_intents_data(ybg_intents_t* it, const uint8_t** blob,
+                     uint64_t* len) {
+  it->blob.clear();
+  for (const auto& r : it->recs) {
+    ybg::Fixed32LEAppend(r.txn_id, &it->blob);
+    ybg::Fixed32LEAppend(r.write_id, &it->blob);
+    ybg::Fixed64LEAppend(r.write_ht, &it->blob);
+    ybg::Fixed32LEAppend((uint32_t)r.key_prefix.size(), &it->blob);
+    ybg::Fixed32LEAppend((uint32_t)r.value_body.size(), &it->blob);
+    it->blob.insert(it->blob.end(), r.key_prefix.begin(),
+                    r.key_prefix.end());
+    it->blob.insert(it->blob.end(), r.value_body.begin(),
+                    r.value_body.end());
+  }
+  *blob = it->blob.data();
+  *len = it->blob.size();
+  return 0;
+}
+
+void ybg_intents_destroy(ybg_intents_t* it) { delete it; }
+
+int ybg_merge_intents(const uint8_t* blocks, const uint64_t* offsets,
+                      uint64_t n_blocks, int kv_format,
+                      const uint8_t* intents, uint64_t intents_len,
+                      const ybg_txn_status_t* txns, uint32_t n_txns,
+                      uint8_t** out_blocks, uint64_t** out_offsets,
+                      uint64_t* out_n_blocks, uint64_t* out_total) {
+  using ybg::Buf;
+  // --- parse + resolve (ProcessIntent + the status cache lookup) ---
+  std::vector<std::pair<Buf, Buf>> resolved;  // (internal key, value)
+  {
+    uint64_t o = 0, idx = 0;
+    while (o < intents_len) {
+      if (o + 24 > intents_len) return 9;
+      uint32_t txn_id, write_id, klen, vlen;
+      uint64_t wht;
+      memcpy(&txn_id, intents + o, 4);
+      memcpy(&write_id, intents + o + 4, 4);
+      memcpy(&wht, intents + o + 8, 8);
+      memcpy(&klen, intents + o + 16, 4);
+      memcpy(&vlen, intents + o + 20, 4);
+      o += 24;
+      if (o + klen + vlen > intents_len) return 9;
+      const uint8_t* kp = intents + o;
+      const uint8_t* vp = intents + o + klen;
+      o += klen + vlen;
+      const ybg_txn_status_t* ts = nullptr;
+      for (uint32_t t = 0; t < n_txns; ++t)
+        if (txns[t].txn_id == txn_id) { ts = &txns[t]; break; }
+      if (!ts) return 9;  // unknown transaction: caller bug
+      if (ts->status != 1) continue;  // pending/aborted: invisible
+      Buf ik(kp, kp + klen);
+      ik.push_back(ybg::kHybridTimeByte);
+      ybg::DocHtAppend(ts->commit_ht, write_id, &ik);
+      ybg::Fixed64LEAppend(((ybg::kIntentSeqBase + idx) << 8) |
+                               ybg::kTypeValue,
+                           &ik);
+      Buf v;
+      v.push_back(ybg::kHybridTimeByte);
+      ybg::DocHtAppend(wht, write_id, &v);
+      v.insert(v.end(), vp, vp + vlen);
+      resolved.emplace_back(std::move(ik), std::move(v));
+      ++idx;
+    }
+  }
+  std::stable_sort(resolved.begin(), resolved.end(),
+                   [](const std::pair<Buf, Buf>& a,
+                      const std::pair<Buf, Buf>& b) {
+                     return ybg::IKeyCmp(a.first, b.first) < 0;
+                   });
+
+  // --- block assignment by first keys ---
+  std::vector<Buf> first_keys(n_blocks);
+  for (uint64_t b = 0; b < n_blocks; ++b) {
+    Buf val;
+    const uint8_t* blk = blocks + offsets[b];
+    size_t sz = offsets[b + 1] - offsets[b];
+    if (sz < 8) return 3;
+    uint32_t nr;
+    memcpy(&nr, blk + sz - 4, 4);
+    if (nr == 0 || (uint64_t)nr * 4 + 4 > sz) return 3;
+    const uint8_t* lim = blk + sz - 4 - (uint64_t)nr * 4;
+    if (!ybg::DecodeEntryHost(kv_format, blk, lim, &first_keys[b], &val))
+      return 3;
+  }
+  std::vector<std::vector<size_t>> per_block(n_blocks);
+  for (size_t i = 0; i < resolved.size(); ++i) {
+    // last block whose first key <= intent key (upper_bound - 1)
+    uint64_t lo = 0, hi = n_blocks;
+    while (lo < hi) {
+      uint64_t mid = (lo + hi) / 2;
+      if (ybg::IKeyCmp(first_keys[mid], resolved[i].first) <= 0)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    uint64_t b = lo > 0 ? lo - 1 : 0;
+    per_block[b].push_back(i);
+  }
+
+  // --- rebuild affected blocks, copy the rest ---
+  Buf out;
+  std::vector<uint64_t> ooff;
+  ooff.push_back(0);
+  for (uint64_t b = 0; b < n_blocks; ++b) {
+    const uint8_t* blk = blocks + offsets[b];
+    size_t sz = offsets[b + 1] - offsets[b];
+    if (per_block[b].empty()) {
+      out.insert(out.end(), blk, blk + sz);
+      ooff.push_back(out.size());
+      continue;
+    }
+    std::vector<std::pair<Buf, Buf>> entries;
+    int rc = ybg::DecodeBlockHost(blk, sz, kv_format, &entries);
+    if (rc) return rc;
+    ybg::BlockBuilder bb(16, (ybg_kv_format_t)kv_format);
+    size_t ei = 0;
+    size_t ii = 0;
+    const auto& ids = per_block[b];
+    while (ei < entries.size() || ii < ids.size()) {
+      bool take_intent;
+      if (ei >= entries.size()) take_intent = true;
+      else if (ii >= ids.size()) take_intent = false;
+      else
+        take_intent =
+            ybg::IKeyCmp(resolved[ids[ii]].first, entries[ei].first) < 0;
+      const auto& e = take_intent ? resolved[ids[ii]] : entries[ei];
+      bb.Add(e.first.data(), e.first.size(), e.second.data(),
+             e.second.size());
+      if (take_intent) ++ii; else ++ei;
+    }
+    bb.Finish();
+    out.insert(out.end(), bb.data().begin(), bb.data().end());
+    ooff.push_back(out.size());
+  }
+  uint8_t* ob = (uint8_t*)malloc(out.size() ? out.size() : 1);
+  memcpy(ob, out.data(), out.size());
+  uint64_t* oo = (uint64_t*)malloc(ooff.size() * 8);
+  memcpy(oo, ooff.data(), ooff.size() * 8);
+  *out_blocks = ob;
+  *out_offsets = oo;
+  *out_n_blocks = n_blocks;
+  *out_total = out.size();
+  return 0;
+}
+
+}  // extern "C"
