@@ -172,6 +172,16 @@ typedef struct {
   /* GROUP BY (config #5 "GROUP-BY-key partial aggregates"):
    * 0 = plain aggregates, else 1 + value-column index to group on. */
   int32_t group_col;
+  /* Scan direction (doc_rowwise_iterator.cc:690-818 FetchNextImpl is
+   * direction-templated; SkipFutureRecords<Direction::kBackward>,
+   * intent_aware_iterator.cc:1319ff). On this engine the bandwidth-bound
+   * full scan is direction-NEUTRAL (every block is scanned in parallel
+   * and visibility picks the same newest version either way); backward is
+   * a DELIVERY-ORDER property of the boundary: rows come back in
+   * descending DocKey order, row_limit pages deliver the highest keys
+   * first, and the paging state becomes an EXCLUSIVE upper bound for the
+   * resumed scan. Aggregate results are unaffected by direction. */
+  int32_t backward;
 } ybg_scan_spec_t;
 
 /* ---- scan handle --------------------------------------------------------- */

@@ -80,6 +80,7 @@ class ScanSpec(C.Structure):
         ("emit_rows", C.c_int32),
         ("row_limit", C.c_uint64),
         ("group_col", C.c_int32),  # 0 = none, else 1 + value column index
+        ("backward", C.c_int32),   # descending delivery order
     ]
 
 

@@ -30,12 +30,19 @@ int GpuDocRowwiseIterator::MaterializeBatch() {
   int rc = yb_gpu_scan_next_batch(handle_, &batch_);
   if (rc) return rc;
   // Restore tablet key order: the kernels emit rows in completion order
-  // with a scan-position sort key (interval, entry offset).
+  // with a scan-position sort key (interval, entry offset). Backward
+  // scans deliver in DESCENDING key order (FetchNextImpl<kBackward>,
+  // doc_rowwise_iterator.cc:690-818) — the parallel scan itself is
+  // direction-neutral, direction is a delivery property here.
   order_.resize(batch_.n_rows);
   for (uint64_t i = 0; i < batch_.n_rows; ++i) order_[i] = i;
   const uint64_t* sk = batch_.sort_key;
-  std::sort(order_.begin(), order_.end(),
-            [sk](uint64_t a, uint64_t b) { return sk[a] < sk[b]; });
+  if (spec_.backward)
+    std::sort(order_.begin(), order_.end(),
+              [sk](uint64_t a, uint64_t b) { return sk[a] > sk[b]; });
+  else
+    std::sort(order_.begin(), order_.end(),
+              [sk](uint64_t a, uint64_t b) { return sk[a] < sk[b]; });
   pos_ = 0;
   batch_ready_ = true;
   return 0;
@@ -106,7 +113,11 @@ int GpuDocRowwiseIterator::PagingState(uint8_t* key_out, size_t cap,
                                        size_t* len) {
   *len = 0;
   if (!batch_ready_ || !limit_ || limit_ >= batch_.n_rows) return 0;
-  size_t n = EncodeRowKeyDyn(order_[limit_]);
+  // forward: the first undelivered key resumes as an INCLUSIVE lower
+  // bound; backward: the LAST DELIVERED key resumes as the EXCLUSIVE
+  // upper bound of the next (still backward) page
+  size_t n = EncodeRowKeyDyn(
+      spec_.backward ? order_[limit_ - 1] : order_[limit_]);
   if (!n) return -1;        // row key failed to encode (corrupt state)
   if (n > cap) return -1;   // explicit error, never a silent empty state
   memcpy(key_out, keybuf_.data(), n);
@@ -137,9 +148,10 @@ int GpuDocRowwiseIterator::SeekTuple(const uint8_t* dockey, size_t len) {
     int rc = MaterializeBatch();
     if (rc) return -rc;
   }
-  // rows are sorted by tablet key order: binary search on encoded keys
-  // (keys encode into the growable keybuf_ — long string key columns must
-  // not truncate, or the search order breaks)
+  // rows are sorted by tablet key order (descending for backward scans):
+  // binary search on encoded keys (keys encode into the growable keybuf_
+  // — long string key columns must not truncate, or the order breaks)
+  const int dir = spec_.backward ? -1 : 1;
   uint64_t lo = 0, hi = batch_.n_rows;
   while (lo < hi) {
     uint64_t mid = (lo + hi) / 2;
@@ -147,7 +159,7 @@ int GpuDocRowwiseIterator::SeekTuple(const uint8_t* dockey, size_t len) {
     if (!n) return -1;
     int cmp = memcmp(keybuf_.data(), dockey, n < len ? n : len);
     if (cmp == 0) cmp = (n < len) ? -1 : (n > len ? 1 : 0);
-    if (cmp < 0) lo = mid + 1;
+    if (dir * cmp < 0) lo = mid + 1;
     else hi = mid;
   }
   if (lo >= batch_.n_rows) return 1;
