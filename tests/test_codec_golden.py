@@ -280,3 +280,136 @@ def test_block_roundtrip_pathological_keys():
     for fmt in (y.ENC_SHARED_PREFIX, y.ENC_THREE_SHARED_PARTS):
         got = _roundtrip_block(fmt, entries)
         assert got == expect, f"fmt={fmt}"
+
+
+# ---------------------------------------------------------------------------
+# Hand-derived three_shared_parts golden bytes. Every expected byte below is
+# derived BY HAND from the reference's encoding definition
+# (rocksdb/table/block_builder_internal.h:100-239 EncodeThreeSharedPartsSizes,
+# block_builder.cc:118-262 CalculateComponents/MaxSharedRun), independently
+# of both the generator and the oracle — pinning the encoder's component
+# CHOICES (maximal shared run, left/right preference, last8 reuse/+0x100)
+# with literal bytes instead of round-trips.
+# ---------------------------------------------------------------------------
+
+def _raw_builder_block(entries):
+    import ctypes as C
+    lib = y.product()
+    create = y._sig(lib, "ybg_builder_create", C.c_void_p,
+                    [C.POINTER(y.Schema), C.c_int, C.c_size_t, C.c_int])
+    add_raw = y._sig(lib, "ybg_builder_add_raw", C.c_int,
+                     [C.c_void_p, C.POINTER(C.c_uint8), C.c_size_t,
+                      C.c_uint64, C.POINTER(C.c_uint8), C.c_uint64])
+    fin = y._sig(lib, "ybg_builder_finish", C.c_int,
+                 [C.c_void_p, C.POINTER(C.POINTER(C.c_uint8)),
+                  C.POINTER(C.POINTER(C.c_uint64)), C.POINTER(C.c_uint64),
+                  C.POINTER(C.c_uint64), C.POINTER(C.c_uint64)])
+    schema = y.make_schema([y.KT_INT64], [(10, y.T_INT64, 1)])
+    h = create(C.byref(schema), y.ENC_THREE_SHARED_PARTS, 1 << 20, 16)
+    for ukey, seq, val in entries:
+        kb = (C.c_uint8 * len(ukey)).from_buffer_copy(ukey)
+        vb = (C.c_uint8 * len(val)).from_buffer_copy(val)
+        assert add_raw(h, kb, len(ukey), seq, vb, len(val)) == 0
+    data = C.POINTER(C.c_uint8)()
+    offs = C.POINTER(C.c_uint64)()
+    nb = C.c_uint64()
+    tot = C.c_uint64()
+    ne = C.c_uint64()
+    assert fin(h, C.byref(data), C.byref(offs), C.byref(nb), C.byref(tot),
+               C.byref(ne)) == 0
+    assert nb.value == 1
+    return bytes(data[0:tot.value])
+
+
+def _le64(v):
+    return v.to_bytes(8, "little")
+
+
+def test_three_shared_parts_hand_derived_bytes():
+    """Restart (case 2.0), 2.1.1 with ns2=0, FREQUENT, and the general
+    2.1.2 form with an ns1 delta — literal expected bytes."""
+    blk = _raw_builder_block([
+        (b"ABCDEFGHI", 0x50, b"v0"),
+        (b"ABCDEFGHJ", 0x51, b"v1"),       # last8 = prev + 0x100 (inc)
+        (b"ABCDXFGHK", 0x52, b"v2"),       # frequent: ns1=1 mid="FGH" ns2=1
+        (b"ABQRSTUVWXYZ", 0x52, b"v3"),    # general: ns1=10 prev_ns1=7 d1=3
+    ])
+    want = b"".join([
+        # e0 restart: e1=len(v0)<<2=0x08; no-reuse e2=key_size<<1 with
+        # key_size=17 (9 user + 8 suffix) -> 0x22; full key; value
+        bytes([0x08, 0x22]) + b"ABCDEFGHI" + _le64((0x50 << 8) | 1) + b"v0",
+        # e1 2.1.1: sp=8 ("ABCDEFGH"), lhs rest "I" vs rhs rest "J":
+        # no shared run -> ns1=1, ns2=0, deltas 0; last8 incremented.
+        # e1=(2<<2)|inc(2)=0x0A; e2=0b01|ns1<<3=0x09; leb(sp)=0x08; "J"
+        bytes([0x0A, 0x09, 0x08]) + b"J" + b"v1",
+        # e2 FREQUENT: sp=4, prev rest "EFGHJ" vs "XFGHK": max shared run
+        # "FGH" from the left -> ns1=1 ("X"), middle=3, ns2=1 ("K"),
+        # deltas 0, last8 incremented again.
+        # e1=(2<<2)|2|1=0x0B; leb(sp)=0x04; "X"; "K"
+        bytes([0x0B, 0x04]) + b"XK" + b"v2",
+        # e3 GENERAL 2.1.2: sp=2, lhs rest (minus reused last8)
+        # "CDXFGHK" (7) vs rhs "QRSTUVWXYZ" (10): no shared run ->
+        # ns1=10, prev_ns1=7 -> ns1_delta=3; ns2=0; last8 reused equal
+        # (same seq). e1=2<<2=0x08; e2=0b11|reuse<<2|d1<<3=0x0F;
+        # leb(ns1)=0x0A; svarint(+3)=0x83; leb(sp)=0x02; 10 key bytes
+        bytes([0x08, 0x0F, 0x0A, 0x83, 0x02]) + b"QRSTUVWXYZ" + b"v3",
+    ])
+    assert blk[:len(want)] == want, (blk[:len(want)].hex(), want.hex())
+    # trailer: one restart at 0
+    assert blk[-8:] == (0).to_bytes(4, "little") + (1).to_bytes(4, "little")
+    # and the oracle's decoder reconstructs the exact keys (independent
+    # restatement of block.cc:287-346)
+    _assert_oracle_roundtrip(blk, [
+        b"ABCDEFGHI" + _le64((0x50 << 8) | 1),
+        b"ABCDEFGHJ" + _le64((0x51 << 8) | 1),
+        b"ABCDXFGHK" + _le64((0x52 << 8) | 1),
+        b"ABQRSTUVWXYZ" + _le64((0x52 << 8) | 1),
+    ], [b"v0", b"v1", b"v2", b"v3"])
+
+
+def test_three_shared_parts_d2_case_bytes():
+    """Case 2.1.1 with ns2_delta == 1 (the d2 bit): the new key's second
+    non-shared part is one byte longer than the previous key's."""
+    blk = _raw_builder_block([
+        (b"AAEFGHJ", 0x60, b"w0"),
+        (b"AAWFGHKL", 0x61, b"w1"),
+    ])
+    want = b"".join([
+        # restart: key_size = 15 -> e2 = 0x1E
+        bytes([0x08, 0x1E]) + b"AAEFGHJ" + _le64((0x60 << 8) | 1) + b"w0",
+        # sp=2; lhs "EFGHJ" vs rhs "WFGHKL": left run "FGH" (len 3) at
+        # offset 1 -> ns1=1 ("W"), middle=3, prev_ns2=1 ("J"),
+        # ns2=2 ("KL") -> d2=1; inc. e1=(2<<2)|2=0x0A;
+        # e2=0b01|d2<<2|ns1<<3|ns2<<6=0x8D; leb(sp)=0x02
+        bytes([0x0A, 0x8D, 0x02]) + b"W" + b"KL" + b"w1",
+    ])
+    assert blk[:len(want)] == want, (blk[:len(want)].hex(), want.hex())
+    _assert_oracle_roundtrip(blk, [
+        b"AAEFGHJ" + _le64((0x60 << 8) | 1),
+        b"AAWFGHKL" + _le64((0x61 << 8) | 1),
+    ], [b"w0", b"w1"])
+
+
+def _assert_oracle_roundtrip(blk, keys, values):
+    import ctypes as C
+    lib = y.product()
+    f = y._sig(lib, "ybg_decode_block", C.c_int,
+               [C.POINTER(C.c_uint8), C.c_uint64, C.c_int,
+                C.POINTER(C.c_uint8), C.c_uint64, C.POINTER(C.c_uint32),
+                C.POINTER(C.c_uint8), C.c_uint64, C.POINTER(C.c_uint32),
+                C.c_uint64, C.POINTER(C.c_uint64)])
+    data = (C.c_uint8 * len(blk)).from_buffer_copy(blk)
+    kb = (C.c_uint8 * 4096)()
+    vb = (C.c_uint8 * 4096)()
+    kl = (C.c_uint32 * 64)()
+    vl = (C.c_uint32 * 64)()
+    n = C.c_uint64()
+    assert f(data, len(blk), y.ENC_THREE_SHARED_PARTS, kb, 4096, kl, vb,
+             4096, vl, 64, C.byref(n)) == 0
+    assert n.value == len(keys)
+    ko = vo = 0
+    for i, (k, v) in enumerate(zip(keys, values)):
+        assert bytes(kb[ko:ko + kl[i]]) == k, i
+        assert bytes(vb[vo:vo + vl[i]]) == v, i
+        ko += kl[i]
+        vo += vl[i]

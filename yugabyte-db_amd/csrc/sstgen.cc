@@ -1209,4 +1209,32 @@ int ybg_merge_intents(const uint8_t* blocks, const uint64_t* offsets,
   return 0;
 }
 
+
+/* Test hook: decode a data block's entries with the host decoder the
+ * intent merge uses (DecodeBlockHost). Flattened output:
+ * keys/values concatenated + per-entry lengths. */
+int ybg_decode_block(const uint8_t* blk, uint64_t size, int kv_format,
+                     uint8_t* keys_out, uint64_t keys_cap,
+                     uint32_t* key_lens, uint8_t* vals_out,
+                     uint64_t vals_cap, uint32_t* val_lens,
+                     uint64_t cap_entries, uint64_t* n_entries) {
+  std::vector<std::pair<ybg::Buf, ybg::Buf>> es;
+  int rc = ybg::DecodeBlockHost(blk, size, kv_format, &es);
+  if (rc) return rc;
+  uint64_t ko = 0, vo = 0;
+  *n_entries = es.size();
+  for (size_t i = 0; i < es.size() && i < cap_entries; ++i) {
+    if (ko + es[i].first.size() > keys_cap ||
+        vo + es[i].second.size() > vals_cap)
+      return 8;
+    memcpy(keys_out + ko, es[i].first.data(), es[i].first.size());
+    memcpy(vals_out + vo, es[i].second.data(), es[i].second.size());
+    key_lens[i] = (uint32_t)es[i].first.size();
+    val_lens[i] = (uint32_t)es[i].second.size();
+    ko += es[i].first.size();
+    vo += es[i].second.size();
+  }
+  return 0;
+}
+
 }  // extern "C"
