@@ -127,6 +127,17 @@ typedef enum {
    *   [tuples: ncols x 8-byte LE datum patterns each]
    * is_key_col must be 1; col is ignored. */
   YBG_PRED_IN_TUPLE,
+  /* Option RANGES over a NUMERIC column (key or value) — the reference's
+   * mixed bound options (docdb/hybrid_scan_choices.h:43-77 OptionRange):
+   * bytes = n x 24-byte records
+   *   [u64 LE lo][u64 LE hi][u32 LE flags: bit0 lo incl, bit1 hi incl]
+   *   [u32 pad]
+   * true when the column datum falls in ANY range. Numeric bounds use
+   * the column dtype's bit pattern (int compares signed, double/float
+   * as FP). Range options on the leading range-key column of a
+   * range-sharded table also PRUNE the scanned block set (see
+   * yb_gpu_scan_feed_blocks). */
+  YBG_PRED_IN_RANGE,
 } ybg_pred_op_t;
 
 typedef struct {
@@ -376,6 +387,16 @@ int ybg_builder_finish(ybg_builder_t *b, const uint8_t **data,
 int ybg_builder_finish_sst(ybg_builder_t *b, const uint8_t **data,
                            uint64_t *total_bytes, uint64_t *n_blocks,
                            uint64_t *n_entries);
+
+/* Host block-entry decode + first-key helpers (the intent merge's and
+ * block pruning's decoder; exposed for CPU tests). */
+int ybg_decode_block(const uint8_t *blk, uint64_t size, int kv_format,
+                     uint8_t *keys_out, uint64_t keys_cap,
+                     uint32_t *key_lens, uint8_t *vals_out,
+                     uint64_t vals_cap, uint32_t *val_lens,
+                     uint64_t cap_entries, uint64_t *n_entries);
+int ybg_block_first_key(const uint8_t *blk, uint64_t size, int kv_format,
+                        uint8_t *out, uint64_t cap, uint64_t *len);
 
 /* Parse an SST file and report its data-block (offset, size) handles —
  * the exact parser yb_gpu_scan_feed_sst uses (exposed for CPU tests). */

@@ -215,6 +215,11 @@ typedef enum {
    * (hybrid_scan_choices.h:43-77); bytes = [u32 ncols][u32 colidx x n]
    * [tuples of n x 8-byte LE datums] */
   ORCL_PRED_IN_TUPLE,
+  /* option RANGES over a numeric column (hybrid_scan_choices.h
+   * OptionRange / mixed bound options): bytes = n x 24-byte records
+   * [u64 lo][u64 hi][u32 flags: bit0 lo inclusive, bit1 hi inclusive]
+   * [u32 pad]; true when the datum falls in ANY range. */
+  ORCL_PRED_IN_RANGE,
 } orcl_pred_op_t;
 
 typedef struct {

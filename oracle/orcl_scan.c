@@ -519,6 +519,31 @@ static int pred_eval(const orcl_pred_t *pr, const orcl_schema_t *sc,
     }
     return 0;
   }
+  if (pr->op == ORCL_PRED_IN_RANGE) {
+    /* option ranges (hybrid_scan_choices.h:43-77 OptionRange) */
+    size_t n = pr->bytes_len / 24, i;
+    if (numeric == 0) return 0; /* numeric columns only (open rejects) */
+    for (i = 0; i < n; ++i) {
+      uint64_t lo, hi;
+      uint32_t fl;
+      memcpy(&lo, pr->bytes + 24 * i, 8);
+      memcpy(&hi, pr->bytes + 24 * i + 8, 8);
+      memcpy(&fl, pr->bytes + 24 * i + 16, 4);
+      if (numeric == 1) {
+        int okl = (fl & 1) ? lhs_i >= (int64_t)lo : lhs_i > (int64_t)lo;
+        int okh = (fl & 2) ? lhs_i <= (int64_t)hi : lhs_i < (int64_t)hi;
+        if (okl && okh) return 1;
+      } else {
+        double l, hgh;
+        memcpy(&l, &lo, 8);
+        memcpy(&hgh, &hi, 8);
+        int okl = (fl & 1) ? lhs_f >= l : lhs_f > l;
+        int okh = (fl & 2) ? lhs_f <= hgh : lhs_f < hgh;
+        if (okl && okh) return 1;
+      }
+    }
+    return 0;
+  }
   if (numeric == 1) cmp = lhs_i < rhs_i ? -1 : (lhs_i > rhs_i ? 1 : 0);
   else if (numeric == 2) cmp = lhs_f < rhs_f ? -1 : (lhs_f > rhs_f ? 1 : 0);
   switch (pr->op) {
@@ -530,6 +555,7 @@ static int pred_eval(const orcl_pred_t *pr, const orcl_schema_t *sc,
     case ORCL_PRED_NE: return cmp != 0;
     case ORCL_PRED_IN: return 0;       /* handled above */
     case ORCL_PRED_IN_TUPLE: return 0; /* handled above */
+    case ORCL_PRED_IN_RANGE: return 0; /* handled above */
   }
   return 0;
 }

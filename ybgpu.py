@@ -25,7 +25,7 @@ T_UINT32, T_UINT64, T_FLOAT, T_DOUBLE, T_STRING = 5, 6, 7, 8, 9
 KT_INT64, KT_INT32, KT_STRING = 0, 1, 2
 ENC_SHARED_PREFIX, ENC_THREE_SHARED_PARTS = 0, 1
 PRED_GT, PRED_GE, PRED_LT, PRED_LE, PRED_EQ, PRED_NE, PRED_IN, \
-    PRED_IN_TUPLE = range(8)
+    PRED_IN_TUPLE, PRED_IN_RANGE = range(9)
 (AGG_COUNT, AGG_COUNT_STAR, AGG_SUM_INT64, AGG_SUM_DOUBLE,
  AGG_MIN_INT64, AGG_MAX_INT64, AGG_MIN_DOUBLE, AGG_MAX_DOUBLE) = range(8)
 
@@ -944,3 +944,27 @@ def orcl_scan_intents(data, offsets, n_blocks, schema, spec, intents_blob,
     if rc != 0:
         raise RuntimeError(f"orcl_scan_intents rc={rc}")
     return res
+
+
+def encode_dockey(schema, hash_=0, key_datums=(), key_strs=()):
+    """Encode a DocKey from key-column datums (ybg_encode_dockey) —
+    bounds / paging-state construction helper."""
+    lib = product()
+    f = _sig(lib, "ybg_encode_dockey", C.c_size_t,
+             [C.POINTER(Schema), C.POINTER(Key), C.POINTER(C.c_uint8),
+              C.c_size_t])
+    k = Key()
+    k.hash = hash_
+    keep = []
+    for i, d in enumerate(key_datums):
+        k.datums[i] = d & 0xFFFFFFFFFFFFFFFF
+    for i, s in enumerate(key_strs):
+        if s is not None:
+            buf = C.create_string_buffer(s, len(s))
+            keep.append(buf)
+            k.strs[i] = C.cast(buf, C.POINTER(C.c_uint8))
+            k.str_lens[i] = len(s)
+    out = (C.c_uint8 * 512)()
+    n = f(C.byref(schema), C.byref(k), out, 512)
+    assert n > 0
+    return bytes(out[:n])

@@ -1102,6 +1102,37 @@ DEV bool pred_compare(const PredC& pr, uint64_t datum, const uint8_t* sptr,
                       uint32_t slen, const uint8_t* aux) {
   int dtype = (int)(pr.opdt >> 8);
   int cmp;
+  if (YBG_UNLIKELY((pr.opdt & 0xff) == YBG_PRED_IN_RANGE)) {
+    // option ranges (hybrid_scan_choices.h:43-77 OptionRange): n 24-byte
+    // records; datum packs (count << 32) | aux offset
+    uint32_t n = (uint32_t)(pr.datum >> 32);
+    const uint8_t* rec = aux + (uint32_t)pr.datum;
+    for (uint32_t i = 0; i < n; ++i, rec += 24) {
+      uint64_t lo = load_u64_una(rec);
+      uint64_t hi = load_u64_una(rec + 8);
+      uint32_t fl = (uint32_t)load_u64_una(rec + 16);
+      bool okl, okh;
+      if (dtype == YBG_T_DOUBLE) {
+        double a = __longlong_as_double((long long)datum);
+        double l = __longlong_as_double((long long)lo);
+        double h = __longlong_as_double((long long)hi);
+        okl = (fl & 1) ? a >= l : a > l;
+        okh = (fl & 2) ? a <= h : a < h;
+      } else if (dtype == YBG_T_FLOAT) {
+        float a = __uint_as_float((uint32_t)datum);
+        float l = __uint_as_float((uint32_t)lo);
+        float h = __uint_as_float((uint32_t)hi);
+        okl = (fl & 1) ? a >= l : a > l;
+        okh = (fl & 2) ? a <= h : a < h;
+      } else {
+        int64_t a = (int64_t)datum;
+        okl = (fl & 1) ? a >= (int64_t)lo : a > (int64_t)lo;
+        okh = (fl & 2) ? a <= (int64_t)hi : a < (int64_t)hi;
+      }
+      if (okl && okh) return true;
+    }
+    return false;
+  }
   if (YBG_UNLIKELY((pr.opdt & 0xff) == YBG_PRED_IN)) {
     // membership over the option list (hybrid_scan_choices.h:43-60).
     // datum packs (count << 32) | aux offset. Numeric options are 8-byte
@@ -1511,6 +1542,22 @@ DEV bool eval_key_preds(const DevSpec& sp, const uint8_t* rk, uint32_t rk_len,
       bool hit = false;
       for (uint32_t k2 = 0; k2 < n; ++k2)
         if (d == load_u64_una(aux + pr.rhs_off + 8ull * k2)) hit = true;
+      if (!hit) return false;
+      continue;
+    } else if (pr.op == YBG_PRED_IN_RANGE) {
+      // option ranges on a numeric key column
+      uint32_t n = pr.str_len / 24;
+      bool hit = false;
+      int64_t a = (int64_t)d;
+      for (uint32_t k2 = 0; k2 < n && !hit; ++k2) {
+        const uint8_t* rec = aux + pr.rhs_off + 24ull * k2;
+        int64_t lo = (int64_t)load_u64_una(rec);
+        int64_t hi2 = (int64_t)load_u64_una(rec + 8);
+        uint32_t fl = (uint32_t)load_u64_una(rec + 16);
+        bool okl = (fl & 1) ? a >= lo : a > lo;
+        bool okh = (fl & 2) ? a <= hi2 : a < hi2;
+        hit = okl && okh;
+      }
       if (!hit) return false;
       continue;
     } else {
@@ -3155,7 +3202,9 @@ inline void build_dev_spec(const ybg_scan_spec_t* spec, DevSpec* dp,
     }
     int dt = d.cols[pr.col].dtype;
     pc.opdt = (uint32_t)pr.op | ((uint32_t)dt << 8);
-    if (pr.op == YBG_PRED_IN) {
+    if (pr.op == YBG_PRED_IN_RANGE) {
+      pc.datum = (((uint64_t)(pr.str_len / 24) << 32) | pr.rhs_off);
+    } else if (pr.op == YBG_PRED_IN) {
       uint64_t cnt;
       if (dt == YBG_T_STRING) {
         // [u32 len][bytes] records

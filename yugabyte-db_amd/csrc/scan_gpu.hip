@@ -875,6 +875,16 @@ int yb_gpu_scan_open(const ybg_scan_spec_t* spec, ybg_scan_t** out) {
       }
       continue;
     }
+    if (p.op == YBG_PRED_IN_RANGE) {
+      bool is_str = p.is_key_col
+                        ? spec->schema.key_types[p.col] == YBG_KT_STRING
+                        : spec->schema.value_cols[p.col].dtype ==
+                              YBG_T_STRING;
+      if (is_str || !p.bytes || p.bytes_len == 0 || p.bytes_len % 24)
+        return set_err(9, "IN_RANGE needs a numeric column and n x "
+                          "24-byte range records");
+      continue;
+    }
     if (p.op != YBG_PRED_IN) continue;
     bool is_str = p.is_key_col
                       ? spec->schema.key_types[p.col] == YBG_KT_STRING
@@ -920,9 +930,210 @@ int yb_gpu_scan_open(const ybg_scan_spec_t* spec, ybg_scan_t** out) {
   return 0;
 }
 
+namespace {
+
+// ---------------------------------------------------------------------
+// Index-guided block pruning (docdb/hybrid_scan_choices.cc seek plans +
+// rocksdb index-based block selection, index_reader.cc): a bounded or
+// option-constrained scan only needs the blocks whose key range
+// intersects the allowed key set. Block separators = each block's first
+// internal key (the role the SST index keys play in the reference).
+// Pruning is a strict superset selection: the per-row bound checks and
+// option FILTERS still run, so a conservative selection cannot change
+// results — parity tests pin it against unpruned scans and the oracle.
+// ---------------------------------------------------------------------
+
+struct KeyRange {
+  std::vector<uint8_t> lo;  // inclusive prefix; empty = -inf
+  std::vector<uint8_t> hi;  // exclusive prefix; empty = +inf
+};
+
+// encoded key-prefix successor: the smallest byte string > p as a prefix
+std::vector<uint8_t> prefix_succ(std::vector<uint8_t> p) {
+  while (!p.empty()) {
+    if (p.back() != 0xff) {
+      p.back() += 1;
+      return p;
+    }
+    p.pop_back();
+  }
+  return p;  // empty = +inf
+}
+
+void enc_key_int(int kt, uint64_t v, std::vector<uint8_t>* out) {
+  if (kt == YBG_KT_INT64) {
+    out->push_back(0x49);  // kInt64B
+    uint64_t u = v ^ 0x8000000000000000ull;
+    for (int i = 7; i >= 0; --i) out->push_back((uint8_t)(u >> (8 * i)));
+  } else {
+    out->push_back(0x48);  // kInt32B
+    uint32_t u = (uint32_t)v ^ 0x80000000u;
+    for (int i = 3; i >= 0; --i) out->push_back((uint8_t)(u >> (8 * i)));
+  }
+}
+
+// prefix vs full-key compare: memcmp over min length; a proper prefix
+// sorts first (all keys starting with it come after or at it)
+int pfx_cmp(const std::vector<uint8_t>& a, const uint8_t* b, size_t blen) {
+  size_t n = a.size() < blen ? a.size() : blen;
+  int c = memcmp(a.data(), b, n);
+  if (c) return c;
+  return a.size() == blen ? 0 : (a.size() < blen ? -1 : 1);
+}
+
+// Returns true (and fills keep) when the spec allows pruning AND at least
+// one block can be skipped.
+bool compute_block_selection(const ybg_scan_spec_t& spec,
+                             const uint8_t* blocks, const uint64_t* offsets,
+                             uint64_t n_blocks,
+                             std::vector<uint8_t>* keep) {
+  std::vector<KeyRange> allowed;
+  KeyRange bounds;
+  bool have_bounds = false;
+  if (spec.lower_bound_len) {
+    bounds.lo.assign(spec.lower_bound,
+                     spec.lower_bound + spec.lower_bound_len);
+    have_bounds = true;
+  }
+  if (spec.upper_bound_len) {
+    bounds.hi.assign(spec.upper_bound,
+                     spec.upper_bound + spec.upper_bound_len);
+    have_bounds = true;
+  }
+  // leading-range-key options on range-sharded tables: derive key-prefix
+  // ranges from IN / IN_RANGE on key column 0 (a seek plan the reference
+  // builds in HybridScanChoices)
+  std::vector<KeyRange> opts;
+  if (!spec.schema.has_hash &&
+      (spec.schema.key_types[0] == YBG_KT_INT64 ||
+       spec.schema.key_types[0] == YBG_KT_INT32)) {
+    for (int i = 0; i < spec.num_preds; ++i) {
+      const ybg_pred_t& p = spec.preds[i];
+      if (!p.is_key_col || p.col != 0 || !p.bytes) continue;
+      if (p.op == YBG_PRED_IN) {
+        for (uint64_t o = 0; o + 8 <= p.bytes_len; o += 8) {
+          uint64_t v;
+          memcpy(&v, p.bytes + o, 8);
+          KeyRange r;
+          enc_key_int(spec.schema.key_types[0], v, &r.lo);
+          r.hi = prefix_succ(r.lo);
+          opts.push_back(std::move(r));
+        }
+      } else if (p.op == YBG_PRED_IN_RANGE) {
+        for (uint64_t o = 0; o + 24 <= p.bytes_len; o += 24) {
+          uint64_t lo, hi;
+          memcpy(&lo, p.bytes + o, 8);
+          memcpy(&hi, p.bytes + o + 8, 8);
+          KeyRange r;  // conservative: inclusive both ends as prefixes
+          enc_key_int(spec.schema.key_types[0], lo, &r.lo);
+          std::vector<uint8_t> h;
+          enc_key_int(spec.schema.key_types[0], hi, &h);
+          r.hi = prefix_succ(h);
+          opts.push_back(std::move(r));
+        }
+      }
+      if (!opts.empty()) break;  // one option pred drives the seek plan
+    }
+  }
+  if (!have_bounds && opts.empty()) return false;
+  if (opts.empty()) {
+    allowed.push_back(std::move(bounds));
+  } else {
+    // intersect every option range with the scan bounds
+    for (auto& r : opts) {
+      KeyRange c = r;
+      if (!bounds.lo.empty() &&
+          (c.lo.empty() || c.lo < bounds.lo))
+        c.lo = bounds.lo;
+      if (!bounds.hi.empty() &&
+          (c.hi.empty() || bounds.hi < c.hi))
+        c.hi = bounds.hi;
+      allowed.push_back(std::move(c));
+    }
+  }
+
+  // block separators: first internal key per block
+  std::vector<std::vector<uint8_t>> fk(n_blocks);
+  for (uint64_t b = 0; b < n_blocks; ++b) {
+    uint8_t buf[256];
+    uint64_t len = 0;
+    if (ybg_block_first_key(blocks + offsets[b],
+                            offsets[b + 1] - offsets[b], spec.kv_format,
+                            buf, sizeof(buf), &len))
+      return false;  // undecodable: do not prune
+    fk[b].assign(buf, buf + len);
+  }
+  keep->assign(n_blocks, 0);
+  bool any_skip = false;
+  for (uint64_t b = 0; b < n_blocks; ++b) {
+    // block b covers [fk[b], fk[b+1]) (+inf for the last block); since a
+    // row can straddle into the next block's first interval via the tail
+    // walk only within one row, range intersection on separators is the
+    // correct granularity (rows are whole within the allowed key set)
+    bool hit = false;
+    for (const auto& r : allowed) {
+      bool lo_ok =
+          r.lo.empty() || b + 1 >= n_blocks ||
+          pfx_cmp(r.lo, fk[b + 1].data(), fk[b + 1].size()) < 0;
+      bool hi_ok = r.hi.empty() ||
+                   pfx_cmp(r.hi, fk[b].data(), fk[b].size()) > 0;
+      if (lo_ok && hi_ok) {
+        hit = true;
+        break;
+      }
+    }
+    (*keep)[b] = hit ? 1 : 0;
+    if (!hit) any_skip = true;
+  }
+  return any_skip;
+}
+
+}  // namespace
+
+// Test hook: the block selection feed_blocks applies (1 byte per block;
+// returns 1 when pruning engaged). CPU tests feed the kept subset through
+// the simulator and compare against unpruned oracle scans.
+int ybg_test_block_selection(const ybg_scan_spec_t* spec,
+                             const uint8_t* blocks, const uint64_t* offsets,
+                             uint64_t n_blocks, uint8_t* keep_out) {
+  std::vector<uint8_t> keep;
+  if (!compute_block_selection(*spec, blocks, offsets, n_blocks, &keep))
+    return 0;
+  memcpy(keep_out, keep.data(), n_blocks);
+  return 1;
+}
+
 int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
                             const uint64_t* offsets, uint64_t n_blocks,
                             int device) {
+  // index-guided pruning (host-bytes path): upload only the blocks whose
+  // key range intersects the scan's bounds / leading-key option ranges
+  std::vector<uint8_t> pruned_blob;
+  std::vector<uint64_t> pruned_off;
+  if (!device && n_blocks) {
+    std::vector<uint8_t> keep;
+    if (compute_block_selection(s->spec, blocks, offsets, n_blocks,
+                                &keep)) {
+      pruned_off.push_back(0);
+      for (uint64_t b = 0; b < n_blocks; ++b) {
+        if (!keep[b]) continue;
+        pruned_blob.insert(pruned_blob.end(), blocks + offsets[b],
+                           blocks + offsets[b + 1]);
+        pruned_off.push_back(pruned_blob.size());
+      }
+      if (pruned_off.size() <= 1) {
+        // nothing selected: feed one empty-ish view is invalid; keep a
+        // single block so the scan machinery has a well-formed table
+        // (its rows are filtered per-row anyway)
+        uint64_t b0 = 0;
+        pruned_blob.assign(blocks + offsets[b0], blocks + offsets[b0 + 1]);
+        pruned_off.push_back(pruned_blob.size());
+      }
+      blocks = pruned_blob.data();
+      offsets = pruned_off.data();
+      n_blocks = pruned_off.size() - 1;
+    }
+  }
   s->n_blocks = n_blocks;
   s->total_bytes = offsets[n_blocks];
   if (device) {

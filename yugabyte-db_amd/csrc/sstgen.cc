@@ -1237,4 +1237,23 @@ int ybg_decode_block(const uint8_t* blk, uint64_t size, int kv_format,
   return 0;
 }
 
+
+/* First internal key of a data block (the restart entry) — feed-time
+ * block pruning uses these as the block separators, the role the SST
+ * index's keys play in the reference (rocksdb/table/index_reader.cc). */
+int ybg_block_first_key(const uint8_t* blk, uint64_t size, int kv_format,
+                        uint8_t* out, uint64_t cap, uint64_t* len) {
+  if (size < 8) return 3;
+  uint32_t nr;
+  memcpy(&nr, blk + size - 4, 4);
+  if (nr == 0 || (uint64_t)nr * 4 + 4 > size) return 3;
+  const uint8_t* lim = blk + size - 4 - (uint64_t)nr * 4;
+  ybg::Buf key, val;
+  if (!ybg::DecodeEntryHost(kv_format, blk, lim, &key, &val)) return 3;
+  if (key.size() > cap) return 8;
+  memcpy(out, key.data(), key.size());
+  *len = key.size();
+  return 0;
+}
+
 }  // extern "C"
