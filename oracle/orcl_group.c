@@ -127,6 +127,20 @@ static int grp_cb(const orcl_row_t *row, void *arg) {
       case ORCL_AGG_MAX_INT64:
         if ((int64_t)d > s->vals[a]) s->vals[a] = (int64_t)d;
         break;
+      case ORCL_AGG_MIN_DOUBLE: {
+        double v, cur;
+        memcpy(&v, &d, 8);
+        memcpy(&cur, &s->vals[a], 8);
+        if (s->cnts[a] == 0 || v < cur) memcpy(&s->vals[a], &v, 8);
+        break;
+      }
+      case ORCL_AGG_MAX_DOUBLE: {
+        double v, cur;
+        memcpy(&v, &d, 8);
+        memcpy(&cur, &s->vals[a], 8);
+        if (s->cnts[a] == 0 || v > cur) memcpy(&s->vals[a], &v, 8);
+        break;
+      }
       default:
         break;
     }

@@ -576,7 +576,8 @@ def _decode_groups(schema, group_col, n, keys, vals, cnts, key_bytes,
             v = vals[g * MAX_AGGS + a]
             if c == 0:
                 row.append(None)
-            elif aggs[a].op == AGG_SUM_DOUBLE:
+            elif aggs[a].op in (AGG_SUM_DOUBLE, AGG_MIN_DOUBLE,
+                                AGG_MAX_DOUBLE):
                 row.append(struct.unpack("<d", struct.pack("<q", v))[0])
             else:
                 row.append(v)

@@ -67,6 +67,23 @@ static inline unsigned long long ybg_atomic_exch_u64(unsigned long long* p,
   *p = v;
   return old;
 }
+static inline unsigned long long ybg_atomic_min_u64(unsigned long long* p,
+                                                    unsigned long long v) {
+  unsigned long long old = *p;
+  if (v < old) *p = v;
+  return old;
+}
+static inline unsigned long long ybg_atomic_max_u64(unsigned long long* p,
+                                                    unsigned long long v) {
+  unsigned long long old = *p;
+  if (v > old) *p = v;
+  return old;
+}
+static inline unsigned ybg_atomic_or_u32(unsigned* p, unsigned v) {
+  unsigned old = *p;
+  *p |= v;
+  return old;
+}
 static inline unsigned long long ybg_atomic_load_u64(unsigned long long* p) {
   return *p;
 }
@@ -91,6 +108,18 @@ __device__ __forceinline__ long long ybg_atomic_max_i64(long long* p,
 __device__ __forceinline__ unsigned long long ybg_atomic_exch_u64(
     unsigned long long* p, unsigned long long v) {
   return atomicExch(p, v);
+}
+__device__ __forceinline__ unsigned long long ybg_atomic_min_u64(
+    unsigned long long* p, unsigned long long v) {
+  return atomicMin(p, v);
+}
+__device__ __forceinline__ unsigned long long ybg_atomic_max_u64(
+    unsigned long long* p, unsigned long long v) {
+  return atomicMax(p, v);
+}
+__device__ __forceinline__ unsigned ybg_atomic_or_u32(unsigned* p,
+                                                      unsigned v) {
+  return atomicOr(p, v);
 }
 __device__ __forceinline__ unsigned long long ybg_atomic_load_u64(
     unsigned long long* p) {
@@ -1968,6 +1997,8 @@ struct GroupCtx {
   unsigned* state;           // [cap + 1]
   long long* vals;           // [(cap + 1) * YBG_MAX_AGGS]
   unsigned long long* cnts;  // [(cap + 1) * YBG_MAX_AGGS]
+  unsigned long long* vals_hi;  // [(cap+1)*MAX_AGGS] SUM_DOUBLE high word
+  unsigned* poison;          // [(cap+1)*MAX_AGGS] non-finite/overflow flag
   uint64_t cap;              // power of two
   unsigned long long* overflow;
   const uint8_t* data;       // block data base (string exemplars)
@@ -2044,6 +2075,64 @@ DEV uint64_t grp_find_or_insert(const GroupCtx& gc, uint64_t kv,
   return ~0ull;
 }
 
+// Order-isomorphic u64 image of a double (sign-magnitude -> biased):
+// unsigned compares/min/max on the image match double ordering, so
+// grouped double MIN/MAX run as plain integer atomics — exact and
+// order-independent (deterministic). NaNs map above +inf.
+DEV uint64_t f64_ordered(uint64_t bits) {
+  return bits ^ ((bits >> 63) ? ~0ull : (1ull << 63));
+}
+DEV uint64_t f64_unordered(uint64_t u) {
+  return u ^ ((u >> 63) ? (1ull << 63) : ~0ull);
+}
+
+// Grouped double SUM accumulates in 128-bit two's-complement FIXED POINT
+// (2^-60 scaling) with carry-propagating integer atomics: bit-exact and
+// order-independent => deterministic run to run and across GPU counts,
+// unlike a float atomic-add. Values outside +-2^66 or non-finite poison
+// the slot (exported as NaN). doc_expr.cc:341-395 semantics otherwise.
+DEV void group_sum_f64_fixed(const GroupCtx& gc, uint64_t slot, int g,
+                             uint64_t dbits) {
+  double v = __longlong_as_double((long long)dbits);
+  if (!(v == v) || v > 7.3786976294838206e19 ||
+      v < -7.3786976294838206e19) {
+    ybg_atomic_or_u32(&gc.poison[slot * YBG_MAX_AGGS + g], 1u);
+    return;
+  }
+  double sc = v * 1152921504606846976.0;  // 2^60
+  __int128 f = (__int128)sc;
+  unsigned long long lo = (unsigned long long)(unsigned __int128)f;
+  unsigned long long hi =
+      (unsigned long long)((unsigned __int128)f >> 64);
+  unsigned long long* vlo =
+      (unsigned long long*)&gc.vals[slot * YBG_MAX_AGGS + g];
+  unsigned long long* vhi = &gc.vals_hi[slot * YBG_MAX_AGGS + g];
+  unsigned long long old = atomicAdd(vlo, lo);
+  if (old + lo < old) atomicAdd(vhi, 1ull);
+  atomicAdd(vhi, hi);
+}
+
+// Final per-slot value decode (shared by the device export kernel and the
+// host simulator's export).
+DEV long long group_export_value(int op, long long raw,
+                                 unsigned long long hi, unsigned poison) {
+  if (op == YBG_AGG_SUM_DOUBLE) {
+    double dv;
+    if (poison) {
+      dv = __longlong_as_double(0x7ff8000000000000ll);  // NaN
+    } else {
+      unsigned __int128 u = ((unsigned __int128)hi << 64) |
+                            (unsigned long long)raw;
+      __int128 sf = (__int128)u;
+      dv = (double)sf * 8.673617379884035e-19;  // 2^-60
+    }
+    return (long long)__double_as_longlong(dv);
+  }
+  if (op == YBG_AGG_MIN_DOUBLE || op == YBG_AGG_MAX_DOUBLE)
+    return (long long)f64_unordered((uint64_t)raw);
+  return raw;
+}
+
 // Deferred head-row record for the single-pass GROUP kernel: the head
 // row's raw operands (ownership unknown until the lane relay / cont-flag
 // resolution). agg_datum is fixed-size for a stable memory layout; the
@@ -2095,8 +2184,7 @@ DEV void group_accum(const DevSpec& sp, const GroupCtx& gc,
         atomicAdd((unsigned long long*)&v[g], d);
         break;
       case YBG_AGG_SUM_DOUBLE:
-        ybg_atomic_add_f64((double*)&v[g],
-                           __longlong_as_double((long long)d));
+        group_sum_f64_fixed(gc, slot, g, d);
         break;
       case YBG_AGG_MIN_INT64:
         ybg_atomic_min_i64(&v[g], (long long)d);
@@ -2104,8 +2192,14 @@ DEV void group_accum(const DevSpec& sp, const GroupCtx& gc,
       case YBG_AGG_MAX_INT64:
         ybg_atomic_max_i64(&v[g], (long long)d);
         break;
+      case YBG_AGG_MIN_DOUBLE:
+        ybg_atomic_min_u64((unsigned long long*)&v[g], f64_ordered(d));
+        break;
+      case YBG_AGG_MAX_DOUBLE:
+        ybg_atomic_max_u64((unsigned long long*)&v[g], f64_ordered(d));
+        break;
       default:
-        break;  // MIN/MAX double unsupported in group mode (open() rejects)
+        break;
     }
     // COUNT ops carry their contribution count in the value itself; the
     // export reconstructs cnt = val for them (one fewer atomic per row)
@@ -2154,14 +2248,19 @@ DEV void group_accum_rec(const DevSpec& sp, const GroupCtx& gc,
         atomicAdd((unsigned long long*)&v[g], d);
         break;
       case YBG_AGG_SUM_DOUBLE:
-        ybg_atomic_add_f64((double*)&v[g],
-                           __longlong_as_double((long long)d));
+        group_sum_f64_fixed(gc, slot, g, d);
         break;
       case YBG_AGG_MIN_INT64:
         ybg_atomic_min_i64(&v[g], (long long)d);
         break;
       case YBG_AGG_MAX_INT64:
         ybg_atomic_max_i64(&v[g], (long long)d);
+        break;
+      case YBG_AGG_MIN_DOUBLE:
+        ybg_atomic_min_u64((unsigned long long*)&v[g], f64_ordered(d));
+        break;
+      case YBG_AGG_MAX_DOUBLE:
+        ybg_atomic_max_u64((unsigned long long*)&v[g], f64_ordered(d));
         break;
       default:
         break;

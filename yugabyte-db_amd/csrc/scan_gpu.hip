@@ -349,8 +349,14 @@ __global__ void k_group_init(DevSpec sp, GroupCtx gc) {
       if (sp.aggs[g].op == YBG_AGG_MIN_INT64) init = 0x7fffffffffffffffll;
       else if (sp.aggs[g].op == YBG_AGG_MAX_INT64)
         init = (long long)0x8000000000000000ll;
+      else if (sp.aggs[g].op == YBG_AGG_MIN_DOUBLE)
+        init = (long long)~0ull;  // ordered-u64 domain max
+      else if (sp.aggs[g].op == YBG_AGG_MAX_DOUBLE)
+        init = 0;                 // ordered-u64 domain min
       gc.vals[i * YBG_MAX_AGGS + g] = init;
       gc.cnts[i * YBG_MAX_AGGS + g] = 0;
+      gc.vals_hi[i * YBG_MAX_AGGS + g] = 0;
+      gc.poison[i * YBG_MAX_AGGS + g] = 0;
     }
   }
 }
@@ -508,7 +514,10 @@ __global__ void k_group_export(DevSpec sp, GroupCtx gc,
     out_keys[slot] = kv;
     for (int g = 0; g < sp.num_aggs; ++g) {
       int op = sp.agg_op[g];
-      long long v = gc.vals[i * YBG_MAX_AGGS + g];
+      long long v = group_export_value(
+          op, gc.vals[i * YBG_MAX_AGGS + g],
+          gc.vals_hi[i * YBG_MAX_AGGS + g],
+          gc.poison[i * YBG_MAX_AGGS + g]);
       out_vals[slot * YBG_MAX_AGGS + g] = v;
       // COUNT ops skip the per-row cnt atomic: cnt == val by definition
       out_cnts[slot * YBG_MAX_AGGS + g] =
@@ -1640,11 +1649,6 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
   if (!s->d_data) return set_err(4, "feed_blocks not called");
   if (s->dspec.group_col < 0)
     return set_err(9, "spec.group_col not set");
-  for (int g = 0; g < s->dspec.num_aggs; ++g) {
-    int op = s->dspec.aggs[g].op;
-    if (op == YBG_AGG_MIN_DOUBLE || op == YBG_AGG_MAX_DOUBLE)
-      return set_err(9, "grouped MIN/MAX over double not supported yet");
-  }
   if (!s->gc.gkey) {
     s->group_cap = 1ull << 20;  // 1M groups
     HIP_TRY(hipMalloc(&s->gc.gkey,
@@ -1652,6 +1656,10 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
     HIP_TRY(hipMalloc(&s->gc.state, (s->group_cap + 1) * sizeof(unsigned)));
     HIP_TRY(hipMalloc(&s->gc.vals, (s->group_cap + 1) * YBG_MAX_AGGS * 8));
     HIP_TRY(hipMalloc(&s->gc.cnts, (s->group_cap + 1) * YBG_MAX_AGGS * 8));
+    HIP_TRY(hipMalloc(&s->gc.vals_hi,
+                      (s->group_cap + 1) * YBG_MAX_AGGS * 8));
+    HIP_TRY(hipMalloc(&s->gc.poison,
+                      (s->group_cap + 1) * YBG_MAX_AGGS * 4));
     HIP_TRY(hipMalloc(&s->gc.overflow, sizeof(unsigned long long)));
     HIP_TRY(hipMalloc(&s->d_g_counters, 3 * sizeof(unsigned long long)));
     s->gc.cap = s->group_cap;
@@ -1809,6 +1817,8 @@ int yb_gpu_scan_close(ybg_scan_t* s) {
   if (s->gc.state) HIP_WARN(hipFree(s->gc.state));
   if (s->gc.vals) HIP_WARN(hipFree(s->gc.vals));
   if (s->gc.cnts) HIP_WARN(hipFree(s->gc.cnts));
+  if (s->gc.vals_hi) HIP_WARN(hipFree(s->gc.vals_hi));
+  if (s->gc.poison) HIP_WARN(hipFree(s->gc.poison));
   if (s->gc.overflow) HIP_WARN(hipFree(s->gc.overflow));
   if (s->d_gheads) HIP_WARN(hipFree(s->d_gheads));
   if (s->d_g_counters) HIP_WARN(hipFree(s->d_g_counters));

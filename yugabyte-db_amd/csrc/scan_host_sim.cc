@@ -378,12 +378,18 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
   std::vector<unsigned> state(gcap + 1, 0);
   std::vector<long long> vals((gcap + 1) * YBG_MAX_AGGS, 0);
   std::vector<unsigned long long> cnts((gcap + 1) * YBG_MAX_AGGS, 0);
+  std::vector<unsigned long long> vals_hi((gcap + 1) * YBG_MAX_AGGS, 0);
+  std::vector<unsigned> poison((gcap + 1) * YBG_MAX_AGGS, 0);
   for (uint64_t i = 0; i <= gcap; ++i)
     for (int g = 0; g < d.num_aggs; ++g) {
       if (d.aggs[g].op == YBG_AGG_MIN_INT64)
         vals[i * YBG_MAX_AGGS + g] = 0x7fffffffffffffffll;
       else if (d.aggs[g].op == YBG_AGG_MAX_INT64)
         vals[i * YBG_MAX_AGGS + g] = (long long)0x8000000000000000ll;
+      else if (d.aggs[g].op == YBG_AGG_MIN_DOUBLE)
+        vals[i * YBG_MAX_AGGS + g] = (long long)~0ull;
+      else if (d.aggs[g].op == YBG_AGG_MAX_DOUBLE)
+        vals[i * YBG_MAX_AGGS + g] = 0;
     }
   unsigned long long overflow = 0;
   GroupCtx gc;
@@ -391,6 +397,8 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
   gc.state = state.data();
   gc.vals = vals.data();
   gc.cnts = cnts.data();
+  gc.vals_hi = vals_hi.data();
+  gc.poison = poison.data();
   gc.cap = gcap;
   gc.overflow = &overflow;
   gc.data = data;
@@ -446,7 +454,9 @@ int ybg_sim_group(const ybg_scan_spec_t* spec, const uint8_t* data,
     keys_out[n] = kv;
     for (int g = 0; g < YBG_MAX_AGGS; ++g) {
       int op = g < d.num_aggs ? d.agg_op[g] : -1;
-      long long v = vals[i * YBG_MAX_AGGS + g];
+      long long v = group_export_value(op, vals[i * YBG_MAX_AGGS + g],
+                                       vals_hi[i * YBG_MAX_AGGS + g],
+                                       poison[i * YBG_MAX_AGGS + g]);
       vals_out[n * YBG_MAX_AGGS + g] = v;
       cnts_out[n * YBG_MAX_AGGS + g] =
           (op == YBG_AGG_COUNT_STAR || op == YBG_AGG_COUNT)
