@@ -969,3 +969,34 @@ def encode_dockey(schema, hash_=0, key_datums=(), key_strs=()):
     n = f(C.byref(schema), C.byref(k), out, 512)
     assert n > 0
     return bytes(out[:n])
+
+
+def lz4_compress(data):
+    """Host lz4 compressor (generator codec; tests)."""
+    lib = product()
+    f = _sig(lib, "ybg_lz4_compress", C.c_int64,
+             [C.POINTER(C.c_uint8), C.c_uint64, C.POINTER(C.c_uint8),
+              C.c_uint64])
+    src = (C.c_uint8 * len(data)).from_buffer_copy(data)
+    cap = len(data) * 2 + 64
+    dst = (C.c_uint8 * cap)()
+    n = f(src, len(data), dst, cap)
+    assert n > 0
+    return bytes(dst[:n])
+
+
+
+def lz4_uncompress(data, cap):
+    """Host lz4 decompressor (same code the GPU kernel runs)."""
+    lib = product()
+    f = _sig(lib, "ybg_lz4_uncompress", C.c_int64,
+             [C.POINTER(C.c_uint8), C.c_uint64, C.POINTER(C.c_uint8),
+              C.c_uint64])
+    src = (C.c_uint8 * len(data)).from_buffer_copy(data)
+    dst = (C.c_uint8 * cap)()
+    n = f(src, len(data), dst, cap)
+    if n < 0:
+        raise RuntimeError("lz4_uncompress failed")
+    return bytes(dst[:n])
+
+

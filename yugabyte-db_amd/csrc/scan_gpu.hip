@@ -44,6 +44,7 @@
 #include "scan_device.h"
 #include "sst_internal.h"
 #include "snappy_dev.h"
+#include "lz4_dev.h"
 
 using namespace ybgdev;
 
@@ -725,8 +726,10 @@ __global__ __launch_bounds__(256) void k_reduce(
 }
 
 // Per-block decompression at feed time: one thread per data block (~1M
-// blocks >> threads), serial snappy decode per thread (snappy_dev.h).
-// type 0 blocks copy through in u64 chunks.
+// blocks >> threads), serial decode per thread — snappy (type 1,
+// snappy_dev.h) or LZ4 (type 4, lz4_dev.h, varint32 raw-length framing
+// per rocksdb util/compression.h). type 0 blocks copy through in u64
+// chunks.
 __global__ __launch_bounds__(256) void k_snappy(
     const uint8_t* __restrict__ src_blob,
     const uint64_t* __restrict__ src_off,
@@ -745,6 +748,14 @@ __global__ __launch_bounds__(256) void k_snappy(
     if (types[i] == 1) {
       if (ybsnappy::snappy_uncompress(src, n, dst, cap) < 0)
         atomicAdd(err, 1ull);
+    } else if (types[i] == 4) {
+      // skip the varint32 raw-length prefix (validated host-side)
+      uint64_t o = 0;
+      while (o < n && (src[o] & 0x80)) ++o;
+      ++o;
+      if (o > n || yblz4::lz4_uncompress(src + o, n - o, dst, cap) !=
+                       (int64_t)cap)
+        atomicAdd(err, 1ull);
     } else {
       for (uint64_t b = 0; b + 8 <= n; b += 8) {
         uint64_t w;
@@ -753,6 +764,55 @@ __global__ __launch_bounds__(256) void k_snappy(
       }
       for (uint64_t b = n & ~7ull; b < n; ++b) dst[b] = src[b];
     }
+  }
+}
+
+// Data-block trailer verification on DEVICE (util/crc32c crc32c.cc,
+// masked per format.h:212): one thread per block, slicing-by-4 over LDS
+// tables built at launch. Replaces the ~2.2 GB/s host verify on the SST
+// feed path.
+__global__ __launch_bounds__(256) void k_crc32c(
+    const uint8_t* __restrict__ blob, const uint64_t* __restrict__ offs,
+    const uint64_t* __restrict__ szs, uint64_t n_blocks,
+    unsigned long long* __restrict__ err) {
+  __shared__ uint32_t t[4][256];
+  for (int i = threadIdx.x; i < 256; i += blockDim.x) {
+    uint32_t c = i;
+    for (int k = 0; k < 8; ++k)
+      c = (c & 1) ? (0x82F63B78u ^ (c >> 1)) : (c >> 1);
+    t[0][i] = c;
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < 256; i += blockDim.x) {
+    uint32_t c = t[0][i];
+    for (int s = 1; s < 4; ++s) {
+      c = t[0][c & 0xff] ^ (c >> 8);
+      t[s][i] = c;
+    }
+  }
+  __syncthreads();
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (; i < n_blocks; i += stride) {
+    const uint8_t* p = blob + offs[i];
+    uint64_t n = szs[i] + 1;  // block bytes + trailer type byte
+    uint32_t c = 0xffffffffu;
+    while (n >= 4) {
+      uint32_t lo;
+      __builtin_memcpy(&lo, p, 4);
+      lo ^= c;
+      c = t[3][lo & 0xff] ^ t[2][(lo >> 8) & 0xff] ^
+          t[1][(lo >> 16) & 0xff] ^ t[0][lo >> 24];
+      p += 4;
+      n -= 4;
+    }
+    for (uint64_t k = 0; k < n; ++k)
+      c = t[0][(c ^ p[k]) & 0xff] ^ (c >> 8);
+    c ^= 0xffffffffu;
+    uint32_t masked = ((c >> 15) | (c << 17)) + 0xa282ead8u;  // kCrcMaskDelta
+    uint32_t want;
+    __builtin_memcpy(&want, blob + offs[i] + szs[i] + 1, 4);
+    if (masked != want) atomicAdd(err, 1ull);
   }
 }
 
@@ -1291,46 +1351,36 @@ int ybg_sst_index(const uint8_t* file, uint64_t size, int verify,
 // Feed a complete SST file: parse footer/index (sst_format.cc), verify
 // block checksums, strip the per-block trailers into the concatenated
 // block layout and hand over to the existing feed path.
+// Feed a complete SST file: footer/index parse host-side (tiny), then
+// EVERYTHING block-sized runs on device — k_crc32c verifies every data
+// block's masked-crc32c trailer, k_snappy decompresses snappy/LZ4 blocks
+// (or copies type-0 blocks) into the concatenated layout the scan
+// consumes (rocksdb/table/format.cc responsibilities).
 int yb_gpu_scan_feed_sst(ybg_scan_t* s, const uint8_t* file, uint64_t size,
                          int verify_checksums) {
   std::vector<uint64_t> offs, szs;
   std::vector<uint8_t> typs;
   std::string err;
-  int rc = ybsst::parse_sst(file, size, verify_checksums, &offs, &szs, &typs,
-                            &err);
+  // verify mode 2: footer + index checksums host-side only; the
+  // data-block trailers are verified by k_crc32c below
+  int rc = ybsst::parse_sst(file, size, verify_checksums ? 2 : 0, &offs,
+                            &szs, &typs, &err);
   if (rc) return set_err(rc, err);
   if (offs.empty()) return set_err(3, "SST file holds no data blocks");
-  bool any_comp = false;
-  for (uint8_t t : typs) any_comp |= (t == 1);
-  if (!any_comp) {
-    uint64_t total = 0;
-    for (uint64_t sz : szs) total += sz;
-    std::vector<uint8_t> blocks;
-    blocks.reserve(total);
-    std::vector<uint64_t> boff(offs.size() + 1);
-    boff[0] = 0;
-    for (size_t i = 0; i < offs.size(); ++i) {
-      blocks.insert(blocks.end(), file + offs[i], file + offs[i] + szs[i]);
-      boff[i + 1] = blocks.size();
-    }
-    return yb_gpu_scan_feed_blocks(s, blocks.data(), boff.data(),
-                                   offs.size(), 0);
-  }
-  // On-GPU decompression: upload the data-block region once, decompress
-  // every block into its final slot (uncompressed lengths come from each
-  // snappy preamble), then the normal device-resident feed path runs.
   uint64_t n = offs.size();
   std::vector<uint64_t> un_len(n), boff(n + 1);
   boff[0] = 0;
   for (uint64_t i = 0; i < n; ++i) {
-    if (typs[i] == 1) {
+    if (typs[i] == 1 || typs[i] == 4) {
+      // uncompressed length: snappy uvarint preamble / rocksdb LZ4
+      // varint32 framing (both LEB128)
       const uint8_t* p = file + offs[i];
       const uint8_t* lim = p + szs[i];
       uint64_t ulen = 0;
       int shift = 0;
       for (;;) {
         if (p >= lim || shift > 28)
-          return set_err(3, "corrupt snappy preamble");
+          return set_err(3, "corrupt compressed-block length preamble");
         uint8_t b = *p++;
         ulen |= (uint64_t)(b & 0x7f) << shift;
         if (!(b & 0x80)) break;
@@ -1363,11 +1413,26 @@ int yb_gpu_scan_feed_sst(ybg_scan_t* s, const uint8_t* file, uint64_t size,
                     hipMemcpyHostToDevice));
   HIP_TRY(hipMalloc(&d_types, n));
   HIP_TRY(hipMemcpy(d_types, typs.data(), n, hipMemcpyHostToDevice));
-  HIP_TRY(hipMalloc(&d_out, total_un + 256));
-  HIP_TRY(hipMemset(d_out + total_un, 0, 256));
   HIP_TRY(hipMalloc(&d_err, 8));
   HIP_TRY(hipMemset(d_err, 0, 8));
   int sgrid = (int)std::min<uint64_t>((n + 255) / 256, 4096);
+  if (verify_checksums) {
+    hipLaunchKernelGGL(k_crc32c, dim3(sgrid), dim3(256), 0, s->stream,
+                       d_blob, d_meta, d_meta + n, n, d_err);
+    unsigned long long h_err = 0;
+    HIP_TRY(hipMemcpyAsync(&h_err, d_err, 8, hipMemcpyDeviceToHost,
+                           s->stream));
+    HIP_TRY(hipStreamSynchronize(s->stream));
+    if (h_err) {
+      HIP_WARN(hipFree(d_blob));
+      HIP_WARN(hipFree(d_meta));
+      HIP_WARN(hipFree(d_types));
+      HIP_WARN(hipFree(d_err));
+      return set_err(3, "data block checksum mismatch (device verify)");
+    }
+  }
+  HIP_TRY(hipMalloc(&d_out, total_un + 256));
+  HIP_TRY(hipMemset(d_out + total_un, 0, 256));
   hipLaunchKernelGGL(k_snappy, dim3(sgrid), dim3(256), 0, s->stream, d_blob,
                      d_meta, d_meta + n, d_meta + 2 * n, d_types, n, d_out,
                      total_un, d_err);
@@ -1381,7 +1446,7 @@ int yb_gpu_scan_feed_sst(ybg_scan_t* s, const uint8_t* file, uint64_t size,
   HIP_WARN(hipFree(d_err));
   if (h_err) {
     HIP_WARN(hipFree(d_out));
-    return set_err(3, "snappy block decompression failed");
+    return set_err(3, "block decompression failed");
   }
   rc = yb_gpu_scan_feed_blocks(s, d_out, boff.data(), n, 1);
   if (rc) {

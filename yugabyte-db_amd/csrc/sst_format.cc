@@ -142,11 +142,12 @@ int parse_sst(const uint8_t* file, uint64_t size, int verify,
     const uint8_t* b = file + (*offsets)[i];
     uint64_t n = (*sizes)[i];
     uint8_t type = (*types)[i];
-    if (type > 1) {
-      *err = "unsupported block compression type (none/snappy only)";
+    if (type != 0 && type != 1 && type != 4) {
+      *err = "unsupported block compression type (none/snappy/lz4 only)";
       return 3;
     }
-    if (verify) {
+    if (verify == 1) {  // verify 2 = footer/index only (data-block
+                        // trailers are verified on DEVICE by k_crc32c)
       uint32_t crc = crc32c_extend(crc32c_value(b, n), &type, 1);
       if (crc32c_mask(crc) != fixed32(b + n + 1)) {
         *err = "data block checksum mismatch";

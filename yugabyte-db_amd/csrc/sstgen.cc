@@ -11,6 +11,7 @@
 #include "codec.h"
 #include "sst_internal.h"
 #include "snappy_dev.h"
+#include "lz4_dev.h"
 
 #include <algorithm>
 #include <atomic>
@@ -568,6 +569,20 @@ int ybg_builder_finish_sst2(ybg_builder_t* b, int compression,
         append_block_t(cbuf.data(), (uint64_t)cn, 1, &h_off[i], &h_sz[i]);
         continue;
       }
+    } else if (compression == 4 /* kLZ4Compression */) {
+      // rocksdb LZ4 framing (compress_format_version 2):
+      // varint32(raw length) then the LZ4 block (util/compression.h)
+      ybg::Buf framed;
+      ybg::Leb128Append(bn, &framed);
+      cbuf.resize(bn + bn / 2 + 64);
+      int64_t cn =
+          yblz4::lz4_compress(bp, bn, cbuf.data(), cbuf.size());
+      if (cn > 0 && framed.size() + (uint64_t)cn < bn) {
+        framed.insert(framed.end(), cbuf.data(), cbuf.data() + cn);
+        append_block_t(framed.data(), framed.size(), 4, &h_off[i],
+                       &h_sz[i]);
+        continue;
+      }
     }
     append_block(bp, bn, &h_off[i], &h_sz[i]);
   }
@@ -631,6 +646,14 @@ int ybg_builder_finish_sst(ybg_builder_t* b, const uint8_t** data,
 int64_t ybg_snappy_compress(const uint8_t* src, uint64_t n, uint8_t* dst,
                             uint64_t cap) {
   return ybsnappy::snappy_compress(src, n, dst, cap);
+}
+int64_t ybg_lz4_compress(const uint8_t* src, uint64_t n, uint8_t* dst,
+                         uint64_t cap) {
+  return yblz4::lz4_compress(src, n, dst, cap);
+}
+int64_t ybg_lz4_uncompress(const uint8_t* src, uint64_t n, uint8_t* dst,
+                           uint64_t cap) {
+  return yblz4::lz4_uncompress(src, n, dst, cap);
 }
 int64_t ybg_snappy_uncompress(const uint8_t* src, uint64_t n, uint8_t* dst,
                               uint64_t cap) {
