@@ -1264,10 +1264,18 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
   HIP_WARN(hipFree(d_counts));
   HIP_WARN(hipFree(d_err));
 
-  // default 2 intervals/batch: the measured optimum for the filtersum
-  // workload (15.6 vs 15.1 Grows/s at 1; MVCC prefers 4 at +11%, 16+
-  // degrades both — tail imbalance and per-wave L2 span grow with C)
-  s->ivb = 2;
+  // Adaptive batch size: aim for ~2 grid-stride passes over a full launch
+  // (span = 8192 workgroups x 256 threads). Reproduces the measured
+  // optima — 2 for the 6.5M-interval filtersum tablet (15.6 vs 15.1
+  // Grows/s at 1), 8 for the 32M-interval MVCC tablet (3.5 vs 3.2 at 2);
+  // larger batches degrade both (tail imbalance + per-wave L2 span).
+  {
+    const uint64_t span = 8192ull * kThreads;
+    uint64_t ivb = (s->n_ivs + span) / (2 * span);
+    if (ivb < 1) ivb = 1;
+    if (ivb > 64) ivb = 64;
+    s->ivb = ivb;
+  }
   if (const char* e = getenv("YBG_IVB")) {
     long v = atol(e);
     if (v >= 1 && v <= 4096) s->ivb = (uint64_t)v;
