@@ -216,3 +216,19 @@ def test_intents_gpu():
         o = y.orcl_scan_intents(built[0], built[1], built[2], osc, ospec,
                                 blob, blen, txns, ntx)
         assert _res(g) == _res(o), (read, local)
+
+
+def test_restart_via_global_limit_only():
+    """Regression (round-2 GPU soak): a committed intent visible through
+    the GLOBAL limit needs restart tracking even when local_limit == read
+    — the device gate used to consider only the local window."""
+    schema, built = _base_tablet(60)
+    it = y.Intents(schema)
+    it.add_packed_row(1200, 9, [(y.T_INT64, 300), (y.T_INT64, 5)],
+                      hash_=0, key_datums=(300,))
+    # read == local == 1500, global 3500; commit 2000 in (read, global]:
+    # intent time 1200 <= local -> global rule -> VISIBLE with commit >
+    # read -> restart data required
+    r = _cross_check(schema, built, it, {9: ("c", 2000)}, 1500, 1500, 3500)
+    assert r.rows_scanned == 61
+    assert r.restart_ht_len > 0

@@ -3255,9 +3255,20 @@ inline void build_dev_spec(const ybg_scan_spec_t* spec, DevSpec* dp,
     int cmp = memcmp(spec->read_time.local_limit, spec->read_time.read, n);
     bool local_smaller = cmp < 0 || (cmp == 0 && ll < rl);
     d.reg_lim = local_smaller ? d.local_lim : d.read;
-    // restart possible iff regular_limit != read (local_limit > read):
-    // encoded order reversed, so local_smaller means local > read
-    d.track_restart = local_smaller ? 1 : 0;
+    // Restart tracking (the reference's UpdateMaxSeenHt is unconditional;
+    // GetReadRestartData reports when max-seen > read). A visible record
+    // with commit > read can exist when EITHER limit extends beyond read:
+    // regular records through local_limit, committed-intent records
+    // through GLOBAL_limit (the :1249-1267 rule) — gating on the local
+    // window alone missed intent-carrying records when local == read <
+    // global (caught by the round-2 GPU/oracle soak). Encoded order is
+    // reversed, so "limit beyond read" == enc(limit) < enc(read).
+    int gl = spec->read_time.global_limit_len;
+    int n2 = gl < rl ? gl : rl;
+    int cmp2 = memcmp(spec->read_time.global_limit, spec->read_time.read,
+                      n2);
+    bool global_beyond = cmp2 < 0 || (cmp2 == 0 && gl < rl);
+    d.track_restart = (local_smaller || global_beyond) ? 1 : 0;
   }
   uint32_t pos = 0;
   auto put = [&](const uint8_t* b, uint64_t n) {
