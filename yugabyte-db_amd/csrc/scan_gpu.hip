@@ -299,22 +299,27 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan_fast(
 
 constexpr int kEmitThreads = 128;
 
+// Row buffers live in DYNAMIC shared memory sized to the actual column
+// count at launch: the static worst-case (kEmitThreads * YBG_MAX_COLS
+// datum + length slots = 64 KB) pinned the kernel at 1 wave/SIMD; a
+// 4-column schema now fits ~4 workgroups per CU.
 __global__ __launch_bounds__(kEmitThreads) void k_emit(
     const uint8_t* __restrict__ data,
     const uint64_t* __restrict__ block_offsets,
     const Interval* __restrict__ ivs, uint64_t n_ivs,
     const uint8_t* __restrict__ aux, uint8_t* __restrict__ rk_save_buf,
-    EmitCtx ec, unsigned long long* __restrict__ err_counter) {
+    EmitCtx ec, unsigned long long* __restrict__ err_counter, int nc_pad) {
   const DevSpec& sp = c_spec;
   __shared__ uint8_t key_scratch[kEmitThreads * kKeyCap];
-  __shared__ uint64_t rowbuf[kEmitThreads * YBG_MAX_COLS];
-  __shared__ uint32_t lenbuf[kEmitThreads * YBG_MAX_COLS];
   __shared__ uint64_t bht_scratch[kEmitThreads * 6];
+  extern __shared__ uint64_t emit_dyn[];  // rowbuf | lenbuf
+  uint64_t* rowbuf = emit_dyn;
+  uint32_t* lenbuf = (uint32_t*)(emit_dyn + (size_t)kEmitThreads * nc_pad);
   uint64_t* bht = bht_scratch + (size_t)threadIdx.x * 6;
   bht[5] = 0;
   uint8_t* key = key_scratch + (size_t)threadIdx.x * kKeyCap;
-  uint64_t* rb = rowbuf + (size_t)threadIdx.x * YBG_MAX_COLS;
-  uint32_t* lb = lenbuf + (size_t)threadIdx.x * YBG_MAX_COLS;
+  uint64_t* rb = rowbuf + (size_t)threadIdx.x * nc_pad;
+  uint32_t* lb = lenbuf + (size_t)threadIdx.x * nc_pad;
   const uint32_t gtid = blockIdx.x * kEmitThreads + threadIdx.x;
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
   const uint64_t span = (uint64_t)gridDim.x * kEmitThreads;
@@ -1667,9 +1672,13 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
     HIP_TRY(hipMalloc(&rk_extra, need_rk));
     rk_area = rk_extra;
   }
-  hipLaunchKernelGGL(k_emit, dim3(egrid), dim3(kEmitThreads), 0, s->stream,
-                     s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
-                     s->d_aux, rk_area, ec, s->d_em_counters + 3);
+  int nc_pad = nc > 0 ? nc : 1;
+  size_t emit_dyn_bytes =
+      (size_t)kEmitThreads * nc_pad * (sizeof(uint64_t) + sizeof(uint32_t));
+  hipLaunchKernelGGL(k_emit, dim3(egrid), dim3(kEmitThreads),
+                     emit_dyn_bytes, s->stream, s->d_data, s->d_offsets,
+                     s->d_ivs, s->n_ivs, s->d_aux, rk_area, ec,
+                     s->d_em_counters + 3, nc_pad);
   unsigned long long ctr[4];
   HIP_TRY(hipMemcpyAsync(ctr, s->d_em_counters,
                          4 * sizeof(unsigned long long),
