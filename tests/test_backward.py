@@ -157,3 +157,34 @@ def test_backward_tuple_id_seek():
     assert lib.yb_host_iter_next(h, kd, vd, C.byref(nm), C.byref(vl)) == 1
     assert kd[0] == last_key
     lib.yb_host_iter_close(h)
+
+
+@pytest.mark.gpu
+def test_ql_row_form():
+    """FetchNext(QLTableRow*) analog: the same rows keyed by COLUMN ID
+    (ql_rowwise_iterator_interface.h:48-52)."""
+    gpu_scan = _gpu()
+    lib = gpu_scan._lib()
+    lib.yb_host_iter_next_ql.restype = C.c_int
+    lib.yb_host_iter_next_ql.argtypes = [
+        C.c_void_p, C.POINTER(C.c_int32), C.POINTER(C.c_uint64),
+        C.POINTER(C.c_int32), C.POINTER(C.c_uint64), C.POINTER(C.c_uint32),
+        C.POINTER(C.POINTER(C.c_uint8))]
+    schema, built = _tablet(300)
+    spec = _spec(schema)
+    h = lib.yb_host_iter_open(C.byref(spec), built[0], built[1], built[2])
+    kci = (C.c_int32 * y.MAX_KEYCOLS)()
+    kd = (C.c_uint64 * y.MAX_KEYCOLS)()
+    ci = (C.c_int32 * y.MAX_COLS)()
+    vd = (C.c_uint64 * y.MAX_COLS)()
+    nm = C.c_uint32()
+    vl = C.POINTER(C.c_uint8)()
+    n = 0
+    while lib.yb_host_iter_next_ql(h, kci, kd, ci, vd, C.byref(nm),
+                                   C.byref(vl)) == 1:
+        if n == 0:
+            assert list(ci[:2]) == [10, 11]  # schema column ids
+            assert kd[0] == 0 and vd[0] == 0 and vd[1] == 0
+        n += 1
+    lib.yb_host_iter_close(h)
+    assert n == 300

@@ -204,6 +204,34 @@ int yb_host_iter_next(void* h, uint64_t* key_datums, uint64_t* datums,
   return 1;
 }
 
+// YCQL row form — FetchNext(QLTableRow*) analog
+// (ql_rowwise_iterator_interface.h:48-52): the same row keyed by COLUMN
+// ID instead of projection position (QLTableRow::AllocColumn semantics).
+// Fills one (column_id, datum) pair per schema value column plus the key
+// columns' ids/datums; NULL columns report their id with the null bit.
+int yb_host_iter_next_ql(void* h, int32_t* key_col_ids,
+                         uint64_t* key_datums, int32_t* col_ids,
+                         uint64_t* datums, uint32_t* null_mask,
+                         const uint8_t** varlen) {
+  auto* it = static_cast<ybg::GpuDocRowwiseIterator*>(h);
+  ybg::PgRow row;
+  int rc = it->PgFetchNext(&row);
+  if (rc != 1) return rc;
+  const ybg_schema_t& sc = it->schema();
+  int nk = sc.num_hash_cols + sc.num_range_cols;
+  for (int c = 0; c < nk; ++c) {
+    key_col_ids[c] = c;  // key columns are identified by position
+    key_datums[c] = row.key_datums[c];
+  }
+  for (int c = 0; c < sc.num_value_cols; ++c) {
+    col_ids[c] = sc.value_cols[c].column_id;
+    datums[c] = row.datums[c];
+  }
+  *null_mask = row.null_mask;
+  *varlen = row.varlen;
+  return 1;
+}
+
 int yb_host_iter_paging_state(void* h, uint8_t* key_out, size_t cap,
                               size_t* len) {
   return static_cast<ybg::GpuDocRowwiseIterator*>(h)->PagingState(key_out,

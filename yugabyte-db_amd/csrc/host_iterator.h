@@ -50,6 +50,7 @@ class GpuDocRowwiseIterator {
   int SeekTuple(const uint8_t* dockey, size_t len);
 
   const char* LastError() const;
+  const ybg_schema_t& schema() const { return spec_.schema; }
 
  private:
   int MaterializeBatch();
