@@ -243,6 +243,9 @@ struct Interval {
 
 constexpr int kThreads = 256;
 constexpr int kKeyCap = 128;
+// the fast kernel's per-thread LDS key slot (fixed rowkeys only need the
+// rowkey prefix + chunk slack; 80 covers 5 + 8 key cols x 9)
+constexpr int kFastKeyCap = 80;
 // wave partial: entries, scanned, matched, err, {val,cnt} x MAX_AGGS,
 // restart-min {hi, lo, len} (encoded-HT MIN among visible restart
 // candidates; len 0 = none)
@@ -2499,7 +2502,6 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
   for (int g = 0; g < NA; ++g) { av_b[g] = 0; ac_b[g] = 0; }
 
   uint32_t key_len = 0;
-  uint64_t last8 = 0;
   uint64_t thi = 0, tlo = 0;
   bool row_open = false, in_head = true, walked = false;
   bool base_seen = false, found = false;
@@ -2553,7 +2555,9 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
       uint32_t e2 = (uint32_t)((w >> (8 * e1len)) & 0xff);
       if (e2 & 1) YBG_FABORT(4);  // reuse forms: not a self-contained restart
       uint32_t ns1 = e2 >> 1;
-      if (ns1 < 24 || ns1 > kKeyCap - 8) YBG_FABORT(5);
+      // upper bound: the u64-chunk copy below writes up to ns1 rounded up
+      // to 8 into the caller's kFastKeyCap LDS slot
+      if (ns1 < 24 || ns1 > kFastKeyCap - 8) YBG_FABORT(5);
       uint32_t hl = e1len + 1;
       if ((uint64_t)(limit - p) < (uint64_t)hl + ns1 + value_size) YBG_FABORT(6);
       const uint8_t* kp2 = p + hl;
@@ -2574,8 +2578,7 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
         __builtin_memcpy(&key[i], &w8, 8);
       }
       key_len = ns1;
-      last8 = load_u64_una(key + ns1 - 8);
-      tail_from_lds(key, ns1 - 8, &thi, &tlo);
+      tail_from_lds(kp2, ns1 - 8, &thi, &tlo);
       er.value = kp2 + ns1;
       er.value_len = value_size;
       er.shared = 0;
@@ -2720,7 +2723,7 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
         if (ns2) tail_splice(&thi, &tlo, (int32_t)ns2s - T, ns2, s2h, 0);
 #endif
       }
-      last8 += inc;
+      (void)inc;
       rdr.consume(nb);
       er.value = rdr.pos();
       er.value_len = value_size;

@@ -187,7 +187,6 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
 // retry list and re-run by k_scan in retry mode; head resolution is the
 // same global bheads/walked protocol.
 // ---------------------------------------------------------------------------
-constexpr int kFastKeyCap = 80;  // fixed rowkeys: <= 5 + 8 cols x 9
 
 template <int WPS>
 __global__ __launch_bounds__(kThreads, WPS) void k_scan_fast(
@@ -1004,6 +1003,8 @@ int yb_gpu_scan_open(const ybg_scan_spec_t* spec, ybg_scan_t** out) {
   return 0;
 }
 
+}  // extern "C"
+
 namespace {
 
 // ---------------------------------------------------------------------
@@ -1163,6 +1164,8 @@ bool compute_block_selection(const ybg_scan_spec_t& spec,
 }
 
 }  // namespace
+
+extern "C" {
 
 // Test hook: the block selection feed_blocks applies (1 byte per block;
 // returns 1 when pruning engaged). CPU tests feed the kept subset through
