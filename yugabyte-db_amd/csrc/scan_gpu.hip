@@ -75,7 +75,8 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
     const uint8_t* __restrict__ aux, uint8_t* __restrict__ rk_save_buf,
     uint64_t* __restrict__ partials, uint64_t* __restrict__ bheads,
     uint8_t* __restrict__ walked, uint32_t* __restrict__ iv_flags,
-    int write_all_flags, uint64_t ivb,
+    int write_all_flags, uint64_t ivb, uint64_t n_bat,
+    const uint64_t* __restrict__ batch_lo,
     const uint64_t* __restrict__ retry_ids,
     const unsigned long long* __restrict__ retry_n) {
   const DevSpec& sp = c_spec;
@@ -87,9 +88,10 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
   const uint32_t gtid = blockIdx.x * kThreads + threadIdx.x;
   uint8_t* rk_save = rk_save_buf + (size_t)gtid * kKeyCap;
   const uint64_t span = (uint64_t)gridDim.x * kThreads;
-  // batch = ivb CONSECUTIVE intervals scanned as one stream by one thread
-  const uint64_t n_batches = (n_ivs + ivb - 1) / ivb;
-  const uint64_t n_work = retry_ids ? *retry_n : n_batches;
+  // batch = ivb CONSECUTIVE intervals — or one BLOCK's intervals when
+  // batch_lo is set (block-aligned batches keep lanes phase-locked on
+  // the 16-entry restart cadence)
+  const uint64_t n_work = retry_ids ? *retry_n : n_bat;
   constexpr int kHS = 2 * NA + 2;  // bheads stride
 
   uint32_t entries = 0, scanned = 0, matched = 0, errs = 0;
@@ -105,8 +107,8 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
     const uint64_t j = retry_ids ? retry_ids[i0] : i0;
     HeadOut<NA> ho;
     bool walked_next = false;
-    const uint64_t lo = j * ivb;
-    uint64_t hi = lo + ivb;
+    const uint64_t lo = batch_lo ? batch_lo[j] : j * ivb;
+    uint64_t hi = batch_lo ? batch_lo[j + 1] : lo + ivb;
     if (hi > n_ivs) hi = n_ivs;
     if (!scan_one_interval<NA>(sp, data, block_offsets, ivs, n_ivs, lo,
                                aux, key, rk_save, bht, &entries, &scanned,
@@ -194,7 +196,8 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan_fast(
     const uint64_t* __restrict__ block_offsets,
     const Interval* __restrict__ ivs, uint64_t n_ivs,
     uint64_t* __restrict__ partials, uint64_t* __restrict__ bheads,
-    uint8_t* __restrict__ walked, uint64_t ivb,
+    uint8_t* __restrict__ walked, uint64_t ivb, uint64_t n_bat,
+    const uint64_t* __restrict__ batch_lo,
     uint64_t* __restrict__ retry_ids,
     unsigned long long* __restrict__ retry_n) {
   const DevSpec& sp = c_spec;
@@ -205,7 +208,6 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan_fast(
   rmin[2] = 0;
   const uint32_t gtid = blockIdx.x * kThreads + threadIdx.x;
   const uint64_t span = (uint64_t)gridDim.x * kThreads;
-  const uint64_t n_batches = (n_ivs + ivb - 1) / ivb;
   constexpr int NA = 2;
   constexpr int kHS = 2 * NA + 2;
 
@@ -214,11 +216,11 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan_fast(
 #pragma unroll
   for (int g = 0; g < NA; ++g) { agg_val[g] = 0; agg_cnt[g] = 0; }
 
-  for (uint64_t j = gtid; j < n_batches; j += span) {
+  for (uint64_t j = gtid; j < n_bat; j += span) {
     HeadOut<NA> ho;
     bool walked_next = false;
-    const uint64_t lo = j * ivb;
-    uint64_t hi = lo + ivb;
+    const uint64_t lo = batch_lo ? batch_lo[j] : j * ivb;
+    uint64_t hi = batch_lo ? batch_lo[j + 1] : lo + ivb;
     if (hi > n_ivs) hi = n_ivs;
     if (!scan_batch_fast<NA>(sp, data, block_offsets, ivs, n_ivs, lo, hi,
                              key, rmin, &entries, &scanned, &matched,
@@ -859,6 +861,8 @@ struct ybg_scan {
   uint64_t n_ivs = 0;
   uint64_t ivb = 1;       // intervals per batch (one thread's stream)
   uint64_t n_batches = 0;
+  uint64_t* d_batch_lo = nullptr;  // block-aligned batches (YBG_BB=1):
+                                   // batch b = intervals [lo[b], lo[b+1])
   uint64_t n_blocks = 0;
   uint64_t total_bytes = 0;
   uint8_t* d_aux = nullptr;
@@ -1268,7 +1272,11 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
                      s->stream, s->d_data, s->d_offsets, n_blocks, d_base,
                      s->d_ivs);
   HIP_TRY(hipStreamSynchronize(s->stream));
-  HIP_WARN(hipFree(d_base));
+  if (getenv("YBG_BB") && atoi(getenv("YBG_BB")) != 0) {
+    s->d_batch_lo = d_base;  // block-aligned batches own the base table
+  } else {
+    HIP_WARN(hipFree(d_base));
+  }
   HIP_WARN(hipFree(d_counts));
   HIP_WARN(hipFree(d_err));
 
@@ -1288,7 +1296,8 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
     long v = atol(e);
     if (v >= 1 && v <= 4096) s->ivb = (uint64_t)v;
   }
-  s->n_batches = (s->n_ivs + s->ivb - 1) / s->ivb;
+  s->n_batches = s->d_batch_lo ? n_blocks
+                               : (s->n_ivs + s->ivb - 1) / s->ivb;
   uint64_t want = (s->n_batches + kThreads - 1) / kThreads;
   uint64_t cap = 8192;
   if (const char* g = getenv("YBG_GRID")) {
@@ -1300,7 +1309,13 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
   uint64_t span_threads = (uint64_t)s->grid * kThreads;
   s->n_partials = span_threads / 64;
   s->n_heads = s->n_batches;  // per-batch head records (global resolution)
-  s->n_gheads = (s->n_batches + kThreads - 1) / kThreads;
+  {
+    // the GROUP path always uses fixed-ivb batches: size its relay
+    // arrays for whichever decomposition is larger
+    uint64_t n_fixed_b = (s->n_ivs + s->ivb - 1) / s->ivb;
+    uint64_t n_gb = n_fixed_b > s->n_batches ? n_fixed_b : s->n_batches;
+    s->n_gheads = (n_gb + kThreads - 1) / kThreads;
+  }
   s->na_cap = s->dspec.num_aggs <= 2 ? 2 : (s->dspec.num_aggs <= 4 ? 4 : 8);
   s->hstride = 2 * s->na_cap + 2;
   HIP_TRY(hipMalloc(&s->d_rk_save, span_threads * kKeyCap));
@@ -1502,7 +1517,8 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
       hipLaunchKernelGGL(kern, dim3(s->grid), dim3(kThreads), 0, s->stream,
                          s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
                          s->d_partials, s->d_heads, s->d_walked, s->ivb,
-                         s->d_retry, s->d_retry_n);
+                         s->n_batches, s->d_batch_lo, s->d_retry,
+                         s->d_retry_n);
     };
     int fwps = wps;
     if (const char* e = getenv("YBG_FWPS")) {
@@ -1525,14 +1541,15 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
                        s->d_aux, s->d_rk_save,
                        s->d_partials + s->n_partials * kPartialStride,
                        s->d_heads, s->d_walked, nullptr, 0, s->ivb,
-                       s->d_retry, s->d_retry_n);
+                       s->n_batches, s->d_batch_lo, s->d_retry,
+                       s->d_retry_n);
   } else {
     auto launch = [&](auto kern) {
       hipLaunchKernelGGL(kern, dim3(s->grid), dim3(kThreads), 0, s->stream,
                          s->d_data, s->d_offsets, s->d_ivs, s->n_ivs,
                          s->d_aux, s->d_rk_save, s->d_partials, s->d_heads,
-                         s->d_walked, nullptr, 0, s->ivb,
-                         (const uint64_t*)nullptr,
+                         s->d_walked, nullptr, 0, s->ivb, s->n_batches,
+                         s->d_batch_lo, (const uint64_t*)nullptr,
                          (const unsigned long long*)nullptr);
     };
     switch (na * 10 + wps) {
@@ -1647,6 +1664,7 @@ int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
                      s->stream, s->d_data, s->d_offsets, s->d_ivs,
                      s->n_ivs, s->d_aux, s->d_rk_save, s->d_partials,
                      s->d_heads, s->d_walked, s->d_flags_all, 1, s->ivb,
+                     s->n_batches, s->d_batch_lo,
                      (const uint64_t*)nullptr,
                      (const unsigned long long*)nullptr);
   EmitCtx ec;
@@ -1883,6 +1901,7 @@ int yb_gpu_scan_close(ybg_scan_t* s) {
   if (s->d_data_owned && s->d_data) HIP_WARN(hipFree(s->d_data));
   if (s->d_offsets) HIP_WARN(hipFree(s->d_offsets));
   if (s->d_ivs) HIP_WARN(hipFree(s->d_ivs));
+  if (s->d_batch_lo) HIP_WARN(hipFree(s->d_batch_lo));
   if (s->d_aux) HIP_WARN(hipFree(s->d_aux));
   if (s->d_rk_save) HIP_WARN(hipFree(s->d_rk_save));
   if (s->d_partials) HIP_WARN(hipFree(s->d_partials));
