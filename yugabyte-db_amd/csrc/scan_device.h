@@ -2469,6 +2469,17 @@ inline bool fast_eligible(const DevSpec& d) {
 #define YBG_FABORT(n) return 0
 #endif
 
+// YBG_ABL: perf-forensics ablations of scan_batch_fast (results WRONG by
+// design — never in a default build): 1 = skip register-tail patches and
+// force visibility true; 2 = skip packed-value decode; 3 = skip the
+// per-row finalize accumulation.
+#ifndef YBG_ABL
+#define YBG_ABL 0
+#endif
+#ifndef YBG_VWIN
+#define YBG_VWIN 0
+#endif
+
 // Returns 1 = batch done (accumulators updated), 0 = abort (accumulators
 // and ho untouched except bht[3..5] restart candidates, which are benign
 // duplicates under the MIN fold when the batch is retried).
@@ -2689,10 +2700,12 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
                                    : 0;
         for (uint32_t i = 0; i < lds2; ++i)
           key[ns2s + i] = body_byte(hl + ns1 + i);
+#if YBG_ABL != 1
         for (uint32_t i = 0; i < ns1; ++i)
           tail_patch(&thi, &tlo, ukey, sp + i, body_byte(hl + i));
         for (uint32_t i = 0; i < ns2; ++i)
           tail_patch(&thi, &tlo, ukey, ns2s + i, body_byte(hl + ns1 + i));
+#endif
 #else
         uint32_t lds_n = sp < rkb ? (rkb - sp < ns1 ? rkb - sp : ns1) : 0;
         for (uint32_t i = 0; i < lds_n; ++i)
@@ -2757,7 +2770,11 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
     // row boundary
     if (kchg) {
       if (row_open) {
+#if YBG_ABL == 3
+        if (false) {
+#else
         if (found) {
+#endif
           bool hit = (pred_pass & sp.value_pred_mask) == sp.value_pred_mask;
           if (in_head) {
             ho->scanned += 1;
@@ -2806,9 +2823,13 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
     const uint32_t vb0 =
         er.value_len ? (uint32_t)(rdr.peek8() & 0xff) : 0u;
     if (YBG_UNLIKELY(vb0 == kHybridTimeByte)) YBG_FABORT(23);
+#if YBG_ABL == 1
+    const bool visible = true;
+#else
     const bool visible =
         u128_slice_cmp(ht_hi, ht_lo, ht_sz, sp.reg_lim.hi, sp.reg_lim.lo,
                        sp.reg_lim.len) >= 0;
+#endif
     if (visible) {
       if (YBG_UNLIKELY(sp.track_restart)) {
         if (u128_slice_cmp(ht_hi, ht_lo, ht_sz, sp.read.hi, sp.read.lo,
@@ -2824,9 +2845,33 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
       }
       if (!base_seen) {
         base_seen = true;
+#if YBG_ABL == 2
+        if (true) {
+          found = true;
+          pred_pass = sp.value_pred_mask;
+          agg_null = 0;
+          rdr.seek(q);
+          p = q;
+          continue;
+        } else
+#endif
         if (YBG_LIKELY(er.value_len == sp.v2_fixed_len && vb0 == kPackedV2B &&
                        (rdr.peek8() & 0xff8000u) == 0)) {
           const uint8_t* value = rdr.pos();
+#if YBG_VWIN
+          // Window-resident column extraction: the value bytes mostly sit
+          // in the reader registers already — peek them (pure VALU, no
+          // load latency) and fall back to one unaligned load per column
+          // beyond the window. Ablation showed the value extract/eval is
+          // ~half the kernel time, dominated by the dependent L1 loads
+          // this removes.
+          rdr.align8();
+          for (int i = 0; i < sp.num_value_cols; ++i) {
+            const uint32_t act = sp.col_act[i];
+            const uint32_t off = sp.v2_off[i];
+            uint64_t u = off <= 17 ? rdr.peek8_at(off)
+                                   : load_u64_una(value + off);
+#else
           rdr.seek(value + er.value_len);  // next window loads issue now
           // fixed-offset column extraction (decode_packed_v2_fixed, lean
           // eval: typed compares + aggregate capture only)
@@ -2839,6 +2884,7 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
             const uint32_t wi = ob >> 3, sh = (ob & 7) * 8;
             uint64_t u = qw[wi];
             if (sh) u = (u >> sh) | (qw[wi + 1] << (64 - sh));
+#endif
             const uint32_t dt = (act >> kActDtShift) & kActDtM;
             switch ((act >> kActV2Shift) & kActV2M) {
               case 1:
@@ -2876,6 +2922,9 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
             }
           }
           found = true;
+#if YBG_VWIN
+          rdr.seek(value + er.value_len);
+#endif
           p = q;
           continue;  // reader already past the value
         } else if (er.value_len == 1 && vb0 == kTombB) {
