@@ -2476,9 +2476,6 @@ inline bool fast_eligible(const DevSpec& d) {
 #ifndef YBG_ABL
 #define YBG_ABL 0
 #endif
-#ifndef YBG_VWIN
-#define YBG_VWIN 0
-#endif
 
 // Returns 1 = batch done (accumulators updated), 0 = abort (accumulators
 // and ho untouched except bht[3..5] restart candidates, which are benign
@@ -2858,20 +2855,6 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
         if (YBG_LIKELY(er.value_len == sp.v2_fixed_len && vb0 == kPackedV2B &&
                        (rdr.peek8() & 0xff8000u) == 0)) {
           const uint8_t* value = rdr.pos();
-#if YBG_VWIN
-          // Window-resident column extraction: the value bytes mostly sit
-          // in the reader registers already — peek them (pure VALU, no
-          // load latency) and fall back to one unaligned load per column
-          // beyond the window. Ablation showed the value extract/eval is
-          // ~half the kernel time, dominated by the dependent L1 loads
-          // this removes.
-          rdr.align8();
-          for (int i = 0; i < sp.num_value_cols; ++i) {
-            const uint32_t act = sp.col_act[i];
-            const uint32_t off = sp.v2_off[i];
-            uint64_t u = off <= 17 ? rdr.peek8_at(off)
-                                   : load_u64_una(value + off);
-#else
           rdr.seek(value + er.value_len);  // next window loads issue now
           // fixed-offset column extraction (decode_packed_v2_fixed, lean
           // eval: typed compares + aggregate capture only)
@@ -2884,7 +2867,6 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
             const uint32_t wi = ob >> 3, sh = (ob & 7) * 8;
             uint64_t u = qw[wi];
             if (sh) u = (u >> sh) | (qw[wi + 1] << (64 - sh));
-#endif
             const uint32_t dt = (act >> kActDtShift) & kActDtM;
             switch ((act >> kActV2Shift) & kActV2M) {
               case 1:
@@ -2922,9 +2904,6 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
             }
           }
           found = true;
-#if YBG_VWIN
-          rdr.seek(value + er.value_len);
-#endif
           p = q;
           continue;  // reader already past the value
         } else if (er.value_len == 1 && vb0 == kTombB) {
