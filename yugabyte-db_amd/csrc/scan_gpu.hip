@@ -1191,7 +1191,9 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
   // key range intersects the scan's bounds / leading-key option ranges
   std::vector<uint8_t> pruned_blob;
   std::vector<uint64_t> pruned_off;
-  if (!device && n_blocks) {
+  bool noprune = false;
+  if (const char* e = getenv("YBG_NOPRUNE")) noprune = atoi(e) != 0;
+  if (!device && n_blocks && !noprune) {
     std::vector<uint8_t> keep;
     if (compute_block_selection(s->spec, blocks, offsets, n_blocks,
                                 &keep)) {
