@@ -123,11 +123,21 @@ def main():
         # skipped (SkipFutureRecords semantics), the newest visible one is
         # deduped — the visibility work config 4 names, not just dedup
         read_micros = 1_600_000_002_500_000
-    else:  # groupby — BASELINE configs[4]-shaped (mixed row, grouped)
+    else:  # groupby — BASELINE configs[4]: mixed row incl. the 32-byte
+        # string column, range predicates over the int64 and the string
+        # (string EQUALITY parity is pinned in the test suite; over
+        # uniform-random 32-byte strings an equality predicate selects
+        # ~nothing, which would leave the grouped aggregates empty)
         schema = y.make_schema(
             [y.KT_INT64],
-            [(10, y.T_INT64, 1), (11, y.T_INT64, 1), (12, y.T_DOUBLE, 1)])
-        preds = [y.Pred(0, 1, y.PRED_GT, 1 << 38, None, 0)]
+            [(10, y.T_INT64, 1), (11, y.T_INT64, 1), (12, y.T_DOUBLE, 1),
+             (13, y.T_STRING, 1)])
+        _slo = ctypes.create_string_buffer(b"m" * 32, 32)
+        preds = [y.Pred(0, 1, y.PRED_GT, 1 << 38, None, 0),
+                 y.Pred(0, 1, y.PRED_LT, 3 << 38, None, 0),
+                 y.Pred(0, 3, y.PRED_GE, 0,
+                        ctypes.cast(_slo, ctypes.POINTER(ctypes.c_uint8)),
+                        32)]
         aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 1)]
         group_col = 1  # 1 + value col 0
     n_tablets = 1 if n_gpus == 1 and world <= 1 else 8
