@@ -1325,7 +1325,8 @@ int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
                     ((s->n_batches + kThreads - 1) / kThreads) *
                         sizeof(uint32_t)));
   HIP_TRY(hipMalloc(&s->d_result, sizeof(DevResult)));
-  HIP_TRY(hipMalloc(&s->d_chunk, 256 * kPartialStride * sizeof(uint64_t)));
+  HIP_TRY(hipMalloc(&s->d_chunk,
+                    1024 * kPartialStride * sizeof(uint64_t)));
   return 0;
 }
 
@@ -1572,11 +1573,11 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
     }
   }
   HIP_TRY(hipEventRecord(s->ev_mid, s->stream));
-  hipLaunchKernelGGL(k_reduce_pre, dim3(256), dim3(256), 0, s->stream,
+  hipLaunchKernelGGL(k_reduce_pre, dim3(1024), dim3(256), 0, s->stream,
                      s->dspec, s->d_partials, 2 * s->n_partials, s->d_heads,
                      s->d_walked, n_bat, s->hstride, s->d_chunk);
   hipLaunchKernelGGL(k_reduce, dim3(1), dim3(256), 0, s->stream, s->dspec,
-                     s->d_chunk, 256, s->d_heads, s->d_cont, 0,
+                     s->d_chunk, 1024, s->d_heads, s->d_cont, 0,
                      s->d_result);
   HIP_TRY(hipEventRecord(s->ev_end, s->stream));
   s->executed = true;
