@@ -1493,8 +1493,11 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
       s->n_partials * kPartialStride * sizeof(uint64_t), s->stream));
   HIP_TRY(hipEventRecord(s->ev_start, s->stream));
   // dispatch on aggregate-slot capacity (register footprint) and the
-  // waves-per-SIMD occupancy bound (YBG_WPS for tuning; default 3 measured)
-  int wps = 3;
+  // waves-per-SIMD occupancy bound (YBG_WPS for tuning; default 3
+  // measured for the 2-agg shape — the 4-agg kernel at 3 waves carries
+  // 240 B/lane of scratch spills vs 0 at 2 waves, so it defaults to 2;
+  // the 8-agg kernel spills either way and keeps 3)
+  int wps = s->na_cap == 4 ? 2 : 3;
   if (const char* e = getenv("YBG_WPS")) {
     long v = atol(e);
     if (v >= 2 && v <= 6) wps = (int)v;
