@@ -188,3 +188,36 @@ def test_ql_row_form():
         n += 1
     lib.yb_host_iter_close(h)
     assert n == 300
+
+
+@pytest.mark.gpu
+def test_backward_over_merged_intents():
+    """Backward delivery over a feed-time-merged tablet (intents +
+    backward compose): descending rows equal the reversed oracle
+    runtime-merge row set."""
+    gpu_scan = _gpu()
+    lib = gpu_scan._lib()
+    schema = y.make_schema([y.KT_INT64],
+                           [(10, y.T_INT64, 1), (11, y.T_INT64, 1)])
+    b = y.Builder(schema)
+    seq = 1 << 50
+    for r in range(800):
+        seq += 1
+        b.add_packed_row(1000, [(y.T_INT64, r), (y.T_INT64, r * 2)],
+                         hash_=r // 256, key_datums=(r,), seq=seq)
+    built = b.finish()
+    it = y.Intents(schema)
+    it.add_packed_row(1500, 1, [(y.T_INT64, 5000), (y.T_INT64, 1)],
+                      hash_=3, key_datums=(5000,))
+    it.add_row_tombstone(1500, 1, hash_=0, key_datums=(10,))
+    blob, blen = it.blob()
+    txns, ntx = y.make_txns({1: ("c", 2000)})
+    mb, mo, mn, mt = y.merge_intents(built[0], built[1], built[2], blob,
+                                     blen, txns, ntx)
+    spec = _spec(schema, read=3000, backward=1)
+    rows_b, _ = _drain(lib, spec, (mb, mo, mn), schema)
+    spec_f = _spec(schema, read=3000)
+    rows_f, _ = _drain(lib, spec_f, (mb, mo, mn), schema)
+    assert rows_b == list(reversed(rows_f))
+    assert len(rows_b) == 800  # +1 inserted, -1 tombstoned
+    assert rows_b[0][0][0] == 5000  # highest key first

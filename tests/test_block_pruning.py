@@ -105,6 +105,33 @@ def test_in_range_filter_parity():
     assert sres.rows_matched < sres.rows_scanned
 
 
+def test_in_range_double_column():
+    """IN_RANGE over a DOUBLE column (FP range semantics)."""
+    schema = y.make_schema([y.KT_INT64],
+                           [(10, y.T_DOUBLE, 1), (11, y.T_INT64, 1)])
+    b = y.Builder(schema)
+    seq = 1 << 50
+    for r in range(4000):
+        seq += 1
+        b.add_packed_row(1000, [(y.T_DOUBLE, r * 0.25), (y.T_INT64, r)],
+                         hash_=r // 512, key_datums=(r,), seq=seq)
+    built = b.finish()
+    lo = struct.unpack("<Q", struct.pack("<d", 100.0))[0]
+    hi = struct.unpack("<Q", struct.pack("<d", 150.0))[0]
+    blob = struct.pack("<QQII", lo, hi, 1, 0)  # [100.0, 150.0)
+    buf = C.create_string_buffer(blob, len(blob))
+    preds = [y.Pred(0, 0, y.PRED_IN_RANGE, 0,
+                    C.cast(buf, C.POINTER(C.c_uint8)), len(blob))]
+    aggs = [y.Agg(y.AGG_COUNT_STAR, 0), y.Agg(y.AGG_SUM_INT64, 1)]
+    spec = _spec(schema, preds, aggs)
+    sres = y.sim_scan(spec, built[0], built[1], built[2])
+    ores = _oracle_run(schema, (built[0], built[1], built[2]), preds, aggs)
+    # rows with 100.0 <= r*0.25 < 150.0: r in [400, 600)
+    assert sres.rows_matched == ores.rows_matched == 200
+    assert sres.aggs[1].value_i64 == ores.aggs[1].value_i64 \
+        == sum(range(400, 600))
+
+
 def test_option_pruning_selective_scan():
     """<1%-selectivity IN options on the leading range key: most blocks
     are pruned; scanning only the kept subset is bit-exact with the

@@ -232,3 +232,39 @@ def test_restart_via_global_limit_only():
     r = _cross_check(schema, built, it, {9: ("c", 2000)}, 1500, 1500, 3500)
     assert r.rows_scanned == 61
     assert r.restart_ht_len > 0
+
+
+def test_intents_shared_prefix_format():
+    """The feed-time merge re-encodes affected blocks in the tablet's KV
+    format — cover ENC_SHARED_PREFIX end to end."""
+    schema = y.make_schema([y.KT_INT64],
+                           [(10, y.T_INT64, 1), (11, y.T_INT64, 1)])
+    b = y.Builder(schema, kv_format=y.ENC_SHARED_PREFIX)
+    seq = 1 << 50
+    for r in range(500):
+        seq += 1
+        b.add_packed_row(1000, [(y.T_INT64, r), (y.T_INT64, r * 2)],
+                         hash_=r // 256, key_datums=(r,), seq=seq)
+    built = b.finish()
+    it = y.Intents(schema)
+    it.add_packed_row(1500, 1, [(y.T_INT64, 100), (y.T_INT64, 424242)],
+                      hash_=0, key_datums=(100,))
+    blob, blen = it.blob()
+    txns, ntx = y.make_txns({1: ("c", 2000)})
+    mb, mo, mn, mt = y.merge_intents(built[0], built[1], built[2], blob,
+                                     blen, txns, ntx,
+                                     kv_format=y.ENC_SHARED_PREFIX)
+    spec = _spec(schema, 3000,
+                 aggs=[y.Agg(y.AGG_COUNT_STAR, 0),
+                       y.Agg(y.AGG_SUM_INT64, 1)])
+    spec.kv_format = y.ENC_SHARED_PREFIX
+    sres = y.sim_scan(spec, mb, mo, mn)
+    osc = y.orcl_schema_from(schema)
+    ospec = _orcl_spec(3000, aggs=[y.Agg(y.AGG_COUNT_STAR, 0),
+                                   y.Agg(y.AGG_SUM_INT64, 1)])
+    ores = y.orcl_scan_intents(built[0], built[1], built[2], osc, ospec,
+                               blob, blen, txns, ntx,
+                               kv_format=y.ENC_SHARED_PREFIX)
+    assert _res(sres) == _res(ores)
+    assert sres.aggs[1].value_i64 == sum(r * 2 for r in range(500)) \
+        - 200 + 424242
