@@ -374,8 +374,8 @@ __global__ void k_group_init(DevSpec sp, GroupCtx gc) {
 // in-kernel once the lane relay resolves ownership, with workgroup-first
 // intervals going through the global record + cont-flag path
 // (k_group_heads folds those).
-template <int NA>
-__global__ __launch_bounds__(kThreads, 3) void k_group(
+template <int NA, int WPS>
+__global__ __launch_bounds__(kThreads, WPS) void k_group(
     const uint8_t* __restrict__ data,
     const uint64_t* __restrict__ block_offsets,
     const Interval* __restrict__ ivs, uint64_t n_ivs,
@@ -1795,22 +1795,28 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
                            s->n_partials * kPartialStride * sizeof(uint64_t),
                            s->stream));
   }
+  int gwps = 3;
+  if (const char* e = getenv("YBG_GWPS")) {
+    long v = atol(e);
+    if (v == 2 || v == 3) gwps = (int)v;
+  }
+  auto launch_group = [&](auto kern, auto kern_heads) {
+    hipLaunchKernelGGL(kern, dim3(s->grid), dim3(kThreads), 0,
+                       s->stream, s->d_data, s->d_offsets, s->d_ivs,
+                       s->n_ivs, s->d_aux, s->d_rk_save, s->gc, s->d_gheads,
+                       s->d_cont, s->gc.overflow, s->d_partials, s->ivb);
+    hipLaunchKernelGGL(kern_heads, dim3(hgrid), dim3(kThreads), 0,
+                       s->stream, s->gc, s->d_gheads, s->d_cont,
+                       s->n_gheads);
+  };
   if (s->dspec.num_aggs <= 4) {
-    hipLaunchKernelGGL(k_group<4>, dim3(s->grid), dim3(kThreads), 0,
-                       s->stream, s->d_data, s->d_offsets, s->d_ivs,
-                       s->n_ivs, s->d_aux, s->d_rk_save, s->gc, s->d_gheads,
-                       s->d_cont, s->gc.overflow, s->d_partials, s->ivb);
-    hipLaunchKernelGGL(k_group_heads<4>, dim3(hgrid), dim3(kThreads), 0,
-                       s->stream, s->gc, s->d_gheads, s->d_cont,
-                       s->n_gheads);
+    auto kh = k_group_heads<4>;
+    if (gwps == 2) launch_group(k_group<4, 2>, kh);
+    else launch_group(k_group<4, 3>, kh);
   } else {
-    hipLaunchKernelGGL(k_group<8>, dim3(s->grid), dim3(kThreads), 0,
-                       s->stream, s->d_data, s->d_offsets, s->d_ivs,
-                       s->n_ivs, s->d_aux, s->d_rk_save, s->gc, s->d_gheads,
-                       s->d_cont, s->gc.overflow, s->d_partials, s->ivb);
-    hipLaunchKernelGGL(k_group_heads<8>, dim3(hgrid), dim3(kThreads), 0,
-                       s->stream, s->gc, s->d_gheads, s->d_cont,
-                       s->n_gheads);
+    auto kh = k_group_heads<8>;
+    if (gwps == 2) launch_group(k_group<8, 2>, kh);
+    else launch_group(k_group<8, 3>, kh);
   }
   if (s->dspec.track_restart) {
     // fold the wave restart minima into DevResult (num_aggs = 0 spec: only
