@@ -218,8 +218,11 @@ def test_gpu_pruned_option_scan():
     ores = _oracle_run(schema, built, preds, aggs)
     assert g.rows_matched == ores.rows_matched == 3
     assert g.aggs[1].value_i64 == ores.aggs[1].value_i64
-    # pruned: the scan decoded a tiny fraction of the tablet
-    assert g.entries_seen < 2000
+    # pruned: the scan decoded a tiny fraction of the tablet (unless the
+    # A/B knob disabled the selection)
+    import os
+    if not os.environ.get("YBG_NOPRUNE"):
+        assert g.entries_seen < 2000
 
 
 @pytest.mark.gpu
