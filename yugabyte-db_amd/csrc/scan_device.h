@@ -2457,6 +2457,7 @@ inline bool fast_eligible(const DevSpec& d) {
   if (d.fmt != YBG_ENC_THREE_SHARED_PARTS) return false;
   if (!d.fixed_rk_len || d.need_rowkey || d.group_col >= 0) return false;
   if (!d.v2_fixed_len || d.num_value_cols <= 0) return false;
+  if (d.num_value_cols > 8) return false;  // NC template cap
   if (d.num_aggs > 2) return false;
   for (int i = 0; i < d.num_preds; ++i) {
     if (d.preds[i].is_key_col) return false;
@@ -2480,7 +2481,14 @@ inline bool fast_eligible(const DevSpec& d) {
 // Returns 1 = batch done (accumulators updated), 0 = abort (accumulators
 // and ho untouched except bht[3..5] restart candidates, which are benign
 // duplicates under the MIN fold when the batch is retried).
-template <int NA>
+// NC = compile-time value-column cap (4 or 8, >= sp.num_value_cols): the
+// packed-row column loads unroll to NC branch-free word-pair loads that
+// issue back-to-back, so a row costs ONE L1 latency instead of one per
+// column (a runtime-bound loop cannot unroll and serializes the loads on
+// the per-iteration vmcnt wait). Columns [num_value_cols, NC) are
+// zero-padded in the spec (col_act = 0, v2_off = 0) and evaluate to
+// no-ops.
+template <int NA, int NC>
 DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
                         const uint64_t* block_offsets, const Interval* ivs,
                         uint64_t n_ivs, uint64_t j_lo, uint64_t j_hi,
@@ -2857,16 +2865,24 @@ DEV int scan_batch_fast(const DevSpec& sp, const uint8_t* data,
           const uint8_t* value = rdr.pos();
           rdr.seek(value + er.value_len);  // next window loads issue now
           // fixed-offset column extraction (decode_packed_v2_fixed, lean
-          // eval: typed compares + aggregate capture only)
+          // eval) in two fully-unrolled passes: a branch-free load pass
+          // issuing all NC word pairs, then the eval pass over registers
+          // (see the NC doc on scan_batch_fast)
           const uintptr_t a = (uintptr_t)value;
           const uint64_t* qw = (const uint64_t*)(a & ~(uintptr_t)7);
           const uint32_t abase = (uint32_t)(a & 7);
-          for (int i = 0; i < sp.num_value_cols; ++i) {
-            const uint32_t act = sp.col_act[i];
+          uint64_t uv[NC];
+#pragma unroll
+          for (int i = 0; i < NC; ++i) {
             const uint32_t ob = abase + sp.v2_off[i];
-            const uint32_t wi = ob >> 3, sh = (ob & 7) * 8;
-            uint64_t u = qw[wi];
-            if (sh) u = (u >> sh) | (qw[wi + 1] << (64 - sh));
+            const uint32_t wi = ob >> 3, shb = (ob & 7) * 8;
+            const uint64_t w0 = qw[wi], w1 = qw[wi + 1];
+            uv[i] = shb ? (w0 >> shb) | (w1 << (64 - shb)) : w0;
+          }
+#pragma unroll
+          for (int i = 0; i < NC; ++i) {
+            const uint32_t act = sp.col_act[i];
+            uint64_t u = uv[i];
             const uint32_t dt = (act >> kActDtShift) & kActDtM;
             switch ((act >> kActV2Shift) & kActV2M) {
               case 1:

@@ -190,7 +190,7 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
 // same global bheads/walked protocol.
 // ---------------------------------------------------------------------------
 
-template <int WPS>
+template <int WPS, int NC>
 __global__ __launch_bounds__(kThreads, WPS) void k_scan_fast(
     const uint8_t* __restrict__ data,
     const uint64_t* __restrict__ block_offsets,
@@ -222,7 +222,7 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan_fast(
     const uint64_t lo = batch_lo ? batch_lo[j] : j * ivb;
     uint64_t hi = batch_lo ? batch_lo[j + 1] : lo + ivb;
     if (hi > n_ivs) hi = n_ivs;
-    if (!scan_batch_fast<NA>(sp, data, block_offsets, ivs, n_ivs, lo, hi,
+    if (!scan_batch_fast<NA, NC>(sp, data, block_offsets, ivs, n_ivs, lo, hi,
                              key, rmin, &entries, &scanned, &matched,
                              agg_val, agg_cnt, &ho, &walked_next)) {
       unsigned long long slot = atomicAdd(retry_n, 1ull);
@@ -1531,13 +1531,22 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
       long v = atol(e);
       if (v >= 2 && v <= 8) fwps = (int)v;
     }
+    // NC: compile-time column cap for the unrolled load pass (spec is
+    // zero-padded up to it; fast_eligible caps num_value_cols at 8)
+    const bool nc4 = s->dspec.num_value_cols <= 4;
+#define YBG_LAUNCH_FAST(W) \
+  do { \
+    if (nc4) launchf(k_scan_fast<W, 4>); \
+    else launchf(k_scan_fast<W, 8>); \
+  } while (0)
     switch (fwps) {
-      case 2: launchf(k_scan_fast<2>); break;
-      case 4: launchf(k_scan_fast<4>); break;
-      case 5: launchf(k_scan_fast<5>); break;
-      case 6: launchf(k_scan_fast<6>); break;
-      default: launchf(k_scan_fast<3>); break;
+      case 2: YBG_LAUNCH_FAST(2); break;
+      case 4: YBG_LAUNCH_FAST(4); break;
+      case 5: YBG_LAUNCH_FAST(5); break;
+      case 6: YBG_LAUNCH_FAST(6); break;
+      default: YBG_LAUNCH_FAST(3); break;
     }
+#undef YBG_LAUNCH_FAST
     // retry pass: the general kernel over the aborted batch list, into
     // the second partials half
     int rg = std::min(s->grid, 1024);

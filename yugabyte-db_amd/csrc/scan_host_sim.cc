@@ -173,9 +173,16 @@ int ybg_sim_scan_fast(const ybg_scan_spec_t* spec, const uint8_t* data,
     uint32_t e32 = 0, s32 = 0, m32 = 0;
     uint64_t lo = b * ivb;
     uint64_t hi = lo + ivb < n_ivs ? lo + ivb : n_ivs;
-    int rc = scan_batch_fast<2>(d, data, offsets, ivs.data(), n_ivs, lo, hi,
-                                key, bht + 3, &e32, &s32, &m32, agg_val,
-                                agg_cnt, &heads[b], &wn);
+    // NC mirrors the GPU dispatch: 4-column instantiation when it fits
+    int rc = d.num_value_cols <= 4
+                 ? scan_batch_fast<2, 4>(d, data, offsets, ivs.data(), n_ivs,
+                                         lo, hi, key, bht + 3, &e32, &s32,
+                                         &m32, agg_val, agg_cnt, &heads[b],
+                                         &wn)
+                 : scan_batch_fast<2, 8>(d, data, offsets, ivs.data(), n_ivs,
+                                         lo, hi, key, bht + 3, &e32, &s32,
+                                         &m32, agg_val, agg_cnt, &heads[b],
+                                         &wn);
     if (!rc) {
       ++fallbacks;
       uint64_t av8[YBG_MAX_AGGS] = {0}, ac8[YBG_MAX_AGGS] = {0};
