@@ -1800,15 +1800,24 @@ DEV void decode_packed_v2_fixed(const DevSpec& sp, const uint8_t* aux,
   // one shared alignment for the whole body: every column extract is a
   // funnel shift over two aligned words (the words are shared between
   // adjacent columns instead of re-deriving per-column unaligned loads)
+  // software-pipelined one deep: column i+1's word pair issues BEFORE
+  // column i's eval, so the (uniform-trip) loop hides each L1 latency
+  // behind the previous column's eval work instead of serializing on it
   const uintptr_t a = (uintptr_t)value;
   const uint64_t* qw = (const uint64_t*)(a & ~(uintptr_t)7);
   const uint32_t abase = (uint32_t)(a & 7);
-  for (int i = 0; i < sp.num_value_cols; ++i) {
+  const int n = sp.num_value_cols;
+  uint32_t ob = abase + sp.v2_off[0];
+  uint64_t w0 = qw[ob >> 3], w1 = qw[(ob >> 3) + 1];
+  for (int i = 0; i < n; ++i) {
     const uint32_t act = sp.col_act[i];
-    const uint32_t ob = abase + sp.v2_off[i];
-    const uint32_t wi = ob >> 3, sh = (ob & 7) * 8;
-    uint64_t u = qw[wi];
-    if (sh) u = (u >> sh) | (qw[wi + 1] << (64 - sh));
+    const uint32_t sh = (ob & 7) * 8;
+    uint64_t u = sh ? (w0 >> sh) | (w1 << (64 - sh)) : w0;
+    if (i + 1 < n) {
+      ob = abase + sp.v2_off[i + 1];
+      w0 = qw[ob >> 3];
+      w1 = qw[(ob >> 3) + 1];
+    }
     const uint32_t dt = (act >> kActDtShift) & kActDtM;
     switch ((act >> kActV2Shift) & kActV2M) {
       case 1:
