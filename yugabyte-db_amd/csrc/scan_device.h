@@ -218,6 +218,12 @@ struct DevSpec {
   // ('|' + version + flags + fixed bodies; single-byte version assumed) and
   // per-column body offsets from the value start. 0 = fast path off.
   uint32_t v2_fixed_len;
+  // leading columns with STATIC offsets (everything before the first
+  // varlen column, offset-capped at 255): v2_off[0..v2_nfp) are valid and
+  // column v2_nfp starts at v2_tail_off. v2_nfp == num_value_cols when the
+  // whole schema is fixed-width.
+  int32_t v2_nfp;
+  uint32_t v2_tail_off;
   uint8_t v2_off[YBG_MAX_COLS];
   // Read-restart tracking (intent_aware_iterator.cc:815-827 UpdateMaxSeenHt
   // + GetReadRestartData :1400-1410): only possible when local_limit >
@@ -1838,6 +1844,90 @@ DEV void decode_packed_v2_fixed(const DevSpec& sp, const uint8_t* aux,
   }
 }
 
+// Direct-load packed-V2 decode for schemas whose leading v2_nfp columns
+// are fixed-width: the prefix decodes from static offsets with the same
+// software-pipelined word-pair loads as decode_packed_v2_fixed, and the
+// varlen tail walks with direct unaligned loads — no window-consume
+// chain at all (the serial chain is what bounds wide mixed schemas).
+// Caller guarantees ver < 128, flags == 0, v2_nfp > 0; value points at
+// the kPackedV2B byte. Returns false on corruption.
+template <int NA>
+DEV bool decode_packed_v2_mixed(const DevSpec& sp, const uint8_t* aux,
+                                const uint8_t* value, uint32_t value_len,
+                                RowCtxT<NA>* rc) {
+  const uint8_t* end = value + value_len;
+  const int nfp = sp.v2_nfp, n = sp.num_value_cols;
+  if (YBG_UNLIKELY(value + sp.v2_tail_off > end)) return false;
+  const uintptr_t a = (uintptr_t)value;
+  const uint64_t* qw = (const uint64_t*)(a & ~(uintptr_t)7);
+  const uint32_t abase = (uint32_t)(a & 7);
+  uint32_t ob = abase + sp.v2_off[0];
+  uint64_t w0 = qw[ob >> 3], w1 = qw[(ob >> 3) + 1];
+  for (int i = 0; i < nfp; ++i) {
+    const uint32_t act = sp.col_act[i];
+    const uint32_t sh = (ob & 7) * 8;
+    uint64_t u = sh ? (w0 >> sh) | (w1 << (64 - sh)) : w0;
+    if (i + 1 < nfp) {
+      ob = abase + sp.v2_off[i + 1];
+      w0 = qw[ob >> 3];
+      w1 = qw[(ob >> 3) + 1];
+    }
+    const uint32_t dt = (act >> kActDtShift) & kActDtM;
+    switch ((act >> kActV2Shift) & kActV2M) {
+      case 1:
+        u = (dt == YBG_T_INT8) ? (uint64_t)(int64_t)(int8_t)u : (u & 0xff);
+        break;
+      case 2:
+        u = (dt == YBG_T_INT16) ? (uint64_t)(int64_t)(int16_t)u
+                                : (u & 0xffff);
+        break;
+      case 4:
+        u = (dt == YBG_T_INT32) ? (uint64_t)(int64_t)(int32_t)u
+                                : (u & 0xffffffffull);
+        break;
+      default:
+        break;
+    }
+    eval_col(sp, aux, rc, i, false, u, nullptr, 0);
+  }
+  const uint8_t* p = value + sp.v2_tail_off;
+  for (int i = nfp; i < n; ++i) {
+    const uint32_t act = sp.col_act[i];
+    const uint32_t fw = (act >> kActV2Shift) & kActV2M;
+    uint64_t u;
+    memcpy(&u, p, 8);  // unaligned; block tail slack covers the overread
+    if (fw) {
+      if (YBG_UNLIKELY(p + fw > end)) return false;
+      switch (fw) {
+        case 1: u &= 0xff; break;
+        case 2: u &= 0xffff; break;
+        case 4: u &= 0xffffffffull; break;
+        default: break;
+      }
+      const uint32_t dt = (act >> kActDtShift) & kActDtM;
+      if (dt == YBG_T_INT8) u = (uint64_t)(int64_t)(int8_t)u;
+      else if (dt == YBG_T_INT16) u = (uint64_t)(int64_t)(int16_t)u;
+      else if (dt == YBG_T_INT32) u = (uint64_t)(int64_t)(int32_t)u;
+      eval_col(sp, aux, rc, i, false, u, nullptr, 0);
+      p += fw;
+    } else {
+      if (YBG_UNLIKELY(p >= end)) return false;
+      uint32_t flen;
+      if ((u & 1) == 0) {
+        flen = (uint32_t)((u & 0xff) >> 1);
+        p += 1;
+      } else {
+        flen = (uint32_t)((u & 0xffffffffull) >> 1);
+        p += 4;
+      }
+      if (YBG_UNLIKELY(p + flen > end)) return false;
+      eval_col(sp, aux, rc, i, false, 0, p, flen);
+      p += flen;
+    }
+  }
+  return true;
+}
+
 // Visibility + row-state update for one entry. key/rowkey live in LDS.
 // rdr is positioned at the value start. Returns false on corruption.
 template <int NA>
@@ -1908,7 +1998,16 @@ DEV bool process_entry(const DevSpec& sp, const uint8_t* base,
           rc->found = true;
           return true;
         }
-        // dominant shape: packed V2, no control fields — register-window
+        if (sp.v2_nfp > 0 && (rdr->peek8() & 0xff8000u) == 0) {
+          // mixed fixed/varlen schema, no null mask: direct loads for the
+          // static-offset prefix + unaligned walk for the varlen tail
+          rdr->seek(value + value_len);
+          if (!decode_packed_v2_mixed(sp, aux, value, value_len, rc))
+            return false;
+          rc->found = true;
+          return true;
+        }
+        // remaining shapes (leading varlen column) — register-window
         // decode; falls back when a null mask / big schema version appears
         bool done;
         if (!decode_packed_v2_rdr(sp, base, aux, rdr, value_len, rc, &done))
@@ -3409,13 +3508,17 @@ inline void build_dev_spec(const ybg_scan_spec_t* spec, DevSpec* dp,
     // fixed-offset packed-V2 fast path: usable when every value column is
     // fixed-width ('|' + 1-byte version + flags + bodies)
     uint32_t off = 3;
+    int nfp = 0;
     bool all_fixed = sc.num_value_cols > 0;
     for (int c = 0; c < sc.num_value_cols; ++c) {
       if (!d.cols[c].v2_fixed || off > 255) { all_fixed = false; break; }
       d.v2_off[c] = (uint8_t)off;
       off += (uint32_t)d.cols[c].v2_fixed;
+      ++nfp;
     }
     d.v2_fixed_len = all_fixed ? off : 0;
+    d.v2_nfp = nfp;
+    d.v2_tail_off = off;
   }
   d.lower_len = (uint32_t)spec->lower_bound_len;
   d.lower_off = put(spec->lower_bound, spec->lower_bound_len);
