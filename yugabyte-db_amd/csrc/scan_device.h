@@ -1184,8 +1184,15 @@ DEV bool pred_compare(const PredC& pr, uint64_t datum, const uint8_t* sptr,
         lst += 4 + ol;
         if (ol != slen) continue;
         bool eq = true;
-        for (uint32_t k = 0; k < ol; ++k)
-          if (sptr[k] != ob[k]) { eq = false; break; }
+        uint32_t k = 0;
+        for (; k + 8 <= ol; k += 8)
+          if (load_u64_una(sptr + k) != load_u64_una(ob + k)) {
+            eq = false;
+            break;
+          }
+        if (eq)
+          for (; k < ol; ++k)
+            if (sptr[k] != ob[k]) { eq = false; break; }
         if (eq) return true;
       }
       return false;
@@ -1214,12 +1221,24 @@ DEV bool pred_compare(const PredC& pr, uint64_t datum, const uint8_t* sptr,
     uint32_t rlen = (uint32_t)(pr.datum >> 32);
     uint32_t n = slen < rlen ? slen : rlen;
     cmp = 0;
-    for (uint32_t k = 0; k < n; ++k) {
-      if (sptr[k] != rhs[k]) {
-        cmp = sptr[k] < rhs[k] ? -1 : 1;
+    // memcmp order == big-endian word order: compare 8 bytes per step
+    // (byteswapped unaligned loads) instead of a serial byte loop; the
+    // sub-8 tail stays a byte loop (no tail-slack assumption on aux)
+    uint32_t k = 0;
+    for (; k + 8 <= n; k += 8) {
+      uint64_t a = __builtin_bswap64(load_u64_una(sptr + k));
+      uint64_t b = __builtin_bswap64(load_u64_una(rhs + k));
+      if (a != b) {
+        cmp = a < b ? -1 : 1;
         break;
       }
     }
+    if (cmp == 0)
+      for (; k < n; ++k)
+        if (sptr[k] != rhs[k]) {
+          cmp = sptr[k] < rhs[k] ? -1 : 1;
+          break;
+        }
     if (cmp == 0 && slen != rlen) cmp = slen < rlen ? -1 : 1;
   } else if (dtype == YBG_T_DOUBLE) {
     double a = __longlong_as_double((long long)datum);
