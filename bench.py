@@ -153,6 +153,10 @@ def main():
         spec.num_preds = len(preds)
         for i, p in enumerate(preds):
             spec.preds[i] = p
+        # the mvcc workload scans a historical snapshot over 5-version
+        # rows — the caller-known condition the expect_versions hint
+        # models (it selects the version-chain kernel shape)
+        spec.expect_versions = 1 if args.workload == "mvcc" else 0
         spec.num_aggs = len(aggs)
         for i, a in enumerate(aggs):
             spec.aggs[i] = a

@@ -216,6 +216,13 @@ def test_fast_fuzz_vs_general():
             ref = _res(y.sim_scan(spec, built[0], built[1], built[2]))
             fast, nf = y.sim_scan_fast(spec, built[0], built[1], built[2])
             assert _res(fast) == ref, (it, read, local)
+            # the expect_versions hint selects the FUSE decode shape —
+            # results must be bit-identical to the default shape
+            spec.expect_versions = 1
+            fused, nf2 = y.sim_scan_fast(spec, built[0], built[1],
+                                         built[2])
+            assert _res(fused) == ref, (it, read, local, "fused")
+            spec.expect_versions = 0
 
 
 def test_fast_deep_window_dht_tails():

@@ -81,6 +81,7 @@ class ScanSpec(C.Structure):
         ("row_limit", C.c_uint64),
         ("group_col", C.c_int32),  # 0 = none, else 1 + value column index
         ("backward", C.c_int32),   # descending delivery order
+        ("expect_versions", C.c_int32),  # MVCC-heavy hint (kernel shape)
     ]
 
 

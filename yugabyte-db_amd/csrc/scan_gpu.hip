@@ -190,7 +190,7 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan(
 // same global bheads/walked protocol.
 // ---------------------------------------------------------------------------
 
-template <int WPS, int NC>
+template <int WPS, int NC, bool FUSE = false>
 __global__ __launch_bounds__(kThreads, WPS) void k_scan_fast(
     const uint8_t* __restrict__ data,
     const uint64_t* __restrict__ block_offsets,
@@ -222,7 +222,8 @@ __global__ __launch_bounds__(kThreads, WPS) void k_scan_fast(
     const uint64_t lo = batch_lo ? batch_lo[j] : j * ivb;
     uint64_t hi = batch_lo ? batch_lo[j + 1] : lo + ivb;
     if (hi > n_ivs) hi = n_ivs;
-    if (!scan_batch_fast<NA, NC>(sp, data, block_offsets, ivs, n_ivs, lo, hi,
+    if (!scan_batch_fast<NA, NC, FUSE>(sp, data, block_offsets, ivs, n_ivs,
+                                       lo, hi,
                              key, rmin, &entries, &scanned, &matched,
                              agg_val, agg_cnt, &ho, &walked_next)) {
       unsigned long long slot = atomicAdd(retry_n, 1ull);
@@ -1532,8 +1533,12 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
       if (v >= 2 && v <= 8) fwps = (int)v;
     }
     // NC: compile-time column cap for the unrolled load pass (spec is
-    // zero-padded up to it; fast_eligible caps num_value_cols at 8)
+    // zero-padded up to it; fast_eligible caps num_value_cols at 8).
+    // fuse: version-chain decode shape per the spec's expect_versions
+    // hint (instantiated for the default wave count only).
     const bool nc4 = s->dspec.num_value_cols <= 4;
+    bool fuse = s->dspec.fuse_hint != 0;
+    if (const char* e = getenv("YBG_FUSE")) fuse = atoi(e) != 0;
 #define YBG_LAUNCH_FAST(W) \
   do { \
     if (nc4) launchf(k_scan_fast<W, 4>); \
@@ -1544,7 +1549,14 @@ int yb_gpu_scan_execute(ybg_scan_t* s) {
       case 4: YBG_LAUNCH_FAST(4); break;
       case 5: YBG_LAUNCH_FAST(5); break;
       case 6: YBG_LAUNCH_FAST(6); break;
-      default: YBG_LAUNCH_FAST(3); break;
+      default:
+        if (fuse) {
+          if (nc4) launchf(k_scan_fast<3, 4, true>);
+          else launchf(k_scan_fast<3, 8, true>);
+        } else {
+          YBG_LAUNCH_FAST(3);
+        }
+        break;
     }
 #undef YBG_LAUNCH_FAST
     // retry pass: the general kernel over the aborted batch list, into

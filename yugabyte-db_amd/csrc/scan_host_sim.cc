@@ -12,6 +12,7 @@ extern "C" unsigned long long ybg_fast_abort_hist[32] = {0};
 #include "scan_device.h"
 
 #include <cstdlib>
+#include <type_traits>
 #include <vector>
 
 using namespace ybgdev;
@@ -173,16 +174,22 @@ int ybg_sim_scan_fast(const ybg_scan_spec_t* spec, const uint8_t* data,
     uint32_t e32 = 0, s32 = 0, m32 = 0;
     uint64_t lo = b * ivb;
     uint64_t hi = lo + ivb < n_ivs ? lo + ivb : n_ivs;
-    // NC mirrors the GPU dispatch: 4-column instantiation when it fits
-    int rc = d.num_value_cols <= 4
-                 ? scan_batch_fast<2, 4>(d, data, offsets, ivs.data(), n_ivs,
-                                         lo, hi, key, bht + 3, &e32, &s32,
-                                         &m32, agg_val, agg_cnt, &heads[b],
-                                         &wn)
-                 : scan_batch_fast<2, 8>(d, data, offsets, ivs.data(), n_ivs,
-                                         lo, hi, key, bht + 3, &e32, &s32,
-                                         &m32, agg_val, agg_cnt, &heads[b],
-                                         &wn);
+    // NC and FUSE mirror the GPU dispatch so CPU fuzz covers both
+    // compiled shapes of the fast decode
+    auto call_fast = [&](auto fuse_tag) {
+      constexpr bool F = decltype(fuse_tag)::value;
+      return d.num_value_cols <= 4
+                 ? scan_batch_fast<2, 4, F>(d, data, offsets, ivs.data(),
+                                            n_ivs, lo, hi, key, bht + 3,
+                                            &e32, &s32, &m32, agg_val,
+                                            agg_cnt, &heads[b], &wn)
+                 : scan_batch_fast<2, 8, F>(d, data, offsets, ivs.data(),
+                                            n_ivs, lo, hi, key, bht + 3,
+                                            &e32, &s32, &m32, agg_val,
+                                            agg_cnt, &heads[b], &wn);
+    };
+    int rc = d.fuse_hint ? call_fast(std::true_type{})
+                         : call_fast(std::false_type{});
     if (!rc) {
       ++fallbacks;
       uint64_t av8[YBG_MAX_AGGS] = {0}, ac8[YBG_MAX_AGGS] = {0};
