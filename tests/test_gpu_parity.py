@@ -23,9 +23,11 @@ def _gpu():
     return gpu_scan
 
 
-def run_gpu(case, read_micros, preds, aggs, lower=None, upper=None):
+def run_gpu(case, read_micros, preds, aggs, lower=None, upper=None,
+            expect_versions=0):
     gpu_scan = _gpu()
     spec = make_spec(case, read_micros, preds, aggs, lower, upper)
+    spec.expect_versions = expect_versions
     s = gpu_scan.GpuScan(spec)
     s.feed_blocks_host(case["data"], case["offsets"], case["n_blocks"],
                        case["total"])
@@ -49,6 +51,27 @@ def test_gpu_vs_oracle_all_cases(cases):
             except AssertionError as e:
                 raise AssertionError(
                     f"case {case['name']} read={read_micros}: {e}") from e
+
+
+def test_gpu_fused_shape_parity(cases):
+    """spec.expect_versions = 1 dispatches the FUSE variant of the fast
+    kernel (version-chain decode shape) — every case/run must match the
+    oracle exactly like the default shape does."""
+    for case in cases:
+        for run in case["runs"]:
+            read_micros, preds, aggs = run[0], run[1], run[2]
+            lower = run[3] if len(run) > 3 else None
+            upper = run[4] if len(run) > 4 else None
+            gres = run_gpu(case, read_micros, preds, aggs, lower, upper,
+                           expect_versions=1)
+            ores = run_oracle(case, read_micros, preds, aggs, lower, upper)
+            try:
+                check_match(gres, ores, aggs,
+                            check_entries=lower is None and upper is None)
+            except AssertionError as e:
+                raise AssertionError(
+                    f"case {case['name']} read={read_micros} fused: {e}"
+                ) from e
 
 
 def test_gpu_determinism(cases):
