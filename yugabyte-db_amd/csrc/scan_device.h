@@ -238,6 +238,11 @@ struct DevSpec {
 constexpr uint32_t kActPredM = 0xffu;
 constexpr uint32_t kActAggShift = 8, kActAggM = 0xffu;
 constexpr uint32_t kActGroup = 1u << 16;
+// any scan-visible role: predicates, aggregates or the group key. A
+// column with none of these is INERT — decode loops skip its eval_col
+// entirely unless rows are being materialized (emit_datums needs every
+// projected column).
+constexpr uint32_t kActLiveM = 0x1ffffu;
 constexpr uint32_t kActDtShift = 17, kActDtM = 0xfu;
 constexpr uint32_t kActV2Shift = 21, kActV2M = 0xfu;
 
@@ -1794,7 +1799,8 @@ DEV bool decode_packed_v2_rdr(const DevSpec& sp, const uint8_t* base,
       if (dt == YBG_T_INT8) u = (uint64_t)(int64_t)(int8_t)u;
       else if (dt == YBG_T_INT16) u = (uint64_t)(int64_t)(int16_t)u;
       else if (dt == YBG_T_INT32) u = (uint64_t)(int64_t)(int32_t)u;
-      eval_col(sp, aux, rc, i, false, u, nullptr, 0);
+      if ((act & kActLiveM) || rc->emit_datums)
+        eval_col(sp, aux, rc, i, false, u, nullptr, 0);
       rdr->consume(fw);
     } else {
       if (rdr->pos() >= end) return false;
@@ -1808,7 +1814,8 @@ DEV bool decode_packed_v2_rdr(const DevSpec& sp, const uint8_t* base,
         rdr->consume(4);
       }
       if (rdr->pos() + flen > end) return false;
-      eval_col(sp, aux, rc, i, false, 0, rdr->pos(), flen);
+      if ((act & kActLiveM) || rc->emit_datums)
+        eval_col(sp, aux, rc, i, false, 0, rdr->pos(), flen);
       rdr->skip(flen);
     }
   }
@@ -1862,7 +1869,8 @@ DEV void decode_packed_v2_fixed(const DevSpec& sp, const uint8_t* aux,
       default:
         break;
     }
-    eval_col(sp, aux, rc, i, false, u, nullptr, 0);
+    if ((act & kActLiveM) || rc->emit_datums)
+      eval_col(sp, aux, rc, i, false, u, nullptr, 0);
   }
 }
 
@@ -1910,7 +1918,8 @@ DEV bool decode_packed_v2_mixed(const DevSpec& sp, const uint8_t* aux,
       default:
         break;
     }
-    eval_col(sp, aux, rc, i, false, u, nullptr, 0);
+    if ((act & kActLiveM) || rc->emit_datums)
+      eval_col(sp, aux, rc, i, false, u, nullptr, 0);
   }
   const uint8_t* p = value + sp.v2_tail_off;
   for (int i = nfp; i < n; ++i) {
@@ -1930,7 +1939,8 @@ DEV bool decode_packed_v2_mixed(const DevSpec& sp, const uint8_t* aux,
       if (dt == YBG_T_INT8) u = (uint64_t)(int64_t)(int8_t)u;
       else if (dt == YBG_T_INT16) u = (uint64_t)(int64_t)(int16_t)u;
       else if (dt == YBG_T_INT32) u = (uint64_t)(int64_t)(int32_t)u;
-      eval_col(sp, aux, rc, i, false, u, nullptr, 0);
+      if ((act & kActLiveM) || rc->emit_datums)
+        eval_col(sp, aux, rc, i, false, u, nullptr, 0);
       p += fw;
     } else {
       if (YBG_UNLIKELY(p >= end)) return false;
@@ -1943,7 +1953,8 @@ DEV bool decode_packed_v2_mixed(const DevSpec& sp, const uint8_t* aux,
         p += 4;
       }
       if (YBG_UNLIKELY(p + flen > end)) return false;
-      eval_col(sp, aux, rc, i, false, 0, p, flen);
+      if ((act & kActLiveM) || rc->emit_datums)
+        eval_col(sp, aux, rc, i, false, 0, p, flen);
       p += flen;
     }
   }
