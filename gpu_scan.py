@@ -99,6 +99,19 @@ class GpuScan:
         self._check(f(self._h, data, offsets, n_blocks, intents_blob,
                       blob_len, txns, n_txns), "feed_blocks_intents")
 
+    def feed_blocks_bloom(self, data, offsets, n_blocks, filt):
+        """feed_blocks + bloom consultation: a point scan whose pinned key
+        prefix the filter proves absent feeds nothing and reports empty
+        (yb_gpu_scan_feed_blocks_bloom)."""
+        f = self._lib.yb_gpu_scan_feed_blocks_bloom
+        f.restype = C.c_int
+        f.argtypes = [C.c_void_p, C.POINTER(C.c_uint8),
+                      C.POINTER(C.c_uint64), C.c_uint64, C.c_int,
+                      C.POINTER(C.c_uint8), C.c_uint64]
+        fb = (C.c_uint8 * len(filt)).from_buffer_copy(filt) if filt else None
+        self._check(f(self._h, data, offsets, n_blocks, 0, fb,
+                      len(filt) if filt else 0), "feed_blocks_bloom")
+
     def feed_sst(self, file_ptr, size, verify=True):
         """Feed a complete BlockBasedTable SST file (footer + index block
         parsed host-side; block checksums verified when verify)."""

@@ -310,6 +310,36 @@ int yb_gpu_scan_next_batch(ybg_scan_t *s, ybg_row_batch_t *out);
 int yb_gpu_scan_restart_data(ybg_scan_t *s, uint8_t *ht_out,
                              uint32_t *len_out);
 
+/* ---- SST bloom filter (docdb_filter_policy / rocksdb FixedSizeFilter,
+ * SURVEY 8f-2). Exact reference bit format: 64-byte cache-line blocks,
+ * 5-byte trailer, rocksdb::Hash(0xbc9f1d34) with signed-char tail
+ * (rocksdb/util/bloom.cc:43-62,384-452; util/hash.cc:32-77). Keys are
+ * transformed to the DocKeyPart::kUpToHashOrFirstRange prefix
+ * (DocDbAwareV3FilterPolicy, docdb/docdb_filter_policy.cc:110-117).
+ * The filter is an optimization only: scan results are identical with
+ * or without it. */
+uint64_t ybg_filter_slice_size(void);
+/* Build a filter over every distinct key prefix of a finished tablet.
+ * out receives concatenated fixed-size slices (out_len a multiple of
+ * ybg_filter_slice_size()). 0 ok, 8 cap too small, 3 corruption. */
+int ybg_filter_from_sst(const uint8_t *data, const uint64_t *offsets,
+                        uint64_t n_blocks, int kv_format, uint8_t *out,
+                        uint64_t cap, uint64_t *out_len);
+/* KeyMayMatch: 1 = prefix may exist, 0 = definitely absent. */
+int ybg_filter_may_match(const uint8_t *filt, uint64_t filt_len,
+                         const uint8_t *key, uint64_t key_len);
+/* Test hook: extracted kUpToHashOrFirstRange prefix length (0 =
+ * unparseable => always-match). */
+uint64_t ybg_filter_key_prefix_len(const uint8_t *key, uint64_t key_len);
+/* feed_blocks + bloom consultation: when the spec's DocKey bounds pin a
+ * single filter prefix and the filter proves it absent, the scan is
+ * answered empty without uploading any block. filter may be NULL. */
+int yb_gpu_scan_feed_blocks_bloom(ybg_scan_t *s, const uint8_t *blocks,
+                                  const uint64_t *offsets,
+                                  uint64_t n_blocks, int device_resident,
+                                  const uint8_t *filter,
+                                  uint64_t filter_len);
+
 /* Resumable position (pgsql_operation.cc:2796-2806, 2908-2922): for an
  * unlimited scan reports length 0 (complete). row_limit paging requires
  * delivered-row ORDER, which the batch ABI leaves to the row-at-a-time

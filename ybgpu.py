@@ -1001,3 +1001,44 @@ def lz4_uncompress(data, cap):
     return bytes(dst[:n])
 
 
+
+
+# ---- SST bloom filter (docdb_filter_policy / rocksdb FixedSizeFilter) ----
+
+def filter_slice_size():
+    lib = product()
+    return _sig(lib, "ybg_filter_slice_size", C.c_uint64, [])()
+
+
+def filter_from_sst(data, offsets, n_blocks, kv_format=1):
+    """Build the tablet's bloom filter (concatenated fixed-size slices)."""
+    lib = product()
+    f = _sig(lib, "ybg_filter_from_sst", C.c_int,
+             [C.POINTER(C.c_uint8), C.POINTER(C.c_uint64), C.c_uint64,
+              C.c_int, C.POINTER(C.c_uint8), C.c_uint64,
+              C.POINTER(C.c_uint64)])
+    cap = filter_slice_size() * (2 + n_blocks)
+    out = (C.c_uint8 * cap)()
+    out_len = C.c_uint64(0)
+    rc = f(data, offsets, n_blocks, kv_format, out, cap, C.byref(out_len))
+    if rc:
+        raise RuntimeError(f"ybg_filter_from_sst rc={rc}")
+    return bytes(out[:out_len.value])
+
+
+def filter_may_match(filt, key):
+    lib = product()
+    f = _sig(lib, "ybg_filter_may_match", C.c_int,
+             [C.POINTER(C.c_uint8), C.c_uint64, C.POINTER(C.c_uint8),
+              C.c_uint64])
+    fb = (C.c_uint8 * len(filt)).from_buffer_copy(filt)
+    kb = (C.c_uint8 * len(key)).from_buffer_copy(key)
+    return f(fb, len(filt), kb, len(key))
+
+
+def filter_key_prefix_len(key):
+    lib = product()
+    f = _sig(lib, "ybg_filter_key_prefix_len", C.c_uint64,
+             [C.POINTER(C.c_uint8), C.c_uint64])
+    kb = (C.c_uint8 * len(key)).from_buffer_copy(key)
+    return f(kb, len(key))

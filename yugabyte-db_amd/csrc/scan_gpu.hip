@@ -43,6 +43,7 @@
 
 #include "scan_device.h"
 #include "sst_internal.h"
+#include "bloom_filter.h"
 #include "snappy_dev.h"
 #include "lz4_dev.h"
 
@@ -883,6 +884,9 @@ struct ybg_scan {
   uint64_t* d_chunk = nullptr;  // k_reduce_pre output (256 partial records)
   int grid = 0;
   bool executed = false;
+  // bloom filter proved the scan's pinned key prefix absent at feed
+  // time: no data was uploaded, every result surface reports empty
+  bool bloom_rejected = false;
   double last_total_ms = 0, last_decode_ms = 0;
   std::vector<uint8_t> aux_host;
   // emit (next_batch) buffers
@@ -1185,6 +1189,38 @@ int ybg_test_block_selection(const ybg_scan_spec_t* spec,
   return 1;
 }
 
+/* Bloom-aware feed (BloomFilterAwareIterator role,
+ * docdb/docdb_rocksdb_util.cc bloom usage + rocksdb
+ * table/fixed_size_filter_block.cc): when the scan's DocKey bounds pin
+ * one kUpToHashOrFirstRange prefix (a point read on the hashed
+ * components) and the tablet's filter proves that prefix absent, the
+ * scan is answered empty WITHOUT uploading or touching any block.
+ * filter = ybg_filter_from_sst output for this tablet (may be
+ * NULL/empty: never rejects). Results are identical to an unfiltered
+ * feed + scan — the filter only skips provably-empty work. */
+int yb_gpu_scan_feed_blocks_bloom(ybg_scan_t* s, const uint8_t* blocks,
+                                  const uint64_t* offsets,
+                                  uint64_t n_blocks, int device,
+                                  const uint8_t* filter,
+                                  uint64_t filter_len) {
+  const ybg_scan_spec_t& sp = s->spec;
+  if (filter && filter_len && sp.lower_bound_len && sp.upper_bound_len) {
+    size_t pl = ybg::filter_key_prefix_len(sp.lower_bound,
+                                           sp.lower_bound_len);
+    size_t pu = ybg::filter_key_prefix_len(sp.upper_bound,
+                                           sp.upper_bound_len);
+    if (pl && pl == pu &&
+        memcmp(sp.lower_bound, sp.upper_bound, pl) == 0 &&
+        !ybg::bloom_may_match(filter, filter_len, sp.lower_bound,
+                              sp.lower_bound_len,
+                              (uint32_t)ybg_filter_slice_size())) {
+      s->bloom_rejected = true;
+      return 0;
+    }
+  }
+  return yb_gpu_scan_feed_blocks(s, blocks, offsets, n_blocks, device);
+}
+
 int yb_gpu_scan_feed_blocks(ybg_scan_t* s, const uint8_t* blocks,
                             const uint64_t* offsets, uint64_t n_blocks,
                             int device) {
@@ -1485,6 +1521,10 @@ int yb_gpu_scan_feed_sst(ybg_scan_t* s, const uint8_t* file, uint64_t size,
 }
 
 int yb_gpu_scan_execute(ybg_scan_t* s) {
+  if (s->bloom_rejected) {
+    s->executed = true;
+    return 0;
+  }
   if (!s->d_data) return set_err(4, "feed_blocks not called");
   // the retry half of the partials is only partially covered by the retry
   // launch (or not at all on the general dispatch): zero it so the fold
@@ -1620,6 +1660,11 @@ int yb_gpu_scan_wait(ybg_scan_t* s) {
 }
 
 int yb_gpu_scan_aggregate(ybg_scan_t* s, ybg_scan_result_t* out) {
+  if (s->bloom_rejected) {
+    memset(out, 0, sizeof(*out));
+    for (int g = 0; g < s->spec.num_aggs; ++g) out->aggs[g].is_null = 1;
+    return 0;
+  }
   int rc = yb_gpu_scan_wait(s);
   if (rc) return rc;
   DevResult r;
@@ -1659,6 +1704,10 @@ int yb_gpu_scan_aggregate(ybg_scan_t* s, ybg_scan_result_t* out) {
 }
 
 int yb_gpu_scan_next_batch(ybg_scan_t* s, ybg_row_batch_t* out) {
+  if (s->bloom_rejected) {
+    memset(out, 0, sizeof(*out));
+    return 0;
+  }
   if (!s->d_data) return set_err(4, "feed_blocks not called");
   int nk = s->spec.schema.num_hash_cols + s->spec.schema.num_range_cols;
   int nc = s->spec.schema.num_value_cols;
@@ -1894,6 +1943,10 @@ int yb_gpu_scan_group_aggregate(ybg_scan_t* s, uint64_t* keys,
 // serves the GROUP BY path, whose result shape has no restart field.
 int yb_gpu_scan_restart_data(ybg_scan_t* s, uint8_t* ht_out,
                              uint32_t* len_out) {
+  if (s->bloom_rejected) {
+    *len_out = 0;
+    return 0;
+  }
   HIP_TRY(hipStreamSynchronize(s->stream));
   DevResult r;
   HIP_TRY(hipMemcpy(&r, s->d_result, sizeof(r), hipMemcpyDeviceToHost));

@@ -12,6 +12,7 @@
 #include "sst_internal.h"
 #include "snappy_dev.h"
 #include "lz4_dev.h"
+#include "bloom_filter.h"
 
 #include <algorithm>
 #include <atomic>
@@ -1277,6 +1278,57 @@ int ybg_block_first_key(const uint8_t* blk, uint64_t size, int kv_format,
   memcpy(out, key.data(), key.size());
   *len = key.size();
   return 0;
+}
+
+/* ---- SST bloom filter (docdb_filter_policy / rocksdb FixedSizeFilter,
+ * SURVEY §8f-2; see bloom_filter.h for the format citations). The filter
+ * is built from a finished tablet's blocks (the flush-time role of
+ * FixedSizeFilterBlockBuilder) and queried at feed time. ------------- */
+
+uint64_t ybg_filter_slice_size(void) {
+  return ybg::bloom_dims().slice_size;
+}
+
+/* Build the filter over every distinct key prefix in the tablet.
+ * Returns 0 ok (out_len bytes written, a multiple of
+ * ybg_filter_slice_size), 8 if cap is too small, 3 on block corruption. */
+int ybg_filter_from_sst(const uint8_t* data, const uint64_t* offsets,
+                        uint64_t n_blocks, int kv_format, uint8_t* out,
+                        uint64_t cap, uint64_t* out_len) {
+  ybg::BloomBuilder fb;
+  for (uint64_t b = 0; b < n_blocks; ++b) {
+    std::vector<std::pair<ybg::Buf, ybg::Buf>> es;
+    int rc = ybg::DecodeBlockHost(data + offsets[b],
+                                  offsets[b + 1] - offsets[b], kv_format,
+                                  &es);
+    if (rc) return rc;
+    for (auto& e : es) {
+      if (e.first.size() < 8) return 3;
+      /* user key = internal key minus the 8-byte seqno/type suffix */
+      fb.AddDocKey(e.first.data(), e.first.size() - 8);
+    }
+  }
+  ybg::Buf f = fb.Finish();
+  if (f.size() > cap) return 8;
+  memcpy(out, f.data(), f.size());
+  *out_len = f.size();
+  return 0;
+}
+
+/* KeyMayMatch (DocDbAwareV3FilterPolicy semantics): key = encoded
+ * DocKey (no internal suffix). 1 = may exist, 0 = definitely absent. */
+int ybg_filter_may_match(const uint8_t* filt, uint64_t filt_len,
+                         const uint8_t* key, uint64_t key_len) {
+  return ybg::bloom_may_match(filt, filt_len, key, key_len,
+                              ybg::bloom_dims().slice_size)
+             ? 1
+             : 0;
+}
+
+/* Test hook: the kUpToHashOrFirstRange prefix length the transform
+ * extracts (0 = unparseable = always-match). */
+uint64_t ybg_filter_key_prefix_len(const uint8_t* key, uint64_t key_len) {
+  return ybg::filter_key_prefix_len(key, key_len);
 }
 
 }  // extern "C"
