@@ -104,7 +104,7 @@ def test_golden_bits_vs_independent_restatement():
     b = y.Builder(sc)
     seq = 1 << 50
     keys = []
-    for r in range(500):
+    for r in sorted(range(500), key=lambda r: (r % 37, r)):
         seq += 1
         b.add_packed_row(5000, [(y.T_INT64, r)], hash_=r % 37,
                          key_datums=(r,), seq=seq)
@@ -128,7 +128,7 @@ def test_may_match_and_fp_rate():
     sc = _schema()
     b = y.Builder(sc)
     seq = 1 << 50
-    for r in range(3000):
+    for r in sorted(range(3000), key=lambda r: (r % 997, r)):
         seq += 1
         b.add_packed_row(5000, [(y.T_INT64, r)], hash_=r % 997,
                          key_datums=(r,), seq=seq)
@@ -179,7 +179,7 @@ def test_gpu_bloom_pruned_point_scan():
     sc = _schema()
     b = y.Builder(sc)
     seq = 1 << 50
-    for r in range(2000):
+    for r in sorted(range(2000), key=lambda r: (r % 100, r)):
         seq += 1
         b.add_packed_row(5000, [(y.T_INT64, r)], hash_=r % 100,
                          key_datums=(r,), seq=seq)
