@@ -96,8 +96,52 @@ def run(cases=40, seed=20250915):
                 == (o.rows_scanned, o.rows_matched, o.aggs[1].value_i64), it
             assert restart == bytes(o.restart_ht[:o.restart_ht_len])
             n_intents += 1
+    # bloom point-scan soak: random sorted tablets; random point keys
+    # scanned through feed_blocks_bloom must match the oracle over the
+    # same bounds (present keys found, filter-rejected keys empty)
+    import ctypes as C
+    n_bloom = 0
+    rng3 = random.Random(seed + 2)
+    sc = y.make_schema([y.KT_INT64], [(10, y.T_INT64, 1)])
+    for it in range(max(3, cases // 20)):
+        rows = rng3.randint(800, 5000)
+        hmod = rng3.choice([53, 211, 997])
+        b = y.Builder(sc)
+        seq = 1 << 50
+        for r in sorted(range(rows), key=lambda r: (r % hmod, r)):
+            seq += 1
+            b.add_packed_row(5000, [(y.T_INT64, r)], hash_=r % hmod,
+                             key_datums=(r,), seq=seq)
+        data, offsets, nb, total = b.finish()[:4]
+        filt = y.filter_from_sst(data, offsets, nb)
+        for _ in range(20):
+            if rng3.random() < 0.5:
+                r = rng3.randrange(rows)
+                hh, k0, expect = r % hmod, r, 1
+            else:
+                hh, k0 = rng3.randrange(8 * hmod), rng3.randrange(4 * rows)
+                expect = 1 if (k0 < rows and k0 % hmod == hh) else 0
+            lower = y.encode_dockey(sc, hash_=hh, key_datums=(k0,))
+            upper = lower + b"\x00"
+            spec = y.ScanSpec()
+            spec.schema = sc
+            spec.kv_format = y.ENC_THREE_SHARED_PARTS
+            spec.read_time = y.read_time(9000)
+            spec.num_aggs = 1
+            spec.aggs[0] = y.Agg(y.AGG_COUNT_STAR, 0)
+            lb = (C.c_uint8 * len(lower)).from_buffer_copy(lower)
+            ub = (C.c_uint8 * len(upper)).from_buffer_copy(upper)
+            spec.lower_bound, spec.lower_bound_len = lb, len(lower)
+            spec.upper_bound, spec.upper_bound_len = ub, len(upper)
+            s = GpuScan(spec)
+            s.feed_blocks_bloom(data, offsets, nb, filt)
+            s.execute()
+            g = s.aggregates()
+            s.close()
+            assert g.rows_matched == expect, (it, hh, k0, g.rows_matched)
+            n_bloom += 1
     print(f"soak ok: {n_scan} scans, {n_intents} intent merges, "
-          f"seed {seed}")
+          f"{n_bloom} bloom point scans, seed {seed}")
 
 if __name__ == "__main__":
     run(int(sys.argv[1]) if len(sys.argv) > 1 else 40,
