@@ -26,6 +26,16 @@ int GpuDocRowwiseIterator::FeedBlocks(const uint8_t* blocks,
                                  device_resident ? 1 : 0);
 }
 
+int GpuDocRowwiseIterator::FeedBlocksBloom(const uint8_t* blocks,
+                                           const uint64_t* offsets,
+                                           uint64_t n_blocks,
+                                           const uint8_t* filter,
+                                           uint64_t filter_len) {
+  if (open_rc_) return open_rc_;
+  return yb_gpu_scan_feed_blocks_bloom(handle_, blocks, offsets, n_blocks,
+                                       0, filter, filter_len);
+}
+
 int GpuDocRowwiseIterator::MaterializeBatch() {
   int rc = yb_gpu_scan_next_batch(handle_, &batch_);
   if (rc) return rc;

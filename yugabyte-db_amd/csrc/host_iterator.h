@@ -32,6 +32,13 @@ class GpuDocRowwiseIterator {
   int FeedBlocks(const uint8_t* blocks, const uint64_t* offsets,
                  uint64_t n_blocks, bool device_resident);
 
+  // ~ the BloomFilterAwareIterator role (docdb_filter_policy.cc): feed
+  // with the tablet's bloom filter consulted — a point scan the filter
+  // proves empty uploads nothing and every fetch reports end-of-scan.
+  int FeedBlocksBloom(const uint8_t* blocks, const uint64_t* offsets,
+                      uint64_t n_blocks, const uint8_t* filter,
+                      uint64_t filter_len);
+
   // ~ YQLRowwiseIteratorIf::PgFetchNext (row-at-a-time over the GPU batch;
   // rows are delivered in tablet key order). Returns 1 row fetched, 0 end
   // of scan, <0 error.
