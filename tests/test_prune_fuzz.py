@@ -141,3 +141,55 @@ def test_bloom_reject_is_proof_of_absence():
         checked += 1
     assert rejected > 200, rejected  # most random absent prefixes reject
     assert checked == rejected
+
+
+def test_prune_point_scan_on_block_boundaries():
+    """Regression: a point scan on a DocKey that opens a block used to
+    lose that block — the upper bound key+\\x00 sorts below the block's
+    first INTERNAL key (DocKey || '#' DHT || seq) even though its DocKey
+    is in range. Selection must compare DocKeys. Exercise the exact
+    first-DocKey of EVERY block."""
+    sc = y.make_schema([y.KT_INT64], [(10, y.T_INT64, 1)])
+    rows, hmod = 4993, 211
+    b = y.Builder(sc)
+    seq = 1 << 50
+    by_key = sorted(range(rows), key=lambda r: (r % hmod, r))
+    for r in by_key:
+        seq += 1
+        b.add_packed_row(5000, [(y.T_INT64, r)], hash_=r % hmod,
+                         key_datums=(r,), seq=seq)
+    data, offsets, n_blocks, total = b.finish()[:4]
+    lib = y.product()
+    fk = lib.ybg_block_first_key
+    fk.restype = C.c_int
+    fk.argtypes = [C.POINTER(C.c_uint8), C.c_uint64, C.c_int,
+                   C.POINTER(C.c_uint8), C.c_uint64, C.POINTER(C.c_uint64)]
+    for blk in range(n_blocks):
+        out = (C.c_uint8 * 256)()
+        ln = C.c_uint64()
+        rc = fk(C.cast(C.addressof(data.contents) + offsets[blk],
+                       C.POINTER(C.c_uint8)),
+                offsets[blk + 1] - offsets[blk], 1, out, 256, C.byref(ln))
+        assert rc == 0
+        ik = bytes(out[:ln.value])
+        # DocKey = up to and including the second kGroupEnd
+        first = ik.index(b"\x21")
+        dk = ik[:ik.index(b"\x21", first + 1) + 1]
+        lower, upper = dk, dk + b"\x00"
+        spec = y.ScanSpec()
+        spec.schema = sc
+        spec.kv_format = y.ENC_THREE_SHARED_PARTS
+        spec.read_time = y.read_time(9000)
+        spec.num_aggs = 1
+        spec.aggs[0] = y.Agg(y.AGG_COUNT_STAR, 0)
+        lb = (C.c_uint8 * len(lower)).from_buffer_copy(lower)
+        ub = (C.c_uint8 * len(upper)).from_buffer_copy(upper)
+        spec.lower_bound, spec.lower_bound_len = lb, len(lower)
+        spec.upper_bound, spec.upper_bound_len = ub, len(upper)
+        full = _res(y.sim_scan(spec, data, offsets, n_blocks))
+        assert full[1] == 1  # the row exists
+        rc2, keep = _selection(spec, data, offsets, n_blocks)
+        if not rc2:
+            continue
+        sd, so, sn, _ka = _subset(data, offsets, n_blocks, keep)
+        assert _res(y.sim_scan(spec, sd, so, sn)) == full, blk

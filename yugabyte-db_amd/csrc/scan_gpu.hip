@@ -1065,6 +1065,35 @@ int pfx_cmp(const std::vector<uint8_t>& a, const uint8_t* b, size_t blen) {
   return a.size() == blen ? 0 : (a.size() < blen ? -1 : 1);
 }
 
+// Length of the full encoded DocKey at the head of an internal key
+// (hash group + range group, each kGroupEnd-terminated —
+// dockv/doc_key.h:40-63). 0 = unparseable. Bounds are DOCKEYS: comparing
+// them against the raw internal first key (DocKey || '#' DHT || seq)
+// mis-orders exactly when the bound extends the DocKey (e.g. a point
+// scan's upper = key + "\x00" sorts BELOW key's internal entries), so
+// block selection must compare against this prefix, never the raw key.
+size_t dockey_full_len(const uint8_t* k, size_t len) {
+  size_t p = 0;
+  if (len && k[0] == 0x47) {  // kUInt16Hash
+    if (len < 3) return 0;
+    p = 3;
+    while (p < len && k[p] != 0x21) {  // hashed components
+      size_t n = ybg::skip_key_entry(k + p, len - p);
+      if (!n) return 0;
+      p += n;
+    }
+    if (p >= len) return 0;
+    ++p;  // hashed group end
+  }
+  while (p < len && k[p] != 0x21) {  // range components
+    size_t n = ybg::skip_key_entry(k + p, len - p);
+    if (!n) return 0;
+    p += n;
+  }
+  if (p >= len) return 0;
+  return p + 1;  // range group end
+}
+
 // Returns true (and fills keep) when the spec allows pruning AND at least
 // one block can be skipped.
 bool compute_block_selection(const ybg_scan_spec_t& spec,
@@ -1136,7 +1165,9 @@ bool compute_block_selection(const ybg_scan_spec_t& spec,
     }
   }
 
-  // block separators: first internal key per block
+  // block separators: the DOCKEY of each block's first internal key
+  // (see dockey_full_len — raw internal-key bytes mis-order vs DocKey
+  // bounds)
   std::vector<std::vector<uint8_t>> fk(n_blocks);
   for (uint64_t b = 0; b < n_blocks; ++b) {
     uint8_t buf[256];
@@ -1145,20 +1176,23 @@ bool compute_block_selection(const ybg_scan_spec_t& spec,
                             offsets[b + 1] - offsets[b], spec.kv_format,
                             buf, sizeof(buf), &len))
       return false;  // undecodable: do not prune
-    fk[b].assign(buf, buf + len);
+    if (len < 8) return false;
+    size_t dl = dockey_full_len(buf, len - 8);  // minus the seqno suffix
+    if (!dl) return false;  // unknown entry encodings: do not prune
+    fk[b].assign(buf, buf + dl);
   }
   keep->assign(n_blocks, 0);
   bool any_skip = false;
   for (uint64_t b = 0; b < n_blocks; ++b) {
-    // block b covers [fk[b], fk[b+1]) (+inf for the last block); since a
-    // row can straddle into the next block's first interval via the tail
-    // walk only within one row, range intersection on separators is the
-    // correct granularity (rows are whole within the allowed key set)
+    // block b covers DocKeys [fk[b], fk[b+1]] — INCLUSIVE of fk[b+1]:
+    // a row whose first entry opens block b+1 may have entries
+    // straddling from the tail of block b, so lo uses <= on the next
+    // separator (keeps at most one extra block)
     bool hit = false;
     for (const auto& r : allowed) {
       bool lo_ok =
           r.lo.empty() || b + 1 >= n_blocks ||
-          pfx_cmp(r.lo, fk[b + 1].data(), fk[b + 1].size()) < 0;
+          pfx_cmp(r.lo, fk[b + 1].data(), fk[b + 1].size()) <= 0;
       bool hi_ok = r.hi.empty() ||
                    pfx_cmp(r.hi, fk[b].data(), fk[b].size()) > 0;
       if (lo_ok && hi_ok) {
