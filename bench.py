@@ -327,44 +327,61 @@ def main():
         cpu_1t_rows_per_s = sample_rows / st
 
         # all-cores oracle on the SAME tablet in the SAME run (north-star
-        # measurement clause): one persistent fork pool, each worker owns a
-        # contiguous block chunk, warm-up pass touches each worker's OWN
-        # chunk (first-touch faults out of the timed region), then the
-        # whole tablet is scanned in parallel; best wall of 3 passes.
+        # measurement clause): fork pools, each worker a contiguous block
+        # chunk, warm-up over own chunks before timing. The host's
+        # parallel throughput PEAKS below the logical core count (2-node
+        # NUMA + per-core saturation: measured 210 M rows/s at 32
+        # workers vs 122 M at 256 — scripts/cpu_baseline_sweep.py), so
+        # the worker count is PROBED and the best figure reported — the
+        # baseline is "the best this host can do", not "all cores".
         import multiprocessing as _mp
-        nproc = os.cpu_count() or 1
-        nproc = min(nproc, nb)
-        cuts = [nb * i // nproc for i in range(nproc + 1)]
-        ranges = [(cuts[i], cuts[i + 1]) for i in range(nproc)
-                  if cuts[i + 1] > cuts[i]]
+        ncpu = os.cpu_count() or 1
+        probe = sorted({min(c, ncpu, nb)
+                        for c in (32, 64, 128, ncpu)
+                        if min(c, ncpu, nb) >= 1})
         _CPU_G.update(data=data, offsets=offsets, sc=osc, spec=ospec)
         ctx = _mp.get_context("fork")
-        with ctx.Pool(len(ranges)) as pool:
-            pool.map(_cpu_worker, ranges)  # warm-up over own chunks
+
+        def _passes(nproc, n_timed):
+            cuts = [nb * i // nproc for i in range(nproc + 1)]
+            ranges = [(cuts[i], cuts[i + 1]) for i in range(nproc)
+                      if cuts[i + 1] > cuts[i]]
             best = None
-            for _ in range(3):
-                t0 = time.time()
-                out_w = pool.map(_cpu_worker, ranges)
-                wall = time.time() - t0
-                if best is None or wall < best[0]:
-                    best = (wall, out_w)
-        wall, out_w = best
+            with ctx.Pool(len(ranges)) as pool:
+                pool.map(_cpu_worker, ranges)  # warm-up over own chunks
+                for _ in range(n_timed):
+                    t0 = time.time()
+                    out_w = pool.map(_cpu_worker, ranges)
+                    wall = time.time() - t0
+                    if best is None or wall < best[0]:
+                        best = (wall, out_w)
+            return len(ranges), best
+
+        results = {}
+        for c in probe:
+            nw, (wall, out_w) = _passes(c, 1)
+            results[nw] = (wall, out_w)
+        best_n = min(results, key=lambda n: results[n][0])
+        nw, (wall, out_w) = _passes(best_n, 2)
+        if results[best_n][0] < wall:
+            wall, out_w = results[best_n]
         par_rows = sum(r[0] for r in out_w)
         cpu_rows_per_s = par_rows / wall
         cpu_baseline = {
             "value": cpu_rows_per_s,
             "unit": "rows/s",
-            "cores": len(ranges),
+            "cores": nw,
             "kind": "port",
-            "sample": f"full tablet ({par_rows} rows) x best-of-3 parallel "
-                      f"passes, {wall:.2f}s wall; single-thread "
-                      f"{cpu_1t_rows_per_s/1e6:.1f} Mrows/s over "
-                      f"{sample_blocks} blocks ({st:.1f}s)",
+            "sample": f"full tablet ({par_rows} rows), worker count probed"
+                      f" over {probe} on {ncpu} logical cores, best at "
+                      f"{nw} workers, best-of-3 passes, {wall:.2f}s wall; "
+                      f"single-thread {cpu_1t_rows_per_s/1e6:.1f} Mrows/s "
+                      f"over {sample_blocks} blocks ({st:.1f}s)",
             "single_thread_value": cpu_1t_rows_per_s,
         }
-        log(f"[cpu baseline] {cpu_rows_per_s/1e6:.1f} Mrows/s on "
-            f"{len(ranges)} cores; {cpu_1t_rows_per_s/1e6:.2f} Mrows/s "
-            f"single-thread")
+        log(f"[cpu baseline] {cpu_rows_per_s/1e6:.1f} Mrows/s at {nw} "
+            f"workers (probed {probe}); {cpu_1t_rows_per_s/1e6:.2f} "
+            f"Mrows/s single-thread")
 
     if rank == 0:
         out = {
