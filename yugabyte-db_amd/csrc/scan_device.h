@@ -2293,6 +2293,9 @@ struct GroupHead {
 template <int NA>
 DEV void group_accum(const DevSpec& sp, const GroupCtx& gc,
                      const RowCtxT<NA>& rc) {
+#if YBG_GABL == 1
+  return;
+#endif
   uint64_t slot;
   if (rc.grp_null) {
     slot = gc.cap;  // the NULL-key group
@@ -2309,6 +2312,9 @@ DEV void group_accum(const DevSpec& sp, const GroupCtx& gc,
       return;
     }
   }
+#if YBG_GABL == 2
+  return;
+#endif
   long long* v = gc.vals + slot * YBG_MAX_AGGS;
   unsigned long long* c = gc.cnts + slot * YBG_MAX_AGGS;
 #pragma unroll
@@ -2373,6 +2379,9 @@ DEV void group_accum_rec(const DevSpec& sp, const GroupCtx& gc,
       return;
     }
   }
+#if YBG_GABL == 2
+  return;
+#endif
   long long* v = gc.vals + slot * YBG_MAX_AGGS;
   unsigned long long* c = gc.cnts + slot * YBG_MAX_AGGS;
 #pragma unroll
@@ -2617,6 +2626,13 @@ inline bool fast_eligible(const DevSpec& d) {
 // per-row finalize accumulation.
 #ifndef YBG_ABL
 #define YBG_ABL 0
+#endif
+
+// YBG_GABL: grouped-kernel ablations (results WRONG by design, never in
+// a default build): 1 = skip the group hash table entirely (measures the
+// scan/eval side alone); 2 = probe/insert but skip the per-agg atomics.
+#ifndef YBG_GABL
+#define YBG_GABL 0
 #endif
 
 // Returns 1 = batch done (accumulators updated), 0 = abort (accumulators
