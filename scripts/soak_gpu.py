@@ -140,8 +140,56 @@ def run(cases=40, seed=20250915):
             s.close()
             assert g.rows_matched == expect, (it, hh, k0, g.rows_matched)
             n_bloom += 1
+    # SST-file soak: finish_sst with random compression (none / snappy /
+    # LZ4) + feed_sst (footer/index parse, device crc32c verify, on-GPU
+    # decompression) vs the oracle over the same builder's raw blocks
+    n_sst = 0
+    rng4 = random.Random(seed + 3)
+    sc2 = y.make_schema([y.KT_INT64],
+                        [(10, y.T_INT64, 1), (11, y.T_INT64, 1)])
+    for it in range(max(3, cases // 25)):
+        rows = rng4.randint(2000, 20000)
+        comp = rng4.choice([0, 1, 4])
+        b = y.Builder(sc2)
+        seq = 1 << 50
+        for r in range(rows):
+            seq += 1
+            b.add_packed_row(5000, [(y.T_INT64, r), (y.T_INT64, r * 7)],
+                             hash_=r // 128, key_datums=(r,), seq=seq)
+        sst_ptr, sst_total, _, _ = b.finish_sst(compression=comp)
+        data2, offsets2, nb2, total2 = b.finish()[:4]
+        for _ in range(3):
+            thr = rng4.randint(0, rows)
+            spec = y.ScanSpec()
+            spec.schema = sc2
+            spec.kv_format = y.ENC_THREE_SHARED_PARTS
+            spec.read_time = y.read_time(9000)
+            spec.num_preds = 1
+            spec.preds[0] = y.Pred(0, 0, y.PRED_GE, thr, None, 0)
+            spec.num_aggs = 2
+            spec.aggs[0] = y.Agg(y.AGG_COUNT_STAR, 0)
+            spec.aggs[1] = y.Agg(y.AGG_SUM_INT64, 1)
+            s = GpuScan(spec)
+            s.feed_sst(sst_ptr, sst_total, verify=True)
+            s.execute()
+            g = s.aggregates()
+            s.close()
+            osc2 = y.orcl_schema_from(sc2)
+            ospec = y.OrclScanSpec()
+            ospec.read_time = y.orcl_read_time(9000)
+            ospec.num_preds = 1
+            ospec.preds[0] = y.OrclPred(0, 0, y.PRED_GE, thr, None, 0)
+            ospec.num_aggs = 2
+            ospec.aggs[0] = y.OrclAgg(y.AGG_COUNT_STAR, 0)
+            ospec.aggs[1] = y.OrclAgg(y.AGG_SUM_INT64, 1)
+            o, _ = y.orcl_scan(data2, offsets2, nb2, osc2, ospec)
+            assert (g.rows_scanned, g.rows_matched, g.aggs[1].value_i64) \
+                == (o.rows_scanned, o.rows_matched,
+                    o.aggs[1].value_i64), (it, comp, thr)
+            n_sst += 1
     print(f"soak ok: {n_scan} scans, {n_intents} intent merges, "
-          f"{n_bloom} bloom point scans, seed {seed}")
+          f"{n_bloom} bloom point scans, {n_sst} sst-file scans "
+          f"(none/snappy/lz4), seed {seed}")
 
 if __name__ == "__main__":
     run(int(sys.argv[1]) if len(sys.argv) > 1 else 40,
