@@ -217,3 +217,36 @@ def test_gpu_bloom_pruned_point_scan():
     assert miss.rows_matched == 0
     assert miss.entries_seen == 0  # nothing was fed, let alone scanned
     assert miss.aggs[0].is_null == 1
+
+
+def test_filter_abi_edge_cases():
+    """ABI error paths: undersized output cap (rc=8), corrupt block
+    (rc=3), and broken/odd filter blobs never reject (fail-open, like
+    the reference reader's broken-filter handling, bloom.cc:183-188)."""
+    import ctypes as C
+    lib = y.product()
+    f = lib.ybg_filter_from_sst
+    f.restype = C.c_int
+    f.argtypes = [C.POINTER(C.c_uint8), C.POINTER(C.c_uint64), C.c_uint64,
+                  C.c_int, C.POINTER(C.c_uint8), C.c_uint64,
+                  C.POINTER(C.c_uint64)]
+    sc = _schema()
+    b = y.Builder(sc)
+    seq = 1 << 50
+    for r in range(200):
+        seq += 1
+        b.add_packed_row(5000, [(y.T_INT64, r)], hash_=r % 7,
+                         key_datums=(r,), seq=seq)
+    data, offsets, n_blocks, total = b.finish()[:4]
+    out = (C.c_uint8 * 16)()
+    ln = C.c_uint64(0)
+    assert f(data, offsets, n_blocks, 1, out, 16, C.byref(ln)) == 8
+    garbage = (C.c_uint8 * 64).from_buffer_copy(b"\x00" * 64)
+    goff = (C.c_uint64 * 2)(0, 64)
+    big = (C.c_uint8 * 20000)()
+    assert f(C.cast(garbage, C.POINTER(C.c_uint8)), goff, 1, 1, big,
+             20000, C.byref(ln)) == 3
+    key = _enc_key(sc, 3, 3)
+    # truncated / non-multiple-of-slice filters fail OPEN (may match)
+    assert y.filter_may_match(b"\x01\x02\x03", key) == 1
+    assert y.filter_may_match(b"\x00" * 100, key) == 1
